@@ -1,0 +1,115 @@
+"""Fused MoE layer (reference: layers/moe/fused_moe_triton/layer.py).
+
+Round-1 compute path: expert-loop gather/scatter over hipBLASLt GEMMs —
+correct on CPU and GPU, TP-sharded intermediate dim. The CDNA4 grouped
+MFMA GEMM kernel (sorted token ids, block-aligned, tuned per (E, N))
+replaces the loop in the optimization pass. EP: contiguous expert shards
+with a trailing EP all-reduce (reference layer.py:686-735 — the
+reference's EP comm is allgather+allreduce, not all-to-all).
+"""
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from gllm_amd import ops
+from gllm_amd.parallel import (get_ep_rank, get_ep_size, get_tp_rank,
+                               get_tp_size, tensor_parallel_all_reduce)
+
+
+class FusedMoE(nn.Module):
+    def __init__(self, num_experts: int, top_k: int, hidden_size: int,
+                 intermediate_size: int, renormalize: bool = True,
+                 use_ep: bool = False, params_dtype=None):
+        super().__init__()
+        dtype = params_dtype or torch.get_default_dtype()
+        self.num_experts = num_experts
+        self.top_k = top_k
+        self.hidden_size = hidden_size
+        self.renormalize = renormalize
+        self.use_ep = use_ep
+
+        if use_ep:
+            ep_rank, ep_size = get_ep_rank(), get_ep_size()
+            base, rem = divmod(num_experts, ep_size)
+            counts = [base + (1 if r < rem else 0) for r in range(ep_size)]
+            self.expert_start = sum(counts[:ep_rank])
+            self.num_local_experts = counts[ep_rank]
+            self.intermediate_per_rank = intermediate_size
+        else:
+            self.expert_start = 0
+            self.num_local_experts = num_experts
+            tp = get_tp_size()
+            assert intermediate_size % tp == 0
+            self.intermediate_per_rank = intermediate_size // tp
+
+        I = self.intermediate_per_rank
+        self.w13_weight = nn.Parameter(
+            torch.empty(self.num_local_experts, 2 * I, hidden_size,
+                        dtype=dtype), requires_grad=False)
+        self.w2_weight = nn.Parameter(
+            torch.empty(self.num_local_experts, hidden_size, I, dtype=dtype),
+            requires_grad=False)
+        self.w13_weight.weight_loader = self._load_w13
+        self.w2_weight.weight_loader = self._load_w2
+
+    # ---- loading: per-expert pulls with EP ownership / TP sharding ----
+    def _local_expert(self, expert_id: int) -> Optional[int]:
+        lid = expert_id - self.expert_start
+        if 0 <= lid < self.num_local_experts:
+            return lid
+        return None
+
+    def _load_w13(self, param, loaded, expert_id: int, shard_id: int):
+        """shard_id: 0 = gate (w1), 1 = up (w3)."""
+        lid = self._local_expert(expert_id)
+        if lid is None:
+            return
+        I = self.intermediate_per_rank
+        if self.use_ep:
+            shard = loaded
+        else:
+            shard = loaded.narrow(0, get_tp_rank() * I, I)
+        param.data[lid].narrow(0, shard_id * I, I).copy_(shard)
+
+    def _load_w2(self, param, loaded, expert_id: int):
+        lid = self._local_expert(expert_id)
+        if lid is None:
+            return
+        I = self.intermediate_per_rank
+        if self.use_ep:
+            shard = loaded
+        else:
+            shard = loaded.narrow(1, get_tp_rank() * I, I)
+        param.data[lid].copy_(shard)
+
+    # ---- forward ----
+    def forward(self, x: torch.Tensor,
+                router_logits: torch.Tensor) -> torch.Tensor:
+        T = x.shape[0]
+        weights, ids = ops.topk_softmax(router_logits, self.top_k,
+                                        self.renormalize)  # [T,K], [T,K]
+        weights = weights.to(x.dtype)
+        out = torch.zeros_like(x)
+        flat_ids = ids.long().flatten()                    # [T*K]
+        flat_rows = torch.arange(T, device=x.device).repeat_interleave(
+            self.top_k)
+        for lid in range(self.num_local_experts):
+            eid = self.expert_start + lid
+            sel = flat_ids == eid
+            if not bool(sel.any()):
+                continue
+            rows = flat_rows[sel]
+            xe = x.index_select(0, rows)
+            h = ops.silu_and_mul(F.linear(xe, self.w13_weight[lid]))
+            ye = F.linear(h, self.w2_weight[lid])
+            w = weights.flatten()[sel].unsqueeze(-1)
+            out.index_add_(0, rows, ye * w)
+        if self.use_ep:
+            from gllm_amd.parallel.state import ep_all_reduce
+            out = ep_all_reduce(out)
+        else:
+            out = tensor_parallel_all_reduce(out)
+        return out

@@ -1,0 +1,142 @@
+"""Token sampler: greedy fast path, temperature, top-k/top-p, repetition
+penalty, logprobs (reference: layers/sampler.py + repetition_penalty.py).
+
+Round-1 note: top-k/top-p runs as a masked-sort composite on GPU; the
+fused single-pass HIP sampling kernel replaces it in a later pass.
+"""
+
+import dataclasses
+from typing import List, Optional
+
+import torch
+
+from gllm_amd.ops import torch_ref
+
+
+@dataclasses.dataclass
+class SamplingMetadata:
+    temperatures: torch.Tensor          # [B] float32 (0 => greedy row)
+    top_ps: torch.Tensor                # [B] float32
+    top_ks: torch.Tensor                # [B] int32 (-1 disabled)
+    penalties: torch.Tensor             # [B] float32
+    all_greedy: bool
+    any_penalty: bool
+    # token history per row for repetition penalty (cpu LongTensors)
+    token_id_rows: Optional[List[torch.Tensor]] = None
+    max_logprobs: int = 0               # >0 => return top-k logprobs
+    generators: Optional[List[Optional[torch.Generator]]] = None
+
+
+@dataclasses.dataclass
+class SamplerOutput:
+    next_tokens: torch.Tensor           # [B] int64 (device)
+    logprobs: Optional[torch.Tensor] = None        # [B] chosen-token logprob
+    topk_logprobs: Optional[torch.Tensor] = None   # [B, K]
+    topk_token_ids: Optional[torch.Tensor] = None  # [B, K]
+
+
+class Sampler(torch.nn.Module):
+    def forward(self, logits: torch.Tensor,
+                meta: SamplingMetadata) -> SamplerOutput:
+        # logits: [B, V] (already gathered to full vocab)
+        if meta.any_penalty and meta.token_id_rows is not None:
+            logits = torch_ref.apply_repetition_penalty(
+                logits.float(), meta.token_id_rows, meta.penalties)
+        if meta.all_greedy:
+            next_tokens = logits.argmax(dim=-1)
+            return self._with_logprobs(logits, next_tokens, meta)
+
+        logits = logits.float()
+        temps = meta.temperatures.clamp_min(1e-5).unsqueeze(-1)
+        scaled = logits / temps
+        probs = torch.softmax(scaled, dim=-1)
+        probs = self._apply_top_k_top_p(probs, meta.top_ks, meta.top_ps)
+        sampled = self._multinomial(probs, meta)
+        # greedy rows override
+        greedy_rows = meta.temperatures == 0.0
+        if greedy_rows.any():
+            sampled = torch.where(greedy_rows, logits.argmax(dim=-1), sampled)
+        return self._with_logprobs(logits, sampled, meta)
+
+    @staticmethod
+    def _apply_top_k_top_p(probs: torch.Tensor, top_ks: torch.Tensor,
+                           top_ps: torch.Tensor) -> torch.Tensor:
+        B, V = probs.shape
+        need_k = bool((top_ks > 0).any())
+        need_p = bool((top_ps < 1.0).any())
+        if not need_k and not need_p:
+            return probs
+        sorted_probs, idx = probs.sort(dim=-1, descending=True)
+        if need_k:
+            ks = torch.where(top_ks > 0, top_ks, torch.full_like(top_ks, V))
+            rank = torch.arange(V, device=probs.device).unsqueeze(0)
+            sorted_probs = sorted_probs.masked_fill(
+                rank >= ks.unsqueeze(-1), 0.0)
+        if need_p:
+            cum = sorted_probs.cumsum(dim=-1)
+            # keep tokens whose cumulative mass (exclusive) < top_p
+            exclusive = cum - sorted_probs
+            sorted_probs = sorted_probs.masked_fill(
+                exclusive > top_ps.unsqueeze(-1), 0.0)
+        out = torch.zeros_like(probs)
+        out.scatter_(-1, idx, sorted_probs)
+        return out / out.sum(-1, keepdim=True).clamp_min(1e-20)
+
+    @staticmethod
+    def _multinomial(probs: torch.Tensor, meta: SamplingMetadata):
+        gens = meta.generators
+        if gens and any(g is not None for g in gens):
+            outs = []
+            for i in range(probs.shape[0]):
+                outs.append(torch.multinomial(
+                    probs[i:i + 1], 1, generator=gens[i]).squeeze(-1))
+            return torch.cat(outs)
+        return torch.multinomial(probs, 1).squeeze(-1)
+
+    @staticmethod
+    def _with_logprobs(logits: torch.Tensor, next_tokens: torch.Tensor,
+                       meta: SamplingMetadata) -> SamplerOutput:
+        if meta.max_logprobs <= 0:
+            return SamplerOutput(next_tokens)
+        logp = torch.log_softmax(logits.float(), dim=-1)
+        chosen = logp.gather(-1, next_tokens.unsqueeze(-1)).squeeze(-1)
+        k = meta.max_logprobs
+        topv, topi = logp.topk(k, dim=-1)
+        return SamplerOutput(next_tokens, chosen, topv, topi)
+
+
+def build_sampling_metadata(items, device) -> SamplingMetadata:
+    """Build metadata from the scheduled batch items (one row per item)."""
+    temps, tps, tks, pens, rows, gens = [], [], [], [], [], []
+    max_lp = 0
+    any_pen = False
+    for it in items:
+        sp = it.seq.sampling
+        temps.append(sp.temperature)
+        tps.append(sp.top_p)
+        tks.append(sp.top_k)
+        pens.append(sp.repetition_penalty)
+        if sp.repetition_penalty != 1.0:
+            any_pen = True
+            rows.append(torch.tensor(it.seq.token_ids, dtype=torch.long,
+                                     device=device))
+        else:
+            rows.append(torch.empty(0, dtype=torch.long, device=device))
+        if sp.logprobs:
+            max_lp = max(max_lp, sp.logprobs)
+        if sp.seed is not None:
+            g = torch.Generator(device=device)
+            g.manual_seed(sp.seed + it.seq.num_output_tokens)
+            gens.append(g)
+        else:
+            gens.append(None)
+    all_greedy = all(t == 0.0 for t in temps)
+    return SamplingMetadata(
+        temperatures=torch.tensor(temps, dtype=torch.float32, device=device),
+        top_ps=torch.tensor(tps, dtype=torch.float32, device=device),
+        top_ks=torch.tensor(tks, dtype=torch.int32, device=device),
+        penalties=torch.tensor(pens, dtype=torch.float32, device=device),
+        all_greedy=all_greedy, any_penalty=any_pen,
+        token_id_rows=rows if any_pen else None,
+        max_logprobs=max_lp,
+        generators=gens if any(g is not None for g in gens) else None)

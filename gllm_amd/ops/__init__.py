@@ -1,0 +1,127 @@
+"""Custom-op facade — the single switch point between backends.
+
+Mirrors the role of the reference's ``_custom_ops.py`` ("single point
+where we can swap backends", _custom_ops.py:1-10): every layer calls
+through here. Dispatch rule:
+
+* CUDA (= HIP/ROCm) tensors -> the in-tree gfx950 extension
+  ``gllm_amd._kernels``. If the extension is missing on a GPU box this
+  RAISES — no silent eager fallback (a GPU run must exercise the native
+  kernels).
+* CPU tensors -> the PyTorch reference implementations (ops.torch_ref),
+  which double as the numerics oracle for the HIP kernels.
+"""
+
+from typing import Optional
+
+import torch
+
+from gllm_amd.ops import torch_ref
+
+_K = None
+_K_ERR: Optional[str] = None
+
+
+def _load_kernels():
+    global _K, _K_ERR
+    if _K is not None or _K_ERR is not None:
+        return _K
+    try:
+        from gllm_amd import _kernels  # built in-tree by setup.py
+        _K = _kernels
+    except ImportError as e:  # pragma: no cover
+        _K_ERR = str(e)
+    return _K
+
+
+def _gpu_kernels():
+    k = _load_kernels()
+    if k is None:
+        raise RuntimeError(
+            "gllm_amd._kernels HIP extension not built but a CUDA tensor "
+            "was passed — build with `python setup.py build_ext --inplace` "
+            f"(PYTORCH_ROCM_ARCH=gfx950). Import error: {_K_ERR}")
+    return k
+
+
+def has_kernels() -> bool:
+    return _load_kernels() is not None
+
+
+# --------------------------------------------------------------- norm
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _gpu_kernels().rmsnorm(out, x, weight, eps)
+        return out
+    return torch_ref.rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(x: torch.Tensor, residual: torch.Tensor,
+                      weight: torch.Tensor, eps: float):
+    if x.is_cuda:
+        _gpu_kernels().fused_add_rmsnorm(x, residual, weight, eps)
+        return x, residual
+    return torch_ref.fused_add_rmsnorm(x, residual, weight, eps)
+
+
+# --------------------------------------------------------------- activation
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        d = x.shape[-1] // 2
+        out = torch.empty(x.shape[:-1] + (d,), dtype=x.dtype, device=x.device)
+        _gpu_kernels().silu_and_mul(out, x)
+        return out
+    return torch_ref.silu_and_mul(x)
+
+
+# --------------------------------------------------------------- rope
+def rotary_embedding(positions: torch.Tensor, q: torch.Tensor,
+                     k: torch.Tensor, head_dim: int,
+                     cos_sin_cache: torch.Tensor, is_neox: bool = True):
+    if q.is_cuda:
+        _gpu_kernels().rotary_embedding(positions, q, k, head_dim,
+                                        cos_sin_cache, is_neox)
+        return q, k
+    return torch_ref.rotary_embedding(positions, q, k, head_dim,
+                                      cos_sin_cache, is_neox)
+
+
+# --------------------------------------------------------------- kv cache
+def reshape_and_cache(k: torch.Tensor, v: torch.Tensor,
+                      k_cache: torch.Tensor, v_cache: torch.Tensor,
+                      slot_mapping: torch.Tensor):
+    if k.is_cuda:
+        _gpu_kernels().reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        return
+    torch_ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+# --------------------------------------------------------------- attention
+def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
+                    v_cache: torch.Tensor, block_table: torch.Tensor,
+                    seq_lens: torch.Tensor, query_start_loc: torch.Tensor,
+                    scale: float, max_query_len: int = 1,
+                    out: Optional[torch.Tensor] = None,
+                    sliding_window: int = 0) -> torch.Tensor:
+    if q.is_cuda:
+        kern = _gpu_kernels()
+        if out is None:
+            out = torch.empty_like(q)
+        if max_query_len == 1:
+            kern.paged_attention_decode(
+                out, q, k_cache, v_cache, block_table, seq_lens, scale,
+                sliding_window)
+        else:
+            kern.paged_attention_prefill(
+                out, q, k_cache, v_cache, block_table, seq_lens,
+                query_start_loc, scale, sliding_window)
+        return out
+    return torch_ref.paged_attention(q, k_cache, v_cache, block_table,
+                                     seq_lens, query_start_loc, scale,
+                                     out=out, sliding_window=sliding_window)
+
+
+# --------------------------------------------------------------- moe
+def topk_softmax(gating: torch.Tensor, topk: int, renormalize: bool = True):
+    return torch_ref.topk_softmax(gating, topk, renormalize)

@@ -1,0 +1,153 @@
+"""Pure-PyTorch reference implementations of every custom op.
+
+These are (a) the CPU execution path (config 1: plumbing without a GPU)
+and (b) the numerics oracle the HIP kernels are unit-tested against
+(reference keeps `forward_native` impls for this purpose,
+layers/layernorm.py:88-127). All math in fp32 for a stable oracle.
+"""
+
+import math
+from typing import Optional
+
+import torch
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    dtype = x.dtype
+    xf = x.float()
+    var = xf.pow(2).mean(-1, keepdim=True)
+    out = xf * torch.rsqrt(var + eps)
+    return (out * weight.float()).to(dtype)
+
+
+def fused_add_rmsnorm(x: torch.Tensor, residual: torch.Tensor,
+                      weight: torch.Tensor, eps: float):
+    """residual += x; x = rmsnorm(residual). In-place on both args."""
+    residual.add_(x)
+    x.copy_(rmsnorm(residual, weight, eps))
+    return x, residual
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    d = x.shape[-1] // 2
+    return (torch.nn.functional.silu(x[..., :d].float())
+            * x[..., d:].float()).to(x.dtype)
+
+
+def rotary_embedding(positions: torch.Tensor, q: torch.Tensor,
+                     k: torch.Tensor, head_dim: int,
+                     cos_sin_cache: torch.Tensor, is_neox: bool = True):
+    """In-place RoPE on q [T, Hq*D] and k [T, Hk*D].
+
+    cos_sin_cache: [max_pos, rot_dim] laid out [cos(rot/2) | sin(rot/2)].
+    """
+    rot_dim = cos_sin_cache.shape[-1]
+    half = rot_dim // 2
+    cs = cos_sin_cache[positions].float()       # [T, rot_dim]
+    cos = cs[:, :half]                          # [T, half]
+    sin = cs[:, half:]
+    for t in (q, k):
+        T = t.shape[0]
+        x = t.view(T, -1, head_dim)
+        rot = x[..., :rot_dim].float()
+        if is_neox:
+            x1, x2 = rot[..., :half], rot[..., half:]
+            c = cos.unsqueeze(1)
+            s = sin.unsqueeze(1)
+            o1 = x1 * c - x2 * s
+            o2 = x2 * c + x1 * s
+            out = torch.cat([o1, o2], dim=-1)
+        else:
+            x1, x2 = rot[..., 0::2], rot[..., 1::2]
+            c = cos.unsqueeze(1)
+            s = sin.unsqueeze(1)
+            o1 = x1 * c - x2 * s
+            o2 = x2 * c + x1 * s
+            out = torch.stack([o1, o2], dim=-1).flatten(-2)
+        x[..., :rot_dim].copy_(out.to(t.dtype))
+    return q, k
+
+
+def reshape_and_cache(k: torch.Tensor, v: torch.Tensor,
+                      k_cache: torch.Tensor, v_cache: torch.Tensor,
+                      slot_mapping: torch.Tensor):
+    """Scatter per-token K/V [T, Hkv, D] into paged caches
+    [num_pages, page_size, Hkv, D] at flat slot indices."""
+    page_size = k_cache.shape[1]
+    pages = torch.div(slot_mapping, page_size, rounding_mode="floor")
+    offs = slot_mapping % page_size
+    k_cache[pages, offs] = k.to(k_cache.dtype)
+    v_cache[pages, offs] = v.to(v_cache.dtype)
+
+
+def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
+                    v_cache: torch.Tensor, block_table: torch.Tensor,
+                    seq_lens: torch.Tensor, query_start_loc: torch.Tensor,
+                    scale: float, out: Optional[torch.Tensor] = None,
+                    sliding_window: int = 0) -> torch.Tensor:
+    """Varlen causal attention over the paged KV cache.
+
+    q:            [T, Hq, D] — all new tokens of the batch, ragged by seq
+    k/v_cache:    [num_pages, page_size, Hkv, D]
+    block_table:  [B, max_pages] int — page ids per seq
+    seq_lens:     [B] — TOTAL context length per seq (past + new)
+    query_start_loc: [B+1] — ragged boundaries into q
+    Causal offset: query token i of seq b attends to cache positions
+    [0, seq_len - q_len + i].
+    """
+    T, Hq, D = q.shape
+    Hkv = k_cache.shape[2]
+    page_size = k_cache.shape[1]
+    group = Hq // Hkv
+    if out is None:
+        out = torch.empty_like(q)
+    B = seq_lens.shape[0]
+    for b in range(B):
+        qs, qe = int(query_start_loc[b]), int(query_start_loc[b + 1])
+        q_len = qe - qs
+        if q_len == 0:
+            continue
+        s_len = int(seq_lens[b])
+        n_pages = -(-s_len // page_size)
+        pages = block_table[b, :n_pages].long()
+        k = k_cache[pages].reshape(-1, Hkv, D)[:s_len].float()  # [S, Hkv, D]
+        v = v_cache[pages].reshape(-1, Hkv, D)[:s_len].float()
+        qq = q[qs:qe].float()                                   # [L, Hq, D]
+        # [Hq, L, S]
+        scores = torch.einsum("lhd,shd->hls", qq,
+                              k.repeat_interleave(group, dim=1)) * scale
+        # causal mask with past offset
+        past = s_len - q_len
+        pos_q = torch.arange(q_len, device=q.device).unsqueeze(1) + past
+        pos_k = torch.arange(s_len, device=q.device).unsqueeze(0)
+        mask = pos_k <= pos_q                                   # [L, S]
+        if sliding_window > 0:
+            mask &= pos_k > (pos_q - sliding_window)
+        scores.masked_fill_(~mask.unsqueeze(0), float("-inf"))
+        p = torch.softmax(scores, dim=-1)
+        o = torch.einsum("hls,shd->lhd", p,
+                         v.repeat_interleave(group, dim=1))
+        out[qs:qe] = o.to(out.dtype)
+    return out
+
+
+def topk_softmax(gating: torch.Tensor, topk: int, renormalize: bool = True):
+    """MoE routing: softmax then top-k. Returns (weights [T,K], ids [T,K])."""
+    probs = torch.softmax(gating.float(), dim=-1)
+    weights, ids = torch.topk(probs, topk, dim=-1)
+    if renormalize:
+        weights = weights / weights.sum(-1, keepdim=True)
+    return weights, ids.to(torch.int32)
+
+
+def apply_repetition_penalty(logits: torch.Tensor, token_ids_per_row,
+                             penalties: torch.Tensor) -> torch.Tensor:
+    """logits [B, V]; token_ids_per_row: list of LongTensors; penalties [B]."""
+    for i, toks in enumerate(token_ids_per_row):
+        p = float(penalties[i])
+        if p == 1.0 or toks.numel() == 0:
+            continue
+        row = logits[i]
+        vals = row[toks]
+        row[toks] = torch.where(vals > 0, vals / p, vals * p)
+    return logits

@@ -1,0 +1,36 @@
+"""Per-forward batch metadata handed through the model to every layer.
+
+Equivalent in role to the reference's InputData (gllm/input_data.py) GPU
+tensors, but passed explicitly down ``model.forward`` instead of being
+read from a global.
+"""
+
+import dataclasses
+from typing import List, Optional
+
+import torch
+
+
+@dataclasses.dataclass
+class ForwardContext:
+    # --- ragged batch geometry ---
+    num_tokens: int
+    positions: torch.Tensor          # [T] int64
+    slot_mapping: torch.Tensor       # [T] int64 (flat KV slot per new token)
+    block_table: torch.Tensor        # [B, max_pages] int32
+    seq_lens: torch.Tensor           # [B] int32 (total context incl. chunk)
+    query_start_loc: torch.Tensor    # [B+1] int32
+    max_query_len: int               # 1 => pure decode
+    max_seq_len: int
+    # --- KV cache (this stage's layers) ---
+    k_caches: List[torch.Tensor]     # per local layer [pages, page, Hkv, D]
+    v_caches: List[torch.Tensor]
+    # --- sampling metadata (set on the last PP stage) ---
+    # rows of the hidden states whose logits we need: query_start_loc[1:]-1
+    logits_indices: Optional[torch.Tensor] = None
+    # profile/warmup run: attention may be skipped (no KV yet)
+    is_profile_run: bool = False
+
+    @property
+    def is_pure_decode(self) -> bool:
+        return self.max_query_len == 1

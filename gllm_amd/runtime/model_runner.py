@@ -1,0 +1,171 @@
+"""ModelRunner: owns the model, KV cache, forward orchestration, sampling.
+
+Parity: reference model_runner.py init/profile/KV-sizing/step_once
+(:491-552, :1482-1523, :1809-1875). hipGraph decode capture lands in
+runtime/graph_runner.py and is driven from here.
+"""
+
+import math
+from typing import List, Optional
+
+import torch
+
+from gllm_amd.config import EngineConfig
+from gllm_amd.core.kv_cache import (KVCacheSpec, MemoryManager,
+                                    PrefixMemoryManager)
+from gllm_amd.core.scheduler import ScheduledBatch
+from gllm_amd.layers.sampler import Sampler, build_sampling_metadata
+from gllm_amd.logger import logger
+from gllm_amd.models.loader import load_hf_config, load_model
+from gllm_amd.runtime.batch_builder import BatchBuilder
+from gllm_amd.runtime.forward_context import ForwardContext
+
+
+class ModelRunner:
+    def __init__(self, config: EngineConfig):
+        self.config = config
+        self.device = config.device
+        self.model = None
+        self.hf_config = None
+        self.k_caches: List[torch.Tensor] = []
+        self.v_caches: List[torch.Tensor] = []
+        self.memory_manager: Optional[MemoryManager] = None
+        self.sampler = Sampler()
+        self.builder: Optional[BatchBuilder] = None
+        self.graph_runner = None
+        self.kv_dtype = None
+
+    # ------------------------------------------------------------------
+    def init(self, num_pages_override: Optional[int] = None):
+        cfg = self.config
+        if cfg.device.startswith("cuda"):
+            torch.cuda.set_device(cfg.device if ":" in cfg.device
+                                  else "cuda:0")
+        self.model, self.hf_config = load_model(cfg, cfg.device)
+        self.kv_dtype = cfg.torch_dtype()
+        num_pages = num_pages_override or self._size_kv_cache()
+        self._allocate_kv(num_pages)
+        mgr_cls = PrefixMemoryManager if cfg.enable_prefix_caching \
+            else MemoryManager
+        self.memory_manager = mgr_cls(num_pages, cfg.page_size)
+        self.builder = BatchBuilder(cfg.page_size, cfg.device)
+        if cfg.use_graph and cfg.device.startswith("cuda"):
+            from gllm_amd.runtime.graph_runner import GraphRunner
+            self.graph_runner = GraphRunner(self)
+            self.graph_runner.capture_all()
+        return self
+
+    # ------------------------------------------------------------------
+    def kv_spec(self) -> KVCacheSpec:
+        hf = self.hf_config
+        num_local_layers = self.model.num_local_layers
+        total_kv = getattr(hf, "num_key_value_heads",
+                           hf.num_attention_heads)
+        from gllm_amd.parallel import get_tp_size
+        tp = get_tp_size()
+        kv_per_rank = max(1, total_kv // tp)
+        head_dim = getattr(hf, "head_dim", None) or \
+            hf.hidden_size // hf.num_attention_heads
+        return KVCacheSpec(num_local_layers, kv_per_rank, head_dim,
+                           self.config.page_size,
+                           dtype_bytes=self.kv_dtype.itemsize)
+
+    def _size_kv_cache(self) -> int:
+        cfg = self.config
+        spec = self.kv_spec()
+        if not cfg.device.startswith("cuda"):
+            # CPU test path: small fixed pool
+            return 512
+        self._profile_run()
+        free, total = torch.cuda.mem_get_info()
+        usable = total * cfg.gpu_memory_util - (total - free)
+        num_pages = max(64, int(usable // spec.bytes_per_page))
+        # min across ranks so every rank sizes identically
+        from gllm_amd.parallel import get_world_size
+        if get_world_size() > 1:
+            import torch.distributed as dist
+            t = torch.tensor([num_pages], dtype=torch.int64,
+                             device=self.device)
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            num_pages = int(t.item())
+        logger.info("KV cache: %d pages x %d tokens (%.1f GiB, %d layers)",
+                    num_pages, cfg.page_size,
+                    num_pages * spec.bytes_per_page / 2**30, spec.num_layers)
+        return num_pages
+
+    def _profile_run(self):
+        """Peak-activation dummy forward to reserve torch workspace before
+        sizing the KV cache (reference model_runner.py:1482-1523)."""
+        cfg = self.config
+        T = min(cfg.maxp, cfg.profile_batch)
+        input_ids = torch.zeros(T, dtype=torch.long, device=self.device)
+        positions = torch.arange(T, dtype=torch.long, device=self.device)
+        fctx = ForwardContext(
+            num_tokens=T, positions=positions,
+            slot_mapping=torch.zeros(T, dtype=torch.long,
+                                     device=self.device),
+            block_table=torch.zeros((1, 1), dtype=torch.int32,
+                                    device=self.device),
+            seq_lens=torch.full((1,), T, dtype=torch.int32,
+                                device=self.device),
+            query_start_loc=torch.tensor([0, T], dtype=torch.int32,
+                                         device=self.device),
+            max_query_len=T, max_seq_len=T, k_caches=[], v_caches=[],
+            is_profile_run=True)
+        fctx.logits_indices = torch.tensor([T - 1], device=self.device)
+        with torch.no_grad():
+            hidden, residual = self._stage_forward(input_ids, positions, fctx)
+            if self.model.is_last_stage:
+                self.model.compute_logits(hidden, fctx)
+        torch.cuda.synchronize()
+
+    def _allocate_kv(self, num_pages: int):
+        spec = self.kv_spec()
+        shape = (num_pages, spec.page_size, spec.num_kv_heads, spec.head_dim)
+        self.k_caches = [torch.zeros(shape, dtype=self.kv_dtype,
+                                     device=self.device)
+                         for _ in range(spec.num_layers)]
+        self.v_caches = [torch.zeros(shape, dtype=self.kv_dtype,
+                                     device=self.device)
+                         for _ in range(spec.num_layers)]
+
+    # ------------------------------------------------------------------
+    def _stage_forward(self, input_ids, positions, fctx,
+                       hidden_states=None, residual=None):
+        return self.model(input_ids, positions, fctx,
+                          hidden_states=hidden_states, residual=residual)
+
+    @torch.no_grad()
+    def step_first_stage(self, batch: ScheduledBatch):
+        """Stage-0 forward. PP=1: returns sampled tokens. PP>1: returns
+        (hidden, residual, fctx) for the PP send."""
+        tokens, fctx = self.builder.build(
+            batch, self.k_caches, self.v_caches,
+            need_logits=self.model.is_last_stage)
+        if (self.graph_runner is not None and self.model.is_last_stage
+                and fctx.max_query_len == 1
+                and self.graph_runner.can_replay(len(batch.items))):
+            return self.graph_runner.replay(batch, tokens, fctx)
+        hidden, residual = self._stage_forward(tokens, fctx.positions, fctx)
+        if self.model.is_last_stage:
+            return self._sample(batch, hidden, fctx)
+        return hidden, residual, fctx
+
+    @torch.no_grad()
+    def step_mid_stage(self, batch: ScheduledBatch, hidden, residual):
+        """PP stage > 0: forward received hidden states."""
+        _, fctx = self.builder.build(
+            batch, self.k_caches, self.v_caches,
+            need_logits=self.model.is_last_stage)
+        hidden, residual = self._stage_forward(
+            None, fctx.positions, fctx, hidden_states=hidden,
+            residual=residual)
+        if self.model.is_last_stage:
+            return self._sample(batch, hidden, fctx)
+        return hidden, residual, fctx
+
+    def _sample(self, batch: ScheduledBatch, hidden, fctx):
+        logits = self.model.compute_logits(hidden, fctx)
+        meta = build_sampling_metadata(batch.items, logits.device)
+        out = self.sampler(logits, meta)
+        return out

@@ -1,0 +1,144 @@
+"""Per-request state: tokens, KV page table, chunked-prefill cursors.
+
+Capability parity with the reference Sequence (gllm/sequence.py): token
+accumulation, chunked-prefill cursors (computed_token_num /
+to_compute_token_num), preempt/resume, incremental detokenization,
+per-page prefix-hash cache. Re-designed as a plain dataclass-style object
+with explicit SamplingParams.
+"""
+
+import dataclasses
+from typing import List, Optional
+
+
+@dataclasses.dataclass
+class SamplingParams:
+    temperature: float = 1.0
+    top_p: float = 1.0
+    top_k: int = -1                    # -1 = disabled
+    repetition_penalty: float = 1.0
+    max_tokens: int = 512
+    min_tokens: int = 0
+    ignore_eos: bool = False
+    stop: Optional[List[str]] = None
+    stop_token_ids: Optional[List[int]] = None
+    logprobs: Optional[int] = None     # top-k logprobs to return per token
+    prompt_logprobs: Optional[int] = None
+    seed: Optional[int] = None
+
+    @property
+    def is_greedy(self) -> bool:
+        return self.temperature == 0.0
+
+
+class Sequence:
+    FINISH_LENGTH = "length"
+    FINISH_STOP = "stop"
+    FINISH_ABORT = "abort"
+
+    def __init__(self, seq_id: int, prompt_token_ids: List[int],
+                 sampling: Optional[SamplingParams] = None,
+                 eos_token_id: Optional[int] = None,
+                 arrival_time: float = 0.0):
+        self.seq_id = seq_id
+        self.token_ids: List[int] = list(prompt_token_ids)
+        self.prompt_len = len(prompt_token_ids)
+        # raw_prompt_len differs from prompt_len once multimodal sentinel
+        # expansion exists; identical for text.
+        self.raw_prompt_len = self.prompt_len
+        self.sampling = sampling or SamplingParams()
+        self.eos_token_id = eos_token_id
+        self.arrival_time = arrival_time
+
+        # --- paged KV state ---
+        self.page_table: List[int] = []
+        # tokens whose KV is already in cache (advances by chunk)
+        self.computed_token_num: int = 0
+        # tokens scheduled to be computed this tick (set by the scheduler)
+        self.to_compute_token_num: int = 0
+        # pages at the head that came from the prefix cache
+        self.num_cached_pages: int = 0
+        # per-page chained hashes (prefix cache keys), computed lazily
+        self.page_hashes: List[int] = []
+
+        # --- output state ---
+        self.finish_reason: Optional[str] = None
+        # incremental detokenize cursor (index into token_ids)
+        self.detok_offset: int = self.prompt_len
+        self.detok_text: str = ""
+        # first-token timestamp for TTFT accounting
+        self.first_token_time: Optional[float] = None
+
+        # repetition-penalty mask slot / SSM slot (set by managers when used)
+        self.penalty_slot: int = -1
+        self.ssm_slot: int = -1
+
+    # ---- basic accounting ----
+    def __len__(self) -> int:
+        return len(self.token_ids)
+
+    @property
+    def output_len(self) -> int:
+        return self.sampling.max_tokens
+
+    @property
+    def num_output_tokens(self) -> int:
+        return len(self.token_ids) - self.prompt_len
+
+    @property
+    def computed_prompt(self) -> bool:
+        """True once the whole prompt's KV is computed (decode phase)."""
+        return self.computed_token_num >= self.prompt_len
+
+    @property
+    def last_token(self) -> int:
+        return self.token_ids[-1]
+
+    @property
+    def is_finished(self) -> bool:
+        return self.finish_reason is not None
+
+    # ---- decoding-loop transitions ----
+    def append_token(self, token_id: int) -> None:
+        self.token_ids.append(token_id)
+
+    def advance_computed(self) -> None:
+        """Commit this tick's chunk after the forward pass."""
+        self.computed_token_num += self.to_compute_token_num
+        self.to_compute_token_num = 0
+
+    def check_finish(self) -> None:
+        if self.finish_reason is not None:
+            return
+        n_out = self.num_output_tokens
+        if n_out >= self.sampling.max_tokens:
+            self.finish_reason = self.FINISH_LENGTH
+            return
+        if n_out < max(1, self.sampling.min_tokens):
+            return
+        last = self.token_ids[-1]
+        if (not self.sampling.ignore_eos and self.eos_token_id is not None
+                and last == self.eos_token_id):
+            self.finish_reason = self.FINISH_STOP
+            return
+        if self.sampling.stop_token_ids and last in self.sampling.stop_token_ids:
+            self.finish_reason = self.FINISH_STOP
+
+    def preempt(self) -> None:
+        """Drop all computed KV; the seq will re-prefill from scratch
+        (reference: sequence.py:156-169 — recompute, not swap)."""
+        self.computed_token_num = 0
+        self.to_compute_token_num = 0
+        self.num_cached_pages = 0
+        self.page_table = []
+        self.page_hashes = []
+
+    # ---- output tokens view ----
+    @property
+    def output_token_ids(self) -> List[int]:
+        return self.token_ids[self.prompt_len:]
+
+    def __repr__(self):
+        return (f"Sequence(id={self.seq_id}, len={len(self.token_ids)}, "
+                f"computed={self.computed_token_num}, "
+                f"pages={len(self.page_table)}, finish={self.finish_reason})")
