@@ -1,0 +1,178 @@
+"""Flagship serving benchmark (driver contract).
+
+Measures the BASELINE.json metric: output tokens/sec (whole node) for
+Qwen2.5-32B at PP = N on MI355X, synthetic data, random-init (dummy)
+weights, bf16. A "step" is one pipeline tick of the serving engine
+(schedule -> forward -> sample -> commit) with a saturated continuous
+decode batch; `value` is sampled output tokens/sec for the WHOLE model
+(one model spanning all N GPUs via PP), so scaling is "strong".
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W]
+N>1 is launched by the driver via torch.distributed.run (one rank per
+GPU over RCCL); ranks read RANK/WORLD_SIZE from the env.
+"""
+
+import argparse
+import json
+import os
+import sys
+import tempfile
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+QWEN25_32B = {
+    "architectures": ["Qwen2ForCausalLM"],
+    "model_type": "qwen2",
+    "hidden_size": 5120,
+    "intermediate_size": 27648,
+    "num_hidden_layers": 64,
+    "num_attention_heads": 40,
+    "num_key_value_heads": 8,
+    "vocab_size": 152064,
+    "max_position_embeddings": 32768,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 1000000.0,
+    "tie_word_embeddings": False,
+    "eos_token_id": 151643,
+}
+
+SMALL_DEBUG = {  # --model debug: quick bring-up config
+    **QWEN25_32B,
+    "hidden_size": 1024, "intermediate_size": 2816,
+    "num_hidden_layers": 8, "num_attention_heads": 8,
+    "num_key_value_heads": 2, "vocab_size": 32000,
+}
+
+
+def write_model_dir(cfg_json):
+    d = tempfile.mkdtemp(prefix="bench_model_")
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump(cfg_json, f)
+    return d
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=32)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--batch", type=int, default=64)
+    ap.add_argument("--prompt-len", type=int, default=1024)
+    ap.add_argument("--model", type=str, default="qwen2.5-32b",
+                    choices=["qwen2.5-32b", "debug"])
+    ap.add_argument("--page-size", type=int, default=16)
+    ap.add_argument("--schedule", type=str, default="token_throttling")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    assert world == args.gpus or world == 1, \
+        f"WORLD_SIZE {world} != --gpus {args.gpus}"
+    n = max(world, 1)
+
+    use_gpu = torch.cuda.is_available()
+    device = f"cuda:{int(os.environ.get('LOCAL_RANK', rank))}" if use_gpu \
+        else "cpu"
+    model_json = QWEN25_32B if args.model == "qwen2.5-32b" else SMALL_DEBUG
+    model_dir = write_model_dir(model_json)
+
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.pp_engine import PPEngine
+    from gllm_amd.sequence import SamplingParams, Sequence
+
+    cfg = EngineConfig(
+        model=model_dir, load_format="dummy",
+        dtype="bfloat16" if use_gpu else "float32",
+        device=device, pp_size=n, page_size=args.page_size,
+        schedule_method=args.schedule, maxp=8192, maxd=1024,
+        use_graph=False, enable_prefix_caching=False,
+        master_addr=os.environ.get("MASTER_ADDR", "127.0.0.1"),
+        master_port=int(os.environ.get("MASTER_PORT", "29500")),
+    )
+    eng = PPEngine(cfg, num_pages_override=None if use_gpu else 512)
+
+    # ---- synthetic load: batch of fixed prompts, unbounded decode ----
+    g = torch.Generator().manual_seed(1234)
+    vocab = model_json["vocab_size"]
+    seqs = []
+    t_submit = time.time()
+    for i in range(args.batch):
+        ids = torch.randint(1, vocab - 1, (args.prompt_len,),
+                            generator=g).tolist()
+        sp = SamplingParams(temperature=0.0, ignore_eos=True,
+                            max_tokens=10 ** 9)
+        s = Sequence(i, ids, sp, eos_token_id=None, arrival_time=t_submit)
+        seqs.append(s)
+    eng.add_requests(seqs)
+
+    # ---- ramp: run until every seq has produced its first token ----
+    ttfts = {}
+    while len(ttfts) < len(seqs):
+        eng.step_tick()
+        now = time.time()
+        for s in seqs:
+            if s.seq_id not in ttfts and s.num_output_tokens > 0:
+                ttfts[s.seq_id] = (now - t_submit) * 1000.0
+        if not eng.scheduler.has_work():
+            break
+    ttft_sorted = sorted(ttfts.values())
+    ttft_p50 = ttft_sorted[len(ttft_sorted) // 2] if ttft_sorted else None
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        eng.step_tick()
+
+    # ---- timed region ----
+    eng.barrier_sync()
+    t0 = time.time()
+    sampled = 0
+    for _ in range(args.steps):
+        sampled += eng.step_tick()
+    eng.barrier_sync()
+    elapsed = time.time() - t0
+
+    # max elapsed over ranks
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    value = sampled / elapsed
+    if rank == 0:
+        out = {
+            "metric": "output tokens/sec (whole node)",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "ttft_p50_ms": round(ttft_p50, 1) if ttft_p50 else None,
+            "config": {
+                "model": "Qwen2.5-32B" if args.model == "qwen2.5-32b"
+                else "debug-0.2B",
+                "global_batch": args.batch,
+                "seq_len": args.prompt_len,
+                "parallelism": f"pp{n}",
+                "schedule": args.schedule,
+            },
+        }
+        print(json.dumps(out))
+    if world > 1:
+        import torch.distributed as dist
+        dist.barrier()
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,153 @@
+"""Pipeline-parallel engine: one process per GPU over RCCL/xGMI.
+
+MI355X-native design (differs from the reference's driver/follower
+split, worker.py:354-391 + dist_schedule.py): EVERY rank runs an
+identical deterministic replica of the scheduler on the same request
+stream, so batch geometry is known everywhere and no schedule broadcast
+/ delta-payload mirror is needed. Only two things move between ranks:
+
+  * PP legs: hidden_states + residual [T, hidden] bf16, point-to-point
+    send/recv — exactly one xGMI link per stage pair;
+  * sampled tokens: broadcast from the last stage on a DEDICATED
+    process group, so token broadcasts never interleave with the p2p
+    stream on one communicator (ordering safety).
+
+Pipelining: stage 0 keeps up to pp_size micro-batches in flight
+(scheduler.batch_running bound); each rank processes batches in FIFO
+order, so the pipeline fills across ranks.
+"""
+
+import time
+from collections import deque
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+from gllm_amd.config import EngineConfig
+from gllm_amd.core.scheduler import ScheduledBatch, Scheduler
+from gllm_amd.logger import logger
+from gllm_amd.runtime.model_runner import ModelRunner
+from gllm_amd.sequence import Sequence
+
+
+class PPEngine:
+    def __init__(self, config: EngineConfig,
+                 num_pages_override: Optional[int] = None):
+        from gllm_amd import parallel as P
+        self.P = P
+        P.init_distributed(config)
+        self.config = config
+        self.runner = ModelRunner(config).init(
+            num_pages_override=num_pages_override)
+        self.scheduler = Scheduler(config, self.runner.memory_manager)
+        self.pp_rank = P.get_pp_rank()
+        self.pp_size = P.get_pp_size()
+        self.is_first = P.is_first_pp_rank()
+        self.is_last = P.is_last_pp_rank()
+        self.device = config.device
+        self.hidden = self.runner.hf_config.hidden_size
+        self.dtype = config.torch_dtype()
+        # dedicated communicator for token broadcasts
+        if P.get_world_size() > 1:
+            self.token_group = dist.new_group(list(range(P.get_world_size())))
+            self.last_stage_rank = (self.pp_size - 1) * config.stage_size
+        else:
+            self.token_group = None
+        # in-flight (batch, pending-state) on this rank
+        self.inflight = deque()
+
+    # ------------------------------------------------------------------
+    def add_requests(self, seqs: List[Sequence]) -> None:
+        self.scheduler.add_seqs(seqs)
+
+    def _launch(self, batch: ScheduledBatch) -> None:
+        """Run this rank's part of the forward for ``batch``."""
+        if self.pp_size == 1:
+            out = self.runner.step_first_stage(batch)
+            self.inflight.append((batch, out))
+            return
+        if self.is_first:
+            hidden, residual, _ = self.runner.step_first_stage(batch)
+            dist.send(hidden.contiguous(), dst=self.P.get_next_pp_rank())
+            dist.send(residual.contiguous(), dst=self.P.get_next_pp_rank())
+            self.inflight.append((batch, None))
+        else:
+            T = batch.num_tokens
+            shape = (T, self.hidden)
+            hidden = torch.empty(shape, dtype=self.dtype, device=self.device)
+            residual = torch.empty(shape, dtype=self.dtype,
+                                   device=self.device)
+            src = self.P.get_prev_pp_rank()
+            dist.recv(hidden, src=src)
+            dist.recv(residual, src=src)
+            out = self.runner.step_mid_stage(batch, hidden, residual)
+            if self.is_last:
+                self.inflight.append((batch, out))
+            else:
+                h2, r2, _ = out
+                dist.send(h2.contiguous(), dst=self.P.get_next_pp_rank())
+                dist.send(r2.contiguous(), dst=self.P.get_next_pp_rank())
+                self.inflight.append((batch, None))
+
+    def _complete_oldest(self) -> List[Sequence]:
+        batch, out = self.inflight.popleft()
+        B = len(batch.items)
+        if self.pp_size == 1:
+            tokens = out.next_tokens.tolist()
+        else:
+            if self.is_last:
+                tok = out.next_tokens.to(self.device)
+            else:
+                tok = torch.empty(B, dtype=torch.long, device=self.device)
+            dist.broadcast(tok, src=self.last_stage_rank,
+                           group=self.token_group)
+            tokens = tok.tolist()
+        return self.scheduler.process_output(batch, tokens)
+
+    # ------------------------------------------------------------------
+    def run_until_done(self, max_steps: Optional[int] = None):
+        """Drive the loop until all requests finish. Returns finished seqs."""
+        done = []
+        steps = 0
+        while self.scheduler.has_work():
+            launched = False
+            while len(self.inflight) < max(1, self.pp_size):
+                b = self.scheduler.schedule_once()
+                if b is None:
+                    break
+                self._launch(b)
+                launched = True
+            if self.inflight:
+                done.extend(self._complete_oldest())
+            elif not launched:
+                break
+            steps += 1
+            if max_steps is not None and steps >= max_steps:
+                break
+        return done
+
+    def step_tick(self) -> int:
+        """One pipeline tick: launch as many batches as fit, complete the
+        oldest. Returns decode tokens committed this tick."""
+        while len(self.inflight) < max(1, self.pp_size):
+            b = self.scheduler.schedule_once()
+            if b is None:
+                break
+            self._launch(b)
+        if not self.inflight:
+            return 0
+        batch = self.inflight[0][0]
+        n_sampled = sum(1 for it in batch.items if it.ends_prompt)
+        self._complete_oldest()
+        return n_sampled
+
+    def drain(self) -> None:
+        while self.inflight:
+            self._complete_oldest()
+
+    def barrier_sync(self) -> None:
+        if self.token_group is not None:
+            dist.barrier()
+        if self.device.startswith("cuda"):
+            torch.cuda.synchronize()

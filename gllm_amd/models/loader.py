@@ -69,17 +69,36 @@ def iterate_safetensors(model_path: str
 
 
 def dummy_init(model: torch.nn.Module, seed: int = 0) -> None:
-    """Random-init all params (reference --load-format dummy). Seeded so
-    every TP rank holding a replicated param gets identical values, and
-    numerically tame so bf16 forward passes stay finite."""
-    gen = torch.Generator()
-    for name, p in sorted(model.named_parameters()):
-        gen.manual_seed(seed ^ (hash(name) & 0x7FFFFFFF))
+    """Random-init all params (reference --load-format dummy).
+
+    Seeding must be (a) process-deterministic (crc32, not the salted
+    built-in hash) and (b) PP-stage-aligned: a param named layers.0.* on
+    stage 1 is GLOBAL layer `layer_start`, so the same global layer gets
+    the same weights regardless of the pipeline split. TP shards still
+    differ by rank (narrowed from differently-seeded full tensors is NOT
+    done here — each rank draws its own shard; TP-replicated params get
+    identical draws)."""
+    import zlib
+    layer_start = getattr(model, "layer_start", 0)
+
+    def global_name(name: str) -> str:
+        if name.startswith("layers."):
+            parts = name.split(".")
+            parts[1] = str(int(parts[1]) + layer_start)
+            return ".".join(parts)
+        return name
+
+    params = sorted(model.named_parameters())
+    on_gpu = any(p.is_cuda for _, p in params)
+    gen = (torch.Generator(device="cuda") if on_gpu else torch.Generator())
+    for name, p in params:
+        gname = global_name(name)
+        gen.manual_seed(seed ^ zlib.crc32(gname.encode()))
         with torch.no_grad():
             if p.dim() >= 2:
-                t = torch.empty(p.shape, dtype=torch.float32)
-                t.normal_(0.0, 0.02, generator=gen)
-                p.data.copy_(t.to(p.dtype))
+                # draw in-place on device (32B params via a CPU RNG would
+                # take minutes); bf16 normal_ is supported on ROCm
+                p.data.normal_(0.0, 0.02, generator=gen)
             elif "norm" in name or "weight" in name and p.dim() == 1 and \
                     "bias" not in name:
                 p.data.fill_(1.0)

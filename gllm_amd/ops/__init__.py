@@ -115,7 +115,7 @@ def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
         else:
             kern.paged_attention_prefill(
                 out, q, k_cache, v_cache, block_table, seq_lens,
-                query_start_loc, scale, sliding_window)
+                query_start_loc.int(), max_query_len, scale, sliding_window)
         return out
     return torch_ref.paged_attention(q, k_cache, v_cache, block_table,
                                      seq_lens, query_start_loc, scale,

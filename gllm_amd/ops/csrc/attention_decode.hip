@@ -1,0 +1,289 @@
+// Paged decode attention (single query token per sequence) for gfx950.
+//
+// Memory-bound: the job is to stream each sequence's K/V pages once at
+// near-HBM rate. Structure:
+//   grid = (B, Hkv, num_splits), block = 256 threads (4 waves).
+//   Each workgroup serves the whole GQA group (G = Hq/Hkv q-heads) of one
+//   kv head over a contiguous token split, so K/V bytes are read once for
+//   all G heads. Per CHUNK of tokens:
+//     A: lane-groups of D/8 lanes each load one K row (16 B/lane) and a
+//        V row (staged to LDS), dot against the G q-vectors in registers,
+//        group-reduce, scores -> LDS.
+//     B1: per-head online-softmax update (m, l) + p = exp(s - m) -> LDS.
+//     B2: (h, d)-mapped threads accumulate acc += p[h][t] * V_lds[t][d].
+//   Split partials written fp32 + LSE; merge kernel combines splits and
+//   converts to bf16 (reference semantics: merge_state / split-KV decode,
+//   SURVEY.md §2.4).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+namespace {
+
+constexpr int BLOCK = 256;
+
+template <int D, int G, int CHUNK>
+__global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
+    float *__restrict__ partial_out,  // [S, B, Hq, D]
+    float *__restrict__ partial_lse,  // [S, B, Hq]
+    const __hip_bfloat16 *__restrict__ q,        // [B, Hq, D]
+    const __hip_bfloat16 *__restrict__ k_cache,  // [P, ps, Hkv, D]
+    const __hip_bfloat16 *__restrict__ v_cache,
+    const int *__restrict__ block_table,  // [B, max_pages]
+    const int *__restrict__ seq_lens,     // [B]
+    int max_pages, int page_size, int num_kv_heads, float scale,
+    int num_splits) {
+  const int b = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int split = blockIdx.z;
+  const int Hq = num_kv_heads * G;
+  const int seq_len = seq_lens[b];
+
+  // split token range (page-aligned chunks not required; rows are
+  // addressed per token through the block table)
+  const int split_len = (seq_len + num_splits - 1) / num_splits;
+  const int t0 = split * split_len;
+  const int t1 = min(t0 + split_len, seq_len);
+  const int lane_per_tok = D / 8;            // 16 lanes for D=128
+  const int toks_per_iter = BLOCK / lane_per_tok;
+
+  // LDS: V tile + scores + p + softmax state
+  __shared__ __hip_bfloat16 v_tile[CHUNK][D];
+  __shared__ float s_scores[G][CHUNK];
+  __shared__ float s_m[G], s_l[G], s_alpha[G];
+
+  const int tid = threadIdx.x;
+  const int tok_slot = tid / lane_per_tok;   // which token in the iter
+  const int dlane = tid % lane_per_tok;      // which 8-elem d-chunk
+  const int d_off = dlane * 8;
+
+  if (t0 >= t1) {
+    // empty split: publish -inf lse so the merge ignores it
+    if (tid < G) partial_lse[(long)split * gridDim.x * Hq + (long)b * Hq +
+                             kvh * G + tid] = -INFINITY;
+    return;
+  }
+
+  // ---- q fragments in registers: G heads x 8 elems of this lane's slice
+  float qreg[G][8];
+#pragma unroll
+  for (int h = 0; h < G; ++h) {
+    const __hip_bfloat16 *qp =
+        q + ((long)b * Hq + kvh * G + h) * D + d_off;
+    shortx8 p = *reinterpret_cast<const shortx8 *>(qp);
+    unpack8<__hip_bfloat16>(p, qreg[h]);
+  }
+
+  // ---- per-thread output accumulators: (h, d) pairs
+  constexpr int ACC = (G * D + BLOCK - 1) / BLOCK;  // e.g. 4 for G=8, D=128
+  float acc[ACC];
+#pragma unroll
+  for (int i = 0; i < ACC; ++i) acc[i] = 0.f;
+  if (tid < G) { s_m[tid] = -INFINITY; s_l[tid] = 0.f; }
+  __syncthreads();
+
+  const int *bt = block_table + (long)b * max_pages;
+
+  for (int c0 = t0; c0 < t1; c0 += CHUNK) {
+    const int c_len = min(CHUNK, t1 - c0);
+    // ---------- phase A: scores + V staging ----------
+    for (int it = 0; it < (c_len + toks_per_iter - 1) / toks_per_iter; ++it) {
+      const int tt = it * toks_per_iter + tok_slot;  // token within chunk
+      float dot[G];
+#pragma unroll
+      for (int h = 0; h < G; ++h) dot[h] = 0.f;
+      if (tt < c_len) {
+        const int tok = c0 + tt;
+        const long row = ((long)bt[tok / page_size] * page_size +
+                          tok % page_size);
+        const __hip_bfloat16 *kp =
+            k_cache + (row * num_kv_heads + kvh) * D + d_off;
+        const __hip_bfloat16 *vp =
+            v_cache + (row * num_kv_heads + kvh) * D + d_off;
+        shortx8 kv8 = *reinterpret_cast<const shortx8 *>(kp);
+        shortx8 vv8 = *reinterpret_cast<const shortx8 *>(vp);
+        float kf[8];
+        unpack8<__hip_bfloat16>(kv8, kf);
+        *reinterpret_cast<shortx8 *>(&v_tile[tt][d_off]) = vv8;
+#pragma unroll
+        for (int h = 0; h < G; ++h) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) dot[h] += qreg[h][j] * kf[j];
+        }
+      }
+      // reduce within the token's lane group; leader writes the score
+#pragma unroll
+      for (int h = 0; h < G; ++h) {
+        float v = group_reduce_sum<D / 8>(dot[h]);
+        if (dlane == 0 && tt < c_len) s_scores[h][tt] = v * scale;
+      }
+    }
+    __syncthreads();
+
+    // ---------- phase B1: online softmax per head ----------
+    {
+      // threads per head: largest power of 2 <= BLOCK/G, capped at 64
+      // (shfl width), so shfl groups stay lane-aligned for any G (incl. 5)
+      constexpr int TPH0 = BLOCK / G;
+      constexpr int TPH = TPH0 >= 64 ? 64 : (TPH0 >= 32 ? 32 :
+                          (TPH0 >= 16 ? 16 : (TPH0 >= 8 ? 8 : 4)));
+      const int h = tid / TPH;
+      const int j = tid % TPH;
+      if (h < G) {
+        float mx = -INFINITY;
+        for (int t = j; t < c_len; t += TPH) mx = fmaxf(mx, s_scores[h][t]);
+#pragma unroll
+        for (int off = TPH / 2; off > 0; off >>= 1)
+          mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+        const float m_old = s_m[h];
+        const float m_new = fmaxf(m_old, mx);
+        float psum = 0.f;
+        for (int t = j; t < c_len; t += TPH) {
+          const float p = __expf(s_scores[h][t] - m_new);
+          s_scores[h][t] = p;                 // reuse scores LDS as p
+          psum += p;
+        }
+#pragma unroll
+        for (int off = TPH / 2; off > 0; off >>= 1)
+          psum += __shfl_xor(psum, off, 64);
+        if (j == 0) {
+          const float alpha =
+              (m_old == -INFINITY) ? 0.f : __expf(m_old - m_new);
+          s_alpha[h] = alpha;
+          s_l[h] = s_l[h] * alpha + psum;
+          s_m[h] = m_new;
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---------- phase B2: PV accumulation ----------
+#pragma unroll
+    for (int i = 0; i < ACC; ++i) {
+      const int flat = tid + i * BLOCK;       // (h, d) index
+      if (flat >= G * D) break;
+      const int h = flat / D;
+      const int d = flat % D;
+      float a = acc[i] * s_alpha[h];
+      for (int t = 0; t < c_len; ++t) {
+        a += s_scores[h][t] *
+             __bfloat162float(v_tile[t][d]);
+      }
+      acc[i] = a;
+    }
+    __syncthreads();
+  }
+
+  // ---------- epilogue: normalized split partial + lse ----------
+#pragma unroll
+  for (int i = 0; i < ACC; ++i) {
+    const int flat = tid + i * BLOCK;
+    if (flat >= G * D) break;
+    const int h = flat / D;
+    const int d = flat % D;
+    const float l = s_l[h];
+    partial_out[(((long)split * gridDim.x + b) * Hq + kvh * G + h) * D + d] =
+        (l > 0.f) ? acc[i] / l : 0.f;
+  }
+  if (tid < G) {
+    const float l = s_l[tid];
+    partial_lse[(long)split * gridDim.x * Hq + (long)b * Hq + kvh * G + tid] =
+        (l > 0.f) ? s_m[tid] + __logf(l) : -INFINITY;
+  }
+}
+
+// Merge split partials: out[b,h,:] = sum_s w_s * partial[s,b,h,:]
+template <int D>
+__global__ void decode_merge_kernel(
+    __hip_bfloat16 *__restrict__ out,       // [B, Hq, D]
+    const float *__restrict__ partial_out,  // [S, B, Hq, D]
+    const float *__restrict__ partial_lse,  // [S, B, Hq]
+    int num_splits, int Hq) {
+  const int b = blockIdx.x, h = blockIdx.y;
+  const int d = threadIdx.x;
+  const int B = gridDim.x;
+  float mx = -INFINITY;
+  for (int s = 0; s < num_splits; ++s)
+    mx = fmaxf(mx, partial_lse[((long)s * B + b) * Hq + h]);
+  float denom = 0.f, o = 0.f;
+  for (int s = 0; s < num_splits; ++s) {
+    const float lse = partial_lse[((long)s * B + b) * Hq + h];
+    if (lse == -INFINITY) continue;
+    const float w = __expf(lse - mx);
+    denom += w;
+    o += w * partial_out[(((long)s * B + b) * Hq + h) * D + d];
+  }
+  out[((long)b * Hq + h) * D + d] =
+      __float2bfloat16(denom > 0.f ? o / denom : 0.f);
+}
+
+template <int D, int G>
+void launch_decode(torch::Tensor &out, const torch::Tensor &q,
+                   const torch::Tensor &k_cache, const torch::Tensor &v_cache,
+                   const torch::Tensor &block_table,
+                   const torch::Tensor &seq_lens, float scale,
+                   int max_seq_len) {
+  const int B = q.size(0);
+  const int Hq = q.size(1);
+  const int Hkv = k_cache.size(2);
+  constexpr int CHUNK = 128;
+  // split heuristic: enough workgroups to fill 256 CUs x 8 XCDs
+  int splits = 1;
+  const int base_wgs = B * Hkv;
+  while (splits < 16 && base_wgs * splits < 640 &&
+         splits * 2 <= (max_seq_len + CHUNK - 1) / CHUNK)
+    splits *= 2;
+  auto opts = q.options().dtype(at::kFloat);
+  auto partial = torch::empty({splits, B, Hq, D}, opts);
+  auto lse = torch::empty({splits, B, Hq}, opts);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL((paged_decode_kernel<D, G, CHUNK>),
+                     dim3(B, Hkv, splits), dim3(BLOCK), 0, stream,
+                     partial.data_ptr<float>(), lse.data_ptr<float>(),
+                     (const __hip_bfloat16 *)q.data_ptr(),
+                     (const __hip_bfloat16 *)k_cache.data_ptr(),
+                     (const __hip_bfloat16 *)v_cache.data_ptr(),
+                     block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
+                     (int)block_table.size(1), (int)k_cache.size(1), Hkv,
+                     scale, splits);
+  HIP_CHECK_KERNEL();
+  hipLaunchKernelGGL((decode_merge_kernel<D>), dim3(B, Hq), dim3(D), 0,
+                     stream, (__hip_bfloat16 *)out.data_ptr(),
+                     partial.data_ptr<float>(), lse.data_ptr<float>(), splits,
+                     Hq);
+  HIP_CHECK_KERNEL();
+}
+
+}  // namespace
+
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_table, torch::Tensor seq_lens,
+                            double scale, long sliding_window) {
+  TORCH_CHECK(sliding_window == 0, "sliding window: not yet");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "decode attn: bf16 only");
+  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(block_table.scalar_type() == at::kInt);
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
+  const int D = q.size(2);
+  const int Hq = q.size(1);
+  const int Hkv = k_cache.size(2);
+  TORCH_CHECK(Hq % Hkv == 0);
+  const int G = Hq / Hkv;
+  const int max_seq = k_cache.size(0) * k_cache.size(1);  // upper bound
+  const float s = (float)scale;
+
+#define CASE(DD, GG)                                                       \
+  if (D == DD && G == GG) {                                                \
+    launch_decode<DD, GG>(out, q, k_cache, v_cache, block_table, seq_lens, \
+                          s, max_seq);                                     \
+    return;                                                                \
+  }
+  CASE(128, 1) CASE(128, 2) CASE(128, 4) CASE(128, 5) CASE(128, 8)
+  CASE(64, 1) CASE(64, 2) CASE(64, 4) CASE(64, 8)
+  CASE(128, 16)
+#undef CASE
+  TORCH_CHECK(false, "paged_attention_decode: unsupported head_dim=", D,
+              " group=", G);
+}

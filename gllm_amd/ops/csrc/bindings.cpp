@@ -1,0 +1,39 @@
+// Python bindings for the gfx950 kernel extension.
+#include <torch/extension.h>
+
+void rmsnorm(torch::Tensor out, torch::Tensor in, torch::Tensor weight,
+             double eps);
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                       torch::Tensor weight, double eps);
+void silu_and_mul(torch::Tensor out, torch::Tensor x);
+void rotary_embedding(torch::Tensor positions, torch::Tensor q,
+                      torch::Tensor k, long head_dim,
+                      torch::Tensor cos_sin_cache, bool is_neox);
+void reshape_and_cache(torch::Tensor k, torch::Tensor v,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor slot_mapping);
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_table, torch::Tensor seq_lens,
+                            double scale, long sliding_window);
+void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
+                             torch::Tensor k_cache, torch::Tensor v_cache,
+                             torch::Tensor block_table,
+                             torch::Tensor seq_lens,
+                             torch::Tensor query_start_loc,
+                             long max_query_len, double scale,
+                             long sliding_window);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (gfx950)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
+        "residual += x; x = rmsnorm(residual)");
+  m.def("silu_and_mul", &silu_and_mul, "silu(x[:d]) * x[d:]");
+  m.def("rotary_embedding", &rotary_embedding, "in-place RoPE on q,k");
+  m.def("reshape_and_cache", &reshape_and_cache,
+        "scatter K/V into paged cache");
+  m.def("paged_attention_decode", &paged_attention_decode,
+        "split-KV paged decode attention");
+  m.def("paged_attention_prefill", &paged_attention_prefill,
+        "varlen MFMA paged prefill attention");
+}

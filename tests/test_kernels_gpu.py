@@ -1,0 +1,175 @@
+"""HIP kernel numerics vs the torch fp32 oracle (run on MI355X)."""
+
+import math
+
+import pytest
+import torch
+
+from gllm_amd.ops import torch_ref as R
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def kernels():
+    from gllm_amd import ops
+    assert ops.has_kernels(), "HIP extension must be built"
+    return ops
+
+
+def assert_close_bf16(out, ref, atol=2e-2, rtol=2e-2, frac=1e-3):
+    """bf16 kernel vs fp32 oracle: allow bf16 rounding noise."""
+    out = out.float().cpu()
+    ref = ref.float().cpu()
+    bad = (~torch.isclose(out, ref, atol=atol, rtol=rtol)).float().mean()
+    assert bad < frac, (
+        f"{bad*100:.3f}% mismatched; max abs err "
+        f"{(out-ref).abs().max():.4f}")
+
+
+def test_rmsnorm(kernels):
+    torch.manual_seed(0)
+    x = torch.randn(512, 4096, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(4096, dtype=torch.bfloat16, device="cuda")
+    out = kernels.rmsnorm(x, w, 1e-6)
+    ref = R.rmsnorm(x.float().cpu(), w.float().cpu(), 1e-6)
+    assert_close_bf16(out, ref)
+
+
+def test_fused_add_rmsnorm(kernels):
+    torch.manual_seed(1)
+    x = torch.randn(256, 4096, dtype=torch.bfloat16, device="cuda")
+    r = torch.randn(256, 4096, dtype=torch.bfloat16, device="cuda")
+    x_ref = x.float().cpu().clone()
+    r_ref = r.float().cpu().clone()
+    w = torch.randn(4096, dtype=torch.bfloat16, device="cuda")
+    kernels.fused_add_rmsnorm(x, r, w, 1e-6)
+    r2 = (x_ref + r_ref)
+    x2 = R.rmsnorm(r2.to(torch.bfloat16).float(), w.float().cpu(), 1e-6)
+    assert_close_bf16(r, r2)
+    assert_close_bf16(x, x2)
+
+
+def test_silu_and_mul(kernels):
+    torch.manual_seed(2)
+    x = torch.randn(333, 2 * 1536, dtype=torch.bfloat16, device="cuda")
+    out = kernels.silu_and_mul(x)
+    ref = R.silu_and_mul(x.float().cpu())
+    assert_close_bf16(out, ref)
+
+
+@pytest.mark.parametrize("is_neox", [True, False])
+def test_rope(kernels, is_neox):
+    torch.manual_seed(3)
+    T, Hq, Hk, D = 100, 8, 2, 128
+    inv = 1.0 / (10000 ** (torch.arange(0, D, 2).float() / D))
+    t = torch.arange(4096).float()
+    freqs = torch.outer(t, inv)
+    cache = torch.cat([freqs.cos(), freqs.sin()], -1).cuda()
+    q = torch.randn(T, Hq * D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(T, Hk * D, dtype=torch.bfloat16, device="cuda")
+    pos = torch.randint(0, 4000, (T,), device="cuda")
+    q_ref, k_ref = R.rotary_embedding(
+        pos.cpu(), q.float().cpu().clone(), k.float().cpu().clone(), D,
+        cache.cpu(), is_neox)
+    kernels.rotary_embedding(pos, q, k, D, cache, is_neox)
+    assert_close_bf16(q, q_ref)
+    assert_close_bf16(k, k_ref)
+
+
+def test_reshape_and_cache(kernels):
+    torch.manual_seed(4)
+    T, H, D, ps, P = 64, 8, 128, 16, 32
+    k = torch.randn(T, H, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, H, D, dtype=torch.bfloat16, device="cuda")
+    kc = torch.zeros(P, ps, H, D, dtype=torch.bfloat16, device="cuda")
+    vc = torch.zeros(P, ps, H, D, dtype=torch.bfloat16, device="cuda")
+    slots = torch.randperm(P * ps, device="cuda")[:T]
+    kernels.reshape_and_cache(k, v, kc, vc, slots)
+    kr = torch.zeros_like(kc).cpu()
+    vr = torch.zeros_like(vc).cpu()
+    R.reshape_and_cache(k.cpu(), v.cpu(), kr, vr, slots.cpu())
+    assert torch.equal(kc.cpu(), kr)
+    assert torch.equal(vc.cpu(), vr)
+
+
+def _mk_paged(B, Hkv, D, ps, ctx_lens, seed=0):
+    torch.manual_seed(seed)
+    max_pages = max(-(-c // ps) for c in ctx_lens)
+    total_pages = sum(-(-c // ps) for c in ctx_lens) + 1
+    k_cache = torch.randn(total_pages, ps, Hkv, D, dtype=torch.bfloat16,
+                          device="cuda")
+    v_cache = torch.randn(total_pages, ps, Hkv, D, dtype=torch.bfloat16,
+                          device="cuda")
+    bt = torch.zeros(B, max_pages, dtype=torch.int32, device="cuda")
+    next_page = 1
+    for b, c in enumerate(ctx_lens):
+        n = -(-c // ps)
+        bt[b, :n] = torch.arange(next_page, next_page + n)
+        next_page += n
+    return k_cache, v_cache, bt
+
+
+@pytest.mark.parametrize("G", [4, 5, 8])
+def test_decode_attention(kernels, G):
+    B, Hkv, D, ps = 5, 4, 128, 16
+    Hq = G * Hkv
+    ctx = [1, 17, 160, 1000, 333]
+    k_cache, v_cache, bt = _mk_paged(B, Hkv, D, ps, ctx, seed=G)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device="cuda")
+    seq_lens = torch.tensor(ctx, dtype=torch.int32, device="cuda")
+    qsl = torch.arange(B + 1, dtype=torch.int32, device="cuda")
+    from gllm_amd import ops
+    out = ops.paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl,
+                              1.0 / math.sqrt(D), max_query_len=1)
+    ref = R.paged_attention(q.float().cpu(), k_cache.float().cpu(),
+                            v_cache.float().cpu(), bt.cpu(), seq_lens.cpu(),
+                            qsl.cpu(), 1.0 / math.sqrt(D))
+    assert_close_bf16(out, ref, frac=2e-3)
+
+
+@pytest.mark.parametrize("case", [
+    # (q_lens, ctx_lens): ctx >= q (past = ctx - q)
+    ([64], [64]),
+    ([1, 64, 17], [100, 64, 333]),       # mixed decode + prefill + chunk
+    ([200, 1], [200, 77]),
+    ([130], [1030]),                     # long past (chunked context)
+])
+def test_prefill_attention(kernels, case):
+    q_lens, ctx = case
+    B, Hkv, D, ps = len(q_lens), 2, 128, 16
+    G = 4
+    Hq = G * Hkv
+    k_cache, v_cache, bt = _mk_paged(B, Hkv, D, ps, ctx, seed=7)
+    T = sum(q_lens)
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+    seq_lens = torch.tensor(ctx, dtype=torch.int32, device="cuda")
+    qsl = torch.tensor([0] + list(torch.cumsum(
+        torch.tensor(q_lens), 0)), dtype=torch.int32, device="cuda")
+    from gllm_amd import ops
+    out = ops.paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl,
+                              1.0 / math.sqrt(D),
+                              max_query_len=max(q_lens))
+    ref = R.paged_attention(q.float().cpu(), k_cache.float().cpu(),
+                            v_cache.float().cpu(), bt.cpu(), seq_lens.cpu(),
+                            qsl.cpu(), 1.0 / math.sqrt(D))
+    assert_close_bf16(out, ref, frac=2e-3)
+
+
+def test_prefill_attention_head_dim_64(kernels):
+    B, Hkv, D, ps, G = 2, 2, 64, 16, 2
+    Hq = G * Hkv
+    ctx = [96, 40]
+    q_lens = [96, 40]
+    k_cache, v_cache, bt = _mk_paged(B, Hkv, D, ps, ctx, seed=9)
+    T = sum(q_lens)
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+    seq_lens = torch.tensor(ctx, dtype=torch.int32, device="cuda")
+    qsl = torch.tensor([0, 96, 136], dtype=torch.int32, device="cuda")
+    from gllm_amd import ops
+    out = ops.paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl,
+                              1.0 / math.sqrt(D), max_query_len=96)
+    ref = R.paged_attention(q.float().cpu(), k_cache.float().cpu(),
+                            v_cache.float().cpu(), bt.cpu(), seq_lens.cpu(),
+                            qsl.cpu(), 1.0 / math.sqrt(D))
+    assert_close_bf16(out, ref, frac=2e-3)
