@@ -29,9 +29,9 @@ class Attention(nn.Module):
     def forward(self, q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                 fctx: ForwardContext) -> torch.Tensor:
         T = q.shape[0]
-        q = q.view(T, self.num_heads, self.head_dim)
-        k = k.view(T, self.num_kv_heads, self.head_dim)
-        v = v.view(T, self.num_kv_heads, self.head_dim)
+        q = q.unflatten(-1, (self.num_heads, self.head_dim))
+        k = k.unflatten(-1, (self.num_kv_heads, self.head_dim))
+        v = v.unflatten(-1, (self.num_kv_heads, self.head_dim))
         if fctx.is_profile_run:
             # Peak-memory profile pass runs before the KV cache exists
             # (reference layers/attention.py:112-116): skip attention math,

@@ -75,9 +75,11 @@ class DenseAttention(nn.Module):
         q, k, v = self.qkv_proj(hidden)
         if self.qk_norm:
             T = q.shape[0]
-            q = self.q_norm(q.view(T, -1, self.head_dim)).view(T, -1)
-            k = self.k_norm(k.view(T, -1, self.head_dim)).view(T, -1)
-        q, k = self.rotary_emb(positions, q.contiguous(), k.contiguous())
+            q = self.q_norm(
+                q.contiguous().view(T, -1, self.head_dim)).view(T, -1)
+            k = self.k_norm(
+                k.contiguous().view(T, -1, self.head_dim)).view(T, -1)
+        q, k = self.rotary_emb(positions, q, k)
         o = self.attn(q, k, v, fctx)
         return self.o_proj(o)
 

@@ -107,7 +107,7 @@ def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
     if q.is_cuda:
         kern = _gpu_kernels()
         if out is None:
-            out = torch.empty_like(q)
+            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         if max_query_len == 1:
             kern.paged_attention_decode(
                 out, q, k_cache, v_cache, block_table, seq_lens, scale,

@@ -34,7 +34,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     const int *__restrict__ block_table,  // [B, max_pages]
     const int *__restrict__ seq_lens,     // [B]
     int max_pages, int page_size, int num_kv_heads, float scale,
-    int num_splits) {
+    int num_splits, long q_stride) {
   const int b = blockIdx.x;
   const int kvh = blockIdx.y;
   const int split = blockIdx.z;
@@ -71,7 +71,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
 #pragma unroll
   for (int h = 0; h < G; ++h) {
     const __hip_bfloat16 *qp =
-        q + ((long)b * Hq + kvh * G + h) * D + d_off;
+        q + (long)b * q_stride + (kvh * G + h) * D + d_off;
     shortx8 p = *reinterpret_cast<const shortx8 *>(qp);
     unpack8<__hip_bfloat16>(p, qreg[h]);
   }
@@ -224,7 +224,7 @@ void launch_decode(torch::Tensor &out, const torch::Tensor &q,
                    const torch::Tensor &k_cache, const torch::Tensor &v_cache,
                    const torch::Tensor &block_table,
                    const torch::Tensor &seq_lens, float scale,
-                   int max_seq_len) {
+                   int max_seq_len, long q_stride) {
   const int B = q.size(0);
   const int Hq = q.size(1);
   const int Hkv = k_cache.size(2);
@@ -247,7 +247,7 @@ void launch_decode(torch::Tensor &out, const torch::Tensor &q,
                      (const __hip_bfloat16 *)v_cache.data_ptr(),
                      block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                      (int)block_table.size(1), (int)k_cache.size(1), Hkv,
-                     scale, splits);
+                     scale, splits, q_stride);
   HIP_CHECK_KERNEL();
   hipLaunchKernelGGL((decode_merge_kernel<D>), dim3(B, Hq), dim3(D), 0,
                      stream, (__hip_bfloat16 *)out.data_ptr(),
@@ -264,7 +264,9 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             double scale, long sliding_window) {
   TORCH_CHECK(sliding_window == 0, "sliding window: not yet");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "decode attn: bf16 only");
-  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(out.is_contiguous());
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2),
+              "q per-token row must be contiguous");
   TORCH_CHECK(block_table.scalar_type() == at::kInt);
   TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
   const int D = q.size(2);
@@ -278,7 +280,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
 #define CASE(DD, GG)                                                       \
   if (D == DD && G == GG) {                                                \
     launch_decode<DD, GG>(out, q, k_cache, v_cache, block_table, seq_lens, \
-                          s, max_seq);                                     \
+                          s, max_seq, q.stride(0));                        \
     return;                                                                \
   }
   CASE(128, 1) CASE(128, 2) CASE(128, 4) CASE(128, 5) CASE(128, 8)

@@ -49,7 +49,8 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
     const int *__restrict__ block_table,         // [B, max_pages]
     const int *__restrict__ seq_lens,            // [B]
     const int *__restrict__ qsl,                 // [B+1]
-    int max_pages, int page_size, int Hq, int num_kv_heads, float scale) {
+    int max_pages, int page_size, int Hq, int num_kv_heads, float scale,
+    long q_stride) {
   const int tile = blockIdx.x;
   const int b = blockIdx.y;
   const int h = blockIdx.z;
@@ -81,7 +82,7 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
   const bool qrow_valid = qrow_local < q_len;
   {
     const __hip_bfloat16 *qp =
-        q + ((long)(q_start + qrow_local) * Hq + h) * D + lhi * 8;
+        q + (long)(q_start + qrow_local) * q_stride + h * D + lhi * 8;
 #pragma unroll
     for (int kt = 0; kt < KT; ++kt) {
       if (qrow_valid)
@@ -251,7 +252,9 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              long sliding_window) {
   TORCH_CHECK(sliding_window == 0, "sliding window: not yet");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "prefill attn: bf16 only");
-  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(out.is_contiguous());
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2),
+              "q per-token row must be contiguous");
   const int D = q.size(2);
   const int Hq = q.size(1);
   const int Hkv = k_cache.size(2);
@@ -270,7 +273,7 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                        query_start_loc.data_ptr<int>(),
                        (int)block_table.size(1), (int)k_cache.size(1), Hq,
-                       Hkv, (float)scale);
+                       Hkv, (float)scale, q.stride(0));
   } else if (D == 64) {
     hipLaunchKernelGGL((paged_prefill_kernel<64>),
                        dim3(q_tiles, B, Hq), dim3(BLOCK), 0, stream,
@@ -281,7 +284,7 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                        query_start_loc.data_ptr<int>(),
                        (int)block_table.size(1), (int)k_cache.size(1), Hq,
-                       Hkv, (float)scale);
+                       Hkv, (float)scale, q.stride(0));
   } else {
     TORCH_CHECK(false, "prefill attn: head_dim ", D, " unsupported");
   }

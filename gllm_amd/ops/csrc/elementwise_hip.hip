@@ -40,7 +40,8 @@ template <typename T, bool NEOX>
 __global__ void rope_kernel(const long *__restrict__ positions,
                             T *__restrict__ q, T *__restrict__ k,
                             const float *__restrict__ cache, int rot_dim,
-                            int head_dim, int num_q_heads, int num_k_heads) {
+                            int head_dim, int num_q_heads, int num_k_heads,
+                            long q_stride, long k_stride) {
   const long t = blockIdx.x;
   const float *cs = cache + positions[t] * rot_dim;
   const int half = rot_dim / 2;
@@ -49,9 +50,8 @@ __global__ void rope_kernel(const long *__restrict__ positions,
     const int h = idx / half;
     const int p = idx % half;           // rotation pair index
     T *base = (h < num_q_heads)
-                  ? q + t * (long)num_q_heads * head_dim + h * head_dim
-                  : k + t * (long)num_k_heads * head_dim +
-                        (h - num_q_heads) * head_dim;
+                  ? q + t * q_stride + h * head_dim
+                  : k + t * k_stride + (h - num_q_heads) * head_dim;
     const float c = cs[p], s = cs[half + p];
     int i1, i2;
     if (NEOX) { i1 = p; i2 = p + half; }
@@ -69,14 +69,14 @@ template <typename T>
 __global__ void reshape_and_cache_kernel(
     const T *__restrict__ k, const T *__restrict__ v, T *__restrict__ k_cache,
     T *__restrict__ v_cache, const long *__restrict__ slot_mapping,
-    int row_elems /* H*D */, int page_size) {
+    int row_elems /* H*D */, int page_size, long k_stride, long v_stride) {
   const long t = blockIdx.x;
   const long slot = slot_mapping[t];
   const long page = slot / page_size, off = slot % page_size;
   const long dst = (page * page_size + off) * (long)row_elems;
   const int nvec = row_elems / 8;
-  const shortx8 *ks = reinterpret_cast<const shortx8 *>(k + t * (long)row_elems);
-  const shortx8 *vs = reinterpret_cast<const shortx8 *>(v + t * (long)row_elems);
+  const shortx8 *ks = reinterpret_cast<const shortx8 *>(k + t * k_stride);
+  const shortx8 *vs = reinterpret_cast<const shortx8 *>(v + t * v_stride);
   shortx8 *kd = reinterpret_cast<shortx8 *>(k_cache + dst);
   shortx8 *vd = reinterpret_cast<shortx8 *>(v_cache + dst);
   for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
@@ -117,7 +117,9 @@ void rotary_embedding(torch::Tensor positions, torch::Tensor q,
   const long T = positions.size(0);
   const int rot_dim = cos_sin_cache.size(-1);
   const int hq = q.size(-1) / head_dim, hk = k.size(-1) / head_dim;
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous());
+  TORCH_CHECK(q.stride(-1) == 1 && k.stride(-1) == 1,
+              "rope: innermost dim must be contiguous");
+  const long qs = q.stride(0), ks = k.stride(0);
   TORCH_CHECK(q.scalar_type() == at::kBFloat16,
               "rope: bf16 only for now");
   TORCH_CHECK(cos_sin_cache.scalar_type() == at::kFloat);
@@ -129,7 +131,7 @@ void rotary_embedding(torch::Tensor positions, torch::Tensor q,
                        (__hip_bfloat16 *)q.data_ptr(),
                        (__hip_bfloat16 *)k.data_ptr(),
                        cos_sin_cache.data_ptr<float>(), rot_dim,
-                       (int)head_dim, hq, hk);
+                       (int)head_dim, hq, hk, qs, ks);
   } else {
     hipLaunchKernelGGL((rope_kernel<__hip_bfloat16, false>), dim3(T),
                        dim3(256), 0, stream,
@@ -137,7 +139,7 @@ void rotary_embedding(torch::Tensor positions, torch::Tensor q,
                        (__hip_bfloat16 *)q.data_ptr(),
                        (__hip_bfloat16 *)k.data_ptr(),
                        cos_sin_cache.data_ptr<float>(), rot_dim,
-                       (int)head_dim, hq, hk);
+                       (int)head_dim, hq, hk, qs, ks);
   }
   HIP_CHECK_KERNEL();
 }
@@ -150,7 +152,9 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v,
   const int row = k.size(1) * k.size(2);
   const int page_size = k_cache.size(1);
   TORCH_CHECK(row % 8 == 0);
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(k.stride(-1) == 1 && v.stride(-1) == 1 &&
+              k.stride(1) == k.size(2) && v.stride(1) == v.size(2),
+              "reshape_and_cache: per-token row must be contiguous");
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
   TORCH_CHECK(k.scalar_type() == at::kBFloat16);
@@ -162,6 +166,7 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v,
                      (const __hip_bfloat16 *)v.data_ptr(),
                      (__hip_bfloat16 *)k_cache.data_ptr(),
                      (__hip_bfloat16 *)v_cache.data_ptr(),
-                     slot_mapping.data_ptr<long>(), row, page_size);
+                     slot_mapping.data_ptr<long>(), row, page_size,
+                     k.stride(0), v.stride(0));
   HIP_CHECK_KERNEL();
 }

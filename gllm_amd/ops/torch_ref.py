@@ -48,7 +48,7 @@ def rotary_embedding(positions: torch.Tensor, q: torch.Tensor,
     sin = cs[:, half:]
     for t in (q, k):
         T = t.shape[0]
-        x = t.view(T, -1, head_dim)
+        x = t.unflatten(-1, (-1, head_dim))
         rot = x[..., :rot_dim].float()
         if is_neox:
             x1, x2 = rot[..., :half], rot[..., half:]
