@@ -103,7 +103,7 @@ def main():
         ids = torch.randint(1, vocab - 1, (args.prompt_len,),
                             generator=g).tolist()
         sp = SamplingParams(temperature=0.0, ignore_eos=True,
-                            max_tokens=10 ** 9)
+                            max_tokens=4096)
         s = Sequence(i, ids, sp, eos_token_id=None, arrival_time=t_submit)
         seqs.append(s)
     eng.add_requests(seqs)

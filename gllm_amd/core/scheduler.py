@@ -147,6 +147,10 @@ class Scheduler:
         if not seq.computed_prompt and not self._in_flight(seq):
             return 0
         remaining = seq.output_len - seq.num_output_tokens
+        # clamp by the context cap: a request asking for 1e9 tokens must not
+        # reserve more KV than the model could ever hold
+        cap = (self.config.model_max_length or 32768) - len(seq.token_ids)
+        remaining = min(remaining, max(0, cap))
         if remaining <= 0:
             return 0
         projected = self.new_token_ratio * remaining
