@@ -1,32 +1,158 @@
-"""hipGraph-captured decode buckets (reference: CUDA-graph capture at
-model_runner.py:1525-1615).
+"""hipGraph-captured decode buckets.
 
-Captures the full decode forward (embed -> layers -> logits -> sample
-metadata-independent part) for power-of-2 batch buckets into hipGraphs
-replayed from persistent input buffers. Round-1: implemented after the
-eager GPU path is validated; ``can_replay`` returns False until capture
-runs.
+Parity: reference CUDA-graph capture (model_runner.py:1525-1615) with
+the dummy-page padding trick (input_data.py:611-671), re-done on HIP:
+torch.cuda.CUDAGraph IS hipGraph on ROCm. Pure-decode batches replay a
+captured graph for the smallest bucket >= B; everything else runs eager.
+
+Capture covers embed -> layers -> final norm (hidden out). Logits GEMM
+and sampling stay eager (cheap, metadata-dependent). All graphs share
+one memory pool; inputs live in persistent device buffers fed from
+persistent pinned staging.
 """
 
-from typing import Dict, List
+from typing import Dict, List, Optional
 
+import numpy as np
 import torch
 
-DECODE_BUCKETS = [1, 2, 4, 8, 16, 32, 48, 64, 96, 128, 192, 256]
+from gllm_amd.logger import logger
+from gllm_amd.runtime.forward_context import ForwardContext
+
+DECODE_BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256, 384, 512]
 
 
 class GraphRunner:
     def __init__(self, runner):
         self.runner = runner
+        cfg = runner.config
+        self.device = cfg.device
+        self.buckets = [b for b in DECODE_BUCKETS if b <= cfg.max_graph_bs]
         self.graphs: Dict[int, torch.cuda.CUDAGraph] = {}
+        self.hidden_out: Dict[int, torch.Tensor] = {}
         self.captured = False
+        self.max_bs = self.buckets[-1]
+        max_len = cfg.model_max_length or 32768
+        self.pages_cap = -(-max_len // cfg.page_size)
+        # the reserved scratch page (never allocated to a real seq)
+        self.dummy_page = runner.num_kv_pages_total - 1
 
+        B = self.max_bs
+        dev = self.device
+        self.in_ids = torch.zeros(B, dtype=torch.long, device=dev)
+        self.positions = torch.zeros(B, dtype=torch.long, device=dev)
+        self.slots = torch.zeros(B, dtype=torch.long, device=dev)
+        self.block_table = torch.zeros((B, self.pages_cap),
+                                       dtype=torch.int32, device=dev)
+        self.seq_lens = torch.ones(B, dtype=torch.int32, device=dev)
+        self.qsl = torch.arange(B + 1, dtype=torch.int32, device=dev)
+        # pinned staging (one flat int64 area reused per copy)
+        self.pin_ids = torch.zeros(B, dtype=torch.long).pin_memory()
+        self.pin_pos = torch.zeros(B, dtype=torch.long).pin_memory()
+        self.pin_slots = torch.zeros(B, dtype=torch.long).pin_memory()
+        self.pin_seq_lens = torch.zeros(B, dtype=torch.int32).pin_memory()
+        self.pin_bt = torch.zeros((B, self.pages_cap),
+                                  dtype=torch.int32).pin_memory()
+
+    def _fctx_for(self, bs: int) -> ForwardContext:
+        return ForwardContext(
+            num_tokens=bs,
+            positions=self.positions[:bs],
+            slot_mapping=self.slots[:bs],
+            block_table=self.block_table[:bs],
+            seq_lens=self.seq_lens[:bs],
+            query_start_loc=self.qsl[:bs + 1],
+            max_query_len=1, max_seq_len=1,
+            k_caches=self.runner.k_caches,
+            v_caches=self.runner.v_caches)
+
+    @torch.no_grad()
     def capture_all(self):
-        # Implemented in the hipGraph pass (after eager GPU validation).
-        self.captured = False
+        runner = self.runner
+        # dummy metadata: every row points at the scratch page
+        self.block_table.fill_(0)
+        self.block_table[:, 0] = self.dummy_page
+        self.seq_lens.fill_(1)
+        base = self.dummy_page * runner.config.page_size
+        ps = runner.config.page_size
+        self.slots.copy_(base + torch.arange(self.max_bs) % ps)
+        pool = torch.cuda.graphs.graph_pool_handle()
+        stream = torch.cuda.Stream()
+        torch.cuda.synchronize()
+        # capture largest first so the shared pool is sized once
+        for bs in reversed(self.buckets):
+            fctx = self._fctx_for(bs)
+            with torch.cuda.stream(stream):
+                for _ in range(2):  # warmup outside capture
+                    runner.model(self.in_ids[:bs], self.positions[:bs], fctx)
+            torch.cuda.current_stream().wait_stream(stream)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, pool=pool, stream=stream):
+                hidden, _ = runner.model(self.in_ids[:bs],
+                                         self.positions[:bs], fctx)
+            self.graphs[bs] = g
+            self.hidden_out[bs] = hidden
+        torch.cuda.synchronize()
+        self.captured = True
+        logger.info("hipGraph capture done: %d decode buckets (max bs %d)",
+                    len(self.graphs), self.max_bs)
 
-    def can_replay(self, batch_size: int) -> bool:
-        return False
+    # ------------------------------------------------------------------
+    def can_replay(self, batch) -> bool:
+        if not self.captured:
+            return False
+        items = batch.items
+        if len(items) > self.max_bs:
+            return False
+        return all(it.num_tokens == 1 and it.start >= it.seq.prompt_len
+                   for it in items)
 
-    def replay(self, batch, tokens, fctx):  # pragma: no cover
-        raise NotImplementedError
+    def _bucket(self, n: int) -> int:
+        for b in self.buckets:
+            if b >= n:
+                return b
+        raise AssertionError
+
+    @torch.no_grad()
+    def replay(self, batch):
+        items = batch.items
+        B = len(items)
+        bs = self._bucket(B)
+        ps = self.runner.config.page_size
+        ids = self.pin_ids.numpy()
+        pos = self.pin_pos.numpy()
+        slots = self.pin_slots.numpy()
+        seq_lens = self.pin_seq_lens.numpy()
+        bt = self.pin_bt.numpy()
+        max_pages_used = 0
+        for i, it in enumerate(items):
+            seq = it.seq
+            ids[i] = seq.token_ids[it.start]
+            pos[i] = it.start
+            page_tab = seq.page_table
+            n_pages = len(page_tab)
+            bt[i, :n_pages] = page_tab
+            max_pages_used = max(max_pages_used, n_pages)
+            slots[i] = page_tab[it.start // ps] * ps + it.start % ps
+            seq_lens[i] = it.start + 1
+        dummy_base = self.dummy_page * ps
+        for i in range(B, bs):
+            ids[i] = 0
+            pos[i] = 0
+            bt[i, 0] = self.dummy_page
+            slots[i] = dummy_base + i % ps
+            seq_lens[i] = 1
+        # H2D into the captured buffers
+        self.in_ids[:bs].copy_(self.pin_ids[:bs], non_blocking=True)
+        self.positions[:bs].copy_(self.pin_pos[:bs], non_blocking=True)
+        self.slots[:bs].copy_(self.pin_slots[:bs], non_blocking=True)
+        self.seq_lens[:bs].copy_(self.pin_seq_lens[:bs], non_blocking=True)
+        npg = max(1, max_pages_used)
+        self.block_table[:bs, :npg].copy_(self.pin_bt[:bs, :npg],
+                                          non_blocking=True)
+        self.graphs[bs].replay()
+        hidden = self.hidden_out[bs][:B]
+        fctx = self._fctx_for(bs)
+        fctx.logits_indices = None  # hidden already one row per seq
+        return self.runner._sample(batch, hidden, fctx)

@@ -45,11 +45,16 @@ class ModelRunner:
         self.kv_dtype = cfg.torch_dtype()
         num_pages = num_pages_override or self._size_kv_cache()
         self._allocate_kv(num_pages)
+        self.num_kv_pages_total = num_pages
         mgr_cls = PrefixMemoryManager if cfg.enable_prefix_caching \
             else MemoryManager
-        self.memory_manager = mgr_cls(num_pages, cfg.page_size)
+        # the LAST page is reserved as the hipGraph dummy/scratch page
+        # (reference input_data.py:611-671 dummy-page padding trick)
+        self.memory_manager = mgr_cls(num_pages - 1, cfg.page_size)
         self.builder = BatchBuilder(cfg.page_size, cfg.device)
-        if cfg.use_graph and cfg.device.startswith("cuda"):
+        from gllm_amd.parallel import get_pp_size, get_tp_size
+        if (cfg.use_graph and cfg.device.startswith("cuda")
+                and get_pp_size() == 1 and get_tp_size() == 1):
             from gllm_amd.runtime.graph_runner import GraphRunner
             self.graph_runner = GraphRunner(self)
             self.graph_runner.capture_all()
@@ -139,13 +144,12 @@ class ModelRunner:
     def step_first_stage(self, batch: ScheduledBatch):
         """Stage-0 forward. PP=1: returns sampled tokens. PP>1: returns
         (hidden, residual, fctx) for the PP send."""
+        if self.graph_runner is not None and \
+                self.graph_runner.can_replay(batch):
+            return self.graph_runner.replay(batch)
         tokens, fctx = self.builder.build(
             batch, self.k_caches, self.v_caches,
             need_logits=self.model.is_last_stage)
-        if (self.graph_runner is not None and self.model.is_last_stage
-                and fctx.max_query_len == 1
-                and self.graph_runner.can_replay(len(batch.items))):
-            return self.graph_runner.replay(batch, tokens, fctx)
         hidden, residual = self._stage_forward(tokens, fctx.positions, fctx)
         if self.model.is_last_stage:
             return self._sample(batch, hidden, fctx)
