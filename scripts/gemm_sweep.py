@@ -1,0 +1,58 @@
+"""Decode/prefill GEMM shape sweep: hipBLASLt default vs TunableOp.
+
+Shapes are Qwen2.5-32B projections. Run on an MI355X:
+    python scripts/gemm_sweep.py             # default heuristics
+    PYTORCH_TUNABLEOP_ENABLED=1 PYTORCH_TUNABLEOP_TUNING=1 \
+        python scripts/gemm_sweep.py         # online-tuned
+Prints effective TB/s of weight streaming per shape (the decode-regime
+bound) and TF/s.
+"""
+
+import os
+import sys
+import time
+
+import torch
+import torch.nn.functional as F
+
+SHAPES = [
+    # (name, N out, K in)
+    ("qkv", 7168, 5120),
+    ("o", 5120, 5120),
+    ("gate_up", 55296, 5120),
+    ("down", 5120, 27648),
+    ("lm_head", 152064, 5120),
+]
+MS = [32, 64, 128, 256, 512, 8192]
+
+
+def bench_shape(M, N, K, iters=20):
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+    for _ in range(3):
+        F.linear(x, w)
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(iters):
+        F.linear(x, w)
+    torch.cuda.synchronize()
+    dt = (time.time() - t0) / iters
+    bytes_w = 2.0 * N * K
+    tbps = bytes_w / dt / 1e12
+    tf = 2.0 * M * N * K / dt / 1e12
+    return dt * 1e6, tbps, tf
+
+
+def main():
+    print(f"TunableOp={os.environ.get('PYTORCH_TUNABLEOP_ENABLED', '0')}")
+    for M in MS:
+        for name, N, K in SHAPES:
+            if M == 8192 and name == "lm_head":
+                continue
+            us, tbps, tf = bench_shape(M, N, K)
+            print(f"M={M:5d} {name:8s} [{N:6d}x{K:5d}] {us:9.1f} us  "
+                  f"W-stream {tbps:5.2f} TB/s  {tf:7.1f} TF/s")
+
+
+if __name__ == "__main__":
+    main()
