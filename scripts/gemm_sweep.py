@@ -56,3 +56,27 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def bench_shape_nn(M, N, K, iters=20):
+    """x [M,K] @ W [K,N] (pre-transposed weight, NN layout)."""
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(K, N, dtype=torch.bfloat16, device="cuda")
+    for _ in range(3):
+        torch.matmul(x, w)
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(iters):
+        torch.matmul(x, w)
+    torch.cuda.synchronize()
+    dt = (time.time() - t0) / iters
+    return dt * 1e6, 2.0 * N * K / dt / 1e12, 2.0 * M * N * K / dt / 1e12
+
+
+if os.environ.get("SWEEP_NN"):
+    print("=== NN layout (x @ W[K,N]) ===")
+    for M in [32, 64, 128, 256]:
+        for name, N, K in SHAPES:
+            us, tbps, tf = bench_shape_nn(M, N, K)
+            print(f"NN M={M:5d} {name:8s} [{K:5d}x{N:6d}] {us:9.1f} us  "
+                  f"W-stream {tbps:5.2f} TB/s  {tf:7.1f} TF/s")
