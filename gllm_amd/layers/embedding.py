@@ -75,7 +75,8 @@ class ParallelLMHead(nn.Module):
         self.weight = embedding.weight
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
-        logits = F.linear(hidden, self.weight)
+        from gllm_amd import ops
+        logits = ops.linear(hidden, self.weight)
         if get_tp_size() > 1:
             logits = tensor_parallel_all_gather(logits, dim=-1)
         return logits[..., :self.vocab_size]

@@ -13,6 +13,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from gllm_amd import ops
 from gllm_amd.parallel import (get_tp_rank, get_tp_size,
                                tensor_parallel_all_reduce)
 
@@ -52,7 +53,7 @@ class ReplicatedLinear(LinearBase):
         param.data.copy_(loaded)
 
     def forward(self, x):
-        return F.linear(x, self.weight, self.bias)
+        return ops.linear(x, self.weight, self.bias)
 
 
 class ColumnParallelLinear(LinearBase):
@@ -74,7 +75,7 @@ class ColumnParallelLinear(LinearBase):
         _narrow_copy(param.data, loaded, 0, get_tp_rank(), get_tp_size())
 
     def forward(self, x):
-        out = F.linear(x, self.weight, self.bias)
+        out = ops.linear(x, self.weight, self.bias)
         if self.gather_output:
             from gllm_amd.parallel import tensor_parallel_all_gather
             out = tensor_parallel_all_gather(out, dim=-1)
@@ -106,7 +107,7 @@ class MergedColumnParallelLinear(LinearBase):
             loaded.narrow(0, tp_rank * size, size))
 
     def forward(self, x):
-        return F.linear(x, self.weight, self.bias)
+        return ops.linear(x, self.weight, self.bias)
 
 
 class QKVParallelLinear(LinearBase):
@@ -151,7 +152,7 @@ class QKVParallelLinear(LinearBase):
             loaded.narrow(0, src_rank * size, size))
 
     def forward(self, x):
-        out = F.linear(x, self.weight, self.bias)
+        out = ops.linear(x, self.weight, self.bias)
         return out.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
 
 
@@ -178,7 +179,7 @@ class RowParallelLinear(LinearBase):
         param.data.copy_(loaded)
 
     def forward(self, x):
-        out = F.linear(x, self.weight)
+        out = ops.linear(x, self.weight)
         if self.reduce_results:
             out = tensor_parallel_all_reduce(out)
         # bias applied once, after the reduction

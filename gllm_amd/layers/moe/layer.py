@@ -103,8 +103,8 @@ class FusedMoE(nn.Module):
                 continue
             rows = flat_rows[sel]
             xe = x.index_select(0, rows)
-            h = ops.silu_and_mul(F.linear(xe, self.w13_weight[lid]))
-            ye = F.linear(h, self.w2_weight[lid])
+            h = ops.silu_and_mul(ops.linear(xe, self.w13_weight[lid]))
+            ye = ops.linear(h, self.w2_weight[lid])
             w = weights.flatten()[sel].unsqueeze(-1)
             out.index_add_(0, rows, ye * w)
         if self.use_ep:

@@ -125,3 +125,48 @@ def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
 # --------------------------------------------------------------- moe
 def topk_softmax(gating: torch.Tensor, topk: int, renormalize: bool = True):
     return torch_ref.topk_softmax(gating, topk, renormalize)
+
+
+# --------------------------------------------------------------- gemm
+_SKINNY_WS: dict = {}
+SKINNY_MAX_M = 256
+
+
+def _skinny_splitk(M: int, N: int, K: int) -> int:
+    n_wg = -(-N // 64) * -(-M // 64)
+    s = 1
+    while s < 16 and n_wg * s < 512 and (K // (s * 2)) >= 64:
+        s *= 2
+    k_slice = ((-(-K // s)) + 63) // 64 * 64
+    return -(-K // k_slice)
+
+
+def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
+                bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out = x @ w.T via the gfx950 weight-streaming kernel (M<=256-ish)."""
+    M, K = x.shape
+    N = w.shape[0]
+    splitk = _skinny_splitk(M, N, K)
+    need = splitk * M * N
+    key = x.device.index or 0
+    ws = _SKINNY_WS.get(key)
+    if ws is None or ws.numel() < need:
+        ws = torch.empty(need, dtype=torch.float32, device=x.device)
+        _SKINNY_WS[key] = ws
+    out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    _gpu_kernels().skinny_gemm(out, x, w, None, ws, splitk)
+    if bias is not None:
+        out += bias
+    return out
+
+
+def linear(x: torch.Tensor, w: torch.Tensor,
+           bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """GEMM dispatch: the skinny weight-streaming kernel for decode-size
+    M on GPU, hipBLASLt otherwise."""
+    import torch.nn.functional as F
+    if (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 2
+            and x.is_contiguous() and 0 < x.shape[0] <= SKINNY_MAX_M
+            and w.shape[0] >= 1024 and w.shape[1] % 64 == 0):
+        return skinny_gemm(x, w, bias)
+    return F.linear(x, w, bias)

@@ -16,6 +16,9 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_table, torch::Tensor seq_lens,
                             double scale, long sliding_window);
+void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                 c10::optional<torch::Tensor> bias,
+                 torch::Tensor workspace, long splitk);
 void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor k_cache, torch::Tensor v_cache,
                              torch::Tensor block_table,
@@ -36,4 +39,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "split-KV paged decode attention");
   m.def("paged_attention_prefill", &paged_attention_prefill,
         "varlen MFMA paged prefill attention");
+  m.def("skinny_gemm", &skinny_gemm,
+        "decode-regime weight-streaming GEMM (M<=256)");
 }

@@ -80,3 +80,22 @@ if os.environ.get("SWEEP_NN"):
             us, tbps, tf = bench_shape_nn(M, N, K)
             print(f"NN M={M:5d} {name:8s} [{K:5d}x{N:6d}] {us:9.1f} us  "
                   f"W-stream {tbps:5.2f} TB/s  {tf:7.1f} TF/s")
+
+
+if os.environ.get("SWEEP_SKINNY"):
+    from gllm_amd import ops
+    print("=== skinny_gemm ===")
+    for M in [32, 64, 128, 256]:
+        for name, N, K in SHAPES:
+            x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+            w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+            for _ in range(3):
+                ops.skinny_gemm(x, w)
+            torch.cuda.synchronize()
+            t0 = time.time()
+            for _ in range(20):
+                ops.skinny_gemm(x, w)
+            torch.cuda.synchronize()
+            dt = (time.time() - t0) / 20
+            print(f"SK M={M:5d} {name:8s} {dt*1e6:9.1f} us  "
+                  f"W-stream {2.0*N*K/dt/1e12:5.2f} TB/s")
