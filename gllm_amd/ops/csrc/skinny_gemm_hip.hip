@@ -3,21 +3,24 @@
 // bf16 in / fp32 accumulate / bf16 out, M <= 256.
 //
 // Why it exists: at decode batch sizes hipBLASLt reaches only 1-3.8 TB/s
-// of weight streaming on Qwen-32B projection shapes (profiles/
-// r01_qwen32b_bench_kernels.md; scripts/gemm_sweep.py) while the chip
-// sustains ~6.3 TB/s. In this regime the GEMM is a pure weight stream:
-// the kernel's only job is to read W once at full bandwidth.
+// of weight streaming on Qwen-32B projection shapes while the chip
+// sustains ~6.3 TB/s (scripts/gemm_sweep.py, profiles/). The GEMM is a
+// pure weight stream; the kernel's job is to read W once at full rate.
 //
-// Structure: grid = (N/BN, ceil(M/BM), SPLITK), 256 threads (4 waves).
-//   Each workgroup owns a BN=64-column x BM=64-row output tile and a
-//   contiguous K-slice. W tile [BN][BK] and x tile [BM][BK] are staged in
-//   LDS (XOR-swizzled rows for conflict-free ds_read_b128), MFMA
-//   16x16x32 per wave (wave = one 16-row M-tile x all 64 N-cols).
-//   Register-staged T14 pipeline: next tile's global loads issue before
-//   the compute phase, LDS writes land after the barrier.
-//   Split-K partials are fp32 [SPLITK, M, N]; a reduce kernel sums them
-//   (+bias) to bf16 — the slab round trip costs <<10% of W traffic at
-//   these shapes.
+// v2 structure (guide T3+T4: glds ring with counted vmcnt):
+//   grid = (N/BN, SPLITK), block = 256 (4 waves). One workgroup owns a
+//   BN=64-column tile for ALL M rows (MB 64-row blocks) and a contiguous
+//   K-slice, so W is streamed exactly once regardless of M.
+//   Per K-step, a tile { W[64][64] | x[MB*64][64] } is DMA'd straight
+//   into an LDS ring via __builtin_amdgcn_global_load_lds (16 B/lane),
+//   with the T2 XOR swizzle applied on the per-lane SOURCE address
+//   (glds writes lane-linear; rule 21). The ring keeps 2 tiles in
+//   flight across raw s_barriers with counted s_waitcnt vmcnt(N) — the
+//   +40..83% pattern of cdna_hip_programming.md §5 'Pipelining across
+//   barriers'. MFMA 16x16x32: wave w computes rows of every M-block
+//   (4 C-frags per block), B-fragments straight from the swizzled W
+//   image.
+//   Split-K partials fp32 [SPLITK, M, N]; reduce kernel sums + bias.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -29,24 +32,36 @@ typedef __attribute__((ext_vector_type(4))) float sg_f4;
 namespace {
 
 constexpr int BLOCK = 256;
-constexpr int BM = 64;
 constexpr int BN = 64;
 constexpr int BK = 64;
+constexpr int ROW_B = BK * 2;           // tile row bytes (128)
 
 DEV_INLINE int swz(int row, int byte_off) {
-  return byte_off ^ ((row & 7) << 4);  // BK*2=128-B rows: 8-slot spread
+  // 8 16-B slots per 128-B row; spread the 16-lane b128 column read
+  return byte_off ^ ((row & 7) << 4);
 }
 
+// One glds instruction moves 64 lanes x 16 B = 1 KiB, lane-linear at
+// lds_base + lane*16. The tile image is linear rows of 128 B; the
+// source address carries the inverse swizzle.
+DEV_INLINE void glds16(const __hip_bfloat16 *gsrc, char *lds_ptr) {
+  __builtin_amdgcn_global_load_lds(
+      reinterpret_cast<const unsigned int *>(gsrc),
+      reinterpret_cast<unsigned int *>(lds_ptr), 16, 0, 0);
+}
+
+template <int MB, int RING>
 __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
     float *__restrict__ partial,          // [SPLITK, M, N]
     const __hip_bfloat16 *__restrict__ x, // [M, K]
     const __hip_bfloat16 *__restrict__ w, // [N, K]
     int M, int N, int K, int k_slice) {
+  constexpr int TILE_B = (BN + MB * 64) * ROW_B;   // bytes per ring slot
   const int n0 = blockIdx.x * BN;
-  const int m0 = blockIdx.y * BM;
-  const int z = blockIdx.z;
+  const int z = blockIdx.y;
   const int k_begin = z * k_slice;
   const int k_end = min(K, k_begin + k_slice);
+  const int nkt = (k_end - k_begin + BK - 1) / BK;  // K divisible by BK
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -54,98 +69,104 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
   const int l16 = lane & 15;
   const int lhi = lane >> 4;
 
-  __shared__ __hip_bfloat16 w_tile[BN * BK];
-  __shared__ __hip_bfloat16 x_tile[BM * BK];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
 
-  // staging assignment: 256 threads x 32 B = 8 KB per tile.
-  // thread covers tile row r = tid/8 (x2 iters of 32 rows... BN=64 rows,
-  // 8 chunks of 16B per 128-B row): r = tid/8 + it*32, chunk c = tid%8.
-  const int srow = tid >> 3;          // 0..31
-  const int schunk = tid & 7;         // 16-B chunk within the row
-
-  sg_f4 acc[BN / 16];                 // 4 C fragments (16 rows x 64 cols)
+  // ---- glds source mapping -------------------------------------------
+  // Tile image: W rows [0,64) then x rows [0, MB*64), 128 B each, row
+  // r's bytes XOR-swizzled. glds j of wave w covers image bytes
+  // [(w*G + j)*1024, ...+1024), lane l -> byte p = base + l*16:
+  // row = p/128, col = (p%128) ^ swz(row).
+  constexpr int GL_PER_WAVE = TILE_B / 1024 / 4;   // glds per wave/tile
+  const __hip_bfloat16 *gsrc[GL_PER_WAVE];
+  {
+    const long wrow_stride = K;  // elements
 #pragma unroll
-  for (int i = 0; i < BN / 16; ++i) acc[i] = sg_f4{0, 0, 0, 0};
-
-  const int nkt = (k_end - k_begin + BK - 1) / BK;
-  shortx8 wreg[2], xreg[2];
-
-  auto load_tile = [&](int kt, shortx8 *wr, shortx8 *xr) {
-    const int kb = k_begin + kt * BK;
-#pragma unroll
-    for (int it = 0; it < 2; ++it) {
-      const int row = srow + it * 32;
-      const int kk = kb + schunk * 8;
-      // W row n0+row; pad rows read row 0 (results discarded via N guard)
-      const int wn = min(n0 + row, N - 1);
-      wr[it] = (kk + 8 <= k_end)
-                   ? *reinterpret_cast<const shortx8 *>(
-                         w + (long)wn * K + kk)
-                   : shortx8{0, 0, 0, 0, 0, 0, 0, 0};
-      const int xm = m0 + row;
-      xr[it] = (xm < M && kk + 8 <= k_end)
-                   ? *reinterpret_cast<const shortx8 *>(
-                         x + (long)xm * K + kk)
-                   : shortx8{0, 0, 0, 0, 0, 0, 0, 0};
-    }
-  };
-
-  auto store_tile = [&](const shortx8 *wr, const shortx8 *xr) {
-#pragma unroll
-    for (int it = 0; it < 2; ++it) {
-      const int row = srow + it * 32;
-      *reinterpret_cast<shortx8 *>(
-          reinterpret_cast<char *>(&w_tile[row * BK]) +
-          swz(row, schunk * 16)) = wr[it];
-      *reinterpret_cast<shortx8 *>(
-          reinterpret_cast<char *>(&x_tile[row * BK]) +
-          swz(row, schunk * 16)) = xr[it];
-    }
-  };
-
-  // prologue: tile 0
-  load_tile(0, wreg, xreg);
-  store_tile(wreg, xreg);
-  __syncthreads();
-
-  for (int kt = 0; kt < nkt; ++kt) {
-    // issue next tile's loads before computing (T14 issue-early)
-    if (kt + 1 < nkt) load_tile(kt + 1, wreg, xreg);
-
-    // compute from LDS tile kt
-#pragma unroll
-    for (int ks = 0; ks < BK / 32; ++ks) {
-      // A fragment: x rows m = wave*16 + l16, k = ks*32 + lhi*8
-      const int arow = wave * 16 + l16;
-      sg_bf8 afrag = *reinterpret_cast<const sg_bf8 *>(
-          reinterpret_cast<const char *>(&x_tile[arow * BK]) +
-          swz(arow, (ks * 32 + lhi * 8) * 2));
-#pragma unroll
-      for (int nt = 0; nt < BN / 16; ++nt) {
-        const int brow = nt * 16 + l16;   // W row (output col)
-        sg_bf8 bfrag = *reinterpret_cast<const sg_bf8 *>(
-            reinterpret_cast<const char *>(&w_tile[brow * BK]) +
-            swz(brow, (ks * 32 + lhi * 8) * 2));
-        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag, bfrag, acc[nt], 0, 0, 0);
+    for (int j = 0; j < GL_PER_WAVE; ++j) {
+      const int p = (wave * GL_PER_WAVE + j) * 1024 + lane * 16;
+      const int row = p / ROW_B;
+      const int col = swz(row, p % ROW_B);         // involution
+      if (row < BN) {
+        const int n = min(n0 + row, N - 1);
+        gsrc[j] = w + (long)n * wrow_stride + col / 2;
+      } else {
+        const int m = min(row - BN, M - 1);
+        gsrc[j] = x + (long)m * K + col / 2;
       }
-    }
-    __syncthreads();
-    if (kt + 1 < nkt) {
-      store_tile(wreg, xreg);
-      __syncthreads();
     }
   }
 
-  // epilogue: fp32 partial
+  auto stage = [&](int kt, int slot) {
+    const int kb = k_begin + kt * BK;
+    char *base = smem + slot * TILE_B;
+#pragma unroll
+    for (int j = 0; j < GL_PER_WAVE; ++j)
+      glds16(gsrc[j] + kb, base + (wave * GL_PER_WAVE + j) * 1024);
+  };
+
+  // ---- accumulators ---------------------------------------------------
+  sg_f4 acc[MB][BN / 16];
+#pragma unroll
+  for (int mb = 0; mb < MB; ++mb)
+#pragma unroll
+    for (int nt = 0; nt < BN / 16; ++nt) acc[mb][nt] = sg_f4{0, 0, 0, 0};
+
+  // ---- prologue: fill the ring ---------------------------------------
+  const int pre = min(RING - 1, nkt);
+  for (int t = 0; t < pre; ++t) stage(t, t % RING);
+
+  for (int kt = 0; kt < nkt; ++kt) {
+    const int slot = kt % RING;
+    if (kt + RING - 1 < nkt) stage(kt + RING - 1, (kt + RING - 1) % RING);
+    // wait until tile kt's glds landed: outstanding = this wave's glds
+    // for tiles (kt+1 .. kt+RING-1) may stay in flight.
+    if (RING == 3) {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * GL_PER_WAVE)
+                   : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(1 * GL_PER_WAVE)
+                   : "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+    const char *wbase = smem + slot * TILE_B;
+    const char *xbase = wbase + BN * ROW_B;
+#pragma unroll
+    for (int ks = 0; ks < BK / 32; ++ks) {
+      // B fragments: lane holds W[n = nt*16 + l16][k = ks*32 + lhi*8..]
+      sg_bf8 bfrag[BN / 16];
+#pragma unroll
+      for (int nt = 0; nt < BN / 16; ++nt) {
+        const int brow = nt * 16 + l16;
+        bfrag[nt] = *reinterpret_cast<const sg_bf8 *>(
+            wbase + brow * ROW_B + swz(brow, (ks * 32 + lhi * 8) * 2));
+      }
+#pragma unroll
+      for (int mb = 0; mb < MB; ++mb) {
+        const int arow = mb * 64 + wave * 16 + l16;
+        sg_bf8 afrag = *reinterpret_cast<const sg_bf8 *>(
+            xbase + arow * ROW_B + swz(arow, (ks * 32 + lhi * 8) * 2));
+#pragma unroll
+        for (int nt = 0; nt < BN / 16; ++nt)
+          acc[mb][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, bfrag[nt], acc[mb][nt], 0, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // ---- epilogue -------------------------------------------------------
   float *base = partial + (long)z * M * N;
 #pragma unroll
-  for (int nt = 0; nt < BN / 16; ++nt) {
+  for (int mb = 0; mb < MB; ++mb) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = m0 + wave * 16 + lhi * 4 + r;
-      const int n = n0 + nt * 16 + l16;
-      if (m < M && n < N) base[(long)m * N + n] = acc[nt][r];
+    for (int nt = 0; nt < BN / 16; ++nt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = mb * 64 + wave * 16 + lhi * 4 + r;
+        const int n = n0 + nt * 16 + l16;
+        if (m < M && n < N) base[(long)m * N + n] = acc[mb][nt][r];
+      }
     }
   }
 }
@@ -163,6 +184,17 @@ __global__ void skinny_reduce_kernel(
   }
 }
 
+template <int MB, int RING>
+void launch_skinny(float *partial, const __hip_bfloat16 *x,
+                   const __hip_bfloat16 *w, int M, int N, int K, int k_slice,
+                   int splitk, hipStream_t stream) {
+  constexpr int TILE_B = (BN + MB * 64) * 2 * BK;
+  const int lds = RING * TILE_B;
+  hipLaunchKernelGGL((skinny_gemm_kernel<MB, RING>),
+                     dim3((N + BN - 1) / BN, splitk), dim3(BLOCK), lds,
+                     stream, partial, x, w, M, N, K, k_slice);
+}
+
 }  // namespace
 
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
@@ -172,13 +204,13 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
               w.scalar_type() == at::kBFloat16);
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
-  TORCH_CHECK(K % 8 == 0);
-  const int n_wg = ((N + BN - 1) / BN) * ((M + BM - 1) / BM);
+  TORCH_CHECK(K % BK == 0, "skinny_gemm: K must be a multiple of 64");
+  TORCH_CHECK(M <= 256, "skinny_gemm: M <= 256");
+  const int n_wg = (N + BN - 1) / BN;
   int splitk = (int)splitk_arg;
   if (splitk <= 0) {
     splitk = 1;
-    while (splitk < 16 && n_wg * splitk < 512 &&
-           (K / (splitk * 2)) >= BK)
+    while (splitk < 16 && n_wg * splitk < 512 && (K / (splitk * 2)) >= BK)
       splitk *= 2;
   }
   int k_slice = (K + splitk - 1) / splitk;
@@ -188,13 +220,17 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(workspace.numel() >= (long)splitk * M * N,
               "skinny_gemm workspace too small");
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  hipLaunchKernelGGL(skinny_gemm_kernel,
-                     dim3((N + BN - 1) / BN, (M + BM - 1) / BM, splitk),
-                     dim3(BLOCK), 0, stream,
-                     workspace.data_ptr<float>(),
-                     (const __hip_bfloat16 *)x.data_ptr(),
-                     (const __hip_bfloat16 *)w.data_ptr(), M, N, K, k_slice);
+  auto *ws = workspace.data_ptr<float>();
+  auto *xp = (const __hip_bfloat16 *)x.data_ptr();
+  auto *wp = (const __hip_bfloat16 *)w.data_ptr();
+  if (M <= 64)
+    launch_skinny<1, 3>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
+  else if (M <= 128)
+    launch_skinny<2, 3>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
+  else
+    launch_skinny<4, 2>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
   HIP_CHECK_KERNEL();
+
   const float *bias_ptr = nullptr;
   if (bias.has_value()) {
     TORCH_CHECK(bias->scalar_type() == at::kFloat);
@@ -205,7 +241,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const long grid = std::min<long>((total + block * 4 - 1) / (block * 4),
                                    2048);
   hipLaunchKernelGGL(skinny_reduce_kernel, dim3(grid), dim3(block), 0,
-                     stream, (__hip_bfloat16 *)out.data_ptr(),
-                     workspace.data_ptr<float>(), bias_ptr, M, N, splitk);
+                     stream, (__hip_bfloat16 *)out.data_ptr(), ws, bias_ptr,
+                     M, N, splitk);
   HIP_CHECK_KERNEL();
 }
