@@ -117,14 +117,20 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
   for (int kt = 0; kt < nkt; ++kt) {
     const int slot = kt % RING;
     if (kt + RING - 1 < nkt) stage(kt + RING - 1, (kt + RING - 1) % RING);
-    // wait until tile kt's glds landed: outstanding = this wave's glds
-    // for tiles (kt+1 .. kt+RING-1) may stay in flight.
-    if (RING == 3) {
+    // Wait until tile kt's glds landed. vmcnt must equal the loads still
+    // allowed in flight = GL_PER_WAVE x (tiles staged BEYOND kt). Near
+    // the slice tail fewer tiles are ahead, so the count shrinks — a
+    // fixed steady-state count would let the wait pass while tile kt's
+    // own DMA is still flying (stale LDS reads).
+    const int ahead = min(nkt - 1 - kt, RING - 1);
+    if (RING == 3 && ahead == 2) {
       asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * GL_PER_WAVE)
                    : "memory");
-    } else {
+    } else if (ahead == 1) {
       asm volatile("s_waitcnt vmcnt(%0)" ::"i"(1 * GL_PER_WAVE)
                    : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
     __builtin_amdgcn_s_barrier();
 

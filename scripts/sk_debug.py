@@ -1,0 +1,24 @@
+import os, sys
+sys.path.insert(0, "/root/repo")
+import torch
+from gllm_amd import ops
+torch.manual_seed(0)
+for (M,N,K) in [(1,7168,5120),(17,5120,5120),(64,64,128),(64,64,64),(16,64,192),(64,5120,27648),(256,152064,5120)]:
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.1
+    out = ops.skinny_gemm(x, w).float()
+    ref = x.float() @ w.float().T
+    diff = (out - ref).abs()
+    rel = (diff / ref.abs().clamp_min(1.0))
+    bad = rel > 3e-2
+    print(f"M={M} N={N} K={K}: maxrel={float(rel.max()):.4f} frac_bad={float(bad.float().mean()):.4f}")
+    if bad.any():
+        bi = bad.nonzero()[:8]
+        print("  sample bad idx:", bi.tolist())
+        r, c = bi[0]
+        print(f"  out={out[r,c]:.4f} ref={ref[r,c]:.4f}")
+        # column pattern of badness
+        colbad = bad.any(0).nonzero().flatten()
+        rowbad = bad.any(1).nonzero().flatten()
+        print(f"  bad cols: {colbad[:16].tolist()}{'...' if colbad.numel()>16 else ''} ({colbad.numel()} total)")
+        print(f"  bad rows: {rowbad[:16].tolist()} ({rowbad.numel()} total)")
