@@ -1,0 +1,45 @@
+"""Incremental detokenization (reference: sequence.py:130-147).
+
+Standard two-cursor scheme: decode from ``prefix_offset`` and emit only
+the stable part, holding back text while the tail could still change
+(byte-fallback / incomplete UTF-8 / merged tokens)."""
+
+from typing import List, Optional, Tuple
+
+
+class IncrementalDetokenizer:
+    def __init__(self, tokenizer):
+        self.tok = tokenizer
+        self.token_ids: List[int] = []
+        self.prefix_offset = 0
+        self.read_offset = 0
+        self.text = ""
+
+    def append(self, token_id: int) -> str:
+        """Add one token; return newly stabilized text (may be '')."""
+        self.token_ids.append(token_id)
+        prefix_text = self.tok.decode(
+            self.token_ids[self.prefix_offset:self.read_offset],
+            skip_special_tokens=False)
+        new_text = self.tok.decode(self.token_ids[self.prefix_offset:],
+                                   skip_special_tokens=False)
+        if new_text.endswith("�"):
+            # incomplete byte sequence: hold back
+            return ""
+        delta = new_text[len(prefix_text):]
+        self.prefix_offset = self.read_offset
+        self.read_offset = len(self.token_ids)
+        self.text += delta
+        return delta
+
+
+def check_stop_strings(text: str, stops: Optional[List[str]]
+                       ) -> Tuple[bool, str]:
+    """Return (hit, truncated_text)."""
+    if not stops:
+        return False, text
+    for s in stops:
+        i = text.find(s)
+        if i >= 0:
+            return True, text[:i]
+    return False, text

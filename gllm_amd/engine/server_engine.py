@@ -1,0 +1,201 @@
+"""Frontend async engine: request intake, worker spawn, output streams.
+
+Parity: reference llm_engine.py (worker spawn, allocate_seq, schedule
+plumbing) + async_llm_engine.py (AsyncStream, abort-on-disconnect), on
+the replicated-scheduler worker design (engine/worker.py)."""
+
+import asyncio
+import dataclasses
+import threading
+import time
+from typing import AsyncIterator, Dict, List, Optional
+
+import torch.multiprocessing as mp
+
+from gllm_amd.config import EngineConfig
+from gllm_amd.engine.detokenizer import (IncrementalDetokenizer,
+                                         check_stop_strings)
+from gllm_amd.engine.ipc import FrontendComm
+from gllm_amd.logger import logger
+from gllm_amd.models.loader import load_hf_config
+from gllm_amd.sequence import SamplingParams
+from gllm_amd.utils.id_allocator import IDAllocator
+
+
+@dataclasses.dataclass
+class StreamChunk:
+    seq_id: int
+    token_id: int
+    text: str
+    finish_reason: Optional[str]
+    n_output_tokens: int
+
+
+class RequestState:
+    def __init__(self, seq_id: int, prompt_len: int, sampling: SamplingParams,
+                 tokenizer, loop):
+        self.seq_id = seq_id
+        self.prompt_len = prompt_len
+        self.sampling = sampling
+        self.detok = IncrementalDetokenizer(tokenizer) if tokenizer else None
+        self.queue: asyncio.Queue = asyncio.Queue()
+        self.loop = loop
+        self.n_tokens = 0
+        self.finished = False
+        self.created = time.time()
+        self.first_token_time: Optional[float] = None
+
+
+class AsyncLLMEngine:
+    """Owns the frontend side: tokenizer, worker processes, streams."""
+
+    def __init__(self, config: EngineConfig, base_port: int = 0):
+        # base_port kept for CLI compat; the queue transport ignores it
+        self.config = config
+        self.hf_config = load_hf_config(config.model)
+        self.tokenizer = self._load_tokenizer(config.model)
+        eos = getattr(self.hf_config, "eos_token_id", None)
+        if self.tokenizer is not None and self.tokenizer.eos_token_id is not None:
+            eos = self.tokenizer.eos_token_id
+        self.eos_token_id = eos[0] if isinstance(eos, list) else eos
+        self.comm = FrontendComm(config.world_size)
+        self.seq_ids = IDAllocator(1 << 20)
+        self.requests: Dict[int, RequestState] = {}
+        self._intake_lock = threading.Lock()
+        self._procs: List = []
+        self._ready = False
+        self._output_thread: Optional[threading.Thread] = None
+        self._stopping = False
+        self.loop: Optional[asyncio.AbstractEventLoop] = None
+
+    @staticmethod
+    def _load_tokenizer(model_path: str):
+        import os
+        if not any(os.path.exists(os.path.join(model_path, f))
+                   for f in ("tokenizer.json", "tokenizer.model",
+                             "tokenizer_config.json")):
+            return None
+        from transformers import AutoTokenizer
+        return AutoTokenizer.from_pretrained(model_path,
+                                             trust_remote_code=True)
+
+    # ------------------------------------------------------------------
+    def start(self) -> None:
+        ctx = mp.get_context("spawn")
+        ready_q = ctx.Queue()
+        from gllm_amd.engine.worker import run_worker
+        for r in range(self.config.world_size):
+            rq, oq = self.comm.worker_endpoints(r)
+            p = ctx.Process(target=run_worker,
+                            args=(r, self.config, rq, oq, ready_q),
+                            daemon=True)
+            p.start()
+            self._procs.append(p)
+        for _ in range(self.config.world_size):
+            status, rank = ready_q.get()  # blocks until model loaded
+            if status != "ready":
+                raise RuntimeError(f"worker {rank} failed to start")
+        self._ready = True
+        self._output_thread = threading.Thread(target=self._output_loop,
+                                               daemon=True)
+        self._output_thread.start()
+        logger.info("engine ready: %d worker(s)", self.config.world_size)
+
+    def stop(self) -> None:
+        self._stopping = True
+        try:
+            self.comm.send_to_all("cmd", "shutdown")
+        except Exception:
+            pass
+        for p in self._procs:
+            p.join(timeout=30)
+
+    # ------------------------------------------------------------------
+    def _output_loop(self) -> None:
+        while not self._stopping:
+            msg = self.comm.recv_output(timeout_ms=200)
+            if msg is None:
+                continue
+            kind, outs, _stats = msg
+            if kind != "out":
+                continue
+            for seq_id, token_id, finish in outs:
+                st = self.requests.get(seq_id)
+                if st is None or st.finished:
+                    continue
+                self._handle_token(st, token_id, finish)
+
+    def _handle_token(self, st: RequestState, token_id: int,
+                      finish: Optional[str]) -> None:
+        if st.first_token_time is None:
+            st.first_token_time = time.time()
+        text = ""
+        if token_id >= 0:
+            st.n_tokens += 1
+            if st.detok is not None:
+                text = st.detok.append(token_id)
+                hit, _trunc = check_stop_strings(st.detok.text,
+                                                 st.sampling.stop)
+                if hit and finish is None:
+                    finish = "stop"
+                    self.abort([st.seq_id])
+        if finish is not None:
+            st.finished = True
+        chunk = StreamChunk(st.seq_id, token_id, text, finish, st.n_tokens)
+        st.loop.call_soon_threadsafe(st.queue.put_nowait, chunk)
+        if st.finished:
+            st.loop.call_soon_threadsafe(st.queue.put_nowait, None)
+
+    # ------------------------------------------------------------------
+    def add_request(self, token_ids: List[int],
+                    sampling: SamplingParams) -> RequestState:
+        loop = asyncio.get_event_loop()
+        with self._intake_lock:
+            seq_id = self.seq_ids.allocate()
+            st = RequestState(seq_id, len(token_ids), sampling,
+                              self.tokenizer, loop)
+            self.requests[seq_id] = st
+            self.comm.send_to_all("req", {
+                "seq_id": seq_id,
+                "token_ids": token_ids,
+                "sampling": dataclasses.asdict(sampling),
+                "eos_token_id": self.eos_token_id,
+            })
+        return st
+
+    def abort(self, seq_ids: List[int]) -> None:
+        with self._intake_lock:
+            self.comm.send_to_all("abort", list(seq_ids))
+
+    def send_command(self, name: str) -> None:
+        with self._intake_lock:
+            self.comm.send_to_all("cmd", name)
+
+    def release(self, st: RequestState) -> None:
+        self.requests.pop(st.seq_id, None)
+        self.seq_ids.free(st.seq_id)
+
+    # ------------------------------------------------------------------
+    async def generate_stream(self, token_ids: List[int],
+                              sampling: SamplingParams
+                              ) -> AsyncIterator[StreamChunk]:
+        st = self.add_request(token_ids, sampling)
+        try:
+            while True:
+                chunk = await st.queue.get()
+                if chunk is None:
+                    break
+                yield chunk
+        finally:
+            if not st.finished:
+                self.abort([st.seq_id])
+            self.release(st)
+
+    def encode(self, prompt: str) -> List[int]:
+        assert self.tokenizer is not None, "no tokenizer available"
+        return self.tokenizer.encode(prompt)
+
+    def apply_chat_template(self, messages, **kwargs) -> List[int]:
+        assert self.tokenizer is not None
+        return self.tokenizer.apply_chat_template(
+            messages, add_generation_prompt=True, **kwargs)

@@ -1,0 +1,166 @@
+"""Per-GPU serving worker: replicated-scheduler event loop.
+
+Design (MI355X-native, replaces the reference's column-driver /
+follower-mirror split, worker.py + dist_schedule.py):
+
+Every rank runs an IDENTICAL deterministic scheduler over the SAME
+ordered request stream. The only cross-rank coordination is:
+  1. intake sync — global rank 0 decides how many queued control
+     messages to admit before each scheduling round and broadcasts that
+     count (one tiny collective); every rank then consumes exactly that
+     many messages from its own in-order zmq queue, so all schedulers
+     stay in lockstep;
+  2. PP hidden-state legs (RCCL send/recv over one xGMI link each);
+  3. sampled-token broadcast from the last stage.
+Nothing else moves: no schedule payloads, no follower KV mirrors.
+
+Global rank 0 is the output rank: after the token broadcast it owns the
+full output stream and pushes (seq_id, token, finish) to the frontend.
+"""
+
+import os
+import time
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+from gllm_amd.config import EngineConfig
+from gllm_amd.engine.ipc import WorkerComm
+from gllm_amd.engine.pp_engine import PPEngine
+from gllm_amd.logger import logger
+from gllm_amd.sequence import SamplingParams, Sequence
+
+
+class ServingWorker(PPEngine):
+    def __init__(self, config: EngineConfig, req_queue, out_queue):
+        super().__init__(config)
+        from gllm_amd.parallel import get_rank, get_world_size
+        self.rank = get_rank()
+        self.world = get_world_size()
+        self.is_output_rank = self.rank == 0
+        self.comm = WorkerComm(req_queue, out_queue, self.is_output_rank)
+        self._intake_buf = torch.zeros(1, dtype=torch.int64)
+        if self.world > 1 and config.device.startswith("cuda"):
+            self._intake_buf = self._intake_buf.to(config.device)
+        self.profiler = None
+        self.shutdown = False
+
+    # ------------------------------------------------------------------
+    def _apply_messages(self, msgs: List[tuple]) -> None:
+        for kind, idx, payload in msgs:
+            if kind == "req":
+                seq = Sequence(payload["seq_id"], payload["token_ids"],
+                               SamplingParams(**payload["sampling"]),
+                               eos_token_id=payload.get("eos_token_id"))
+                self.scheduler.add_seqs([seq])
+            elif kind == "abort":
+                self.scheduler.abort_seqs(payload)
+            elif kind == "cmd":
+                self._handle_cmd(payload)
+
+    def _handle_cmd(self, name: str) -> None:
+        if name == "shutdown":
+            self.shutdown = True
+        elif name == "start_profile":
+            d = os.environ.get("GLLM_TORCH_PROFILER_DIR", "/tmp/gllm_prof")
+            os.makedirs(d, exist_ok=True)
+            self.profiler = torch.profiler.profile(
+                activities=[torch.profiler.ProfilerActivity.CPU,
+                            torch.profiler.ProfilerActivity.CUDA],
+                with_stack=False)
+            self.profiler.__enter__()
+            logger.info("profiler started -> %s", d)
+        elif name == "stop_profile" and self.profiler is not None:
+            self.profiler.__exit__(None, None, None)
+            d = os.environ.get("GLLM_TORCH_PROFILER_DIR", "/tmp/gllm_prof")
+            trace = os.path.join(d, f"rank{self.rank}.json")
+            try:
+                self.profiler.export_chrome_trace(trace)
+            except Exception as e:  # pragma: no cover
+                logger.warning("trace export failed: %s", e)
+            self.profiler = None
+            logger.info("profiler stopped; trace at %s", trace)
+
+    # ------------------------------------------------------------------
+    def _sync_intake(self, block: bool) -> None:
+        """Rank 0 counts queued messages (optionally waiting for at least
+        one); every rank admits exactly that count, in order."""
+        if self.world == 1:
+            if block and not self.comm.poll(100):
+                return
+            self._apply_messages(self.comm.drain())
+            return
+        if self.rank == 0:
+            if block:
+                self.comm.poll(100)
+            msgs = self.comm.drain()
+            self._intake_buf.fill_(len(msgs))
+            dist.broadcast(self._intake_buf, src=0, group=self.token_group)
+            self._apply_messages(msgs)
+        else:
+            dist.broadcast(self._intake_buf, src=0, group=self.token_group)
+            n = int(self._intake_buf.item())
+            if n:
+                self._apply_messages(self.comm.recv_blocking(n))
+
+    # ------------------------------------------------------------------
+    def _complete_oldest(self) -> List[Sequence]:
+        batch, _ = self.inflight[0]
+        finished = super()._complete_oldest()
+        # tokens just appended: emit each sampled token
+        outs = []
+        if self.is_output_rank:
+            for item in batch.items:
+                if not item.ends_prompt:
+                    continue
+                seq = item.seq
+                if seq.finish_reason == Sequence.FINISH_ABORT:
+                    outs.append((seq.seq_id, -1, seq.finish_reason))
+                else:
+                    outs.append((seq.seq_id, seq.token_ids[-1],
+                                 seq.finish_reason))
+            if outs:
+                self.comm.send_output(("out", outs, {}))
+        return finished
+
+    # ------------------------------------------------------------------
+    def run_loop(self) -> None:
+        logger.info("worker %d ready (pp=%d)", self.rank, self.pp_size)
+        while not self.shutdown:
+            has_work = self.scheduler.has_work()
+            self._sync_intake(block=not has_work)
+            if self.shutdown:
+                break
+            if not self.scheduler.has_work():
+                continue
+            while len(self.inflight) < max(1, self.pp_size):
+                b = self.scheduler.schedule_once()
+                if b is None:
+                    break
+                self._launch(b)
+            if self.inflight:
+                self._complete_oldest()
+        self.drain()
+        self.comm.close()
+        logger.info("worker %d shut down", self.rank)
+
+
+def run_worker(rank: int, config: EngineConfig, req_queue, out_queue,
+               ready_queue=None) -> None:
+    os.environ["RANK"] = str(rank)
+    os.environ.setdefault("LOCAL_RANK", str(rank))
+    os.environ["MASTER_ADDR"] = config.master_addr
+    os.environ["MASTER_PORT"] = str(config.master_port)
+    if config.device.startswith("cuda"):
+        config.device = f"cuda:{rank % max(1, torch.cuda.device_count())}"
+    try:
+        worker = ServingWorker(config, req_queue, out_queue)
+        if ready_queue is not None:
+            ready_queue.put(("ready", rank))
+        worker.run_loop()
+    except Exception as e:  # pragma: no cover
+        logger.exception("worker %d died: %s", rank, e)
+        if ready_queue is not None:
+            ready_queue.put(("dead", rank))
+        raise

@@ -1,0 +1,286 @@
+"""OpenAI-compatible API server (reference: entrypoints/api_server.py).
+
+Routes: /v1/chat/completions, /v1/completions, /v1/models, /health,
+/version, /server_info, /start_profile, /stop_profile.
+
+    python -m gllm_amd.entrypoints.api_server --model <hf_dir> \
+        --pp-size 4 --schedule-method token_throttling
+"""
+
+import argparse
+import asyncio
+import json
+import time
+from typing import Optional
+
+from gllm_amd import __version__
+from gllm_amd.config import EngineConfig
+from gllm_amd.engine.server_engine import AsyncLLMEngine
+from gllm_amd.entrypoints.protocol import (
+    ChatCompletionRequest, ChatCompletionResponse,
+    ChatCompletionResponseChoice, ChatCompletionStreamChoice,
+    ChatCompletionStreamResponse, ChatMessage, CompletionRequest,
+    CompletionResponse, CompletionResponseChoice, DeltaMessage, ModelCard,
+    ModelList, UsageInfo)
+from gllm_amd.logger import logger
+from gllm_amd.sequence import SamplingParams
+
+engine: Optional[AsyncLLMEngine] = None
+served_model = ""
+
+
+def build_app():
+    from fastapi import FastAPI, Request
+    from fastapi.responses import JSONResponse, StreamingResponse
+
+    app = FastAPI(title="gllm_amd")
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok"}
+
+    @app.get("/version")
+    async def version():
+        return {"version": __version__}
+
+    @app.get("/server_info")
+    async def server_info():
+        c = engine.config
+        return {"model": c.model, "pp_size": c.pp_size, "tp_size": c.tp_size,
+                "dp_size": c.dp_size, "schedule_method": c.schedule_method,
+                "page_size": c.page_size}
+
+    @app.get("/v1/models")
+    async def models():
+        return ModelList(data=[ModelCard(id=served_model)]).model_dump()
+
+    @app.post("/start_profile")
+    async def start_profile():
+        engine.send_command("start_profile")
+        return {"status": "started"}
+
+    @app.post("/stop_profile")
+    async def stop_profile():
+        engine.send_command("stop_profile")
+        return {"status": "stopped"}
+
+    def _sampling_from(req, default_max: int = 512) -> SamplingParams:
+        stops = req.stop
+        if isinstance(stops, str):
+            stops = [stops]
+        max_tokens = getattr(req, "max_completion_tokens", None) or \
+            req.max_tokens or default_max
+        rp = req.repetition_penalty
+        if rp is None:
+            rp = 1.0
+        return SamplingParams(
+            temperature=req.temperature if req.temperature is not None
+            else 1.0,
+            top_p=req.top_p if req.top_p is not None else 1.0,
+            top_k=req.top_k if req.top_k is not None else -1,
+            repetition_penalty=rp,
+            max_tokens=max_tokens,
+            min_tokens=req.min_tokens or 0,
+            ignore_eos=req.ignore_eos,
+            stop=stops, stop_token_ids=req.stop_token_ids,
+            seed=req.seed)
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(req: ChatCompletionRequest, raw: Request):
+        messages = [m.model_dump(exclude_none=True) for m in req.messages]
+        kwargs = req.chat_template_kwargs or {}
+        if req.tools:
+            kwargs["tools"] = [t.model_dump() for t in req.tools]
+        try:
+            token_ids = engine.apply_chat_template(messages, **kwargs)
+        except Exception as e:
+            return JSONResponse(status_code=400,
+                                content={"error": str(e)})
+        sampling = _sampling_from(req)
+        if req.stream:
+            return StreamingResponse(
+                _chat_stream(req, raw, token_ids, sampling),
+                media_type="text/event-stream")
+        text, finish, n_out = await _collect(raw, token_ids, sampling)
+        message = ChatMessage(role="assistant", content=text)
+        if req.tools:
+            from gllm_amd.tokenizers.tool_parsers import parse_tool_calls
+            parsed_text, calls = parse_tool_calls(text, served_model)
+            if calls:
+                message = ChatMessage(role="assistant",
+                                      content=parsed_text or None,
+                                      tool_calls=calls)
+                finish = "tool_calls"
+        resp = ChatCompletionResponse(
+            model=req.model or served_model,
+            choices=[ChatCompletionResponseChoice(
+                index=0, message=message, finish_reason=finish)],
+            usage=UsageInfo(prompt_tokens=len(token_ids),
+                            completion_tokens=n_out,
+                            total_tokens=len(token_ids) + n_out))
+        return resp.model_dump()
+
+    async def _collect(raw, token_ids, sampling):
+        text_parts = []
+        finish = None
+        n_out = 0
+        async for chunk in engine.generate_stream(token_ids, sampling):
+            if await raw.is_disconnected():
+                break
+            text_parts.append(chunk.text)
+            finish = chunk.finish_reason or finish
+            n_out = chunk.n_output_tokens
+        text = "".join(text_parts)
+        from gllm_amd.engine.detokenizer import check_stop_strings
+        _, text = check_stop_strings(text, sampling.stop)
+        return text, finish, n_out
+
+    async def _chat_stream(req, raw, token_ids, sampling):
+        resp_id = None
+        first = True
+        n_out = 0
+        async for chunk in engine.generate_stream(token_ids, sampling):
+            if await raw.is_disconnected():
+                break
+            delta = DeltaMessage(content=chunk.text)
+            if first:
+                delta.role = "assistant"
+                first = False
+            n_out = chunk.n_output_tokens
+            out = ChatCompletionStreamResponse(
+                model=req.model or served_model,
+                choices=[ChatCompletionStreamChoice(
+                    index=0, delta=delta,
+                    finish_reason=chunk.finish_reason)])
+            if resp_id is None:
+                resp_id = out.id
+            else:
+                out.id = resp_id
+            yield f"data: {out.model_dump_json(exclude_none=True)}\n\n"
+        if req.stream_options and req.stream_options.get("include_usage"):
+            out = ChatCompletionStreamResponse(
+                id=resp_id or "", model=req.model or served_model,
+                choices=[],
+                usage=UsageInfo(prompt_tokens=len(token_ids),
+                                completion_tokens=n_out,
+                                total_tokens=len(token_ids) + n_out))
+            yield f"data: {out.model_dump_json(exclude_none=True)}\n\n"
+        yield "data: [DONE]\n\n"
+
+    @app.post("/v1/completions")
+    async def completions(req: CompletionRequest, raw: Request):
+        prompts = req.prompt
+        if isinstance(prompts, str):
+            prompts = [prompts]
+        elif prompts and isinstance(prompts[0], int):
+            prompts = [prompts]
+        sampling = _sampling_from(req, default_max=16)
+        if req.stream:
+            token_ids = prompts[0] if isinstance(prompts[0], list) \
+                else engine.encode(prompts[0])
+            return StreamingResponse(
+                _completion_stream(req, raw, token_ids, sampling),
+                media_type="text/event-stream")
+        choices = []
+        total_p = total_c = 0
+        for i, p in enumerate(prompts):
+            token_ids = p if isinstance(p, list) else engine.encode(p)
+            text, finish, n_out = await _collect(raw, token_ids, sampling)
+            if req.echo and not isinstance(p, list):
+                text = p + text
+            choices.append(CompletionResponseChoice(
+                index=i, text=text, finish_reason=finish))
+            total_p += len(token_ids)
+            total_c += n_out
+        resp = CompletionResponse(
+            model=req.model or served_model, choices=choices,
+            usage=UsageInfo(prompt_tokens=total_p, completion_tokens=total_c,
+                            total_tokens=total_p + total_c))
+        return resp.model_dump()
+
+    async def _completion_stream(req, raw, token_ids, sampling):
+        async for chunk in engine.generate_stream(token_ids, sampling):
+            if await raw.is_disconnected():
+                break
+            resp = CompletionResponse(
+                model=req.model or served_model,
+                choices=[CompletionResponseChoice(
+                    index=0, text=chunk.text,
+                    finish_reason=chunk.finish_reason)])
+            yield f"data: {resp.model_dump_json(exclude_none=True)}\n\n"
+        yield "data: [DONE]\n\n"
+
+    return app
+
+
+def make_arg_parser():
+    p = argparse.ArgumentParser(description="gllm_amd OpenAI API server")
+    p.add_argument("--model", type=str, required=True)
+    p.add_argument("--served-model-name", type=str, default=None)
+    p.add_argument("--host", type=str, default="0.0.0.0")
+    p.add_argument("--port", type=int, default=8000)
+    p.add_argument("--zmq-port", type=int, default=28700)
+    p.add_argument("--pp-size", "--pp", type=int, default=1)
+    p.add_argument("--tp-size", "--tp", type=int, default=1)
+    p.add_argument("--dp-size", "--dp", type=int, default=1)
+    p.add_argument("--use-ep", action="store_true")
+    p.add_argument("--load-format", choices=["auto", "dummy"],
+                   default="auto")
+    p.add_argument("--dtype", type=str, default="bfloat16")
+    p.add_argument("--schedule-method", type=str,
+                   choices=["split_pd", "chunked_prefill",
+                            "token_throttling"],
+                   default="token_throttling")
+    p.add_argument("--maxp", type=int, default=8192)
+    p.add_argument("--maxd", type=int, default=1024)
+    p.add_argument("--minp", type=int, default=512)
+    p.add_argument("--iterp", type=int, default=16)
+    p.add_argument("--page-size", type=int, default=16)
+    p.add_argument("--gpu-memory-util", type=float, default=0.9)
+    p.add_argument("--model-max-length", type=int, default=None)
+    p.add_argument("--disable-prefix-caching", action="store_true")
+    p.add_argument("--enforce-eager", action="store_true")
+    p.add_argument("--max-graph-bs", type=int, default=256)
+    p.add_argument("--assigned-layers", type=str, default=None)
+    p.add_argument("--master-port", type=int, default=29500)
+    p.add_argument("--seed", type=int, default=0)
+    return p
+
+
+def config_from_args(args) -> EngineConfig:
+    return EngineConfig(
+        model=args.model, load_format=args.load_format, dtype=args.dtype,
+        pp_size=args.pp_size, tp_size=args.tp_size, dp_size=args.dp_size,
+        use_ep=args.use_ep, schedule_method=args.schedule_method,
+        maxp=args.maxp, maxd=args.maxd, minp=args.minp, iterp=args.iterp,
+        page_size=args.page_size, gpu_memory_util=args.gpu_memory_util,
+        model_max_length=args.model_max_length,
+        enable_prefix_caching=not args.disable_prefix_caching,
+        enforce_eager=args.enforce_eager, max_graph_bs=args.max_graph_bs,
+        assigned_layers=args.assigned_layers,
+        master_port=args.master_port, seed=args.seed,
+        device="cuda" if _has_gpu() else "cpu")
+
+
+def _has_gpu() -> bool:
+    import torch
+    return torch.cuda.is_available()
+
+
+def main():
+    global engine, served_model
+    args = make_arg_parser().parse_args()
+    served_model = args.served_model_name or args.model
+    config = config_from_args(args)
+    engine = AsyncLLMEngine(config, base_port=args.zmq_port)
+    engine.start()
+    app = build_app()
+    import uvicorn
+    try:
+        uvicorn.run(app, host=args.host, port=args.port, log_level="info")
+    finally:
+        engine.stop()
+
+
+if __name__ == "__main__":
+    main()
