@@ -160,13 +160,26 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
     return out
 
 
+def _use_skinny(M: int, N: int, K: int) -> bool:
+    """Measured dispatch rule (scripts/gemm_sweep.py on MI355X r01):
+    the skinny kernel beats hipBLASLt on K-deep reductions at any decode
+    M (down_proj: 1.9 -> 4.1 TB/s) and on every N<=8k shape once M>64
+    (library tiles collapse to ~1.1-1.6 TB/s there); the library keeps
+    wide-N shapes (gate_up, lm_head) and small-M small-K projections."""
+    if K % 64 != 0 or N < 1024 or M > SKINNY_MAX_M:
+        return False
+    if K >= 2 * N:
+        return True
+    return M > 64 and N <= 8192
+
+
 def linear(x: torch.Tensor, w: torch.Tensor,
            bias: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """GEMM dispatch: the skinny weight-streaming kernel for decode-size
-    M on GPU, hipBLASLt otherwise."""
+    """GEMM dispatch: the skinny weight-streaming kernel where measured
+    faster, hipBLASLt otherwise."""
     import torch.nn.functional as F
     if (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 2
-            and x.is_contiguous() and 0 < x.shape[0] <= SKINNY_MAX_M
-            and w.shape[0] >= 1024 and w.shape[1] % 64 == 0):
+            and x.is_contiguous()
+            and _use_skinny(x.shape[0], w.shape[0], w.shape[1])):
         return skinny_gemm(x, w, bias)
     return F.linear(x, w, bias)
