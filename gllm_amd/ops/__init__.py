@@ -144,9 +144,11 @@ def _skinny_splitk(M: int, N: int, K: int) -> int:
 def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
                 bias: Optional[torch.Tensor] = None) -> torch.Tensor:
     """out = x @ w.T via the gfx950 weight-streaming kernel (M<=256-ish)."""
+    import os
     M, K = x.shape
     N = w.shape[0]
-    splitk = _skinny_splitk(M, N, K)
+    splitk = int(os.environ.get("SK_SPLITK", "0")) or \
+        _skinny_splitk(M, N, K)
     need = splitk * M * N
     key = x.device.index or 0
     ws = _SKINNY_WS.get(key)

@@ -24,6 +24,7 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <cstdlib>
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 sg_bf8;
@@ -123,7 +124,10 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
     // fixed steady-state count would let the wait pass while tile kt's
     // own DMA is still flying (stale LDS reads).
     const int ahead = min(nkt - 1 - kt, RING - 1);
-    if (RING == 3 && ahead == 2) {
+    if (RING >= 4 && ahead == 3) {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(3 * GL_PER_WAVE)
+                   : "memory");
+    } else if (RING >= 3 && ahead == 2) {
       asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * GL_PER_WAVE)
                    : "memory");
     } else if (ahead == 1) {
@@ -229,12 +233,25 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   auto *ws = workspace.data_ptr<float>();
   auto *xp = (const __hip_bfloat16 *)x.data_ptr();
   auto *wp = (const __hip_bfloat16 *)w.data_ptr();
-  if (M <= 64)
-    launch_skinny<1, 3>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
-  else if (M <= 128)
-    launch_skinny<2, 3>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
-  else
+  static int ring_env = [] {
+    const char *e = getenv("SK_RING");
+    return e ? atoi(e) : 0;
+  }();
+  if (M <= 64) {
+    const int r = ring_env ? ring_env : 3;
+    if (r == 2) launch_skinny<1, 2>(ws, xp, wp, M, N, K, k_slice, splitk,
+                                    stream);
+    else if (r == 4) launch_skinny<1, 4>(ws, xp, wp, M, N, K, k_slice,
+                                         splitk, stream);
+    else launch_skinny<1, 3>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
+  } else if (M <= 128) {
+    const int r = ring_env ? ring_env : 3;
+    if (r == 2) launch_skinny<2, 2>(ws, xp, wp, M, N, K, k_slice, splitk,
+                                    stream);
+    else launch_skinny<2, 3>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
+  } else {
     launch_skinny<4, 2>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
+  }
   HIP_CHECK_KERNEL();
 
   const float *bias_ptr = nullptr;

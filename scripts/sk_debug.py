@@ -22,3 +22,31 @@ for (M,N,K) in [(1,7168,5120),(17,5120,5120),(64,64,128),(64,64,64),(16,64,192),
         rowbad = bad.any(1).nonzero().flatten()
         print(f"  bad cols: {colbad[:16].tolist()}{'...' if colbad.numel()>16 else ''} ({colbad.numel()} total)")
         print(f"  bad rows: {rowbad[:16].tolist()} ({rowbad.numel()} total)")
+
+# MB=2 / MB=4 engine-like shapes + graph capture repro
+def graph_repro():
+    M, N, K = 256, 5120, 27648
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.1
+    ops.skinny_gemm(x, w)  # warm ws
+    torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        out = ops.skinny_gemm(x, w)
+    for _ in range(3):
+        g.replay()
+    torch.cuda.synchronize()
+    ref = x.float() @ w.float().T
+    rel = ((out.float() - ref).abs() / ref.abs().clamp_min(1.0)).max()
+    print(f"graph repro M={M}: maxrel={float(rel):.4f}")
+
+if os.environ.get("SK_EXTRA"):
+    for (M,N,K) in [(100,5120,5120),(128,7168,5120),(200,5120,27648),
+                    (256,7168,5120),(256,5120,5120),(65,5120,27648)]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.1
+        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.1
+        out = ops.skinny_gemm(x, w).float()
+        ref = x.float() @ w.float().T
+        rel = ((out - ref).abs() / ref.abs().clamp_min(1.0)).max()
+        print(f"extra M={M} N={N} K={K}: maxrel={float(rel):.4f}")
+    graph_repro()
