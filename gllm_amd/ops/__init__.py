@@ -168,6 +168,9 @@ def _use_skinny(M: int, N: int, K: int) -> bool:
     M (down_proj: 1.9 -> 4.1 TB/s) and on every N<=8k shape once M>64
     (library tiles collapse to ~1.1-1.6 TB/s there); the library keeps
     wide-N shapes (gate_up, lm_head) and small-M small-K projections."""
+    import os
+    if os.environ.get("GLLM_DISABLE_SKINNY"):
+        return False
     if K % 64 != 0 or N < 1024 or M > SKINNY_MAX_M:
         return False
     if K >= 2 * N:
