@@ -50,3 +50,53 @@ if os.environ.get("SK_EXTRA"):
         rel = ((out - ref).abs() / ref.abs().clamp_min(1.0)).max()
         print(f"extra M={M} N={N} K={K}: maxrel={float(rel):.4f}")
     graph_repro()
+
+def graph_repro_multi():
+    """Mimic GraphRunner: shared pool, side stream, multiple buckets,
+    qkv(+bias)/o/down skinny + gate_up library per forward."""
+    import torch.nn.functional as F
+    K = 5120
+    w_qkv = torch.randn(7168, K, dtype=torch.bfloat16, device="cuda") * 0.05
+    b_qkv = torch.randn(7168, dtype=torch.bfloat16, device="cuda") * 0.05
+    w_o = torch.randn(K, K, dtype=torch.bfloat16, device="cuda") * 0.05
+    w_gu = torch.randn(55296, K, dtype=torch.bfloat16, device="cuda") * 0.05
+    w_dn = torch.randn(K, 27648, dtype=torch.bfloat16, device="cuda") * 0.05
+
+    def fwd(x):
+        q = ops.linear(x, w_qkv, b_qkv)
+        o = ops.linear(q[:, :K].contiguous(), w_o)
+        g = ops.linear(o, w_gu)
+        h = ops.silu_and_mul(g)
+        y = ops.linear(h.contiguous(), w_dn)
+        return y
+
+    pool = torch.cuda.graphs.graph_pool_handle()
+    stream = torch.cuda.Stream()
+    graphs, outs, ins = {}, {}, {}
+    for bs in [512, 384, 256, 192, 128, 96, 64, 32, 16, 8, 4, 2, 1]:
+        x = torch.randn(bs, K, dtype=torch.bfloat16, device="cuda") * 0.05
+        ins[bs] = x
+        with torch.cuda.stream(stream):
+            for _ in range(2):
+                fwd(x)
+        torch.cuda.current_stream().wait_stream(stream)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g, pool=pool, stream=stream):
+            outs[bs] = fwd(x)
+        graphs[bs] = g
+    torch.cuda.synchronize()
+    for rep in range(5):
+        for bs in [256, 64, 128, 1, 512]:
+            graphs[bs].replay()
+    torch.cuda.synchronize()
+    for bs in [256, 64]:
+        ref = fwd(ins[bs])
+        graphs[bs].replay()
+        torch.cuda.synchronize()
+        d = (outs[bs].float() - ref.float()).abs().max()
+        print(f"multi-graph bs={bs}: maxdiff={float(d):.4f}")
+    print("graph_repro_multi OK")
+
+if os.environ.get("SK_GRAPH2"):
+    graph_repro_multi()
