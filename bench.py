@@ -66,6 +66,7 @@ def main():
     ap.add_argument("--page-size", type=int, default=16)
     ap.add_argument("--schedule", type=str, default="token_throttling")
     ap.add_argument("--no-graph", action="store_true")
+    ap.add_argument("--max-graph-bs", type=int, default=512)
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -89,7 +90,7 @@ def main():
         dtype="bfloat16" if use_gpu else "float32",
         device=device, pp_size=n, page_size=args.page_size,
         schedule_method=args.schedule, maxp=8192, maxd=1024,
-        use_graph=not args.no_graph, max_graph_bs=512,
+        use_graph=not args.no_graph, max_graph_bs=args.max_graph_bs,
         enable_prefix_caching=False,
         master_addr=os.environ.get("MASTER_ADDR", "127.0.0.1"),
         master_port=int(os.environ.get("MASTER_PORT", "29500")),

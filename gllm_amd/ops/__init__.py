@@ -171,7 +171,8 @@ def _use_skinny(M: int, N: int, K: int) -> bool:
     import os
     if os.environ.get("GLLM_DISABLE_SKINNY"):
         return False
-    if K % 64 != 0 or N < 1024 or M > SKINNY_MAX_M:
+    max_m = int(os.environ.get("GLLM_SKINNY_MAX_M", "0")) or SKINNY_MAX_M
+    if K % 64 != 0 or N < 1024 or M > max_m:
         return False
     if K >= 2 * N:
         return True
