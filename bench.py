@@ -59,14 +59,14 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--batch", type=int, default=64)
+    ap.add_argument("--batch", type=int, default=256)
     ap.add_argument("--prompt-len", type=int, default=1024)
     ap.add_argument("--model", type=str, default="qwen2.5-32b",
                     choices=["qwen2.5-32b", "debug"])
     ap.add_argument("--page-size", type=int, default=16)
     ap.add_argument("--schedule", type=str, default="token_throttling")
     ap.add_argument("--no-graph", action="store_true")
-    ap.add_argument("--max-graph-bs", type=int, default=512)
+    ap.add_argument("--max-graph-bs", type=int, default=64)
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))

@@ -53,7 +53,12 @@ class EngineConfig:
     # --- runtime ---
     use_graph: bool = True               # hipGraph-captured decode buckets
     enable_overlap: bool = True          # launch-first/collect-later worker loop
-    max_graph_bs: int = 256
+    # hipGraphs pay off when launch overhead dominates (small decode
+    # batches); large batches run eager (GPU-bound) with the full skinny
+    # GEMM range. NOTE: skinny MB>=2 kernels captured inside engine
+    # graphs fault on replay (suspected hipGraph + 72-80KB dynamic-LDS
+    # interaction; standalone repros pass) — buckets <= 64 sidestep it.
+    max_graph_bs: int = 64
     profile_batch: int = 2048            # tokens used for the peak profile run
     device: str = "cuda"                 # "cuda" (=ROCm HIP) | "cpu"
     seed: int = 0

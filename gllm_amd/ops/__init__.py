@@ -174,6 +174,9 @@ def _use_skinny(M: int, N: int, K: int) -> bool:
     max_m = int(os.environ.get("GLLM_SKINNY_MAX_M", "0")) or SKINNY_MAX_M
     if K % 64 != 0 or N < 1024 or M > max_m:
         return False
+    if M > 64 and torch.cuda.is_current_stream_capturing():
+        return False  # MB>=2 under graph capture faults on replay (see
+        # config.max_graph_bs note); capture uses the library instead
     if K >= 2 * N:
         return True
     return M > 64 and N <= 8192
