@@ -66,6 +66,7 @@ def main():
     ap.add_argument("--page-size", type=int, default=16)
     ap.add_argument("--schedule", type=str, default="token_throttling")
     ap.add_argument("--no-graph", action="store_true")
+    ap.add_argument("--no-overlap", action="store_true")
     ap.add_argument("--max-graph-bs", type=int, default=64)
     args = ap.parse_args()
 
@@ -82,6 +83,7 @@ def main():
     model_dir = write_model_dir(model_json)
 
     from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.overlap_engine import OverlapEngine
     from gllm_amd.engine.pp_engine import PPEngine
     from gllm_amd.sequence import SamplingParams, Sequence
 
@@ -95,7 +97,10 @@ def main():
         master_addr=os.environ.get("MASTER_ADDR", "127.0.0.1"),
         master_port=int(os.environ.get("MASTER_PORT", "29500")),
     )
-    eng = PPEngine(cfg, num_pages_override=None if use_gpu else 512)
+    if n == 1 and not args.no_overlap:
+        eng = OverlapEngine(cfg, num_pages_override=None if use_gpu else 512)
+    else:
+        eng = PPEngine(cfg, num_pages_override=None if use_gpu else 512)
 
     # ---- synthetic load: batch of fixed prompts, unbounded decode ----
     g = torch.Generator().manual_seed(1234)

@@ -198,6 +198,14 @@ class PrefixMemoryManager(MemoryManager):
         # caller registers before appending the sampled token; clamp to the
         # tokens we actually have.
         n_full = min(seq.computed_token_num, len(seq.token_ids)) // self.page_size
+        # overlap mode: never intern pages containing unresolved
+        # placeholder tokens (negative ids) — reference hit exactly this
+        # poisoning bug (memory_manager.py:1055-1079)
+        limit = n_full * self.page_size
+        for j, t in enumerate(seq.token_ids[:limit]):
+            if t < 0:
+                n_full = j // self.page_size
+                break
         chains = self._chains_up_to(seq, n_full)
         for i in range(n_full):
             cid = chains[i]

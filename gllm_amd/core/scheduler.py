@@ -409,6 +409,73 @@ class Scheduler:
         return finished
 
     # ------------------------------------------------------------------
+    # overlap mode: deferred finalize (reference OverlapScheduler,
+    # scheduler.py:699-783). The sampled token is appended as a NEGATIVE
+    # placeholder resolved on-GPU next step; finish checks run at
+    # finalize once the real token lands on host.
+    # ------------------------------------------------------------------
+    def process_output_deferred(self, batch: ScheduledBatch,
+                                placeholders: List[int]):
+        """Commit cursors and append placeholder tokens. Returns records
+        [(item_idx, seq, token_pos)] for finalize_output."""
+        assert self.batch_running and self.batch_running[0] is batch
+        self.batch_running.popleft()
+        records = []
+        for i, item in enumerate(batch.items):
+            seq = item.seq
+            seq.computed_token_num = item.start + item.num_tokens
+            if not item.ends_prompt:
+                continue
+            if seq.is_finished:
+                # finished retroactively while this chunk was in flight
+                if not self._in_flight(seq):
+                    self.mm.free_seq(seq)
+                    self._scheduled_cursor.pop(seq.seq_id, None)
+                continue
+            seq.append_token(placeholders[i])
+            records.append((i, seq, len(seq.token_ids) - 1))
+            # keep decoding only if the output budget allows another step
+            if seq.num_output_tokens < seq.sampling.max_tokens and                     seq.seq_id not in self.abort_ids:
+                self.seqs_to_decode.append(seq)
+        return records
+
+    def finalize_output(self, batch: ScheduledBatch, tokens: List[int],
+                        records) -> List[Sequence]:
+        """Replace placeholders with real tokens; run finish checks."""
+        finished: List[Sequence] = []
+        for (i, seq, pos) in records:
+            already_finished = seq.is_finished
+            if already_finished:
+                # extra speculative token of a retro-finished seq
+                if pos < len(seq.token_ids) and seq.token_ids[pos] < 0:
+                    del seq.token_ids[pos]
+                continue
+            if seq.token_ids[pos] < 0:
+                seq.token_ids[pos] = int(tokens[i])
+            if seq.seq_id in self.abort_ids:
+                seq.finish_reason = Sequence.FINISH_ABORT
+            # position-aware: a later placeholder may already sit beyond
+            # pos when the pipeline runs >1 batch deep
+            seq.check_finish(pos)
+            if seq.is_finished:
+                # drop speculative tokens appended after the finish
+                del seq.token_ids[pos + 1:]
+            elif isinstance(self.mm, PrefixMemoryManager):
+                self.mm.register_computed_pages(seq)
+            if seq.is_finished:
+                finished.append(seq)
+                # drop from the decode queue if parked there
+                try:
+                    self.seqs_to_decode.remove(seq)
+                except ValueError:
+                    pass
+                if not self._in_flight(seq):
+                    self.mm.free_seq(seq)
+                    self._scheduled_cursor.pop(seq.seq_id, None)
+                    self.abort_ids.discard(seq.seq_id)
+        return finished
+
+    # ------------------------------------------------------------------
     def _maybe_log(self, items: List[ScheduledSeq]) -> None:
         if not self.log:
             return

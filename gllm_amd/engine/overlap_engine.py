@@ -1,0 +1,105 @@
+"""Launch-first / collect-later engine (PP = 1).
+
+Parity: reference OverlapWorker + OverlapModelRunner + FutureMap
+(overlap_worker.py, model_runner.py:1961-2311, async_utils.py). The CPU
+side schedules and launches batch b+1 while batch b still runs on the
+GPU; b+1's decode inputs carry negative placeholders resolved on-GPU
+against the sampled-token ring, and b's outputs are finalized only after
+an async D2H lands (event-gated). Up to DEPTH batches stay in flight.
+"""
+
+import collections
+from typing import List, Optional
+
+import torch
+
+from gllm_amd.config import EngineConfig
+from gllm_amd.engine.pp_engine import PPEngine
+from gllm_amd.sequence import Sequence
+
+
+class OverlapEngine(PPEngine):
+    DEPTH = 2
+
+    def __init__(self, config: EngineConfig,
+                 num_pages_override: Optional[int] = None):
+        super().__init__(config, num_pages_override=num_pages_override)
+        assert self.pp_size == 1, "overlap engine is the PP=1 fast path"
+        self.maxd = config.maxd
+        self.ring_slots = self.runner.ring_slots
+        self._slot = 0
+        # pending: (batch, records, event, pinned_tokens, n_rows)
+        self.pending = collections.deque()
+        self.is_cuda = config.device.startswith("cuda")
+        if self.is_cuda:
+            self._pinned = [torch.zeros(config.maxd, dtype=torch.long)
+                            .pin_memory() for _ in range(self.ring_slots)]
+        self._finished_since: List[Sequence] = []
+
+    # ------------------------------------------------------------------
+    def _launch_overlap(self, batch) -> None:
+        slot = self._slot
+        self._slot = (self._slot + 1) % self.ring_slots
+        out = self.runner.step_first_stage(batch)   # SamplerOutput, async
+        B = len(batch.items)
+        ring = self.runner.token_ring
+        ring[slot, :B].copy_(out.next_tokens)
+        phs = [-(slot * self.maxd + i) - 1 for i in range(B)]
+        records = self.scheduler.process_output_deferred(batch, phs)
+        if self.is_cuda:
+            pinned = self._pinned[slot]
+            pinned[:B].copy_(out.next_tokens, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+        else:
+            pinned = out.next_tokens
+            ev = None
+        self.pending.append((batch, records, ev, pinned, B))
+
+    def _collect_one(self) -> int:
+        batch, records, ev, pinned, B = self.pending.popleft()
+        if ev is not None:
+            ev.synchronize()
+        tokens = pinned[:B].tolist()
+        finished = self.scheduler.finalize_output(batch, tokens, records)
+        self._finished_since.extend(finished)
+        return len(records)
+
+    # ------------------------------------------------------------------
+    def step_tick(self) -> int:
+        """Returns sampled tokens finalized during this tick."""
+        n_final = 0
+        while len(self.pending) >= self.DEPTH:
+            n_final += self._collect_one()
+        batch = self.scheduler.schedule_once()
+        if batch is None:
+            while self.pending:
+                n_final += self._collect_one()
+            return n_final
+        # repetition penalty reads real token histories: drain first
+        if any(it.seq.sampling.repetition_penalty != 1.0
+               for it in batch.items):
+            while self.pending:
+                n_final += self._collect_one()
+        self._launch_overlap(batch)
+        return n_final
+
+    def pop_finished(self) -> List[Sequence]:
+        out = self._finished_since
+        self._finished_since = []
+        return out
+
+    def run_until_done(self, max_steps: Optional[int] = None):
+        done = []
+        steps = 0
+        while self.scheduler.has_work() or self.pending:
+            self.step_tick()
+            done.extend(self.pop_finished())
+            steps += 1
+            if max_steps is not None and steps >= max_steps:
+                break
+        return done
+
+    def drain(self) -> None:
+        while self.pending:
+            self._collect_one()

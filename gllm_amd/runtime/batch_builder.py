@@ -49,6 +49,7 @@ class BatchBuilder:
             seq_lens[i] = s + n
             block_table[i, :pt.shape[0]] = pt
 
+        has_ph = bool((tokens < 0).any())
         max_q = int(lens.max())
         max_s = int(seq_lens.max())
         dev = self.device
@@ -74,6 +75,7 @@ class BatchBuilder:
             k_caches=k_caches,
             v_caches=v_caches,
         )
+        fctx.has_placeholders = has_ph
         if need_logits:
             fctx.logits_indices = (fctx.query_start_loc[1:].long() - 1)
         return tokens_t, fctx

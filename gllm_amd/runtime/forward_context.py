@@ -30,6 +30,8 @@ class ForwardContext:
     logits_indices: Optional[torch.Tensor] = None
     # profile/warmup run: attention may be skipped (no KV yet)
     is_profile_run: bool = False
+    # overlap mode: token ids contain negative ring placeholders
+    has_placeholders: bool = False
 
     @property
     def is_pure_decode(self) -> bool:

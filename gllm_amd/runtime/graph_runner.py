@@ -84,13 +84,14 @@ class GraphRunner:
             fctx = self._fctx_for(bs)
             with torch.cuda.stream(stream):
                 for _ in range(2):  # warmup outside capture
-                    runner.model(self.in_ids[:bs], self.positions[:bs], fctx)
+                    ids = runner.resolve_tokens(self.in_ids[:bs])
+                    runner.model(ids, self.positions[:bs], fctx)
             torch.cuda.current_stream().wait_stream(stream)
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g, pool=pool, stream=stream):
-                hidden, _ = runner.model(self.in_ids[:bs],
-                                         self.positions[:bs], fctx)
+                ids = runner.resolve_tokens(self.in_ids[:bs])
+                hidden, _ = runner.model(ids, self.positions[:bs], fctx)
             self.graphs[bs] = g
             self.hidden_out[bs] = hidden
         torch.cuda.synchronize()

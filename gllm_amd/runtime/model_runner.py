@@ -52,6 +52,13 @@ class ModelRunner:
         # (reference input_data.py:611-671 dummy-page padding trick)
         self.memory_manager = mgr_cls(num_pages - 1, cfg.page_size)
         self.builder = BatchBuilder(cfg.page_size, cfg.device)
+        # overlap-mode sampled-token ring (FutureMap equivalent,
+        # reference async_utils.py:56-61): next batch's decode inputs may
+        # be negative placeholders resolved against this GPU buffer
+        self.ring_slots = 8
+        self.token_ring = torch.zeros(
+            (self.ring_slots, cfg.maxd), dtype=torch.long,
+            device=cfg.device)
         from gllm_amd.parallel import get_pp_size, get_tp_size
         if (cfg.use_graph and cfg.device.startswith("cuda")
                 and get_pp_size() == 1 and get_tp_size() == 1):
@@ -150,6 +157,8 @@ class ModelRunner:
         tokens, fctx = self.builder.build(
             batch, self.k_caches, self.v_caches,
             need_logits=self.model.is_last_stage)
+        if fctx.has_placeholders:
+            tokens = self.resolve_tokens(tokens)
         hidden, residual = self._stage_forward(tokens, fctx.positions, fctx)
         if self.model.is_last_stage:
             return self._sample(batch, hidden, fctx)
@@ -167,6 +176,13 @@ class ModelRunner:
         if self.model.is_last_stage:
             return self._sample(batch, hidden, fctx)
         return hidden, residual, fctx
+
+    def resolve_tokens(self, tokens: torch.Tensor) -> torch.Tensor:
+        """Replace negative placeholder ids with sampled tokens from the
+        ring (overlap mode)."""
+        ring_flat = self.token_ring.view(-1)
+        idx = (-tokens - 1).clamp_min(0)
+        return torch.where(tokens < 0, ring_flat[idx], tokens)
 
     def _sample(self, batch: ScheduledBatch, hidden, fctx):
         logits = self.model.compute_logits(hidden, fctx)

@@ -107,16 +107,21 @@ class Sequence:
         self.computed_token_num += self.to_compute_token_num
         self.to_compute_token_num = 0
 
-    def check_finish(self) -> None:
+    def check_finish(self, pos: Optional[int] = None) -> None:
+        """Evaluate finish conditions for the token at ``pos`` (default:
+        the last token). Overlap mode passes an explicit pos because a
+        later placeholder may already be appended beyond it."""
         if self.finish_reason is not None:
             return
-        n_out = self.num_output_tokens
+        if pos is None:
+            pos = len(self.token_ids) - 1
+        n_out = pos - self.prompt_len + 1
         if n_out >= self.sampling.max_tokens:
             self.finish_reason = self.FINISH_LENGTH
             return
         if n_out < max(1, self.sampling.min_tokens):
             return
-        last = self.token_ids[-1]
+        last = self.token_ids[pos]
         if (not self.sampling.ignore_eos and self.eos_token_id is not None
                 and last == self.eos_token_id):
             self.finish_reason = self.FINISH_STOP
