@@ -250,7 +250,10 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                                     stream);
     else launch_skinny<2, 3>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
   } else {
-    launch_skinny<4, 2>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
+    const int r = ring_env ? ring_env : 2;
+    if (r >= 3) launch_skinny<4, 3>(ws, xp, wp, M, N, K, k_slice, splitk,
+                                    stream);
+    else launch_skinny<4, 2>(ws, xp, wp, M, N, K, k_slice, splitk, stream);
   }
   HIP_CHECK_KERNEL();
 
