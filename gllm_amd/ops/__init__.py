@@ -133,7 +133,9 @@ SKINNY_MAX_M = 256
 
 
 def _skinny_splitk(M: int, N: int, K: int) -> int:
-    n_wg = -(-N // 64) * -(-M // 64)
+    # v2 kernel grid is (N/64, splitk) — M-blocks live INSIDE one
+    # workgroup, so the fill target counts only N tiles
+    n_wg = -(-N // 64)
     s = 1
     while s < 16 and n_wg * s < 512 and (K // (s * 2)) >= 64:
         s *= 2
