@@ -27,6 +27,7 @@ class RequestOutput:
         self.token_ids = seq.output_token_ids
         self.finish_reason = seq.finish_reason
         self.text = text
+        self.logprobs = seq.out_logprobs or None
 
 
 class LLM:
@@ -86,6 +87,8 @@ class LLM:
         if batch is None:
             return []
         out = self.runner.step_first_stage(batch)
+        from gllm_amd.engine.pp_engine import PPEngine
+        PPEngine._stash_logprobs(batch, out)
         tokens = out.next_tokens.tolist()
         finished = self.scheduler.process_output(batch, tokens)
         for s in finished:

@@ -90,10 +90,26 @@ class PPEngine:
                 dist.send(r2.contiguous(), dst=self.P.get_next_pp_rank())
                 self.inflight.append((batch, None))
 
+    @staticmethod
+    def _stash_logprobs(batch, out) -> None:
+        if out.logprobs is None:
+            return
+        lp = out.logprobs.tolist()
+        tv = out.topk_logprobs.tolist()
+        ti = out.topk_token_ids.tolist()
+        for i, item in enumerate(batch.items):
+            if not item.ends_prompt:
+                continue
+            if item.seq.sampling.logprobs:
+                k = item.seq.sampling.logprobs
+                item.seq.out_logprobs.append(
+                    (lp[i], dict(zip(ti[i][:k], tv[i][:k]))))
+
     def _complete_oldest(self) -> List[Sequence]:
         batch, out = self.inflight.popleft()
         B = len(batch.items)
         if self.pp_size == 1:
+            self._stash_logprobs(batch, out)
             tokens = out.next_tokens.tolist()
         else:
             if self.is_last:

@@ -72,6 +72,9 @@ class Sequence:
         # repetition-penalty mask slot / SSM slot (set by managers when used)
         self.penalty_slot: int = -1
         self.ssm_slot: int = -1
+        # per-output-token logprobs (filled when sampling.logprobs is set):
+        # list of (chosen_logprob, {token_id: logprob} top-k)
+        self.out_logprobs: List[tuple] = []
 
     # ---- basic accounting ----
     def __len__(self) -> int:

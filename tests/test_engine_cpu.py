@@ -77,3 +77,14 @@ def test_token_throttling_engine(tiny_config):
     outs = llm.generate([list(range(1, 20)), list(range(5, 30))],
                         [greedy(5), greedy(5)])
     assert all(len(o.token_ids) == 5 for o in outs)
+
+
+def test_logprobs_returned(llm):
+    sp = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True,
+                        logprobs=3)
+    out = llm.generate([[1, 2, 3, 4]], [sp])[0]
+    assert out.logprobs is not None and len(out.logprobs) == 4
+    chosen, top = out.logprobs[0]
+    assert chosen <= 0.0 and len(top) == 3
+    # the chosen (greedy) token must be the top-1 entry
+    assert out.token_ids[0] in top
