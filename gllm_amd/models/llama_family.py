@@ -45,6 +45,9 @@ class DenseAttention(nn.Module):
                  dtype=None):
         super().__init__()
         hidden = cfg.hidden_size
+        sw = getattr(cfg, "sliding_window", None)
+        use_sw = getattr(cfg, "use_sliding_window", sw is not None)
+        self.sliding_window = int(sw) if (sw and use_sw) else 0
         self.total_heads = cfg.num_attention_heads
         self.total_kv_heads = getattr(cfg, "num_key_value_heads",
                                       self.total_heads)
@@ -69,7 +72,8 @@ class DenseAttention(nn.Module):
             self.k_norm = RMSNorm(self.head_dim, eps)
         self.attn = Attention(
             layer_idx, self.qkv_proj.num_heads, self.qkv_proj.num_kv_heads,
-            self.head_dim, self.head_dim ** -0.5)
+            self.head_dim, self.head_dim ** -0.5,
+            sliding_window=self.sliding_window)
 
     def forward(self, positions, hidden, fctx: ForwardContext):
         q, k, v = self.qkv_proj(hidden)
@@ -255,3 +259,9 @@ class Qwen2ForCausalLM(LlamaFamilyForCausalLM):
 class Qwen3ForCausalLM(LlamaFamilyForCausalLM):
     qkv_bias = False
     qk_norm = True
+
+
+class MistralForCausalLM(LlamaFamilyForCausalLM):
+    """Llama structure + sliding-window attention (read from config)."""
+    qkv_bias = False
+    qk_norm = False

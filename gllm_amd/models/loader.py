@@ -22,7 +22,12 @@ def get_arch_registry():
     from gllm_amd.models.moe_family import (MixtralForCausalLM,
                                             Qwen2MoeForCausalLM,
                                             Qwen3MoeForCausalLM)
+    from gllm_amd.models.chatglm import ChatGLMForCausalLM
+    from gllm_amd.models.llama_family import MistralForCausalLM
     return {
+        "ChatGLMModel": ChatGLMForCausalLM,
+        "ChatGLMForConditionalGeneration": ChatGLMForCausalLM,
+        "MistralForCausalLM": MistralForCausalLM,
         "LlamaForCausalLM": LlamaForCausalLM,
         "Qwen2ForCausalLM": Qwen2ForCausalLM,
         "Qwen3ForCausalLM": Qwen3ForCausalLM,

@@ -34,17 +34,20 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     const int *__restrict__ block_table,  // [B, max_pages]
     const int *__restrict__ seq_lens,     // [B]
     int max_pages, int page_size, int num_kv_heads, float scale,
-    int num_splits, long q_stride) {
+    int num_splits, long q_stride, int window) {
   const int b = blockIdx.x;
   const int kvh = blockIdx.y;
   const int split = blockIdx.z;
   const int Hq = num_kv_heads * G;
   const int seq_len = seq_lens[b];
+  // sliding window: only the last `window` tokens are visible
+  const int w_begin = (window > 0) ? max(0, seq_len - window) : 0;
+  const int visible = seq_len - w_begin;
 
   // split token range (page-aligned chunks not required; rows are
   // addressed per token through the block table)
-  const int split_len = (seq_len + num_splits - 1) / num_splits;
-  const int t0 = split * split_len;
+  const int split_len = (visible + num_splits - 1) / num_splits;
+  const int t0 = w_begin + split * split_len;
   const int t1 = min(t0 + split_len, seq_len);
   const int lane_per_tok = D / 8;            // 16 lanes for D=128
   const int toks_per_iter = BLOCK / lane_per_tok;
@@ -224,7 +227,7 @@ void launch_decode(torch::Tensor &out, const torch::Tensor &q,
                    const torch::Tensor &k_cache, const torch::Tensor &v_cache,
                    const torch::Tensor &block_table,
                    const torch::Tensor &seq_lens, float scale,
-                   int max_seq_len, long q_stride) {
+                   int max_seq_len, long q_stride, int window) {
   const int B = q.size(0);
   const int Hq = q.size(1);
   const int Hkv = k_cache.size(2);
@@ -247,7 +250,7 @@ void launch_decode(torch::Tensor &out, const torch::Tensor &q,
                      (const __hip_bfloat16 *)v_cache.data_ptr(),
                      block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                      (int)block_table.size(1), (int)k_cache.size(1), Hkv,
-                     scale, splits, q_stride);
+                     scale, splits, q_stride, window);
   HIP_CHECK_KERNEL();
   hipLaunchKernelGGL((decode_merge_kernel<D>), dim3(B, Hq), dim3(D), 0,
                      stream, (__hip_bfloat16 *)out.data_ptr(),
@@ -262,7 +265,6 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_table, torch::Tensor seq_lens,
                             double scale, long sliding_window) {
-  TORCH_CHECK(sliding_window == 0, "sliding window: not yet");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "decode attn: bf16 only");
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2),
@@ -280,7 +282,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
 #define CASE(DD, GG)                                                       \
   if (D == DD && G == GG) {                                                \
     launch_decode<DD, GG>(out, q, k_cache, v_cache, block_table, seq_lens, \
-                          s, max_seq, q.stride(0));                        \
+                          s, max_seq, q.stride(0), (int)sliding_window);   \
     return;                                                                \
   }
   CASE(128, 1) CASE(128, 2) CASE(128, 4) CASE(128, 5) CASE(128, 8)

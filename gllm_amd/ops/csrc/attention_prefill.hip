@@ -49,7 +49,7 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
     const int *__restrict__ seq_lens,            // [B]
     const int *__restrict__ qsl,                 // [B+1]
     int max_pages, int page_size, int Hq, int num_kv_heads, float scale,
-    long q_stride) {
+    long q_stride, int window) {
   const int tile = blockIdx.x;
   const int b = blockIdx.y;
   const int h = blockIdx.z;
@@ -102,8 +102,14 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
   const int *bt = block_table + (long)b * max_pages;
   // causal upper bound for this q tile
   const int kv_max = min(seq_len, past + min(q_len, (tile + 1) * BQ));
+  // sliding window lower bound for the tile's FIRST row
+  int kv_lo = 0;
+  if (window > 0) {
+    const int first_qpos = past + tile * BQ;
+    kv_lo = max(0, first_qpos - window + 1) / BKV * BKV;
+  }
 
-  for (int kv0 = 0; kv0 < kv_max; kv0 += BKV) {
+  for (int kv0 = kv_lo; kv0 < kv_max; kv0 += BKV) {
     const int kv_len = min(BKV, kv_max - kv0);
     // ---------- stage K (swizzled) and V^T ----------
     __syncthreads();
@@ -169,6 +175,10 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
       const int kvp0 = kv0 + l16, kvp1 = kv0 + 16 + l16;
       if (qrow >= q_len || kvp0 > qpos || kvp0 >= seq_len) s0 = -INFINITY;
       if (qrow >= q_len || kvp1 > qpos || kvp1 >= seq_len) s1 = -INFINITY;
+      if (window > 0) {
+        if (kvp0 <= qpos - window) s0 = -INFINITY;
+        if (kvp1 <= qpos - window) s1 = -INFINITY;
+      }
       // row max over the 16 lanes of this row group, both fragments
       float mx = fmaxf(s0, s1);
 #pragma unroll
@@ -249,7 +259,6 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor query_start_loc,
                              long max_query_len, double scale,
                              long sliding_window) {
-  TORCH_CHECK(sliding_window == 0, "sliding window: not yet");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "prefill attn: bf16 only");
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2),
@@ -272,7 +281,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                        query_start_loc.data_ptr<int>(),
                        (int)block_table.size(1), (int)k_cache.size(1), Hq,
-                       Hkv, (float)scale, q.stride(0));
+                       Hkv, (float)scale, q.stride(0),
+                       (int)sliding_window);
   } else if (D == 64) {
     hipLaunchKernelGGL((paged_prefill_kernel<64>),
                        dim3(q_tiles, B, Hq), dim3(BLOCK), 0, stream,
@@ -283,7 +293,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
                        query_start_loc.data_ptr<int>(),
                        (int)block_table.size(1), (int)k_cache.size(1), Hq,
-                       Hkv, (float)scale, q.stride(0));
+                       Hkv, (float)scale, q.stride(0),
+                       (int)sliding_window);
   } else {
     TORCH_CHECK(false, "prefill attn: head_dim ", D, " unsupported");
   }

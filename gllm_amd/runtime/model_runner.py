@@ -71,13 +71,19 @@ class ModelRunner:
     def kv_spec(self) -> KVCacheSpec:
         hf = self.hf_config
         num_local_layers = self.model.num_local_layers
-        total_kv = getattr(hf, "num_key_value_heads",
-                           hf.num_attention_heads)
-        from gllm_amd.parallel import get_tp_size
-        tp = get_tp_size()
-        kv_per_rank = max(1, total_kv // tp)
-        head_dim = getattr(hf, "head_dim", None) or \
-            hf.hidden_size // hf.num_attention_heads
+        # prefer the model's own attention geometry (configs name kv
+        # heads differently, e.g. ChatGLM's multi_query_group_num)
+        attn = getattr(self.model, "kv_geometry", None)
+        if attn is not None:
+            kv_per_rank, head_dim = attn
+        else:
+            total_kv = getattr(hf, "num_key_value_heads",
+                               hf.num_attention_heads)
+            from gllm_amd.parallel import get_tp_size
+            tp = get_tp_size()
+            kv_per_rank = max(1, total_kv // tp)
+            head_dim = getattr(hf, "head_dim", None) or \
+                hf.hidden_size // hf.num_attention_heads
         return KVCacheSpec(num_local_layers, kv_per_rank, head_dim,
                            self.config.page_size,
                            dtype_bytes=self.kv_dtype.itemsize)

@@ -88,3 +88,58 @@ def test_logprobs_returned(llm):
     assert chosen <= 0.0 and len(top) == 3
     # the chosen (greedy) token must be the top-1 entry
     assert out.token_ids[0] in top
+
+
+CHATGLM_TINY = {
+    "architectures": ["ChatGLMModel"],
+    "model_type": "chatglm",
+    "hidden_size": 64,
+    "ffn_hidden_size": 128,
+    "num_layers": 2,
+    "num_attention_heads": 4,
+    "multi_query_attention": True,
+    "multi_query_group_num": 2,
+    "kv_channels": 16,
+    "padded_vocab_size": 128,
+    "seq_length": 2048,
+    "layernorm_epsilon": 1e-5,
+    "add_qkv_bias": True,
+    "rope_ratio": 1.0,
+    "eos_token_id": 0,
+    "vocab_size": 128,
+}
+
+MISTRAL_TINY = {
+    "architectures": ["MistralForCausalLM"],
+    "model_type": "mistral",
+    "hidden_size": 64,
+    "intermediate_size": 128,
+    "num_hidden_layers": 2,
+    "num_attention_heads": 4,
+    "num_key_value_heads": 2,
+    "vocab_size": 128,
+    "max_position_embeddings": 2048,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 10000.0,
+    "sliding_window": 8,
+    "eos_token_id": 0,
+}
+
+
+@pytest.mark.parametrize("cfg_json", [CHATGLM_TINY, MISTRAL_TINY],
+                         ids=["chatglm", "mistral_swa"])
+def test_more_arches_generate(tmp_path, cfg_json):
+    import json
+    d = tmp_path / cfg_json["model_type"]
+    d.mkdir()
+    with open(d / "config.json", "w") as f:
+        json.dump(cfg_json, f)
+    from gllm_amd.config import EngineConfig
+    cfg = EngineConfig(model=str(d), load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4,
+                       enable_prefix_caching=False)
+    llm2 = LLM(config=cfg, num_pages_override=128)
+    out = llm2.generate([list(range(1, 30))], [greedy(6)])
+    assert len(out[0].token_ids) == 6
+    out2 = llm2.generate([list(range(1, 30))], [greedy(6)])
+    assert out2[0].token_ids == out[0].token_ids
