@@ -173,3 +173,34 @@ def test_prefill_attention_head_dim_64(kernels):
                             v_cache.float().cpu(), bt.cpu(), seq_lens.cpu(),
                             qsl.cpu(), 1.0 / math.sqrt(D))
     assert_close_bf16(out, ref, frac=2e-3)
+
+
+def test_sliding_window_prefill_and_decode(kernels):
+    B, Hkv, D, ps, G, W = 2, 2, 128, 16, 4, 32
+    Hq = G * Hkv
+    ctx = [200, 77]
+    q_lens = [200, 1]
+    k_cache, v_cache, bt = _mk_paged(B, Hkv, D, ps, ctx, seed=11)
+    T = sum(q_lens)
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+    seq_lens = torch.tensor(ctx, dtype=torch.int32, device="cuda")
+    qsl = torch.tensor([0, 200, 201], dtype=torch.int32, device="cuda")
+    from gllm_amd import ops
+    out = ops.paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl,
+                              1.0 / math.sqrt(D), max_query_len=200,
+                              sliding_window=W)
+    ref = R.paged_attention(q.float().cpu(), k_cache.float().cpu(),
+                            v_cache.float().cpu(), bt.cpu(), seq_lens.cpu(),
+                            qsl.cpu(), 1.0 / math.sqrt(D), sliding_window=W)
+    assert_close_bf16(out, ref, frac=2e-3)
+    # decode-only path
+    q1 = torch.randn(B, Hq, D, dtype=torch.bfloat16, device="cuda")
+    qsl1 = torch.arange(B + 1, dtype=torch.int32, device="cuda")
+    out1 = ops.paged_attention(q1, k_cache, v_cache, bt, seq_lens, qsl1,
+                               1.0 / math.sqrt(D), max_query_len=1,
+                               sliding_window=W)
+    ref1 = R.paged_attention(q1.float().cpu(), k_cache.float().cpu(),
+                             v_cache.float().cpu(), bt.cpu(),
+                             seq_lens.cpu(), qsl1.cpu(),
+                             1.0 / math.sqrt(D), sliding_window=W)
+    assert_close_bf16(out1, ref1, frac=2e-3)
