@@ -441,8 +441,11 @@ class Scheduler:
 
     def finalize_output(self, batch: ScheduledBatch, tokens: List[int],
                         records) -> List[Sequence]:
-        """Replace placeholders with real tokens; run finish checks."""
+        """Replace placeholders with real tokens; run finish checks.
+        Also fills ``self.last_emissions`` with (seq_id, token, finish)
+        tuples for the serving worker's output stream."""
         finished: List[Sequence] = []
+        self.last_emissions = []
         for (i, seq, pos) in records:
             already_finished = seq.is_finished
             if already_finished:
@@ -462,6 +465,12 @@ class Scheduler:
                 del seq.token_ids[pos + 1:]
             elif isinstance(self.mm, PrefixMemoryManager):
                 self.mm.register_computed_pages(seq)
+            if seq.finish_reason == Sequence.FINISH_ABORT:
+                self.last_emissions.append((seq.seq_id, -1,
+                                            seq.finish_reason))
+            else:
+                self.last_emissions.append((seq.seq_id, seq.token_ids[pos],
+                                            seq.finish_reason))
             if seq.is_finished:
                 finished.append(seq)
                 # drop from the decode queue if parked there

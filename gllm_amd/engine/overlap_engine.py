@@ -35,6 +35,9 @@ class OverlapEngine(PPEngine):
             self._pinned = [torch.zeros(config.maxd, dtype=torch.long)
                             .pin_memory() for _ in range(self.ring_slots)]
         self._finished_since: List[Sequence] = []
+        # serving hook: called after each finalize with the scheduler's
+        # last_emissions list
+        self.on_finalized = None
 
     # ------------------------------------------------------------------
     def _launch_overlap(self, batch) -> None:
@@ -63,6 +66,9 @@ class OverlapEngine(PPEngine):
         tokens = pinned[:B].tolist()
         finished = self.scheduler.finalize_output(batch, tokens, records)
         self._finished_since.extend(finished)
+        if self.on_finalized is not None and \
+                getattr(self.scheduler, "last_emissions", None):
+            self.on_finalized(self.scheduler.last_emissions)
         return len(records)
 
     # ------------------------------------------------------------------

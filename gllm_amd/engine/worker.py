@@ -27,14 +27,17 @@ import torch.distributed as dist
 
 from gllm_amd.config import EngineConfig
 from gllm_amd.engine.ipc import WorkerComm
+from gllm_amd.engine.overlap_engine import OverlapEngine
 from gllm_amd.engine.pp_engine import PPEngine
 from gllm_amd.logger import logger
 from gllm_amd.sequence import SamplingParams, Sequence
 
 
-class ServingWorker(PPEngine):
-    def __init__(self, config: EngineConfig, req_queue, out_queue):
-        super().__init__(config)
+class ServingMixin:
+    """Intake sync, control commands and output emission shared by the
+    sync (PP) and overlap serving workers."""
+
+    def _init_serving(self, config: EngineConfig, req_queue, out_queue):
         from gllm_amd.parallel import get_rank, get_world_size
         self.rank = get_rank()
         self.world = get_world_size()
@@ -105,6 +108,17 @@ class ServingWorker(PPEngine):
                 self._apply_messages(self.comm.recv_blocking(n))
 
     # ------------------------------------------------------------------
+    def close(self):
+        self.comm.close()
+
+
+class ServingWorker(ServingMixin, PPEngine):
+    """Synchronous serving worker (any PP size)."""
+
+    def __init__(self, config: EngineConfig, req_queue, out_queue):
+        PPEngine.__init__(self, config)
+        self._init_serving(config, req_queue, out_queue)
+
     def _complete_oldest(self) -> List[Sequence]:
         batch, _ = self.inflight[0]
         finished = super()._complete_oldest()
@@ -124,9 +138,8 @@ class ServingWorker(PPEngine):
                 self.comm.send_output(("out", outs, {}))
         return finished
 
-    # ------------------------------------------------------------------
     def run_loop(self) -> None:
-        logger.info("worker %d ready (pp=%d)", self.rank, self.pp_size)
+        logger.info("worker %d ready (pp=%d, sync)", self.rank, self.pp_size)
         while not self.shutdown:
             has_work = self.scheduler.has_work()
             self._sync_intake(block=not has_work)
@@ -146,6 +159,33 @@ class ServingWorker(PPEngine):
         logger.info("worker %d shut down", self.rank)
 
 
+class OverlapServingWorker(ServingMixin, OverlapEngine):
+    """PP=1 serving worker on the launch-first/collect-later engine."""
+
+    def __init__(self, config: EngineConfig, req_queue, out_queue):
+        OverlapEngine.__init__(self, config)
+        self._init_serving(config, req_queue, out_queue)
+        if self.is_output_rank:
+            self.on_finalized = self._emit
+
+    def _emit(self, emissions) -> None:
+        self.comm.send_output(("out", list(emissions), {}))
+
+    def run_loop(self) -> None:
+        logger.info("worker %d ready (overlap)", self.rank)
+        while not self.shutdown:
+            has_work = self.scheduler.has_work() or bool(self.pending)
+            self._sync_intake(block=not has_work)
+            if self.shutdown:
+                break
+            if not (self.scheduler.has_work() or self.pending):
+                continue
+            self.step_tick()
+        self.drain()
+        self.comm.close()
+        logger.info("worker %d shut down", self.rank)
+
+
 def run_worker(rank: int, config: EngineConfig, req_queue, out_queue,
                ready_queue=None) -> None:
     os.environ["RANK"] = str(rank)
@@ -155,7 +195,10 @@ def run_worker(rank: int, config: EngineConfig, req_queue, out_queue,
     if config.device.startswith("cuda"):
         config.device = f"cuda:{rank % max(1, torch.cuda.device_count())}"
     try:
-        worker = ServingWorker(config, req_queue, out_queue)
+        use_overlap = (config.world_size == 1 and config.enable_overlap
+                       and config.device.startswith("cuda"))
+        cls = OverlapServingWorker if use_overlap else ServingWorker
+        worker = cls(config, req_queue, out_queue)
         if ready_queue is not None:
             ready_queue.put(("ready", rank))
         worker.run_loop()
