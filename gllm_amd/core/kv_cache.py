@@ -27,15 +27,19 @@ from gllm_amd.utils.id_allocator import IDAllocator
 class KVCacheSpec:
     num_layers: int          # layers on THIS pipeline stage
     num_kv_heads: int        # per TP rank
-    head_dim: int
+    head_dim: int            # K head dim
     page_size: int
     dtype_bytes: int = 2     # bf16
+    v_head_dim: int = 0      # 0 => same as head_dim (MLA: Dv != Dk)
+
+    def __post_init__(self):
+        if not self.v_head_dim:
+            self.v_head_dim = self.head_dim
 
     @property
     def bytes_per_page(self) -> int:
-        # K + V
-        return (2 * self.num_layers * self.page_size * self.num_kv_heads
-                * self.head_dim * self.dtype_bytes)
+        return (self.num_layers * self.page_size * self.num_kv_heads
+                * (self.head_dim + self.v_head_dim) * self.dtype_bytes)
 
 
 class MemoryManager:

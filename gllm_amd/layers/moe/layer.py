@@ -88,9 +88,15 @@ class FusedMoE(nn.Module):
     # ---- forward ----
     def forward(self, x: torch.Tensor,
                 router_logits: torch.Tensor) -> torch.Tensor:
-        T = x.shape[0]
         weights, ids = ops.topk_softmax(router_logits, self.top_k,
                                         self.renormalize)  # [T,K], [T,K]
+        return self.forward_routed(x, weights, ids)
+
+    def forward_routed(self, x: torch.Tensor, weights: torch.Tensor,
+                       ids: torch.Tensor) -> torch.Tensor:
+        """Expert compute with externally computed routing (DeepSeek
+        grouped-topk / routed scaling paths)."""
+        T = x.shape[0]
         weights = weights.to(x.dtype)
         out = torch.zeros_like(x)
         flat_ids = ids.long().flatten()                    # [T*K]

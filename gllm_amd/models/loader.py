@@ -24,7 +24,11 @@ def get_arch_registry():
                                             Qwen3MoeForCausalLM)
     from gllm_amd.models.chatglm import ChatGLMForCausalLM
     from gllm_amd.models.llama_family import MistralForCausalLM
+    from gllm_amd.models.deepseek_v2 import (DeepseekV2ForCausalLM,
+                                             DeepseekV3ForCausalLM)
     return {
+        "DeepseekV2ForCausalLM": DeepseekV2ForCausalLM,
+        "DeepseekV3ForCausalLM": DeepseekV3ForCausalLM,
         "ChatGLMModel": ChatGLMForCausalLM,
         "ChatGLMForConditionalGeneration": ChatGLMForCausalLM,
         "MistralForCausalLM": MistralForCausalLM,

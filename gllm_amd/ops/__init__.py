@@ -127,6 +127,27 @@ def topk_softmax(gating: torch.Tensor, topk: int, renormalize: bool = True):
     return torch_ref.topk_softmax(gating, topk, renormalize)
 
 
+def grouped_topk(scores, topk, n_group, topk_group, renormalize=True,
+                 scoring="softmax", e_bias=None):
+    dev = scores.device
+    w, i = torch_ref.grouped_topk(scores.cpu(), topk, n_group, topk_group,
+                                  renormalize, scoring,
+                                  e_bias.cpu() if e_bias is not None
+                                  else None)
+    return w.to(dev), i.to(dev)
+
+
+# --------------------------------------------------------------- mla
+def mla_paged_attention(q, k_cache, v_cache, block_table, seq_lens,
+                        query_start_loc, scale):
+    if q.is_cuda:
+        raise NotImplementedError(
+            "MLA paged attention HIP kernel lands in round 2 — the "
+            "DeepSeek family currently runs on the CPU reference path")
+    return torch_ref.mla_paged_attention(q, k_cache, v_cache, block_table,
+                                         seq_lens, query_start_loc, scale)
+
+
 # --------------------------------------------------------------- gemm
 _SKINNY_WS: dict = {}
 SKINNY_MAX_M = 256

@@ -131,6 +131,71 @@ def paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
     return out
 
 
+def mla_paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
+                        v_cache: torch.Tensor, block_table: torch.Tensor,
+                        seq_lens: torch.Tensor,
+                        query_start_loc: torch.Tensor,
+                        scale: float) -> torch.Tensor:
+    """Varlen causal attention with asymmetric head dims (MLA):
+    q/k have Dk (nope+rope), v has Dv. Same paged layout as
+    paged_attention; returns [T, H, Dv]."""
+    T, H, Dk = q.shape
+    Dv = v_cache.shape[3]
+    page_size = k_cache.shape[1]
+    out = torch.empty(T, H, Dv, dtype=q.dtype, device=q.device)
+    B = seq_lens.shape[0]
+    for b in range(B):
+        qs, qe = int(query_start_loc[b]), int(query_start_loc[b + 1])
+        q_len = qe - qs
+        if q_len == 0:
+            continue
+        s_len = int(seq_lens[b])
+        n_pages = -(-s_len // page_size)
+        pages = block_table[b, :n_pages].long()
+        k = k_cache[pages].reshape(-1, H, Dk)[:s_len].float()
+        v = v_cache[pages].reshape(-1, H, Dv)[:s_len].float()
+        qq = q[qs:qe].float()
+        scores = torch.einsum("lhd,shd->hls", qq, k) * scale
+        past = s_len - q_len
+        pos_q = torch.arange(q_len).unsqueeze(1) + past
+        pos_k = torch.arange(s_len).unsqueeze(0)
+        scores.masked_fill_(~(pos_k <= pos_q).unsqueeze(0), float("-inf"))
+        p = torch.softmax(scores, dim=-1)
+        o = torch.einsum("hls,shd->lhd", p, v)
+        out[qs:qe] = o.to(out.dtype)
+    return out
+
+
+def grouped_topk(scores: torch.Tensor, topk: int, n_group: int,
+                 topk_group: int, renormalize: bool = True,
+                 scoring: str = "softmax",
+                 e_bias: Optional[torch.Tensor] = None):
+    """DeepSeek grouped routing (reference: layers/moe/topk.py
+    group_limited_greedy / noaux_tc). ``e_bias`` set => noaux_tc: sigmoid
+    scores + bias for SELECTION, original sigmoid scores as weights."""
+    T, E = scores.shape
+    if scoring == "sigmoid":
+        probs = scores.float().sigmoid()
+    else:
+        probs = torch.softmax(scores.float(), dim=-1)
+    sel = probs + e_bias.float() if e_bias is not None else probs
+    g = sel.view(T, n_group, E // n_group)
+    if e_bias is not None:
+        group_scores = g.topk(2, dim=-1)[0].sum(-1)     # noaux_tc
+    else:
+        group_scores = g.max(dim=-1).values
+    grp_idx = group_scores.topk(topk_group, dim=-1)[1]  # [T, topk_group]
+    mask = torch.zeros(T, n_group, dtype=torch.bool)
+    mask.scatter_(1, grp_idx, True)
+    mask = mask.unsqueeze(-1).expand(T, n_group, E // n_group)
+    sel = sel.masked_fill(~mask.reshape(T, E), float("-inf"))
+    ids = sel.topk(topk, dim=-1)[1]
+    weights = probs.gather(1, ids)
+    if renormalize:
+        weights = weights / weights.sum(-1, keepdim=True).clamp_min(1e-20)
+    return weights, ids.to(torch.int32)
+
+
 def topk_softmax(gating: torch.Tensor, topk: int, renormalize: bool = True):
     """MoE routing: softmax then top-k. Returns (weights [T,K], ids [T,K])."""
     probs = torch.softmax(gating.float(), dim=-1)

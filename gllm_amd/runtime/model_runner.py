@@ -75,8 +75,16 @@ class ModelRunner:
         # heads differently, e.g. ChatGLM's multi_query_group_num)
         attn = getattr(self.model, "kv_geometry", None)
         if attn is not None:
-            kv_per_rank, head_dim = attn
-        else:
+            if len(attn) == 3:
+                kv_per_rank, head_dim, v_dim = attn
+            else:
+                kv_per_rank, head_dim = attn
+                v_dim = head_dim
+            return KVCacheSpec(num_local_layers, kv_per_rank, head_dim,
+                               self.config.page_size,
+                               dtype_bytes=self.kv_dtype.itemsize,
+                               v_head_dim=v_dim)
+        if True:
             total_kv = getattr(hf, "num_key_value_heads",
                                hf.num_attention_heads)
             from gllm_amd.parallel import get_tp_size
@@ -139,11 +147,14 @@ class ModelRunner:
 
     def _allocate_kv(self, num_pages: int):
         spec = self.kv_spec()
-        shape = (num_pages, spec.page_size, spec.num_kv_heads, spec.head_dim)
-        self.k_caches = [torch.zeros(shape, dtype=self.kv_dtype,
+        kshape = (num_pages, spec.page_size, spec.num_kv_heads,
+                  spec.head_dim)
+        vshape = (num_pages, spec.page_size, spec.num_kv_heads,
+                  spec.v_head_dim)
+        self.k_caches = [torch.zeros(kshape, dtype=self.kv_dtype,
                                      device=self.device)
                          for _ in range(spec.num_layers)]
-        self.v_caches = [torch.zeros(shape, dtype=self.kv_dtype,
+        self.v_caches = [torch.zeros(vshape, dtype=self.kv_dtype,
                                      device=self.device)
                          for _ in range(spec.num_layers)]
 
