@@ -78,11 +78,13 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     unpack8<__hip_bfloat16>(p, qreg[h]);
   }
 
-  // ---- per-thread output accumulators: (h, d) pairs
-  constexpr int ACC = (G * D + BLOCK - 1) / BLOCK;  // e.g. 4 for G=8, D=128
-  float acc[ACC];
+  // ---- per-thread output accumulators: (h, d-quad) units so the PV
+  // phase reads V 4 elems per ds_read (8 B) instead of scalar b32
+  constexpr int NQUAD = G * D / 4;
+  constexpr int ACC = (NQUAD + BLOCK - 1) / BLOCK;
+  floatx4 acc[ACC];
 #pragma unroll
-  for (int i = 0; i < ACC; ++i) acc[i] = 0.f;
+  for (int i = 0; i < ACC; ++i) acc[i] = floatx4{0.f, 0.f, 0.f, 0.f};
   if (tid < G) { s_m[tid] = -INFINITY; s_l[tid] = 0.f; }
   __syncthreads();
 
@@ -161,17 +163,26 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     }
     __syncthreads();
 
-    // ---------- phase B2: PV accumulation ----------
+    // ---------- phase B2: PV accumulation (vectorized V reads) ----------
 #pragma unroll
     for (int i = 0; i < ACC; ++i) {
-      const int flat = tid + i * BLOCK;       // (h, d) index
-      if (flat >= G * D) break;
-      const int h = flat / D;
-      const int d = flat % D;
-      float a = acc[i] * s_alpha[h];
+      const int quad = tid + i * BLOCK;       // (h, d/4) index
+      if (quad >= NQUAD) break;
+      const int h = (quad * 4) / D;
+      const int d0 = (quad * 4) % D;
+      const float al = s_alpha[h];
+      floatx4 a = acc[i];
+      a.x *= al; a.y *= al; a.z *= al; a.w *= al;
       for (int t = 0; t < c_len; ++t) {
-        a += s_scores[h][t] *
-             __bfloat162float(v_tile[t][d]);
+        const float p = s_scores[h][t];
+        const shortx4 v4 =
+            *reinterpret_cast<const shortx4 *>(&v_tile[t][d0]);
+        const __hip_bfloat16 *ve =
+            reinterpret_cast<const __hip_bfloat16 *>(&v4);
+        a.x += p * __bfloat162float(ve[0]);
+        a.y += p * __bfloat162float(ve[1]);
+        a.z += p * __bfloat162float(ve[2]);
+        a.w += p * __bfloat162float(ve[3]);
       }
       acc[i] = a;
     }
@@ -181,13 +192,17 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   // ---------- epilogue: normalized split partial + lse ----------
 #pragma unroll
   for (int i = 0; i < ACC; ++i) {
-    const int flat = tid + i * BLOCK;
-    if (flat >= G * D) break;
-    const int h = flat / D;
-    const int d = flat % D;
+    const int quad = tid + i * BLOCK;
+    if (quad >= NQUAD) break;
+    const int h = (quad * 4) / D;
+    const int d0 = (quad * 4) % D;
     const float l = s_l[h];
-    partial_out[(((long)split * gridDim.x + b) * Hq + kvh * G + h) * D + d] =
-        (l > 0.f) ? acc[i] / l : 0.f;
+    const float inv = (l > 0.f) ? 1.f / l : 0.f;
+    float *dst = partial_out +
+        (((long)split * gridDim.x + b) * Hq + kvh * G + h) * D + d0;
+    floatx4 o = acc[i];
+    o.x *= inv; o.y *= inv; o.z *= inv; o.w *= inv;
+    *reinterpret_cast<floatx4 *>(dst) = o;
   }
   if (tid < G) {
     const float l = s_l[tid];
