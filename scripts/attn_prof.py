@@ -31,3 +31,31 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def decode_bench():
+    B, Hq, Hkv, D, ps = 256, 40, 8, 128, 16
+    S = 1024
+    n_pages = B * (S // ps) + 1
+    torch.manual_seed(0)
+    k_cache = torch.randn(n_pages, ps, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v_cache = torch.randn(n_pages, ps, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    bt = torch.arange(1, n_pages, dtype=torch.int32, device="cuda").reshape(B, S // ps)
+    q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device="cuda")
+    seq_lens = torch.full((B,), S, dtype=torch.int32, device="cuda")
+    qsl = torch.arange(B + 1, dtype=torch.int32, device="cuda")
+    sc = D ** -0.5
+    for _ in range(3):
+        ops.paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl, sc, max_query_len=1)
+    torch.cuda.synchronize()
+    t0 = time.time(); it = 30
+    for _ in range(it):
+        ops.paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl, sc, max_query_len=1)
+    torch.cuda.synchronize()
+    dt = (time.time() - t0) / it
+    kv_bytes = 2.0 * B * S * Hkv * D * 2
+    print(f"decode attn B{B}xS{S}: {dt*1e3:.3f} ms  KV {kv_bytes/dt/1e12:.2f} TB/s")
+
+
+if os.environ.get("ATTN_DECODE"):
+    decode_bench()
