@@ -67,6 +67,9 @@ class AsyncLLMEngine:
         self._output_thread: Optional[threading.Thread] = None
         self._stopping = False
         self.loop: Optional[asyncio.AbstractEventLoop] = None
+        self.latest_stats: Dict = {}
+        self.request_counter = 0
+        self.token_counter = 0
 
     @staticmethod
     def _load_tokenizer(model_path: str):
@@ -117,8 +120,12 @@ class AsyncLLMEngine:
             if msg is None:
                 continue
             kind, outs, _stats = msg
+            if kind == "stats":
+                self.latest_stats = outs
+                continue
             if kind != "out":
                 continue
+            self.token_counter += sum(1 for _, t, _f in outs if t >= 0)
             for seq_id, token_id, finish in outs:
                 st = self.requests.get(seq_id)
                 if st is None or st.finished:
@@ -155,6 +162,7 @@ class AsyncLLMEngine:
             st = RequestState(seq_id, len(token_ids), sampling,
                               self.tokenizer, loop)
             self.requests[seq_id] = st
+            self.request_counter += 1
             self.comm.send_to_all("req", {
                 "seq_id": seq_id,
                 "token_ids": token_ids,

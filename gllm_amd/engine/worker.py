@@ -34,10 +34,11 @@ from gllm_amd.sequence import SamplingParams, Sequence
 
 
 class ServingMixin:
-    """Intake sync, control commands and output emission shared by the
-    sync (PP) and overlap serving workers."""
+    """Intake sync, control commands, stats and output emission shared by
+    the sync (PP) and overlap serving workers."""
 
     def _init_serving(self, config: EngineConfig, req_queue, out_queue):
+        self._last_stats = 0.0
         from gllm_amd.parallel import get_rank, get_world_size
         self.rank = get_rank()
         self.world = get_world_size()
@@ -108,6 +109,25 @@ class ServingMixin:
                 self._apply_messages(self.comm.recv_blocking(n))
 
     # ------------------------------------------------------------------
+    def _maybe_send_stats(self) -> None:
+        if not self.is_output_rank:
+            return
+        now = time.time()
+        if now - self._last_stats < 1.0:
+            return
+        self._last_stats = now
+        sched = self.scheduler
+        mm = self.runner.memory_manager
+        stats = {
+            "num_waiting": len(sched.seqs_to_prefill),
+            "num_running": sched.get_num_decode_seqs(),
+            "num_preempted_total": sched.num_preempt_seqs,
+            "kv_memory_util_pct": round(mm.get_memory_util(), 2),
+            "kv_pages_free": mm.get_num_free_pages(),
+            "prefix_cache_hit_rate_pct": round(mm.get_cache_hit_rate(), 2),
+        }
+        self.comm.send_output(("stats", stats, {}))
+
     def close(self):
         self.comm.close()
 
@@ -154,6 +174,7 @@ class ServingWorker(ServingMixin, PPEngine):
                 self._launch(b)
             if self.inflight:
                 self._complete_oldest()
+            self._maybe_send_stats()
         self.drain()
         self.comm.close()
         logger.info("worker %d shut down", self.rank)
@@ -181,6 +202,7 @@ class OverlapServingWorker(ServingMixin, OverlapEngine):
             if not (self.scheduler.has_work() or self.pending):
                 continue
             self.step_tick()
+            self._maybe_send_stats()
         self.drain()
         self.comm.close()
         logger.info("worker %d shut down", self.rank)

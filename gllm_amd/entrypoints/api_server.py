@@ -50,6 +50,30 @@ def build_app():
                 "dp_size": c.dp_size, "schedule_method": c.schedule_method,
                 "page_size": c.page_size}
 
+    @app.get("/stats")
+    async def stats():
+        return {"engine": engine.latest_stats,
+                "requests_total": engine.request_counter,
+                "tokens_total": engine.token_counter,
+                "active_requests": len(engine.requests)}
+
+    @app.get("/metrics")
+    async def metrics():
+        from fastapi.responses import PlainTextResponse
+        st = engine.latest_stats or {}
+        lines = [
+            "# TYPE gllm_requests_total counter",
+            f"gllm_requests_total {engine.request_counter}",
+            "# TYPE gllm_generated_tokens_total counter",
+            f"gllm_generated_tokens_total {engine.token_counter}",
+            "# TYPE gllm_active_requests gauge",
+            f"gllm_active_requests {len(engine.requests)}",
+        ]
+        for k, v in st.items():
+            lines.append(f"# TYPE gllm_{k} gauge")
+            lines.append(f"gllm_{k} {v}")
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.get("/v1/models")
     async def models():
         return ModelList(data=[ModelCard(id=served_model)]).model_dump()
