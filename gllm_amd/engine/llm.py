@@ -52,17 +52,9 @@ class LLM:
 
     @staticmethod
     def _load_tokenizer(model_path: str):
-        if not model_path:
-            return None
-        import os
-        if not any(os.path.exists(os.path.join(model_path, f))
-                   for f in ("tokenizer.json", "tokenizer.model",
-                             "tokenizer_config.json")):
-            return None
+        from gllm_amd.utils.tokenizer import load_tokenizer
         try:
-            from transformers import AutoTokenizer
-            return AutoTokenizer.from_pretrained(model_path,
-                                                 trust_remote_code=True)
+            return load_tokenizer(model_path)
         except Exception as e:  # pragma: no cover
             logger.warning("tokenizer load failed: %s", e)
             return None
@@ -134,6 +126,7 @@ class LLM:
     def chat(self, messages, sampling_params: Optional[SamplingParams] = None
              ) -> RequestOutput:
         assert self.tokenizer is not None
-        ids = self.tokenizer.apply_chat_template(
-            messages, add_generation_prompt=True)
+        text = self.tokenizer.apply_chat_template(
+            messages, add_generation_prompt=True, tokenize=False)
+        ids = self.tokenizer.encode(text)
         return self.generate([ids], [sampling_params or SamplingParams()])[0]

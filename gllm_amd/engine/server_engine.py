@@ -73,14 +73,8 @@ class AsyncLLMEngine:
 
     @staticmethod
     def _load_tokenizer(model_path: str):
-        import os
-        if not any(os.path.exists(os.path.join(model_path, f))
-                   for f in ("tokenizer.json", "tokenizer.model",
-                             "tokenizer_config.json")):
-            return None
-        from transformers import AutoTokenizer
-        return AutoTokenizer.from_pretrained(model_path,
-                                             trust_remote_code=True)
+        from gllm_amd.utils.tokenizer import load_tokenizer
+        return load_tokenizer(model_path)
 
     # ------------------------------------------------------------------
     def start(self) -> None:
@@ -205,5 +199,11 @@ class AsyncLLMEngine:
 
     def apply_chat_template(self, messages, **kwargs) -> List[int]:
         assert self.tokenizer is not None
-        return self.tokenizer.apply_chat_template(
-            messages, add_generation_prompt=True, **kwargs)
+        # render to text then encode: apply_chat_template's tokenize=True
+        # return type varies across transformers versions (list vs dict)
+        text = self.tokenizer.apply_chat_template(
+            messages, add_generation_prompt=True, tokenize=False, **kwargs)
+        ids = self.tokenizer.encode(text)
+        if not ids:
+            raise ValueError("empty prompt after tokenization")
+        return ids

@@ -1,0 +1,100 @@
+"""HTTP-level API server test: FastAPI TestClient against a real CPU
+engine with a tiny WordLevel tokenizer built on the fly."""
+
+import json
+
+import pytest
+
+
+def _mk_model_dir(tmp_path):
+    d = tmp_path / "tinytok"
+    d.mkdir()
+    cfg = {
+        "architectures": ["Qwen2ForCausalLM"],
+        "model_type": "qwen2",
+        "hidden_size": 64,
+        "intermediate_size": 128,
+        "num_hidden_layers": 2,
+        "num_attention_heads": 4,
+        "num_key_value_heads": 2,
+        "vocab_size": 128,
+        "max_position_embeddings": 2048,
+        "rms_norm_eps": 1e-6,
+        "rope_theta": 10000.0,
+        "eos_token_id": 1,
+    }
+    with open(d / "config.json", "w") as f:
+        json.dump(cfg, f)
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+    vocab = {"[UNK]": 0, "</s>": 1}
+    for i in range(2, 128):
+        vocab[f"w{i}"] = i
+    tok = Tokenizer(WordLevel(vocab, unk_token="[UNK]"))
+    tok.pre_tokenizer = Whitespace()
+    tok.save(str(d / "tokenizer.json"))
+    with open(d / "tokenizer_config.json", "w") as f:
+        json.dump({
+            "tokenizer_class": "PreTrainedTokenizerFast",
+            "eos_token": "</s>",
+            "unk_token": "[UNK]",
+            "model_max_length": 2048,
+            "chat_template": (
+                "{% for m in messages %}{{ m.content }} {% endfor %}"),
+        }, f)
+    return str(d)
+
+
+@pytest.mark.timeout(300)
+def test_api_server_routes(tmp_path):
+    from fastapi.testclient import TestClient
+    import gllm_amd.entrypoints.api_server as srv
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+
+    d = _mk_model_dir(tmp_path)
+    cfg = EngineConfig(model=d, load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       master_port=29690,
+                       enable_prefix_caching=False)
+    srv.engine = AsyncLLMEngine(cfg)
+    srv.served_model = "tiny"
+    srv.engine.start()
+    try:
+        app = srv.build_app()
+        client = TestClient(app)
+        assert client.get("/health").json()["status"] == "ok"
+        assert client.get("/v1/models").json()["data"][0]["id"] == "tiny"
+        assert "pp_size" in client.get("/server_info").json()
+
+        r = client.post("/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "w5 w6 w7"}],
+            "max_tokens": 6, "temperature": 0.0, "ignore_eos": True})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["choices"][0]["message"]["role"] == "assistant"
+        assert body["usage"]["completion_tokens"] == 6
+
+        # streaming
+        with client.stream("POST", "/v1/chat/completions", json={
+                "messages": [{"role": "user", "content": "w9 w10"}],
+                "max_tokens": 4, "temperature": 0.0, "stream": True,
+                "ignore_eos": True}) as r2:
+            lines = [ln for ln in r2.iter_lines() if ln]
+        assert lines[-1] == "data: [DONE]"
+        assert len([l for l in lines if l.startswith("data: {")]) >= 4
+
+        # completions
+        r3 = client.post("/v1/completions", json={
+            "prompt": "w3 w4", "max_tokens": 3, "temperature": 0.0,
+            "ignore_eos": True})
+        assert r3.status_code == 200, r3.text
+        assert r3.json()["usage"]["completion_tokens"] == 3
+
+        # metrics + stats
+        m = client.get("/metrics").text
+        assert "gllm_requests_total" in m
+        assert client.get("/stats").json()["requests_total"] >= 3
+    finally:
+        srv.engine.stop()
