@@ -65,6 +65,9 @@ def main():
                     choices=["qwen2.5-32b", "debug"])
     ap.add_argument("--page-size", type=int, default=16)
     ap.add_argument("--schedule", type=str, default="token_throttling")
+    ap.add_argument("--qps", type=float, default=0.0,
+                    help=">0: pace request arrivals at this rate for the "
+                         "TTFT measurement instead of a burst")
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--no-overlap", action="store_true")
     ap.add_argument("--max-graph-bs", type=int, default=64)
@@ -114,18 +117,40 @@ def main():
                             max_tokens=4096)
         s = Sequence(i, ids, sp, eos_token_id=None, arrival_time=t_submit)
         seqs.append(s)
-    eng.add_requests(seqs)
-
-    # ---- ramp: run until every seq has produced its first token ----
+    # ---- ramp: arrivals (burst or paced at --qps) until every seq has
+    # produced its first token; TTFT measured from each seq's arrival ----
     ttfts = {}
-    while len(ttfts) < len(seqs):
-        eng.step_tick()
-        now = time.time()
-        for s in seqs:
-            if s.seq_id not in ttfts and s.num_output_tokens > 0:
-                ttfts[s.seq_id] = (now - t_submit) * 1000.0
-        if not eng.scheduler.has_work():
-            break
+    arrivals = {}
+    if args.qps > 0:
+        to_release = list(seqs)
+        next_t = time.time()
+        while len(ttfts) < len(seqs):
+            now = time.time()
+            while to_release and now >= next_t:
+                s = to_release.pop(0)
+                arrivals[s.seq_id] = now
+                eng.add_requests([s])
+                next_t += 1.0 / args.qps
+                now = time.time()
+            eng.step_tick()
+            now = time.time()
+            for s in seqs:
+                if s.seq_id in arrivals and s.seq_id not in ttfts \
+                        and s.num_output_tokens > 0:
+                    ttfts[s.seq_id] = (now - arrivals[s.seq_id]) * 1000.0
+            if not to_release and not eng.scheduler.has_work() and \
+                    not getattr(eng, "pending", None):
+                break
+    else:
+        eng.add_requests(seqs)
+        while len(ttfts) < len(seqs):
+            eng.step_tick()
+            now = time.time()
+            for s in seqs:
+                if s.seq_id not in ttfts and s.num_output_tokens > 0:
+                    ttfts[s.seq_id] = (now - t_submit) * 1000.0
+            if not eng.scheduler.has_work():
+                break
     ttft_sorted = sorted(ttfts.values())
     ttft_p50 = ttft_sorted[len(ttft_sorted) // 2] if ttft_sorted else None
 
@@ -166,6 +191,7 @@ def main():
             "dtype": "bf16" if use_gpu else "fp32",
             "data": "synthetic",
             "ttft_p50_ms": round(ttft_p50, 1) if ttft_p50 else None,
+            "ttft_qps": args.qps if args.qps > 0 else "burst",
             "config": {
                 "model": "Qwen2.5-32B" if args.model == "qwen2.5-32b"
                 else "debug-0.2B",
