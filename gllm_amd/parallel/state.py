@@ -163,19 +163,19 @@ def tensor_parallel_all_reduce(t: torch.Tensor) -> torch.Tensor:
 
 
 def tensor_parallel_all_gather(t: torch.Tensor, dim: int = -1) -> torch.Tensor:
+    """All-gather across TP, concatenating along ``dim`` (0 or -1)."""
     if _TP_SIZE == 1:
         return t
-    if dim != 0:
-        t = t.transpose(0, dim).contiguous() if dim not in (-1,) else t
-    if dim in (-1, t.dim() - 1):
-        out = torch.empty((_TP_SIZE,) + tuple(t.shape), dtype=t.dtype,
-                          device=t.device)
-        dist.all_gather_into_tensor(out, t.contiguous(), group=_TP_GROUP)
-        return torch.cat([out[i] for i in range(_TP_SIZE)], dim=-1)
-    out = torch.empty((_TP_SIZE,) + tuple(t.shape), dtype=t.dtype,
-                      device=t.device)
-    dist.all_gather_into_tensor(out, t.contiguous(), group=_TP_GROUP)
-    return out.flatten(0, 1)
+    t = t.contiguous()
+    # all_gather_into_tensor concatenates along dim 0 of a flat output
+    out = torch.empty((_TP_SIZE * t.shape[0],) + tuple(t.shape[1:]),
+                      dtype=t.dtype, device=t.device)
+    dist.all_gather_into_tensor(out, t, group=_TP_GROUP)
+    if dim == 0:
+        return out
+    assert dim in (-1, t.dim() - 1), "only dim 0 / -1 supported"
+    chunks = out.reshape((_TP_SIZE,) + tuple(t.shape))
+    return torch.cat(list(chunks.unbind(0)), dim=-1)
 
 
 def ep_all_reduce(t: torch.Tensor) -> torch.Tensor:
