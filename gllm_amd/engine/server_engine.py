@@ -96,7 +96,27 @@ class AsyncLLMEngine:
         self._output_thread = threading.Thread(target=self._output_loop,
                                                daemon=True)
         self._output_thread.start()
+        self._watchdog = threading.Thread(target=self._watch_workers,
+                                          daemon=True)
+        self._watchdog.start()
         logger.info("engine ready: %d worker(s)", self.config.world_size)
+
+    def _watch_workers(self) -> None:
+        """Worker death is fatal (reference worker.py:988-999 /
+        llm_engine.py:349-352): fail every pending stream, then stop."""
+        import time as _t
+        while not self._stopping:
+            for p in self._procs:
+                if not p.is_alive() and p.exitcode not in (0, None):
+                    logger.error("worker %s died (exit %s) — failing %d "
+                                 "pending requests", p.pid, p.exitcode,
+                                 len(self.requests))
+                    for st in list(self.requests.values()):
+                        if not st.finished:
+                            self._handle_token(st, -1, "abort")
+                    self._stopping = True
+                    return
+            _t.sleep(0.5)
 
     def stop(self) -> None:
         self._stopping = True
