@@ -119,6 +119,10 @@ def main():
         seqs.append(s)
     # ---- ramp: arrivals (burst or paced at --qps) until every seq has
     # produced its first token; TTFT measured from each seq's arrival ----
+    # Paced arrivals are wall-clock driven and would desynchronize the
+    # replicated schedulers across ranks: force burst for multi-GPU runs.
+    if world > 1:
+        args.qps = 0.0
     ttfts = {}
     arrivals = {}
     if args.qps > 0:
