@@ -49,6 +49,9 @@ class MemoryManager:
         self.num_pages = num_pages
         self.page_size = page_size
         self.allocator = IDAllocator(num_pages)
+        # auxiliary per-seq resources released with the seq (penalty
+        # mask slots, later SSM slots): list of callables(seq)
+        self.free_hooks = []
 
     # ---- stats ----
     def get_num_free_pages(self) -> int:
@@ -89,6 +92,8 @@ class MemoryManager:
     def free_seq(self, seq: Sequence) -> None:
         self.allocator.free_many(seq.page_table)
         seq.page_table = []
+        for hook in self.free_hooks:
+            hook(seq)
 
     # ---- slot mapping helper ----
     def slots_for(self, seq: Sequence) -> List[int]:
@@ -228,3 +233,5 @@ class PrefixMemoryManager(MemoryManager):
                 self.allocator.free(p)
         seq.page_table = []
         seq.num_cached_pages = 0
+        for hook in self.free_hooks:
+            hook(seq)

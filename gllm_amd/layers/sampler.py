@@ -21,8 +21,14 @@ class SamplingMetadata:
     penalties: torch.Tensor             # [B] float32
     all_greedy: bool
     any_penalty: bool
-    # token history per row for repetition penalty (cpu LongTensors)
+    # token history per row for repetition penalty (cpu LongTensors),
+    # used only as the fallback when no mask pool slot is available
     token_id_rows: Optional[List[torch.Tensor]] = None
+    # persistent mask pool path: pool ref + per-row slot ids (-1 = none)
+    penalty_pool: Optional[object] = None
+    penalty_slots: Optional[torch.Tensor] = None   # [B] long
+    # rows that actually sample a kept token (ends_prompt), host list
+    sample_rows: Optional[List[int]] = None
     max_logprobs: int = 0               # >0 => return top-k logprobs
     generators: Optional[List[Optional[torch.Generator]]] = None
 
@@ -39,9 +45,8 @@ class Sampler(torch.nn.Module):
     def forward(self, logits: torch.Tensor,
                 meta: SamplingMetadata) -> SamplerOutput:
         # logits: [B, V] (already gathered to full vocab)
-        if meta.any_penalty and meta.token_id_rows is not None:
-            logits = torch_ref.apply_repetition_penalty(
-                logits.float(), meta.token_id_rows, meta.penalties)
+        if meta.any_penalty:
+            logits = self._apply_penalties(logits, meta)
         if meta.all_greedy:
             next_tokens = logits.argmax(dim=-1)
             return self._with_logprobs(logits, next_tokens, meta)
@@ -57,6 +62,22 @@ class Sampler(torch.nn.Module):
         if greedy_rows.any():
             sampled = torch.where(greedy_rows, logits.argmax(dim=-1), sampled)
         return self._with_logprobs(logits, sampled, meta)
+
+    @staticmethod
+    def _apply_penalties(logits: torch.Tensor,
+                         meta: SamplingMetadata) -> torch.Tensor:
+        logits = logits.float()
+        if meta.penalty_pool is not None and meta.penalty_slots is not None:
+            slots = meta.penalty_slots.clamp_min(0)
+            rows = meta.penalty_pool.mask[slots, :logits.shape[1]]
+            pen = meta.penalties.to(logits.device).unsqueeze(1)
+            active = (rows != 0) & (pen != 1.0) &                 (meta.penalty_slots.unsqueeze(1) >= 0)
+            adj = torch.where(logits > 0, logits / pen, logits * pen)
+            logits = torch.where(active, adj, logits)
+        if meta.token_id_rows is not None:
+            logits = torch_ref.apply_repetition_penalty(
+                logits, meta.token_id_rows, meta.penalties)
+        return logits
 
     @staticmethod
     def _apply_top_k_top_p(probs: torch.Tensor, top_ks: torch.Tensor,
@@ -105,23 +126,43 @@ class Sampler(torch.nn.Module):
         return SamplerOutput(next_tokens, chosen, topv, topi)
 
 
-def build_sampling_metadata(items, device) -> SamplingMetadata:
-    """Build metadata from the scheduled batch items (one row per item)."""
-    temps, tps, tks, pens, rows, gens = [], [], [], [], [], []
+def build_sampling_metadata(items, device,
+                            penalty_pool=None) -> SamplingMetadata:
+    """Build metadata from the scheduled batch items (one row per item).
+
+    With a ``penalty_pool``, penalized seqs get persistent device mask
+    slots (seeded from the prompt on first use, updated on-GPU after
+    every sample); the host token-row fallback covers pool exhaustion."""
+    temps, tps, tks, pens, gens = [], [], [], [], []
+    rows: List[torch.Tensor] = []
+    slots: List[int] = []
+    sample_rows: List[int] = []
     max_lp = 0
     any_pen = False
-    for it in items:
+    need_rows = False
+    for i, it in enumerate(items):
         sp = it.seq.sampling
         temps.append(sp.temperature)
         tps.append(sp.top_p)
         tks.append(sp.top_k)
         pens.append(sp.repetition_penalty)
+        if it.ends_prompt:
+            sample_rows.append(i)
+        slot = -1
         if sp.repetition_penalty != 1.0:
             any_pen = True
-            rows.append(torch.tensor(it.seq.token_ids, dtype=torch.long,
-                                     device=device))
+            if penalty_pool is not None:
+                slot = penalty_pool.ensure(it.seq)
+            if slot < 0:
+                need_rows = True
+                rows.append(torch.tensor(it.seq.token_ids,
+                                         dtype=torch.long, device=device))
+            else:
+                rows.append(torch.empty(0, dtype=torch.long,
+                                        device=device))
         else:
             rows.append(torch.empty(0, dtype=torch.long, device=device))
+        slots.append(slot)
         if sp.logprobs:
             max_lp = max(max_lp, sp.logprobs)
         if sp.seed is not None:
@@ -137,6 +178,10 @@ def build_sampling_metadata(items, device) -> SamplingMetadata:
         top_ks=torch.tensor(tks, dtype=torch.int32, device=device),
         penalties=torch.tensor(pens, dtype=torch.float32, device=device),
         all_greedy=all_greedy, any_penalty=any_pen,
-        token_id_rows=rows if any_pen else None,
+        token_id_rows=rows if need_rows else None,
+        penalty_pool=penalty_pool if any_pen else None,
+        penalty_slots=torch.tensor(slots, dtype=torch.long, device=device)
+        if any_pen else None,
+        sample_rows=sample_rows,
         max_logprobs=max_lp,
         generators=gens if any(g is not None for g in gens) else None)

@@ -59,6 +59,11 @@ class ModelRunner:
         self.token_ring = torch.zeros(
             (self.ring_slots, cfg.maxd), dtype=torch.long,
             device=cfg.device)
+        from gllm_amd.core.penalty import PenaltyPool
+        self.penalty_pool = PenaltyPool(
+            cfg.maxd + 64, getattr(self.hf_config, "vocab_size", 32000),
+            cfg.device)
+        self.memory_manager.free_hooks.append(self.penalty_pool.free)
         from gllm_amd.parallel import get_pp_size, get_tp_size
         if (cfg.use_graph and cfg.device.startswith("cuda")
                 and get_pp_size() == 1 and get_tp_size() == 1):
@@ -203,6 +208,20 @@ class ModelRunner:
 
     def _sample(self, batch: ScheduledBatch, hidden, fctx):
         logits = self.model.compute_logits(hidden, fctx)
-        meta = build_sampling_metadata(batch.items, logits.device)
+        meta = build_sampling_metadata(batch.items, logits.device,
+                                       penalty_pool=self.penalty_pool)
         out = self.sampler(logits, meta)
+        if meta.any_penalty and meta.penalty_slots is not None:
+            # GPU-side mask update for rows that really sampled (no host
+            # sync: tokens may still be in flight under overlap)
+            rows = [i for i in meta.sample_rows
+                    if int(meta.penalty_slots[i]) >= 0]                 if not logits.is_cuda else meta.sample_rows
+            if rows:
+                idx = torch.tensor(rows, dtype=torch.long,
+                                   device=logits.device)
+                slots = meta.penalty_slots.index_select(0, idx)
+                ok = slots >= 0
+                self.penalty_pool.append(slots[ok],
+                                         out.next_tokens.index_select(
+                                             0, idx)[ok])
         return out

@@ -143,3 +143,21 @@ def test_more_arches_generate(tmp_path, cfg_json):
     assert len(out[0].token_ids) == 6
     out2 = llm2.generate([list(range(1, 30))], [greedy(6)])
     assert out2[0].token_ids == out[0].token_ids
+
+
+def test_penalty_pool_used_and_freed(llm):
+    pen = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True,
+                         repetition_penalty=2.0)
+    pool = llm.runner.penalty_pool
+    free0 = pool.alloc.num_free
+    out = llm.generate([[3, 4, 5]], [pen])[0]
+    assert len(out.token_ids) == 6
+    # slot released with the seq
+    assert pool.alloc.num_free == free0
+    # mask rows contain prompt + generated tokens for the next alloc
+    from gllm_amd.sequence import Sequence
+    s = Sequence(123, [3, 4, 5], pen)
+    slot = pool.ensure(s)
+    assert slot >= 0
+    assert pool.mask[slot, 3] == 1 and pool.mask[slot, 4] == 1
+    pool.free(s)
