@@ -44,10 +44,11 @@ DEV_INLINE int swz(int row, int byte_off) {
 // One glds instruction moves 64 lanes x 16 B = 1 KiB, lane-linear at
 // lds_base + lane*16. The tile image is linear rows of 128 B; the
 // source address carries the inverse swizzle.
+template <int AUX>
 DEV_INLINE void glds16(const __hip_bfloat16 *gsrc, char *lds_ptr) {
   __builtin_amdgcn_global_load_lds(
       reinterpret_cast<const unsigned int *>(gsrc),
-      reinterpret_cast<unsigned int *>(lds_ptr), 16, 0, 0);
+      reinterpret_cast<unsigned int *>(lds_ptr), 16, 0, AUX);
 }
 
 template <int MB, int RING>
@@ -95,12 +96,21 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
     }
   }
 
+  // W rows are streamed ONCE per CU -> non-temporal (aux=2) lifts the
+  // per-CU LDS-DMA landing cadence (guide row nt-weights: -18% landed
+  // latency, chip 6.5-6.8 TB/s). x is re-read by every N-workgroup
+  // through L2 -> default policy.
   auto stage = [&](int kt, int slot) {
     const int kb = k_begin + kt * BK;
     char *base = smem + slot * TILE_B;
 #pragma unroll
-    for (int j = 0; j < GL_PER_WAVE; ++j)
-      glds16(gsrc[j] + kb, base + (wave * GL_PER_WAVE + j) * 1024);
+    for (int j = 0; j < GL_PER_WAVE; ++j) {
+      const int off = (wave * GL_PER_WAVE + j) * 1024;
+      if (off < BN * ROW_B)
+        glds16<2>(gsrc[j] + kb, base + off);
+      else
+        glds16<0>(gsrc[j] + kb, base + off);
+    }
   };
 
   // ---- accumulators ---------------------------------------------------
