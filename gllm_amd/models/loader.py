@@ -26,7 +26,10 @@ def get_arch_registry():
     from gllm_amd.models.llama_family import MistralForCausalLM
     from gllm_amd.models.deepseek_v2 import (DeepseekV2ForCausalLM,
                                              DeepseekV3ForCausalLM)
+    from gllm_amd.models.hybrid_gdn import Qwen3_5ForCausalLM
     return {
+        "Qwen3_5ForCausalLM": Qwen3_5ForCausalLM,
+        "Qwen3NextForCausalLM": Qwen3_5ForCausalLM,
         "DeepseekV2ForCausalLM": DeepseekV2ForCausalLM,
         "DeepseekV3ForCausalLM": DeepseekV3ForCausalLM,
         "ChatGLMModel": ChatGLMForCausalLM,

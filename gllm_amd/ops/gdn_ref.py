@@ -1,0 +1,104 @@
+"""Gated DeltaNet (GDN) linear-attention reference ops.
+
+Parity target: the reference's vendored FLA Triton kernels
+(layers/ops/fla/: fused_gdn_gating, fused_recurrent_gated_delta_rule,
+chunk_gated_delta_rule) and causal_conv1d (layers/ops/mamba/). These
+torch implementations are the round-1 compute path (CPU and GPU) and
+the numerics oracle for the round-2 HIP chunked kernels.
+
+Recurrence per v-head (state S [Dv, Dk]; q/k L2-normalized, q scaled):
+    S  = S * exp(g_t)
+    u  = beta_t * (v_t - S @ k_t)
+    S  = S + outer(u, k_t)
+    o_t = S @ q_t
+GVA: v-heads group over k-heads (Hv = G * Hk); gating g/beta are
+per-v-head.
+"""
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+
+def gdn_gating(A_log: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
+               dt_bias: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """g = -exp(A_log) * softplus(a + dt_bias); beta = sigmoid(b).
+    a, b: [T, Hv]; A_log, dt_bias: [Hv]."""
+    g = -A_log.float().exp() * F.softplus(a.float() + dt_bias.float())
+    beta = torch.sigmoid(b.float())
+    return g, beta
+
+
+def l2norm(x: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    return x / torch.sqrt((x * x).sum(-1, keepdim=True) + eps)
+
+
+def causal_conv1d_prefill(x: torch.Tensor, weight: torch.Tensor,
+                          conv_state: torch.Tensor,
+                          has_initial_state: bool) -> torch.Tensor:
+    """Depthwise causal conv over one sequence chunk with state carry.
+
+    x: [T, C]; weight: [C, K]; conv_state: [C, K-1] (updated in place).
+    Returns silu(conv(x)) [T, C].
+    """
+    T, C = x.shape
+    K = weight.shape[1]
+    xt = x.float().T                       # [C, T]
+    if has_initial_state:
+        ctx = torch.cat([conv_state.float(), xt], dim=1)
+    else:
+        ctx = torch.cat([torch.zeros(C, K - 1, dtype=torch.float32,
+                                     device=x.device), xt], dim=1)
+    out = F.conv1d(ctx.unsqueeze(0), weight.float().unsqueeze(1),
+                   groups=C).squeeze(0)    # [C, T]
+    # save the last K-1 inputs for the next chunk
+    conv_state.copy_(ctx[:, -(K - 1):].to(conv_state.dtype))
+    return F.silu(out).T.to(x.dtype)
+
+
+def causal_conv1d_update(x: torch.Tensor, weight: torch.Tensor,
+                         conv_state: torch.Tensor) -> torch.Tensor:
+    """Single-token decode step. x: [C]; conv_state [C, K-1] rolls."""
+    K = weight.shape[1]
+    ctx = torch.cat([conv_state.float(), x.float().unsqueeze(1)], dim=1)
+    out = (ctx * weight.float()).sum(-1)
+    conv_state.copy_(ctx[:, 1:].to(conv_state.dtype))
+    return F.silu(out).to(x.dtype)
+
+
+def gated_delta_rule(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     g: torch.Tensor, beta: torch.Tensor, scale: float,
+                     state: torch.Tensor) -> torch.Tensor:
+    """Sequential gated delta rule over one sequence.
+
+    q, k: [T, Hk, Dk]; v: [T, Hv, Dv]; g, beta: [T, Hv];
+    state: [Hv, Dv, Dk] fp32, updated IN PLACE. Returns o [T, Hv, Dv].
+    """
+    T, Hk, Dk = q.shape
+    Hv, Dv = v.shape[1], v.shape[2]
+    G = Hv // Hk
+    qn = (l2norm(q.float()) * scale).repeat_interleave(G, dim=1)
+    kn = l2norm(k.float()).repeat_interleave(G, dim=1)   # [T, Hv, Dk]
+    vf = v.float()
+    S = state.float()
+    outs = []
+    for t in range(T):
+        S = S * g[t].exp().view(Hv, 1, 1)
+        kt = kn[t]                                        # [Hv, Dk]
+        u = vf[t] - torch.einsum("hvk,hk->hv", S, kt)
+        u = u * beta[t].unsqueeze(-1)
+        S = S + torch.einsum("hv,hk->hvk", u, kt)
+        outs.append(torch.einsum("hvk,hk->hv", S, qn[t]))
+    state.copy_(S.to(state.dtype))
+    return torch.stack(outs).to(v.dtype)
+
+
+def rmsnorm_gated(x: torch.Tensor, z: torch.Tensor, weight: torch.Tensor,
+                  eps: float) -> torch.Tensor:
+    """out = rmsnorm(x) * w * silu(z)  (norm applied before the gate;
+    reference RMSNormGated with norm_before_gate=True). x, z: [T, D]."""
+    xf = x.float()
+    var = xf.pow(2).mean(-1, keepdim=True)
+    n = xf * torch.rsqrt(var + eps) * weight.float()
+    return (n * F.silu(z.float())).to(x.dtype)

@@ -32,6 +32,10 @@ class ForwardContext:
     is_profile_run: bool = False
     # overlap mode: token ids contain negative ring placeholders
     has_placeholders: bool = False
+    # hybrid GDN models: per-seq SSM state slots (core/ssm.py)
+    ssm_pool: Optional[object] = None
+    ssm_slots: Optional[List[int]] = None
+    ssm_has_init: Optional[List[bool]] = None
 
     @property
     def is_pure_decode(self) -> bool:

@@ -43,6 +43,12 @@ class ModelRunner:
                                   else "cuda:0")
         self.model, self.hf_config = load_model(cfg, cfg.device)
         self.kv_dtype = cfg.torch_dtype()
+        if getattr(self.model, "ssm_spec", None) is not None:
+            if cfg.enable_prefix_caching:
+                logger.info("hybrid GDN model: disabling prefix caching "
+                            "(state snapshots land in round 2)")
+                cfg.enable_prefix_caching = False
+            cfg.use_graph = False
         num_pages = num_pages_override or self._size_kv_cache()
         self._allocate_kv(num_pages)
         self.num_kv_pages_total = num_pages
@@ -64,6 +70,12 @@ class ModelRunner:
             cfg.maxd + 64, getattr(self.hf_config, "vocab_size", 32000),
             cfg.device)
         self.memory_manager.free_hooks.append(self.penalty_pool.free)
+        self.ssm_pool = None
+        if getattr(self.model, "ssm_spec", None) is not None:
+            from gllm_amd.core.ssm import SSMPool
+            self.ssm_pool = SSMPool(self.model.ssm_spec, cfg.maxd + 64,
+                                    cfg.device, dtype=self.kv_dtype)
+            self.memory_manager.free_hooks.append(self.ssm_pool.free)
         from gllm_amd.parallel import get_pp_size, get_tp_size
         if (cfg.use_graph and cfg.device.startswith("cuda")
                 and get_pp_size() == 1 and get_tp_size() == 1):
@@ -75,7 +87,10 @@ class ModelRunner:
     # ------------------------------------------------------------------
     def kv_spec(self) -> KVCacheSpec:
         hf = self.hf_config
-        num_local_layers = self.model.num_local_layers
+        # hybrid models allocate KV only for their full-attention layers
+        num_local_layers = getattr(self.model, "num_kv_layers", None)
+        if num_local_layers is None:
+            num_local_layers = self.model.num_local_layers
         # prefer the model's own attention geometry (configs name kv
         # heads differently, e.g. ChatGLM's multi_query_group_num)
         attn = getattr(self.model, "kv_geometry", None)
@@ -179,6 +194,7 @@ class ModelRunner:
         tokens, fctx = self.builder.build(
             batch, self.k_caches, self.v_caches,
             need_logits=self.model.is_last_stage)
+        self._attach_ssm(batch, fctx)
         if fctx.has_placeholders:
             tokens = self.resolve_tokens(tokens)
         hidden, residual = self._stage_forward(tokens, fctx.positions, fctx)
@@ -192,12 +208,21 @@ class ModelRunner:
         _, fctx = self.builder.build(
             batch, self.k_caches, self.v_caches,
             need_logits=self.model.is_last_stage)
+        self._attach_ssm(batch, fctx)
         hidden, residual = self._stage_forward(
             None, fctx.positions, fctx, hidden_states=hidden,
             residual=residual)
         if self.model.is_last_stage:
             return self._sample(batch, hidden, fctx)
         return hidden, residual, fctx
+
+    def _attach_ssm(self, batch, fctx) -> None:
+        if self.ssm_pool is None:
+            return
+        fctx.ssm_pool = self.ssm_pool
+        fctx.ssm_slots = [self.ssm_pool.ensure(it.seq)
+                          for it in batch.items]
+        fctx.ssm_has_init = [it.start > 0 for it in batch.items]
 
     def resolve_tokens(self, tokens: torch.Tensor) -> torch.Tensor:
         """Replace negative placeholder ids with sampled tokens from the

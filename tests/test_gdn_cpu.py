@@ -1,0 +1,129 @@
+"""GDN (Gated DeltaNet) reference ops + hybrid model tests."""
+
+import json
+
+import pytest
+import torch
+
+from gllm_amd.ops import gdn_ref as G
+
+
+def test_gating_formula():
+    A_log = torch.tensor([0.0, 1.0])
+    a = torch.zeros(3, 2)
+    b = torch.zeros(3, 2)
+    dt = torch.zeros(2)
+    g, beta = G.gdn_gating(A_log, a, b, dt)
+    sp0 = torch.nn.functional.softplus(torch.tensor(0.0))
+    assert torch.allclose(g[0], torch.tensor([-sp0, -torch.e * sp0]),
+                          atol=1e-5)
+    assert torch.allclose(beta, torch.full((3, 2), 0.5))
+
+
+def test_conv_chunked_equals_full():
+    torch.manual_seed(0)
+    C, K, T = 6, 4, 13
+    w = torch.randn(C, K)
+    x = torch.randn(T, C)
+    st_full = torch.zeros(C, K - 1)
+    full = G.causal_conv1d_prefill(x, w, st_full, False)
+    # two chunks with state carry
+    st = torch.zeros(C, K - 1)
+    o1 = G.causal_conv1d_prefill(x[:5], w, st, False)
+    o2 = G.causal_conv1d_prefill(x[5:], w, st, True)
+    assert torch.allclose(torch.cat([o1, o2]), full, atol=1e-5)
+    # decode updates equal the tail
+    st2 = torch.zeros(C, K - 1)
+    o3 = G.causal_conv1d_prefill(x[:12], w, st2, False)
+    o4 = G.causal_conv1d_update(x[12], w, st2)
+    assert torch.allclose(o4, full[12], atol=1e-5)
+    assert torch.allclose(st2, st, atol=1e-5)
+
+
+def test_delta_rule_chunked_equals_full():
+    torch.manual_seed(1)
+    T, Hk, Hv, Dk, Dv = 11, 2, 4, 8, 6
+    q = torch.randn(T, Hk, Dk)
+    k = torch.randn(T, Hk, Dk)
+    v = torch.randn(T, Hv, Dv)
+    g = -torch.rand(T, Hv) * 0.3
+    beta = torch.sigmoid(torch.randn(T, Hv))
+    S_full = torch.zeros(Hv, Dv, Dk)
+    full = G.gated_delta_rule(q, k, v, g, beta, 0.5, S_full)
+    S = torch.zeros(Hv, Dv, Dk)
+    o1 = G.gated_delta_rule(q[:4], k[:4], v[:4], g[:4], beta[:4], 0.5, S)
+    o2 = G.gated_delta_rule(q[4:], k[4:], v[4:], g[4:], beta[4:], 0.5, S)
+    assert torch.allclose(torch.cat([o1, o2]), full, atol=1e-4)
+    assert torch.allclose(S, S_full, atol=1e-4)
+
+
+HYBRID_TINY = {
+    "architectures": ["Qwen3_5ForCausalLM"],
+    "model_type": "qwen3_5",
+    "hidden_size": 64,
+    "intermediate_size": 128,
+    "num_hidden_layers": 4,
+    "full_attention_interval": 2,     # 2 GDN + 2 full-attn layers
+    "num_attention_heads": 4,
+    "num_key_value_heads": 2,
+    "head_dim": 16,
+    "attn_output_gate": True,
+    "partial_rotary_factor": 0.5,
+    "linear_num_value_heads": 4,
+    "linear_num_key_heads": 2,
+    "linear_key_head_dim": 8,
+    "linear_value_head_dim": 8,
+    "linear_conv_kernel_dim": 4,
+    "vocab_size": 128,
+    "max_position_embeddings": 2048,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 10000.0,
+    "eos_token_id": 0,
+}
+
+
+def _mk_llm(tmp_path, maxp=64, name="h"):
+    d = tmp_path / name
+    d.mkdir(exist_ok=True)
+    with open(d / "config.json", "w") as f:
+        json.dump(HYBRID_TINY, f)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    cfg = EngineConfig(model=str(d), load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=maxp,
+                       enable_prefix_caching=True)  # auto-disabled
+    return LLM(config=cfg, num_pages_override=128)
+
+
+def test_hybrid_generates_and_chunked_prefill_state_carry(tmp_path):
+    from gllm_amd.sequence import SamplingParams
+    sp = [SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)]
+    prompt = [list(range(1, 30))]
+    llm = _mk_llm(tmp_path, maxp=64, name="full")
+    ref = llm.generate(prompt, sp)[0].token_ids
+    assert len(ref) == 6
+    # prefix caching must have been auto-disabled for hybrid
+    from gllm_amd.core.kv_cache import PrefixMemoryManager
+    assert not isinstance(llm.runner.memory_manager, PrefixMemoryManager)
+    # KV allocated only for the full-attention layers
+    assert len(llm.runner.k_caches) == 2
+
+    # chunked prefill must carry conv + recurrent state across chunks
+    llm2 = _mk_llm(tmp_path, maxp=8, name="chunked")
+    out = llm2.generate(prompt, sp)[0].token_ids
+    assert out == ref
+
+    # slots released
+    assert llm.runner.ssm_pool.alloc.num_used == 0
+
+
+def test_hybrid_batched_equals_single(tmp_path):
+    from gllm_amd.sequence import SamplingParams
+    llm = _mk_llm(tmp_path, name="b")
+    sp = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
+    p1, p2 = [1, 2, 3, 4, 5, 6, 7], [9, 10, 11]
+    batched = llm.generate([p1, p2], [sp, sp])
+    s1 = llm.generate([p1], [sp])[0].token_ids
+    s2 = llm.generate([p2], [sp])[0].token_ids
+    assert batched[0].token_ids == s1
+    assert batched[1].token_ids == s2
