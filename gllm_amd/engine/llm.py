@@ -61,10 +61,55 @@ class LLM:
 
     # ------------------------------------------------------------------
     def allocate_seq(self, prompt_token_ids: List[int],
-                     sampling: Optional[SamplingParams] = None) -> Sequence:
-        return Sequence(self.seq_id_alloc.allocate(), prompt_token_ids,
-                        sampling=sampling, eos_token_id=self.eos_token_id,
-                        arrival_time=time.time())
+                     sampling: Optional[SamplingParams] = None,
+                     mm_input: Optional[dict] = None) -> Sequence:
+        seq = Sequence(self.seq_id_alloc.allocate(), prompt_token_ids,
+                       sampling=sampling, eos_token_id=self.eos_token_id,
+                       arrival_time=time.time())
+        if mm_input is not None:
+            self._prepare_mm(seq, mm_input)
+        return seq
+
+    def _prepare_mm(self, seq: Sequence, mm_input: dict) -> None:
+        """Offline multimodal admission: run the vision tower, record
+        image spans and MRoPE positions.
+
+        mm_input: {"pixel_values": [L, C*tps*ps*ps], "grids":
+        [(t,h,w), ...]} or {"embeds": [N, hidden], "grids": [...]}.
+        """
+        model = self.runner.model
+        grids = mm_input["grids"]
+        if "embeds" in mm_input:
+            embeds = mm_input["embeds"]
+        else:
+            import torch as _t
+            with _t.no_grad():
+                embeds = model.encode_images(mm_input["pixel_values"],
+                                             grids).cpu()
+        seq.mm_embeds = embeds
+        img_tok = model.image_token_id
+        # spans = maximal runs of the image token
+        spans = []
+        i = 0
+        toks = seq.token_ids
+        while i < len(toks):
+            if toks[i] == img_tok:
+                j = i
+                while j < len(toks) and toks[j] == img_tok:
+                    j += 1
+                spans.append((i, j - i))
+                i = j
+            else:
+                i += 1
+        assert sum(n for _, n in spans) == embeds.shape[0], \
+            (spans, embeds.shape)
+        seq.mm_spans = spans
+        from gllm_amd.layers.mrope import MRotaryEmbedding
+        pos, delta = MRotaryEmbedding.get_input_positions(
+            toks, img_tok, grids,
+            spatial_merge_size=model.spatial_merge_size)
+        seq.mrope_positions = pos
+        seq.mrope_delta = delta
 
     def add_requests(self, seqs: List[Sequence]) -> None:
         self.scheduler.add_seqs(seqs)
@@ -91,22 +136,25 @@ class LLM:
     def generate(self,
                  prompts: Optional[Seq[Union[str, List[int]]]] = None,
                  sampling_params: Optional[Union[SamplingParams,
-                                                 List[SamplingParams]]] = None
+                                                 List[SamplingParams]]] = None,
+                 mm_inputs: Optional[List[Optional[dict]]] = None
                  ) -> List[RequestOutput]:
         assert prompts is not None
         if isinstance(sampling_params, SamplingParams) or \
                 sampling_params is None:
             sampling_params = [sampling_params or SamplingParams()
                                ] * len(prompts)
+        if mm_inputs is None:
+            mm_inputs = [None] * len(prompts)
         seqs = []
-        for p, sp in zip(prompts, sampling_params):
+        for p, sp, mm in zip(prompts, sampling_params, mm_inputs):
             if isinstance(p, str):
                 assert self.tokenizer is not None, \
                     "string prompts need a tokenizer"
                 ids = self.tokenizer.encode(p)
             else:
                 ids = list(p)
-            seqs.append(self.allocate_seq(ids, sp))
+            seqs.append(self.allocate_seq(ids, sp, mm_input=mm))
         self.add_requests(seqs)
         pending = {s.seq_id for s in seqs}
         while pending:

@@ -142,8 +142,11 @@ _ROPE_CACHE: Dict[Tuple, RotaryEmbedding] = {}
 def get_rope(head_dim: int, rot_dim: int, max_position: int, base: float,
              is_neox: bool = True, rope_scaling: Optional[dict] = None
              ) -> RotaryEmbedding:
+    def _freeze(v):
+        return tuple(v) if isinstance(v, list) else v
     key = (head_dim, rot_dim, max_position, base, is_neox,
-           tuple(sorted(rope_scaling.items())) if rope_scaling else None)
+           tuple(sorted((k, _freeze(v)) for k, v in rope_scaling.items()))
+           if rope_scaling else None)
     if key in _ROPE_CACHE:
         return _ROPE_CACHE[key]
     if rope_scaling is None:
@@ -175,6 +178,11 @@ def get_rope(head_dim: int, rot_dim: int, max_position: int, base: float,
                 attn_factor=rope_scaling.get("attention_factor", 1.0) or 1.0,
                 mscale=rope_scaling.get("mscale"),
                 mscale_all_dim=rope_scaling.get("mscale_all_dim"))
+        elif rtype in ("mrope", "default") and \
+                rope_scaling.get("mrope_section"):
+            from gllm_amd.layers.mrope import MRotaryEmbedding
+            rope = MRotaryEmbedding(head_dim, rot_dim, max_position, base,
+                                    list(rope_scaling["mrope_section"]))
         elif rtype == "default":
             rope = RotaryEmbedding(head_dim, rot_dim, max_position, base,
                                    is_neox)

@@ -167,6 +167,12 @@ class LlamaFamilyForCausalLM(nn.Module):
                 residual: Optional[torch.Tensor] = None):
         if self.is_first_stage:
             hidden_states = self.embed_tokens(input_ids)
+            if fctx.mm_rows is not None:
+                # vision-embedding merge (reference embed_input_ids /
+                # disagg_set_embedding, model_runner.py:931-994)
+                hidden_states = hidden_states.index_copy(
+                    0, fctx.mm_rows,
+                    fctx.mm_embeds.to(hidden_states.dtype))
             residual = None
         for layer in self.layers:
             hidden_states, residual = layer(positions, hidden_states,

@@ -27,7 +27,10 @@ def get_arch_registry():
     from gllm_amd.models.deepseek_v2 import (DeepseekV2ForCausalLM,
                                              DeepseekV3ForCausalLM)
     from gllm_amd.models.hybrid_gdn import Qwen3_5ForCausalLM
+    from gllm_amd.models.qwen2_vl import Qwen2VLForCausalLM
     return {
+        "Qwen2VLForConditionalGeneration": Qwen2VLForCausalLM,
+        "Qwen2_5_VLForConditionalGeneration": Qwen2VLForCausalLM,
         "Qwen3_5ForCausalLM": Qwen3_5ForCausalLM,
         "Qwen3NextForCausalLM": Qwen3_5ForCausalLM,
         "DeepseekV2ForCausalLM": DeepseekV2ForCausalLM,

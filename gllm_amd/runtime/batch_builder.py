@@ -21,7 +21,9 @@ class BatchBuilder:
         self.pin = pin and device != "cpu"
 
     def build(self, batch: ScheduledBatch, k_caches, v_caches,
-              need_logits: bool = True) -> Tuple[torch.Tensor, ForwardContext]:
+              need_logits: bool = True,
+              use_mrope: bool = False) -> Tuple[torch.Tensor,
+                                                ForwardContext]:
         items = batch.items
         ps = self.page_size
         B = len(items)
@@ -32,7 +34,7 @@ class BatchBuilder:
         np.cumsum(lens, out=qsl[1:])
 
         tokens = np.empty(T, dtype=np.int64)
-        positions = np.empty(T, dtype=np.int64)
+        positions = np.empty((3, T) if use_mrope else T, dtype=np.int64)
         slots = np.empty(T, dtype=np.int64)
         seq_lens = np.empty(B, dtype=np.int32)
         max_pages = max(-(-(it.start + it.num_tokens) // ps) for it in items)
@@ -42,7 +44,20 @@ class BatchBuilder:
             s, n = it.start, it.num_tokens
             o = qsl[i]
             tokens[o:o + n] = it.seq.token_ids[s:s + n]
-            positions[o:o + n] = np.arange(s, s + n)
+            if use_mrope:
+                seq = it.seq
+                mp = seq.mrope_positions
+                for j in range(n):
+                    pos = s + j
+                    if mp is not None and pos < seq.prompt_len:
+                        positions[:, o + j] = mp[:, pos].numpy()
+                    elif mp is not None:
+                        positions[:, o + j] = seq.mrope_delta + \
+                            (pos - seq.prompt_len)
+                    else:
+                        positions[:, o + j] = pos
+            else:
+                positions[o:o + n] = np.arange(s, s + n)
             pt = np.asarray(it.seq.page_table, dtype=np.int64)
             pos = np.arange(s, s + n)
             slots[o:o + n] = pt[pos // ps] * ps + pos % ps

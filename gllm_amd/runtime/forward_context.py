@@ -36,6 +36,10 @@ class ForwardContext:
     ssm_pool: Optional[object] = None
     ssm_slots: Optional[List[int]] = None
     ssm_has_init: Optional[List[bool]] = None
+    # multimodal: rows of this batch whose embeddings come from the
+    # vision tower (replaced after embed_tokens on the first stage)
+    mm_rows: Optional[torch.Tensor] = None     # [N] long
+    mm_embeds: Optional[torch.Tensor] = None   # [N, hidden]
 
     @property
     def is_pure_decode(self) -> bool:

@@ -49,6 +49,11 @@ class ModelRunner:
                             "(state snapshots land in round 2)")
                 cfg.enable_prefix_caching = False
             cfg.use_graph = False
+        self.uses_mrope = bool(getattr(self.model, "uses_mrope", False))
+        if self.uses_mrope:
+            # [3,B] graph position buffers + mm content hashing: round 2
+            cfg.use_graph = False
+            cfg.enable_prefix_caching = False
         num_pages = num_pages_override or self._size_kv_cache()
         self._allocate_kv(num_pages)
         self.num_kv_pages_total = num_pages
@@ -193,8 +198,10 @@ class ModelRunner:
             return self.graph_runner.replay(batch)
         tokens, fctx = self.builder.build(
             batch, self.k_caches, self.v_caches,
-            need_logits=self.model.is_last_stage)
+            need_logits=self.model.is_last_stage,
+            use_mrope=self.uses_mrope)
         self._attach_ssm(batch, fctx)
+        self._attach_mm(batch, fctx)
         if fctx.has_placeholders:
             tokens = self.resolve_tokens(tokens)
         hidden, residual = self._stage_forward(tokens, fctx.positions, fctx)
@@ -207,7 +214,8 @@ class ModelRunner:
         """PP stage > 0: forward received hidden states."""
         _, fctx = self.builder.build(
             batch, self.k_caches, self.v_caches,
-            need_logits=self.model.is_last_stage)
+            need_logits=self.model.is_last_stage,
+            use_mrope=self.uses_mrope)
         self._attach_ssm(batch, fctx)
         hidden, residual = self._stage_forward(
             None, fctx.positions, fctx, hidden_states=hidden,
@@ -223,6 +231,31 @@ class ModelRunner:
         fctx.ssm_slots = [self.ssm_pool.ensure(it.seq)
                           for it in batch.items]
         fctx.ssm_has_init = [it.start > 0 for it in batch.items]
+
+    def _attach_mm(self, batch, fctx) -> None:
+        """Collect vision-embedding rows for this batch's prefill chunks
+        (chunk-aware: an image span may straddle chunk boundaries)."""
+        rows, embeds = [], []
+        qsl_off = 0
+        for it in batch.items:
+            seq = it.seq
+            if seq.mm_embeds is not None and it.start < seq.prompt_len:
+                emb_off = 0
+                for (span_s, span_n) in seq.mm_spans:
+                    lo = max(span_s, it.start)
+                    hi = min(span_s + span_n, it.start + it.num_tokens)
+                    if lo < hi:
+                        rows.extend(range(qsl_off + lo - it.start,
+                                          qsl_off + hi - it.start))
+                        embeds.append(
+                            seq.mm_embeds[emb_off + lo - span_s:
+                                          emb_off + hi - span_s])
+                    emb_off += span_n
+            qsl_off += it.num_tokens
+        if rows:
+            fctx.mm_rows = torch.tensor(rows, dtype=torch.long,
+                                        device=self.device)
+            fctx.mm_embeds = torch.cat(embeds).to(self.device)
 
     def resolve_tokens(self, tokens: torch.Tensor) -> torch.Tensor:
         """Replace negative placeholder ids with sampled tokens from the

@@ -76,6 +76,15 @@ class Sequence:
         # list of (chosen_logprob, {token_id: logprob} top-k)
         self.out_logprobs: List[tuple] = []
 
+        # --- multimodal (set by the engine at admission) ---
+        # concatenated vision embeddings for all image spans [N, hidden]
+        self.mm_embeds = None
+        # (token_start, length) spans of image placeholder runs
+        self.mm_spans: List[tuple] = []
+        # MRoPE position table [3, prompt_len] + decode delta
+        self.mrope_positions = None
+        self.mrope_delta: int = 0
+
     # ---- basic accounting ----
     def __len__(self) -> int:
         return len(self.token_ids)
