@@ -110,18 +110,29 @@ class Scheduler:
             self.seqs_to_prefill.append(seq)
 
     def abort_seqs(self, seq_ids) -> None:
-        self.abort_ids.update(seq_ids)
+        ids = set(seq_ids)
         # drop queued (not in-flight) seqs immediately
         for q in (self.seqs_to_prefill, self.seqs_to_decode):
-            keep = [s for s in q if s.seq_id not in self.abort_ids or
-                    self._in_flight(s)]
-            dropped = [s for s in q if s not in keep]
+            keep, dropped = [], []
+            for s in q:
+                if s.seq_id in ids and not self._in_flight(s):
+                    dropped.append(s)
+                else:
+                    keep.append(s)
             q.clear()
             q.extend(keep)
             for s in dropped:
                 s.finish_reason = Sequence.FINISH_ABORT
                 self.mm.free_seq(s)
-                self.abort_ids.discard(s.seq_id)
+                ids.discard(s.seq_id)
+        # Only ids this scheduler actually has in flight go to the
+        # deferred set — an unknown id already finished here or belongs
+        # to another DP replica; registering it would abort a FUTURE
+        # request if the frontend recycles the seq id.
+        known = set(self._scheduled_cursor) | \
+            {s.seq_id for s in self.seqs_to_prefill} | \
+            {s.seq_id for s in self.seqs_to_decode}
+        self.abort_ids.update(ids & known)
 
     def _in_flight(self, seq: Sequence) -> bool:
         return seq.seq_id in self._scheduled_cursor and \

@@ -43,6 +43,13 @@ class PPEngine:
         self.scheduler = Scheduler(config, self.runner.memory_manager)
         self.pp_rank = P.get_pp_rank()
         self.pp_size = P.get_pp_size()
+        self.dp_rank = P.get_dp_rank()
+        self.dp_size = P.get_dp_size()
+        # round-1 bound: DP attention (replica parallelism) is PP=1;
+        # multi-stage DP lockstep is in ROADMAP.md
+        assert self.dp_size == 1 or self.pp_size == 1, \
+            "dp_size > 1 requires pp_size == 1"
+        self._dp_global_work = 1
         self.is_first = P.is_first_pp_rank()
         self.is_last = P.is_last_pp_rank()
         self.device = config.device
@@ -121,9 +128,52 @@ class PPEngine:
             tokens = tok.tolist()
         return self.scheduler.process_output(batch, tokens)
 
+    # ---------------------------------------------------- DP attention
+    def dp_forward(self, batch: Optional[ScheduledBatch]) -> bool:
+        """One lockstep DP-attention round (reference worker.py:750-889
+        re-shaped for replicated schedulers): every replica enters a
+        metadata barrier with (tokens scheduled this round, has pending
+        work); if any replica scheduled tokens, ALL replicas forward —
+        an idle one runs a 1-token dummy so the MoE DP-gather and EP
+        all-reduce collectives stay matched. Returns True if a forward
+        ran this round."""
+        P = self.P
+        nt = batch.num_tokens if batch is not None else 0
+        counts, flags = P.dp_meta_barrier(nt, self.scheduler.has_work())
+        self._dp_global_work = sum(flags) + sum(counts)
+        if max(counts) == 0:
+            return False
+        P.set_dp_forward_counts(counts)
+        try:
+            if batch is None:
+                self.runner.step_dummy()
+            else:
+                out = self.runner.step_first_stage(batch)
+                self.inflight.append((batch, out))
+        finally:
+            P.set_dp_forward_counts(None)
+        return True
+
+    def _run_until_done_dp(self, max_steps: Optional[int] = None):
+        done, steps = [], 0
+        while True:
+            b = self.scheduler.schedule_once() \
+                if self.scheduler.has_work() else None
+            ran = self.dp_forward(b)
+            if self.inflight:
+                done.extend(self._complete_oldest())
+            if not ran and self._dp_global_work == 0:
+                break
+            steps += 1
+            if max_steps is not None and steps >= max_steps:
+                break
+        return done
+
     # ------------------------------------------------------------------
     def run_until_done(self, max_steps: Optional[int] = None):
         """Drive the loop until all requests finish. Returns finished seqs."""
+        if self.dp_size > 1:
+            return self._run_until_done_dp(max_steps)
         done = []
         steps = 0
         while self.scheduler.has_work():

@@ -39,10 +39,19 @@ class ServingMixin:
 
     def _init_serving(self, config: EngineConfig, req_queue, out_queue):
         self._last_stats = 0.0
-        from gllm_amd.parallel import get_rank, get_world_size
+        from gllm_amd.parallel import (get_dp_rank, get_dp_size, get_rank,
+                                       get_tp_rank, get_world_size)
         self.rank = get_rank()
         self.world = get_world_size()
-        self.is_output_rank = self.rank == 0
+        self.dp_rank = get_dp_rank()
+        self.dp_size = get_dp_size()
+        # DP: every replica emits its own outputs (tp rank 0 of the
+        # replica; DP requires pp == 1). Non-DP: global rank 0.
+        if self.dp_size > 1:
+            self.is_output_rank = get_tp_rank() == 0
+        else:
+            self.is_output_rank = self.rank == 0
+        self._req_counter = 0  # deterministic DP round-robin routing
         self.comm = WorkerComm(req_queue, out_queue, self.is_output_rank)
         self._intake_buf = torch.zeros(1, dtype=torch.int64)
         if self.world > 1 and config.device.startswith("cuda"):
@@ -54,6 +63,14 @@ class ServingMixin:
     def _apply_messages(self, msgs: List[tuple]) -> None:
         for kind, idx, payload in msgs:
             if kind == "req":
+                # every rank sees every request in the same order;
+                # DP replicas take theirs round-robin (reference
+                # llm_engine.py:490-519 routes at the frontend — here
+                # routing is a deterministic function of the stream)
+                target = self._req_counter % self.dp_size
+                self._req_counter += 1
+                if target != self.dp_rank:
+                    continue
                 seq = Sequence(payload["seq_id"], payload["token_ids"],
                                SamplingParams(**payload["sampling"]),
                                eos_token_id=payload.get("eos_token_id"))
@@ -110,7 +127,8 @@ class ServingMixin:
 
     # ------------------------------------------------------------------
     def _maybe_send_stats(self) -> None:
-        if not self.is_output_rank:
+        # one stats stream (rank 0 reports its own replica under DP)
+        if not self.is_output_rank or self.rank != 0:
             return
         now = time.time()
         if now - self._last_stats < 1.0:
@@ -158,7 +176,32 @@ class ServingWorker(ServingMixin, PPEngine):
                 self.comm.send_output(("out", outs, {}))
         return finished
 
+    def _run_loop_dp(self) -> None:
+        """DP-attention serving loop: lockstep rounds over the replica
+        grid (idle replicas run dummies so MoE collectives match)."""
+        logger.info("worker %d ready (dp=%d of %d, lockstep)",
+                    self.rank, self.dp_rank, self.dp_size)
+        while not self.shutdown:
+            # rank 0 blocks on intake only when the LAST barrier said the
+            # whole grid is idle (its own replica included)
+            block = (not self.scheduler.has_work() and not self.inflight
+                     and self._dp_global_work == 0)
+            self._sync_intake(block=block)
+            if self.shutdown:
+                break
+            b = self.scheduler.schedule_once() \
+                if self.scheduler.has_work() else None
+            self.dp_forward(b)
+            if self.inflight:
+                self._complete_oldest()
+            self._maybe_send_stats()
+        self.drain()
+        self.comm.close()
+        logger.info("worker %d shut down", self.rank)
+
     def run_loop(self) -> None:
+        if self.dp_size > 1:
+            return self._run_loop_dp()
         logger.info("worker %d ready (pp=%d, sync)", self.rank, self.pp_size)
         while not self.shutdown:
             has_work = self.scheduler.has_work()

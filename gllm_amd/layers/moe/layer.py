@@ -96,8 +96,22 @@ class FusedMoE(nn.Module):
                        ids: torch.Tensor) -> torch.Tensor:
         """Expert compute with externally computed routing (DeepSeek
         grouped-topk / routed scaling paths)."""
+        from gllm_amd.parallel import get_dp_size
+        if self.use_ep and get_dp_size() > 1:
+            return self._forward_dp(x, weights, ids)
+        out = self._expert_loop(x, weights.to(x.dtype), ids)
+        if self.use_ep:
+            from gllm_amd.parallel.state import ep_all_reduce
+            out = ep_all_reduce(out)
+        else:
+            out = tensor_parallel_all_reduce(out)
+        return out
+
+    def _expert_loop(self, x: torch.Tensor, weights: torch.Tensor,
+                     ids: torch.Tensor) -> torch.Tensor:
+        """Local expert shard over the (possibly DP-gathered) batch;
+        returns the PARTIAL output (no collectives)."""
         T = x.shape[0]
-        weights = weights.to(x.dtype)
         out = torch.zeros_like(x)
         flat_ids = ids.long().flatten()                    # [T*K]
         flat_rows = torch.arange(T, device=x.device).repeat_interleave(
@@ -113,9 +127,39 @@ class FusedMoE(nn.Module):
             ye = ops.linear(h, self.w2_weight[lid])
             w = weights.flatten()[sel].unsqueeze(-1)
             out.index_add_(0, rows, ye * w)
-        if self.use_ep:
-            from gllm_amd.parallel.state import ep_all_reduce
-            out = ep_all_reduce(out)
-        else:
-            out = tensor_parallel_all_reduce(out)
         return out
+
+    def _forward_dp(self, x: torch.Tensor, weights: torch.Tensor,
+                    ids: torch.Tensor) -> torch.Tensor:
+        """DP-attention MoE: experts span replicas (EP = DP x TP), so the
+        routed batch is the union over replicas (reference
+        models/utils.py:39-96 dp_ep_moe_routed). Every replica pads its
+        rows to the round's max count (published by the engine's
+        dp_meta_barrier), all-gathers tokens + routing over the DP
+        group, computes its local expert shard on the GLOBAL batch, and
+        one EP all-reduce over the stage yields the full output
+        everywhere; each replica slices back its own rows. Padding rows
+        carry expert id -1 (never matched) so they contribute zeros."""
+        from gllm_amd.parallel import (dp_all_gather, get_dp_forward_counts,
+                                       get_dp_rank)
+        from gllm_amd.parallel.state import ep_all_reduce
+        counts = get_dp_forward_counts()
+        assert counts is not None, \
+            "DP MoE forward entered without dp_meta_barrier counts"
+        T = x.shape[0]
+        maxc = max(counts + [T])
+
+        def pad(t, fill):
+            if T == maxc:
+                return t.contiguous()
+            tail = torch.full((maxc - T,) + tuple(t.shape[1:]), fill,
+                              dtype=t.dtype, device=t.device)
+            return torch.cat([t, tail], dim=0)
+
+        gx = dp_all_gather(pad(x, 0))
+        gw = dp_all_gather(pad(weights.to(x.dtype), 0))
+        gids = dp_all_gather(pad(ids.long(), -1))
+        out = self._expert_loop(gx, gw, gids)
+        out = ep_all_reduce(out)
+        start = get_dp_rank() * maxc
+        return out[start:start + T]

@@ -6,4 +6,6 @@ from gllm_amd.parallel.state import (  # noqa: F401
     is_first_pp_rank, is_last_pp_rank,
     tensor_parallel_all_reduce, tensor_parallel_all_gather,
     send_pp_data, recv_pp_data, get_prev_pp_rank, get_next_pp_rank,
+    dp_meta_barrier, dp_all_gather, set_dp_forward_counts,
+    get_dp_forward_counts,
 )

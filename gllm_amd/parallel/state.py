@@ -185,6 +185,53 @@ def ep_all_reduce(t: torch.Tensor) -> torch.Tensor:
     return t
 
 
+# ------------------------------------------------------- DP attention
+# Replica parallelism (reference dist_utils.py:266-352 + worker.py
+# _schedule_forward_dp): each DP replica owns its own scheduler + KV;
+# MoE layers span replicas via EP = DP x TP, so every forward round is
+# entered in lockstep. The per-round metadata barrier gathers
+# (num_tokens_this_round, has_pending_work) from each replica; idle
+# replicas run a 1-token dummy forward so the MoE collectives match.
+# The per-forward token counts are published module-globally for the
+# MoE layer's padded gather (set around each forward by the engine).
+_DP_FWD_COUNTS: Optional[List[int]] = None
+
+
+def dp_meta_barrier(num_tokens: int, has_work: bool):
+    """All-gather (num_tokens, has_work) across the DP group. Returns
+    (counts, flags) lists indexed by dp rank."""
+    if _DP_SIZE == 1:
+        return [num_tokens], [1 if has_work else 0]
+    dev = _DEVICE if str(_DEVICE).startswith("cuda") else "cpu"
+    local = torch.tensor([num_tokens, 1 if has_work else 0],
+                         dtype=torch.int64, device=dev)
+    out = torch.empty(2 * _DP_SIZE, dtype=torch.int64, device=dev)
+    dist.all_gather_into_tensor(out, local, group=_DP_GROUP)
+    pairs = out.view(_DP_SIZE, 2).tolist()
+    return [p[0] for p in pairs], [p[1] for p in pairs]
+
+
+def dp_all_gather(t: torch.Tensor) -> torch.Tensor:
+    """All-gather rows across the DP group (uniform row count on every
+    replica — callers pad to max first): [R, ...] -> [dp*R, ...]."""
+    if _DP_SIZE == 1:
+        return t
+    t = t.contiguous()
+    out = torch.empty((_DP_SIZE * t.shape[0],) + tuple(t.shape[1:]),
+                      dtype=t.dtype, device=t.device)
+    dist.all_gather_into_tensor(out, t, group=_DP_GROUP)
+    return out
+
+
+def set_dp_forward_counts(counts: Optional[List[int]]) -> None:
+    global _DP_FWD_COUNTS
+    _DP_FWD_COUNTS = counts
+
+
+def get_dp_forward_counts() -> Optional[List[int]]:
+    return _DP_FWD_COUNTS
+
+
 # ---------------------------------------------------------------- PP p2p
 def send_pp_data(tensors: List[torch.Tensor], dst: int) -> None:
     """Send hidden_states (+residual) to the next stage. One xGMI link

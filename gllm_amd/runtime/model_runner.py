@@ -170,6 +170,27 @@ class ModelRunner:
                 self.model.compute_logits(hidden, fctx)
         torch.cuda.synchronize()
 
+    @torch.no_grad()
+    def step_dummy(self) -> None:
+        """1-token lockstep dummy forward for DP attention (reference
+        worker.py:750-889): an idle replica must still enter every MoE
+        collective of the round. ``is_profile_run`` makes attention a
+        no-op (no KV touched); MoE layers run and join the DP gather."""
+        T = 1
+        dev = self.device
+        fctx = ForwardContext(
+            num_tokens=T,
+            positions=torch.zeros(T, dtype=torch.long, device=dev),
+            slot_mapping=torch.zeros(T, dtype=torch.long, device=dev),
+            block_table=torch.zeros((1, 1), dtype=torch.int32, device=dev),
+            seq_lens=torch.ones(1, dtype=torch.int32, device=dev),
+            query_start_loc=torch.tensor([0, T], dtype=torch.int32,
+                                         device=dev),
+            max_query_len=T, max_seq_len=T, k_caches=[], v_caches=[],
+            is_profile_run=True)
+        input_ids = torch.zeros(T, dtype=torch.long, device=dev)
+        self._stage_forward(input_ids, fctx.positions, fctx)
+
     def _allocate_kv(self, num_pages: int):
         spec = self.kv_spec()
         kshape = (num_pages, spec.page_size, spec.num_kv_heads,

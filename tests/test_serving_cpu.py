@@ -66,3 +66,34 @@ def test_serving_roundtrip(tmp_path, pp_size, port):
         assert [c.token_id for c in chunks2] == toks
     finally:
         eng.stop()
+
+
+@pytest.mark.timeout(300)
+def test_serving_dp2_replicas(tmp_path):
+    """DP-attention serving: two replicas take the request stream
+    round-robin in lockstep rounds; a dense dummy-init model is
+    replica-identical, so both replicas must emit the same tokens."""
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+    cfg = EngineConfig(model=_model_dir(tmp_path), load_format="dummy",
+                       device="cpu", dtype="float32", page_size=4,
+                       dp_size=2, maxp=64, maxd=32, master_port=29658,
+                       enable_prefix_caching=False)
+    eng = AsyncLLMEngine(cfg, base_port=28760)
+    eng.start()
+    try:
+        sp = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
+
+        async def two():
+            a = asyncio.ensure_future(_agen(eng, [1, 2, 3, 4, 5], sp))
+            b = asyncio.ensure_future(_agen(eng, [1, 2, 3, 4, 5], sp))
+            return await asyncio.gather(a, b)
+
+        r1, r2 = asyncio.new_event_loop().run_until_complete(two())
+        assert len(r1) == 5 and r1 == r2
+    finally:
+        eng.stop()
+
+
+async def _agen(engine, token_ids, sampling):
+    return [c.token_id
+            async for c in engine.generate_stream(token_ids, sampling)]
