@@ -51,7 +51,10 @@ def _run_ep_rank(rank, tmp, port, q):
                       MASTER_PORT=str(port))
     torch.set_num_threads(1)
     from gllm_amd.parallel import init_distributed
-    init_distributed(rank=rank, pp_size=1, dp_size=2, tp_size=1,
+    # EP spans dp*tp; tp=2 gives 2 EP ranks with REPLICATED activations
+    # (dp=2 would mean DP attention: per-replica batches + MoE gather,
+    # covered by tests/test_dp_cpu.py)
+    init_distributed(rank=rank, pp_size=1, dp_size=1, tp_size=2,
                      master_port=port)
     torch.manual_seed(0)
     E, K, H, I, T = 8, 2, 32, 64, 6
