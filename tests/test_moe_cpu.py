@@ -75,7 +75,9 @@ def _run_ep_rank(rank, tmp, port, q):
     if rank == 0:
         w13 = torch.cat([full_w1, full_w3], dim=1)
         ref = _ref_moe(x, w13, full_w2, logits, K)
-        q.put((out, ref))
+        # plain lists: tensors over mp.Queue ride /dev/shm files that
+        # vanish when the child exits before the parent reads them
+        q.put((out.tolist(), ref.tolist()))
     import torch.distributed as dist
     dist.barrier()
     dist.destroy_process_group()
@@ -94,6 +96,7 @@ def test_ep_two_ranks_matches_dense(tmp_path):
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
+    out, ref = torch.tensor(out), torch.tensor(ref)
     assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
 
 
