@@ -31,9 +31,16 @@ class EngineConfig:
     master_addr: str = "127.0.0.1"
     master_port: int = 29500
     # multi-node launch: "normal" (single node) | "master" | "slave"
+    # (engine/multinode.py — control plane over TCP relay_port, data
+    # plane over the torch.distributed rendezvous at master_port)
     launch_mode: str = "normal"
     nnodes: int = 1
     node_rank: int = 0
+    # global ranks hosted by THIS node (master/slave modes); None in
+    # normal mode = all ranks
+    worker_ranks: Optional[List[int]] = None
+    # control-plane TCP port on the master node; None = master_port + 1
+    relay_port: Optional[int] = None
 
     # --- KV cache ---
     page_size: int = 16

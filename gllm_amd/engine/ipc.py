@@ -24,10 +24,17 @@ import torch.multiprocessing as mp
 
 
 class FrontendComm:
-    def __init__(self, world_size: int):
+    def __init__(self, world_size: int,
+                 local_ranks: Optional[List[int]] = None):
         ctx = mp.get_context("spawn")
-        self.req_queues = [ctx.Queue() for _ in range(world_size)]
+        # master launch mode: only this node's ranks get local queues;
+        # the remote_sender hook (engine/multinode.py MasterRelay) ships
+        # the same ordered stream to slave nodes over TCP
+        self.local_ranks = list(local_ranks) if local_ranks is not None \
+            else list(range(world_size))
+        self.req_queues = {r: ctx.Queue() for r in self.local_ranks}
         self.out_queue = ctx.Queue()
+        self.remote_sender = None
         self._msg_idx = 0
 
     def worker_endpoints(self, rank: int):
@@ -38,8 +45,10 @@ class FrontendComm:
         idx = self._msg_idx
         self._msg_idx += 1
         msg = (kind, idx, payload)
-        for q in self.req_queues:
+        for q in self.req_queues.values():
             q.put(msg)
+        if self.remote_sender is not None:
+            self.remote_sender(msg)
         return idx
 
     def recv_output(self, timeout_ms: Optional[int] = None):

@@ -58,7 +58,10 @@ class AsyncLLMEngine:
         if self.tokenizer is not None and self.tokenizer.eos_token_id is not None:
             eos = self.tokenizer.eos_token_id
         self.eos_token_id = eos[0] if isinstance(eos, list) else eos
-        self.comm = FrontendComm(config.world_size)
+        local_ranks = config.worker_ranks \
+            if config.launch_mode == "master" else None
+        self.comm = FrontendComm(config.world_size, local_ranks=local_ranks)
+        self._relay = None
         self.seq_ids = IDAllocator(1 << 20)
         self.requests: Dict[int, RequestState] = {}
         self._intake_lock = threading.Lock()
@@ -81,14 +84,29 @@ class AsyncLLMEngine:
         ctx = mp.get_context("spawn")
         ready_q = ctx.Queue()
         from gllm_amd.engine.worker import run_worker
-        for r in range(self.config.world_size):
+        cfg = self.config
+        local_ranks = self.comm.local_ranks
+        if cfg.launch_mode == "master":
+            # control-plane relay for slave nodes (data plane = RCCL
+            # rendezvous at master_addr:master_port, cross-node already)
+            from gllm_amd.engine.multinode import MasterRelay
+            remote = sorted(set(range(cfg.world_size)) - set(local_ranks))
+            self._relay = MasterRelay("0.0.0.0",
+                                      cfg.relay_port or cfg.master_port + 1,
+                                      remote, self.comm.out_queue)
+            self.comm.remote_sender = self._relay.broadcast
+        for r in local_ranks:
             rq, oq = self.comm.worker_endpoints(r)
             p = ctx.Process(target=run_worker,
                             args=(r, self.config, rq, oq, ready_q),
                             daemon=True)
             p.start()
             self._procs.append(p)
-        for _ in range(self.config.world_size):
+        if self._relay is not None:
+            # slaves must be connected before workers can finish their
+            # torch.distributed init; accept them while workers load
+            self._relay.wait_for_slaves()
+        for _ in local_ranks:
             status, rank = ready_q.get()  # blocks until model loaded
             if status != "ready":
                 raise RuntimeError(f"worker {rank} failed to start")
@@ -126,6 +144,8 @@ class AsyncLLMEngine:
             pass
         for p in self._procs:
             p.join(timeout=30)
+        if self._relay is not None:
+            self._relay.close()
 
     # ------------------------------------------------------------------
     def _output_loop(self) -> None:
@@ -137,6 +157,14 @@ class AsyncLLMEngine:
             if kind == "stats":
                 self.latest_stats = outs
                 continue
+            if kind == "worker_dead":
+                logger.error("remote worker %s died — failing %d pending "
+                             "requests", outs, len(self.requests))
+                for st in list(self.requests.values()):
+                    if not st.finished:
+                        self._handle_token(st, -1, "abort")
+                self._stopping = True
+                return
             if kind != "out":
                 continue
             self.token_counter += sum(1 for _, t, _f in outs if t >= 0)

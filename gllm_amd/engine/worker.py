@@ -254,11 +254,15 @@ class OverlapServingWorker(ServingMixin, OverlapEngine):
 def run_worker(rank: int, config: EngineConfig, req_queue, out_queue,
                ready_queue=None) -> None:
     os.environ["RANK"] = str(rank)
-    os.environ.setdefault("LOCAL_RANK", str(rank))
+    # master/slave launch: this node hosts config.worker_ranks; the
+    # device index is the rank's position within them
+    local = rank - min(config.worker_ranks) if config.worker_ranks \
+        else rank
+    os.environ.setdefault("LOCAL_RANK", str(local))
     os.environ["MASTER_ADDR"] = config.master_addr
     os.environ["MASTER_PORT"] = str(config.master_port)
     if config.device.startswith("cuda"):
-        config.device = f"cuda:{rank % max(1, torch.cuda.device_count())}"
+        config.device = f"cuda:{local % max(1, torch.cuda.device_count())}"
     try:
         use_overlap = (config.world_size == 1 and config.enable_overlap
                        and config.device.startswith("cuda"))

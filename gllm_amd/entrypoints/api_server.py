@@ -266,7 +266,18 @@ def make_arg_parser():
     p.add_argument("--enforce-eager", action="store_true")
     p.add_argument("--max-graph-bs", type=int, default=256)
     p.add_argument("--assigned-layers", type=str, default=None)
+    p.add_argument("--master-addr", type=str, default="127.0.0.1")
     p.add_argument("--master-port", type=int, default=29500)
+    # multi-node: master hosts the frontend + --worker-ranks; each slave
+    # node hosts its own --worker-ranks and bridges to the master's
+    # control-plane relay (engine/multinode.py)
+    p.add_argument("--launch-mode", choices=["normal", "master", "slave"],
+                   default="normal")
+    p.add_argument("--worker-ranks", type=str, default=None,
+                   help="comma-separated global ranks on this node, "
+                        "e.g. '0,1' (master/slave modes)")
+    p.add_argument("--relay-port", type=int, default=None,
+                   help="control-plane TCP port (default master_port+1)")
     p.add_argument("--seed", type=int, default=0)
     return p
 
@@ -282,7 +293,11 @@ def config_from_args(args) -> EngineConfig:
         enable_prefix_caching=not args.disable_prefix_caching,
         enforce_eager=args.enforce_eager, max_graph_bs=args.max_graph_bs,
         assigned_layers=args.assigned_layers,
-        master_port=args.master_port, seed=args.seed,
+        master_addr=args.master_addr, master_port=args.master_port,
+        launch_mode=args.launch_mode,
+        worker_ranks=[int(r) for r in args.worker_ranks.split(",")]
+        if args.worker_ranks else None,
+        relay_port=args.relay_port, seed=args.seed,
         device="cuda" if _has_gpu() else "cpu")
 
 
@@ -296,6 +311,11 @@ def main():
     args = make_arg_parser().parse_args()
     served_model = args.served_model_name or args.model
     config = config_from_args(args)
+    if config.launch_mode == "slave":
+        # no HTTP frontend on slave nodes: run workers + control bridge
+        from gllm_amd.engine.multinode import run_slave_node
+        run_slave_node(config)
+        return
     engine = AsyncLLMEngine(config, base_port=args.zmq_port)
     engine.start()
     app = build_app()

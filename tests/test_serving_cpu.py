@@ -97,3 +97,49 @@ def test_serving_dp2_replicas(tmp_path):
 async def _agen(engine, token_ids, sampling):
     return [c.token_id
             async for c in engine.generate_stream(token_ids, sampling)]
+
+
+def _slave_main(model_dir):
+    import torch
+    torch.set_num_threads(1)
+    from gllm_amd.engine.multinode import run_slave_node
+    run_slave_node(_mn_cfg(model_dir, "slave", [1]))
+
+
+def _mn_cfg(model_dir, mode, ranks):
+    return EngineConfig(model=model_dir, load_format="dummy", device="cpu",
+                        dtype="float32", page_size=4, pp_size=2,
+                        maxp=64, maxd=32, master_port=29675,
+                        launch_mode=mode, worker_ranks=ranks,
+                        relay_port=28795, enable_prefix_caching=False)
+
+
+@pytest.mark.timeout(300)
+def test_multinode_master_slave(tmp_path):
+    """Master node hosts PP rank 0 + the frontend; a 'slave node'
+    (separate process tree on localhost) hosts PP rank 1, bridged over
+    the TCP control-plane relay. Covers engine/multinode.py end to end:
+    hello/ready handshake, ordered request fan-out, token outputs
+    (output rank 0 lives on the master; intake lockstep crosses nodes
+    via the gloo broadcast)."""
+    import multiprocessing
+    d = _model_dir(tmp_path)
+    ctx = multiprocessing.get_context("spawn")
+    slave = ctx.Process(target=_slave_main, args=(d,))  # spawns workers
+    slave.start()
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+    eng = AsyncLLMEngine(_mn_cfg(d, "master", [0]))
+    try:
+        eng.start()
+        sp = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
+        chunks = _collect(eng, [1, 2, 3, 4, 5], sp)
+        toks = [c.token_id for c in chunks]
+        assert len(toks) == 5
+        chunks2 = _collect(eng, [1, 2, 3, 4, 5], sp)
+        assert [c.token_id for c in chunks2] == toks
+    finally:
+        eng.stop()
+        slave.join(timeout=60)
+        if slave.is_alive():
+            slave.terminate()
+    assert slave.exitcode == 0
