@@ -1,6 +1,7 @@
 """Hybrid Gated-DeltaNet + full-attention decoder (Qwen3.5 / Qwen3-Next
-family). Parity target: reference models/qwen3_5.py (+qwen3_5_moe.py for
-the MoE variant, round 2).
+family). Parity target: reference models/qwen3_5.py + qwen3_5_moe.py
+(the MoE variant is config-driven here: ``num_experts`` swaps the dense
+MLP for a routed MoE block on sparse layers).
 
 Layer interleave from ``layer_types`` (or ``full_attention_interval``):
 linear_attention layers run the GDN recurrence over per-seq SSM state
@@ -203,7 +204,7 @@ class HybridFullAttention(nn.Module):
 
 class HybridDecoderLayer(nn.Module):
     def __init__(self, cfg, layer_type: str, ssm_or_kv_idx: int,
-                 dtype=None):
+                 dtype=None, mlp: nn.Module = None):
         super().__init__()
         eps = getattr(cfg, "rms_norm_eps", 1e-6)
         self.layer_type = layer_type
@@ -213,8 +214,8 @@ class HybridDecoderLayer(nn.Module):
         else:
             self.self_attn = HybridFullAttention(cfg, ssm_or_kv_idx,
                                                  dtype=dtype)
-        self.mlp = DenseMLP(cfg.hidden_size, cfg.intermediate_size,
-                            dtype=dtype)
+        self.mlp = mlp if mlp is not None else DenseMLP(
+            cfg.hidden_size, cfg.intermediate_size, dtype=dtype)
         self.input_layernorm = RMSNorm(cfg.hidden_size, eps)
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, eps)
 
@@ -234,6 +235,32 @@ class HybridDecoderLayer(nn.Module):
 
 
 class Qwen3_5ForCausalLM(nn.Module):
+    """Hybrid GDN decoder; the MoE variant (Qwen3-Next / Qwen3.5-MoE,
+    reference qwen3_5_moe.py) is config-driven — layers with
+    ``num_experts > 1`` (minus ``mlp_only_layers`` /
+    ``decoder_sparse_step`` exceptions) get a routed MoE MLP."""
+
+    @staticmethod
+    def _is_sparse_layer(cfg, global_idx: int) -> bool:
+        if getattr(cfg, "num_experts", 0) in (0, 1, None):
+            return False
+        step = getattr(cfg, "decoder_sparse_step", 1) or 1
+        mlp_only = getattr(cfg, "mlp_only_layers", []) or []
+        return global_idx not in mlp_only and (global_idx + 1) % step == 0
+
+    def _make_mlp(self, cfg, engine_config, global_idx, dtype):
+        if not self._is_sparse_layer(cfg, global_idx):
+            return None  # HybridDecoderLayer builds a DenseMLP
+        from gllm_amd.models.moe_family import MoEBlock
+        return MoEBlock(
+            cfg, engine_config, dtype=dtype,
+            num_experts=cfg.num_experts,
+            top_k=cfg.num_experts_per_tok,
+            moe_intermediate=cfg.moe_intermediate_size,
+            shared_intermediate=getattr(
+                cfg, "shared_expert_intermediate_size", 0) or 0,
+            norm_topk_prob=getattr(cfg, "norm_topk_prob", True))
+
     def __init__(self, cfg, engine_config):
         super().__init__()
         self.cfg = cfg
@@ -257,13 +284,14 @@ class Qwen3_5ForCausalLM(nn.Module):
         for g in range(self.layer_start, self.layer_end):
             t = types[g]
             self._local_types.append(t)
+            mlp = self._make_mlp(cfg, engine_config, g, dtype)
             if t == "linear_attention":
                 layers.append(HybridDecoderLayer(cfg, t, ssm_idx,
-                                                 dtype=dtype))
+                                                 dtype=dtype, mlp=mlp))
                 ssm_idx += 1
             else:
                 layers.append(HybridDecoderLayer(cfg, t, kv_idx,
-                                                 dtype=dtype))
+                                                 dtype=dtype, mlp=mlp))
                 kv_idx += 1
         self.layers = nn.ModuleList(layers)
         self.num_kv_layers = kv_idx

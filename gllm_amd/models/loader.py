@@ -33,6 +33,7 @@ def get_arch_registry():
         "Qwen2_5_VLForConditionalGeneration": Qwen2VLForCausalLM,
         "Qwen3_5ForCausalLM": Qwen3_5ForCausalLM,
         "Qwen3NextForCausalLM": Qwen3_5ForCausalLM,
+        "Qwen3_5MoeForCausalLM": Qwen3_5ForCausalLM,  # MoE via config
         "DeepseekV2ForCausalLM": DeepseekV2ForCausalLM,
         "DeepseekV3ForCausalLM": DeepseekV3ForCausalLM,
         "ChatGLMModel": ChatGLMForCausalLM,

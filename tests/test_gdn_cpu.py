@@ -127,3 +127,43 @@ def test_hybrid_batched_equals_single(tmp_path):
     s2 = llm.generate([p2], [sp])[0].token_ids
     assert batched[0].token_ids == s1
     assert batched[1].token_ids == s2
+
+
+HYBRID_MOE_TINY = {
+    **HYBRID_TINY,
+    "architectures": ["Qwen3NextForCausalLM"],
+    "model_type": "qwen3_next",
+    "num_experts": 4,
+    "num_experts_per_tok": 2,
+    "moe_intermediate_size": 48,
+    "shared_expert_intermediate_size": 64,
+    "norm_topk_prob": True,
+    "decoder_sparse_step": 1,
+    "mlp_only_layers": [0],           # layer 0 keeps a dense MLP
+}
+
+
+def test_hybrid_moe_generates(tmp_path):
+    """Qwen3-Next-style hybrid: GDN + full attention + routed MoE MLP
+    (reference qwen3_5_moe.py)."""
+    import json as _json
+    d = tmp_path / "hmoe"
+    d.mkdir()
+    with open(d / "config.json", "w") as f:
+        _json.dump(HYBRID_MOE_TINY, f)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    from gllm_amd.sequence import SamplingParams
+    cfg = EngineConfig(model=str(d), load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       enable_prefix_caching=False)
+    llm = LLM(config=cfg, num_pages_override=128)
+    from gllm_amd.models.moe_family import MoEBlock
+    from gllm_amd.models.llama_family import DenseMLP
+    layers = llm.runner.model.layers
+    assert isinstance(layers[0].mlp, DenseMLP)      # mlp_only_layers
+    assert isinstance(layers[1].mlp, MoEBlock)
+    sp = [SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)]
+    o1 = llm.generate([[1, 2, 3, 4, 5, 6]], sp)[0].token_ids
+    o2 = llm.generate([[1, 2, 3, 4, 5, 6]], sp)[0].token_ids
+    assert len(o1) == 5 and o1 == o2
