@@ -88,12 +88,19 @@ class MLAAttention(nn.Module):
                                    getattr(cfg, "rope_theta", 10000.0),
                                    is_neox=True, rope_scaling=rope_scaling)
 
+    def _dsa_select(self, positions, hidden, q_resid, fctx):
+        """DSA hook (DeepSeek-V3.2): per-query top-k token positions, or
+        None for dense MLA. Overridden in models/deepseek_v32.py."""
+        return None
+
     def forward(self, positions, hidden, fctx: ForwardContext):
         T = hidden.shape[0]
         H = self.num_heads
         if self.q_lora_rank:
-            q = self.q_b_proj(self.q_a_layernorm(self.q_a_proj(hidden)))
+            q_resid = self.q_a_layernorm(self.q_a_proj(hidden))
+            q = self.q_b_proj(q_resid)
         else:
+            q_resid = None
             q = self.q_proj(hidden)
         q = q.view(T, H, self.qk_dim)
         q_nope, q_pe = q.split([self.qk_nope, self.qk_rope], dim=-1)
@@ -121,9 +128,10 @@ class MLAAttention(nn.Module):
         k_cache = fctx.k_caches[self.layer_idx]
         v_cache = fctx.v_caches[self.layer_idx]
         ops.reshape_and_cache(k, v, k_cache, v_cache, fctx.slot_mapping)
+        topk_pos = self._dsa_select(positions, hidden, q_resid, fctx)
         out = ops.mla_paged_attention(
             qf, k_cache, v_cache, fctx.block_table, fctx.seq_lens,
-            fctx.query_start_loc, self.scale)
+            fctx.query_start_loc, self.scale, topk_positions=topk_pos)
         return self.o_proj(out.reshape(T, -1))
 
 
@@ -184,10 +192,11 @@ class DeepseekMoE(nn.Module):
 
 class DeepseekDecoderLayer(nn.Module):
     def __init__(self, cfg, engine_config, layer_idx, global_idx,
-                 dtype=None):
+                 dtype=None, attn_cls=None):
         super().__init__()
         eps = getattr(cfg, "rms_norm_eps", 1e-6)
-        self.self_attn = MLAAttention(cfg, layer_idx, dtype=dtype)
+        self.self_attn = (attn_cls or MLAAttention)(cfg, layer_idx,
+                                                    dtype=dtype)
         first_dense = getattr(cfg, "first_k_dense_replace", 0)
         step = getattr(cfg, "moe_layer_freq", 1)
         is_moe = (getattr(cfg, "n_routed_experts", None)
@@ -214,6 +223,8 @@ class DeepseekDecoderLayer(nn.Module):
 
 
 class DeepseekV2ForCausalLM(nn.Module):
+    attn_cls = None  # default MLAAttention; V3.2 swaps in the DSA variant
+
     def __init__(self, cfg, engine_config):
         super().__init__()
         self.cfg = cfg
@@ -231,7 +242,8 @@ class DeepseekV2ForCausalLM(nn.Module):
                 cfg.vocab_size, cfg.hidden_size, params_dtype=dtype)
         self.layers = nn.ModuleList([
             DeepseekDecoderLayer(cfg, engine_config, local,
-                                 self.layer_start + local, dtype=dtype)
+                                 self.layer_start + local, dtype=dtype,
+                                 attn_cls=self.attn_cls)
             for local in range(self.layer_end - self.layer_start)])
         if self.is_last_stage:
             self.norm = RMSNorm(cfg.hidden_size,

@@ -1,0 +1,152 @@
+"""DeepSeek-V3.2: DeepSeek-V3 + DeepSeek Sparse Attention (DSA).
+
+Parity target: reference models/deepseek_v32.py. V3.2 is V3 (MLA +
+grouped-topk MoE + YaRN) plus a per-layer **lightning indexer**: a
+cheap side path (``self_attn.indexer.{wq_b, wk, k_norm,
+weights_proj}``) that scores every cached key against each query with
+``Σ_h w[q,h] · ReLU(scale · q[q,h,:] · k[p,:])`` and keeps only the top
+``index_topk`` (2048) token positions, fed as a sparse mask into MLA
+attention. For any causal horizon <= index_topk the top-k selects every
+key, so DSA is exactly dense there — the correctness oracle
+(reference deepseek_v32.py:455-470).
+
+MI355X round-1 scope: the fp32 torch selection path over a paged
+index-K cache parallel to the KV pool ([pages, page_size,
+index_head_dim], runtime/model_runner.py sizes it from
+``model.index_head_dim``), exact on CPU; selection returns per-query
+TOKEN POSITIONS (-1 padded) that ops.mla_paged_attention masks with.
+Round 2 (ROADMAP.md): the gfx950 tile-static scorer (graph-safe fixed
+position tiles like the reference's _INDEX_SCORE_TILE), the fp8 e4m3
+index cache + MFMA scoring kernel, and sparse gather-attention over
+physical slots. The indexer's rope is NON-interleaved (neox) on the
+first qk_rope dims — a deliberate reference quirk we match.
+"""
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from gllm_amd.layers.layernorm import RMSNorm  # noqa: F401 (doc parity)
+from gllm_amd.layers.linear import ReplicatedLinear
+from gllm_amd.layers.rotary import get_rope
+from gllm_amd.models.deepseek_v2 import (DeepseekV2ForCausalLM,
+                                         MLAAttention)
+
+
+class DSALightningIndexer(nn.Module):
+    """Reference deepseek_v32.py:86-232. Replicated (not TP-sharded):
+    the indexer is cheap and its per-head scores are summed, so
+    sharding buys little and complicates the top-k."""
+
+    def __init__(self, cfg, dtype=None):
+        super().__init__()
+        self.n_heads = cfg.index_n_heads
+        self.head_dim = cfg.index_head_dim
+        self.qk_rope = cfg.qk_rope_head_dim
+        self.index_topk = cfg.index_topk
+        self.softmax_scale = self.head_dim ** -0.5
+        q_lora = cfg.q_lora_rank
+        hidden = cfg.hidden_size
+        self.wq_b = ReplicatedLinear(q_lora, self.n_heads * self.head_dim,
+                                     params_dtype=dtype)
+        self.wk = ReplicatedLinear(hidden, self.head_dim,
+                                   params_dtype=dtype)
+        # LayerNorm (with bias) — unlike the RMSNorms elsewhere
+        self.k_norm = nn.LayerNorm(self.head_dim, eps=1e-6)
+        if dtype is not None:
+            self.k_norm = self.k_norm.to(dtype)
+        self.weights_proj = ReplicatedLinear(hidden, self.n_heads,
+                                             params_dtype=dtype)
+        rope_scaling = getattr(cfg, "rope_scaling", None)
+        max_pos = getattr(cfg, "max_position_embeddings", 32768)
+        self.rotary_emb = get_rope(self.qk_rope, self.qk_rope, max_pos,
+                                   getattr(cfg, "rope_theta", 10000.0),
+                                   is_neox=True, rope_scaling=rope_scaling)
+
+    def compute_qk(self, positions, hidden, q_resid):
+        """(q [T, Hi, D], k [T, D]) with k_norm + neox rope on the first
+        qk_rope dims."""
+        T = hidden.shape[0]
+        q = self.wq_b(q_resid).view(T, self.n_heads, self.head_dim)
+        k = self.k_norm(self.wk(hidden).float()).to(q.dtype)
+        q_rot = q[..., :self.qk_rope].reshape(
+            T, self.n_heads * self.qk_rope).contiguous()
+        k_rot = k[..., :self.qk_rope].contiguous()
+        q_rot, k_rot = self.rotary_emb(positions, q_rot, k_rot)
+        q = torch.cat([q_rot.view(T, self.n_heads, self.qk_rope),
+                       q[..., self.qk_rope:]], dim=-1)
+        k = torch.cat([k_rot, k[..., self.qk_rope:]], dim=-1)
+        return q, k
+
+    def head_weights(self, hidden):
+        """[T, Hi] fp32, with the n_heads**-0.5 normalization folded in."""
+        return self.weights_proj(hidden).float() * (self.n_heads ** -0.5)
+
+    def score(self, q, k, weights):
+        """[num_q, num_k] fp32 = Σ_h w·ReLU(scale·q·k)."""
+        s = torch.matmul(q.float(), k.float().t()) * self.softmax_scale
+        return torch.einsum("qhk,qh->qk", F.relu(s), weights)
+
+
+class DSAMLAAttention(MLAAttention):
+    """MLA attention + lightning indexer: stores this step's index keys
+    into the paged index cache, then selects each query's top-k token
+    positions (reference deepseek_v32.py:637-737 forward wiring)."""
+
+    def __init__(self, cfg, layer_idx, dtype=None):
+        super().__init__(cfg, layer_idx, dtype=dtype)
+        self.indexer = DSALightningIndexer(cfg, dtype=dtype)
+
+    @torch.no_grad()
+    def _dsa_select(self, positions, hidden, q_resid,
+                    fctx) -> Optional[torch.Tensor]:
+        idx_caches = getattr(fctx, "idx_caches", None)
+        if idx_caches is None:
+            return None  # profile run / cache not allocated
+        idx_q, idx_k = self.indexer.compute_qk(positions, hidden, q_resid)
+        weights = self.indexer.head_weights(hidden)
+        cache = idx_caches[self.layer_idx]       # [pages, page_sz, D]
+        page_sz = cache.shape[1]
+        flat = cache.view(-1, cache.shape[2])
+        flat.index_copy_(0, fctx.slot_mapping,
+                         idx_k.to(cache.dtype))
+
+        T = hidden.shape[0]
+        topk = self.indexer.index_topk
+        out = torch.full((T, topk), -1, dtype=torch.int32)
+        qsl = fctx.query_start_loc
+        for b in range(fctx.seq_lens.shape[0]):
+            qs, qe = int(qsl[b]), int(qsl[b + 1])
+            q_len = qe - qs
+            if q_len == 0:
+                continue
+            s_len = int(fctx.seq_lens[b])
+            n_pages = -(-s_len // page_sz)
+            pages = fctx.block_table[b, :n_pages].long()
+            keys = cache[pages].reshape(-1, cache.shape[2])[:s_len]
+            logits = self.indexer.score(idx_q[qs:qe], keys,
+                                        weights[qs:qe])  # [q_len, s_len]
+            past = s_len - q_len
+            pos_q = torch.arange(q_len).unsqueeze(1) + past
+            pos_k = torch.arange(s_len).unsqueeze(0)
+            logits = logits.masked_fill(pos_k > pos_q, float("-inf"))
+            k_sel = min(topk, s_len)
+            top = logits.topk(k_sel, dim=-1)
+            sel = top.indices.to(torch.int32)
+            # rows with fewer valid keys than k_sel picked -inf slots
+            sel = torch.where(torch.isinf(top.values), sel.new_full((), -1),
+                              sel)
+            out[qs:qe, :k_sel] = sel
+        return out
+
+
+class DeepseekV32ForCausalLM(DeepseekV2ForCausalLM):
+    attn_cls = DSAMLAAttention
+
+    def __init__(self, cfg, engine_config):
+        super().__init__(cfg, engine_config)
+        # runtime/model_runner.py sizes the parallel paged index-K cache
+        # from this (one per local layer, [pages, page_size, D_idx])
+        self.index_head_dim = cfg.index_head_dim

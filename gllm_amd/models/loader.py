@@ -26,6 +26,7 @@ def get_arch_registry():
     from gllm_amd.models.llama_family import MistralForCausalLM
     from gllm_amd.models.deepseek_v2 import (DeepseekV2ForCausalLM,
                                              DeepseekV3ForCausalLM)
+    from gllm_amd.models.deepseek_v32 import DeepseekV32ForCausalLM
     from gllm_amd.models.hybrid_gdn import Qwen3_5ForCausalLM
     from gllm_amd.models.qwen2_vl import Qwen2VLForCausalLM
     return {
@@ -36,6 +37,7 @@ def get_arch_registry():
         "Qwen3_5MoeForCausalLM": Qwen3_5ForCausalLM,  # MoE via config
         "DeepseekV2ForCausalLM": DeepseekV2ForCausalLM,
         "DeepseekV3ForCausalLM": DeepseekV3ForCausalLM,
+        "DeepseekV32ForCausalLM": DeepseekV32ForCausalLM,
         "ChatGLMModel": ChatGLMForCausalLM,
         "ChatGLMForConditionalGeneration": ChatGLMForCausalLM,
         "MistralForCausalLM": MistralForCausalLM,

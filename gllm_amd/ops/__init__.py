@@ -139,13 +139,14 @@ def grouped_topk(scores, topk, n_group, topk_group, renormalize=True,
 
 # --------------------------------------------------------------- mla
 def mla_paged_attention(q, k_cache, v_cache, block_table, seq_lens,
-                        query_start_loc, scale):
+                        query_start_loc, scale, topk_positions=None):
     if q.is_cuda:
         raise NotImplementedError(
             "MLA paged attention HIP kernel lands in round 2 — the "
             "DeepSeek family currently runs on the CPU reference path")
     return torch_ref.mla_paged_attention(q, k_cache, v_cache, block_table,
-                                         seq_lens, query_start_loc, scale)
+                                         seq_lens, query_start_loc, scale,
+                                         topk_positions=topk_positions)
 
 
 # --------------------------------------------------------------- gemm

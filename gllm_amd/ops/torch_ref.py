@@ -135,10 +135,19 @@ def mla_paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
                         v_cache: torch.Tensor, block_table: torch.Tensor,
                         seq_lens: torch.Tensor,
                         query_start_loc: torch.Tensor,
-                        scale: float) -> torch.Tensor:
+                        scale: float,
+                        topk_positions: Optional[torch.Tensor] = None
+                        ) -> torch.Tensor:
     """Varlen causal attention with asymmetric head dims (MLA):
     q/k have Dk (nope+rope), v has Dv. Same paged layout as
-    paged_attention; returns [T, H, Dv]."""
+    paged_attention; returns [T, H, Dv].
+
+    ``topk_positions`` (DeepSeek Sparse Attention, reference
+    deepseek_v32.py:235-792): [T, topk] int per-QUERY token POSITIONS
+    within the query's own sequence (-1 padded). When given, each query
+    row attends ONLY to its selected positions (intersected with the
+    causal mask). For topk >= seq_len the selection covers every key,
+    making sparse == dense — the DSA correctness oracle."""
     T, H, Dk = q.shape
     Dv = v_cache.shape[3]
     page_size = k_cache.shape[1]
@@ -159,7 +168,15 @@ def mla_paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
         past = s_len - q_len
         pos_q = torch.arange(q_len).unsqueeze(1) + past
         pos_k = torch.arange(s_len).unsqueeze(0)
-        scores.masked_fill_(~(pos_k <= pos_q).unsqueeze(0), float("-inf"))
+        mask = pos_k <= pos_q                          # [q_len, s_len]
+        if topk_positions is not None:
+            sel = topk_positions[qs:qe].long()         # [q_len, topk]
+            allowed = torch.zeros(q_len, s_len, dtype=torch.bool)
+            valid = (sel >= 0) & (sel < s_len)
+            rows = torch.arange(q_len).unsqueeze(1).expand_as(sel)
+            allowed[rows[valid], sel[valid]] = True
+            mask = mask & allowed
+        scores.masked_fill_(~mask.unsqueeze(0), float("-inf"))
         p = torch.softmax(scores, dim=-1)
         o = torch.einsum("hls,shd->lhd", p, v)
         out[qs:qe] = o.to(out.dtype)

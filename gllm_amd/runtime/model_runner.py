@@ -34,6 +34,8 @@ class ModelRunner:
         self.builder: Optional[BatchBuilder] = None
         self.graph_runner = None
         self.kv_dtype = None
+        self.idx_caches = None
+        self.index_head_dim = None
 
     # ------------------------------------------------------------------
     def init(self, num_pages_override: Optional[int] = None):
@@ -54,6 +56,11 @@ class ModelRunner:
             # [3,B] graph position buffers + mm content hashing: round 2
             cfg.use_graph = False
             cfg.enable_prefix_caching = False
+        self.index_head_dim = getattr(self.model, "index_head_dim", None)
+        if self.index_head_dim:
+            # DSA selector round-1 runs the eager per-seq torch path;
+            # the graph-safe tile-static scorer is round 2 (ROADMAP.md)
+            cfg.use_graph = False
         num_pages = num_pages_override or self._size_kv_cache()
         self._allocate_kv(num_pages)
         self.num_kv_pages_total = num_pages
@@ -203,6 +210,14 @@ class ModelRunner:
         self.v_caches = [torch.zeros(vshape, dtype=self.kv_dtype,
                                      device=self.device)
                          for _ in range(spec.num_layers)]
+        # DSA (DeepSeek-V3.2): paged index-K cache parallel to the KV
+        # pool, one per local layer (reference memory_manager.py:334-362)
+        self.idx_caches = None
+        if self.index_head_dim:
+            ishape = (num_pages, spec.page_size, self.index_head_dim)
+            self.idx_caches = [torch.zeros(ishape, dtype=self.kv_dtype,
+                                           device=self.device)
+                               for _ in range(spec.num_layers)]
 
     # ------------------------------------------------------------------
     def _stage_forward(self, input_ids, positions, fctx,
@@ -223,6 +238,7 @@ class ModelRunner:
             use_mrope=self.uses_mrope)
         self._attach_ssm(batch, fctx)
         self._attach_mm(batch, fctx)
+        fctx.idx_caches = self.idx_caches
         if fctx.has_placeholders:
             tokens = self.resolve_tokens(tokens)
         hidden, residual = self._stage_forward(tokens, fctx.positions, fctx)
@@ -238,6 +254,7 @@ class ModelRunner:
             need_logits=self.model.is_last_stage,
             use_mrope=self.uses_mrope)
         self._attach_ssm(batch, fctx)
+        fctx.idx_caches = self.idx_caches
         hidden, residual = self._stage_forward(
             None, fctx.positions, fctx, hidden_states=hidden,
             residual=residual)
