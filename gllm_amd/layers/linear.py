@@ -39,6 +39,25 @@ class LinearBase(nn.Module):
                                      requires_grad=False)
         else:
             self.register_parameter("bias", None)
+        # set by quantization/fp8.py convert_linear_to_fp8
+        self.fp8_block = None
+        self._w_dq = None
+
+    def _gemm(self, x, bias=None):
+        """GEMM entry for all subclasses. fp8-block weights are
+        dequantized once (lazily, cached) into the compute dtype —
+        the fp8 MFMA GEMM consuming them natively is round 2
+        (layers/quantization/fp8.py docstring)."""
+        w = self.weight
+        if self.fp8_block is not None:
+            if self._w_dq is None:
+                from gllm_amd.layers.quantization.fp8 import \
+                    dequant_block_fp8
+                self._w_dq = dequant_block_fp8(
+                    w, self.weight_scale_inv, self.fp8_block,
+                    x.dtype).to(x.device)
+            w = self._w_dq
+        return ops.linear(x, w, bias)
 
 
 class ReplicatedLinear(LinearBase):
@@ -53,7 +72,7 @@ class ReplicatedLinear(LinearBase):
         param.data.copy_(loaded)
 
     def forward(self, x):
-        return ops.linear(x, self.weight, self.bias)
+        return self._gemm(x, self.bias)
 
 
 class ColumnParallelLinear(LinearBase):
@@ -75,7 +94,7 @@ class ColumnParallelLinear(LinearBase):
         _narrow_copy(param.data, loaded, 0, get_tp_rank(), get_tp_size())
 
     def forward(self, x):
-        out = ops.linear(x, self.weight, self.bias)
+        out = self._gemm(x, self.bias)
         if self.gather_output:
             from gllm_amd.parallel import tensor_parallel_all_gather
             out = tensor_parallel_all_gather(out, dim=-1)
@@ -107,7 +126,7 @@ class MergedColumnParallelLinear(LinearBase):
             loaded.narrow(0, tp_rank * size, size))
 
     def forward(self, x):
-        return ops.linear(x, self.weight, self.bias)
+        return self._gemm(x, self.bias)
 
 
 class QKVParallelLinear(LinearBase):
@@ -152,7 +171,7 @@ class QKVParallelLinear(LinearBase):
             loaded.narrow(0, src_rank * size, size))
 
     def forward(self, x):
-        out = ops.linear(x, self.weight, self.bias)
+        out = self._gemm(x, self.bias)
         return out.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
 
 
@@ -179,7 +198,7 @@ class RowParallelLinear(LinearBase):
         param.data.copy_(loaded)
 
     def forward(self, x):
-        out = ops.linear(x, self.weight)
+        out = self._gemm(x)
         if self.reduce_results:
             out = tensor_parallel_all_reduce(out)
         # bias applied once, after the reduction

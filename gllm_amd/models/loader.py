@@ -144,6 +144,15 @@ def load_model(engine_config, device: str = "cpu"):
     t0 = time.time()
     cfg = load_hf_config(engine_config.model)
     model = create_model(cfg, engine_config, device)
+    qcfg = getattr(cfg, "quantization_config", None)
+    if qcfg is not None and not isinstance(qcfg, dict):
+        qcfg = getattr(qcfg, "to_dict", lambda: vars(qcfg))()
+    if qcfg and qcfg.get("quant_method") == "fp8" and \
+            engine_config.load_format != "dummy":
+        from gllm_amd.layers.quantization.fp8 import convert_model_to_fp8
+        n = convert_model_to_fp8(model, qcfg)
+        logger.info("fp8 block-quant checkpoint: converted %d linears "
+                    "(block %s)", n, qcfg.get("weight_block_size"))
     if engine_config.load_format == "dummy":
         dummy_init(model, engine_config.seed)
     else:
