@@ -28,6 +28,7 @@ class RequestOutput:
         self.finish_reason = seq.finish_reason
         self.text = text
         self.logprobs = seq.out_logprobs or None
+        self.prompt_logprobs = seq.prompt_logprobs_out or None
 
 
 class LLM:

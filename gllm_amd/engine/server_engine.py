@@ -29,6 +29,8 @@ class StreamChunk:
     text: str
     finish_reason: Optional[str]
     n_output_tokens: int
+    # attached to the FINAL chunk when sampling.prompt_logprobs was set
+    prompt_logprobs: Optional[list] = None
 
 
 class RequestState:
@@ -44,6 +46,7 @@ class RequestState:
         self.finished = False
         self.created = time.time()
         self.first_token_time: Optional[float] = None
+        self.prompt_logprobs: Optional[list] = None
 
 
 class AsyncLLMEngine:
@@ -157,6 +160,13 @@ class AsyncLLMEngine:
             if kind == "stats":
                 self.latest_stats = outs
                 continue
+            if kind == "plp":
+                # prompt logprobs arrive just before the finish token
+                for seq_id, plp in outs:
+                    st = self.requests.get(seq_id)
+                    if st is not None:
+                        st.prompt_logprobs = plp
+                continue
             if kind == "worker_dead":
                 logger.error("remote worker %s died — failing %d pending "
                              "requests", outs, len(self.requests))
@@ -190,7 +200,9 @@ class AsyncLLMEngine:
                     self.abort([st.seq_id])
         if finish is not None:
             st.finished = True
-        chunk = StreamChunk(st.seq_id, token_id, text, finish, st.n_tokens)
+        chunk = StreamChunk(st.seq_id, token_id, text, finish, st.n_tokens,
+                            prompt_logprobs=st.prompt_logprobs
+                            if finish is not None else None)
         st.loop.call_soon_threadsafe(st.queue.put_nowait, chunk)
         if st.finished:
             st.loop.call_soon_threadsafe(st.queue.put_nowait, None)

@@ -52,6 +52,7 @@ class ServingMixin:
         else:
             self.is_output_rank = self.rank == 0
         self._req_counter = 0  # deterministic DP round-robin routing
+        self._seqs_by_id = {}  # live seqs (for prompt-logprob emission)
         self.comm = WorkerComm(req_queue, out_queue, self.is_output_rank)
         self._intake_buf = torch.zeros(1, dtype=torch.int64)
         if self.world > 1 and config.device.startswith("cuda"):
@@ -74,6 +75,7 @@ class ServingMixin:
                 seq = Sequence(payload["seq_id"], payload["token_ids"],
                                SamplingParams(**payload["sampling"]),
                                eos_token_id=payload.get("eos_token_id"))
+                self._seqs_by_id[seq.seq_id] = seq
                 self.scheduler.add_seqs([seq])
             elif kind == "abort":
                 self.scheduler.abort_seqs(payload)
@@ -125,6 +127,13 @@ class ServingMixin:
             if n:
                 self._apply_messages(self.comm.recv_blocking(n))
 
+    def _emit_prompt_logprobs(self, seq: Sequence) -> None:
+        """Ship prompt logprobs BEFORE the finish token so the frontend
+        attaches them to the final stream chunk."""
+        if seq.sampling.prompt_logprobs and seq.prompt_logprobs_out:
+            self.comm.send_output(
+                ("plp", [(seq.seq_id, seq.prompt_logprobs_out)], {}))
+
     # ------------------------------------------------------------------
     def _maybe_send_stats(self) -> None:
         # one stats stream (rank 0 reports its own replica under DP)
@@ -172,8 +181,12 @@ class ServingWorker(ServingMixin, PPEngine):
                 else:
                     outs.append((seq.seq_id, seq.token_ids[-1],
                                  seq.finish_reason))
+                if seq.finish_reason:
+                    self._emit_prompt_logprobs(seq)
             if outs:
                 self.comm.send_output(("out", outs, {}))
+        for s in finished:
+            self._seqs_by_id.pop(s.seq_id, None)
         return finished
 
     def _run_loop_dp(self) -> None:
@@ -233,7 +246,13 @@ class OverlapServingWorker(ServingMixin, OverlapEngine):
             self.on_finalized = self._emit
 
     def _emit(self, emissions) -> None:
-        self.comm.send_output(("out", list(emissions), {}))
+        emissions = list(emissions)
+        for seq_id, _tok, fin in emissions:
+            if fin:
+                seq = self._seqs_by_id.pop(seq_id, None)
+                if seq is not None:
+                    self._emit_prompt_logprobs(seq)
+        self.comm.send_output(("out", emissions, {}))
 
     def run_loop(self) -> None:
         logger.info("worker %d ready (overlap)", self.rank)

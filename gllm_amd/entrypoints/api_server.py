@@ -107,7 +107,18 @@ def build_app():
             min_tokens=req.min_tokens or 0,
             ignore_eos=req.ignore_eos,
             stop=stops, stop_token_ids=req.stop_token_ids,
-            seed=req.seed)
+            seed=req.seed,
+            prompt_logprobs=getattr(req, "prompt_logprobs", None))
+
+    def _vary_seed(sampling: SamplingParams, j: int) -> SamplingParams:
+        """Choice j of an n>1 request: distinct seed per choice when one
+        was given, shared params otherwise."""
+        import dataclasses as _dc
+        if j == 0:
+            return sampling
+        return _dc.replace(sampling,
+                           seed=None if sampling.seed is None
+                           else sampling.seed + j)
 
     @app.post("/v1/chat/completions")
     async def chat_completions(req: ChatCompletionRequest, raw: Request):
@@ -125,56 +136,100 @@ def build_app():
             return StreamingResponse(
                 _chat_stream(req, raw, token_ids, sampling),
                 media_type="text/event-stream")
-        text, finish, n_out = await _collect(raw, token_ids, sampling)
-        message = ChatMessage(role="assistant", content=text)
-        if req.tools:
-            from gllm_amd.tokenizers.tool_parsers import parse_tool_calls
-            parsed_text, calls = parse_tool_calls(text, served_model)
-            if calls:
-                message = ChatMessage(role="assistant",
-                                      content=parsed_text or None,
-                                      tool_calls=calls)
-                finish = "tool_calls"
+        n = req.n or 1
+        import asyncio as _aio
+        results = await _aio.gather(*[
+            _collect(raw, token_ids, _vary_seed(sampling, j))
+            for j in range(n)])
+        choices = []
+        total_out = 0
+        for j, (text, finish, n_out, _plp) in enumerate(results):
+            total_out += n_out
+            message = ChatMessage(role="assistant", content=text)
+            if req.tools:
+                from gllm_amd.tokenizers.tool_parsers import \
+                    parse_tool_calls
+                parsed_text, calls = parse_tool_calls(text, served_model)
+                if calls:
+                    message = ChatMessage(role="assistant",
+                                          content=parsed_text or None,
+                                          tool_calls=calls)
+                    finish = "tool_calls"
+            choices.append(ChatCompletionResponseChoice(
+                index=j, message=message, finish_reason=finish))
         resp = ChatCompletionResponse(
             model=req.model or served_model,
-            choices=[ChatCompletionResponseChoice(
-                index=0, message=message, finish_reason=finish)],
+            choices=choices,
             usage=UsageInfo(prompt_tokens=len(token_ids),
-                            completion_tokens=n_out,
-                            total_tokens=len(token_ids) + n_out))
+                            completion_tokens=total_out,
+                            total_tokens=len(token_ids) + total_out))
         return resp.model_dump()
 
     async def _collect(raw, token_ids, sampling):
         text_parts = []
         finish = None
         n_out = 0
+        plp = None
         async for chunk in engine.generate_stream(token_ids, sampling):
             if await raw.is_disconnected():
                 break
             text_parts.append(chunk.text)
             finish = chunk.finish_reason or finish
             n_out = chunk.n_output_tokens
+            plp = chunk.prompt_logprobs or plp
         text = "".join(text_parts)
         from gllm_amd.engine.detokenizer import check_stop_strings
         _, text = check_stop_strings(text, sampling.stop)
-        return text, finish, n_out
+        return text, finish, n_out, plp
+
+    async def _merged_stream(raw, token_ids, sampling, n):
+        """Run n generations concurrently, yield (choice_idx, chunk) in
+        arrival order (n>1 streaming: interleaved choices)."""
+        import asyncio as _aio
+        if n <= 1:
+            async for chunk in engine.generate_stream(token_ids, sampling):
+                if await raw.is_disconnected():
+                    return
+                yield 0, chunk
+            return
+        q: _aio.Queue = _aio.Queue()
+
+        async def pump(j):
+            async for chunk in engine.generate_stream(
+                    token_ids, _vary_seed(sampling, j)):
+                await q.put((j, chunk))
+            await q.put((j, None))
+
+        tasks = [_aio.ensure_future(pump(j)) for j in range(n)]
+        live = n
+        try:
+            while live:
+                j, chunk = await q.get()
+                if chunk is None:
+                    live -= 1
+                    continue
+                if await raw.is_disconnected():
+                    return
+                yield j, chunk
+        finally:
+            for t in tasks:
+                t.cancel()
 
     async def _chat_stream(req, raw, token_ids, sampling):
         resp_id = None
-        first = True
+        first_for = set()
+        n = req.n or 1
         n_out = 0
-        async for chunk in engine.generate_stream(token_ids, sampling):
-            if await raw.is_disconnected():
-                break
+        async for j, chunk in _merged_stream(raw, token_ids, sampling, n):
             delta = DeltaMessage(content=chunk.text)
-            if first:
+            if j not in first_for:
                 delta.role = "assistant"
-                first = False
-            n_out = chunk.n_output_tokens
+                first_for.add(j)
+            n_out += 1 if chunk.token_id >= 0 else 0
             out = ChatCompletionStreamResponse(
                 model=req.model or served_model,
                 choices=[ChatCompletionStreamChoice(
-                    index=0, delta=delta,
+                    index=j, delta=delta,
                     finish_reason=chunk.finish_reason)])
             if resp_id is None:
                 resp_id = out.id
@@ -205,17 +260,23 @@ def build_app():
             return StreamingResponse(
                 _completion_stream(req, raw, token_ids, sampling),
                 media_type="text/event-stream")
+        n = req.n or 1
         choices = []
         total_p = total_c = 0
         for i, p in enumerate(prompts):
             token_ids = p if isinstance(p, list) else engine.encode(p)
-            text, finish, n_out = await _collect(raw, token_ids, sampling)
-            if req.echo and not isinstance(p, list):
-                text = p + text
-            choices.append(CompletionResponseChoice(
-                index=i, text=text, finish_reason=finish))
+            import asyncio as _aio
+            results = await _aio.gather(*[
+                _collect(raw, token_ids, _vary_seed(sampling, j))
+                for j in range(n)])
+            for j, (text, finish, n_out, plp) in enumerate(results):
+                if req.echo and not isinstance(p, list):
+                    text = p + text
+                choices.append(CompletionResponseChoice(
+                    index=i * n + j, text=text, finish_reason=finish,
+                    prompt_logprobs=plp if req.prompt_logprobs else None))
+                total_c += n_out
             total_p += len(token_ids)
-            total_c += n_out
         resp = CompletionResponse(
             model=req.model or served_model, choices=choices,
             usage=UsageInfo(prompt_tokens=total_p, completion_tokens=total_c,
@@ -223,13 +284,12 @@ def build_app():
         return resp.model_dump()
 
     async def _completion_stream(req, raw, token_ids, sampling):
-        async for chunk in engine.generate_stream(token_ids, sampling):
-            if await raw.is_disconnected():
-                break
+        n = req.n or 1
+        async for j, chunk in _merged_stream(raw, token_ids, sampling, n):
             resp = CompletionResponse(
                 model=req.model or served_model,
                 choices=[CompletionResponseChoice(
-                    index=0, text=chunk.text,
+                    index=j, text=chunk.text,
                     finish_reason=chunk.finish_reason)])
             yield f"data: {resp.model_dump_json(exclude_none=True)}\n\n"
         yield "data: [DONE]\n\n"

@@ -65,6 +65,7 @@ class ChatCompletionRequest(BaseModel):
     tools: Optional[List[ToolDef]] = None
     tool_choice: Optional[Union[str, Dict[str, Any]]] = None
     chat_template_kwargs: Optional[Dict[str, Any]] = None
+    prompt_logprobs: Optional[int] = None
 
 
 class CompletionRequest(BaseModel):
@@ -84,6 +85,7 @@ class CompletionRequest(BaseModel):
     ignore_eos: bool = False
     echo: bool = False
     logprobs: Optional[int] = None
+    prompt_logprobs: Optional[int] = None
 
 
 class UsageInfo(BaseModel):
@@ -133,6 +135,7 @@ class CompletionResponseChoice(BaseModel):
     text: str
     finish_reason: Optional[str] = None
     logprobs: Optional[Dict[str, Any]] = None
+    prompt_logprobs: Optional[List[Any]] = None
 
 
 class CompletionResponse(BaseModel):

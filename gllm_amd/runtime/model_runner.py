@@ -302,7 +302,47 @@ class ModelRunner:
         idx = (-tokens - 1).clamp_min(0)
         return torch.where(tokens < 0, ring_flat[idx], tokens)
 
+    def _prompt_logprobs(self, batch: ScheduledBatch, hidden, fctx):
+        """Prompt logprobs (reference model_runner.py:1700-1807): for
+        seqs requesting them, the prefill-chunk rows' logits score the
+        NEXT prompt token; computed chunk by chunk so chunked prefill
+        accumulates the full prompt. PP=1 scope (the values live where
+        sampling runs)."""
+        want = [i for i, it in enumerate(batch.items)
+                if it.seq.sampling.prompt_logprobs
+                and it.start < it.seq.prompt_len - 1]
+        if not want:
+            return
+        import types
+        qsl = fctx.query_start_loc
+        shim = types.SimpleNamespace(logits_indices=None)
+        for i in want:
+            item = batch.items[i]
+            seq = item.seq
+            k = seq.sampling.prompt_logprobs
+            qs, qe = int(qsl[i]), int(qsl[i + 1])
+            n_rows = qe - qs
+            # row r predicts prompt position item.start + r + 1
+            last = min(n_rows, seq.prompt_len - 1 - item.start)
+            if last <= 0:
+                continue
+            # preemption recompute restarts the prefill: drop entries
+            # this chunk is about to recompute
+            del seq.prompt_logprobs_out[item.start:]
+            logits = self.model.compute_logits(hidden[qs:qs + last], shim)
+            logp = torch.log_softmax(logits.float(), dim=-1)
+            targets = torch.tensor(
+                seq.token_ids[item.start + 1:item.start + 1 + last],
+                device=logp.device)
+            chosen = logp.gather(1, targets.unsqueeze(1)).squeeze(1)
+            topv, topi = logp.topk(min(k, logp.shape[-1]), dim=-1)
+            for r in range(last):
+                seq.prompt_logprobs_out.append(
+                    (float(chosen[r]),
+                     dict(zip(topi[r].tolist(), topv[r].tolist()))))
+
     def _sample(self, batch: ScheduledBatch, hidden, fctx):
+        self._prompt_logprobs(batch, hidden, fctx)
         logits = self.model.compute_logits(hidden, fctx)
         meta = build_sampling_metadata(batch.items, logits.device,
                                        penalty_pool=self.penalty_pool)

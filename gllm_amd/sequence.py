@@ -75,6 +75,9 @@ class Sequence:
         # per-output-token logprobs (filled when sampling.logprobs is set):
         # list of (chosen_logprob, {token_id: logprob} top-k)
         self.out_logprobs: List[tuple] = []
+        # prompt logprobs (sampling.prompt_logprobs): one entry per
+        # prompt position >= 1, (logprob_of_actual_token, {tok: lp})
+        self.prompt_logprobs_out: List[tuple] = []
 
         # --- multimodal (set by the engine at admission) ---
         # concatenated vision embeddings for all image spans [N, hidden]

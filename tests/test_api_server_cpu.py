@@ -92,6 +92,26 @@ def test_api_server_routes(tmp_path):
         assert r3.status_code == 200, r3.text
         assert r3.json()["usage"]["completion_tokens"] == 3
 
+        # n>1: greedy choices must be identical and indexed 0..n-1
+        r4 = client.post("/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "w5 w6"}],
+            "max_tokens": 4, "temperature": 0.0, "n": 2,
+            "ignore_eos": True})
+        assert r4.status_code == 200, r4.text
+        ch = r4.json()["choices"]
+        assert [c["index"] for c in ch] == [0, 1]
+        assert ch[0]["message"]["content"] == ch[1]["message"]["content"]
+        assert r4.json()["usage"]["completion_tokens"] == 8
+
+        # prompt_logprobs on completions: prompt_len-1 entries
+        r5 = client.post("/v1/completions", json={
+            "prompt": "w3 w4 w5 w6", "max_tokens": 2, "temperature": 0.0,
+            "prompt_logprobs": 3, "ignore_eos": True})
+        assert r5.status_code == 200, r5.text
+        plp = r5.json()["choices"][0]["prompt_logprobs"]
+        assert plp is not None and len(plp) == 3
+        assert all(len(entry) == 2 and len(entry[1]) == 3 for entry in plp)
+
         # metrics + stats
         m = client.get("/metrics").text
         assert "gllm_requests_total" in m
