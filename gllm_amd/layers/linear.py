@@ -39,15 +39,16 @@ class LinearBase(nn.Module):
                                      requires_grad=False)
         else:
             self.register_parameter("bias", None)
-        # set by quantization/fp8.py convert_linear_to_fp8
+        # set by quantization/{fp8,int4}.py converters
         self.fp8_block = None
+        self.int4_cfg = None
         self._w_dq = None
 
     def _gemm(self, x, bias=None):
-        """GEMM entry for all subclasses. fp8-block weights are
-        dequantized once (lazily, cached) into the compute dtype —
-        the fp8 MFMA GEMM consuming them natively is round 2
-        (layers/quantization/fp8.py docstring)."""
+        """GEMM entry for all subclasses. fp8-block / int4 weights are
+        dequantized once (lazily, cached) into the compute dtype — the
+        dequant-fused MFMA GEMMs consuming them natively are round 2
+        (layers/quantization/ docstrings)."""
         w = self.weight
         if self.fp8_block is not None:
             if self._w_dq is None:
@@ -56,6 +57,11 @@ class LinearBase(nn.Module):
                 self._w_dq = dequant_block_fp8(
                     w, self.weight_scale_inv, self.fp8_block,
                     x.dtype).to(x.device)
+            w = self._w_dq
+        elif self.int4_cfg is not None:
+            if self._w_dq is None:
+                from gllm_amd.layers.quantization.int4 import dequant_layer
+                self._w_dq = dequant_layer(self, x.dtype).to(x.device)
             w = self._w_dq
         return ops.linear(x, w, bias)
 

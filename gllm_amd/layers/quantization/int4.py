@@ -1,0 +1,205 @@
+"""int4 weight-only quantization (AWQ / GPTQ checkpoint formats).
+
+Parity target: reference layers/moe/fused_moe_triton/layer.py
+Int4MarlinMoEMethod + the gptq-marlin dense path (_custom_ops.py:
+gptq_marlin_repack / moe_wna16_marlin_gemm). MI355X stance: the round-2
+path is an LDS-resident dequant-fused MFMA GEMM (weights stay packed in
+registers/LDS, unpacked on the fly — int4 is memory-bound decode's best
+friend at 4x bf16 density); round-1 executes like fp8: weights load in
+the checkpoint's packed layout and are dequantized lazily (cached) into
+the compute dtype, so numerics are exact w.r.t. the format. TP sharding
+of packed tensors is round 2 — int4 loads require tp_size == 1.
+
+Formats (public conventions):
+  GPTQ (bits=4, no act-order): qweight int32 [K/8, N], K packed
+    little-nibble-first; qzeros int32 [G, N/8] (stored zero is z-1: add
+    1 when unpacking — the autogptq legacy quirk); scales fp16 [G, N];
+    w[k, n] = s[g, n] * (q[k, n] - (z[g, n] + 1)),  g = k // group.
+  AWQ (bits=4): qweight int32 [K, N/8] with the interleaved nibble
+    order [0, 2, 4, 6, 1, 3, 5, 7]; qzeros int32 [G, N/8] same order;
+    scales fp16 [G, N]; w[k, n] = s * (q - z).
+Both store weights K-major ([K, N]); torch linear wants [N, K], so
+dequant transposes at the end.
+"""
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+AWQ_ORDER = (0, 2, 4, 6, 1, 3, 5, 7)
+
+
+def _unpack_nibbles_k(packed: torch.Tensor) -> torch.Tensor:
+    """GPTQ qweight [K/8, N] int32 -> [K, N] int, nibble i of word w is
+    row 8*w + i."""
+    Kp, N = packed.shape
+    out = torch.empty(Kp * 8, N, dtype=torch.int32)
+    p = packed.to(torch.int64) & 0xFFFFFFFF
+    for i in range(8):
+        # word w's nibble i is row 8*w + i == out[i::8][w]
+        out[i::8] = ((p >> (4 * i)) & 0xF).to(torch.int32)
+    return out
+
+
+def _unpack_nibbles_n(packed: torch.Tensor, order=None) -> torch.Tensor:
+    """[R, N/8] int32 -> [R, N] int; nibble i of a word is column
+    8*w + order[i] (AWQ interleave) or 8*w + i (GPTQ zeros)."""
+    R, Np = packed.shape
+    order = order or tuple(range(8))
+    out = torch.empty(R, Np * 8, dtype=torch.int32)
+    p = packed.to(torch.int64) & 0xFFFFFFFF
+    for i, col in enumerate(order):
+        out[:, col::8] = ((p >> (4 * i)) & 0xF).to(torch.int32)
+    return out
+
+
+def dequant_gptq(qweight, qzeros, scales, group_size: int,
+                 dtype: torch.dtype) -> torch.Tensor:
+    """-> [N, K] dense weight."""
+    wq = _unpack_nibbles_k(qweight)                   # [K, N]
+    zeros = _unpack_nibbles_n(qzeros) + 1             # [G, N] (legacy +1)
+    K = wq.shape[0]
+    g = torch.arange(K) // group_size
+    w = (wq.float() - zeros.float()[g]) * scales.float()[g]
+    return w.t().contiguous().to(dtype)
+
+
+def dequant_awq(qweight, qzeros, scales, group_size: int,
+                dtype: torch.dtype) -> torch.Tensor:
+    """-> [N, K] dense weight."""
+    wq = _unpack_nibbles_n(qweight, AWQ_ORDER)        # [K, N]
+    zeros = _unpack_nibbles_n(qzeros, AWQ_ORDER)      # [G, N]
+    K = wq.shape[0]
+    g = torch.arange(K) // group_size
+    w = (wq.float() - zeros.float()[g]) * scales.float()[g]
+    return w.t().contiguous().to(dtype)
+
+
+# --------------------------------------------------------- test packing
+def pack_gptq(w: torch.Tensor, group_size: int):
+    """[N, K] fp -> (qweight, qzeros, scales); round-trip oracle for
+    tests (asymmetric per-group min/max quant)."""
+    N, K = w.shape
+    assert K % group_size == 0 and K % 8 == 0 and N % 8 == 0
+    wk = w.t().float()                                # [K, N]
+    G = K // group_size
+    wg = wk.view(G, group_size, N)
+    mn, mx = wg.min(1).values, wg.max(1).values       # [G, N]
+    scales = ((mx - mn) / 15.0).clamp_min(1e-8)
+    zeros = (-mn / scales).round().clamp(0, 15)       # [G, N]
+    g = torch.arange(K) // group_size
+    q = (wk / scales[g] + zeros[g]).round().clamp(0, 15).to(torch.int64)
+    qweight = torch.zeros(K // 8, N, dtype=torch.int64)
+    for i in range(8):
+        qweight |= q[i::8] << (4 * i)
+    zstore = (zeros.to(torch.int64) - 1).clamp(0, 15)  # legacy z-1
+    qzeros = torch.zeros(G, N // 8, dtype=torch.int64)
+    for i in range(8):
+        qzeros |= zstore[:, i::8] << (4 * i)
+    return (qweight.to(torch.int32), qzeros.to(torch.int32),
+            scales.to(torch.float16))
+
+
+def pack_awq(w: torch.Tensor, group_size: int):
+    N, K = w.shape
+    assert K % group_size == 0 and N % 8 == 0
+    wk = w.t().float()
+    G = K // group_size
+    wg = wk.view(G, group_size, N)
+    mn, mx = wg.min(1).values, wg.max(1).values
+    scales = ((mx - mn) / 15.0).clamp_min(1e-8)
+    zeros = (-mn / scales).round().clamp(0, 15)
+    g = torch.arange(K) // group_size
+    q = (wk / scales[g] + zeros[g]).round().clamp(0, 15).to(torch.int64)
+    qweight = torch.zeros(K, N // 8, dtype=torch.int64)
+    zq = zeros.to(torch.int64)
+    qzeros = torch.zeros(G, N // 8, dtype=torch.int64)
+    for i, col in enumerate(AWQ_ORDER):
+        qweight |= q[:, col::8] << (4 * i)
+        qzeros |= zq[:, col::8] << (4 * i)
+    return (qweight.to(torch.int32), qzeros.to(torch.int32),
+            scales.to(torch.float16))
+
+
+# ------------------------------------------------------ layer conversion
+_QUANT_SUFFIXES = (
+    "q_proj", "k_proj", "v_proj", "o_proj", "qkv_proj",
+    "gate_proj", "up_proj", "down_proj", "gate_up_proj",
+)
+
+
+def convert_linear_to_int4(layer, method: str, group_size: int) -> None:
+    """Replace the dense weight with packed int4 params. Checkpoint
+    names (qweight/qzeros/scales) route to them directly through the
+    models' substring-mapped loaders."""
+    from gllm_amd.layers.linear import (MergedColumnParallelLinear,
+                                        QKVParallelLinear)
+    from gllm_amd.parallel import get_tp_size
+    assert get_tp_size() == 1, \
+        "int4 TP sharding of packed tensors is round 2 (ROADMAP.md)"
+    N, K = layer.weight.shape
+    G = K // group_size
+
+    def _mk_loader(div: int):
+        """Place a sub-projection's packed tensor into its dim-1 slice
+        of the fused param (all int4 tensors concat along N; ``div`` is
+        the N packing factor of that tensor: 8 for qzeros / AWQ qweight,
+        1 for scales / GPTQ qweight)."""
+        if isinstance(layer, MergedColumnParallelLinear):
+            def load(param, loaded, shard_id: int):
+                off = sum(layer.output_sizes[:shard_id]) // div
+                size = layer.output_sizes[shard_id] // div
+                param.data.narrow(1, off, size).copy_(loaded)
+        elif isinstance(layer, QKVParallelLinear):
+            def load(param, loaded, shard_id: str):
+                offs = {"q": 0, "k": layer.q_size,
+                        "v": layer.q_size + layer.kv_size}
+                sizes = {"q": layer.q_size, "k": layer.kv_size,
+                         "v": layer.kv_size}
+                param.data.narrow(1, offs[shard_id] // div,
+                                  sizes[shard_id] // div).copy_(loaded)
+        else:
+            def load(param, loaded, *a):
+                param.data.copy_(loaded)
+        return load
+
+    if method == "gptq":
+        qw_shape, qw_div = (K // 8, N), 1
+    else:
+        qw_shape, qw_div = (K, N // 8), 8
+    layer.qweight = nn.Parameter(
+        torch.zeros(qw_shape, dtype=torch.int32), requires_grad=False)
+    layer.qweight.weight_loader = _mk_loader(qw_div)
+    layer.qzeros = nn.Parameter(
+        torch.zeros(G, N // 8, dtype=torch.int32), requires_grad=False)
+    layer.qzeros.weight_loader = _mk_loader(8)
+    layer.scales = nn.Parameter(
+        torch.zeros(G, N, dtype=torch.float16), requires_grad=False)
+    layer.scales.weight_loader = _mk_loader(1)
+    # drop the dense weight (it has no checkpoint tensor)
+    layer.weight = nn.Parameter(torch.zeros(1, dtype=torch.float32),
+                                requires_grad=False)
+    layer.weight.weight_loader = lambda param, loaded, *a: None
+    layer.int4_cfg = (method, group_size)
+    layer._w_dq = None
+
+
+def convert_model_to_int4(model, quant_config: dict) -> int:
+    from gllm_amd.layers.linear import LinearBase
+    method = quant_config["quant_method"]
+    group = int(quant_config.get("group_size", 128))
+    n = 0
+    for name, mod in model.named_modules():
+        if not isinstance(mod, LinearBase):
+            continue
+        if name.rsplit(".", 1)[-1] in _QUANT_SUFFIXES:
+            convert_linear_to_int4(mod, method, group)
+            n += 1
+    return n
+
+
+def dequant_layer(layer, dtype: torch.dtype) -> torch.Tensor:
+    method, group = layer.int4_cfg
+    fn = dequant_gptq if method == "gptq" else dequant_awq
+    return fn(layer.qweight, layer.qzeros, layer.scales, group, dtype)

@@ -147,12 +147,25 @@ def load_model(engine_config, device: str = "cpu"):
     qcfg = getattr(cfg, "quantization_config", None)
     if qcfg is not None and not isinstance(qcfg, dict):
         qcfg = getattr(qcfg, "to_dict", lambda: vars(qcfg))()
-    if qcfg and qcfg.get("quant_method") == "fp8" and \
-            engine_config.load_format != "dummy":
-        from gllm_amd.layers.quantization.fp8 import convert_model_to_fp8
-        n = convert_model_to_fp8(model, qcfg)
-        logger.info("fp8 block-quant checkpoint: converted %d linears "
-                    "(block %s)", n, qcfg.get("weight_block_size"))
+    if qcfg and engine_config.load_format != "dummy":
+        method = qcfg.get("quant_method")
+        if method == "fp8":
+            from gllm_amd.layers.quantization.fp8 import \
+                convert_model_to_fp8
+            n = convert_model_to_fp8(model, qcfg)
+            logger.info("fp8 block-quant checkpoint: converted %d linears "
+                        "(block %s)", n, qcfg.get("weight_block_size"))
+        elif method in ("awq", "gptq"):
+            from gllm_amd.layers.quantization.int4 import \
+                convert_model_to_int4
+            bits = qcfg.get("bits", qcfg.get("w_bit", 4))
+            assert bits == 4, f"only 4-bit {method} supported"
+            n = convert_model_to_int4(model, {
+                "quant_method": method,
+                "group_size": qcfg.get("group_size",
+                                       qcfg.get("q_group_size", 128))})
+            logger.info("%s int4 checkpoint: converted %d linears",
+                        method, n)
     if engine_config.load_format == "dummy":
         dummy_init(model, engine_config.seed)
     else:

@@ -1,0 +1,135 @@
+"""int4 AWQ/GPTQ weight-only quantization tests.
+
+Same contract as fp8: loading a packed int4 checkpoint must produce
+EXACTLY the outputs of a plain checkpoint holding the manually
+dequantized weights (pack/unpack conventions in
+layers/quantization/int4.py)."""
+
+import json
+import os
+
+import pytest
+import torch
+
+from gllm_amd.layers.quantization.int4 import (dequant_awq, dequant_gptq,
+                                               pack_awq, pack_gptq)
+
+CFG = {
+    "architectures": ["Qwen2ForCausalLM"],
+    "model_type": "qwen2",
+    "hidden_size": 64,
+    "intermediate_size": 128,
+    "num_hidden_layers": 2,
+    "num_attention_heads": 4,
+    "num_key_value_heads": 2,
+    "vocab_size": 128,
+    "max_position_embeddings": 2048,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 10000.0,
+    "tie_word_embeddings": False,
+    "eos_token_id": 0,
+}
+GROUP = 16
+QUANT_KEYS = ("q_proj", "k_proj", "v_proj", "o_proj",
+              "gate_proj", "up_proj", "down_proj")
+PROMPTS = [list(range(1, 18)), [9, 8, 7]]
+MAX_TOKENS = 6
+
+
+@pytest.mark.parametrize("method", ["gptq", "awq"])
+def test_pack_dequant_roundtrip(method):
+    torch.manual_seed(0)
+    w = torch.randn(48, 64)  # [N, K]
+    pack = pack_gptq if method == "gptq" else pack_awq
+    deq = dequant_gptq if method == "gptq" else dequant_awq
+    qw, qz, s = pack(w, GROUP)
+    back = deq(qw, qz, s, GROUP, torch.float32)
+    assert back.shape == w.shape
+    # 4-bit asymmetric per-group quant: max error ~ scale/2
+    err = (back - w).abs().max()
+    assert err < 0.2, err
+
+
+def _base_state_dict():
+    g = torch.Generator().manual_seed(77)
+    H, I, V = CFG["hidden_size"], CFG["intermediate_size"], CFG["vocab_size"]
+    hd = H // CFG["num_attention_heads"]
+    kv = CFG["num_key_value_heads"] * hd
+    sd = {}
+
+    def rnd(*shape):
+        return torch.randn(*shape, generator=g) * 0.08
+
+    sd["model.embed_tokens.weight"] = rnd(V, H)
+    for L in range(CFG["num_hidden_layers"]):
+        p = f"model.layers.{L}."
+        sd[p + "self_attn.q_proj.weight"] = rnd(H, H)
+        sd[p + "self_attn.q_proj.bias"] = rnd(H)
+        sd[p + "self_attn.k_proj.weight"] = rnd(kv, H)
+        sd[p + "self_attn.k_proj.bias"] = rnd(kv)
+        sd[p + "self_attn.v_proj.weight"] = rnd(kv, H)
+        sd[p + "self_attn.v_proj.bias"] = rnd(kv)
+        sd[p + "self_attn.o_proj.weight"] = rnd(H, H)
+        sd[p + "mlp.gate_proj.weight"] = rnd(I, H)
+        sd[p + "mlp.up_proj.weight"] = rnd(I, H)
+        sd[p + "mlp.down_proj.weight"] = rnd(H, I)
+        sd[p + "input_layernorm.weight"] = torch.ones(H) + rnd(H) * 0.05
+        sd[p + "post_attention_layernorm.weight"] = \
+            torch.ones(H) + rnd(H) * 0.05
+    sd["model.norm.weight"] = torch.ones(H) + rnd(H) * 0.05
+    sd["lm_head.weight"] = rnd(V, H)
+    return sd
+
+
+def _is_quantized(name):
+    return name.endswith(".weight") and \
+        any(f".{k}." in name for k in QUANT_KEYS)
+
+
+def _write(d, sd, method):
+    os.makedirs(d, exist_ok=True)
+    cfg = dict(CFG)
+    if method:
+        cfg["quantization_config"] = {"quant_method": method, "bits": 4,
+                                      "group_size": GROUP}
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump(cfg, f)
+    from safetensors.torch import save_file
+    save_file(sd, os.path.join(d, "model.safetensors"))
+
+
+def _gen(model_dir):
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    from gllm_amd.sequence import SamplingParams
+    cfg = EngineConfig(model=model_dir, load_format="auto", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       enable_prefix_caching=False)
+    llm = LLM(config=cfg, num_pages_override=128)
+    sp = [SamplingParams(temperature=0.0, max_tokens=MAX_TOKENS,
+                         ignore_eos=True)] * len(PROMPTS)
+    return [o.token_ids for o in llm.generate(PROMPTS, sp)]
+
+
+@pytest.mark.parametrize("method", ["gptq", "awq"])
+def test_int4_checkpoint_equals_dequantized_twin(tmp_path, method):
+    pack = pack_gptq if method == "gptq" else pack_awq
+    deq = dequant_gptq if method == "gptq" else dequant_awq
+    base = _base_state_dict()
+    q_sd, twin_sd = {}, {}
+    for name, w in base.items():
+        if _is_quantized(name):
+            qw, qz, s = pack(w, GROUP)
+            stem = name[:-len(".weight")]
+            q_sd[stem + ".qweight"] = qw
+            q_sd[stem + ".qzeros"] = qz
+            q_sd[stem + ".scales"] = s
+            twin_sd[name] = deq(qw, qz, s, GROUP, torch.float32)
+        else:
+            q_sd[name] = w
+            twin_sd[name] = w
+    dq = str(tmp_path / f"{method}_q")
+    dt = str(tmp_path / f"{method}_t")
+    _write(dq, q_sd, method)
+    _write(dt, twin_sd, None)
+    assert _gen(dq) == _gen(dt)
