@@ -72,45 +72,8 @@ class LLM:
         return seq
 
     def _prepare_mm(self, seq: Sequence, mm_input: dict) -> None:
-        """Offline multimodal admission: run the vision tower, record
-        image spans and MRoPE positions.
-
-        mm_input: {"pixel_values": [L, C*tps*ps*ps], "grids":
-        [(t,h,w), ...]} or {"embeds": [N, hidden], "grids": [...]}.
-        """
-        model = self.runner.model
-        grids = mm_input["grids"]
-        if "embeds" in mm_input:
-            embeds = mm_input["embeds"]
-        else:
-            import torch as _t
-            with _t.no_grad():
-                embeds = model.encode_images(mm_input["pixel_values"],
-                                             grids).cpu()
-        seq.mm_embeds = embeds
-        img_tok = model.image_token_id
-        # spans = maximal runs of the image token
-        spans = []
-        i = 0
-        toks = seq.token_ids
-        while i < len(toks):
-            if toks[i] == img_tok:
-                j = i
-                while j < len(toks) and toks[j] == img_tok:
-                    j += 1
-                spans.append((i, j - i))
-                i = j
-            else:
-                i += 1
-        assert sum(n for _, n in spans) == embeds.shape[0], \
-            (spans, embeds.shape)
-        seq.mm_spans = spans
-        from gllm_amd.layers.mrope import MRotaryEmbedding
-        pos, delta = MRotaryEmbedding.get_input_positions(
-            toks, img_tok, grids,
-            spatial_merge_size=model.spatial_merge_size)
-        seq.mrope_positions = pos
-        seq.mrope_delta = delta
+        from gllm_amd.multimodal.prepare import prepare_mm_seq
+        prepare_mm_seq(self.runner.model, seq, mm_input)
 
     def add_requests(self, seqs: List[Sequence]) -> None:
         self.scheduler.add_seqs(seqs)

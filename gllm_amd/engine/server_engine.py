@@ -65,6 +65,19 @@ class AsyncLLMEngine:
             if config.launch_mode == "master" else None
         self.comm = FrontendComm(config.world_size, local_ranks=local_ranks)
         self._relay = None
+        # multimodal serving (Qwen2-VL family): native image processor
+        vcfg = getattr(self.hf_config, "vision_config", None)
+        self.image_token_id = getattr(self.hf_config, "image_token_id",
+                                      None)
+        if vcfg is not None and self.image_token_id is not None:
+            from gllm_amd.multimodal.processor import ImageProcessor
+            self.mm_processor = ImageProcessor.from_config(vcfg)
+        else:
+            self.mm_processor = None
+        # literal strings the chat template/tokenizer map to the image
+        # tokens (Qwen2-VL conventions)
+        self.image_pad_str = "<|image_pad|>"
+        self.vision_wrap = ("<|vision_start|>", "<|vision_end|>")
         self.seq_ids = IDAllocator(1 << 20)
         self.requests: Dict[int, RequestState] = {}
         self._intake_lock = threading.Lock()
@@ -209,7 +222,8 @@ class AsyncLLMEngine:
 
     # ------------------------------------------------------------------
     def add_request(self, token_ids: List[int],
-                    sampling: SamplingParams) -> RequestState:
+                    sampling: SamplingParams,
+                    mm: Optional[dict] = None) -> RequestState:
         loop = asyncio.get_event_loop()
         with self._intake_lock:
             seq_id = self.seq_ids.allocate()
@@ -222,6 +236,7 @@ class AsyncLLMEngine:
                 "token_ids": token_ids,
                 "sampling": dataclasses.asdict(sampling),
                 "eos_token_id": self.eos_token_id,
+                "mm": mm,
             })
         return st
 
@@ -239,9 +254,10 @@ class AsyncLLMEngine:
 
     # ------------------------------------------------------------------
     async def generate_stream(self, token_ids: List[int],
-                              sampling: SamplingParams
+                              sampling: SamplingParams,
+                              mm: Optional[dict] = None
                               ) -> AsyncIterator[StreamChunk]:
-        st = self.add_request(token_ids, sampling)
+        st = self.add_request(token_ids, sampling, mm=mm)
         try:
             while True:
                 chunk = await st.queue.get()
@@ -256,6 +272,55 @@ class AsyncLLMEngine:
     def encode(self, prompt: str) -> List[int]:
         assert self.tokenizer is not None, "no tokenizer available"
         return self.tokenizer.encode(prompt)
+
+    # ------------------------------------------------- multimodal intake
+    def extract_images(self, messages: list):
+        """OpenAI multi-part content -> (text-only messages with one
+        image-pad sentinel per image, raw image bytes list)."""
+        from gllm_amd.multimodal.processor import decode_image_url
+        images = []
+        out = []
+        vs, ve = self.vision_wrap
+        for m in messages:
+            content = m.get("content")
+            if not isinstance(content, list):
+                out.append(m)
+                continue
+            parts = []
+            for part in content:
+                ptype = part.get("type")
+                if ptype == "text":
+                    parts.append(part["text"])
+                elif ptype == "image_url":
+                    url = part["image_url"]
+                    if isinstance(url, dict):
+                        url = url["url"]
+                    images.append(decode_image_url(url))
+                    parts.append(f"{vs}{self.image_pad_str}{ve}")
+                else:
+                    raise ValueError(f"unsupported content part {ptype}")
+            out.append({**m, "content": "".join(parts)})
+        return out, images
+
+    def process_images(self, token_ids: List[int], images: list):
+        """Preprocess images and expand each single image-pad sentinel
+        to its per-image token count. Returns (token_ids, mm dict)."""
+        if not images:
+            return token_ids, None
+        assert self.mm_processor is not None, \
+            "model has no vision tower; cannot accept images"
+        import torch
+        from gllm_amd.multimodal.processor import expand_image_tokens
+        pixels, grids, counts = [], [], []
+        for raw in images:
+            p, grid = self.mm_processor(raw)
+            pixels.append(p)
+            grids.append(grid)
+            counts.append(self.mm_processor.num_tokens(grid))
+        token_ids = expand_image_tokens(token_ids, self.image_token_id,
+                                        counts, self.image_token_id)
+        mm = {"pixel_values": torch.cat(pixels, dim=0), "grids": grids}
+        return token_ids, mm
 
     def apply_chat_template(self, messages, **kwargs) -> List[int]:
         assert self.tokenizer is not None

@@ -75,6 +75,9 @@ class ServingMixin:
                 seq = Sequence(payload["seq_id"], payload["token_ids"],
                                SamplingParams(**payload["sampling"]),
                                eos_token_id=payload.get("eos_token_id"))
+                if payload.get("mm") is not None:
+                    from gllm_amd.multimodal.prepare import prepare_mm_seq
+                    prepare_mm_seq(self.runner.model, seq, payload["mm"])
                 self._seqs_by_id[seq.seq_id] = seq
                 self.scheduler.add_seqs([seq])
             elif kind == "abort":

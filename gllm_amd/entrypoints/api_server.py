@@ -127,19 +127,21 @@ def build_app():
         if req.tools:
             kwargs["tools"] = [t.model_dump() for t in req.tools]
         try:
+            messages, images = engine.extract_images(messages)
             token_ids = engine.apply_chat_template(messages, **kwargs)
+            token_ids, mm = engine.process_images(token_ids, images)
         except Exception as e:
             return JSONResponse(status_code=400,
                                 content={"error": str(e)})
         sampling = _sampling_from(req)
         if req.stream:
             return StreamingResponse(
-                _chat_stream(req, raw, token_ids, sampling),
+                _chat_stream(req, raw, token_ids, sampling, mm=mm),
                 media_type="text/event-stream")
         n = req.n or 1
         import asyncio as _aio
         results = await _aio.gather(*[
-            _collect(raw, token_ids, _vary_seed(sampling, j))
+            _collect(raw, token_ids, _vary_seed(sampling, j), mm=mm)
             for j in range(n)])
         choices = []
         total_out = 0
@@ -165,12 +167,13 @@ def build_app():
                             total_tokens=len(token_ids) + total_out))
         return resp.model_dump()
 
-    async def _collect(raw, token_ids, sampling):
+    async def _collect(raw, token_ids, sampling, mm=None):
         text_parts = []
         finish = None
         n_out = 0
         plp = None
-        async for chunk in engine.generate_stream(token_ids, sampling):
+        async for chunk in engine.generate_stream(token_ids, sampling,
+                                                  mm=mm):
             if await raw.is_disconnected():
                 break
             text_parts.append(chunk.text)
@@ -182,12 +185,13 @@ def build_app():
         _, text = check_stop_strings(text, sampling.stop)
         return text, finish, n_out, plp
 
-    async def _merged_stream(raw, token_ids, sampling, n):
+    async def _merged_stream(raw, token_ids, sampling, n, mm=None):
         """Run n generations concurrently, yield (choice_idx, chunk) in
         arrival order (n>1 streaming: interleaved choices)."""
         import asyncio as _aio
         if n <= 1:
-            async for chunk in engine.generate_stream(token_ids, sampling):
+            async for chunk in engine.generate_stream(token_ids, sampling,
+                                                      mm=mm):
                 if await raw.is_disconnected():
                     return
                 yield 0, chunk
@@ -196,7 +200,7 @@ def build_app():
 
         async def pump(j):
             async for chunk in engine.generate_stream(
-                    token_ids, _vary_seed(sampling, j)):
+                    token_ids, _vary_seed(sampling, j), mm=mm):
                 await q.put((j, chunk))
             await q.put((j, None))
 
@@ -215,12 +219,13 @@ def build_app():
             for t in tasks:
                 t.cancel()
 
-    async def _chat_stream(req, raw, token_ids, sampling):
+    async def _chat_stream(req, raw, token_ids, sampling, mm=None):
         resp_id = None
         first_for = set()
         n = req.n or 1
         n_out = 0
-        async for j, chunk in _merged_stream(raw, token_ids, sampling, n):
+        async for j, chunk in _merged_stream(raw, token_ids, sampling, n,
+                                             mm=mm):
             delta = DeltaMessage(content=chunk.text)
             if j not in first_for:
                 delta.role = "assistant"
