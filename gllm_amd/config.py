@@ -41,6 +41,10 @@ class EngineConfig:
     worker_ranks: Optional[List[int]] = None
     # control-plane TCP port on the master node; None = master_port + 1
     relay_port: Optional[int] = None
+    # encoder disaggregation (disagg/): remote vision-encoder address
+    # "host:port", or a discovery server to resolve one from
+    mm_encoder_addr: Optional[str] = None
+    discovery_addr: Optional[str] = None
 
     # --- KV cache ---
     page_size: int = 16

@@ -1,0 +1,171 @@
+"""Vision-encoder server: a standalone process hosting ONLY the ViT.
+
+Parity target: reference disagg/encoder_runtime.py + encoder_engine.py
+(vision-only process, content-hash dedup cache) and
+entrypoints/encoder_server.py. The LM frontend sends EncoderJobs; this
+server runs the tower and returns merged embeddings. A hash-first
+probe protocol avoids resending pixels for cached content:
+
+  client:  EncoderJob(pixel_values=None)   # probe by content hash
+  server:  EncoderResult(embeds=..., cached=True)   on hit
+           EncoderResult(embeds=None)               on miss
+  client:  EncoderJob(pixel_values=...)    # full job
+  server:  EncoderResult(embeds=...)
+
+Round-2 data plane: hipIpc slot-pool WRITEs over xGMI (ROADMAP.md).
+"""
+
+import collections
+import socket
+import threading
+from typing import Optional
+
+import torch
+
+from gllm_amd.disagg.protocol import EncoderJob, EncoderResult
+from gllm_amd.engine.multinode import recv_msg, send_msg
+from gllm_amd.logger import logger
+
+
+class EmbeddingCache:
+    """Content-hash LRU (reference model_runner.py:161-221)."""
+
+    def __init__(self, max_items: int = 256):
+        self.max_items = max_items
+        self._d: "collections.OrderedDict[str, torch.Tensor]" = \
+            collections.OrderedDict()
+        self.hits = 0
+        self.misses = 0
+
+    def get(self, key: str) -> Optional[torch.Tensor]:
+        if key in self._d:
+            self._d.move_to_end(key)
+            self.hits += 1
+            return self._d[key]
+        self.misses += 1
+        return None
+
+    def put(self, key: str, value: torch.Tensor) -> None:
+        self._d[key] = value
+        self._d.move_to_end(key)
+        while len(self._d) > self.max_items:
+            self._d.popitem(last=False)
+
+
+class EncoderServer:
+    def __init__(self, config, host: str = "0.0.0.0", port: int = 29820,
+                 cache_items: int = 256):
+        from gllm_amd.models.loader import load_model
+        from gllm_amd.parallel import init_distributed
+        init_distributed(pp_size=1, dp_size=1, tp_size=1)
+        # full-model construction keeps dummy-init weight names (and so
+        # the crc32 weights) identical to the LM side; only .visual runs
+        self.model, _ = load_model(config, config.device)
+        assert getattr(self.model, "visual", None) is not None, \
+            "model has no vision tower"
+        self.cache = EmbeddingCache(cache_items)
+        self.port = port
+        self._srv = socket.create_server((host, port))
+        self._srv.settimeout(0.5)
+        self._stop = threading.Event()
+
+    def serve_forever(self) -> None:
+        logger.info("encoder server listening on %d", self.port)
+        while not self._stop.is_set():
+            try:
+                conn, _ = self._srv.accept()
+            except socket.timeout:
+                continue
+            except OSError:
+                return
+            threading.Thread(target=self._serve_conn, args=(conn,),
+                             daemon=True).start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        self._srv.close()
+
+    def _serve_conn(self, conn: socket.socket) -> None:
+        try:
+            while not self._stop.is_set():
+                job = recv_msg(conn)
+                if job is None:
+                    return
+                if job == "stats":
+                    send_msg(conn, {"hits": self.cache.hits,
+                                    "misses": self.cache.misses})
+                    continue
+                if job == "shutdown":
+                    send_msg(conn, {"ok": True})
+                    self.stop()
+                    return
+                send_msg(conn, self._run(job))
+        except OSError:
+            pass
+        finally:
+            conn.close()
+
+    def _run(self, job: EncoderJob) -> EncoderResult:
+        emb = self.cache.get(job.content_hash)
+        if emb is not None:
+            return EncoderResult(job.job_id, emb, cached=True)
+        if job.pixel_values is None:
+            return EncoderResult(job.job_id, None)  # probe miss
+        try:
+            with torch.no_grad():
+                emb = self.model.encode_images(job.pixel_values,
+                                               job.grids).cpu()
+        except Exception as e:  # pragma: no cover
+            logger.exception("encoder job %d failed", job.job_id)
+            return EncoderResult(job.job_id, None, error=str(e))
+        self.cache.put(job.content_hash, emb)
+        return EncoderResult(job.job_id, emb)
+
+
+class EncoderClient:
+    """LM-frontend side: probe-by-hash then send pixels on miss."""
+
+    def __init__(self, addr: str):
+        host, port = addr.rsplit(":", 1)
+        self._sock = socket.create_connection((host, int(port)),
+                                              timeout=60)
+        self._lock = threading.Lock()
+        self._job_id = 0
+
+    def encode(self, pixel_values: torch.Tensor, grids) -> torch.Tensor:
+        from gllm_amd.disagg.protocol import content_hash
+        key = content_hash(pixel_values, grids)
+        with self._lock:
+            self._job_id += 1
+            jid = self._job_id
+            send_msg(self._sock, EncoderJob(jid, key, list(grids)))
+            res: EncoderResult = recv_msg(self._sock)
+            if res.embeds is None and res.error is None:
+                send_msg(self._sock,
+                         EncoderJob(jid, key, list(grids), pixel_values))
+                res = recv_msg(self._sock)
+        if res.error:
+            raise RuntimeError(f"encoder job failed: {res.error}")
+        return res.embeds
+
+    def stats(self) -> dict:
+        with self._lock:
+            send_msg(self._sock, "stats")
+            return recv_msg(self._sock)
+
+    def close(self) -> None:
+        self._sock.close()
+
+
+def run_encoder_server(config, host: str = "0.0.0.0", port: int = 29820,
+                       discovery_addr: Optional[str] = None) -> None:
+    """Entry point for ``python -m gllm_amd.entrypoints.encoder_server``."""
+    srv = EncoderServer(config, host, port)
+    if discovery_addr:
+        from gllm_amd.disagg.discovery import (DiscoveryClient,
+                                               start_heartbeat)
+        dc = DiscoveryClient(discovery_addr)
+        my_addr = f"{socket.gethostname()}:{port}"
+        dc.register("encoder", config.model, my_addr)
+        start_heartbeat(dc, "encoder", config.model, my_addr)
+    srv.serve_forever()

@@ -74,6 +74,21 @@ class AsyncLLMEngine:
             self.mm_processor = ImageProcessor.from_config(vcfg)
         else:
             self.mm_processor = None
+        # encoder disaggregation: vision runs in a remote encoder
+        # process; workers receive ready embeddings (disagg/)
+        self.encoder_client = None
+        addr = config.mm_encoder_addr
+        if addr is None and config.discovery_addr:
+            from gllm_amd.disagg.discovery import DiscoveryClient
+            dc = DiscoveryClient(config.discovery_addr)
+            addrs = dc.lookup("encoder", config.model)
+            dc.close()
+            if addrs:
+                addr = addrs[0]
+                logger.info("discovered encoder at %s", addr)
+        if addr:
+            from gllm_amd.disagg.encoder_server import EncoderClient
+            self.encoder_client = EncoderClient(addr)
         # literal strings the chat template/tokenizer map to the image
         # tokens (Qwen2-VL conventions)
         self.image_pad_str = "<|image_pad|>"
@@ -319,8 +334,13 @@ class AsyncLLMEngine:
             counts.append(self.mm_processor.num_tokens(grid))
         token_ids = expand_image_tokens(token_ids, self.image_token_id,
                                         counts, self.image_token_id)
-        mm = {"pixel_values": torch.cat(pixels, dim=0), "grids": grids}
-        return token_ids, mm
+        pixel_values = torch.cat(pixels, dim=0)
+        if self.encoder_client is not None:
+            # disaggregated: the remote encoder runs the tower; workers
+            # get ready embeddings and skip their local tower
+            embeds = self.encoder_client.encode(pixel_values, grids)
+            return token_ids, {"embeds": embeds, "grids": grids}
+        return token_ids, {"pixel_values": pixel_values, "grids": grids}
 
     def apply_chat_template(self, messages, **kwargs) -> List[int]:
         assert self.tokenizer is not None

@@ -129,7 +129,10 @@ def build_app():
         try:
             messages, images = engine.extract_images(messages)
             token_ids = engine.apply_chat_template(messages, **kwargs)
-            token_ids, mm = engine.process_images(token_ids, images)
+            # preprocessing + (disagg) remote encode off the event loop
+            import asyncio as _aio
+            token_ids, mm = await _aio.to_thread(
+                engine.process_images, token_ids, images)
         except Exception as e:
             return JSONResponse(status_code=400,
                                 content={"error": str(e)})

@@ -1,0 +1,125 @@
+"""Encoder-disaggregation tests (CPU): discovery TTL registry, encoder
+server with content-hash dedup, and the full disagg serving path — the
+disaggregated completion must EXACTLY equal the in-process-tower one
+(both sides dummy-init the same crc32-seeded weights)."""
+
+import json
+import multiprocessing as mp
+import socket
+import time
+
+import pytest
+import torch
+
+from tests.test_mm_serving_cpu import VL_CFG, _img_b64, _mk_vl_dir
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def test_discovery_register_lookup_ttl():
+    from gllm_amd.disagg.discovery import DiscoveryClient, DiscoveryServer
+    port = _free_port()
+    srv = DiscoveryServer("127.0.0.1", port, ttl_s=0.5).start()
+    try:
+        c = DiscoveryClient(f"127.0.0.1:{port}")
+        assert c.lookup("encoder", "m") == []
+        c.register("encoder", "m", "hostA:1")
+        c.register("encoder", "m2", "hostB:2")
+        assert c.lookup("encoder", "m") == ["hostA:1"]
+        time.sleep(0.8)  # expire
+        assert c.lookup("encoder", "m") == []
+        c.register("encoder", "m", "hostA:1")
+        c.deregister("encoder", "m", "hostA:1")
+        assert c.lookup("encoder", "m") == []
+        c.close()
+    finally:
+        srv.stop()
+
+
+def _run_encoder(model_dir, port):
+    torch.set_num_threads(1)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.disagg.encoder_server import run_encoder_server
+    cfg = EngineConfig(model=model_dir, load_format="dummy", device="cpu",
+                       dtype="float32")
+    run_encoder_server(cfg, "127.0.0.1", port)
+
+
+@pytest.mark.timeout(300)
+def test_disagg_matches_local_tower(tmp_path):
+    from fastapi.testclient import TestClient
+    import gllm_amd.entrypoints.api_server as srv_mod
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+
+    d = _mk_vl_dir(tmp_path)
+    enc_port = _free_port()
+    ctx = mp.get_context("spawn")
+    enc = ctx.Process(target=_run_encoder, args=(d, enc_port))
+    enc.start()
+
+    def ask(engine, seed):
+        srv_mod.engine = engine
+        srv_mod.served_model = "vl"
+        app = srv_mod.build_app()
+        client = TestClient(app)
+        r = client.post("/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": [
+                {"type": "text", "text": "w5 w6 "},
+                {"type": "image_url", "image_url": {"url": _img_b64(seed)}},
+            ]}],
+            "max_tokens": 5, "temperature": 0.0, "ignore_eos": True})
+        assert r.status_code == 200, r.text
+        return r.json()["choices"][0]["message"]["content"]
+
+    # local-tower reference
+    cfg_local = EngineConfig(model=d, load_format="dummy", device="cpu",
+                             dtype="float32", page_size=4, maxp=64,
+                             master_port=29696,
+                             enable_prefix_caching=False)
+    eng = AsyncLLMEngine(cfg_local)
+    eng.vision_wrap = ("<|vision_start|> ", " <|vision_end|>")
+    eng.start()
+    try:
+        ref = ask(eng, 7)
+    finally:
+        eng.stop()
+
+    # disaggregated: remote encoder, workers get ready embeddings
+    try:
+        cfg_dis = EngineConfig(model=d, load_format="dummy", device="cpu",
+                               dtype="float32", page_size=4, maxp=64,
+                               master_port=29697,
+                               mm_encoder_addr=f"127.0.0.1:{enc_port}",
+                               enable_prefix_caching=False)
+        # wait for the encoder to come up
+        deadline = time.time() + 120
+        while True:
+            try:
+                socket.create_connection(("127.0.0.1", enc_port),
+                                         timeout=1).close()
+                break
+            except OSError:
+                assert time.time() < deadline, "encoder never came up"
+                time.sleep(0.5)
+        eng2 = AsyncLLMEngine(cfg_dis)
+        eng2.vision_wrap = ("<|vision_start|> ", " <|vision_end|>")
+        eng2.start()
+        try:
+            out = ask(eng2, 7)
+            assert out == ref, "disagg path must match local tower"
+            out2 = ask(eng2, 7)  # same image -> encoder cache hit
+            assert out2 == ref
+            stats = eng2.encoder_client.stats()
+            assert stats["hits"] >= 1, stats
+        finally:
+            eng2.stop()
+    finally:
+        enc.terminate()
+        enc.join(timeout=30)
