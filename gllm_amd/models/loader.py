@@ -55,14 +55,60 @@ def load_hf_config(model_path: str):
     if os.path.exists(cfg_file):
         try:
             from transformers import AutoConfig
-            return AutoConfig.from_pretrained(model_path,
-                                              trust_remote_code=True)
+            cfg = AutoConfig.from_pretrained(model_path,
+                                             trust_remote_code=True)
+            return _normalize_rope(_flatten_text_config(cfg))
         except Exception:
             import types
             with open(cfg_file) as f:
                 d = json.load(f)
             return types.SimpleNamespace(**d)
     raise FileNotFoundError(f"no config.json under {model_path}")
+
+
+def _flatten_text_config(cfg):
+    """Newer transformers nest LM fields under ``text_config`` for
+    multimodal configs (e.g. Qwen2_5_VLConfig). The model classes here
+    read flat attributes, so merge text_config up when the top level
+    lacks them."""
+    txt = getattr(cfg, "text_config", None)
+    if txt is None:
+        return cfg
+    try:
+        if getattr(cfg, "num_hidden_layers", None) is not None:
+            return cfg
+    except Exception:
+        pass
+    import types
+    d = dict(txt.to_dict() if hasattr(txt, "to_dict") else vars(txt))
+    try:
+        top = cfg.to_dict()
+    except Exception:
+        top = dict(vars(cfg))
+    for k, v in top.items():
+        if k != "text_config" and k not in d:
+            d[k] = v
+    # top-level identity fields win over text_config leftovers
+    for k in ("architectures", "model_type"):
+        if top.get(k):
+            d[k] = top[k]
+    return types.SimpleNamespace(**d)
+
+
+def _normalize_rope(cfg):
+    """transformers >= 5 renamed rope_scaling -> rope_parameters (with
+    rope_theta folded in); the model classes read the classic names."""
+    if getattr(cfg, "rope_scaling", None) is None:
+        rp = getattr(cfg, "rope_parameters", None)
+        if isinstance(rp, dict) and rp:
+            try:
+                cfg.rope_scaling = dict(rp)
+                if rp.get("rope_theta") is not None and \
+                        getattr(cfg, "rope_theta", None) is None:
+                    cfg.rope_theta = rp["rope_theta"]
+            except Exception:  # frozen config object
+                pass
+    return cfg
 
 
 def iterate_safetensors(model_path: str

@@ -32,8 +32,16 @@ class Qwen2VLForCausalLM(Qwen2ForCausalLM):
         self.spatial_merge_size = getattr(vcfg, "spatial_merge_size", 2) \
             if vcfg is not None else 2
         if self.is_first_stage and vcfg is not None:
-            self.visual = Qwen2VisionTransformer(
-                vcfg, dtype=engine_config.torch_dtype())
+            # Qwen2.5-VL configs carry window attention fields; the
+            # plain Qwen2-VL tower is full-attention LayerNorm/QuickGELU
+            if getattr(vcfg, "window_size", None) is not None:
+                from gllm_amd.models.qwen2_vl_vision import \
+                    Qwen25VisionTransformer
+                self.visual = Qwen25VisionTransformer(
+                    vcfg, dtype=engine_config.torch_dtype())
+            else:
+                self.visual = Qwen2VisionTransformer(
+                    vcfg, dtype=engine_config.torch_dtype())
         else:
             self.visual = None
 

@@ -189,3 +189,140 @@ class Qwen2VisionTransformer(nn.Module):
         for blk in self.blocks:
             x = blk(x, cu, rotary)
         return self.merger(x)
+
+
+# ---------------------------------------------------------------- 2.5-VL
+class Vision25MLP(nn.Module):
+    """SiLU-gated MLP (Qwen2.5-VL blocks; HF names gate/up/down_proj)."""
+
+    def __init__(self, dim, hidden, dtype=None):
+        super().__init__()
+        self.gate_proj = nn.Linear(dim, hidden, bias=True, dtype=dtype)
+        self.up_proj = nn.Linear(dim, hidden, bias=True, dtype=dtype)
+        self.down_proj = nn.Linear(hidden, dim, bias=True, dtype=dtype)
+
+    def forward(self, x):
+        return self.down_proj(F.silu(self.gate_proj(x)) * self.up_proj(x))
+
+
+class Vision25Block(nn.Module):
+    def __init__(self, dim, num_heads, intermediate, dtype=None):
+        super().__init__()
+        from gllm_amd.layers.layernorm import RMSNorm
+        self.norm1 = RMSNorm(dim, 1e-6)
+        self.norm2 = RMSNorm(dim, 1e-6)
+        self.attn = VisionAttention(dim, num_heads, dtype=dtype)
+        self.mlp = Vision25MLP(dim, intermediate, dtype=dtype)
+
+    def forward(self, x, cu_seqlens, rotary):
+        x = x + self.attn(self.norm1(x), cu_seqlens, rotary)
+        x = x + self.mlp(self.norm2(x))
+        return x
+
+
+class PatchMerger25(nn.Module):
+    """RMSNorm ln_q + 2-layer MLP to out_hidden_size (HF names
+    merger.ln_q, merger.mlp.{0,2})."""
+
+    def __init__(self, d_model, context_dim, merge_size, dtype=None):
+        super().__init__()
+        from gllm_amd.layers.layernorm import RMSNorm
+        self.hidden_size = context_dim * (merge_size ** 2)
+        self.ln_q = RMSNorm(context_dim, 1e-6)
+        self.mlp = nn.Sequential(
+            nn.Linear(self.hidden_size, self.hidden_size, dtype=dtype),
+            nn.GELU(),
+            nn.Linear(self.hidden_size, d_model, dtype=dtype))
+
+    def forward(self, x):
+        return self.mlp(self.ln_q(x).view(-1, self.hidden_size))
+
+
+def window_index_thw(t, h, w, merge_size, window_patches):
+    """Qwen2.5-VL window partition (reference qwen2_5_vl.py:537-572):
+    permutation of merge-unit indices grouping them into
+    window_patches x window_patches windows, plus the per-window
+    cumulative PATCH counts (x merge_unit)."""
+    lh, lw = h // merge_size, w // merge_size
+    idx = torch.arange(t * lh * lw).reshape(t, lh, lw)
+    pad_h = (-lh) % window_patches
+    pad_w = (-lw) % window_patches
+    nh = (lh + pad_h) // window_patches
+    nw = (lw + pad_w) // window_patches
+    padded = F.pad(idx, (0, pad_w, 0, pad_h), value=-100)
+    padded = padded.reshape(t, nh, window_patches, nw, window_patches)
+    padded = padded.permute(0, 1, 3, 2, 4).reshape(
+        t, nh * nw, window_patches, window_patches)
+    seqlens = (padded != -100).sum([2, 3]).reshape(-1)
+    flat = padded.reshape(-1)
+    index = flat[flat != -100]
+    cu = seqlens.cumsum(0) * (merge_size ** 2)
+    cu = torch.unique_consecutive(cu.to(torch.int32))
+    return index, cu
+
+
+class Qwen25VisionTransformer(nn.Module):
+    """visual.* of Qwen2.5-VL checkpoints: RMSNorm blocks, SiLU-gated
+    MLP, and WINDOWED attention — all blocks attend within
+    window_size x window_size pixel windows except
+    ``fullatt_block_indexes``, which see the whole image. Rows are
+    permuted into window order (in spatial-merge units) once, attended
+    with per-window cu_seqlens, and un-permuted after the merger
+    (reference qwen2_5_vl.py:440-686)."""
+
+    def __init__(self, vcfg, dtype=None):
+        super().__init__()
+        g = lambda k, d=None: (vcfg.get(k, d) if isinstance(vcfg, dict)
+                               else getattr(vcfg, k, d))
+        self.spatial_merge_size = g("spatial_merge_size", 2)
+        self.patch_size = g("patch_size", 14)
+        self.window_size = g("window_size", 112)
+        self.fullatt_block_indexes = list(g("fullatt_block_indexes",
+                                            []) or [])
+        dim = g("hidden_size")
+        self.patch_embed = PatchEmbed(
+            self.patch_size, g("temporal_patch_size", 2),
+            g("in_channels", 3), dim, dtype=dtype)
+        num_heads = g("num_heads", 16)
+        self.head_dim = dim // num_heads
+        self.rotary = VisionRotaryEmbedding(self.head_dim // 2)
+        self.blocks = nn.ModuleList([
+            Vision25Block(dim, num_heads, g("intermediate_size", dim * 4),
+                          dtype=dtype)
+            for _ in range(g("depth", 32))])
+        self.merger = PatchMerger25(g("out_hidden_size", dim), dim,
+                                    self.spatial_merge_size, dtype=dtype)
+
+    def forward(self, pixel_values: torch.Tensor,
+                grid_thw: List[Tuple[int, int, int]]) -> torch.Tensor:
+        m = self.spatial_merge_size
+        unit = m * m
+        wp = self.window_size // m // self.patch_size
+        x = self.patch_embed(pixel_values)
+        rotary = rot_pos_emb(grid_thw, self.head_dim, m, self.rotary)
+
+        # per-image window permutation over merge units
+        win_index = []
+        cu_window = [torch.zeros(1, dtype=torch.int32)]
+        cu_full = [0]
+        base = 0
+        for t, h, w in grid_thw:
+            idx, cu = window_index_thw(t, h, w, m, wp)
+            win_index.append(idx + base)
+            base += t * (h // m) * (w // m)
+            cu_window.append(cu + int(cu_window[-1][-1]))
+            cu_full.append(cu_full[-1] + t * h * w)
+        win_index = torch.cat(win_index)
+        cu_window = torch.unique_consecutive(torch.cat(cu_window)).tolist()
+
+        L = x.shape[0]
+        x = x.reshape(L // unit, unit, -1)[win_index].reshape(L, -1)
+        rotary = rotary.reshape(L // unit, unit, -1)[win_index] \
+            .reshape(L, -1)
+
+        for i, blk in enumerate(self.blocks):
+            cu = cu_full if i in self.fullatt_block_indexes else cu_window
+            x = blk(x, cu, rotary)
+        out = self.merger(x)
+        reverse = torch.argsort(win_index)
+        return out[reverse]

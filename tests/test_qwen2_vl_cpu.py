@@ -95,3 +95,39 @@ def test_vl_text_only_still_works(tmp_path):
     sp = [SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)]
     out = llm.generate([[1, 2, 3, 4, 5]], sp)[0].token_ids
     assert len(out) == 4
+
+
+VL25_TINY = {
+    **VL_TINY,
+    "architectures": ["Qwen2_5_VLForConditionalGeneration"],
+    "model_type": "qwen2_5_vl",
+    "vision_config": {
+        "depth": 2, "hidden_size": 32, "out_hidden_size": 64,
+        "num_heads": 4, "intermediate_size": 48, "patch_size": 14,
+        "temporal_patch_size": 2, "in_channels": 3,
+        "spatial_merge_size": 2, "window_size": 56,
+        "fullatt_block_indexes": [1],
+    },
+}
+
+
+def test_vl25_generate_with_windowed_tower(tmp_path):
+    from gllm_amd.sequence import SamplingParams
+    d = tmp_path / "vl25"
+    d.mkdir()
+    with open(d / "config.json", "w") as f:
+        json.dump(VL25_TINY, f)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    cfg = EngineConfig(model=str(d), load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64)
+    llm = LLM(config=cfg, num_pages_override=128)
+    from gllm_amd.models.qwen2_vl_vision import Qwen25VisionTransformer
+    assert isinstance(llm.runner.model.visual, Qwen25VisionTransformer)
+    toks = [1, 2] + [150] * 16 + [3]
+    mm = {"pixel_values": torch.randn(64, 3 * 2 * 14 * 14),
+          "grids": [(1, 8, 8)]}
+    sp = [SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)]
+    o1 = llm.generate([toks], sp, mm_inputs=[mm])[0].token_ids
+    o2 = llm.generate([toks], sp, mm_inputs=[mm])[0].token_ids
+    assert len(o1) == 5 and o1 == o2

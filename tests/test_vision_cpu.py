@@ -54,3 +54,38 @@ def test_vision_checkpoint_names_match_hf_layout():
         "merger.ln_q.weight", "merger.mlp.0.weight", "merger.mlp.2.weight",
     ]:
         assert expect in names, expect
+
+
+def test_qwen25_tower_windowed_attention():
+    """Qwen2.5-VL tower: window partition correctness + the full-attn
+    blocks actually change the result vs all-window."""
+    import types
+    import torch
+    from gllm_amd.models.qwen2_vl_vision import (Qwen25VisionTransformer,
+                                                 window_index_thw)
+    # window math: 1x8x8 patches, merge 2, window 2 merge-units
+    idx, cu = window_index_thw(1, 8, 8, 2, 2)
+    assert sorted(idx.tolist()) == list(range(16))
+    assert cu.tolist() == [16, 32, 48, 64]  # 4 windows x 4 units x 4
+
+    vcfg = dict(depth=2, hidden_size=32, out_hidden_size=64, num_heads=4,
+                intermediate_size=48, patch_size=14,
+                temporal_patch_size=2, in_channels=3,
+                spatial_merge_size=2, window_size=56,
+                fullatt_block_indexes=[1])
+    torch.manual_seed(0)
+    tower = Qwen25VisionTransformer(types.SimpleNamespace(**vcfg))
+    px = torch.randn(64, 3 * 2 * 14 * 14)
+    out = tower(px, [(1, 8, 8)])
+    assert out.shape == (16, 64)
+
+    # identical weights, but every block windowed -> different output
+    vcfg2 = dict(vcfg, fullatt_block_indexes=[])
+    torch.manual_seed(0)
+    tower2 = Qwen25VisionTransformer(types.SimpleNamespace(**vcfg2))
+    out2 = tower2(px, [(1, 8, 8)])
+    assert not torch.allclose(out, out2, atol=1e-5), \
+        "full-attention blocks must see beyond their window"
+
+    # permutation sanity: un-permuted output must be deterministic
+    assert torch.allclose(out, tower(px, [(1, 8, 8)]), atol=1e-6)
