@@ -19,10 +19,16 @@ from gllm_amd.layers.rotary import RotaryEmbedding
 
 class MRotaryEmbedding(RotaryEmbedding):
     def __init__(self, head_dim: int, rot_dim: int, max_position: int,
-                 base: float, mrope_section: List[int]):
+                 base: float, mrope_section: List[int],
+                 mrope_interleaved: bool = False):
         assert sum(mrope_section) * 2 == rot_dim, \
             (mrope_section, rot_dim)
         self.mrope_section = mrope_section
+        # Qwen3-VL / Qwen3.5 layout: frequency pair k reads stream
+        # T/H/W by k % 3 (within each section's reach) instead of
+        # contiguous [T | H | W] bands (reference rotary_embedding.py
+        # apply_interleaved_rope, :588-605)
+        self.mrope_interleaved = mrope_interleaved
         super().__init__(head_dim, rot_dim, max_position, base,
                          is_neox=True)
 
@@ -41,15 +47,25 @@ class MRotaryEmbedding(RotaryEmbedding):
         half = self.rot_dim // 2
         cos_full = cs[:, :half]
         sin_full = cs[:, half:]
-        cos_parts, sin_parts = [], []
-        off = 0
-        for i, n in enumerate(self.mrope_section):
-            idx = positions[i]
-            cos_parts.append(cos_full[idx][:, off:off + n])
-            sin_parts.append(sin_full[idx][:, off:off + n])
-            off += n
-        cos = torch.cat(cos_parts, dim=-1).float()   # [T, half]
-        sin = torch.cat(sin_parts, dim=-1).float()
+        if self.mrope_interleaved:
+            sec = self.mrope_section
+            cos = cos_full[positions[0]].clone()     # [T, half]
+            sin = sin_full[positions[0]].clone()
+            for axis in (1, 2):
+                sl = slice(axis, sec[axis] * 3, 3)
+                cos[:, sl] = cos_full[positions[axis]][:, sl]
+                sin[:, sl] = sin_full[positions[axis]][:, sl]
+            cos, sin = cos.float(), sin.float()
+        else:
+            cos_parts, sin_parts = [], []
+            off = 0
+            for i, n in enumerate(self.mrope_section):
+                idx = positions[i]
+                cos_parts.append(cos_full[idx][:, off:off + n])
+                sin_parts.append(sin_full[idx][:, off:off + n])
+                off += n
+            cos = torch.cat(cos_parts, dim=-1).float()   # [T, half]
+            sin = torch.cat(sin_parts, dim=-1).float()
         for t in (q, k):
             T = t.shape[0]
             x = t.unflatten(-1, (-1, self.head_dim))

@@ -181,8 +181,11 @@ def get_rope(head_dim: int, rot_dim: int, max_position: int, base: float,
         elif rtype in ("mrope", "default") and \
                 rope_scaling.get("mrope_section"):
             from gllm_amd.layers.mrope import MRotaryEmbedding
-            rope = MRotaryEmbedding(head_dim, rot_dim, max_position, base,
-                                    list(rope_scaling["mrope_section"]))
+            rope = MRotaryEmbedding(
+                head_dim, rot_dim, max_position, base,
+                list(rope_scaling["mrope_section"]),
+                mrope_interleaved=bool(
+                    rope_scaling.get("mrope_interleaved", False)))
         elif rtype == "default":
             rope = RotaryEmbedding(head_dim, rot_dim, max_position, base,
                                    is_neox)

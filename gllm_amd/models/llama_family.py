@@ -174,9 +174,20 @@ class LlamaFamilyForCausalLM(nn.Module):
                     0, fctx.mm_rows,
                     fctx.mm_embeds.to(hidden_states.dtype))
             residual = None
-        for layer in self.layers:
+        ds = getattr(fctx, "mm_deepstack", None)
+        for li, layer in enumerate(self.layers):
             hidden_states, residual = layer(positions, hidden_states,
                                             residual, fctx)
+            if ds is not None:
+                # Qwen3-VL deepstack: level li of the multiscale vision
+                # features is ADDED at the image rows after decoder
+                # layer li (reference qwen3_vl.py Qwen3LLMModel.forward)
+                g = self.layer_start + li
+                H = hidden_states.shape[-1]
+                if g * H < ds.shape[1]:
+                    hidden_states = hidden_states.index_add(
+                        0, fctx.mm_rows,
+                        ds[:, g * H:(g + 1) * H].to(hidden_states.dtype))
         if self.is_last_stage:
             hidden_states, _ = self.norm(hidden_states, residual)
             return hidden_states, None

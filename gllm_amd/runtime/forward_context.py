@@ -40,6 +40,9 @@ class ForwardContext:
     # vision tower (replaced after embed_tokens on the first stage)
     mm_rows: Optional[torch.Tensor] = None     # [N] long
     mm_embeds: Optional[torch.Tensor] = None   # [N, hidden]
+    # Qwen3-VL deepstack: [N, D*hidden] multiscale features added at the
+    # image rows after decoder layers 0..D-1
+    mm_deepstack: Optional[torch.Tensor] = None
     # DSA (DeepSeek-V3.2): per-layer paged index-K caches
     # [pages, page_size, index_head_dim] (models/deepseek_v32.py)
     idx_caches: Optional[List[torch.Tensor]] = None

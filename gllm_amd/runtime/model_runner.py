@@ -293,7 +293,14 @@ class ModelRunner:
         if rows:
             fctx.mm_rows = torch.tensor(rows, dtype=torch.long,
                                         device=self.device)
-            fctx.mm_embeds = torch.cat(embeds).to(self.device)
+            emb = torch.cat(embeds).to(self.device)
+            H = self.hf_config.hidden_size
+            if emb.shape[1] > H:
+                # Qwen3-VL deepstack: [hidden | D*hidden] multiscale
+                fctx.mm_embeds = emb[:, :H]
+                fctx.mm_deepstack = emb[:, H:]
+            else:
+                fctx.mm_embeds = emb
 
     def resolve_tokens(self, tokens: torch.Tensor) -> torch.Tensor:
         """Replace negative placeholder ids with sampled tokens from the
