@@ -264,6 +264,12 @@ class DeepseekV2ForCausalLM(nn.Module):
                 residual=None):
         if self.is_first_stage:
             hidden_states = self.embed_tokens(input_ids)
+            if fctx.mm_rows is not None:
+                # multimodal merge (Kimi-K2.5 rides the DeepSeek-V3
+                # backbone; models/kimi_k25.py)
+                hidden_states = hidden_states.index_copy(
+                    0, fctx.mm_rows,
+                    fctx.mm_embeds.to(hidden_states.dtype))
             residual = None
         for layer in self.layers:
             hidden_states, residual = layer(positions, hidden_states,

@@ -27,6 +27,7 @@ def get_arch_registry():
     from gllm_amd.models.deepseek_v2 import (DeepseekV2ForCausalLM,
                                              DeepseekV3ForCausalLM)
     from gllm_amd.models.deepseek_v32 import DeepseekV32ForCausalLM
+    from gllm_amd.models.kimi_k25 import KimiK25ForCausalLM
     from gllm_amd.models.hybrid_gdn import Qwen3_5ForCausalLM
     from gllm_amd.models.qwen2_vl import Qwen2VLForCausalLM
     from gllm_amd.models.qwen3_vl import (Qwen3VLForCausalLM,
@@ -42,6 +43,7 @@ def get_arch_registry():
         "DeepseekV2ForCausalLM": DeepseekV2ForCausalLM,
         "DeepseekV3ForCausalLM": DeepseekV3ForCausalLM,
         "DeepseekV32ForCausalLM": DeepseekV32ForCausalLM,
+        "KimiK25ForConditionalGeneration": KimiK25ForCausalLM,
         "ChatGLMModel": ChatGLMForCausalLM,
         "ChatGLMForConditionalGeneration": ChatGLMForCausalLM,
         "MistralForCausalLM": MistralForCausalLM,

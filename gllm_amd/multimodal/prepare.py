@@ -21,9 +21,11 @@ def prepare_mm_seq(model, seq: Sequence, mm_input: dict) -> None:
     need spans + mrope for batch building.
     """
     grids = mm_input["grids"]
+    has_tower = getattr(model, "visual", None) is not None or \
+        getattr(model, "vision_tower", None) is not None
     if "embeds" in mm_input:
         seq.mm_embeds = mm_input["embeds"]
-    elif getattr(model, "visual", None) is not None:
+    elif has_tower:
         with torch.no_grad():
             seq.mm_embeds = model.encode_images(
                 mm_input["pixel_values"], grids).cpu()
@@ -44,9 +46,10 @@ def prepare_mm_seq(model, seq: Sequence, mm_input: dict) -> None:
         assert sum(n for _, n in spans) == seq.mm_embeds.shape[0], \
             (spans, seq.mm_embeds.shape)
     seq.mm_spans = spans
-    from gllm_amd.layers.mrope import MRotaryEmbedding
-    pos, delta = MRotaryEmbedding.get_input_positions(
-        toks, img_tok, grids,
-        spatial_merge_size=model.spatial_merge_size)
-    seq.mrope_positions = pos
-    seq.mrope_delta = delta
+    if getattr(model, "uses_mrope", False):
+        from gllm_amd.layers.mrope import MRotaryEmbedding
+        pos, delta = MRotaryEmbedding.get_input_positions(
+            toks, img_tok, grids,
+            spatial_merge_size=model.spatial_merge_size)
+        seq.mrope_positions = pos
+        seq.mrope_delta = delta

@@ -56,6 +56,12 @@ class ModelRunner:
             # [3,B] graph position buffers + mm content hashing: round 2
             cfg.use_graph = False
             cfg.enable_prefix_caching = False
+        elif getattr(self.model, "visual", None) is not None or \
+                getattr(self.model, "vision_tower", None) is not None:
+            # 1-D-position multimodal (Kimi-K2.5): page reuse keyed on
+            # token ids alone would alias different images' pad runs
+            # (mm content hashing is round 2)
+            cfg.enable_prefix_caching = False
         self.index_head_dim = getattr(self.model, "index_head_dim", None)
         if self.index_head_dim:
             # DSA selector round-1 runs the eager per-seq torch path;
