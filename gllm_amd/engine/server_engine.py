@@ -344,6 +344,18 @@ class AsyncLLMEngine:
 
     def apply_chat_template(self, messages, **kwargs) -> List[int]:
         assert self.tokenizer is not None
+        # DeepSeek-V3.2 ships its own DSML encoder instead of a Jinja
+        # template (tokenizers/deepseek_v32.py)
+        from gllm_amd.tokenizers.deepseek_v32 import (
+            apply_dsv32_chat_template, load_dsv32_encoder)
+        enc = load_dsv32_encoder(self.config.model)
+        if enc is not None:
+            ids = apply_dsv32_chat_template(
+                enc, messages, self.tokenizer,
+                tools=kwargs.pop("tools", None), tokenize=True, **kwargs)
+            if not ids:
+                raise ValueError("empty prompt after tokenization")
+            return ids
         # render to text then encode: apply_chat_template's tokenize=True
         # return type varies across transformers versions (list vs dict)
         text = self.tokenizer.apply_chat_template(
