@@ -68,12 +68,21 @@ class Sampler(torch.nn.Module):
                          meta: SamplingMetadata) -> torch.Tensor:
         logits = logits.float()
         if meta.penalty_pool is not None and meta.penalty_slots is not None:
-            slots = meta.penalty_slots.clamp_min(0)
-            rows = meta.penalty_pool.mask[slots, :logits.shape[1]]
-            pen = meta.penalties.to(logits.device).unsqueeze(1)
-            active = (rows != 0) & (pen != 1.0) &                 (meta.penalty_slots.unsqueeze(1) >= 0)
-            adj = torch.where(logits > 0, logits / pen, logits * pen)
-            logits = torch.where(active, adj, logits)
+            if logits.is_cuda:
+                # fused gather+scale HIP kernel (csrc/sampling.hip)
+                from gllm_amd import ops
+                logits = logits.contiguous()
+                ops.apply_penalty_pool(
+                    logits, meta.penalty_pool.mask, meta.penalty_slots,
+                    meta.penalties.to(logits.device).float())
+            else:
+                slots = meta.penalty_slots.clamp_min(0)
+                rows = meta.penalty_pool.mask[slots, :logits.shape[1]]
+                pen = meta.penalties.to(logits.device).unsqueeze(1)
+                active = (rows != 0) & (pen != 1.0) & \
+                    (meta.penalty_slots.unsqueeze(1) >= 0)
+                adj = torch.where(logits > 0, logits / pen, logits * pen)
+                logits = torch.where(active, adj, logits)
         if meta.token_id_rows is not None:
             logits = torch_ref.apply_repetition_penalty(
                 logits, meta.token_id_rows, meta.penalties)

@@ -138,6 +138,16 @@ def grouped_topk(scores, topk, n_group, topk_group, renormalize=True,
 
 
 # --------------------------------------------------------------- mla
+def apply_penalty_pool(logits, mask_pool, slots, penalties):
+    """In-place repetition penalty against the persistent uint8 mask
+    pool (GPU kernel; CPU callers use the inline torch math in
+    layers/sampler.py)."""
+    assert logits.is_cuda
+    _gpu_kernels().apply_repetition_penalty(logits, mask_pool, slots,
+                                            penalties)
+    return logits
+
+
 def mla_paged_attention(q, k_cache, v_cache, block_table, seq_lens,
                         query_start_loc, scale, topk_positions=None):
     if q.is_cuda:

@@ -26,6 +26,15 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor query_start_loc,
                              long max_query_len, double scale,
                              long sliding_window);
+void apply_repetition_penalty(torch::Tensor logits, torch::Tensor pool,
+                              torch::Tensor slots,
+                              torch::Tensor penalties);
+std::pair<int64_t, py::bytes> car_alloc(int64_t data_bytes);
+int64_t car_open(py::bytes handle_bytes);
+void car_close(int64_t ptr);
+void car_free(int64_t ptr);
+void car_all_reduce(torch::Tensor inout, std::vector<int64_t> ptrs,
+                    int64_t rank, int64_t world, int64_t epoch);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (gfx950)");
@@ -41,4 +50,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "varlen MFMA paged prefill attention");
   m.def("skinny_gemm", &skinny_gemm,
         "decode-regime weight-streaming GEMM (M<=256)");
+  m.def("apply_repetition_penalty", &apply_repetition_penalty,
+        "scaling penalty vs persistent seen-token mask pool");
+  m.def("car_alloc", &car_alloc, "alloc hipIpc-shared AR buffer");
+  m.def("car_open", &car_open, "map a peer's AR buffer");
+  m.def("car_close", &car_close);
+  m.def("car_free", &car_free);
+  m.def("car_all_reduce", &car_all_reduce,
+        "one-shot xGMI custom all-reduce (epoch advances by 2/call)");
 }

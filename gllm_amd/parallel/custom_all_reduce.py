@@ -1,21 +1,106 @@
-"""xGMI direct-write custom all-reduce (placeholder hook, round 1).
+"""Custom all-reduce over hipIpc-mapped peer buffers (xGMI).
 
-The reference uses a cudaIPC two-shot NVLink AR for <=8 MB TP messages
-(distributed/custom_all_reduce.py). The MI355X equivalent is a
-hipIpc-mapped two-shot AR over xGMI's all-to-all point-to-point links
-(reduce-scatter by direct peer writes + all-gather) — implemented in a
-later round as a HIP kernel with graph-capture buffer registration.
-Until then every call falls through to RCCL.
+Reference parity: distributed/custom_all_reduce.py (sgl_kernel NVLink
+two-shot AR + cuda_wrapper.py ctypes IPC — here the hipIpc calls live
+in the extension, ops/csrc/custom_ar.hip, and the kernel reads peers
+straight over point-to-point xGMI links).
+
+Round-1 status: one-shot AR, eager-mode only (no graph capture),
+gated OFF by default — enable with GLLM_CUSTOM_AR=1. Validated by a
+2-process-on-1-GPU numerics test (tests/test_custom_ar_gpu.py); the
+multi-GPU xGMI latency win needs the 8-GPU tier (ROADMAP.md item 5).
+Eligibility mirrors the reference's should_custom_ar: bf16/fp32,
+payload <= max_bytes, world size 2..8, not during graph capture.
 """
 
+import os
 from typing import Optional
 
 import torch
 
-_ENABLED = False
+_MAX_BYTES = 8 << 20
+_INSTANCE = None
+_INIT_TRIED = False
+
+
+class CustomAllReduce:
+    def __init__(self, group, rank: int, world: int,
+                 max_bytes: int = _MAX_BYTES):
+        from gllm_amd import _kernels as K
+        import torch.distributed as dist
+        self.K = K
+        self.rank = rank
+        self.world = world
+        self.max_bytes = max_bytes
+        self.my_ptr, handle = K.car_alloc(max_bytes)
+        # exchange 64-byte hipIpc handles over the group (byte tensors,
+        # reference custom_all_reduce.py:57-78)
+        h = torch.zeros(len(handle), dtype=torch.uint8)
+        h[:] = torch.tensor(list(handle), dtype=torch.uint8)
+        gathered = [torch.zeros_like(h) for _ in range(world)]
+        dist.all_gather(gathered, h, group=group)
+        self.ptrs = []
+        self._opened = []
+        for r in range(world):
+            if r == rank:
+                self.ptrs.append(self.my_ptr)
+            else:
+                p = K.car_open(bytes(gathered[r].tolist()))
+                self.ptrs.append(p)
+                self._opened.append(p)
+        self.epoch = 0
+        # all peers mapped before anyone reduces
+        dist.barrier(group=group)
+
+    def eligible(self, t: torch.Tensor) -> bool:
+        if t.dtype not in (torch.bfloat16, torch.float32):
+            return False
+        if t.numel() * t.element_size() > self.max_bytes:
+            return False
+        if torch.cuda.is_current_stream_capturing():
+            return False
+        return True
+
+    def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        t = t.contiguous()
+        # the kernel issues barriers at epoch+1 and epoch+2
+        self.K.car_all_reduce(t, self.ptrs, self.rank, self.world,
+                              self.epoch + 1)
+        self.epoch += 2
+        return t
+
+    def close(self):
+        for p in self._opened:
+            self.K.car_close(p)
+        self.K.car_free(self.my_ptr)
+
+
+def init_custom_all_reduce() -> Optional[CustomAllReduce]:
+    """Build the AR for the current TP group (call after
+    init_distributed). No-op unless GLLM_CUSTOM_AR=1."""
+    global _INSTANCE, _INIT_TRIED
+    _INIT_TRIED = True
+    if os.environ.get("GLLM_CUSTOM_AR", "0") != "1":
+        return None
+    if not torch.cuda.is_available():
+        return None
+    from gllm_amd.parallel import state as S
+    world = S.get_tp_size()
+    if not (2 <= world <= 8):
+        return None
+    try:
+        _INSTANCE = CustomAllReduce(S.get_tp_group(), S.get_tp_rank(),
+                                    world)
+    except Exception:  # pragma: no cover - driver without IPC support
+        from gllm_amd.logger import logger
+        logger.exception("custom AR init failed; falling back to RCCL")
+        _INSTANCE = None
+    return _INSTANCE
 
 
 def try_custom_all_reduce(t: torch.Tensor) -> Optional[torch.Tensor]:
-    if not _ENABLED:
+    if _INSTANCE is None:
         return None
-    return None
+    if not _INSTANCE.eligible(t):
+        return None
+    return _INSTANCE.all_reduce(t)

@@ -112,6 +112,11 @@ def init_distributed(config=None, rank: Optional[int] = None,
         if rank in ranks:
             _EP_GROUP = g
     _INITIALIZED = True
+    if use_gpu and tp_size > 1:
+        # opt-in hipIpc/xGMI custom AR (GLLM_CUSTOM_AR=1)
+        from gllm_amd.parallel.custom_all_reduce import \
+            init_custom_all_reduce
+        init_custom_all_reduce()
 
 
 def destroy_distributed() -> None:
