@@ -46,6 +46,13 @@ class EngineConfig:
     mm_encoder_addr: Optional[str] = None
     discovery_addr: Optional[str] = None
 
+    # --- MLA (DeepSeek family) ---
+    # "absorbed" (default): 576-dim latent MQA cache, W_UK/W_UV folded
+    # into q/out (models/deepseek_v2.py) — ~70x less KV than
+    # decompressed per-head K/V on real configs.
+    # "decompressed": per-head K/V cache (the numerics cross-check).
+    mla_mode: str = "absorbed"
+
     # --- KV cache ---
     page_size: int = 16
     gpu_memory_util: float = 0.9

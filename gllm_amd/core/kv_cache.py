@@ -31,6 +31,9 @@ class KVCacheSpec:
     page_size: int
     dtype_bytes: int = 2     # bf16
     v_head_dim: int = 0      # 0 => same as head_dim (MLA: Dv != Dk)
+    # absorbed MLA: the v-cache is a zero-copy view of the latent
+    # k-cache's first v_head_dim dims — only k bytes are allocated
+    v_shared: bool = False
 
     def __post_init__(self):
         if not self.v_head_dim:
@@ -38,8 +41,9 @@ class KVCacheSpec:
 
     @property
     def bytes_per_page(self) -> int:
+        v = 0 if self.v_shared else self.v_head_dim
         return (self.num_layers * self.page_size * self.num_kv_heads
-                * (self.head_dim + self.v_head_dim) * self.dtype_bytes)
+                * (self.head_dim + v) * self.dtype_bytes)
 
 
 class MemoryManager:

@@ -35,12 +35,30 @@ def _yarn_mscale(scale: float, mscale: float) -> float:
 
 
 class MLAAttention(nn.Module):
-    def __init__(self, cfg, layer_idx: int, dtype=None):
+    """MLA attention. Two execution modes (EngineConfig.mla_mode):
+
+    * ``absorbed`` (default) — the latent form: the paged cache holds
+      ONE shared row per token, [c_kv (kv_lora_rank) | k_pe (rope)]
+      (576 dims on real configs), attended as MQA. W_UK is folded into
+      the query (q_lat = q_nope @ W_UK) and W_UV applied to the latent
+      attention output; mathematically identical to decompressed
+      (q·(W_UK c) == (W_UK^T q)·c) at ~H*(dn+dv)/(lora+rope) less KV —
+      ~71x for DeepSeek-V3. The v-cache is a zero-copy VIEW of the
+      latent's first kv_lora dims. This layout is the contract for the
+      round-2 gfx950 absorbed-decode kernel (576-dim MQA over the
+      latent cache, reference layers/attention.py:653-925).
+    * ``decompressed`` — per-head K/V cache, the numerics cross-check
+      (tests/test_deepseek_cpu.py asserts both modes emit identical
+      tokens).
+    """
+
+    def __init__(self, cfg, layer_idx: int, dtype=None, absorbed=True):
         super().__init__()
         from gllm_amd.parallel import get_tp_size
         tp = get_tp_size()
         hidden = cfg.hidden_size
         self.layer_idx = layer_idx
+        self.absorbed = absorbed
         self.total_heads = cfg.num_attention_heads
         assert self.total_heads % tp == 0
         self.num_heads = self.total_heads // tp
@@ -93,6 +111,13 @@ class MLAAttention(nn.Module):
         None for dense MLA. Overridden in models/deepseek_v32.py."""
         return None
 
+    def _uk_uv(self):
+        """Per-head absorption views of kv_b_proj's weight:
+        W_UK [H, dn, lora], W_UV [H, dv, lora]."""
+        w = self.kv_b_proj.weight.view(
+            self.num_heads, self.qk_nope + self.v_dim, self.kv_lora_rank)
+        return w[:, :self.qk_nope, :], w[:, self.qk_nope:, :]
+
     def forward(self, positions, hidden, fctx: ForwardContext):
         T = hidden.shape[0]
         H = self.num_heads
@@ -107,9 +132,7 @@ class MLAAttention(nn.Module):
 
         kv_a = self.kv_a_proj_with_mqa(hidden)
         c_kv, k_pe = kv_a.split([self.kv_lora_rank, self.qk_rope], dim=-1)
-        kv = self.kv_b_proj(self.kv_a_layernorm(c_kv.contiguous()))
-        kv = kv.view(T, H, self.qk_nope + self.v_dim)
-        k_nope, v = kv.split([self.qk_nope, self.v_dim], dim=-1)
+        c_kv = self.kv_a_layernorm(c_kv.contiguous())
 
         # rope on q_pe (per head) and the shared k_pe (one "head")
         q_pe = q_pe.reshape(T, H * self.qk_rope).contiguous()
@@ -117,6 +140,12 @@ class MLAAttention(nn.Module):
         q_pe, k_pe = self.rotary_emb(positions, q_pe, k_pe)
         q_pe = q_pe.view(T, H, self.qk_rope)
 
+        if self.absorbed:
+            return self._forward_absorbed(positions, hidden, q_resid,
+                                          q_nope, q_pe, c_kv, k_pe, fctx)
+        kv = self.kv_b_proj(c_kv)
+        kv = kv.view(T, H, self.qk_nope + self.v_dim)
+        k_nope, v = kv.split([self.qk_nope, self.v_dim], dim=-1)
         k = torch.cat(
             [k_nope, k_pe.unsqueeze(1).expand(T, H, self.qk_rope)], dim=-1)
         qf = torch.cat([q_nope, q_pe], dim=-1).contiguous()
@@ -132,6 +161,34 @@ class MLAAttention(nn.Module):
         out = ops.mla_paged_attention(
             qf, k_cache, v_cache, fctx.block_table, fctx.seq_lens,
             fctx.query_start_loc, self.scale, topk_positions=topk_pos)
+        return self.o_proj(out.reshape(T, -1))
+
+    def _forward_absorbed(self, positions, hidden, q_resid, q_nope, q_pe,
+                          c_kv, k_pe, fctx):
+        T, H = q_nope.shape[0], self.num_heads
+        if fctx.is_profile_run:
+            # peak-shaped dummy matching the o_proj input
+            return q_nope.new_zeros(T, H * self.v_dim)
+        w_uk, w_uv = self._uk_uv()
+        # q_lat[t,h,l] = sum_d q_nope[t,h,d] * W_UK[h,d,l]
+        q_lat = torch.einsum("thd,hdl->thl", q_nope.float(),
+                             w_uk.float()).to(q_nope.dtype)
+        qf = torch.cat([q_lat, q_pe], dim=-1).contiguous()  # [T,H,l+r]
+        k_lat = torch.cat([c_kv, k_pe], dim=-1).unsqueeze(1)  # [T,1,l+r]
+        k_cache = fctx.k_caches[self.layer_idx]   # [P,page,1,lora+rope]
+        v_cache = fctx.v_caches[self.layer_idx]   # VIEW: [..., :lora]
+        # write the latent row once (the v view aliases its first lora
+        # dims, so caching k fills both)
+        ops.reshape_and_cache(k_lat.contiguous(),
+                              c_kv.unsqueeze(1).contiguous(),
+                              k_cache, v_cache, fctx.slot_mapping)
+        topk_pos = self._dsa_select(positions, hidden, q_resid, fctx)
+        out_lat = ops.mla_paged_attention(
+            qf, k_cache, v_cache, fctx.block_table, fctx.seq_lens,
+            fctx.query_start_loc, self.scale, topk_positions=topk_pos)
+        # out[t,h,v] = sum_l out_lat[t,h,l] * W_UV[h,v,l]
+        out = torch.einsum("thl,hvl->thv", out_lat.float(),
+                           w_uv.float()).to(out_lat.dtype)
         return self.o_proj(out.reshape(T, -1))
 
 
@@ -195,8 +252,9 @@ class DeepseekDecoderLayer(nn.Module):
                  dtype=None, attn_cls=None):
         super().__init__()
         eps = getattr(cfg, "rms_norm_eps", 1e-6)
-        self.self_attn = (attn_cls or MLAAttention)(cfg, layer_idx,
-                                                    dtype=dtype)
+        self.self_attn = (attn_cls or MLAAttention)(
+            cfg, layer_idx, dtype=dtype,
+            absorbed=engine_config.mla_mode == "absorbed")
         first_dense = getattr(cfg, "first_k_dense_replace", 0)
         step = getattr(cfg, "moe_layer_freq", 1)
         is_moe = (getattr(cfg, "n_routed_experts", None)
@@ -258,7 +316,16 @@ class DeepseekV2ForCausalLM(nn.Module):
     @property
     def kv_geometry(self):
         a = self.layers[0].self_attn
+        if a.absorbed:
+            return 1, a.kv_lora_rank + a.qk_rope, a.kv_lora_rank
         return a.num_heads, a.qk_dim, a.v_dim
+
+    @property
+    def kv_share_latent(self):
+        """Absorbed MLA: v-cache = view of k-cache[..., :kv_lora_rank]
+        (runtime/model_runner.py honors this at allocation)."""
+        a = self.layers[0].self_attn
+        return a.kv_lora_rank if a.absorbed else None
 
     def forward(self, input_ids, positions, fctx, hidden_states=None,
                 residual=None):

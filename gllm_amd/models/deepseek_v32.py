@@ -95,8 +95,8 @@ class DSAMLAAttention(MLAAttention):
     into the paged index cache, then selects each query's top-k token
     positions (reference deepseek_v32.py:637-737 forward wiring)."""
 
-    def __init__(self, cfg, layer_idx, dtype=None):
-        super().__init__(cfg, layer_idx, dtype=dtype)
+    def __init__(self, cfg, layer_idx, dtype=None, absorbed=True):
+        super().__init__(cfg, layer_idx, dtype=dtype, absorbed=absorbed)
         self.indexer = DSALightningIndexer(cfg, dtype=dtype)
 
     @torch.no_grad()

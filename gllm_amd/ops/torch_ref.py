@@ -161,8 +161,13 @@ def mla_paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
         s_len = int(seq_lens[b])
         n_pages = -(-s_len // page_size)
         pages = block_table[b, :n_pages].long()
-        k = k_cache[pages].reshape(-1, H, Dk)[:s_len].float()
-        v = v_cache[pages].reshape(-1, H, Dv)[:s_len].float()
+        Hkv = k_cache.shape[2]
+        k = k_cache[pages].reshape(-1, Hkv, Dk)[:s_len].float()
+        v = v_cache[pages].reshape(-1, Hkv, Dv)[:s_len].float()
+        if Hkv == 1 and H > 1:
+            # absorbed MLA: MQA over the shared latent row
+            k = k.expand(s_len, H, Dk)
+            v = v.expand(s_len, H, Dv)
         qq = q[qs:qe].float()
         scores = torch.einsum("lhd,shd->hls", qq, k) * scale
         past = s_len - q_len

@@ -118,10 +118,12 @@ class ModelRunner:
             else:
                 kv_per_rank, head_dim = attn
                 v_dim = head_dim
+            shared = getattr(self.model, "kv_share_latent", None)
             return KVCacheSpec(num_local_layers, kv_per_rank, head_dim,
                                self.config.page_size,
                                dtype_bytes=self.kv_dtype.itemsize,
-                               v_head_dim=v_dim)
+                               v_head_dim=v_dim,
+                               v_shared=shared is not None)
         if True:
             total_kv = getattr(hf, "num_key_value_heads",
                                hf.num_attention_heads)
@@ -213,9 +215,15 @@ class ModelRunner:
         self.k_caches = [torch.zeros(kshape, dtype=self.kv_dtype,
                                      device=self.device)
                          for _ in range(spec.num_layers)]
-        self.v_caches = [torch.zeros(vshape, dtype=self.kv_dtype,
-                                     device=self.device)
-                         for _ in range(spec.num_layers)]
+        if spec.v_shared:
+            # absorbed MLA: v = zero-copy view of the latent's first
+            # v_head_dim dims
+            self.v_caches = [k[..., :spec.v_head_dim]
+                             for k in self.k_caches]
+        else:
+            self.v_caches = [torch.zeros(vshape, dtype=self.kv_dtype,
+                                         device=self.device)
+                             for _ in range(spec.num_layers)]
         # DSA (DeepSeek-V3.2): paged index-K cache parallel to the KV
         # pool, one per local layer (reference memory_manager.py:334-362)
         self.idx_caches = None

@@ -122,3 +122,35 @@ def test_deepseek_generates(tmp_path, cfg_json):
     llm2 = LLM(config=cfg2, num_pages_override=128)
     out2 = llm2.generate([list(range(1, 25))], [sp])
     assert out2[0].token_ids == out[0].token_ids
+
+
+def test_mla_absorbed_equals_decompressed(tmp_path):
+    """The latent-MQA absorbed path (default) must emit exactly the
+    decompressed per-head path's tokens: q.(W_UK c) == (W_UK^T q).c."""
+    import json as _json
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    from gllm_amd.sequence import SamplingParams
+
+    outs = {}
+    for mode in ("absorbed", "decompressed"):
+        d = tmp_path / mode
+        d.mkdir()
+        with open(d / "config.json", "w") as f:
+            _json.dump(DSV2_TINY, f)
+        cfg = EngineConfig(model=str(d), load_format="dummy",
+                           device="cpu", dtype="float32", page_size=4,
+                           maxp=64, mla_mode=mode,
+                           enable_prefix_caching=False)
+        llm = LLM(config=cfg, num_pages_override=128)
+        if mode == "absorbed":
+            # latent cache: 1 head, lora+rope dims; v is a view
+            k = llm.runner.k_caches[0]
+            v = llm.runner.v_caches[0]
+            assert k.shape[2] == 1 and k.shape[3] == 48 + 8
+            assert v.data_ptr() == k.data_ptr()  # zero-copy
+        sp = [SamplingParams(temperature=0.0, max_tokens=6,
+                             ignore_eos=True)] * 2
+        res = llm.generate([list(range(1, 20)), [5, 6, 7]], sp)
+        outs[mode] = [o.token_ids for o in res]
+    assert outs["absorbed"] == outs["decompressed"]
