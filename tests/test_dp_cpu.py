@@ -139,6 +139,70 @@ def test_dp2_ep_moe_equals_single(tmp_path):
     assert [got[i] for i in range(len(PROMPTS))] == ref
 
 
+def _run_dp_fuzz_rank(rank, model_dir, port, prompts, max_tokens, q):
+    os.environ.update(RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    from gllm_amd.engine.pp_engine import PPEngine
+    from gllm_amd.sequence import SamplingParams, Sequence
+    eng = PPEngine(_mk_cfg(model_dir, 2, True, port),
+                   num_pages_override=128)
+    mine = [(i, p) for i, p in enumerate(prompts) if i % 2 == eng.dp_rank]
+    seqs = [Sequence(i, p, SamplingParams(temperature=0.0,
+                                          max_tokens=max_tokens[i],
+                                          ignore_eos=True))
+            for i, p in mine]
+    # staggered intake: half the requests arrive after the first rounds
+    eng.add_requests(seqs[:len(seqs) // 2 + 1])
+    eng.run_until_done(max_steps=3)
+    eng.add_requests(seqs[len(seqs) // 2 + 1:])
+    eng.run_until_done()
+    q.put((rank, [(s.seq_id, s.output_token_ids) for s in seqs]))
+    import torch.distributed as dist
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("seed,port", [(0, 29763), (1, 29765)])
+def test_dp2_fuzz_staggered_load_equals_single(tmp_path, seed, port):
+    """Randomized skewed workloads + staggered arrival across replicas:
+    dummy-round interleavings must never change results."""
+    rng = torch.Generator().manual_seed(seed)
+
+    def ri(lo, hi):
+        return int(torch.randint(lo, hi, (1,), generator=rng))
+
+    prompts = [[ri(1, 120) for _ in range(ri(3, 24))] for _ in range(7)]
+    max_tokens = [ri(1, 9) for _ in range(7)]
+    d = str(tmp_path / "ckpt")
+    _write_checkpoint(d)
+
+    from gllm_amd.engine.llm import LLM
+    from gllm_amd.sequence import SamplingParams
+    llm = LLM(config=_mk_cfg(d, 1, False, 0), num_pages_override=128)
+    sp = [SamplingParams(temperature=0.0, max_tokens=mt, ignore_eos=True)
+          for mt in max_tokens]
+    ref = [o.token_ids for o in llm.generate(prompts, sp)]
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_dp_fuzz_rank,
+                         args=(r, d, port, prompts, max_tokens, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = {}
+    for _ in range(2):
+        _rank, outs = q.get(timeout=240)
+        for sid, toks in outs:
+            got[sid] = toks
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert [got[i] for i in range(len(prompts))] == ref
+
+
 def test_abort_unknown_seq_id_is_ignored():
     """An abort for a seq this scheduler never saw (another DP replica's,
     or already finished) must not poison a future recycled seq id."""
