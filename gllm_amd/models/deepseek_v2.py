@@ -179,9 +179,7 @@ class MLAAttention(nn.Module):
         v_cache = fctx.v_caches[self.layer_idx]   # VIEW: [..., :lora]
         # write the latent row once (the v view aliases its first lora
         # dims, so caching k fills both)
-        ops.reshape_and_cache(k_lat.contiguous(),
-                              c_kv.unsqueeze(1).contiguous(),
-                              k_cache, v_cache, fctx.slot_mapping)
+        ops.cache_latent(k_lat.contiguous(), k_cache, fctx.slot_mapping)
         topk_pos = self._dsa_select(positions, hidden, q_resid, fctx)
         out_lat = ops.mla_paged_attention(
             qf, k_cache, v_cache, fctx.block_table, fctx.seq_lens,

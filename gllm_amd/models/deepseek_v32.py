@@ -53,10 +53,10 @@ class DSALightningIndexer(nn.Module):
                                      params_dtype=dtype)
         self.wk = ReplicatedLinear(hidden, self.head_dim,
                                    params_dtype=dtype)
-        # LayerNorm (with bias) — unlike the RMSNorms elsewhere
+        # LayerNorm (with bias) — unlike the RMSNorms elsewhere; kept
+        # fp32 (the scorer accumulates fp32, reference keeps the
+        # indexer weighting in fp32)
         self.k_norm = nn.LayerNorm(self.head_dim, eps=1e-6)
-        if dtype is not None:
-            self.k_norm = self.k_norm.to(dtype)
         self.weights_proj = ReplicatedLinear(hidden, self.n_heads,
                                              params_dtype=dtype)
         rope_scaling = getattr(cfg, "rope_scaling", None)
@@ -114,8 +114,9 @@ class DSAMLAAttention(MLAAttention):
                          idx_k.to(cache.dtype))
 
         T = hidden.shape[0]
+        dev = hidden.device
         topk = self.indexer.index_topk
-        out = torch.full((T, topk), -1, dtype=torch.int32)
+        out = torch.full((T, topk), -1, dtype=torch.int32, device=dev)
         qsl = fctx.query_start_loc
         for b in range(fctx.seq_lens.shape[0]):
             qs, qe = int(qsl[b]), int(qsl[b + 1])
@@ -129,8 +130,8 @@ class DSAMLAAttention(MLAAttention):
             logits = self.indexer.score(idx_q[qs:qe], keys,
                                         weights[qs:qe])  # [q_len, s_len]
             past = s_len - q_len
-            pos_q = torch.arange(q_len).unsqueeze(1) + past
-            pos_k = torch.arange(s_len).unsqueeze(0)
+            pos_q = torch.arange(q_len, device=dev).unsqueeze(1) + past
+            pos_k = torch.arange(s_len, device=dev).unsqueeze(0)
             logits = logits.masked_fill(pos_k > pos_q, float("-inf"))
             k_sel = min(topk, s_len)
             top = logits.topk(k_sel, dim=-1)

@@ -148,12 +148,37 @@ def apply_penalty_pool(logits, mask_pool, slots, penalties):
     return logits
 
 
+def cache_latent(k, k_cache, slot_mapping):
+    """Scatter the absorbed-MLA latent row [T, 1, lora+rope] into the
+    paged latent cache (the v 'cache' is a view of its first lora dims,
+    so one write covers both)."""
+    if k.is_cuda:
+        # reuse the K/V scatter kernel with both targets = the latent
+        # cache (same row written twice; a dedicated single-row kernel
+        # rides the round-2 MLA kernel work)
+        _gpu_kernels().reshape_and_cache(k, k, k_cache, k_cache,
+                                         slot_mapping)
+        return
+    page_size = k_cache.shape[1]
+    pages = torch.div(slot_mapping, page_size, rounding_mode="floor")
+    offs = slot_mapping % page_size
+    k_cache[pages, offs] = k.to(k_cache.dtype)
+
+
+_MLA_GPU_WARNED = False
+
+
 def mla_paged_attention(q, k_cache, v_cache, block_table, seq_lens,
                         query_start_loc, scale, topk_positions=None):
-    if q.is_cuda:
-        raise NotImplementedError(
-            "MLA paged attention HIP kernel lands in round 2 — the "
-            "DeepSeek family currently runs on the CPU reference path")
+    global _MLA_GPU_WARNED
+    if q.is_cuda and not _MLA_GPU_WARNED:
+        # NOT a silent fallback: no native MLA kernel exists yet — the
+        # torch path IS the current GPU implementation for the absorbed
+        # 576-dim latent MQA (gfx950 kernel = ROADMAP.md item 7)
+        from gllm_amd.logger import logger
+        logger.warning("absorbed-MLA attention runs the torch path on "
+                       "GPU (gfx950 MLA kernel lands in round 2)")
+        _MLA_GPU_WARNED = True
     return torch_ref.mla_paged_attention(q, k_cache, v_cache, block_table,
                                          seq_lens, query_start_loc, scale,
                                          topk_positions=topk_positions)

@@ -68,20 +68,19 @@ template <typename T>
 __global__ void reshape_and_cache_kernel(
     const T *__restrict__ k, const T *__restrict__ v, T *__restrict__ k_cache,
     T *__restrict__ v_cache, const long *__restrict__ slot_mapping,
-    int row_elems /* H*D */, int page_size, long k_stride, long v_stride) {
+    int k_row_elems /* H*Dk */, int v_row_elems /* H*Dv (MLA: != k) */,
+    int page_size, long k_stride, long v_stride) {
   const long t = blockIdx.x;
   const long slot = slot_mapping[t];
   const long page = slot / page_size, off = slot % page_size;
-  const long dst = (page * page_size + off) * (long)row_elems;
-  const int nvec = row_elems / 8;
+  const long tok = page * page_size + off;
+  const int knvec = k_row_elems / 8, vnvec = v_row_elems / 8;
   const shortx8 *ks = reinterpret_cast<const shortx8 *>(k + t * k_stride);
   const shortx8 *vs = reinterpret_cast<const shortx8 *>(v + t * v_stride);
-  shortx8 *kd = reinterpret_cast<shortx8 *>(k_cache + dst);
-  shortx8 *vd = reinterpret_cast<shortx8 *>(v_cache + dst);
-  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
-    kd[i] = ks[i];
-    vd[i] = vs[i];
-  }
+  shortx8 *kd = reinterpret_cast<shortx8 *>(k_cache + tok * (long)k_row_elems);
+  shortx8 *vd = reinterpret_cast<shortx8 *>(v_cache + tok * (long)v_row_elems);
+  for (int i = threadIdx.x; i < knvec; i += blockDim.x) kd[i] = ks[i];
+  for (int i = threadIdx.x; i < vnvec; i += blockDim.x) vd[i] = vs[i];
 }
 
 }  // namespace
@@ -148,24 +147,28 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v,
                        torch::Tensor slot_mapping) {
   const long T = k.size(0);
   if (T == 0) return;
-  const int row = k.size(1) * k.size(2);
+  const int k_row = k.size(1) * k.size(2);
+  const int v_row = v.size(1) * v.size(2);
   const int page_size = k_cache.size(1);
-  TORCH_CHECK(row % 8 == 0);
+  TORCH_CHECK(k_row % 8 == 0 && v_row % 8 == 0);
   TORCH_CHECK(k.stride(-1) == 1 && v.stride(-1) == 1 &&
               k.stride(1) == k.size(2) && v.stride(1) == v.size(2),
               "reshape_and_cache: per-token row must be contiguous");
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+  TORCH_CHECK(k_cache.size(2) * k_cache.size(3) == k_row &&
+              v_cache.size(2) * v_cache.size(3) == v_row,
+              "cache row dims must match inputs");
   TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
   TORCH_CHECK(k.scalar_type() == at::kBFloat16);
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int block = std::min(256, std::max(64, row / 8));
+  const int block = std::min(256, std::max(64, k_row / 8));
   hipLaunchKernelGGL((reshape_and_cache_kernel<__hip_bfloat16>), dim3(T),
                      dim3(block), 0, stream,
                      (const __hip_bfloat16 *)k.data_ptr(),
                      (const __hip_bfloat16 *)v.data_ptr(),
                      (__hip_bfloat16 *)k_cache.data_ptr(),
                      (__hip_bfloat16 *)v_cache.data_ptr(),
-                     slot_mapping.data_ptr<long>(), row, page_size,
-                     k.stride(0), v.stride(0));
+                     slot_mapping.data_ptr<long>(), k_row, v_row,
+                     page_size, k.stride(0), v.stride(0));
   HIP_CHECK_KERNEL();
 }

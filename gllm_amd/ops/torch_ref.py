@@ -169,16 +169,19 @@ def mla_paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
             k = k.expand(s_len, H, Dk)
             v = v.expand(s_len, H, Dv)
         qq = q[qs:qe].float()
+        dev = q.device
         scores = torch.einsum("lhd,shd->hls", qq, k) * scale
         past = s_len - q_len
-        pos_q = torch.arange(q_len).unsqueeze(1) + past
-        pos_k = torch.arange(s_len).unsqueeze(0)
+        pos_q = torch.arange(q_len, device=dev).unsqueeze(1) + past
+        pos_k = torch.arange(s_len, device=dev).unsqueeze(0)
         mask = pos_k <= pos_q                          # [q_len, s_len]
         if topk_positions is not None:
-            sel = topk_positions[qs:qe].long()         # [q_len, topk]
-            allowed = torch.zeros(q_len, s_len, dtype=torch.bool)
+            sel = topk_positions[qs:qe].long().to(dev)  # [q_len, topk]
+            allowed = torch.zeros(q_len, s_len, dtype=torch.bool,
+                                  device=dev)
             valid = (sel >= 0) & (sel < s_len)
-            rows = torch.arange(q_len).unsqueeze(1).expand_as(sel)
+            rows = torch.arange(q_len,
+                                device=dev).unsqueeze(1).expand_as(sel)
             allowed[rows[valid], sel[valid]] = True
             mask = mask & allowed
         scores.masked_fill_(~mask.unsqueeze(0), float("-inf"))
