@@ -132,7 +132,11 @@ class GatedDeltaNet(nn.Module):
                     mixed[s:e], self.conv1d_weight, conv_states[slot],
                     has_init)
             qd, kd, vd = conv_out.split([k_tp, k_tp, v_tp], dim=-1)
-            o = gdn_ref.gated_delta_rule(
+            # decode: one sequential step; prefill: chunk-parallel WY
+            # form (O(T/64) sequential steps — the round-2 kernel shape)
+            rule = gdn_ref.gated_delta_rule if e - s == 1 \
+                else gdn_ref.gated_delta_rule_chunked
+            o = rule(
                 qd.view(e - s, self.tp_k_heads, self.head_k_dim),
                 kd.view(e - s, self.tp_k_heads, self.head_k_dim),
                 vd.view(e - s, self.tp_v, self.head_v_dim),

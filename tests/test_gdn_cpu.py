@@ -167,3 +167,26 @@ def test_hybrid_moe_generates(tmp_path):
     o1 = llm.generate([[1, 2, 3, 4, 5, 6]], sp)[0].token_ids
     o2 = llm.generate([[1, 2, 3, 4, 5, 6]], sp)[0].token_ids
     assert len(o1) == 5 and o1 == o2
+
+
+def test_chunked_delta_rule_equals_sequential():
+    """The chunk-parallel WY form must match the sequential recurrence
+    (state carry included) across chunk sizes and state carry-in."""
+    import torch
+    from gllm_amd.ops.gdn_ref import (gated_delta_rule,
+                                      gated_delta_rule_chunked)
+    torch.manual_seed(1)
+    T, Hk, Dk, Hv, Dv = 53, 2, 16, 4, 8
+    q = torch.randn(T, Hk, Dk)
+    k = torch.randn(T, Hk, Dk)
+    v = torch.randn(T, Hv, Dv)
+    g = -torch.rand(T, Hv)
+    beta = torch.sigmoid(torch.randn(T, Hv))
+    for chunk in (1, 7, 16, 64, 128):
+        s_ref = torch.randn(Hv, Dv, Dk)
+        s_chk = s_ref.clone()
+        o_ref = gated_delta_rule(q, k, v, g, beta, 0.25, s_ref)
+        o_chk = gated_delta_rule_chunked(q, k, v, g, beta, 0.25, s_chk,
+                                         chunk=chunk)
+        assert torch.allclose(o_ref, o_chk, atol=1e-4), chunk
+        assert torch.allclose(s_ref, s_chk, atol=1e-4), chunk
