@@ -79,9 +79,11 @@ class GatedDeltaNet(nn.Module):
         self.conv1d_weight.weight_loader = self._load_conv
         self.dt_bias = nn.Parameter(torch.ones(self.tp_v),
                                     requires_grad=False)
+        self.dt_bias.weight_loader = self._load_headed
         self.A_log = nn.Parameter(torch.zeros(self.tp_v,
                                               dtype=torch.float32),
                                   requires_grad=False)
+        self.A_log.weight_loader = self._load_headed
         self.norm_weight = nn.Parameter(
             torch.ones(self.head_v_dim, dtype=dtype), requires_grad=False)
         self.out_proj = RowParallelLinear(self.value_dim, cfg.hidden_size,
@@ -95,6 +97,32 @@ class GatedDeltaNet(nn.Module):
         w = loaded.reshape(loaded.shape[0], -1)
         n = w.shape[0] // get_tp_size()
         param.data.copy_(w.narrow(0, get_tp_rank() * n, n))
+
+    @staticmethod
+    def _load_headed(param, loaded):
+        # per-value-head vectors (A_log, dt_bias): TP narrow
+        from gllm_amd.parallel import get_tp_rank, get_tp_size
+        n = loaded.shape[0] // get_tp_size()
+        param.data.copy_(loaded.narrow(0, get_tp_rank() * n, n))
+
+    def load_fused_qkvz(self, loaded):
+        """Checkpoint ships in_proj_qkvz as ONE fused tensor; split into
+        the (q, k, v, z) segments and route each through the merged
+        column loader (reference weight_loader.py GDN fused-proj
+        pre-pass)."""
+        p = self.in_proj_qkvz.weight
+        off = 0
+        for sid, size in enumerate([self.key_dim, self.key_dim,
+                                    self.value_dim, self.value_dim]):
+            p.weight_loader(p, loaded.narrow(0, off, size), sid)
+            off += size
+
+    def load_fused_ba(self, loaded):
+        p = self.in_proj_ba.weight
+        off = 0
+        for sid, size in enumerate([self.num_v_heads, self.num_v_heads]):
+            p.weight_loader(p, loaded.narrow(0, off, size), sid)
+            off += size
 
     def forward(self, hidden: torch.Tensor, fctx: ForwardContext):
         if fctx.is_profile_run or fctx.ssm_pool is None:
@@ -351,9 +379,103 @@ class Qwen3_5ForCausalLM(nn.Module):
         return self.lm_head(rows)
 
     def load_weights(self, weights: Iterable[Tuple[str, torch.Tensor]]):
-        # round-2: checkpoint mapping for real Qwen3.5 weights
-        # (in_proj_qkvz / in_proj_ba / conv1d / A_log / dt_bias / norm /
-        # out_proj names already match — see GatedDeltaNet docstring)
-        raise NotImplementedError(
-            "hybrid GDN checkpoint loading lands in round 2; use "
-            "--load-format dummy")
+        """Qwen3.5 / Qwen3-Next checkpoint mapping: fused-GDN pre-pass
+        (in_proj_qkvz/ba), conv1d/A_log/dt_bias TP shards, gated
+        q_proj (q|gate per head) into the fused qkv, dense + MoE MLPs.
+        """
+        params = dict(self.named_parameters())
+        tied = getattr(self.cfg, "tie_word_embeddings", False)
+        stacked = [("gate_up_proj", "gate_proj", 0),
+                   ("gate_up_proj", "up_proj", 1)]
+        qkv_shards = {"q_proj": "q", "k_proj": "k", "v_proj": "v"}
+        expert_map = [("gate_proj", "w13_weight", 0),
+                      ("up_proj", "w13_weight", 1),
+                      ("down_proj", "w2_weight", None)]
+        for name, w in weights:
+            if name.startswith("model."):
+                name = name[len("model."):]
+            if name.startswith("layers."):
+                parts = name.split(".")
+                g_idx = int(parts[1])
+                if not (self.layer_start <= g_idx < self.layer_end):
+                    continue
+                local = g_idx - self.layer_start
+                parts[1] = str(local)
+                name = ".".join(parts)
+                layer = self.layers[local]
+                rest = ".".join(parts[2:])
+                if rest.startswith("linear_attn."):
+                    la = layer.linear_attn
+                    sub = rest[len("linear_attn."):]
+                    if sub == "in_proj_qkvz.weight":
+                        la.load_fused_qkvz(w)
+                    elif sub == "in_proj_ba.weight":
+                        la.load_fused_ba(w)
+                    elif sub == "conv1d.weight":
+                        la._load_conv(la.conv1d_weight, w)
+                    elif sub in ("A_log", "dt_bias"):
+                        p = getattr(la, sub)
+                        p.weight_loader(p, w)
+                    elif sub == "norm.weight":
+                        la.norm_weight.data.copy_(
+                            w.to(la.norm_weight.dtype))
+                    elif sub == "out_proj.weight":
+                        p = la.out_proj.weight
+                        p.weight_loader(p, w)
+                    continue
+                if rest.startswith("self_attn."):
+                    sub = rest[len("self_attn."):]
+                    base = sub.split(".")[0]
+                    if base in qkv_shards:
+                        p = layer.self_attn.qkv_proj.weight
+                        p.weight_loader(p, w, qkv_shards[base])
+                        continue
+                    # o_proj / q_norm / k_norm fall through by name
+                if ".experts." in name:
+                    eidx = name.split(".").index("experts")
+                    nparts = name.split(".")
+                    expert_id = int(nparts[eidx + 1])
+                    wname = nparts[eidx + 2]
+                    prefix = ".".join(nparts[:eidx + 1])
+                    for ckpt, fused, shard in expert_map:
+                        if wname == ckpt:
+                            p = params[f"{prefix}.{fused}"]
+                            if shard is None:
+                                p.weight_loader(p, w, expert_id)
+                            else:
+                                p.weight_loader(p, w, expert_id, shard)
+                            break
+                    continue
+            elif name.startswith("embed_tokens"):
+                if self.is_first_stage:
+                    p = params["embed_tokens.weight"]
+                    p.weight_loader(p, w)
+                if tied and self.is_last_stage and not self.is_first_stage:
+                    p = params["lm_head.weight"]
+                    p.weight_loader(p, w)
+                continue
+            elif name.startswith("lm_head"):
+                if self.is_last_stage and not tied:
+                    p = params["lm_head.weight"]
+                    p.weight_loader(p, w)
+                continue
+            elif name.startswith("norm."):
+                if not self.is_last_stage:
+                    continue
+            hit = False
+            for fused, ckpt, shard in stacked:
+                if ckpt in name:
+                    tgt = name.replace(ckpt, fused)
+                    if tgt in params:
+                        p = params[tgt]
+                        p.weight_loader(p, w, shard)
+                        hit = True
+                    break
+            if hit:
+                continue
+            if name in params:
+                p = params[name]
+                if hasattr(p, "weight_loader"):
+                    p.weight_loader(p, w)
+                else:
+                    p.data.copy_(w.to(p.dtype))

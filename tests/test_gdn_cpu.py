@@ -190,3 +190,90 @@ def test_chunked_delta_rule_equals_sequential():
                                          chunk=chunk)
         assert torch.allclose(o_ref, o_chk, atol=1e-4), chunk
         assert torch.allclose(s_ref, s_chk, atol=1e-4), chunk
+
+
+def test_hybrid_checkpoint_loading(tmp_path):
+    """Real-checkpoint mapping: fused in_proj_qkvz/ba split, conv1d /
+    A_log / dt_bias, gated q_proj into the fused qkv, dense MLP +
+    norms. Loads a synthetic safetensors checkpoint and checks the
+    parameters landed where the forward reads them."""
+    import os
+    import torch
+    d = tmp_path / "ckpt"
+    d.mkdir()
+    with open(d / "config.json", "w") as f:
+        json.dump(HYBRID_TINY, f)
+    g = torch.Generator().manual_seed(11)
+    H = HYBRID_TINY["hidden_size"]
+    I = HYBRID_TINY["intermediate_size"]
+    V = HYBRID_TINY["vocab_size"]
+    hd = HYBRID_TINY["head_dim"]
+    nh = HYBRID_TINY["num_attention_heads"]
+    nkv = HYBRID_TINY["num_key_value_heads"]
+    nvh = HYBRID_TINY["linear_num_value_heads"]
+    nkh = HYBRID_TINY["linear_num_key_heads"]
+    dk = HYBRID_TINY["linear_key_head_dim"]
+    dv = HYBRID_TINY["linear_value_head_dim"]
+    K = HYBRID_TINY["linear_conv_kernel_dim"]
+    key_dim, value_dim = nkh * dk, nvh * dv
+    conv_dim = 2 * key_dim + value_dim
+
+    def rnd(*s):
+        return torch.randn(*s, generator=g) * 0.05
+
+    sd = {"model.embed_tokens.weight": rnd(V, H),
+          "model.norm.weight": torch.ones(H) + rnd(H) * 0.01,
+          "lm_head.weight": rnd(V, H)}
+    for L in range(HYBRID_TINY["num_hidden_layers"]):
+        p = f"model.layers.{L}."
+        if (L + 1) % HYBRID_TINY["full_attention_interval"]:
+            la = p + "linear_attn."
+            sd[la + "in_proj_qkvz.weight"] = rnd(
+                2 * key_dim + 2 * value_dim, H)
+            sd[la + "in_proj_ba.weight"] = rnd(2 * nvh, H)
+            sd[la + "conv1d.weight"] = rnd(conv_dim, 1, K)
+            sd[la + "A_log"] = rnd(nvh).abs()
+            sd[la + "dt_bias"] = rnd(nvh)
+            sd[la + "norm.weight"] = torch.ones(dv) + rnd(dv) * 0.01
+            sd[la + "out_proj.weight"] = rnd(H, value_dim)
+        else:
+            sa = p + "self_attn."
+            sd[sa + "q_proj.weight"] = rnd(nh * 2 * hd, H)  # q|gate
+            sd[sa + "k_proj.weight"] = rnd(nkv * hd, H)
+            sd[sa + "v_proj.weight"] = rnd(nkv * hd, H)
+            sd[sa + "o_proj.weight"] = rnd(H, nh * hd)
+            sd[sa + "q_norm.weight"] = torch.ones(hd)
+            sd[sa + "k_norm.weight"] = torch.ones(hd)
+        sd[p + "mlp.gate_proj.weight"] = rnd(I, H)
+        sd[p + "mlp.up_proj.weight"] = rnd(I, H)
+        sd[p + "mlp.down_proj.weight"] = rnd(H, I)
+        sd[p + "input_layernorm.weight"] = torch.ones(H)
+        sd[p + "post_attention_layernorm.weight"] = torch.ones(H)
+    from safetensors.torch import save_file
+    save_file(sd, os.path.join(str(d), "model.safetensors"))
+
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    from gllm_amd.sequence import SamplingParams
+    cfg = EngineConfig(model=str(d), load_format="auto", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       enable_prefix_caching=False)
+    llm = LLM(config=cfg, num_pages_override=128)
+    la0 = llm.runner.model.layers[0].linear_attn
+    assert torch.allclose(la0.in_proj_qkvz.weight,
+                          sd["model.layers.0.linear_attn"
+                             ".in_proj_qkvz.weight"])
+    assert torch.allclose(la0.A_log,
+                          sd["model.layers.0.linear_attn.A_log"])
+    assert torch.allclose(
+        la0.conv1d_weight,
+        sd["model.layers.0.linear_attn.conv1d.weight"].reshape(
+            conv_dim, K))
+    at1 = llm.runner.model.layers[1].self_attn
+    assert torch.allclose(
+        at1.qkv_proj.weight[:nh * 2 * hd],
+        sd["model.layers.1.self_attn.q_proj.weight"])
+    sp = [SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)]
+    o1 = llm.generate([[1, 2, 3, 4, 5, 6, 7]], sp)[0].token_ids
+    o2 = llm.generate([[1, 2, 3, 4, 5, 6, 7]], sp)[0].token_ids
+    assert len(o1) == 5 and o1 == o2
