@@ -346,6 +346,12 @@ def make_arg_parser():
                         "e.g. '0,1' (master/slave modes)")
     p.add_argument("--relay-port", type=int, default=None,
                    help="control-plane TCP port (default master_port+1)")
+    # encoder disaggregation (disagg/)
+    p.add_argument("--mm-encoder-addr", type=str, default=None,
+                   help="remote vision-encoder 'host:port'")
+    p.add_argument("--discovery-addr", type=str, default=None,
+                   help="discovery server 'host:port' to resolve an "
+                        "encoder from")
     p.add_argument("--seed", type=int, default=0)
     return p
 
@@ -365,7 +371,9 @@ def config_from_args(args) -> EngineConfig:
         launch_mode=args.launch_mode,
         worker_ranks=[int(r) for r in args.worker_ranks.split(",")]
         if args.worker_ranks else None,
-        relay_port=args.relay_port, seed=args.seed,
+        relay_port=args.relay_port,
+        mm_encoder_addr=args.mm_encoder_addr,
+        discovery_addr=args.discovery_addr, seed=args.seed,
         device="cuda" if _has_gpu() else "cpu")
 
 
