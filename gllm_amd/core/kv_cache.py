@@ -143,13 +143,26 @@ class PrefixMemoryManager(MemoryManager):
             self._intern[key] = cid
         return cid
 
+    def _key_ids(self, seq: Sequence, lo: int, hi: int) -> Tuple[int, ...]:
+        """Token ids for cache KEYS. Multimodal prompts substitute each
+        image-pad run with content-hash pseudo-ids
+        (multimodal/prepare.py builds seq.cache_key_ids) so two prompts
+        with identical text but different pixels never alias a page."""
+        base = seq.cache_key_ids
+        if base is None:
+            return tuple(seq.token_ids[lo:hi])
+        if hi <= len(base):
+            return tuple(base[lo:hi])
+        mixed = list(base) + list(seq.token_ids[len(base):hi])
+        return tuple(mixed[lo:hi])
+
     def _chains_up_to(self, seq: Sequence, n_full: int) -> List[int]:
         """Extend seq.page_hashes to cover the first ``n_full`` pages."""
         chains = seq.page_hashes
         parent = chains[-1] if chains else 0
         for i in range(len(chains), n_full):
-            toks = tuple(
-                seq.token_ids[i * self.page_size:(i + 1) * self.page_size])
+            toks = self._key_ids(seq, i * self.page_size,
+                                 (i + 1) * self.page_size)
             parent = self._chain_id(parent, toks)
             chains.append(parent)
         return chains[:n_full]

@@ -46,6 +46,26 @@ def prepare_mm_seq(model, seq: Sequence, mm_input: dict) -> None:
         assert sum(n for _, n in spans) == seq.mm_embeds.shape[0], \
             (spans, seq.mm_embeds.shape)
     seq.mm_spans = spans
+    # prefix-cache keys: substitute each image run's pad tokens with
+    # content-derived pseudo-ids so identical text + different pixels
+    # never alias a cached page (core/kv_cache.py _key_ids). crc-free
+    # sha digest: deterministic across processes/ranks.
+    import hashlib
+    src = mm_input.get("pixel_values",
+                       mm_input.get("embeds"))
+    if src is not None and spans:
+        keys = list(toks)
+        digest = hashlib.sha256()
+        digest.update(repr([tuple(g) for g in grids]).encode())
+        digest.update(src.detach().to(torch.float32).cpu().numpy()
+                      .tobytes())
+        h = int.from_bytes(digest.digest()[:8], "little")
+        for si, (s, n) in enumerate(spans):
+            base = h ^ (si * 0x9E3779B97F4A7C15 & (1 << 63) - 1)
+            for j in range(n):
+                keys[s + j] = -(
+                    (base + j * 0x100000001B3) & ((1 << 62) - 1)) - 10
+        seq.cache_key_ids = keys
     if getattr(model, "uses_mrope", False):
         from gllm_amd.layers.mrope import MRotaryEmbedding
         pos, delta = MRotaryEmbedding.get_input_positions(

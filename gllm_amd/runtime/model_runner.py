@@ -53,15 +53,10 @@ class ModelRunner:
             cfg.use_graph = False
         self.uses_mrope = bool(getattr(self.model, "uses_mrope", False))
         if self.uses_mrope:
-            # [3,B] graph position buffers + mm content hashing: round 2
+            # [3,B] graph position buffers are round 2; prefix caching
+            # stays ON — image runs get content-hash cache keys
+            # (multimodal/prepare.py + core/kv_cache.py _key_ids)
             cfg.use_graph = False
-            cfg.enable_prefix_caching = False
-        elif getattr(self.model, "visual", None) is not None or \
-                getattr(self.model, "vision_tower", None) is not None:
-            # 1-D-position multimodal (Kimi-K2.5): page reuse keyed on
-            # token ids alone would alias different images' pad runs
-            # (mm content hashing is round 2)
-            cfg.enable_prefix_caching = False
         self.index_head_dim = getattr(self.model, "index_head_dim", None)
         if self.index_head_dim:
             # DSA selector round-1 runs the eager per-seq torch path;

@@ -60,6 +60,10 @@ class Sequence:
         self.num_cached_pages: int = 0
         # per-page chained hashes (prefix cache keys), computed lazily
         self.page_hashes: List[int] = []
+        # multimodal: prompt-length token list with image-pad runs
+        # replaced by content-hash pseudo-ids — the prefix cache keys
+        # (core/kv_cache.py _key_ids); None = text-only
+        self.cache_key_ids: Optional[List[int]] = None
 
         # --- output state ---
         self.finish_reason: Optional[str] = None

@@ -65,7 +65,8 @@ def test_kimi_generate_and_image_sensitivity(tmp_path):
     from gllm_amd.sequence import SamplingParams
     llm = _mk_llm(tmp_path)
     from gllm_amd.core.kv_cache import PrefixMemoryManager
-    assert not isinstance(llm.runner.memory_manager, PrefixMemoryManager)
+    # prefix caching stays on: image runs get content-hash cache keys
+    assert isinstance(llm.runner.memory_manager, PrefixMemoryManager)
     assert not llm.runner.uses_mrope
     toks, mm = _mm()
     sp = [SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)]

@@ -131,3 +131,29 @@ def test_vl25_generate_with_windowed_tower(tmp_path):
     o1 = llm.generate([toks], sp, mm_inputs=[mm])[0].token_ids
     o2 = llm.generate([toks], sp, mm_inputs=[mm])[0].token_ids
     assert len(o1) == 5 and o1 == o2
+
+
+def test_mm_prefix_cache_content_keys(tmp_path):
+    """Prefix caching with multimodal prompts: identical image reuses
+    pages (hit rate > 0, same output); a different image with the SAME
+    token ids must NOT alias the cached pages."""
+    from gllm_amd.sequence import SamplingParams
+    from gllm_amd.core.kv_cache import PrefixMemoryManager
+    llm = _mk_llm(tmp_path, name="vlpfx")
+    assert isinstance(llm.runner.memory_manager, PrefixMemoryManager)
+    # long text prefix so image pads land beyond the first pages too
+    toks = list(range(1, 10)) + [150] * 6 + [3, 4]
+    torch.manual_seed(41)
+    mm1 = {"pixel_values": torch.randn(24, 3 * 2 * 14 * 14),
+           "grids": [(1, 4, 6)]}
+    mm2 = {"pixel_values": torch.randn(24, 3 * 2 * 14 * 14) * 2,
+           "grids": [(1, 4, 6)]}
+    sp = [SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)]
+    o1 = llm.generate([toks], sp, mm_inputs=[mm1])[0].token_ids
+    o1b = llm.generate([toks], sp, mm_inputs=[mm1])[0].token_ids
+    assert o1b == o1
+    assert llm.runner.memory_manager.hit_tokens > 0, \
+        "identical mm request must hit the prefix cache"
+    o2 = llm.generate([toks], sp, mm_inputs=[mm2])[0].token_ids
+    assert o2 != o1, \
+        "different pixels with identical token ids must not alias pages"
