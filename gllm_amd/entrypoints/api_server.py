@@ -97,6 +97,12 @@ def build_app():
         rp = req.repetition_penalty
         if rp is None:
             rp = 1.0
+        if getattr(req, "prompt_logprobs", None) and \
+                engine.config.pp_size > 1:
+            # the values are computed where sampling runs (last stage);
+            # shipping them across stages is a round-2 item
+            raise ValueError(
+                "prompt_logprobs requires pp_size == 1")
         return SamplingParams(
             temperature=req.temperature if req.temperature is not None
             else 1.0,
@@ -136,7 +142,11 @@ def build_app():
         except Exception as e:
             return JSONResponse(status_code=400,
                                 content={"error": str(e)})
-        sampling = _sampling_from(req)
+        try:
+            sampling = _sampling_from(req)
+        except ValueError as e:
+            return JSONResponse(status_code=400,
+                                content={"error": str(e)})
         if req.stream:
             return StreamingResponse(
                 _chat_stream(req, raw, token_ids, sampling, mm=mm),
@@ -261,7 +271,11 @@ def build_app():
             prompts = [prompts]
         elif prompts and isinstance(prompts[0], int):
             prompts = [prompts]
-        sampling = _sampling_from(req, default_max=16)
+        try:
+            sampling = _sampling_from(req, default_max=16)
+        except ValueError as e:
+            return JSONResponse(status_code=400,
+                                content={"error": str(e)})
         if req.stream:
             token_ids = prompts[0] if isinstance(prompts[0], list) \
                 else engine.encode(prompts[0])

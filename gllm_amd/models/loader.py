@@ -132,7 +132,12 @@ def iterate_safetensors(model_path: str
             sd = torch.load(b, map_location="cpu", weights_only=True)
             yield from sd.items()
         return
-    for shard in shards:
+    for i, shard in enumerate(shards):
+        # load-progress reporting (reference llm_engine.py:324-347 uses
+        # shared mp arrays + a bar; per-shard logs serve the same
+        # purpose across worker processes)
+        logger.info("loading weights: shard %d/%d (%s)", i + 1,
+                    len(shards), os.path.basename(shard))
         with safe_open(shard, framework="pt", device="cpu") as f:
             for name in f.keys():
                 yield name, f.get_tensor(name)
