@@ -132,6 +132,11 @@ class PrefixMemoryManager(MemoryManager):
         # stats
         self.lookup_tokens = 0
         self.hit_tokens = 0
+        # hybrid-SSM hooks (runtime/model_runner.py): hit_filter trims a
+        # hit to the deepest boundary whose recurrent-state snapshot
+        # exists; on_register(seq, chains, n_full) takes snapshots
+        self.hit_filter = None
+        self.on_register = None
 
     # ---- interning ----
     def _chain_id(self, parent: int, tokens: Tuple[int, ...]) -> int:
@@ -187,6 +192,9 @@ class PrefixMemoryManager(MemoryManager):
         # forward pass produces a logit row (reference memory_manager.py:992).
         while hit_pages and len(hit_pages) * self.page_size >= seq.prompt_len:
             hit_pages.pop()
+        if self.hit_filter is not None and hit_pages:
+            keep = self.hit_filter(chains[:len(hit_pages)])
+            hit_pages = hit_pages[:keep]
         for page in hit_pages:
             if self.page_ref[page] == 0:
                 # resurrect from the evictable pool
@@ -240,6 +248,8 @@ class PrefixMemoryManager(MemoryManager):
                 self.page_chain[page] = cid
             # last writer wins; identical content either way
             self.chain_to_page[cid] = page
+        if self.on_register is not None and n_full:
+            self.on_register(seq, chains, n_full)
 
     def free_seq(self, seq: Sequence) -> None:
         for p in seq.page_table:

@@ -415,6 +415,11 @@ class Scheduler:
                 if isinstance(self.mm, PrefixMemoryManager):
                     self.mm.register_computed_pages(seq)
                 self.seqs_to_decode.append(seq)
+            elif isinstance(self.mm, PrefixMemoryManager):
+                # mid-prompt chunks publish their full pages too:
+                # concurrent identical prompts share, and hybrid models
+                # snapshot recurrent state at page-aligned chunk ends
+                self.mm.register_computed_pages(seq)
             # chunks that don't end the prompt stay in seqs_to_prefill
             # (they were parked there at schedule time)
         return finished

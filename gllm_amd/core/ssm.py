@@ -1,13 +1,19 @@
-"""SSM working-state pool for hybrid linear-attention models.
+"""SSM working-state + snapshot pools for hybrid linear-attention
+models.
 
-Parity: reference SSMSegment working pool (memory_manager.py:87-256) —
-per-layer conv_state + recurrent-state tensors addressed by a per-seq
-slot. Snapshot pools (prefix-cache state restore) are a round-2 item;
-round 1 disables prefix caching for hybrid models instead.
+Parity: reference SSMSegment (memory_manager.py:87-256) — per-layer
+conv_state + recurrent-state tensors addressed by a per-seq slot, plus
+SNAPSHOT storage keyed by prefix-cache chain id: when a sequence's
+computed length lands exactly on a page boundary at registration time,
+its recurrent state is cloned under that boundary's chain id; a later
+prefix hit restores it (the KV pages alone are not enough for hybrid
+models — the linear-attention layers carry state). The prefix
+manager's hit_filter trims hits to the deepest snapshotted boundary.
 """
 
+import collections
 import dataclasses
-from typing import List
+from typing import List, Optional
 
 import torch
 
@@ -44,9 +50,40 @@ class SSMPool:
             return seq.ssm_slot
         slot = self.alloc.allocate()
         seq.ssm_slot = slot
+        seq.ssm_state_ready = False
         return slot
 
     def free(self, seq: Sequence) -> None:
         if seq.ssm_slot >= 0:
             self.alloc.free(seq.ssm_slot)
             seq.ssm_slot = -1
+            seq.ssm_state_ready = False
+
+    # ---- snapshots (prefix-cache state restore) ----
+    max_snapshots = 64
+
+    @property
+    def snapshots(self):
+        if not hasattr(self, "_snapshots"):
+            self._snapshots = collections.OrderedDict()
+        return self._snapshots
+
+    def has_snapshot(self, chain_id: int) -> bool:
+        return chain_id in self.snapshots
+
+    def snapshot(self, chain_id: int, slot: int) -> None:
+        snaps = self.snapshots
+        snaps[chain_id] = (self.conv_state[:, slot].clone(),
+                           self.ssm_state[:, slot].clone())
+        snaps.move_to_end(chain_id)
+        while len(snaps) > self.max_snapshots:
+            snaps.popitem(last=False)
+
+    def restore(self, chain_id: int, slot: int) -> bool:
+        snap = self.snapshots.get(chain_id)
+        if snap is None:
+            return False
+        self.snapshots.move_to_end(chain_id)
+        self.conv_state[:, slot].copy_(snap[0])
+        self.ssm_state[:, slot].copy_(snap[1])
+        return True

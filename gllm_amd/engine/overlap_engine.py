@@ -25,6 +25,10 @@ class OverlapEngine(PPEngine):
                  num_pages_override: Optional[int] = None):
         super().__init__(config, num_pages_override=num_pages_override)
         assert self.pp_size == 1, "overlap engine is the PP=1 fast path"
+        # deferred finalize registers pages AFTER later batches may have
+        # advanced the recurrent state past the boundary — snapshots
+        # would be stale, so hybrid models skip them under overlap
+        self.runner.ssm_snapshot_enabled = False
         self.maxd = config.maxd
         self.ring_slots = self.runner.ring_slots
         self._slot = 0

@@ -46,10 +46,9 @@ class ModelRunner:
         self.model, self.hf_config = load_model(cfg, cfg.device)
         self.kv_dtype = cfg.torch_dtype()
         if getattr(self.model, "ssm_spec", None) is not None:
-            if cfg.enable_prefix_caching:
-                logger.info("hybrid GDN model: disabling prefix caching "
-                            "(state snapshots land in round 2)")
-                cfg.enable_prefix_caching = False
+            # prefix caching stays ON: page-boundary recurrent-state
+            # snapshots restore on hits (core/ssm.py + the hit_filter
+            # wired below); graphs are the round-2 item
             cfg.use_graph = False
         self.uses_mrope = bool(getattr(self.model, "uses_mrope", False))
         if self.uses_mrope:
@@ -84,11 +83,35 @@ class ModelRunner:
             cfg.device)
         self.memory_manager.free_hooks.append(self.penalty_pool.free)
         self.ssm_pool = None
+        self.ssm_snapshot_enabled = True
         if getattr(self.model, "ssm_spec", None) is not None:
             from gllm_amd.core.ssm import SSMPool
             self.ssm_pool = SSMPool(self.model.ssm_spec, cfg.maxd + 64,
                                     cfg.device, dtype=self.kv_dtype)
             self.memory_manager.free_hooks.append(self.ssm_pool.free)
+            if isinstance(self.memory_manager, PrefixMemoryManager):
+                # hybrid prefix caching: hits only at boundaries whose
+                # recurrent state was snapshotted; snapshots taken when
+                # registration lands exactly on a page boundary
+                pool = self.ssm_pool
+                page = cfg.page_size
+
+                def hit_filter(chains):
+                    for k in range(len(chains), 0, -1):
+                        if pool.has_snapshot(chains[k - 1]):
+                            return k
+                    return 0
+
+                def on_register(seq, chains, n_full):
+                    if not self.ssm_snapshot_enabled:
+                        return
+                    if seq.ssm_slot < 0 or not seq.ssm_state_ready:
+                        return
+                    if seq.computed_token_num == n_full * page:
+                        pool.snapshot(chains[n_full - 1], seq.ssm_slot)
+
+                self.memory_manager.hit_filter = hit_filter
+                self.memory_manager.on_register = on_register
         from gllm_amd.parallel import get_pp_size, get_tp_size
         if (cfg.use_graph and cfg.device.startswith("cuda")
                 and get_pp_size() == 1 and get_tp_size() == 1):
@@ -275,9 +298,22 @@ class ModelRunner:
         if self.ssm_pool is None:
             return
         fctx.ssm_pool = self.ssm_pool
-        fctx.ssm_slots = [self.ssm_pool.ensure(it.seq)
-                          for it in batch.items]
-        fctx.ssm_has_init = [it.start > 0 for it in batch.items]
+        slots, has_init = [], []
+        page = self.config.page_size
+        for it in batch.items:
+            seq = it.seq
+            slot = self.ssm_pool.ensure(seq)
+            if it.start > 0 and not seq.ssm_state_ready:
+                # prefix-cache hit on a fresh slot: restore the boundary
+                # snapshot (the hit_filter guaranteed it exists)
+                chain = seq.page_hashes[it.start // page - 1]
+                ok = self.ssm_pool.restore(chain, slot)
+                assert ok, "prefix hit without an SSM snapshot"
+            seq.ssm_state_ready = True
+            slots.append(slot)
+            has_init.append(it.start > 0)
+        fctx.ssm_slots = slots
+        fctx.ssm_has_init = has_init
 
     def _attach_mm(self, batch, fctx) -> None:
         """Collect vision-embedding rows for this batch's prefill chunks

@@ -76,6 +76,9 @@ class Sequence:
         # repetition-penalty mask slot / SSM slot (set by managers when used)
         self.penalty_slot: int = -1
         self.ssm_slot: int = -1
+        # slot state matches computed_token_num (False on a fresh slot
+        # whose prefix-hit state still needs a snapshot restore)
+        self.ssm_state_ready: bool = False
         # per-output-token logprobs (filled when sampling.logprobs is set):
         # list of (chosen_logprob, {token_id: logprob} top-k)
         self.out_logprobs: List[tuple] = []
