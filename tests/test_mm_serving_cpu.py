@@ -146,3 +146,35 @@ def test_mm_chat_completion_roundtrip(tmp_path):
         assert r.status_code == 200, r.text
     finally:
         srv.engine.stop()
+
+
+def test_vit_output_lru_cache(tmp_path):
+    """Repeated image content skips the tower via the content-hash LRU
+    (reference MultiModalEmbeddingCache)."""
+    import json as _json
+    import torch
+    from gllm_amd import multimodal
+    from gllm_amd.multimodal import prepare as P
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    from gllm_amd.sequence import SamplingParams
+
+    d = tmp_path / "vlc"
+    d.mkdir()
+    cfg_json = {**VL_CFG}
+    with open(d / "config.json", "w") as f:
+        _json.dump(cfg_json, f)
+    cfg = EngineConfig(model=str(d), load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       enable_prefix_caching=False)
+    llm = LLM(config=cfg, num_pages_override=128)
+    torch.manual_seed(0)
+    px = torch.randn(16, 3 * 2 * 14 * 14)
+    mm = {"pixel_values": px, "grids": [(1, 4, 4)]}
+    toks = [2, 3] + [150] * 4 + [4]
+    sp = [SamplingParams(temperature=0.0, max_tokens=3, ignore_eos=True)]
+    before = P.emb_cache_hits
+    o1 = llm.generate([toks], sp, mm_inputs=[mm])[0].token_ids
+    o2 = llm.generate([toks], sp, mm_inputs=[mm])[0].token_ids
+    assert P.emb_cache_hits > before
+    assert o1 == o2
