@@ -55,6 +55,8 @@ class EmbeddingCache:
 class EncoderServer:
     def __init__(self, config, host: str = "0.0.0.0", port: int = 29820,
                  cache_items: int = 256):
+        import os
+
         from gllm_amd.models.loader import load_model
         from gllm_amd.parallel import init_distributed
         init_distributed(pp_size=1, dp_size=1, tp_size=1)
@@ -65,6 +67,12 @@ class EncoderServer:
             "model has no vision tower"
         self.cache = EmbeddingCache(cache_items)
         self.port = port
+        # fault injection (reference GLLM_ENC_FAIL_FIRST_N,
+        # disagg/config.py:13-18): fail the first N real encode jobs so
+        # the client's redispatch path can be exercised
+        self._fail_first_n = int(os.environ.get("GLLM_ENC_FAIL_FIRST_N",
+                                                "0"))
+        self._jobs_seen = 0
         self._srv = socket.create_server((host, port))
         self._srv.settimeout(0.5)
         self._stop = threading.Event()
@@ -111,6 +119,12 @@ class EncoderServer:
             return EncoderResult(job.job_id, emb, cached=True)
         if job.pixel_values is None:
             return EncoderResult(job.job_id, None)  # probe miss
+        self._jobs_seen += 1
+        if self._jobs_seen <= self._fail_first_n:
+            return EncoderResult(job.job_id, None,
+                                 error="injected failure "
+                                       f"({self._jobs_seen}/"
+                                       f"{self._fail_first_n})")
         try:
             with torch.no_grad():
                 emb = self.model.encode_images(job.pixel_values,
@@ -123,30 +137,71 @@ class EncoderServer:
 
 
 class EncoderClient:
-    """LM-frontend side: probe-by-hash then send pixels on miss."""
+    """LM-frontend side: probe-by-hash then send pixels on miss.
 
-    def __init__(self, addr: str):
-        host, port = addr.rsplit(":", 1)
-        self._sock = socket.create_connection((host, int(port)),
-                                              timeout=60)
+    Failure handling (reference DisaggCoordinator watchdog +
+    GLLM_DISAGG_MAX_REDISPATCH): a failed or dropped job is re-sent up
+    to ``max_redispatch`` times (reconnecting on a dead socket) before
+    the request is surfaced as an error — the API layer turns that into
+    a 400/abort, never a hung sequence."""
+
+    def __init__(self, addr: str, max_redispatch: int = None):
+        import os
+        self._addr = addr
+        self._sock = None
+        self._connect()
         self._lock = threading.Lock()
         self._job_id = 0
+        self.max_redispatch = max_redispatch if max_redispatch is not None \
+            else int(os.environ.get("GLLM_DISAGG_MAX_REDISPATCH", "2"))
+
+    def _connect(self):
+        host, port = self._addr.rsplit(":", 1)
+        if self._sock is not None:
+            try:
+                self._sock.close()
+            except OSError:
+                pass
+        self._sock = socket.create_connection((host, int(port)),
+                                              timeout=60)
+
+    def _attempt(self, key, grids, pixel_values):
+        self._job_id += 1
+        jid = self._job_id
+        send_msg(self._sock, EncoderJob(jid, key, list(grids)))
+        res: EncoderResult = recv_msg(self._sock)
+        if res is None:
+            raise ConnectionError("encoder connection closed")
+        if res.embeds is None and res.error is None:
+            send_msg(self._sock,
+                     EncoderJob(jid, key, list(grids), pixel_values))
+            res = recv_msg(self._sock)
+            if res is None:
+                raise ConnectionError("encoder connection closed")
+        if res.error:
+            raise RuntimeError(f"encoder job failed: {res.error}")
+        return res.embeds
 
     def encode(self, pixel_values: torch.Tensor, grids) -> torch.Tensor:
         from gllm_amd.disagg.protocol import content_hash
         key = content_hash(pixel_values, grids)
+        last = None
         with self._lock:
-            self._job_id += 1
-            jid = self._job_id
-            send_msg(self._sock, EncoderJob(jid, key, list(grids)))
-            res: EncoderResult = recv_msg(self._sock)
-            if res.embeds is None and res.error is None:
-                send_msg(self._sock,
-                         EncoderJob(jid, key, list(grids), pixel_values))
-                res = recv_msg(self._sock)
-        if res.error:
-            raise RuntimeError(f"encoder job failed: {res.error}")
-        return res.embeds
+            for attempt in range(self.max_redispatch + 1):
+                try:
+                    return self._attempt(key, grids, pixel_values)
+                except ConnectionError as e:
+                    last = e
+                    logger.warning("encoder redispatch %d/%d: %s",
+                                   attempt + 1, self.max_redispatch, e)
+                    self._connect()
+                except RuntimeError as e:
+                    last = e
+                    logger.warning("encoder redispatch %d/%d: %s",
+                                   attempt + 1, self.max_redispatch, e)
+        raise RuntimeError(
+            f"encoder job failed after {self.max_redispatch + 1} "
+            f"attempts: {last}")
 
     def stats(self) -> dict:
         with self._lock:
