@@ -9,9 +9,11 @@ slots (core/ssm.py); full_attention layers use the paged KV cache with
 DENSE kv-layer indices (a 24-layer stack with 6 softmax layers allocates
 6 KV layers, reference qwen3_5.py header notes).
 
-Round-1 compute path: torch GDN ops (ops/gdn_ref.py — also the oracle
-for the round-2 chunked HIP kernels). Prefix caching is disabled for
-hybrid models (state snapshots are round 2).
+Compute path: torch GDN ops (ops/gdn_ref.py — chunk-parallel WY-form
+prefill + sequential decode; also the oracle for the round-2 HIP
+kernels). Prefix caching is ON: recurrent state snapshots at page
+boundaries restore on hits (core/ssm.py). Checkpoint loading covers
+the fused in_proj_qkvz/ba layout.
 """
 
 from typing import Iterable, List, Tuple

@@ -1,13 +1,13 @@
 """Qwen2-VL: vision tower + mrope language model.
 
 Parity target: reference models/qwen2_5_vl.py (ViT tower, mrope,
-embed_multimodal merge). Round-1 scope: the OFFLINE path — the engine
-accepts pixel patches + grids (or precomputed embeddings) per request,
-runs the tower at admission, and merges embeddings at the image-pad
-rows; prefix caching and hipGraphs are disabled for mrope models
-(content-hash mm caching and [3,B] graph position buffers are round 2,
-as is the server-side image processor).
-"""
+embed_multimodal merge). Full path: offline (pixel patches + grids or
+precomputed embeddings per request) AND serving (native image
+processor + OpenAI image_url content, docs/multimodal.md); the tower
+runs at admission (ViT-output LRU) or in a disaggregated encoder
+process; embeddings merge at the image-pad rows chunk-aware; prefix
+caching uses content-hash keys for image runs. hipGraphs stay off for
+mrope models ([3,B] graph position buffers are round 2)."""
 
 from typing import Iterable, Tuple
 

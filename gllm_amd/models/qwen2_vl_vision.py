@@ -6,9 +6,11 @@ Conv3d patch embed (temporal_patch_size x patch x patch), blocks of
 QuickGELU MLP], and the 2x2 PatchMerger. Full (per-image) attention via
 varlen SDPA over cu_seqlens.
 
-Round-1 status: standalone tested tower; the LM-side embedding merge +
-[3, T] mrope plumbing through the batch builder is the round-2
-multimodal milestone (docs/architecture.md).
+Wired into the full multimodal path (docs/multimodal.md): LM-side
+embedding merge, [3, T] mrope through the batch builder, server-side
+image processor, encoder disaggregation. The 2.5-VL windowed variant
+(Qwen25VisionTransformer) and the Qwen3-VL deepstack tower
+(models/qwen3_vl_vision.py) build on the same pieces.
 """
 
 from typing import List, Tuple
