@@ -48,6 +48,11 @@ class OverlapEngine(PPEngine):
         slot = self._slot
         self._slot = (self._slot + 1) % self.ring_slots
         out = self.runner.step_first_stage(batch)   # SamplerOutput, async
+        if out.logprobs is not None:
+            # a logprob-requesting batch syncs here (.tolist()) — the
+            # overlap win is traded for the feature on those requests
+            from gllm_amd.engine.pp_engine import PPEngine
+            PPEngine._stash_logprobs(batch, out)
         B = len(batch.items)
         ring = self.runner.token_ring
         ring[slot, :B].copy_(out.next_tokens)

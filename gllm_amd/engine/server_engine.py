@@ -31,6 +31,8 @@ class StreamChunk:
     n_output_tokens: int
     # attached to the FINAL chunk when sampling.prompt_logprobs was set
     prompt_logprobs: Optional[list] = None
+    # (chosen_logprob, {token_id: logprob}) when sampling.logprobs
+    logprob: Optional[tuple] = None
 
 
 class RequestState:
@@ -205,15 +207,17 @@ class AsyncLLMEngine:
                 return
             if kind != "out":
                 continue
-            self.token_counter += sum(1 for _, t, _f in outs if t >= 0)
-            for seq_id, token_id, finish in outs:
+            self.token_counter += sum(1 for o in outs if o[1] >= 0)
+            for o in outs:
+                seq_id, token_id, finish = o[0], o[1], o[2]
+                lp = o[3] if len(o) > 3 else None
                 st = self.requests.get(seq_id)
                 if st is None or st.finished:
                     continue
-                self._handle_token(st, token_id, finish)
+                self._handle_token(st, token_id, finish, lp)
 
     def _handle_token(self, st: RequestState, token_id: int,
-                      finish: Optional[str]) -> None:
+                      finish: Optional[str], lp=None) -> None:
         if st.first_token_time is None:
             st.first_token_time = time.time()
         text = ""
@@ -230,7 +234,8 @@ class AsyncLLMEngine:
             st.finished = True
         chunk = StreamChunk(st.seq_id, token_id, text, finish, st.n_tokens,
                             prompt_logprobs=st.prompt_logprobs
-                            if finish is not None else None)
+                            if finish is not None else None,
+                            logprob=lp)
         st.loop.call_soon_threadsafe(st.queue.put_nowait, chunk)
         if st.finished:
             st.loop.call_soon_threadsafe(st.queue.put_nowait, None)

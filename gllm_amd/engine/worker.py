@@ -53,6 +53,7 @@ class ServingMixin:
             self.is_output_rank = self.rank == 0
         self._req_counter = 0  # deterministic DP round-robin routing
         self._seqs_by_id = {}  # live seqs (for prompt-logprob emission)
+        self._emit_counts = {}  # seq_id -> tokens emitted (logprob idx)
         self.comm = WorkerComm(req_queue, out_queue, self.is_output_rank)
         self._intake_buf = torch.zeros(1, dtype=torch.int64)
         if self.world > 1 and config.device.startswith("cuda"):
@@ -130,6 +131,17 @@ class ServingMixin:
             if n:
                 self._apply_messages(self.comm.recv_blocking(n))
 
+    def _token_logprob(self, seq: Sequence):
+        """The just-emitted token's (logprob, {top-k}) when requested
+        (reference sampler logprobs served per stream chunk)."""
+        if not seq.sampling.logprobs or not seq.out_logprobs:
+            return None
+        idx = self._emit_counts.get(seq.seq_id, 0)
+        self._emit_counts[seq.seq_id] = idx + 1
+        if idx < len(seq.out_logprobs):
+            return seq.out_logprobs[idx]
+        return None
+
     def _emit_prompt_logprobs(self, seq: Sequence) -> None:
         """Ship prompt logprobs BEFORE the finish token so the frontend
         attaches them to the final stream chunk."""
@@ -180,16 +192,18 @@ class ServingWorker(ServingMixin, PPEngine):
                     continue
                 seq = item.seq
                 if seq.finish_reason == Sequence.FINISH_ABORT:
-                    outs.append((seq.seq_id, -1, seq.finish_reason))
+                    outs.append((seq.seq_id, -1, seq.finish_reason, None))
                 else:
                     outs.append((seq.seq_id, seq.token_ids[-1],
-                                 seq.finish_reason))
+                                 seq.finish_reason,
+                                 self._token_logprob(seq)))
                 if seq.finish_reason:
                     self._emit_prompt_logprobs(seq)
             if outs:
                 self.comm.send_output(("out", outs, {}))
         for s in finished:
             self._seqs_by_id.pop(s.seq_id, None)
+            self._emit_counts.pop(s.seq_id, None)
         return finished
 
     def _run_loop_dp(self) -> None:
@@ -249,13 +263,18 @@ class OverlapServingWorker(ServingMixin, OverlapEngine):
             self.on_finalized = self._emit
 
     def _emit(self, emissions) -> None:
-        emissions = list(emissions)
-        for seq_id, _tok, fin in emissions:
+        out = []
+        for seq_id, tok, fin in emissions:
+            seq = self._seqs_by_id.get(seq_id)
+            lp = self._token_logprob(seq) if seq is not None and \
+                tok >= 0 else None
+            out.append((seq_id, tok, fin, lp))
             if fin:
-                seq = self._seqs_by_id.pop(seq_id, None)
+                self._seqs_by_id.pop(seq_id, None)
+                self._emit_counts.pop(seq_id, None)
                 if seq is not None:
                     self._emit_prompt_logprobs(seq)
-        self.comm.send_output(("out", emissions, {}))
+        self.comm.send_output(("out", out, {}))
 
     def run_loop(self) -> None:
         logger.info("worker %d ready (overlap)", self.rank)

@@ -103,17 +103,22 @@ def build_app():
             # shipping them across stages is a round-2 item
             raise ValueError(
                 "prompt_logprobs requires pp_size == 1")
+        lp = req.logprobs
+        if isinstance(lp, bool):  # chat: bool + top_logprobs count
+            lp = (getattr(req, "top_logprobs", None) or 1) if lp else None
         return SamplingParams(
             temperature=req.temperature if req.temperature is not None
             else 1.0,
             top_p=req.top_p if req.top_p is not None else 1.0,
             top_k=req.top_k if req.top_k is not None else -1,
+            min_p=getattr(req, "min_p", None) or 0.0,
             repetition_penalty=rp,
             max_tokens=max_tokens,
             min_tokens=req.min_tokens or 0,
             ignore_eos=req.ignore_eos,
             stop=stops, stop_token_ids=req.stop_token_ids,
             seed=req.seed,
+            logprobs=lp,
             prompt_logprobs=getattr(req, "prompt_logprobs", None))
 
     def _vary_seed(sampling: SamplingParams, j: int) -> SamplingParams:
@@ -158,7 +163,7 @@ def build_app():
             for j in range(n)])
         choices = []
         total_out = 0
-        for j, (text, finish, n_out, _plp) in enumerate(results):
+        for j, (text, finish, n_out, _plp, lps) in enumerate(results):
             total_out += n_out
             message = ChatMessage(role="assistant", content=text)
             if req.tools:
@@ -171,7 +176,8 @@ def build_app():
                                           tool_calls=calls)
                     finish = "tool_calls"
             choices.append(ChatCompletionResponseChoice(
-                index=j, message=message, finish_reason=finish))
+                index=j, message=message, finish_reason=finish,
+                logprobs={"content": lps} if lps else None))
         resp = ChatCompletionResponse(
             model=req.model or served_model,
             choices=choices,
@@ -180,11 +186,18 @@ def build_app():
                             total_tokens=len(token_ids) + total_out))
         return resp.model_dump()
 
+    def _lp_entry(chunk):
+        chosen, topk = chunk.logprob
+        return {"token": chunk.text, "logprob": chosen,
+                "top_logprobs": [{"token_id": int(t), "logprob": v}
+                                 for t, v in topk.items()]}
+
     async def _collect(raw, token_ids, sampling, mm=None):
         text_parts = []
         finish = None
         n_out = 0
         plp = None
+        lps = [] if sampling.logprobs else None
         async for chunk in engine.generate_stream(token_ids, sampling,
                                                   mm=mm):
             if await raw.is_disconnected():
@@ -193,10 +206,12 @@ def build_app():
             finish = chunk.finish_reason or finish
             n_out = chunk.n_output_tokens
             plp = chunk.prompt_logprobs or plp
+            if lps is not None and chunk.logprob is not None:
+                lps.append(_lp_entry(chunk))
         text = "".join(text_parts)
         from gllm_amd.engine.detokenizer import check_stop_strings
         _, text = check_stop_strings(text, sampling.stop)
-        return text, finish, n_out, plp
+        return text, finish, n_out, plp, lps
 
     async def _merged_stream(raw, token_ids, sampling, n, mm=None):
         """Run n generations concurrently, yield (choice_idx, chunk) in
@@ -248,7 +263,9 @@ def build_app():
                 model=req.model or served_model,
                 choices=[ChatCompletionStreamChoice(
                     index=j, delta=delta,
-                    finish_reason=chunk.finish_reason)])
+                    finish_reason=chunk.finish_reason,
+                    logprobs={"content": [_lp_entry(chunk)]}
+                    if chunk.logprob is not None else None)])
             if resp_id is None:
                 resp_id = out.id
             else:
@@ -291,11 +308,12 @@ def build_app():
             results = await _aio.gather(*[
                 _collect(raw, token_ids, _vary_seed(sampling, j))
                 for j in range(n)])
-            for j, (text, finish, n_out, plp) in enumerate(results):
+            for j, (text, finish, n_out, plp, lps) in enumerate(results):
                 if req.echo and not isinstance(p, list):
                     text = p + text
                 choices.append(CompletionResponseChoice(
                     index=i * n + j, text=text, finish_reason=finish,
+                    logprobs={"content": lps} if lps else None,
                     prompt_logprobs=plp if req.prompt_logprobs else None))
                 total_c += n_out
             total_p += len(token_ids)

@@ -56,6 +56,7 @@ class ChatCompletionRequest(BaseModel):
     stop_token_ids: Optional[List[int]] = None
     n: int = 1
     seed: Optional[int] = None
+    min_p: Optional[float] = None
     repetition_penalty: Optional[float] = None
     presence_penalty: Optional[float] = 0.0
     frequency_penalty: Optional[float] = 0.0
@@ -81,6 +82,7 @@ class CompletionRequest(BaseModel):
     stop_token_ids: Optional[List[int]] = None
     n: int = 1
     seed: Optional[int] = None
+    min_p: Optional[float] = None
     repetition_penalty: Optional[float] = None
     ignore_eos: bool = False
     echo: bool = False
@@ -98,6 +100,7 @@ class ChatCompletionResponseChoice(BaseModel):
     index: int
     message: ChatMessage
     finish_reason: Optional[str] = None
+    logprobs: Optional[Dict[str, Any]] = None
 
 
 class ChatCompletionResponse(BaseModel):
@@ -119,6 +122,7 @@ class ChatCompletionStreamChoice(BaseModel):
     index: int
     delta: DeltaMessage
     finish_reason: Optional[str] = None
+    logprobs: Optional[Dict[str, Any]] = None
 
 
 class ChatCompletionStreamResponse(BaseModel):

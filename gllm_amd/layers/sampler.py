@@ -21,6 +21,7 @@ class SamplingMetadata:
     penalties: torch.Tensor             # [B] float32
     all_greedy: bool
     any_penalty: bool
+    min_ps: Optional[torch.Tensor] = None  # [B] float32 (0 disabled)
     # token history per row for repetition penalty (cpu LongTensors),
     # used only as the fallback when no mask pool slot is available
     token_id_rows: Optional[List[torch.Tensor]] = None
@@ -56,6 +57,7 @@ class Sampler(torch.nn.Module):
         scaled = logits / temps
         probs = torch.softmax(scaled, dim=-1)
         probs = self._apply_top_k_top_p(probs, meta.top_ks, meta.top_ps)
+        probs = self._apply_min_p(probs, meta.min_ps)
         sampled = self._multinomial(probs, meta)
         # greedy rows override
         greedy_rows = meta.temperatures == 0.0
@@ -87,6 +89,16 @@ class Sampler(torch.nn.Module):
             logits = torch_ref.apply_repetition_penalty(
                 logits, meta.token_id_rows, meta.penalties)
         return logits
+
+    @staticmethod
+    def _apply_min_p(probs: torch.Tensor, min_ps) -> torch.Tensor:
+        """vLLM-style min_p: drop tokens whose probability is below
+        min_p * max_prob of the row (applied after top-k/top-p)."""
+        if min_ps is None or not bool((min_ps > 0).any()):
+            return probs
+        thresh = probs.amax(dim=-1, keepdim=True) * min_ps.unsqueeze(-1)
+        probs = probs.masked_fill(probs < thresh, 0.0)
+        return probs / probs.sum(-1, keepdim=True).clamp_min(1e-20)
 
     @staticmethod
     def _apply_top_k_top_p(probs: torch.Tensor, top_ks: torch.Tensor,
@@ -142,7 +154,7 @@ def build_sampling_metadata(items, device,
     With a ``penalty_pool``, penalized seqs get persistent device mask
     slots (seeded from the prompt on first use, updated on-GPU after
     every sample); the host token-row fallback covers pool exhaustion."""
-    temps, tps, tks, pens, gens = [], [], [], [], []
+    temps, tps, tks, mps, pens, gens = [], [], [], [], [], []
     rows: List[torch.Tensor] = []
     slots: List[int] = []
     sample_rows: List[int] = []
@@ -154,6 +166,7 @@ def build_sampling_metadata(items, device,
         temps.append(sp.temperature)
         tps.append(sp.top_p)
         tks.append(sp.top_k)
+        mps.append(getattr(sp, "min_p", 0.0))
         pens.append(sp.repetition_penalty)
         if it.ends_prompt:
             sample_rows.append(i)
@@ -185,6 +198,7 @@ def build_sampling_metadata(items, device,
         temperatures=torch.tensor(temps, dtype=torch.float32, device=device),
         top_ps=torch.tensor(tps, dtype=torch.float32, device=device),
         top_ks=torch.tensor(tks, dtype=torch.int32, device=device),
+        min_ps=torch.tensor(mps, dtype=torch.float32, device=device),
         penalties=torch.tensor(pens, dtype=torch.float32, device=device),
         all_greedy=all_greedy, any_penalty=any_pen,
         token_id_rows=rows if need_rows else None,

@@ -16,6 +16,7 @@ class SamplingParams:
     temperature: float = 1.0
     top_p: float = 1.0
     top_k: int = -1                    # -1 = disabled
+    min_p: float = 0.0                 # drop p < min_p * max_p (vLLM)
     repetition_penalty: float = 1.0
     max_tokens: int = 512
     min_tokens: int = 0

@@ -118,3 +118,53 @@ def test_api_server_routes(tmp_path):
         assert client.get("/stats").json()["requests_total"] >= 3
     finally:
         srv.engine.stop()
+
+
+@pytest.mark.timeout(300)
+def test_api_token_logprobs(tmp_path):
+    """Per-token logprobs served through chat (bool + top_logprobs) and
+    completions (int), non-stream and stream."""
+    from fastapi.testclient import TestClient
+    import gllm_amd.entrypoints.api_server as srv
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+
+    d = _mk_model_dir(tmp_path)
+    cfg = EngineConfig(model=d, load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       master_port=29693, enable_prefix_caching=False)
+    srv.engine = AsyncLLMEngine(cfg)
+    srv.served_model = "tiny"
+    srv.engine.start()
+    try:
+        app = srv.build_app()
+        client = TestClient(app)
+        r = client.post("/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "w5 w6"}],
+            "max_tokens": 4, "temperature": 0.0, "ignore_eos": True,
+            "logprobs": True, "top_logprobs": 3})
+        assert r.status_code == 200, r.text
+        lp = r.json()["choices"][0]["logprobs"]["content"]
+        assert len(lp) == 4
+        for e in lp:
+            assert e["logprob"] <= 0.0 and len(e["top_logprobs"]) == 3
+
+        r2 = client.post("/v1/completions", json={
+            "prompt": "w3 w4", "max_tokens": 3, "temperature": 0.0,
+            "ignore_eos": True, "logprobs": 2})
+        assert r2.status_code == 200, r2.text
+        lp2 = r2.json()["choices"][0]["logprobs"]["content"]
+        assert len(lp2) == 3 and len(lp2[0]["top_logprobs"]) == 2
+
+        with client.stream("POST", "/v1/chat/completions", json={
+                "messages": [{"role": "user", "content": "w9"}],
+                "max_tokens": 3, "temperature": 0.0, "stream": True,
+                "ignore_eos": True, "logprobs": True}) as r3:
+            lines = [ln for ln in r3.iter_lines()
+                     if ln.startswith("data: {")]
+        bodies = [json.loads(ln[len("data: "):]) for ln in lines]
+        with_lp = [b for b in bodies
+                   if b["choices"] and b["choices"][0].get("logprobs")]
+        assert len(with_lp) == 3
+    finally:
+        srv.engine.stop()

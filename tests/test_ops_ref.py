@@ -113,3 +113,36 @@ def test_topk_softmax_renorm():
     w, ids = R.topk_softmax(g, 2)
     assert ids[0].tolist() == [2, 1]
     assert abs(float(w.sum()) - 1.0) < 1e-6
+
+
+def test_min_p_filter():
+    """vLLM-style min_p: tokens below min_p * max_prob are dropped."""
+    import torch
+    from gllm_amd.layers.sampler import Sampler
+    probs = torch.tensor([[0.5, 0.3, 0.15, 0.05],
+                          [0.25, 0.25, 0.25, 0.25]])
+    out = Sampler._apply_min_p(probs.clone(),
+                               torch.tensor([0.4, 0.0]))
+    # row 0: threshold 0.2 -> keep 0.5, 0.3; renormalized
+    assert out[0, 2] == 0 and out[0, 3] == 0
+    assert abs(out[0, 0] - 0.5 / 0.8) < 1e-6
+    # row 1: disabled -> unchanged
+    assert torch.allclose(out[1], probs[1])
+
+
+def test_min_p_sampling_end_to_end():
+    import torch
+    from gllm_amd.layers.sampler import Sampler, SamplingMetadata
+    B, V = 4, 50
+    torch.manual_seed(0)
+    logits = torch.randn(B, V) * 4
+    meta = SamplingMetadata(
+        temperatures=torch.full((B,), 1.0),
+        top_ps=torch.ones(B), top_ks=torch.full((B,), -1,
+                                                dtype=torch.int32),
+        penalties=torch.ones(B), all_greedy=False, any_penalty=False,
+        min_ps=torch.full((B,), 0.9),
+        generators=[torch.Generator().manual_seed(i) for i in range(B)])
+    out = Sampler()(logits, meta)
+    # min_p=0.9 keeps essentially only the argmax at these scales
+    assert torch.equal(out.next_tokens, logits.argmax(-1))
