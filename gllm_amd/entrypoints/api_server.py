@@ -113,6 +113,9 @@ def build_app():
             top_k=req.top_k if req.top_k is not None else -1,
             min_p=getattr(req, "min_p", None) or 0.0,
             repetition_penalty=rp,
+            presence_penalty=getattr(req, "presence_penalty", 0.0) or 0.0,
+            frequency_penalty=getattr(req, "frequency_penalty", 0.0)
+            or 0.0,
             max_tokens=max_tokens,
             min_tokens=req.min_tokens or 0,
             ignore_eos=req.ignore_eos,

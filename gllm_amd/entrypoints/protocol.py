@@ -84,6 +84,8 @@ class CompletionRequest(BaseModel):
     seed: Optional[int] = None
     min_p: Optional[float] = None
     repetition_penalty: Optional[float] = None
+    presence_penalty: Optional[float] = 0.0
+    frequency_penalty: Optional[float] = 0.0
     ignore_eos: bool = False
     echo: bool = False
     logprobs: Optional[int] = None

@@ -22,6 +22,9 @@ class SamplingMetadata:
     all_greedy: bool
     any_penalty: bool
     min_ps: Optional[torch.Tensor] = None  # [B] float32 (0 disabled)
+    # OpenAI presence/frequency penalties (additive over OUTPUT tokens);
+    # applied on the host token-row path
+    pres_freq_rows: Optional[List] = None  # [(row, out_ids, pres, freq)]
     # token history per row for repetition penalty (cpu LongTensors),
     # used only as the fallback when no mask pool slot is available
     token_id_rows: Optional[List[torch.Tensor]] = None
@@ -88,6 +91,16 @@ class Sampler(torch.nn.Module):
         if meta.token_id_rows is not None:
             logits = torch_ref.apply_repetition_penalty(
                 logits, meta.token_id_rows, meta.penalties)
+        if meta.pres_freq_rows:
+            for row, out_ids, pres, freq in meta.pres_freq_rows:
+                if out_ids.numel() == 0:
+                    continue
+                counts = torch.bincount(out_ids,
+                                        minlength=logits.shape[1]
+                                        ).to(logits.device)
+                seen = counts > 0
+                logits[row] = logits[row] - pres * seen.float() \
+                    - freq * counts.float()
         return logits
 
     @staticmethod
@@ -161,6 +174,7 @@ def build_sampling_metadata(items, device,
     max_lp = 0
     any_pen = False
     need_rows = False
+    pres_freq = []
     for i, it in enumerate(items):
         sp = it.seq.sampling
         temps.append(sp.temperature)
@@ -185,6 +199,13 @@ def build_sampling_metadata(items, device,
         else:
             rows.append(torch.empty(0, dtype=torch.long, device=device))
         slots.append(slot)
+        pres = getattr(sp, "presence_penalty", 0.0) or 0.0
+        freq = getattr(sp, "frequency_penalty", 0.0) or 0.0
+        if pres or freq:
+            any_pen = True
+            out_ids = torch.tensor(
+                it.seq.token_ids[it.seq.prompt_len:], dtype=torch.long)
+            pres_freq.append((i, out_ids, pres, freq))
         if sp.logprobs:
             max_lp = max(max_lp, sp.logprobs)
         if sp.seed is not None:
@@ -206,5 +227,6 @@ def build_sampling_metadata(items, device,
         penalty_slots=torch.tensor(slots, dtype=torch.long, device=device)
         if any_pen else None,
         sample_rows=sample_rows,
+        pres_freq_rows=pres_freq or None,
         max_logprobs=max_lp,
         generators=gens if any(g is not None for g in gens) else None)

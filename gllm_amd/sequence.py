@@ -18,6 +18,8 @@ class SamplingParams:
     top_k: int = -1                    # -1 = disabled
     min_p: float = 0.0                 # drop p < min_p * max_p (vLLM)
     repetition_penalty: float = 1.0
+    presence_penalty: float = 0.0      # additive, over output tokens
+    frequency_penalty: float = 0.0     # additive * count, output tokens
     max_tokens: int = 512
     min_tokens: int = 0
     ignore_eos: bool = False

@@ -146,3 +146,57 @@ def test_min_p_sampling_end_to_end():
     out = Sampler()(logits, meta)
     # min_p=0.9 keeps essentially only the argmax at these scales
     assert torch.equal(out.next_tokens, logits.argmax(-1))
+
+
+def test_presence_frequency_penalties():
+    """OpenAI additive penalties over OUTPUT tokens only."""
+    import torch
+    from gllm_amd.layers.sampler import Sampler, SamplingMetadata
+    B, V = 2, 10
+    logits = torch.zeros(B, V)
+    out_ids0 = torch.tensor([3, 3, 5])   # token 3 twice, 5 once
+    meta = SamplingMetadata(
+        temperatures=torch.zeros(B),
+        top_ps=torch.ones(B),
+        top_ks=torch.full((B,), -1, dtype=torch.int32),
+        penalties=torch.ones(B), all_greedy=True, any_penalty=True,
+        pres_freq_rows=[(0, out_ids0, 0.5, 0.25)])
+    out = Sampler._apply_penalties(logits.clone(), meta)
+    assert abs(out[0, 3] - (-0.5 - 0.25 * 2)) < 1e-6
+    assert abs(out[0, 5] - (-0.5 - 0.25)) < 1e-6
+    assert out[0, 0] == 0.0 and torch.all(out[1] == 0)
+
+
+def test_frequency_penalty_end_to_end(tmp_path):
+    """A strong frequency penalty must change greedy output vs none
+    (and prompt tokens must NOT be penalized: outputs only)."""
+    import json as _json
+    import torch
+    d = tmp_path / "fp"
+    d.mkdir()
+    cfg_json = {
+        "architectures": ["Qwen2ForCausalLM"], "model_type": "qwen2",
+        "hidden_size": 64, "intermediate_size": 128,
+        "num_hidden_layers": 2, "num_attention_heads": 4,
+        "num_key_value_heads": 2, "vocab_size": 128,
+        "max_position_embeddings": 2048, "rms_norm_eps": 1e-6,
+        "rope_theta": 10000.0, "eos_token_id": 0,
+    }
+    with open(d / "config.json", "w") as f:
+        _json.dump(cfg_json, f)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    from gllm_amd.sequence import SamplingParams
+    cfg = EngineConfig(model=str(d), load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       enable_prefix_caching=False)
+    llm = LLM(config=cfg, num_pages_override=128)
+    prompt = [1, 2, 3, 4, 5]
+    base = llm.generate([prompt], [SamplingParams(
+        temperature=0.0, max_tokens=10, ignore_eos=True)])[0].token_ids
+    pen = llm.generate([prompt], [SamplingParams(
+        temperature=0.0, max_tokens=10, ignore_eos=True,
+        frequency_penalty=100.0)])[0].token_ids
+    assert len(set(pen)) == len(pen), \
+        f"freq penalty must forbid repeats, got {pen}"
+    assert base != pen or len(set(base)) == len(base)
