@@ -33,13 +33,14 @@ class IncrementalDetokenizer:
         return delta
 
 
-def check_stop_strings(text: str, stops: Optional[List[str]]
-                       ) -> Tuple[bool, str]:
-    """Return (hit, truncated_text)."""
+def check_stop_strings(text: str, stops: Optional[List[str]],
+                       include_stop: bool = False) -> Tuple[bool, str]:
+    """Return (hit, truncated_text). ``include_stop`` keeps the stop
+    string in the output (OpenAI include_stop_str_in_output)."""
     if not stops:
         return False, text
     for s in stops:
         i = text.find(s)
         if i >= 0:
-            return True, text[:i]
+            return True, text[:i + len(s)] if include_stop else text[:i]
     return False, text

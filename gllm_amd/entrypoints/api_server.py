@@ -120,6 +120,8 @@ def build_app():
             min_tokens=req.min_tokens or 0,
             ignore_eos=req.ignore_eos,
             stop=stops, stop_token_ids=req.stop_token_ids,
+            include_stop_str_in_output=getattr(
+                req, "include_stop_str_in_output", False),
             seed=req.seed,
             logprobs=lp,
             prompt_logprobs=getattr(req, "prompt_logprobs", None))
@@ -213,7 +215,9 @@ def build_app():
                 lps.append(_lp_entry(chunk))
         text = "".join(text_parts)
         from gllm_amd.engine.detokenizer import check_stop_strings
-        _, text = check_stop_strings(text, sampling.stop)
+        _, text = check_stop_strings(
+            text, sampling.stop,
+            include_stop=sampling.include_stop_str_in_output)
         return text, finish, n_out, plp, lps
 
     async def _merged_stream(raw, token_ids, sampling, n, mm=None):

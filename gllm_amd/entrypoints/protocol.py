@@ -67,6 +67,7 @@ class ChatCompletionRequest(BaseModel):
     tool_choice: Optional[Union[str, Dict[str, Any]]] = None
     chat_template_kwargs: Optional[Dict[str, Any]] = None
     prompt_logprobs: Optional[int] = None
+    include_stop_str_in_output: bool = False
 
 
 class CompletionRequest(BaseModel):
@@ -90,6 +91,7 @@ class CompletionRequest(BaseModel):
     echo: bool = False
     logprobs: Optional[int] = None
     prompt_logprobs: Optional[int] = None
+    include_stop_str_in_output: bool = False
 
 
 class UsageInfo(BaseModel):

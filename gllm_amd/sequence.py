@@ -25,6 +25,7 @@ class SamplingParams:
     ignore_eos: bool = False
     stop: Optional[List[str]] = None
     stop_token_ids: Optional[List[int]] = None
+    include_stop_str_in_output: bool = False
     logprobs: Optional[int] = None     # top-k logprobs to return per token
     prompt_logprobs: Optional[int] = None
     seed: Optional[int] = None
