@@ -239,6 +239,9 @@ class ServingWorker(ServingMixin, PPEngine):
             if self.shutdown:
                 break
             if not self.scheduler.has_work():
+                # keep the 1 Hz stats line fresh while idle (the last
+                # busy-loop snapshot may still show reaped seqs)
+                self._maybe_send_stats()
                 continue
             while len(self.inflight) < max(1, self.pp_size):
                 b = self.scheduler.schedule_once()
@@ -284,6 +287,7 @@ class OverlapServingWorker(ServingMixin, OverlapEngine):
             if self.shutdown:
                 break
             if not (self.scheduler.has_work() or self.pending):
+                self._maybe_send_stats()
                 continue
             self.step_tick()
             self._maybe_send_stats()
