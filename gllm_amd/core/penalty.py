@@ -7,7 +7,6 @@ GPU scatters, so the overlap engine can penalize without waiting for
 the sampled token to reach the host.
 """
 
-from typing import Optional
 
 import torch
 

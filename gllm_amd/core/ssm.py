@@ -13,7 +13,6 @@ manager's hit_filter trims hits to the deepest snapshotted boundary.
 
 import collections
 import dataclasses
-from typing import List, Optional
 
 import torch
 

@@ -9,7 +9,7 @@ Entries expire after ``ttl_s`` without a heartbeat.
 import socket
 import threading
 import time
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 from gllm_amd.engine.multinode import recv_msg, send_msg
 from gllm_amd.logger import logger

@@ -10,7 +10,6 @@ Routes: /v1/chat/completions, /v1/completions, /v1/models, /health,
 import argparse
 import asyncio
 import json
-import time
 from typing import Optional
 
 from gllm_amd import __version__

@@ -6,7 +6,6 @@ kernel family (prefill MFMA kernel / split-KV decode kernel, dispatched
 in ops.paged_attention).
 """
 
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -7,7 +7,7 @@ kernels in ops/). Weight sharding happens at load time via a
 ``weight_loader`` attribute attached to each parameter.
 """
 
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.nn as nn

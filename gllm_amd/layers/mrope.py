@@ -10,7 +10,7 @@ Engine status: unit-tested building block for the round-2 multimodal
 path (vision tower + embedding merge + mm scheduler plumbing).
 """
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

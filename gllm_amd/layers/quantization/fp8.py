@@ -24,7 +24,7 @@ Checkpoint format (DeepSeek / Qwen fp8 releases):
 """
 
 import math
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.nn as nn

@@ -22,7 +22,6 @@ Both store weights K-major ([K, N]); torch linear wants [N, K], so
 dequant transposes at the end.
 """
 
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -13,8 +13,7 @@ Input: pixel_values [N_patches, C, ps, ps] (per-patch crops), grids
 [(t, h, w)] in patch units. Output: [sum (h/kh)*(w/kw), text_hidden].
 """
 
-import math
-from typing import List, Tuple
+from typing import List
 
 import torch
 import torch.nn as nn

@@ -8,7 +8,6 @@ a content-hash LRU of merged embeddings."""
 
 import collections
 import hashlib
-from typing import Optional, Tuple
 
 import torch
 

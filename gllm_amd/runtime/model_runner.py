@@ -5,7 +5,6 @@ Parity: reference model_runner.py init/profile/KV-sizing/step_once
 runtime/graph_runner.py and is driven from here.
 """
 
-import math
 from typing import List, Optional
 
 import torch

@@ -9,7 +9,7 @@ DeepSeek/Kimi DSML formats follow in a later pass.
 
 import json
 import re
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 from gllm_amd.entrypoints.protocol import FunctionCall, ToolCall
 

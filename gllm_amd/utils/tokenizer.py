@@ -10,7 +10,6 @@ directly with the config's special tokens + chat template.
 
 import json
 import os
-from typing import Optional
 
 
 def load_tokenizer(model_path: str):
