@@ -83,6 +83,8 @@ class OverlapEngine(PPEngine):
     # ------------------------------------------------------------------
     def step_tick(self) -> int:
         """Returns sampled tokens finalized during this tick."""
+        if self.dp_size > 1:
+            return self._step_tick_dp()
         n_final = 0
         while len(self.pending) >= self.DEPTH:
             n_final += self._collect_one()
@@ -99,6 +101,41 @@ class OverlapEngine(PPEngine):
         self._launch_overlap(batch)
         return n_final
 
+    def _step_tick_dp(self) -> int:
+        """DP-attention overlap tick (reference overlap_worker.py
+        :258-309): one dp_meta_barrier per tick; a replica with nothing
+        scheduled launches a 1-token dummy so the MoE collectives stay
+        matched. Collection stays local (pp=1: no cross-rank traffic in
+        finalize)."""
+        P = self.P
+        n_final = 0
+        while len(self.pending) >= self.DEPTH:
+            n_final += self._collect_one()
+        batch = self.scheduler.schedule_once() \
+            if self.scheduler.has_work() else None
+        nt = batch.num_tokens if batch is not None else 0
+        counts, flags = P.dp_meta_barrier(
+            nt, self.scheduler.has_work() or bool(self.pending))
+        self._dp_global_work = sum(flags) + sum(counts)
+        if max(counts) == 0:
+            while self.pending:
+                n_final += self._collect_one()
+            return n_final
+        if batch is not None and any(
+                it.seq.sampling.repetition_penalty != 1.0
+                for it in batch.items):
+            while self.pending:
+                n_final += self._collect_one()
+        P.set_dp_forward_counts(counts)
+        try:
+            if batch is None:
+                self.runner.step_dummy()
+            else:
+                self._launch_overlap(batch)
+        finally:
+            P.set_dp_forward_counts(None)
+        return n_final
+
     def pop_finished(self) -> List[Sequence]:
         out = self._finished_since
         self._finished_since = []
@@ -107,7 +144,8 @@ class OverlapEngine(PPEngine):
     def run_until_done(self, max_steps: Optional[int] = None):
         done = []
         steps = 0
-        while self.scheduler.has_work() or self.pending:
+        while self.scheduler.has_work() or self.pending or \
+                (self.dp_size > 1 and self._dp_global_work > 0):
             self.step_tick()
             done.extend(self.pop_finished())
             steps += 1

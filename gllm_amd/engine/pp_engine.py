@@ -49,7 +49,7 @@ class PPEngine:
         # multi-stage DP lockstep is in ROADMAP.md
         assert self.dp_size == 1 or self.pp_size == 1, \
             "dp_size > 1 requires pp_size == 1"
-        self._dp_global_work = 1
+        self._dp_global_work = 1 if self.dp_size > 1 else 0
         self.is_first = P.is_first_pp_rank()
         self.is_last = P.is_last_pp_rank()
         self.device = config.device

@@ -280,13 +280,17 @@ class OverlapServingWorker(ServingMixin, OverlapEngine):
         self.comm.send_output(("out", out, {}))
 
     def run_loop(self) -> None:
-        logger.info("worker %d ready (overlap)", self.rank)
+        logger.info("worker %d ready (overlap%s)", self.rank,
+                    f", dp={self.dp_rank} of {self.dp_size}"
+                    if self.dp_size > 1 else "")
         while not self.shutdown:
             has_work = self.scheduler.has_work() or bool(self.pending)
-            self._sync_intake(block=not has_work)
+            self._sync_intake(block=not has_work
+                              and self._dp_global_work == 0)
             if self.shutdown:
                 break
-            if not (self.scheduler.has_work() or self.pending):
+            if not (self.scheduler.has_work() or self.pending
+                    or self._dp_global_work > 0):
                 self._maybe_send_stats()
                 continue
             self.step_tick()
@@ -309,7 +313,9 @@ def run_worker(rank: int, config: EngineConfig, req_queue, out_queue,
     if config.device.startswith("cuda"):
         config.device = f"cuda:{local % max(1, torch.cuda.device_count())}"
     try:
-        use_overlap = (config.world_size == 1 and config.enable_overlap
+        # overlap covers PP=1 worlds (incl. DP replicas and TP shards);
+        # multi-stage pipelines use the sync PP worker
+        use_overlap = (config.pp_size == 1 and config.enable_overlap
                        and config.device.startswith("cuda"))
         cls = OverlapServingWorker if use_overlap else ServingWorker
         worker = cls(config, req_queue, out_queue)

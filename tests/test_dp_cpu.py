@@ -139,6 +139,52 @@ def test_dp2_ep_moe_equals_single(tmp_path):
     assert [got[i] for i in range(len(PROMPTS))] == ref
 
 
+def _run_dp_overlap_rank(rank, model_dir, port, q):
+    os.environ.update(RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    from gllm_amd.engine.overlap_engine import OverlapEngine
+    from gllm_amd.sequence import SamplingParams, Sequence
+    eng = OverlapEngine(_mk_cfg(model_dir, 2, True, port),
+                        num_pages_override=128)
+    mine = [(i, p) for i, p in enumerate(PROMPTS) if i % 2 == eng.dp_rank]
+    seqs = [Sequence(i, p, SamplingParams(temperature=0.0,
+                                          max_tokens=MAX_TOKENS[i],
+                                          ignore_eos=True))
+            for i, p in mine]
+    eng.add_requests(seqs)
+    eng.run_until_done()
+    q.put((rank, [(s.seq_id, s.output_token_ids) for s in seqs]))
+    import torch.distributed as dist
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_overlap_equals_single(tmp_path):
+    """DP + launch-first/collect-later overlap (reference
+    overlap_worker.py:258-309): lockstep dummy launches keep the MoE
+    collectives matched while collection stays local."""
+    d = str(tmp_path / "ckpt")
+    _write_checkpoint(d)
+    ref = _single_reference(d)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_dp_overlap_rank,
+                         args=(r, d, 29767, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    got = {}
+    for _ in range(2):
+        _rank, outs = q.get(timeout=240)
+        for sid, toks in outs:
+            got[sid] = toks
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert [got[i] for i in range(len(PROMPTS))] == ref
+
+
 def _run_dp_fuzz_rank(rank, model_dir, port, prompts, max_tokens, q):
     os.environ.update(RANK=str(rank), MASTER_ADDR="127.0.0.1",
                       MASTER_PORT=str(port))
