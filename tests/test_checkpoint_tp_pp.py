@@ -125,3 +125,44 @@ def test_checkpoint_parallel_equals_single(tmp_path, pp, tp, port):
         p.join(timeout=120)
         assert p.exitcode == 0
     assert got == ref
+
+
+def _run_overlap_tp_rank(rank, model_dir, port, q):
+    os.environ.update(RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    from gllm_amd.engine.overlap_engine import OverlapEngine
+    from gllm_amd.sequence import SamplingParams, Sequence
+    eng = OverlapEngine(_mk_cfg(model_dir, 1, 2, port),
+                        num_pages_override=128)
+    seqs = [Sequence(i, p, SamplingParams(temperature=0.0,
+                                          max_tokens=MAX_TOKENS,
+                                          ignore_eos=True))
+            for i, p in enumerate(PROMPTS)]
+    eng.add_requests(seqs)
+    eng.run_until_done()
+    if rank == 0:
+        q.put([s.output_token_ids for s in seqs])
+    import torch.distributed as dist
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_overlap_tp2_equals_single(tmp_path):
+    """TP=2 under the overlap engine (launch-first/collect-later with
+    replicated samplers) must equal the single-process sync run."""
+    d = str(tmp_path / "ckpt")
+    _write_checkpoint(d)
+    ref = _single_reference(d)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_overlap_tp_rank,
+                         args=(r, d, 29743, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert got == ref
