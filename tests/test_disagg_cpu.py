@@ -169,3 +169,67 @@ def test_encoder_fault_injection_and_redispatch(tmp_path):
     finally:
         enc.terminate()
         enc.join(timeout=30)
+
+
+@pytest.mark.timeout(300)
+def test_disagg_with_prefix_cache(tmp_path):
+    """Encoder disagg + mm prefix caching: the second identical request
+    hits BOTH the encoder's embedding cache and the LM's prefix cache,
+    and still matches exactly."""
+    from fastapi.testclient import TestClient
+    import gllm_amd.entrypoints.api_server as srv_mod
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+
+    d = _mk_vl_dir(tmp_path)
+    enc_port = _free_port()
+    ctx = mp.get_context("spawn")
+    enc = ctx.Process(target=_run_encoder, args=(d, enc_port))
+    enc.start()
+    try:
+        deadline = time.time() + 120
+        while True:
+            try:
+                socket.create_connection(("127.0.0.1", enc_port),
+                                         timeout=1).close()
+                break
+            except OSError:
+                assert time.time() < deadline
+                time.sleep(0.5)
+        cfg = EngineConfig(model=d, load_format="dummy", device="cpu",
+                           dtype="float32", page_size=4, maxp=64,
+                           master_port=29698,
+                           mm_encoder_addr=f"127.0.0.1:{enc_port}",
+                           enable_prefix_caching=True)
+        eng = AsyncLLMEngine(cfg)
+        eng.vision_wrap = ("<|vision_start|> ", " <|vision_end|>")
+        eng.start()
+        srv_mod.engine = eng
+        srv_mod.served_model = "vl"
+        try:
+            client = TestClient(srv_mod.build_app())
+
+            def ask(seed):
+                r = client.post("/v1/chat/completions", json={
+                    "messages": [{"role": "user", "content": [
+                        {"type": "text", "text": "w5 w6 w7 w8 "},
+                        {"type": "image_url",
+                         "image_url": {"url": _img_b64(seed)}},
+                    ]}],
+                    "max_tokens": 4, "temperature": 0.0,
+                    "ignore_eos": True})
+                assert r.status_code == 200, r.text
+                return r.json()["choices"][0]["message"]["content"]
+
+            a1 = ask(5)
+            a2 = ask(5)
+            assert a1 == a2
+            b = ask(77)
+            assert b != a1, "different image must not alias cached pages"
+            stats = eng.encoder_client.stats()
+            assert stats["hits"] >= 1
+        finally:
+            eng.stop()
+    finally:
+        enc.terminate()
+        enc.join(timeout=30)
