@@ -60,6 +60,12 @@ class ModelRunner:
             # DSA selector round-1 runs the eager per-seq torch path;
             # the graph-safe tile-static scorer is round 2 (ROADMAP.md)
             cfg.use_graph = False
+        if cfg.dp_size > 1:
+            # DP rounds gate forwards on dp_meta_barrier counts and the
+            # MoE DP gather runs collectives inside the forward — both
+            # incompatible with capture (graph-captured DP decode with
+            # uniform gather is a round-2 item)
+            cfg.use_graph = False
         num_pages = num_pages_override or self._size_kv_cache()
         self._allocate_kv(num_pages)
         self.num_kv_pages_total = num_pages
