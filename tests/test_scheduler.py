@@ -156,3 +156,19 @@ def test_determinism_two_replicas():
             == [(it.seq.seq_id, it.start, it.num_tokens) for it in bb.items]
         a.process_output(ba, list(range(len(ba.items))))
         b.process_output(bb, list(range(len(bb.items))))
+
+
+def test_pp_layer_range_and_assigned_layers():
+    from gllm_amd.config import EngineConfig
+    cfg = EngineConfig(model="x", pp_size=4)
+    # even split with remainder on the LAST stages (token throttling
+    # wants earlier stages lighter: they also run embed/sampling legs)
+    assert [cfg.pp_layer_range(r, 10) for r in range(4)] == \
+        [(0, 2), (2, 4), (4, 7), (7, 10)]
+    cfg2 = EngineConfig(model="x", pp_size=3, assigned_layers="5,3,2")
+    assert [cfg2.pp_layer_range(r, 10) for r in range(3)] == \
+        [(0, 5), (5, 8), (8, 10)]
+    import pytest as _pt
+    cfg3 = EngineConfig(model="x", pp_size=2, assigned_layers="5,3")
+    with _pt.raises(AssertionError):
+        cfg3.pp_layer_range(0, 10)
