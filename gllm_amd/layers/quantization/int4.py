@@ -7,8 +7,13 @@ path is an LDS-resident dequant-fused MFMA GEMM (weights stay packed in
 registers/LDS, unpacked on the fly — int4 is memory-bound decode's best
 friend at 4x bf16 density); round-1 executes like fp8: weights load in
 the checkpoint's packed layout and are dequantized lazily (cached) into
-the compute dtype, so numerics are exact w.r.t. the format. TP sharding
-of packed tensors is round 2 — int4 loads require tp_size == 1.
+the compute dtype, so numerics are exact w.r.t. the format. Packed
+tensors shard across TP at load time: column-family layers slice the
+stored N axis (checkpoints keep weights K-major, so the output dim is
+dim 1 of every packed tensor), row-parallel slices the stored K axis —
+qweight by packed rows, qzeros/scales by whole quant groups, which
+requires K/tp to be a multiple of group_size (the standard wna16
+restriction; group 128 divides every real K/tp).
 
 Formats (public conventions):
   GPTQ (bits=4, no act-order): qweight int32 [K/8, N], K packed
@@ -131,36 +136,63 @@ _QUANT_SUFFIXES = (
 def convert_linear_to_int4(layer, method: str, group_size: int) -> None:
     """Replace the dense weight with packed int4 params. Checkpoint
     names (qweight/qzeros/scales) route to them directly through the
-    models' substring-mapped loaders."""
+    models' substring-mapped loaders. Under TP each loader takes this
+    rank's slice of the full checkpoint tensor: dim 1 (the stored N
+    axis) for the column family, dim 0 (packed K rows / quant groups)
+    for row-parallel."""
     from gllm_amd.layers.linear import (MergedColumnParallelLinear,
-                                        QKVParallelLinear)
-    from gllm_amd.parallel import get_tp_size
-    assert get_tp_size() == 1, \
-        "int4 TP sharding of packed tensors is round 2 (ROADMAP.md)"
-    N, K = layer.weight.shape
+                                        QKVParallelLinear,
+                                        RowParallelLinear)
+    from gllm_amd.parallel import get_tp_rank, get_tp_size
+    tp, tp_rank = get_tp_size(), get_tp_rank()
+    N, K = layer.weight.shape  # already this rank's shard
+    row = isinstance(layer, RowParallelLinear)
+    if tp > 1:
+        if row:
+            assert K % group_size == 0 and K % 8 == 0, (
+                f"int4 row-parallel needs K/tp divisible by the quant "
+                f"group ({group_size}) and the pack factor: K/tp={K}")
+        elif isinstance(layer, QKVParallelLinear):
+            assert layer.q_size % 8 == 0 and layer.kv_size % 8 == 0
+        elif isinstance(layer, MergedColumnParallelLinear):
+            assert all(s // tp % 8 == 0 for s in layer.output_sizes)
+        else:
+            assert N % 8 == 0, f"int4 column shard N/tp={N} not 8-aligned"
     G = K // group_size
 
     def _mk_loader(div: int):
-        """Place a sub-projection's packed tensor into its dim-1 slice
-        of the fused param (all int4 tensors concat along N; ``div`` is
-        the N packing factor of that tensor: 8 for qzeros / AWQ qweight,
-        1 for scales / GPTQ qweight)."""
+        """Build the loader for one packed tensor. ``div`` is that
+        tensor's N packing factor (8 for qzeros / AWQ qweight, 1 for
+        scales / GPTQ qweight); stored dim 1 holds N/div columns, so a
+        rank's slice of a full tensor is loaded.shape[axis] // tp wide
+        on the shard axis. Merged/QKV additionally place the slice at
+        the sub-projection's offset inside the fused param."""
         if isinstance(layer, MergedColumnParallelLinear):
             def load(param, loaded, shard_id: int):
-                off = sum(layer.output_sizes[:shard_id]) // div
-                size = layer.output_sizes[shard_id] // div
-                param.data.narrow(1, off, size).copy_(loaded)
+                off = sum(layer.output_sizes[:shard_id]) // tp // div
+                size = layer.output_sizes[shard_id] // tp // div
+                param.data.narrow(1, off, size).copy_(
+                    loaded.narrow(1, tp_rank * size, size))
         elif isinstance(layer, QKVParallelLinear):
             def load(param, loaded, shard_id: str):
-                offs = {"q": 0, "k": layer.q_size,
-                        "v": layer.q_size + layer.kv_size}
-                sizes = {"q": layer.q_size, "k": layer.kv_size,
-                         "v": layer.kv_size}
-                param.data.narrow(1, offs[shard_id] // div,
-                                  sizes[shard_id] // div).copy_(loaded)
+                if shard_id == "q":
+                    off, size, src = 0, layer.q_size, tp_rank
+                elif shard_id == "k":
+                    off, size = layer.q_size, layer.kv_size
+                    src = tp_rank // layer.kv_replication
+                else:
+                    off, size = layer.q_size + layer.kv_size, layer.kv_size
+                    src = tp_rank // layer.kv_replication
+                param.data.narrow(1, off // div, size // div).copy_(
+                    loaded.narrow(1, src * (size // div), size // div))
+        elif row:
+            def load(param, loaded, *a):
+                shard = loaded.shape[0] // tp
+                param.data.copy_(loaded.narrow(0, tp_rank * shard, shard))
         else:
             def load(param, loaded, *a):
-                param.data.copy_(loaded)
+                shard = loaded.shape[1] // tp
+                param.data.copy_(loaded.narrow(1, tp_rank * shard, shard))
         return load
 
     if method == "gptq":

@@ -111,6 +111,67 @@ def _gen(model_dir):
     return [o.token_ids for o in llm.generate(PROMPTS, sp)]
 
 
+def _run_int4_tp_rank(rank, model_dir, port, q):
+    os.environ.update(RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.pp_engine import PPEngine
+    from gllm_amd.sequence import SamplingParams, Sequence
+    cfg = EngineConfig(model=model_dir, load_format="auto", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       tp_size=2, master_port=port,
+                       enable_prefix_caching=False)
+    eng = PPEngine(cfg, num_pages_override=128)
+    seqs = [Sequence(i, p, SamplingParams(temperature=0.0,
+                                          max_tokens=MAX_TOKENS,
+                                          ignore_eos=True))
+            for i, p in enumerate(PROMPTS)]
+    eng.add_requests(seqs)
+    eng.run_until_done()
+    if rank == 0:
+        q.put([s.output_token_ids for s in seqs])
+    import torch.distributed as dist
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("method,port", [("gptq", 29771), ("awq", 29781)])
+def test_int4_tp2_equals_single(tmp_path, method, port):
+    """TP=2 sharding of packed int4 tensors (qweight/qzeros/scales
+    sliced along stored N for column layers, along packed K rows and
+    whole quant groups for row-parallel) must reproduce the
+    single-process int4 outputs exactly."""
+    import multiprocessing as mp
+    pack = pack_gptq if method == "gptq" else pack_awq
+    base = _base_state_dict()
+    q_sd = {}
+    for name, w in base.items():
+        if _is_quantized(name):
+            qw, qz, s = pack(w, GROUP)
+            stem = name[:-len(".weight")]
+            q_sd[stem + ".qweight"] = qw
+            q_sd[stem + ".qzeros"] = qz
+            q_sd[stem + ".scales"] = s
+        else:
+            q_sd[name] = w
+    d = str(tmp_path / f"{method}_tp")
+    _write(d, q_sd, method)
+    ref = _gen(d)
+    ctx = mp.get_context("spawn")
+    rq = ctx.Queue()
+    procs = [ctx.Process(target=_run_int4_tp_rank,
+                         args=(r, d, port, rq)) for r in range(2)]
+    for p in procs:
+        p.start()
+    got = rq.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert got == ref
+
+
 @pytest.mark.parametrize("method", ["gptq", "awq"])
 def test_int4_checkpoint_equals_dequantized_twin(tmp_path, method):
     pack = pack_gptq if method == "gptq" else pack_awq
