@@ -54,6 +54,9 @@ class FusedMoE(nn.Module):
             requires_grad=False)
         self.w13_weight.weight_loader = self._load_w13
         self.w2_weight.weight_loader = self._load_w2
+        # set by quantization/fp8.py convert_moe_to_fp8
+        self.fp8_block = None
+        self._dq_cache = None
 
     # ---- loading: per-expert pulls with EP ownership / TP sharding ----
     def _local_expert(self, expert_id: int) -> Optional[int]:
@@ -123,11 +126,35 @@ class FusedMoE(nn.Module):
                 continue
             rows = flat_rows[sel]
             xe = x.index_select(0, rows)
-            h = ops.silu_and_mul(ops.linear(xe, self.w13_weight[lid]))
-            ye = ops.linear(h, self.w2_weight[lid])
+            if self.fp8_block is not None:
+                w13, w2 = self._dequant_expert(lid, x.dtype, x.device)
+            else:
+                w13, w2 = self.w13_weight[lid], self.w2_weight[lid]
+            h = ops.silu_and_mul(ops.linear(xe, w13))
+            ye = ops.linear(h, w2)
             w = weights.flatten()[sel].unsqueeze(-1)
             out.index_add_(0, rows, ye * w)
         return out
+
+    def _dequant_expert(self, lid: int, dtype, device):
+        """fp8 expert weights dequantized once per expert (lazily,
+        cached) into the compute dtype — same round-1 stance as the
+        dense fp8 linears (layers/quantization/fp8.py); the grouped fp8
+        MFMA GEMM consuming the packed layout directly is round 2."""
+        if self._dq_cache is None:
+            self._dq_cache = [None] * self.num_local_experts
+        ent = self._dq_cache[lid]
+        if ent is None:
+            from gllm_amd.layers.quantization.fp8 import dequant_block_fp8
+            w13 = dequant_block_fp8(
+                self.w13_weight[lid], self.w13_weight_scale_inv[lid],
+                self.fp8_block, dtype).to(device)
+            w2 = dequant_block_fp8(
+                self.w2_weight[lid], self.w2_weight_scale_inv[lid],
+                self.fp8_block, dtype).to(device)
+            ent = (w13, w2)
+            self._dq_cache[lid] = ent
+        return ent
 
     def _forward_dp(self, x: torch.Tensor, weights: torch.Tensor,
                     ids: torch.Tensor) -> torch.Tensor:

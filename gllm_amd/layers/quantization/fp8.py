@@ -90,7 +90,7 @@ def dequant_block_fp8(wq: torch.Tensor, scale_inv: torch.Tensor,
 # ----------------------------------------------------- layer conversion
 # Linear projections that fp8 checkpoints quantize (embeddings, lm_head,
 # router gates, norms and the DSA weights_proj stay bf16). MoE expert
-# tensors (w13/w2) are round 2 (grouped fp8 MFMA GEMM).
+# tensors (w13/w2) convert through convert_moe_to_fp8 below.
 _QUANT_SUFFIXES = (
     "q_proj", "k_proj", "v_proj", "o_proj", "qkv_proj",
     "gate_proj", "up_proj", "down_proj", "gate_up_proj",
@@ -189,13 +189,76 @@ def convert_linear_to_fp8(layer, block: Tuple[int, int]) -> None:
     layer._w_dq = None
 
 
+def convert_moe_to_fp8(moe, block: Tuple[int, int]) -> None:
+    """Swap a FusedMoE's stacked expert weights for fp8 + per-expert
+    block scale grids (DeepSeek-V3 style checkpoints quantize every
+    routed expert; the router gate stays high precision). Checkpoint
+    tensors stay per-expert ([I, H] gate/up, [H, I] down; scale grids in
+    128-block space), so the scale loaders mirror the weight loaders'
+    stacking and TP narrowing in BLOCK space — which needs the per-rank
+    intermediate dim to be block-aligned (true for every real fp8 MoE
+    release: moe_intermediate_size is a multiple of 128)."""
+    from gllm_amd.parallel import get_tp_rank
+    bs0, bs1 = block
+    E = moe.num_local_experts
+    I, H = moe.intermediate_per_rank, moe.hidden_size
+    assert I % bs0 == 0 and I % bs1 == 0, (
+        f"fp8 MoE needs the per-rank intermediate dim ({I}) aligned to "
+        f"the quant block {block}")
+
+    for wname in ("w13_weight", "w2_weight"):
+        old = getattr(moe, wname)
+        neww = nn.Parameter(torch.empty(old.shape, dtype=FP8_DTYPE),
+                            requires_grad=False)
+        neww.weight_loader = old.weight_loader
+        # keep registration order/name: assign through setattr
+        setattr(moe, wname, neww)
+
+    w13_scale = nn.Parameter(
+        torch.ones(E, 2 * (I // bs0), math.ceil(H / bs1),
+                   dtype=torch.float32), requires_grad=False)
+    w2_scale = nn.Parameter(
+        torch.ones(E, math.ceil(H / bs0), I // bs1,
+                   dtype=torch.float32), requires_grad=False)
+
+    def load_w13_scale(param, loaded, expert_id: int, shard_id: int):
+        lid = moe._local_expert(expert_id)
+        if lid is None:
+            return
+        ib = I // bs0
+        shard = loaded if moe.use_ep else \
+            loaded.narrow(0, get_tp_rank() * ib, ib)
+        param.data[lid].narrow(0, shard_id * ib, ib).copy_(shard)
+
+    def load_w2_scale(param, loaded, expert_id: int):
+        lid = moe._local_expert(expert_id)
+        if lid is None:
+            return
+        ib = I // bs1
+        shard = loaded if moe.use_ep else \
+            loaded.narrow(1, get_tp_rank() * ib, ib)
+        param.data[lid].copy_(shard)
+
+    w13_scale.weight_loader = load_w13_scale
+    w2_scale.weight_loader = load_w2_scale
+    moe.w13_weight_scale_inv = w13_scale
+    moe.w2_weight_scale_inv = w2_scale
+    moe.fp8_block = tuple(block)
+    moe._dq_cache = None
+
+
 def convert_model_to_fp8(model, quant_config: dict) -> int:
-    """Walk the model and convert every checkpoint-quantized linear.
-    Returns the number of converted layers."""
+    """Walk the model and convert every checkpoint-quantized linear and
+    MoE expert bank. Returns the number of converted modules."""
     from gllm_amd.layers.linear import LinearBase
+    from gllm_amd.layers.moe.layer import FusedMoE
     block = tuple(quant_config.get("weight_block_size") or (128, 128))
     n = 0
     for name, mod in model.named_modules():
+        if isinstance(mod, FusedMoE):
+            convert_moe_to_fp8(mod, block)
+            n += 1
+            continue
         if not isinstance(mod, LinearBase):
             continue
         leaf = name.rsplit(".", 1)[-1]

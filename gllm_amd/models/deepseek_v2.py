@@ -386,9 +386,13 @@ class DeepseekV2ForCausalLM(nn.Module):
                 eidx = parts.index("experts")
                 expert_id = int(parts[eidx + 1])
                 wname = parts[eidx + 2]
+                suffix = parts[eidx + 3] if len(parts) > eidx + 3 \
+                    else "weight"
                 prefix = ".".join(parts[:eidx + 1])
                 for ckpt, fused, shard in expert_map:
                     if wname == ckpt:
+                        if suffix == "weight_scale_inv":  # fp8 scales
+                            fused = fused + "_scale_inv"
                         p = params[f"{prefix}.{fused}"]
                         if shard is None:
                             p.weight_loader(p, w, expert_id)

@@ -162,10 +162,13 @@ class MoEFamilyForCausalLM(LlamaFamilyForCausalLM):
             eidx = parts.index("experts")
             expert_id = int(parts[eidx + 1])
             wname = parts[eidx + 2]
+            suffix = parts[eidx + 3] if len(parts) > eidx + 3 else "weight"
             prefix = ".".join(parts[:eidx + 1])
             prefix = self._canonical_moe_name(prefix)
             for ckpt, fused, shard in self.expert_params_mapping:
                 if wname == ckpt:
+                    if suffix == "weight_scale_inv":  # fp8 expert scales
+                        fused = fused + "_scale_inv"
                     p = params[f"{prefix}.{fused}"]
                     if shard is None:
                         p.weight_loader(p, w, expert_id)

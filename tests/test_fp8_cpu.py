@@ -202,3 +202,136 @@ def test_fp8_tp2_scale_sharding_equals_single(tmp_path):
         p.join(timeout=120)
         assert p.exitcode == 0
     assert got == ref
+
+
+# --------------------------------------------------------- fp8 MoE experts
+MOE_CFG = {
+    "architectures": ["Qwen2MoeForCausalLM"],
+    "model_type": "qwen2_moe",
+    "hidden_size": 64,
+    "intermediate_size": 128,
+    "num_hidden_layers": 2,
+    "num_attention_heads": 4,
+    "num_key_value_heads": 2,
+    "num_experts": 4,
+    "num_experts_per_tok": 2,
+    "moe_intermediate_size": 48,
+    "shared_expert_intermediate_size": 96,
+    "norm_topk_prob": False,
+    "decoder_sparse_step": 1,
+    "vocab_size": 128,
+    "max_position_embeddings": 2048,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 10000.0,
+    "tie_word_embeddings": False,
+    "eos_token_id": 0,
+}
+MOE_BLOCK = (8, 8)  # moe_intermediate/tp = 24 stays block-aligned at tp=2
+
+
+def _moe_state_dict():
+    g = torch.Generator().manual_seed(321)
+    c = MOE_CFG
+    H, V = c["hidden_size"], c["vocab_size"]
+    hd = H // c["num_attention_heads"]
+    kv = c["num_key_value_heads"] * hd
+    Im, Is = c["moe_intermediate_size"], c["shared_expert_intermediate_size"]
+    sd = {}
+
+    def rnd(*shape):
+        return torch.randn(*shape, generator=g) * 0.08
+
+    sd["model.embed_tokens.weight"] = rnd(V, H)
+    for L in range(c["num_hidden_layers"]):
+        p = f"model.layers.{L}."
+        sd[p + "self_attn.q_proj.weight"] = rnd(H, H)
+        sd[p + "self_attn.q_proj.bias"] = rnd(H)
+        sd[p + "self_attn.k_proj.weight"] = rnd(kv, H)
+        sd[p + "self_attn.k_proj.bias"] = rnd(kv)
+        sd[p + "self_attn.v_proj.weight"] = rnd(kv, H)
+        sd[p + "self_attn.v_proj.bias"] = rnd(kv)
+        sd[p + "self_attn.o_proj.weight"] = rnd(H, H)
+        sd[p + "mlp.gate.weight"] = rnd(c["num_experts"], H)
+        for e in range(c["num_experts"]):
+            ep = p + f"mlp.experts.{e}."
+            sd[ep + "gate_proj.weight"] = rnd(Im, H)
+            sd[ep + "up_proj.weight"] = rnd(Im, H)
+            sd[ep + "down_proj.weight"] = rnd(H, Im)
+        sd[p + "mlp.shared_expert.gate_proj.weight"] = rnd(Is, H)
+        sd[p + "mlp.shared_expert.up_proj.weight"] = rnd(Is, H)
+        sd[p + "mlp.shared_expert.down_proj.weight"] = rnd(H, Is)
+        sd[p + "mlp.shared_expert_gate.weight"] = rnd(1, H)
+        sd[p + "input_layernorm.weight"] = torch.ones(H) + rnd(H) * 0.05
+        sd[p + "post_attention_layernorm.weight"] = \
+            torch.ones(H) + rnd(H) * 0.05
+    sd["model.norm.weight"] = torch.ones(H) + rnd(H) * 0.05
+    sd["lm_head.weight"] = rnd(V, H)
+    return sd
+
+
+def _moe_is_quantized(name):
+    if not name.endswith(".weight"):
+        return False
+    if name.endswith("shared_expert_gate.weight") or \
+            name.endswith("mlp.gate.weight"):
+        return False  # routers stay high precision (DeepSeek convention)
+    return any(f".{k}." in name for k in QUANT_KEYS)
+
+
+def _write_moe(d, sd, quantized):
+    os.makedirs(d, exist_ok=True)
+    cfg = dict(MOE_CFG)
+    if quantized:
+        cfg["quantization_config"] = {"quant_method": "fp8",
+                                      "fmt": "e4m3",
+                                      "weight_block_size": list(MOE_BLOCK)}
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump(cfg, f)
+    from safetensors.torch import save_file
+    save_file(sd, os.path.join(d, "model.safetensors"))
+
+
+def _make_moe_checkpoints(tmp_path):
+    base = _moe_state_dict()
+    fp8_sd, deq_sd = {}, {}
+    for name, w in base.items():
+        if _moe_is_quantized(name):
+            q, s = block_quant_fp8(w, block=MOE_BLOCK)
+            fp8_sd[name] = q
+            fp8_sd[name + "_scale_inv"] = s
+            deq_sd[name] = dequant_block_fp8(q, s, MOE_BLOCK, torch.float32)
+        else:
+            fp8_sd[name] = w
+            deq_sd[name] = w
+    d8 = str(tmp_path / "moe_fp8")
+    dq = str(tmp_path / "moe_deq")
+    _write_moe(d8, fp8_sd, quantized=True)
+    _write_moe(dq, deq_sd, quantized=False)
+    return d8, dq
+
+
+def test_fp8_moe_checkpoint_equals_dequantized_twin(tmp_path):
+    """fp8 expert banks (w13/w2 + per-expert block scale grids, routed
+    through the experts.<e>.<proj>.weight_scale_inv names) must load and
+    execute exactly like the dequantized twin checkpoint."""
+    d8, dq = _make_moe_checkpoints(tmp_path)
+    assert _gen_tokens(d8) == _gen_tokens(dq)
+
+
+@pytest.mark.timeout(300)
+def test_fp8_moe_tp2_equals_single(tmp_path):
+    """TP=2 over the fp8 MoE checkpoint — covers the block-space TP
+    narrowing of the expert scale grids (intermediate-dim shards)."""
+    d8, _ = _make_moe_checkpoints(tmp_path)
+    ref = _gen_tokens(d8)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_tp_rank, args=(r, d8, 29695, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert got == ref
