@@ -54,8 +54,9 @@ class FusedMoE(nn.Module):
             requires_grad=False)
         self.w13_weight.weight_loader = self._load_w13
         self.w2_weight.weight_loader = self._load_w2
-        # set by quantization/fp8.py convert_moe_to_fp8
+        # set by quantization/ converters (convert_moe_to_{fp8,int4})
         self.fp8_block = None
+        self.int4_cfg = None
         self._dq_cache = None
 
     # ---- loading: per-expert pulls with EP ownership / TP sharding ----
@@ -126,7 +127,7 @@ class FusedMoE(nn.Module):
                 continue
             rows = flat_rows[sel]
             xe = x.index_select(0, rows)
-            if self.fp8_block is not None:
+            if self.fp8_block is not None or self.int4_cfg is not None:
                 w13, w2 = self._dequant_expert(lid, x.dtype, x.device)
             else:
                 w13, w2 = self.w13_weight[lid], self.w2_weight[lid]
@@ -145,13 +146,24 @@ class FusedMoE(nn.Module):
             self._dq_cache = [None] * self.num_local_experts
         ent = self._dq_cache[lid]
         if ent is None:
-            from gllm_amd.layers.quantization.fp8 import dequant_block_fp8
-            w13 = dequant_block_fp8(
-                self.w13_weight[lid], self.w13_weight_scale_inv[lid],
-                self.fp8_block, dtype).to(device)
-            w2 = dequant_block_fp8(
-                self.w2_weight[lid], self.w2_weight_scale_inv[lid],
-                self.fp8_block, dtype).to(device)
+            if self.fp8_block is not None:
+                from gllm_amd.layers.quantization.fp8 import \
+                    dequant_block_fp8
+                w13 = dequant_block_fp8(
+                    self.w13_weight[lid], self.w13_weight_scale_inv[lid],
+                    self.fp8_block, dtype).to(device)
+                w2 = dequant_block_fp8(
+                    self.w2_weight[lid], self.w2_weight_scale_inv[lid],
+                    self.fp8_block, dtype).to(device)
+            else:
+                from gllm_amd.layers.quantization.int4 import (dequant_awq,
+                                                               dequant_gptq)
+                method, group = self.int4_cfg
+                fn = dequant_gptq if method == "gptq" else dequant_awq
+                w13 = fn(self.w13_qweight[lid], self.w13_qzeros[lid],
+                         self.w13_scales[lid], group, dtype).to(device)
+                w2 = fn(self.w2_qweight[lid], self.w2_qzeros[lid],
+                        self.w2_scales[lid], group, dtype).to(device)
             ent = (w13, w2)
             self._dq_cache[lid] = ent
         return ent

@@ -38,7 +38,7 @@ def _unpack_nibbles_k(packed: torch.Tensor) -> torch.Tensor:
     """GPTQ qweight [K/8, N] int32 -> [K, N] int, nibble i of word w is
     row 8*w + i."""
     Kp, N = packed.shape
-    out = torch.empty(Kp * 8, N, dtype=torch.int32)
+    out = torch.empty(Kp * 8, N, dtype=torch.int32, device=packed.device)
     p = packed.to(torch.int64) & 0xFFFFFFFF
     for i in range(8):
         # word w's nibble i is row 8*w + i == out[i::8][w]
@@ -51,7 +51,7 @@ def _unpack_nibbles_n(packed: torch.Tensor, order=None) -> torch.Tensor:
     8*w + order[i] (AWQ interleave) or 8*w + i (GPTQ zeros)."""
     R, Np = packed.shape
     order = order or tuple(range(8))
-    out = torch.empty(R, Np * 8, dtype=torch.int32)
+    out = torch.empty(R, Np * 8, dtype=torch.int32, device=packed.device)
     p = packed.to(torch.int64) & 0xFFFFFFFF
     for i, col in enumerate(order):
         out[:, col::8] = ((p >> (4 * i)) & 0xF).to(torch.int32)
@@ -64,7 +64,7 @@ def dequant_gptq(qweight, qzeros, scales, group_size: int,
     wq = _unpack_nibbles_k(qweight)                   # [K, N]
     zeros = _unpack_nibbles_n(qzeros) + 1             # [G, N] (legacy +1)
     K = wq.shape[0]
-    g = torch.arange(K) // group_size
+    g = torch.arange(K, device=wq.device) // group_size
     w = (wq.float() - zeros.float()[g]) * scales.float()[g]
     return w.t().contiguous().to(dtype)
 
@@ -75,7 +75,7 @@ def dequant_awq(qweight, qzeros, scales, group_size: int,
     wq = _unpack_nibbles_n(qweight, AWQ_ORDER)        # [K, N]
     zeros = _unpack_nibbles_n(qzeros, AWQ_ORDER)      # [G, N]
     K = wq.shape[0]
-    g = torch.arange(K) // group_size
+    g = torch.arange(K, device=wq.device) // group_size
     w = (wq.float() - zeros.float()[g]) * scales.float()[g]
     return w.t().contiguous().to(dtype)
 
@@ -216,12 +216,80 @@ def convert_linear_to_int4(layer, method: str, group_size: int) -> None:
     layer._w_dq = None
 
 
+def convert_moe_to_int4(moe, method: str, group_size: int) -> None:
+    """Swap a FusedMoE's stacked expert weights for packed int4 banks
+    (AWQ Mixtral / GPTQ Qwen-MoE releases quantize every routed expert;
+    the router gate stays dense). Checkpoint tensors are per-expert in
+    the dense formats' K-major layout (w13: K=hidden, N=intermediate —
+    column-like, TP slices stored N; w2: K=intermediate — row-like, TP
+    slices packed K rows and whole quant groups), so stacking offsets
+    and shards must be pack- (8) and group-aligned."""
+    from gllm_amd.parallel import get_tp_rank, get_tp_size
+    tp_rank = get_tp_rank()
+    tp = 1 if moe.use_ep else get_tp_size()
+    E = moe.num_local_experts
+    I, H = moe.intermediate_per_rank, moe.hidden_size
+    g = group_size
+    assert H % g == 0 and H % 8 == 0 and I % g == 0 and I % 8 == 0, (
+        f"int4 MoE needs hidden ({H}) and per-rank intermediate ({I}) "
+        f"aligned to the quant group ({g}) and the pack factor")
+
+    def mk(shape, dtype, loader):
+        p = nn.Parameter(torch.zeros(shape, dtype=dtype),
+                         requires_grad=False)
+        p.weight_loader = loader
+        return p
+
+    def col_loader(param, loaded, expert_id: int, shard_id: int):
+        # w13 tensors: place this rank's stored-N slice at the
+        # sub-projection's offset (dim 1 of the per-expert tensor)
+        lid = moe._local_expert(expert_id)
+        if lid is None:
+            return
+        size = loaded.shape[1] // tp
+        shard = loaded.narrow(1, tp_rank * size, size) if tp > 1 else loaded
+        param.data[lid].narrow(1, shard_id * size, size).copy_(shard)
+
+    def row_loader(param, loaded, expert_id: int):
+        # w2 tensors: this rank's stored-K row slice (packed rows for
+        # qweight, whole quant groups for qzeros/scales)
+        lid = moe._local_expert(expert_id)
+        if lid is None:
+            return
+        size = loaded.shape[0] // tp
+        shard = loaded.narrow(0, tp_rank * size, size) if tp > 1 else loaded
+        param.data[lid].copy_(shard)
+
+    if method == "gptq":
+        w13_qw, w2_qw = (E, H // 8, 2 * I), (E, I // 8, H)
+    else:
+        w13_qw, w2_qw = (E, H, 2 * I // 8), (E, I, H // 8)
+    moe.w13_qweight = mk(w13_qw, torch.int32, col_loader)
+    moe.w13_qzeros = mk((E, H // g, 2 * I // 8), torch.int32, col_loader)
+    moe.w13_scales = mk((E, H // g, 2 * I), torch.float16, col_loader)
+    moe.w2_qweight = mk(w2_qw, torch.int32, row_loader)
+    moe.w2_qzeros = mk((E, I // g, H // 8), torch.int32, row_loader)
+    moe.w2_scales = mk((E, I // g, H), torch.float16, row_loader)
+    # drop the dense banks (no checkpoint tensor feeds them)
+    for wname in ("w13_weight", "w2_weight"):
+        dead = nn.Parameter(torch.zeros(1), requires_grad=False)
+        dead.weight_loader = lambda param, loaded, *a: None
+        setattr(moe, wname, dead)
+    moe.int4_cfg = (method, group_size)
+    moe._dq_cache = None
+
+
 def convert_model_to_int4(model, quant_config: dict) -> int:
     from gllm_amd.layers.linear import LinearBase
+    from gllm_amd.layers.moe.layer import FusedMoE
     method = quant_config["quant_method"]
     group = int(quant_config.get("group_size", 128))
     n = 0
     for name, mod in model.named_modules():
+        if isinstance(mod, FusedMoE):
+            convert_moe_to_int4(mod, method, group)
+            n += 1
+            continue
         if not isinstance(mod, LinearBase):
             continue
         if name.rsplit(".", 1)[-1] in _QUANT_SUFFIXES:

@@ -393,6 +393,8 @@ class DeepseekV2ForCausalLM(nn.Module):
                     if wname == ckpt:
                         if suffix == "weight_scale_inv":  # fp8 scales
                             fused = fused + "_scale_inv"
+                        elif suffix in ("qweight", "qzeros", "scales"):
+                            fused = fused.replace("weight", suffix)  # int4
                         p = params[f"{prefix}.{fused}"]
                         if shard is None:
                             p.weight_loader(p, w, expert_id)

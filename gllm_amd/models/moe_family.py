@@ -169,6 +169,8 @@ class MoEFamilyForCausalLM(LlamaFamilyForCausalLM):
                 if wname == ckpt:
                     if suffix == "weight_scale_inv":  # fp8 expert scales
                         fused = fused + "_scale_inv"
+                    elif suffix in ("qweight", "qzeros", "scales"):
+                        fused = fused.replace("weight", suffix)  # int4
                     p = params[f"{prefix}.{fused}"]
                     if shard is None:
                         p.weight_loader(p, w, expert_id)

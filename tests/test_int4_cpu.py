@@ -194,3 +194,143 @@ def test_int4_checkpoint_equals_dequantized_twin(tmp_path, method):
     _write(dq, q_sd, method)
     _write(dt, twin_sd, None)
     assert _gen(dq) == _gen(dt)
+
+
+# --------------------------------------------------------- int4 MoE experts
+MOE_CFG = {
+    "architectures": ["Qwen2MoeForCausalLM"],
+    "model_type": "qwen2_moe",
+    "hidden_size": 64,
+    "intermediate_size": 128,
+    "num_hidden_layers": 2,
+    "num_attention_heads": 4,
+    "num_key_value_heads": 2,
+    "num_experts": 4,
+    "num_experts_per_tok": 2,
+    "moe_intermediate_size": 48,
+    "shared_expert_intermediate_size": 96,
+    "norm_topk_prob": False,
+    "decoder_sparse_step": 1,
+    "vocab_size": 128,
+    "max_position_embeddings": 2048,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 10000.0,
+    "tie_word_embeddings": False,
+    "eos_token_id": 0,
+}
+MOE_GROUP = 8  # moe_intermediate/tp = 24 stays group- and pack-aligned
+
+
+def _moe_state_dict():
+    g = torch.Generator().manual_seed(432)
+    c = MOE_CFG
+    H, V = c["hidden_size"], c["vocab_size"]
+    hd = H // c["num_attention_heads"]
+    kv = c["num_key_value_heads"] * hd
+    Im, Is = c["moe_intermediate_size"], c["shared_expert_intermediate_size"]
+    sd = {}
+
+    def rnd(*shape):
+        return torch.randn(*shape, generator=g) * 0.08
+
+    sd["model.embed_tokens.weight"] = rnd(V, H)
+    for L in range(c["num_hidden_layers"]):
+        p = f"model.layers.{L}."
+        sd[p + "self_attn.q_proj.weight"] = rnd(H, H)
+        sd[p + "self_attn.q_proj.bias"] = rnd(H)
+        sd[p + "self_attn.k_proj.weight"] = rnd(kv, H)
+        sd[p + "self_attn.k_proj.bias"] = rnd(kv)
+        sd[p + "self_attn.v_proj.weight"] = rnd(kv, H)
+        sd[p + "self_attn.v_proj.bias"] = rnd(kv)
+        sd[p + "self_attn.o_proj.weight"] = rnd(H, H)
+        sd[p + "mlp.gate.weight"] = rnd(c["num_experts"], H)
+        for e in range(c["num_experts"]):
+            ep = p + f"mlp.experts.{e}."
+            sd[ep + "gate_proj.weight"] = rnd(Im, H)
+            sd[ep + "up_proj.weight"] = rnd(Im, H)
+            sd[ep + "down_proj.weight"] = rnd(H, Im)
+        sd[p + "mlp.shared_expert.gate_proj.weight"] = rnd(Is, H)
+        sd[p + "mlp.shared_expert.up_proj.weight"] = rnd(Is, H)
+        sd[p + "mlp.shared_expert.down_proj.weight"] = rnd(H, Is)
+        sd[p + "mlp.shared_expert_gate.weight"] = rnd(1, H)
+        sd[p + "input_layernorm.weight"] = torch.ones(H) + rnd(H) * 0.05
+        sd[p + "post_attention_layernorm.weight"] = \
+            torch.ones(H) + rnd(H) * 0.05
+    sd["model.norm.weight"] = torch.ones(H) + rnd(H) * 0.05
+    sd["lm_head.weight"] = rnd(V, H)
+    return sd
+
+
+def _moe_is_quantized(name):
+    if not name.endswith(".weight"):
+        return False
+    if name.endswith("shared_expert_gate.weight") or \
+            name.endswith("mlp.gate.weight"):
+        return False  # routers stay dense
+    return any(f".{k}." in name for k in QUANT_KEYS)
+
+
+def _write_moe(d, sd, method):
+    os.makedirs(d, exist_ok=True)
+    cfg = dict(MOE_CFG)
+    if method:
+        cfg["quantization_config"] = {"quant_method": method, "bits": 4,
+                                      "group_size": MOE_GROUP}
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump(cfg, f)
+    from safetensors.torch import save_file
+    save_file(sd, os.path.join(d, "model.safetensors"))
+
+
+def _make_moe_checkpoints(tmp_path, method):
+    pack = pack_gptq if method == "gptq" else pack_awq
+    deq = dequant_gptq if method == "gptq" else dequant_awq
+    base = _moe_state_dict()
+    q_sd, twin_sd = {}, {}
+    for name, w in base.items():
+        if _moe_is_quantized(name):
+            qw, qz, s = pack(w, MOE_GROUP)
+            stem = name[:-len(".weight")]
+            q_sd[stem + ".qweight"] = qw
+            q_sd[stem + ".qzeros"] = qz
+            q_sd[stem + ".scales"] = s
+            twin_sd[name] = deq(qw, qz, s, MOE_GROUP, torch.float32)
+        else:
+            q_sd[name] = w
+            twin_sd[name] = w
+    dq = str(tmp_path / f"{method}_moe_q")
+    dt = str(tmp_path / f"{method}_moe_t")
+    _write_moe(dq, q_sd, method)
+    _write_moe(dt, twin_sd, None)
+    return dq, dt
+
+
+@pytest.mark.parametrize("method", ["gptq", "awq"])
+def test_int4_moe_checkpoint_equals_dequantized_twin(tmp_path, method):
+    """Packed int4 expert banks (w13/w2 qweight/qzeros/scales routed
+    through the experts.<e>.<proj>.<qtensor> names) must load and
+    execute exactly like the dequantized twin checkpoint."""
+    dq, dt = _make_moe_checkpoints(tmp_path, method)
+    assert _gen(dq) == _gen(dt)
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("method,port", [("gptq", 29772), ("awq", 29782)])
+def test_int4_moe_tp2_equals_single(tmp_path, method, port):
+    """TP=2 over the packed int4 MoE checkpoint — covers stored-N
+    slicing of the w13 banks and packed-K / whole-group slicing of the
+    w2 banks."""
+    import multiprocessing as mp
+    dq, _ = _make_moe_checkpoints(tmp_path, method)
+    ref = _gen(dq)
+    ctx = mp.get_context("spawn")
+    rq = ctx.Queue()
+    procs = [ctx.Process(target=_run_int4_tp_rank,
+                         args=(r, dq, port, rq)) for r in range(2)]
+    for p in procs:
+        p.start()
+    got = rq.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert got == ref
