@@ -45,20 +45,29 @@ class PPEngine:
         self.pp_size = P.get_pp_size()
         self.dp_rank = P.get_dp_rank()
         self.dp_size = P.get_dp_size()
-        # round-1 bound: DP attention (replica parallelism) is PP=1;
-        # multi-stage DP lockstep is in ROADMAP.md
-        assert self.dp_size == 1 or self.pp_size == 1, \
-            "dp_size > 1 requires pp_size == 1"
         self._dp_global_work = 1 if self.dp_size > 1 else 0
         self.is_first = P.is_first_pp_rank()
         self.is_last = P.is_last_pp_rank()
         self.device = config.device
         self.hidden = self.runner.hf_config.hidden_size
         self.dtype = config.torch_dtype()
-        # dedicated communicator for token broadcasts
+        # dedicated communicator for token broadcasts — one per DP
+        # replica (a replica's pipeline spans all pp stages x tp ranks
+        # of its dp index; tokens differ per replica, so the broadcast
+        # must not cross replicas)
         if P.get_world_size() > 1:
-            self.token_group = dist.new_group(list(range(P.get_world_size())))
-            self.last_stage_rank = (self.pp_size - 1) * config.stage_size
+            stage = config.stage_size
+            tokg = None
+            for dp in range(self.dp_size):
+                ranks = [pp * stage + dp * config.tp_size + t
+                         for pp in range(self.pp_size)
+                         for t in range(config.tp_size)]
+                g = dist.new_group(ranks)
+                if P.get_rank() in ranks:
+                    tokg = g
+            self.token_group = tokg
+            self.last_stage_rank = (self.pp_size - 1) * stage + \
+                self.dp_rank * config.tp_size
         else:
             self.token_group = None
         # in-flight (batch, pending-state) on this rank
@@ -131,12 +140,15 @@ class PPEngine:
     # ---------------------------------------------------- DP attention
     def dp_forward(self, batch: Optional[ScheduledBatch]) -> bool:
         """One lockstep DP-attention round (reference worker.py:750-889
-        re-shaped for replicated schedulers): every replica enters a
-        metadata barrier with (tokens scheduled this round, has pending
-        work); if any replica scheduled tokens, ALL replicas forward —
-        an idle one runs a 1-token dummy so the MoE DP-gather and EP
-        all-reduce collectives stay matched. Returns True if a forward
-        ran this round."""
+        _schedule_forward_dp{,_pp} re-shaped for replicated schedulers):
+        every replica enters a metadata barrier with (tokens scheduled
+        this round, has pending work); if any replica scheduled tokens,
+        ALL replicas forward — an idle one runs a 1-token dummy so the
+        MoE DP-gather and EP all-reduce collectives stay matched. Under
+        PP each stage runs the barrier on its OWN per-stage DP group
+        with identical (replicated-scheduler) inputs, so every stage of
+        a replica takes the same branch; dummies run each stage's local
+        layers with no p2p. Returns True if a forward ran this round."""
         P = self.P
         nt = batch.num_tokens if batch is not None else 0
         counts, flags = P.dp_meta_barrier(nt, self.scheduler.has_work())
@@ -148,8 +160,7 @@ class PPEngine:
             if batch is None:
                 self.runner.step_dummy()
             else:
-                out = self.runner.step_first_stage(batch)
-                self.inflight.append((batch, out))
+                self._launch(batch)
         finally:
             P.set_dp_forward_counts(None)
         return True

@@ -226,8 +226,17 @@ class ModelRunner:
                                          device=dev),
             max_query_len=T, max_seq_len=T, k_caches=[], v_caches=[],
             is_profile_run=True)
-        input_ids = torch.zeros(T, dtype=torch.long, device=dev)
-        self._stage_forward(input_ids, fctx.positions, fctx)
+        if self.model.is_first_stage:
+            input_ids = torch.zeros(T, dtype=torch.long, device=dev)
+            self._stage_forward(input_ids, fctx.positions, fctx)
+        else:
+            # DP under PP: idle mid/last stages run their local layers on
+            # zero hidden states (content irrelevant — only the MoE
+            # collectives must be entered)
+            z = torch.zeros(T, self.hf_config.hidden_size,
+                            dtype=self.config.torch_dtype(), device=dev)
+            self._stage_forward(None, fctx.positions, fctx,
+                                hidden_states=z, residual=z.clone())
 
     def _allocate_kv(self, num_pages: int):
         spec = self.kv_spec()

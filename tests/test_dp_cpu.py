@@ -268,3 +268,58 @@ def test_abort_unknown_seq_id_is_ignored():
     sched.add_seqs([seq])
     b = sched.schedule_once()
     assert b is not None and not seq.finish_reason
+
+
+def _run_dp_pp_rank(rank, model_dir, port, q):
+    os.environ.update(RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    torch.set_num_threads(1)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.pp_engine import PPEngine
+    from gllm_amd.sequence import SamplingParams, Sequence
+    cfg = EngineConfig(model=model_dir, load_format="auto", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       pp_size=2, dp_size=2, use_ep=True, master_port=port,
+                       enable_prefix_caching=False)
+    eng = PPEngine(cfg, num_pages_override=128)
+    assert eng.dp_size == 2 and eng.pp_size == 2
+    # every stage of a replica adds the replica's round-robin share
+    # (replicated schedulers)
+    mine = [(i, p) for i, p in enumerate(PROMPTS) if i % 2 == eng.dp_rank]
+    seqs = [Sequence(i, p, SamplingParams(temperature=0.0,
+                                          max_tokens=MAX_TOKENS[i],
+                                          ignore_eos=True))
+            for i, p in mine]
+    eng.add_requests(seqs)
+    eng.run_until_done()
+    q.put((rank, [(s.seq_id, s.output_token_ids) for s in seqs]))
+    import torch.distributed as dist
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_pp2_ep_moe_equals_single(tmp_path):
+    """DP across a 2-stage pipeline (reference worker.py
+    _schedule_forward_dp_pp): per-stage DP meta barriers, stage-local
+    dummy forwards on idle replicas, per-replica token broadcast
+    groups. Skewed workloads exercise the dummy path on both stages."""
+    d = str(tmp_path / "ckpt")
+    _write_checkpoint(d)
+    ref = _single_reference(d)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_dp_pp_rank, args=(r, d, 29763, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    got = {}
+    for _ in range(4):
+        rank, outs = q.get(timeout=240)
+        for seq_id, toks in outs:
+            got.setdefault(seq_id, toks)
+            assert got[seq_id] == toks, f"rank disagreement on {seq_id}"
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert [got[i] for i in range(len(PROMPTS))] == ref
