@@ -8,8 +8,9 @@ from typing import List, Optional, Tuple
 
 
 class IncrementalDetokenizer:
-    def __init__(self, tokenizer):
+    def __init__(self, tokenizer, skip_special_tokens: bool = True):
         self.tok = tokenizer
+        self.skip_special = skip_special_tokens
         self.token_ids: List[int] = []
         self.prefix_offset = 0
         self.read_offset = 0
@@ -20,9 +21,9 @@ class IncrementalDetokenizer:
         self.token_ids.append(token_id)
         prefix_text = self.tok.decode(
             self.token_ids[self.prefix_offset:self.read_offset],
-            skip_special_tokens=False)
+            skip_special_tokens=self.skip_special)
         new_text = self.tok.decode(self.token_ids[self.prefix_offset:],
-                                   skip_special_tokens=False)
+                                   skip_special_tokens=self.skip_special)
         if new_text.endswith("�"):
             # incomplete byte sequence: hold back
             return ""

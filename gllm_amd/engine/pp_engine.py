@@ -68,8 +68,13 @@ class PPEngine:
             self.token_group = tokg
             self.last_stage_rank = (self.pp_size - 1) * stage + \
                 self.dp_rank * config.tp_size
+            # all-ranks control channel (serving intake broadcast from
+            # global rank 0) — kept off the p2p default group
+            self.ctrl_group = dist.new_group(
+                list(range(P.get_world_size())))
         else:
             self.token_group = None
+            self.ctrl_group = None
         # in-flight (batch, pending-state) on this rank
         self.inflight = deque()
 

@@ -41,7 +41,10 @@ class RequestState:
         self.seq_id = seq_id
         self.prompt_len = prompt_len
         self.sampling = sampling
-        self.detok = IncrementalDetokenizer(tokenizer) if tokenizer else None
+        self.detok = IncrementalDetokenizer(
+            tokenizer, skip_special_tokens=getattr(
+                sampling, "skip_special_tokens", True)) \
+            if tokenizer else None
         self.queue: asyncio.Queue = asyncio.Queue()
         self.loop = loop
         self.n_tokens = 0

@@ -123,10 +123,10 @@ class ServingMixin:
                 self.comm.poll(100)
             msgs = self.comm.drain()
             self._intake_buf.fill_(len(msgs))
-            dist.broadcast(self._intake_buf, src=0, group=self.token_group)
+            dist.broadcast(self._intake_buf, src=0, group=self.ctrl_group)
             self._apply_messages(msgs)
         else:
-            dist.broadcast(self._intake_buf, src=0, group=self.token_group)
+            dist.broadcast(self._intake_buf, src=0, group=self.ctrl_group)
             n = int(self._intake_buf.item())
             if n:
                 self._apply_messages(self.comm.recv_blocking(n))

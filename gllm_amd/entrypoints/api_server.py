@@ -105,6 +105,23 @@ def build_app():
         lp = req.logprobs
         if isinstance(lp, bool):  # chat: bool + top_logprobs count
             lp = (getattr(req, "top_logprobs", None) or 1) if lp else None
+        bias = None
+        if getattr(req, "logit_bias", None):
+            bias = {int(k): float(v) for k, v in req.logit_bias.items()}
+        bad_words_ids = None
+        if getattr(req, "bad_words", None):
+            tok = engine.tokenizer
+            if tok is None:
+                raise ValueError("bad_words needs a tokenizer")
+            bad_words_ids = []
+            for w in req.bad_words:
+                # ban both the word-initial and the mid-text piece
+                # sequences (tokenizers split " word" and "word"
+                # differently)
+                for v in {w, " " + w}:
+                    ids = tok.encode(v, add_special_tokens=False)
+                    if ids:
+                        bad_words_ids.append(list(ids))
         return SamplingParams(
             temperature=req.temperature if req.temperature is not None
             else 1.0,
@@ -123,7 +140,18 @@ def build_app():
                 req, "include_stop_str_in_output", False),
             seed=req.seed,
             logprobs=lp,
-            prompt_logprobs=getattr(req, "prompt_logprobs", None))
+            prompt_logprobs=getattr(req, "prompt_logprobs", None),
+            logit_bias=bias,
+            allowed_token_ids=getattr(req, "allowed_token_ids", None),
+            bad_words_token_ids=bad_words_ids,
+            skip_special_tokens=getattr(req, "skip_special_tokens", True))
+
+    def _truncate(req, token_ids):
+        """OpenAI truncate_prompt_tokens: keep only the LAST t prompt
+        tokens (skipped for multimodal prompts — cutting an image span
+        would desync the embedding merge)."""
+        t = getattr(req, "truncate_prompt_tokens", None)
+        return token_ids[-t:] if t else token_ids
 
     def _vary_seed(sampling: SamplingParams, j: int) -> SamplingParams:
         """Choice j of an n>1 request: distinct seed per choice when one
@@ -148,6 +176,8 @@ def build_app():
             import asyncio as _aio
             token_ids, mm = await _aio.to_thread(
                 engine.process_images, token_ids, images)
+            if not mm:
+                token_ids = _truncate(req, token_ids)
         except Exception as e:
             return JSONResponse(status_code=400,
                                 content={"error": str(e)})
@@ -302,6 +332,7 @@ def build_app():
         if req.stream:
             token_ids = prompts[0] if isinstance(prompts[0], list) \
                 else engine.encode(prompts[0])
+            token_ids = _truncate(req, token_ids)
             return StreamingResponse(
                 _completion_stream(req, raw, token_ids, sampling),
                 media_type="text/event-stream")
@@ -310,6 +341,7 @@ def build_app():
         total_p = total_c = 0
         for i, p in enumerate(prompts):
             token_ids = p if isinstance(p, list) else engine.encode(p)
+            token_ids = _truncate(req, token_ids)
             import asyncio as _aio
             results = await _aio.gather(*[
                 _collect(raw, token_ids, _vary_seed(sampling, j))

@@ -68,6 +68,11 @@ class ChatCompletionRequest(BaseModel):
     chat_template_kwargs: Optional[Dict[str, Any]] = None
     prompt_logprobs: Optional[int] = None
     include_stop_str_in_output: bool = False
+    logit_bias: Optional[Dict[str, float]] = None
+    allowed_token_ids: Optional[List[int]] = None
+    bad_words: Optional[List[str]] = None
+    skip_special_tokens: bool = True
+    truncate_prompt_tokens: Optional[int] = None
 
 
 class CompletionRequest(BaseModel):
@@ -92,6 +97,11 @@ class CompletionRequest(BaseModel):
     logprobs: Optional[int] = None
     prompt_logprobs: Optional[int] = None
     include_stop_str_in_output: bool = False
+    logit_bias: Optional[Dict[str, float]] = None
+    allowed_token_ids: Optional[List[int]] = None
+    bad_words: Optional[List[str]] = None
+    skip_special_tokens: bool = True
+    truncate_prompt_tokens: Optional[int] = None
 
 
 class UsageInfo(BaseModel):

@@ -33,6 +33,12 @@ class SamplingMetadata:
     penalty_slots: Optional[torch.Tensor] = None   # [B] long
     # rows that actually sample a kept token (ends_prompt), host list
     sample_rows: Optional[List[int]] = None
+    # per-row logits adjustments (host-built, device tensors):
+    # logit_bias additive entries, allowed-token whitelists and
+    # bad-words bans (-inf outside/on the listed ids)
+    bias_rows: Optional[List] = None     # [(row, ids, vals)]
+    allowed_rows: Optional[List] = None  # [(row, ids)]
+    ban_rows: Optional[List] = None      # [(row, ids)]
     max_logprobs: int = 0               # >0 => return top-k logprobs
     generators: Optional[List[Optional[torch.Generator]]] = None
 
@@ -51,6 +57,8 @@ class Sampler(torch.nn.Module):
         # logits: [B, V] (already gathered to full vocab)
         if meta.any_penalty:
             logits = self._apply_penalties(logits, meta)
+        if meta.bias_rows or meta.allowed_rows or meta.ban_rows:
+            logits = self._apply_row_adjust(logits, meta)
         if meta.all_greedy:
             next_tokens = logits.argmax(dim=-1)
             return self._with_logprobs(logits, next_tokens, meta)
@@ -101,6 +109,25 @@ class Sampler(torch.nn.Module):
                 seen = counts > 0
                 logits[row] = logits[row] - pres * seen.float() \
                     - freq * counts.float()
+        return logits
+
+    @staticmethod
+    def _apply_row_adjust(logits: torch.Tensor,
+                          meta: SamplingMetadata) -> torch.Tensor:
+        """logit_bias / allowed_token_ids / bad_words, applied before
+        the greedy argmax and the softmax alike."""
+        logits = logits.float()
+        if meta.bias_rows:
+            for row, ids, vals in meta.bias_rows:
+                logits[row, ids] += vals
+        if meta.allowed_rows:
+            for row, ids in meta.allowed_rows:
+                keep = logits[row, ids].clone()
+                logits[row] = float("-inf")
+                logits[row, ids] = keep
+        if meta.ban_rows:
+            for row, ids in meta.ban_rows:
+                logits[row, ids] = float("-inf")
         return logits
 
     @staticmethod
@@ -175,8 +202,30 @@ def build_sampling_metadata(items, device,
     any_pen = False
     need_rows = False
     pres_freq = []
+    bias_rows, allowed_rows, ban_rows = [], [], []
     for i, it in enumerate(items):
         sp = it.seq.sampling
+        if sp.logit_bias:
+            ids = torch.tensor(list(sp.logit_bias.keys()),
+                               dtype=torch.long, device=device)
+            vals = torch.tensor(list(sp.logit_bias.values()),
+                                dtype=torch.float32, device=device)
+            bias_rows.append((i, ids, vals))
+        if sp.allowed_token_ids:
+            allowed_rows.append((i, torch.tensor(
+                sp.allowed_token_ids, dtype=torch.long, device=device)))
+        if sp.bad_words_token_ids:
+            ctx = it.seq.token_ids
+            banned = set()
+            for w in sp.bad_words_token_ids:
+                if not w:
+                    continue
+                if len(w) == 1 or (len(w) - 1 <= len(ctx) and
+                                   ctx[len(ctx) - len(w) + 1:] == w[:-1]):
+                    banned.add(w[-1])
+            if banned:
+                ban_rows.append((i, torch.tensor(
+                    sorted(banned), dtype=torch.long, device=device)))
         temps.append(sp.temperature)
         tps.append(sp.top_p)
         tks.append(sp.top_k)
@@ -228,5 +277,8 @@ def build_sampling_metadata(items, device,
         if any_pen else None,
         sample_rows=sample_rows,
         pres_freq_rows=pres_freq or None,
+        bias_rows=bias_rows or None,
+        allowed_rows=allowed_rows or None,
+        ban_rows=ban_rows or None,
         max_logprobs=max_lp,
         generators=gens if any(g is not None for g in gens) else None)

@@ -8,7 +8,7 @@ with explicit SamplingParams.
 """
 
 import dataclasses
-from typing import List, Optional
+from typing import Dict, List, Optional
 
 
 @dataclasses.dataclass
@@ -29,6 +29,15 @@ class SamplingParams:
     logprobs: Optional[int] = None     # top-k logprobs to return per token
     prompt_logprobs: Optional[int] = None
     seed: Optional[int] = None
+    # OpenAI logit_bias: token id -> additive bias (applied every step)
+    logit_bias: Optional[Dict[int, float]] = None
+    # restrict sampling to this token set (OpenAI allowed_token_ids)
+    allowed_token_ids: Optional[List[int]] = None
+    # banned token SEQUENCES (pre-tokenized bad_words): the last token
+    # of a sequence is masked whenever the preceding tokens match the
+    # tail of the generated context
+    bad_words_token_ids: Optional[List[List[int]]] = None
+    skip_special_tokens: bool = True   # detokenization flag
 
     @property
     def is_greedy(self) -> bool:

@@ -168,3 +168,95 @@ def test_api_token_logprobs(tmp_path):
         assert len(with_lp) == 3
     finally:
         srv.engine.stop()
+
+
+@pytest.mark.timeout(300)
+def test_api_sampling_controls(tmp_path):
+    """logit_bias / allowed_token_ids / bad_words / skip_special_tokens /
+    truncate_prompt_tokens are SERVED (change the sampled stream), not
+    just accepted."""
+    from fastapi.testclient import TestClient
+    import gllm_amd.entrypoints.api_server as srv
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+
+    d = _mk_model_dir(tmp_path)
+    cfg = EngineConfig(model=d, load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       master_port=29692,
+                       enable_prefix_caching=False)
+    srv.engine = AsyncLLMEngine(cfg)
+    srv.served_model = "tiny"
+    srv.engine.start()
+    try:
+        app = srv.build_app()
+        client = TestClient(app)
+
+        # allowed_token_ids: greedy decode restricted to one token
+        r = client.post("/v1/completions", json={
+            "prompt": "w3 w4", "max_tokens": 4, "temperature": 0.0,
+            "ignore_eos": True, "allowed_token_ids": [42]})
+        assert r.status_code == 200, r.text
+        assert r.json()["choices"][0]["text"].split() == ["w42"] * 4
+
+        # logit_bias: +100 forces the token everywhere
+        r = client.post("/v1/completions", json={
+            "prompt": "w3 w4", "max_tokens": 3, "temperature": 0.0,
+            "ignore_eos": True, "logit_bias": {"37": 100.0}})
+        assert r.json()["choices"][0]["text"].split() == ["w37"] * 3
+
+        # bad_words: the unconstrained first token must change
+        base = client.post("/v1/completions", json={
+            "prompt": "w3 w4", "max_tokens": 1, "temperature": 0.0,
+            "ignore_eos": True}).json()["choices"][0]["text"].strip()
+        banned = client.post("/v1/completions", json={
+            "prompt": "w3 w4", "max_tokens": 1, "temperature": 0.0,
+            "ignore_eos": True,
+            "bad_words": [base]}).json()["choices"][0]["text"].strip()
+        assert banned != base
+
+        # skip_special_tokens: force the eos token into the stream
+        body = {"prompt": "w3 w4", "max_tokens": 2, "temperature": 0.0,
+                "ignore_eos": True, "logit_bias": {"1": 100.0}}
+        hide = client.post("/v1/completions", json=body).json()
+        assert "</s>" not in hide["choices"][0]["text"]
+        show = client.post("/v1/completions",
+                           json={**body, "skip_special_tokens": False}
+                           ).json()
+        assert "</s>" in show["choices"][0]["text"]
+
+        # truncate_prompt_tokens: usage reflects the kept tail
+        r = client.post("/v1/completions", json={
+            "prompt": "w3 w4 w5 w6 w7", "max_tokens": 1,
+            "temperature": 0.0, "ignore_eos": True,
+            "truncate_prompt_tokens": 2})
+        assert r.json()["usage"]["prompt_tokens"] == 2
+    finally:
+        srv.engine.stop()
+
+
+def test_bad_words_sequence_matching():
+    """Multi-token bad words ban only the final token and only when the
+    preceding context matches (vLLM semantics)."""
+    from types import SimpleNamespace
+    import torch
+    from gllm_amd.layers.sampler import Sampler, build_sampling_metadata
+    from gllm_amd.sequence import SamplingParams
+
+    def mk_item(ctx, bad):
+        sp = SamplingParams(temperature=0.0, bad_words_token_ids=bad)
+        seq = SimpleNamespace(sampling=sp, token_ids=list(ctx),
+                              prompt_len=len(ctx), num_output_tokens=0)
+        return SimpleNamespace(seq=seq, ends_prompt=True)
+
+    V = 16
+    logits = torch.zeros(2, V)
+    logits[:, 7] = 5.0          # argmax would be 7 everywhere
+    logits[:, 3] = 4.0          # runner-up
+    # row 0: context ends with [5, 6] and bad word is [5, 6, 7] -> 7 banned
+    # row 1: context does NOT match -> 7 allowed
+    items = [mk_item([1, 5, 6], [[5, 6, 7]]),
+             mk_item([1, 2, 4], [[5, 6, 7]])]
+    meta = build_sampling_metadata(items, "cpu")
+    out = Sampler()(logits.clone(), meta)
+    assert out.next_tokens.tolist() == [3, 7]
