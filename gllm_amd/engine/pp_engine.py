@@ -140,7 +140,43 @@ class PPEngine:
             dist.broadcast(tok, src=self.last_stage_rank,
                            group=self.token_group)
             tokens = tok.tolist()
-        return self.scheduler.process_output(batch, tokens)
+            # per-token logprobs ride the same channel when requested
+            # (reference worker.py:682-700 sends (next_tokens, logprobs,
+            # prompt_logprobs) across stages); every rank derives the
+            # same max-k from its replicated scheduler state
+            max_lp = max((it.seq.sampling.logprobs or 0)
+                         for it in batch.items)
+            if max_lp > 0:
+                from gllm_amd.layers.sampler import SamplerOutput
+                if self.is_last:
+                    lp = out.logprobs.float().to(self.device)
+                    topv = out.topk_logprobs.float().to(self.device)
+                    topi = out.topk_token_ids.to(self.device)
+                else:
+                    lp = torch.empty(B, dtype=torch.float32,
+                                     device=self.device)
+                    topv = torch.empty(B, max_lp, dtype=torch.float32,
+                                       device=self.device)
+                    topi = torch.empty(B, max_lp, dtype=torch.long,
+                                       device=self.device)
+                for t in (lp, topv, topi):
+                    dist.broadcast(t, src=self.last_stage_rank,
+                                   group=self.token_group)
+                self._stash_logprobs(batch, SamplerOutput(tok, lp, topv,
+                                                          topi))
+        finished = self.scheduler.process_output(batch, tokens)
+        if self.pp_size > 1:
+            # prompt logprobs accumulate on the last stage (where
+            # sampling runs); ship them to the other stages at finish
+            for s in finished:
+                if s.sampling.prompt_logprobs:
+                    obj = [s.prompt_logprobs_out if self.is_last else None]
+                    dist.broadcast_object_list(
+                        obj, src=self.last_stage_rank,
+                        group=self.token_group)
+                    if not self.is_last:
+                        s.prompt_logprobs_out = obj[0]
+        return finished
 
     # ---------------------------------------------------- DP attention
     def dp_forward(self, batch: Optional[ScheduledBatch]) -> bool:

@@ -45,10 +45,11 @@ class ServingMixin:
         self.world = get_world_size()
         self.dp_rank = get_dp_rank()
         self.dp_size = get_dp_size()
-        # DP: every replica emits its own outputs (tp rank 0 of the
-        # replica; DP requires pp == 1). Non-DP: global rank 0.
+        # DP: every replica emits its own outputs (pp-0/tp-0 rank of
+        # the replica). Non-DP: global rank 0.
         if self.dp_size > 1:
-            self.is_output_rank = get_tp_rank() == 0
+            from gllm_amd.parallel import get_pp_rank
+            self.is_output_rank = get_tp_rank() == 0 and get_pp_rank() == 0
         else:
             self.is_output_rank = self.rank == 0
         self._req_counter = 0  # deterministic DP round-robin routing

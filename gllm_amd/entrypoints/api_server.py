@@ -96,12 +96,6 @@ def build_app():
         rp = req.repetition_penalty
         if rp is None:
             rp = 1.0
-        if getattr(req, "prompt_logprobs", None) and \
-                engine.config.pp_size > 1:
-            # the values are computed where sampling runs (last stage);
-            # shipping them across stages is a round-2 item
-            raise ValueError(
-                "prompt_logprobs requires pp_size == 1")
         lp = req.logprobs
         if isinstance(lp, bool):  # chat: bool + top_logprobs count
             lp = (getattr(req, "top_logprobs", None) or 1) if lp else None
