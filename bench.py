@@ -46,6 +46,70 @@ SMALL_DEBUG = {  # --model debug: quick bring-up config
     "num_key_value_heads": 2, "vocab_size": 32000,
 }
 
+MIXTRAL_8X7B = {  # BASELINE config: Mixtral 8x7B PP + EP over xGMI
+    "architectures": ["MixtralForCausalLM"],
+    "model_type": "mixtral",
+    "hidden_size": 4096,
+    "intermediate_size": 14336,
+    "num_hidden_layers": 32,
+    "num_attention_heads": 32,
+    "num_key_value_heads": 8,
+    "num_local_experts": 8,
+    "num_experts_per_tok": 2,
+    "vocab_size": 32000,
+    "max_position_embeddings": 32768,
+    "rms_norm_eps": 1e-5,
+    "rope_theta": 1000000.0,
+    "tie_word_embeddings": False,
+    "eos_token_id": 2,
+}
+
+MIXTRAL_DEBUG = {  # CPU-testable MoE path
+    **MIXTRAL_8X7B,
+    "hidden_size": 512, "intermediate_size": 1024,
+    "num_hidden_layers": 4, "num_attention_heads": 8,
+    "num_key_value_heads": 2, "vocab_size": 32000,
+}
+
+DEEPSEEK_V3 = {  # BASELINE config: DeepSeek-V3 PP=8 + EP, absorbed MLA
+    "architectures": ["DeepseekV3ForCausalLM"],
+    "model_type": "deepseek_v3",
+    "hidden_size": 7168,
+    "intermediate_size": 18432,
+    "moe_intermediate_size": 2048,
+    "num_hidden_layers": 61,
+    "first_k_dense_replace": 3,
+    "num_attention_heads": 128,
+    "num_key_value_heads": 128,
+    "n_routed_experts": 256,
+    "n_shared_experts": 1,
+    "num_experts_per_tok": 8,
+    "n_group": 8,
+    "topk_group": 4,
+    "routed_scaling_factor": 2.5,
+    "scoring_func": "sigmoid",
+    "norm_topk_prob": True,
+    "q_lora_rank": 1536,
+    "kv_lora_rank": 512,
+    "qk_nope_head_dim": 128,
+    "qk_rope_head_dim": 64,
+    "v_head_dim": 128,
+    "vocab_size": 129280,
+    "max_position_embeddings": 163840,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 10000.0,
+    "tie_word_embeddings": False,
+    "eos_token_id": 1,
+}
+
+MODELS = {
+    "qwen2.5-32b": ("Qwen2.5-32B", QWEN25_32B),
+    "debug": ("debug-0.2B", SMALL_DEBUG),
+    "mixtral-8x7b": ("Mixtral-8x7B", MIXTRAL_8X7B),
+    "mixtral-debug": ("mixtral-debug", MIXTRAL_DEBUG),
+    "deepseek-v3": ("DeepSeek-V3", DEEPSEEK_V3),
+}
+
 
 def write_model_dir(cfg_json):
     d = tempfile.mkdtemp(prefix="bench_model_")
@@ -62,7 +126,13 @@ def main():
     ap.add_argument("--batch", type=int, default=256)
     ap.add_argument("--prompt-len", type=int, default=1024)
     ap.add_argument("--model", type=str, default="qwen2.5-32b",
-                    choices=["qwen2.5-32b", "debug"])
+                    choices=sorted(MODELS))
+    ap.add_argument("--tp", type=int, default=1,
+                    help="TP degree; PP = gpus // tp (BASELINE's MoE "
+                         "configs pair PP with EP over the TP ranks)")
+    ap.add_argument("--use-ep", action="store_true",
+                    help="expert parallelism over the dp*tp ranks of "
+                         "each stage")
     ap.add_argument("--page-size", type=int, default=16)
     ap.add_argument("--schedule", type=str, default="token_throttling")
     ap.add_argument("--qps", type=float, default=0.0,
@@ -82,8 +152,10 @@ def main():
     use_gpu = torch.cuda.is_available()
     device = f"cuda:{int(os.environ.get('LOCAL_RANK', rank))}" if use_gpu \
         else "cpu"
-    model_json = QWEN25_32B if args.model == "qwen2.5-32b" else SMALL_DEBUG
+    model_name, model_json = MODELS[args.model]
     model_dir = write_model_dir(model_json)
+    assert n % args.tp == 0, (n, args.tp)
+    pp = n // args.tp
 
     from gllm_amd.config import EngineConfig
     from gllm_amd.engine.overlap_engine import OverlapEngine
@@ -93,7 +165,8 @@ def main():
     cfg = EngineConfig(
         model=model_dir, load_format="dummy",
         dtype="bfloat16" if use_gpu else "float32",
-        device=device, pp_size=n, page_size=args.page_size,
+        device=device, pp_size=pp, tp_size=args.tp, use_ep=args.use_ep,
+        page_size=args.page_size,
         schedule_method=args.schedule, maxp=8192, maxd=1024,
         use_graph=not args.no_graph, max_graph_bs=args.max_graph_bs,
         enable_prefix_caching=False,
@@ -197,11 +270,12 @@ def main():
             "ttft_p50_ms": round(ttft_p50, 1) if ttft_p50 else None,
             "ttft_qps": args.qps if args.qps > 0 else "burst",
             "config": {
-                "model": "Qwen2.5-32B" if args.model == "qwen2.5-32b"
-                else "debug-0.2B",
+                "model": model_name,
                 "global_batch": args.batch,
                 "seq_len": args.prompt_len,
-                "parallelism": f"pp{n}",
+                "parallelism": f"pp{pp}" +
+                (f"tp{args.tp}" if args.tp > 1 else "") +
+                ("ep" if args.use_ep else ""),
                 "schedule": args.schedule,
             },
         }
