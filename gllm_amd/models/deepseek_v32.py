@@ -117,13 +117,14 @@ class DSAMLAAttention(MLAAttention):
         dev = hidden.device
         topk = self.indexer.index_topk
         out = torch.full((T, topk), -1, dtype=torch.int32, device=dev)
-        qsl = fctx.query_start_loc
-        for b in range(fctx.seq_lens.shape[0]):
-            qs, qe = int(qsl[b]), int(qsl[b + 1])
+        qsl = fctx.host_qsl()
+        host_lens = fctx.host_seq_lens()
+        for b in range(len(host_lens)):
+            qs, qe = qsl[b], qsl[b + 1]
             q_len = qe - qs
             if q_len == 0:
                 continue
-            s_len = int(fctx.seq_lens[b])
+            s_len = host_lens[b]
             n_pages = -(-s_len // page_sz)
             pages = fctx.block_table[b, :n_pages].long()
             keys = cache[pages].reshape(-1, cache.shape[2])[:s_len]

@@ -145,9 +145,9 @@ class GatedDeltaNet(nn.Module):
 
         out_core = torch.empty(T, self.tp_v, self.head_v_dim,
                                dtype=hidden.dtype, device=hidden.device)
-        qsl = fctx.query_start_loc
+        qsl = fctx.host_qsl()
         for i in range(len(fctx.ssm_slots)):
-            s, e = int(qsl[i]), int(qsl[i + 1])
+            s, e = qsl[i], qsl[i + 1]
             slot = int(fctx.ssm_slots[i])
             has_init = bool(fctx.ssm_has_init[i])
             if not has_init:

@@ -91,6 +91,8 @@ class BatchBuilder:
             v_caches=v_caches,
         )
         fctx.has_placeholders = has_ph
+        fctx.seq_lens_cpu = seq_lens.tolist()
+        fctx.query_start_loc_cpu = qsl.tolist()
         if need_logits:
             fctx.logits_indices = (fctx.query_start_loc[1:].long() - 1)
         return tokens_t, fctx

@@ -46,7 +46,23 @@ class ForwardContext:
     # DSA (DeepSeek-V3.2): per-layer paged index-K caches
     # [pages, page_size, index_head_dim] (models/deepseek_v32.py)
     idx_caches: Optional[List[torch.Tensor]] = None
+    # host copies of the ragged geometry (filled by the batch builder
+    # from its numpy staging for free) so per-seq torch paths — DSA
+    # selection, GDN chunking, prompt logprobs — iterate without a
+    # device sync per int() read
+    seq_lens_cpu: Optional[List[int]] = None
+    query_start_loc_cpu: Optional[List[int]] = None
 
     @property
     def is_pure_decode(self) -> bool:
         return self.max_query_len == 1
+
+    def host_qsl(self) -> List[int]:
+        if self.query_start_loc_cpu is None:
+            self.query_start_loc_cpu = self.query_start_loc.tolist()
+        return self.query_start_loc_cpu
+
+    def host_seq_lens(self) -> List[int]:
+        if self.seq_lens_cpu is None:
+            self.seq_lens_cpu = self.seq_lens.tolist()
+        return self.seq_lens_cpu

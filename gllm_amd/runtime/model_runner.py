@@ -380,13 +380,13 @@ class ModelRunner:
         if not want:
             return
         import types
-        qsl = fctx.query_start_loc
+        qsl = fctx.host_qsl()
         shim = types.SimpleNamespace(logits_indices=None)
         for i in want:
             item = batch.items[i]
             seq = item.seq
             k = seq.sampling.prompt_logprobs
-            qs, qe = int(qsl[i]), int(qsl[i + 1])
+            qs, qe = qsl[i], qsl[i + 1]
             n_rows = qe - qs
             # row r predicts prompt position item.start + r + 1
             last = min(n_rows, seq.prompt_len - 1 - item.start)
