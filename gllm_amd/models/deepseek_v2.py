@@ -160,7 +160,9 @@ class MLAAttention(nn.Module):
         topk_pos = self._dsa_select(positions, hidden, q_resid, fctx)
         out = ops.mla_paged_attention(
             qf, k_cache, v_cache, fctx.block_table, fctx.seq_lens,
-            fctx.query_start_loc, self.scale, topk_positions=topk_pos)
+            fctx.query_start_loc, self.scale, topk_positions=topk_pos,
+            seq_lens_cpu=fctx.host_seq_lens(),
+            query_start_loc_cpu=fctx.host_qsl())
         return self.o_proj(out.reshape(T, -1))
 
     def _forward_absorbed(self, positions, hidden, q_resid, q_nope, q_pe,
@@ -183,7 +185,9 @@ class MLAAttention(nn.Module):
         topk_pos = self._dsa_select(positions, hidden, q_resid, fctx)
         out_lat = ops.mla_paged_attention(
             qf, k_cache, v_cache, fctx.block_table, fctx.seq_lens,
-            fctx.query_start_loc, self.scale, topk_positions=topk_pos)
+            fctx.query_start_loc, self.scale, topk_positions=topk_pos,
+            seq_lens_cpu=fctx.host_seq_lens(),
+            query_start_loc_cpu=fctx.host_qsl())
         # out[t,h,v] = sum_l out_lat[t,h,l] * W_UV[h,v,l]
         out = torch.einsum("thl,hvl->thv", out_lat.float(),
                            w_uv.float()).to(out_lat.dtype)

@@ -169,7 +169,8 @@ _MLA_GPU_WARNED = False
 
 
 def mla_paged_attention(q, k_cache, v_cache, block_table, seq_lens,
-                        query_start_loc, scale, topk_positions=None):
+                        query_start_loc, scale, topk_positions=None,
+                        seq_lens_cpu=None, query_start_loc_cpu=None):
     global _MLA_GPU_WARNED
     if q.is_cuda and not _MLA_GPU_WARNED:
         # NOT a silent fallback: no native MLA kernel exists yet — the
@@ -179,9 +180,10 @@ def mla_paged_attention(q, k_cache, v_cache, block_table, seq_lens,
         logger.warning("absorbed-MLA attention runs the torch path on "
                        "GPU (gfx950 MLA kernel lands in round 2)")
         _MLA_GPU_WARNED = True
-    return torch_ref.mla_paged_attention(q, k_cache, v_cache, block_table,
-                                         seq_lens, query_start_loc, scale,
-                                         topk_positions=topk_positions)
+    return torch_ref.mla_paged_attention(
+        q, k_cache, v_cache, block_table, seq_lens, query_start_loc,
+        scale, topk_positions=topk_positions, seq_lens_cpu=seq_lens_cpu,
+        query_start_loc_cpu=query_start_loc_cpu)
 
 
 # --------------------------------------------------------------- gemm

@@ -136,7 +136,9 @@ def mla_paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
                         seq_lens: torch.Tensor,
                         query_start_loc: torch.Tensor,
                         scale: float,
-                        topk_positions: Optional[torch.Tensor] = None
+                        topk_positions: Optional[torch.Tensor] = None,
+                        seq_lens_cpu: Optional[list] = None,
+                        query_start_loc_cpu: Optional[list] = None
                         ) -> torch.Tensor:
     """Varlen causal attention with asymmetric head dims (MLA):
     q/k have Dk (nope+rope), v has Dv. Same paged layout as
@@ -153,12 +155,15 @@ def mla_paged_attention(q: torch.Tensor, k_cache: torch.Tensor,
     page_size = k_cache.shape[1]
     out = torch.empty(T, H, Dv, dtype=q.dtype, device=q.device)
     B = seq_lens.shape[0]
+    # host geometry (when the caller has it) avoids 2B+1 device syncs
+    qsl_h = query_start_loc_cpu or query_start_loc.tolist()
+    lens_h = seq_lens_cpu or seq_lens.tolist()
     for b in range(B):
-        qs, qe = int(query_start_loc[b]), int(query_start_loc[b + 1])
+        qs, qe = qsl_h[b], qsl_h[b + 1]
         q_len = qe - qs
         if q_len == 0:
             continue
-        s_len = int(seq_lens[b])
+        s_len = lens_h[b]
         n_pages = -(-s_len // page_size)
         pages = block_table[b, :n_pages].long()
         Hkv = k_cache.shape[2]
