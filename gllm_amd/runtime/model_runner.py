@@ -66,6 +66,14 @@ class ModelRunner:
             # incompatible with capture (graph-captured DP decode with
             # uniform gather is a round-2 item)
             cfg.use_graph = False
+        else:
+            from gllm_amd.layers.moe.layer import FusedMoE
+            if any(isinstance(m, FusedMoE) for m in self.model.modules()):
+                # the round-1 expert dispatch reads the per-expert
+                # segment table on the host (an illegal sync under
+                # stream capture); the graph-safe path is the round-2
+                # grouped MFMA GEMM over sorted tokens
+                cfg.use_graph = False
         num_pages = num_pages_override or self._size_kv_cache()
         self._allocate_kv(num_pages)
         self.num_kv_pages_total = num_pages
