@@ -416,6 +416,11 @@ def make_arg_parser():
     p.add_argument("--discovery-addr", type=str, default=None,
                    help="discovery server 'host:port' to resolve an "
                         "encoder from")
+    p.add_argument("--mla-mode", choices=["absorbed", "decompressed"],
+                   default="absorbed",
+                   help="MLA execution form (reference --mla-backend): "
+                        "absorbed = 576-dim latent MQA cache, "
+                        "decompressed = per-head K/V")
     p.add_argument("--seed", type=int, default=0)
     return p
 
@@ -438,6 +443,7 @@ def config_from_args(args) -> EngineConfig:
         relay_port=args.relay_port,
         mm_encoder_addr=args.mm_encoder_addr,
         discovery_addr=args.discovery_addr, seed=args.seed,
+        mla_mode=args.mla_mode,
         device="cuda" if _has_gpu() else "cpu")
 
 
