@@ -94,6 +94,34 @@ def test_serving_dp2_replicas(tmp_path):
         eng.stop()
 
 
+@pytest.mark.timeout(300)
+def test_serving_dp2_pp2_grid(tmp_path):
+    """DP x PP serving grid (4 workers: 2 replicas x 2 stages): lockstep
+    rounds with per-stage barriers; only each replica's pp-0 rank emits.
+    Dense dummy-init replicas are identical, so both requests (routed
+    round-robin to different replicas) must emit the same tokens."""
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+    cfg = EngineConfig(model=_model_dir(tmp_path), load_format="dummy",
+                       device="cpu", dtype="float32", page_size=4,
+                       pp_size=2, dp_size=2, maxp=64, maxd=32,
+                       master_port=29659,
+                       enable_prefix_caching=False)
+    eng = AsyncLLMEngine(cfg, base_port=28761)
+    eng.start()
+    try:
+        sp = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
+
+        async def two():
+            a = asyncio.ensure_future(_agen(eng, [1, 2, 3, 4, 5], sp))
+            b = asyncio.ensure_future(_agen(eng, [1, 2, 3, 4, 5], sp))
+            return await asyncio.gather(a, b)
+
+        r1, r2 = asyncio.new_event_loop().run_until_complete(two())
+        assert len(r1) == 5 and r1 == r2
+    finally:
+        eng.stop()
+
+
 async def _agen(engine, token_ids, sampling):
     return [c.token_id
             async for c in engine.generate_stream(token_ids, sampling)]
