@@ -366,8 +366,13 @@ class AsyncLLMEngine:
             return ids
         # render to text then encode: apply_chat_template's tokenize=True
         # return type varies across transformers versions (list vs dict)
+        agp = kwargs.pop("add_generation_prompt", True)
+        cfm = kwargs.pop("continue_final_message", False)
+        if cfm:
+            agp = False  # mutually exclusive (OpenAI/transformers rule)
         text = self.tokenizer.apply_chat_template(
-            messages, add_generation_prompt=True, tokenize=False, **kwargs)
+            messages, add_generation_prompt=agp,
+            continue_final_message=cfm, tokenize=False, **kwargs)
         ids = self.tokenizer.encode(text)
         if not ids:
             raise ValueError("empty prompt after tokenization")

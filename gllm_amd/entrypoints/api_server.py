@@ -161,6 +161,10 @@ def build_app():
     async def chat_completions(req: ChatCompletionRequest, raw: Request):
         messages = [m.model_dump(exclude_none=True) for m in req.messages]
         kwargs = req.chat_template_kwargs or {}
+        kwargs.setdefault("add_generation_prompt",
+                          req.add_generation_prompt)
+        if req.continue_final_message:
+            kwargs["continue_final_message"] = True
         if req.tools:
             kwargs["tools"] = [t.model_dump() for t in req.tools]
         try:

@@ -66,6 +66,8 @@ class ChatCompletionRequest(BaseModel):
     tools: Optional[List[ToolDef]] = None
     tool_choice: Optional[Union[str, Dict[str, Any]]] = None
     chat_template_kwargs: Optional[Dict[str, Any]] = None
+    add_generation_prompt: bool = True
+    continue_final_message: bool = False
     prompt_logprobs: Optional[int] = None
     include_stop_str_in_output: bool = False
     logit_bias: Optional[Dict[str, float]] = None

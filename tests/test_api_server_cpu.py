@@ -260,3 +260,51 @@ def test_bad_words_sequence_matching():
     meta = build_sampling_metadata(items, "cpu")
     out = Sampler()(logits.clone(), meta)
     assert out.next_tokens.tolist() == [3, 7]
+
+
+@pytest.mark.timeout(300)
+def test_chat_template_knobs(tmp_path):
+    """add_generation_prompt / continue_final_message reach the chat
+    template (prompt token count shifts by the generation-prompt
+    marker)."""
+    import os
+
+    from fastapi.testclient import TestClient
+    import gllm_amd.entrypoints.api_server as srv
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+
+    d = _mk_model_dir(tmp_path)
+    # template that actually honors the knob
+    with open(os.path.join(d, "tokenizer_config.json")) as f:
+        tc = json.load(f)
+    tc["chat_template"] = ("{% for m in messages %}{{ m.content }} "
+                           "{% endfor %}"
+                           "{% if add_generation_prompt %}w99 {% endif %}")
+    with open(os.path.join(d, "tokenizer_config.json"), "w") as f:
+        json.dump(tc, f)
+    cfg = EngineConfig(model=d, load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       master_port=29693,
+                       enable_prefix_caching=False)
+    srv.engine = AsyncLLMEngine(cfg)
+    srv.served_model = "tiny"
+    srv.engine.start()
+    try:
+        client = TestClient(srv.build_app())
+        body = {"messages": [{"role": "user", "content": "w5 w6"}],
+                "max_tokens": 1, "temperature": 0.0, "ignore_eos": True}
+        with_gp = client.post("/v1/chat/completions", json=body).json()
+        without = client.post("/v1/chat/completions", json={
+            **body, "add_generation_prompt": False}).json()
+        assert with_gp["usage"]["prompt_tokens"] == \
+            without["usage"]["prompt_tokens"] + 1
+        cont = client.post("/v1/chat/completions", json={
+            **body,
+            "messages": [{"role": "user", "content": "w5 w6"},
+                         {"role": "assistant", "content": "w7 w8"}],
+            "continue_final_message": True}).json()
+        # continued prompt = both messages, no generation marker
+        assert cont["usage"]["prompt_tokens"] == 4
+    finally:
+        srv.engine.stop()
