@@ -334,3 +334,54 @@ def test_int4_moe_tp2_equals_single(tmp_path, method, port):
         p.join(timeout=120)
         assert p.exitcode == 0
     assert got == ref
+
+
+@pytest.mark.timeout(300)
+def test_int4_tp2_kv_replication_equals_single(tmp_path):
+    """tp=2 over num_key_value_heads=1: the packed qkv loaders must
+    replicate the shared K/V shard (src_rank = tp_rank //
+    kv_replication) instead of slicing it."""
+    import multiprocessing as mp
+    cfg = dict(CFG)
+    cfg["num_key_value_heads"] = 1
+    base = _base_state_dict()
+    hd = CFG["hidden_size"] // CFG["num_attention_heads"]
+    g = torch.Generator().manual_seed(99)
+    for L in range(CFG["num_hidden_layers"]):
+        p = f"model.layers.{L}.self_attn."
+        base[p + "k_proj.weight"] = torch.randn(
+            hd, CFG["hidden_size"], generator=g) * 0.08
+        base[p + "k_proj.bias"] = torch.randn(hd, generator=g) * 0.08
+        base[p + "v_proj.weight"] = torch.randn(
+            hd, CFG["hidden_size"], generator=g) * 0.08
+        base[p + "v_proj.bias"] = torch.randn(hd, generator=g) * 0.08
+    q_sd = {}
+    for name, w in base.items():
+        if _is_quantized(name):
+            qw, qz, s = pack_gptq(w, GROUP)
+            stem = name[:-len(".weight")]
+            q_sd[stem + ".qweight"] = qw
+            q_sd[stem + ".qzeros"] = qz
+            q_sd[stem + ".scales"] = s
+        else:
+            q_sd[name] = w
+    d = str(tmp_path / "gptq_kvrep")
+    os.makedirs(d, exist_ok=True)
+    cfg["quantization_config"] = {"quant_method": "gptq", "bits": 4,
+                                  "group_size": GROUP}
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump(cfg, f)
+    from safetensors.torch import save_file
+    save_file(q_sd, os.path.join(d, "model.safetensors"))
+    ref = _gen(d)
+    ctx = mp.get_context("spawn")
+    rq = ctx.Queue()
+    procs = [ctx.Process(target=_run_int4_tp_rank,
+                         args=(r, d, 29791, rq)) for r in range(2)]
+    for p in procs:
+        p.start()
+    got = rq.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert got == ref
