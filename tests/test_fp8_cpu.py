@@ -335,3 +335,50 @@ def test_fp8_moe_tp2_equals_single(tmp_path):
         p.join(timeout=120)
         assert p.exitcode == 0
     assert got == ref
+
+
+@pytest.mark.timeout(300)
+def test_fp8_tp2_kv_replication_equals_single(tmp_path):
+    """tp=2 over num_key_value_heads=1: the block-space scale loader
+    must replicate the shared K/V scale rows like the weight loader."""
+    cfg = dict(CFG)
+    cfg["num_key_value_heads"] = 1
+    base = _base_state_dict()
+    hd = CFG["hidden_size"] // CFG["num_attention_heads"]
+    g = torch.Generator().manual_seed(77)
+    for L in range(CFG["num_hidden_layers"]):
+        p = f"model.layers.{L}.self_attn."
+        base[p + "k_proj.weight"] = torch.randn(
+            hd, CFG["hidden_size"], generator=g) * 0.08
+        base[p + "k_proj.bias"] = torch.randn(hd, generator=g) * 0.08
+        base[p + "v_proj.weight"] = torch.randn(
+            hd, CFG["hidden_size"], generator=g) * 0.08
+        base[p + "v_proj.bias"] = torch.randn(hd, generator=g) * 0.08
+    fp8_sd = {}
+    for name, w in base.items():
+        if _is_quantized(name):
+            q, s = block_quant_fp8(w, block=BLOCK)
+            fp8_sd[name] = q
+            fp8_sd[name + "_scale_inv"] = s
+        else:
+            fp8_sd[name] = w
+    d = str(tmp_path / "fp8_kvrep")
+    os.makedirs(d, exist_ok=True)
+    cfg["quantization_config"] = {"quant_method": "fp8", "fmt": "e4m3",
+                                  "weight_block_size": list(BLOCK)}
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump(cfg, f)
+    from safetensors.torch import save_file
+    save_file(fp8_sd, os.path.join(d, "model.safetensors"))
+    ref = _gen_tokens(d)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_run_tp_rank, args=(r, d, 29697, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert got == ref

@@ -385,3 +385,24 @@ def test_int4_tp2_kv_replication_equals_single(tmp_path):
         p.join(timeout=120)
         assert p.exitcode == 0
     assert got == ref
+
+
+def test_pack_dequant_property_sweep():
+    """Property sweep over shapes/groups: dequant(pack(w)) stays within
+    half a quant step of w for both formats (hypothesis-style grid,
+    deterministic seeds)."""
+    for seed, (N, K, g) in enumerate([(8, 16, 8), (24, 32, 16),
+                                      (16, 64, 32), (40, 48, 8),
+                                      (8, 128, 128)]):
+        torch.manual_seed(seed)
+        w = torch.randn(N, K) * (0.01 + seed)
+        for pack, deq in ((pack_gptq, dequant_gptq),
+                          (pack_awq, dequant_awq)):
+            qw, qz, s = pack(w, g)
+            back = deq(qw, qz, s, g, torch.float32)
+            # bound: half a step from rounding + up to half a step
+            # from the rounded zero-point (plus the GPTQ stored-z-1
+            # clamp at z=0) => one full quant step per group
+            step = s.float().repeat_interleave(g, dim=0).t()  # [N, K]
+            assert ((back - w).abs() <= step + 1e-5).all(), \
+                (seed, N, K, g)
