@@ -91,13 +91,15 @@ def pack_gptq(w: torch.Tensor, group_size: int):
     wg = wk.view(G, group_size, N)
     mn, mx = wg.min(1).values, wg.max(1).values       # [G, N]
     scales = ((mx - mn) / 15.0).clamp_min(1e-8)
-    zeros = (-mn / scales).round().clamp(0, 15)       # [G, N]
+    # the stored-z-1 convention cannot represent z = 0 (real GPTQ
+    # packers keep z in [1, 15]); clamping low keeps dequant's +1 exact
+    zeros = (-mn / scales).round().clamp(1, 15)       # [G, N]
     g = torch.arange(K) // group_size
     q = (wk / scales[g] + zeros[g]).round().clamp(0, 15).to(torch.int64)
     qweight = torch.zeros(K // 8, N, dtype=torch.int64)
     for i in range(8):
         qweight |= q[i::8] << (4 * i)
-    zstore = (zeros.to(torch.int64) - 1).clamp(0, 15)  # legacy z-1
+    zstore = zeros.to(torch.int64) - 1                 # legacy z-1
     qzeros = torch.zeros(G, N // 8, dtype=torch.int64)
     for i in range(8):
         qzeros |= zstore[:, i::8] << (4 * i)
