@@ -361,13 +361,25 @@ def build_app():
 
     async def _completion_stream(req, raw, token_ids, sampling):
         n = req.n or 1
+        total_c = 0
         async for j, chunk in _merged_stream(raw, token_ids, sampling, n):
+            if chunk.finish_reason is not None:
+                total_c += chunk.n_output_tokens
             resp = CompletionResponse(
                 model=req.model or served_model,
                 choices=[CompletionResponseChoice(
                     index=j, text=chunk.text,
                     finish_reason=chunk.finish_reason)])
             yield f"data: {resp.model_dump_json(exclude_none=True)}\n\n"
+        opts = getattr(req, "stream_options", None) or {}
+        if opts.get("include_usage"):
+            usage = CompletionResponse(
+                model=req.model or served_model, choices=[],
+                usage=UsageInfo(
+                    prompt_tokens=len(token_ids) * n,
+                    completion_tokens=total_c,
+                    total_tokens=len(token_ids) * n + total_c))
+            yield f"data: {usage.model_dump_json(exclude_none=True)}\n\n"
         yield "data: [DONE]\n\n"
 
     return app

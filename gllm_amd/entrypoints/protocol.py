@@ -96,6 +96,7 @@ class CompletionRequest(BaseModel):
     frequency_penalty: Optional[float] = 0.0
     ignore_eos: bool = False
     echo: bool = False
+    stream_options: Optional[Dict[str, Any]] = None
     logprobs: Optional[int] = None
     prompt_logprobs: Optional[int] = None
     include_stop_str_in_output: bool = False
