@@ -308,3 +308,35 @@ def test_chat_template_knobs(tmp_path):
         assert cont["usage"]["prompt_tokens"] == 4
     finally:
         srv.engine.stop()
+
+
+@pytest.mark.timeout(300)
+def test_completions_stream_usage(tmp_path):
+    """stream_options.include_usage on /v1/completions emits a final
+    usage chunk before [DONE]."""
+    from fastapi.testclient import TestClient
+    import gllm_amd.entrypoints.api_server as srv
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.server_engine import AsyncLLMEngine
+
+    d = _mk_model_dir(tmp_path)
+    cfg = EngineConfig(model=d, load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       master_port=29698,
+                       enable_prefix_caching=False)
+    srv.engine = AsyncLLMEngine(cfg)
+    srv.served_model = "tiny"
+    srv.engine.start()
+    try:
+        client = TestClient(srv.build_app())
+        with client.stream("POST", "/v1/completions", json={
+                "prompt": [1, 2, 3], "max_tokens": 4, "temperature": 0.0,
+                "stream": True, "ignore_eos": True,
+                "stream_options": {"include_usage": True}}) as r:
+            lines = [ln for ln in r.iter_lines()
+                     if ln.startswith("data: {")]
+        last = json.loads(lines[-1][6:])
+        assert last["usage"]["completion_tokens"] == 4
+        assert last["usage"]["prompt_tokens"] == 3
+    finally:
+        srv.engine.stop()
