@@ -20,7 +20,7 @@ full output stream and pushes (seq_id, token, finish) to the frontend.
 
 import os
 import time
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist

@@ -14,7 +14,7 @@ import os
 import pytest
 import torch
 
-from gllm_amd.layers.quantization.fp8 import (FP8_MAX, block_quant_fp8,
+from gllm_amd.layers.quantization.fp8 import (block_quant_fp8,
                                               dequant_block_fp8,
                                               per_token_group_quant_fp8)
 

@@ -115,7 +115,6 @@ def test_pp2_ships_logprobs_to_rank0(tmp_path):
     """Under PP>1 sampling runs on the last stage; per-token logprobs
     ride the token broadcast and prompt logprobs ship at finish, so
     rank 0 serves the same values a single-process run computes."""
-    import json as _json
     import multiprocessing as mp
 
     from gllm_amd.sequence import SamplingParams
