@@ -129,12 +129,10 @@ def topk_softmax(gating: torch.Tensor, topk: int, renormalize: bool = True):
 
 def grouped_topk(scores, topk, n_group, topk_group, renormalize=True,
                  scoring="softmax", e_bias=None):
-    dev = scores.device
-    w, i = torch_ref.grouped_topk(scores.cpu(), topk, n_group, topk_group,
-                                  renormalize, scoring,
-                                  e_bias.cpu() if e_bias is not None
-                                  else None)
-    return w.to(dev), i.to(dev)
+    # device-resident (a CPU round-trip here would sync every MoE
+    # layer of every DeepSeek step)
+    return torch_ref.grouped_topk(scores, topk, n_group, topk_group,
+                                  renormalize, scoring, e_bias)
 
 
 # --------------------------------------------------------------- mla

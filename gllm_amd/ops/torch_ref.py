@@ -215,7 +215,7 @@ def grouped_topk(scores: torch.Tensor, topk: int, n_group: int,
     else:
         group_scores = g.max(dim=-1).values
     grp_idx = group_scores.topk(topk_group, dim=-1)[1]  # [T, topk_group]
-    mask = torch.zeros(T, n_group, dtype=torch.bool)
+    mask = torch.zeros(T, n_group, dtype=torch.bool, device=scores.device)
     mask.scatter_(1, grp_idx, True)
     mask = mask.unsqueeze(-1).expand(T, n_group, E // n_group)
     sel = sel.masked_fill(~mask.reshape(T, E), float("-inf"))
