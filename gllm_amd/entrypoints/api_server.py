@@ -99,9 +99,22 @@ def build_app():
         lp = req.logprobs
         if isinstance(lp, bool):  # chat: bool + top_logprobs count
             lp = (getattr(req, "top_logprobs", None) or 1) if lp else None
+        vocab = getattr(engine.hf_config, "vocab_size", None)
         bias = None
         if getattr(req, "logit_bias", None):
             bias = {int(k): float(v) for k, v in req.logit_bias.items()}
+            if vocab is not None:
+                bad = [k for k in bias if not 0 <= k < vocab]
+                if bad:
+                    raise ValueError(
+                        f"logit_bias token id(s) {bad[:5]} out of range "
+                        f"[0, {vocab})")
+        allowed = getattr(req, "allowed_token_ids", None)
+        if allowed and vocab is not None:
+            bad = [i for i in allowed if not 0 <= i < vocab]
+            if bad:
+                raise ValueError(
+                    f"allowed_token_ids {bad[:5]} out of range [0, {vocab})")
         bad_words_ids = None
         if getattr(req, "bad_words", None):
             tok = engine.tokenizer
@@ -136,7 +149,7 @@ def build_app():
             logprobs=lp,
             prompt_logprobs=getattr(req, "prompt_logprobs", None),
             logit_bias=bias,
-            allowed_token_ids=getattr(req, "allowed_token_ids", None),
+            allowed_token_ids=allowed,
             bad_words_token_ids=bad_words_ids,
             skip_special_tokens=getattr(req, "skip_special_tokens", True))
 
@@ -376,9 +389,9 @@ def build_app():
             usage = CompletionResponse(
                 model=req.model or served_model, choices=[],
                 usage=UsageInfo(
-                    prompt_tokens=len(token_ids) * n,
+                    prompt_tokens=len(token_ids),
                     completion_tokens=total_c,
-                    total_tokens=len(token_ids) * n + total_c))
+                    total_tokens=len(token_ids) + total_c))
             yield f"data: {usage.model_dump_json(exclude_none=True)}\n\n"
         yield "data: [DONE]\n\n"
 

@@ -74,7 +74,7 @@ class ChatCompletionRequest(BaseModel):
     allowed_token_ids: Optional[List[int]] = None
     bad_words: Optional[List[str]] = None
     skip_special_tokens: bool = True
-    truncate_prompt_tokens: Optional[int] = None
+    truncate_prompt_tokens: Optional[int] = Field(default=None, ge=1)
 
 
 class CompletionRequest(BaseModel):
@@ -104,7 +104,7 @@ class CompletionRequest(BaseModel):
     allowed_token_ids: Optional[List[int]] = None
     bad_words: Optional[List[str]] = None
     skip_special_tokens: bool = True
-    truncate_prompt_tokens: Optional[int] = None
+    truncate_prompt_tokens: Optional[int] = Field(default=None, ge=1)
 
 
 class UsageInfo(BaseModel):

@@ -155,12 +155,19 @@ class FusedMoE(nn.Module):
         return out
 
     def _dequant_expert(self, lid: int, dtype, device):
-        """fp8 expert weights dequantized once per expert (lazily,
-        cached) into the compute dtype — same round-1 stance as the
-        dense fp8 linears (layers/quantization/fp8.py); the grouped fp8
-        MFMA GEMM consuming the packed layout directly is round 2."""
+        """Quantized expert weights dequantized into the compute dtype.
+
+        The FULL local bank materializes on first touch (not lazily
+        per-expert): the first forward is the model runner's profile
+        run, so the dequant footprint is resident before KV-cache
+        sizing and can never OOM mid-serving (advisor r1 finding).
+        The grouped fp8 MFMA GEMM consuming the packed layout directly
+        replaces this."""
         if self._dq_cache is None:
             self._dq_cache = [None] * self.num_local_experts
+            for other in range(self.num_local_experts):
+                if other != lid:
+                    self._dequant_expert(other, dtype, device)
         ent = self._dq_cache[lid]
         if ent is None:
             if self.fp8_block is not None:

@@ -83,6 +83,27 @@ __global__ void reshape_and_cache_kernel(
   for (int i = threadIdx.x; i < vnvec; i += blockDim.x) vd[i] = vs[i];
 }
 
+// fp32 variant: floatx4 (16 B) vector loads instead of shortx8.
+__global__ void silu_and_mul_f32_kernel(float *__restrict__ out,
+                                        const float *__restrict__ x,
+                                        long rows, int d) {
+  const int nvec = d / 4;
+  const long total = rows * nvec;
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const long row = idx / nvec;
+    const int i = idx % nvec;
+    floatx4 a = reinterpret_cast<const floatx4 *>(x + row * 2 * d)[i];
+    floatx4 b = reinterpret_cast<const floatx4 *>(x + row * 2 * d + d)[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float s = a[j] / (1.f + __expf(-a[j]));
+      a[j] = s * b[j];
+    }
+    reinterpret_cast<floatx4 *>(out + row * d)[i] = a;
+  }
+}
+
 }  // namespace
 
 void silu_and_mul(torch::Tensor out, torch::Tensor x) {
@@ -103,6 +124,10 @@ void silu_and_mul(torch::Tensor out, torch::Tensor x) {
     hipLaunchKernelGGL((silu_and_mul_kernel<__half>), dim3(grid), dim3(block),
                        0, stream, (__half *)out.data_ptr(),
                        (const __half *)x.data_ptr(), rows, d);
+  } else if (x.scalar_type() == at::kFloat) {
+    hipLaunchKernelGGL((silu_and_mul_f32_kernel), dim3(grid), dim3(block),
+                       0, stream, out.data_ptr<float>(),
+                       x.data_ptr<float>(), rows, d);
   } else {
     TORCH_CHECK(false, "silu_and_mul: unsupported dtype");
   }

@@ -231,6 +231,21 @@ def test_api_sampling_controls(tmp_path):
             "temperature": 0.0, "ignore_eos": True,
             "truncate_prompt_tokens": 2})
         assert r.json()["usage"]["prompt_tokens"] == 2
+
+        # out-of-range ids are a 400, not an engine crash (advisor r1)
+        r = client.post("/v1/completions", json={
+            "prompt": "w3", "max_tokens": 1,
+            "logit_bias": {"99999": 5.0}})
+        assert r.status_code == 400 and "out of range" in r.text
+        r = client.post("/v1/completions", json={
+            "prompt": "w3", "max_tokens": 1,
+            "allowed_token_ids": [-3]})
+        assert r.status_code == 400 and "out of range" in r.text
+        # non-positive truncate_prompt_tokens is rejected by pydantic
+        r = client.post("/v1/completions", json={
+            "prompt": "w3", "max_tokens": 1,
+            "truncate_prompt_tokens": 0})
+        assert r.status_code == 422
     finally:
         srv.engine.stop()
 
