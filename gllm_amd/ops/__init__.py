@@ -149,13 +149,15 @@ def apply_penalty_pool(logits, mask_pool, slots, penalties):
 def cache_latent(k, k_cache, slot_mapping):
     """Scatter the absorbed-MLA latent row [T, 1, lora+rope] into the
     paged latent cache (the v 'cache' is a view of its first lora dims,
-    so one write covers both)."""
+    so one write covers both). Reference concat_and_cache_mla
+    (cache_kernels.py:161)."""
     if k.is_cuda:
-        # reuse the K/V scatter kernel with both targets = the latent
-        # cache (same row written twice; a dedicated single-row kernel
-        # rides the round-2 MLA kernel work)
-        _gpu_kernels().reshape_and_cache(k, k, k_cache, k_cache,
-                                         slot_mapping)
+        if k_cache.shape[-1] == 576:
+            _gpu_kernels().cache_latent(
+                k.reshape(k.shape[0], -1), k_cache, slot_mapping)
+        else:  # non-standard latent width: plain K/V scatter, twice
+            _gpu_kernels().reshape_and_cache(k, k, k_cache, k_cache,
+                                             slot_mapping)
         return
     page_size = k_cache.shape[1]
     pages = torch.div(slot_mapping, page_size, rounding_mode="floor")
@@ -169,14 +171,34 @@ _MLA_GPU_WARNED = False
 def mla_paged_attention(q, k_cache, v_cache, block_table, seq_lens,
                         query_start_loc, scale, topk_positions=None,
                         seq_lens_cpu=None, query_start_loc_cpu=None):
+    """Varlen causal attention with asymmetric head dims (MLA).
+
+    GPU + absorbed layout (Hkv=1, 576-dim latent) -> the gfx950 MQA
+    MFMA kernel (one kernel serves decode w/ split-KV, chunked prefill
+    and mixed batches — reference attention.py:366-446,653-925 roles).
+    DSA sparse selection (topk_positions) and the non-absorbed debug
+    layout stay on the torch path."""
     global _MLA_GPU_WARNED
+    if (q.is_cuda and topk_positions is None and k_cache.shape[2] == 1
+            and q.shape[-1] == 576 and v_cache.shape[-1] == 512
+            and q.dtype == torch.bfloat16):
+        qsl_h = query_start_loc_cpu or query_start_loc.tolist()
+        lens_h = seq_lens_cpu or seq_lens.tolist()
+        max_q = max((qsl_h[i + 1] - qsl_h[i] for i in range(len(lens_h))),
+                    default=1)
+        max_s = max(lens_h, default=1)
+        T, H = q.shape[0], q.shape[1]
+        out = torch.empty(T, H, 512, dtype=q.dtype, device=q.device)
+        qsl_i = query_start_loc.int() \
+            if query_start_loc.dtype != torch.int32 else query_start_loc
+        _gpu_kernels().mla_paged_attention(
+            out, q.contiguous(), k_cache, block_table, seq_lens, qsl_i,
+            max_q, scale, max_s)
+        return out
     if q.is_cuda and not _MLA_GPU_WARNED:
-        # NOT a silent fallback: no native MLA kernel exists yet — the
-        # torch path IS the current GPU implementation for the absorbed
-        # 576-dim latent MQA (gfx950 kernel = ROADMAP.md item 7)
         from gllm_amd.logger import logger
-        logger.warning("absorbed-MLA attention runs the torch path on "
-                       "GPU (gfx950 MLA kernel lands in round 2)")
+        logger.warning("MLA attention falls back to the torch path on "
+                       "GPU (sparse/non-absorbed layout)")
         _MLA_GPU_WARNED = True
     return torch_ref.mla_paged_attention(
         q, k_cache, v_cache, block_table, seq_lens, query_start_loc,

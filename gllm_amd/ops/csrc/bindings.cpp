@@ -26,6 +26,13 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor query_start_loc,
                              long max_query_len, double scale,
                              long sliding_window);
+void mla_paged_attention(torch::Tensor out, torch::Tensor q,
+                         torch::Tensor k_cache, torch::Tensor block_table,
+                         torch::Tensor seq_lens,
+                         torch::Tensor query_start_loc, long max_query_len,
+                         double scale, long max_seq_len);
+void cache_latent(torch::Tensor k, torch::Tensor k_cache,
+                  torch::Tensor slot_mapping);
 void apply_repetition_penalty(torch::Tensor logits, torch::Tensor pool,
                               torch::Tensor slots,
                               torch::Tensor penalties);
@@ -50,6 +57,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "varlen MFMA paged prefill attention");
   m.def("skinny_gemm", &skinny_gemm,
         "decode-regime weight-streaming GEMM (M<=256)");
+  m.def("mla_paged_attention", &mla_paged_attention,
+        "absorbed-MLA varlen attention over the 576-dim latent cache");
+  m.def("cache_latent", &cache_latent,
+        "scatter the per-token latent row into the paged cache");
   m.def("apply_repetition_penalty", &apply_repetition_penalty,
         "scaling penalty vs persistent seen-token mask pool");
   m.def("car_alloc", &car_alloc, "alloc hipIpc-shared AR buffer");

@@ -204,3 +204,63 @@ def test_sliding_window_prefill_and_decode(kernels):
                              seq_lens.cpu(), qsl1.cpu(),
                              1.0 / math.sqrt(D), sliding_window=W)
     assert_close_bf16(out1, ref1, frac=2e-3)
+
+
+# ---------------------------------------------------------------- MLA
+def _mk_latent_paged(B, ps, ctx_lens, seed=0, DK=576):
+    torch.manual_seed(seed)
+    max_pages = max(-(-c // ps) for c in ctx_lens)
+    total_pages = sum(-(-c // ps) for c in ctx_lens) + 1
+    k_cache = torch.randn(total_pages, ps, 1, DK, dtype=torch.bfloat16,
+                          device="cuda")
+    bt = torch.zeros(B, max_pages, dtype=torch.int32, device="cuda")
+    next_page = 1
+    for b, c in enumerate(ctx_lens):
+        n = -(-c // ps)
+        bt[b, :n] = torch.arange(next_page, next_page + n)
+        next_page += n
+    return k_cache, bt
+
+
+@pytest.mark.parametrize("case", [
+    # (q_lens, ctx_lens, H)
+    ([1, 1, 1], [5, 900, 333], 128),        # decode, split-KV path
+    ([1], [4096], 128),                     # long decode, deep splits
+    ([64, 17], [100, 333], 128),            # chunked prefill w/ past
+    ([1, 130, 1], [77, 1030, 16], 16),      # mixed batch, TP=8 heads
+    ([33], [33], 8),                        # fresh prefill, small H
+])
+def test_mla_attention(kernels, case):
+    q_lens, ctx, H = case
+    B, ps, DK, DV = len(q_lens), 16, 576, 512
+    k_cache, bt = _mk_latent_paged(B, ps, ctx, seed=11)
+    v_cache = k_cache[..., :DV]
+    T = sum(q_lens)
+    q = torch.randn(T, H, DK, dtype=torch.bfloat16, device="cuda")
+    seq_lens = torch.tensor(ctx, dtype=torch.int32, device="cuda")
+    qsl = torch.tensor([0] + list(torch.cumsum(
+        torch.tensor(q_lens), 0)), dtype=torch.int32, device="cuda")
+    from gllm_amd import ops
+    scale = 1.0 / math.sqrt(DK)
+    out = ops.mla_paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl,
+                                  scale, seq_lens_cpu=ctx,
+                                  query_start_loc_cpu=[0] + list(
+                                      torch.cumsum(torch.tensor(q_lens),
+                                                   0).tolist()))
+    ref = R.mla_paged_attention(q.float().cpu(), k_cache.float().cpu(),
+                                v_cache.float().cpu(), bt.cpu(),
+                                seq_lens.cpu(), qsl.cpu(), scale)
+    assert_close_bf16(out, ref, frac=2e-3)
+
+
+def test_cache_latent(kernels):
+    torch.manual_seed(3)
+    T, ps, P, DK = 33, 16, 12, 576
+    k = torch.randn(T, 1, DK, dtype=torch.bfloat16, device="cuda")
+    kc = torch.zeros(P, ps, 1, DK, dtype=torch.bfloat16, device="cuda")
+    slots = torch.randperm(P * ps, device="cuda")[:T]
+    from gllm_amd import ops
+    ops.cache_latent(k, kc, slots)
+    kr = torch.zeros(P, ps, 1, DK, dtype=torch.bfloat16)
+    R.reshape_and_cache(k.cpu(), k.cpu(), kr, kr, slots.cpu())
+    assert torch.equal(kc.cpu(), kr)
