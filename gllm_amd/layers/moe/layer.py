@@ -116,6 +116,10 @@ class FusedMoE(nn.Module):
         """Local expert shard over the (possibly DP-gathered) batch;
         returns the PARTIAL output (no collectives).
 
+        GPU + bf16 experts -> the grouped MFMA GEMM pipeline
+        (ops.fused_moe): device-side align, no host syncs, hipGraph-
+        safe. CPU and quantized banks keep the sort+segment host loop.
+
         Host analogue of the reference's moe_align_block_size
         (_custom_ops.py): ONE stable sort groups the (token, expert)
         pairs by expert and ONE bincount transfer gives the segment
@@ -124,6 +128,12 @@ class FusedMoE(nn.Module):
         ``sel.any()`` form synced once per expert: 256/layer on
         DeepSeek-V3). DP padding rows carry expert id -1 and land in
         segment 0 of the shifted bincount (skipped)."""
+        if (x.is_cuda and self.fp8_block is None and self.int4_cfg is None
+                and x.dtype == torch.bfloat16 and ops.has_kernels()):
+            return ops.fused_moe(
+                x.contiguous(), self.w13_weight, self.w2_weight,
+                weights, ids, expert_start=self.expert_start,
+                num_global_experts=self.num_experts)
         T = x.shape[0]
         out = torch.zeros_like(x)
         flat_ids = ids.long().flatten()                    # [T*K]

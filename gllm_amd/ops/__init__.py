@@ -135,6 +135,53 @@ def grouped_topk(scores, topk, n_group, topk_group, renormalize=True,
                                   renormalize, scoring, e_bias)
 
 
+def fused_moe(x: torch.Tensor, w13: torch.Tensor, w2: torch.Tensor,
+              topk_weights: torch.Tensor, topk_ids: torch.Tensor,
+              expert_start: int = 0,
+              num_global_experts: Optional[int] = None) -> torch.Tensor:
+    """Grouped-GEMM fused MoE on the gfx950 kernels (GPU, bf16).
+
+    Reference fused_experts_impl (fused_moe_triton/fused_moe.py:779):
+    align -> GEMM1 -> silu_and_mul -> GEMM2 (weighted scatter) -> sum.
+    Fully device-resident and hipGraph-safe (no host syncs; grids are
+    sized by upper bound and gated on the device-side block count).
+    Returns the PARTIAL output for this rank's expert shard (callers
+    run their TP/EP all-reduce on top).
+    """
+    T, K = x.shape
+    E_local, two_i, _ = w13.shape
+    inter = two_i // 2
+    topk = topk_ids.shape[1]
+    n_pairs = T * topk
+    # block_m heuristic: sparse decode routing -> small blocks
+    block_m = 16 if n_pairs < 8 * E_local or n_pairs <= 512 else 64
+    cap = n_pairs + E_local * (block_m - 1) + 1
+    max_blocks = (n_pairs + block_m - 1) // block_m + E_local
+    dev = x.device
+    k = _gpu_kernels()
+    ids32 = topk_ids.int().contiguous()
+    sorted_ids = torch.empty(cap, dtype=torch.int32, device=dev)
+    expert_blocks = torch.empty(max_blocks, dtype=torch.int32, device=dev)
+    n_post = torch.empty(1, dtype=torch.int32, device=dev)
+    k.moe_align(ids32, E_local, expert_start, block_m, sorted_ids,
+                expert_blocks, n_post)
+    rows_pad = cap - 1 + block_m  # >= n_post_pad upper bound
+    inter1 = torch.empty(rows_pad, two_i, dtype=x.dtype, device=dev)
+    k.moe_gemm(inter1, x, w13, sorted_ids, expert_blocks, n_post, None,
+               n_pairs, topk, block_m, False)
+    act = silu_and_mul(inter1)
+    # zero-filled: pairs routed to non-local experts (EP shards) or to
+    # the DP padding id -1 are dropped by the align kernel and must
+    # contribute zeros
+    pair_out = x.new_zeros(n_pairs, K)
+    k.moe_gemm(pair_out, act, w2, sorted_ids, expert_blocks, n_post,
+               topk_weights.float().contiguous(), n_pairs, topk, block_m,
+               True)
+    out = torch.empty(T, K, dtype=x.dtype, device=dev)
+    k.moe_sum(out, pair_out, topk)
+    return out
+
+
 # --------------------------------------------------------------- mla
 def apply_penalty_pool(logits, mask_pool, slots, penalties):
     """In-place repetition penalty against the persistent uint8 mask

@@ -33,6 +33,15 @@ void mla_paged_attention(torch::Tensor out, torch::Tensor q,
                          double scale, long max_seq_len);
 void cache_latent(torch::Tensor k, torch::Tensor k_cache,
                   torch::Tensor slot_mapping);
+void moe_align(torch::Tensor topk_ids, long E_local, long expert_start,
+               long block_m, torch::Tensor sorted_ids,
+               torch::Tensor expert_blocks, torch::Tensor n_post_pad);
+void moe_gemm(torch::Tensor C, torch::Tensor A, torch::Tensor W,
+              torch::Tensor sorted_ids, torch::Tensor expert_blocks,
+              torch::Tensor n_post_pad,
+              c10::optional<torch::Tensor> topk_weights, long n_pairs,
+              long topk, long block_m, bool scatter);
+void moe_sum(torch::Tensor out, torch::Tensor pair_out, long topk);
 void apply_repetition_penalty(torch::Tensor logits, torch::Tensor pool,
                               torch::Tensor slots,
                               torch::Tensor penalties);
@@ -61,6 +70,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "absorbed-MLA varlen attention over the 576-dim latent cache");
   m.def("cache_latent", &cache_latent,
         "scatter the per-token latent row into the paged cache");
+  m.def("moe_align", &moe_align,
+        "device-side moe_align_block_size (graph-safe)");
+  m.def("moe_gemm", &moe_gemm,
+        "grouped MFMA GEMM over sorted (token, expert) pairs");
+  m.def("moe_sum", &moe_sum, "sum pair outputs over topk");
   m.def("apply_repetition_penalty", &apply_repetition_penalty,
         "scaling penalty vs persistent seen-token mask pool");
   m.def("car_alloc", &car_alloc, "alloc hipIpc-shared AR buffer");

@@ -8,6 +8,8 @@ attention kernel at DeepSeek-V3 shapes (H=128, latent 576/512).
 import math
 import time
 
+import sys, os
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import torch
 
 from gllm_amd import ops
