@@ -67,12 +67,19 @@ class ModelRunner:
             # uniform gather is a round-2 item)
             cfg.use_graph = False
         else:
+            from gllm_amd import ops
             from gllm_amd.layers.moe.layer import FusedMoE
-            if any(isinstance(m, FusedMoE) for m in self.model.modules()):
-                # the round-1 expert dispatch reads the per-expert
-                # segment table on the host (an illegal sync under
-                # stream capture); the graph-safe path is the round-2
-                # grouped MFMA GEMM over sorted tokens
+            moe_mods = [m for m in self.model.modules()
+                        if isinstance(m, FusedMoE)]
+            if moe_mods and not (cfg.device.startswith("cuda")
+                                 and ops.has_kernels()
+                                 and all(m.fp8_block is None
+                                         and m.int4_cfg is None
+                                         for m in moe_mods)):
+                # bf16 MoE on GPU runs the device-resident grouped
+                # MFMA GEMM pipeline (ops.fused_moe) and is capture-
+                # safe; quantized banks still run the host segment
+                # loop (illegal sync under stream capture)
                 cfg.use_graph = False
         num_pages = num_pages_override or self._size_kv_cache()
         self._allocate_kv(num_pages)
