@@ -67,6 +67,20 @@ class Sampler(torch.nn.Module):
         temps = meta.temperatures.clamp_min(1e-5).unsqueeze(-1)
         scaled = logits / temps
         probs = torch.softmax(scaled, dim=-1)
+        if probs.is_cuda:
+            # fused sorting-free radix-select kernel (in-place),
+            # replaces the [B, V] sort composite below
+            from gllm_amd import ops
+            if ops.has_kernels():
+                probs = probs.contiguous()
+                ops.topk_topp_filter(probs, meta.top_ks,
+                                     meta.top_ps, meta.min_ps)
+                sampled = self._multinomial(probs, meta)
+                greedy_rows = meta.temperatures == 0.0
+                if greedy_rows.any():
+                    sampled = torch.where(greedy_rows,
+                                          logits.argmax(dim=-1), sampled)
+                return self._with_logprobs(logits, sampled, meta)
         probs = self._apply_top_k_top_p(probs, meta.top_ks, meta.top_ps)
         probs = self._apply_min_p(probs, meta.min_ps)
         sampled = self._multinomial(probs, meta)

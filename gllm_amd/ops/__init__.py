@@ -182,6 +182,20 @@ def fused_moe(x: torch.Tensor, w13: torch.Tensor, w2: torch.Tensor,
     return out
 
 
+# --------------------------------------------------------------- sampling
+def topk_topp_filter(probs: torch.Tensor, top_ks: torch.Tensor,
+                     top_ps: torch.Tensor,
+                     min_ps: Optional[torch.Tensor] = None):
+    """In-place fused top-k/top-p/min-p filter + renormalize on [B, V]
+    fp32 probs (GPU kernel; sorting-free 3-level radix select). CPU
+    callers use the torch sort composite in layers/sampler.py."""
+    assert probs.is_cuda
+    _gpu_kernels().topk_topp_filter(
+        probs, top_ks.int(), top_ps.float(),
+        min_ps.float() if min_ps is not None else None)
+    return probs
+
+
 # --------------------------------------------------------------- mla
 def apply_penalty_pool(logits, mask_pool, slots, penalties):
     """In-place repetition penalty against the persistent uint8 mask

@@ -45,6 +45,9 @@ void moe_sum(torch::Tensor out, torch::Tensor pair_out, long topk);
 void apply_repetition_penalty(torch::Tensor logits, torch::Tensor pool,
                               torch::Tensor slots,
                               torch::Tensor penalties);
+void topk_topp_filter(torch::Tensor probs, torch::Tensor top_ks,
+                      torch::Tensor top_ps,
+                      c10::optional<torch::Tensor> min_ps);
 std::pair<int64_t, py::bytes> car_alloc(int64_t data_bytes);
 int64_t car_open(py::bytes handle_bytes);
 void car_close(int64_t ptr);
@@ -77,6 +80,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_sum", &moe_sum, "sum pair outputs over topk");
   m.def("apply_repetition_penalty", &apply_repetition_penalty,
         "scaling penalty vs persistent seen-token mask pool");
+  m.def("topk_topp_filter", &topk_topp_filter,
+        "fused sorting-free top-k/top-p/min-p filter + renormalize");
   m.def("car_alloc", &car_alloc, "alloc hipIpc-shared AR buffer");
   m.def("car_open", &car_open, "map a peer's AR buffer");
   m.def("car_close", &car_close);

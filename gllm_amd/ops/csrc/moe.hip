@@ -100,11 +100,13 @@ __global__ __launch_bounds__(GEMM_BLOCK) void moe_gemm_kernel(
   const int l16 = lane & 15;
   const int lhi = lane >> 4;
 
-  // wave tiling over the BM x BN block
+  // wave tiling over the BM x BN block: (BM/16) x 4 sixteen-wide tiles
+  // split over 4 waves -> each wave owns one m sub-tile row and
+  // BN/(4/WM*16) = WM consecutive n tiles
   constexpr int WM = BM / 16;            // m sub-tiles (1, 2 or 4)
-  constexpr int WN_TILES = 4 / WM;       // 16-wide n tiles per wave
+  constexpr int WN_TILES = WM;           // 16-wide n tiles per wave
   const int wm = wave % WM;
-  const int wn = wave / WM;
+  const int wn = wave / WM;              // n-tile group (4/WM groups)
 
   constexpr int APAD = 8;
   __shared__ __hip_bfloat16 a_tile[2][BM * (BK + APAD)];

@@ -332,3 +332,55 @@ def test_fused_moe_graph_capture(kernels):
     g.replay()
     torch.cuda.synchronize()
     assert torch.equal(out, eager)
+
+
+# ------------------------------------------------------------ sampling
+@pytest.mark.parametrize("case", [
+    # (B, V, k, p, min_p)
+    (8, 32000, 50, 0.9, 0.0),
+    (4, 152064, -1, 0.8, 0.0),      # top-p only, big vocab
+    (16, 32000, 5, 1.0, 0.0),       # top-k only
+    (8, 32000, -1, 1.0, 0.05),      # min-p only
+    (8, 32000, 200, 0.95, 0.02),    # all three
+    (3, 32000, -1, 1.0, 0.0),       # disabled rows: untouched
+])
+def test_topk_topp_filter(kernels, case):
+    B, V, k, p, mp = case
+    torch.manual_seed(B * V + k)
+    from gllm_amd import ops
+    from gllm_amd.layers.sampler import Sampler
+    logits = torch.randn(B, V, device="cuda") * 3
+    probs = torch.softmax(logits, dim=-1)
+    ks = torch.full((B,), k, dtype=torch.int32, device="cuda")
+    ps = torch.full((B,), p, dtype=torch.float32, device="cuda")
+    mps = torch.full((B,), mp, dtype=torch.float32, device="cuda")
+    ref = Sampler._apply_top_k_top_p(probs.clone(), ks, ps)
+    ref = Sampler._apply_min_p(ref, mps)
+    out = probs.clone()
+    ops.topk_topp_filter(out, ks, ps, mps)
+    # exact same kept set (no ties in random fp32) and same renorm
+    assert ((out > 0) == (ref > 0)).all(), (
+        f"kept-set mismatch: {int((out > 0).sum())} vs "
+        f"{int((ref > 0).sum())}")
+    assert torch.allclose(out, ref, atol=1e-5, rtol=1e-4)
+
+
+def test_topk_topp_filter_mixed_rows(kernels):
+    """Per-row parameters differ (the serving case)."""
+    torch.manual_seed(0)
+    from gllm_amd import ops
+    from gllm_amd.layers.sampler import Sampler
+    B, V = 6, 50000
+    probs = torch.softmax(torch.randn(B, V, device="cuda") * 2, dim=-1)
+    ks = torch.tensor([1, 50, -1, 3000, -1, 10], dtype=torch.int32,
+                      device="cuda")
+    ps = torch.tensor([1.0, 0.9, 0.5, 0.99, 1.0, 1.0],
+                      dtype=torch.float32, device="cuda")
+    mps = torch.tensor([0.0, 0.0, 0.0, 0.01, 0.0, 0.0],
+                       dtype=torch.float32, device="cuda")
+    ref = Sampler._apply_top_k_top_p(probs.clone(), ks, ps)
+    ref = Sampler._apply_min_p(ref, mps)
+    out = probs.clone()
+    ops.topk_topp_filter(out, ks, ps, mps)
+    assert ((out > 0) == (ref > 0)).all()
+    assert torch.allclose(out, ref, atol=1e-5, rtol=1e-4)
