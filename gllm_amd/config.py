@@ -71,12 +71,11 @@ class EngineConfig:
     # --- runtime ---
     use_graph: bool = True               # hipGraph-captured decode buckets
     enable_overlap: bool = True          # launch-first/collect-later worker loop
-    # hipGraphs pay off when launch overhead dominates (small decode
-    # batches); large batches run eager (GPU-bound) with the full skinny
-    # GEMM range. NOTE: skinny MB>=2 kernels captured inside engine
-    # graphs fault on replay (suspected hipGraph + 72-80KB dynamic-LDS
-    # interaction; standalone repros pass) — buckets <= 64 sidestep it.
-    max_graph_bs: int = 64
+    # hipGraph decode buckets now cover the full skinny GEMM range:
+    # the r1 MB>=2 replay fault traced to dynamic LDS under capture
+    # (kernel uses static LDS since r2) and to the workspace growing
+    # mid-capture (warmup at the largest bucket sizes it first).
+    max_graph_bs: int = 512
     profile_batch: int = 2048            # tokens used for the peak profile run
     device: str = "cuda"                 # "cuda" (=ROCm HIP) | "cpu"
     seed: int = 0

@@ -295,6 +295,13 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
     key = x.device.index or 0
     ws = _SKINNY_WS.get(key)
     if ws is None or ws.numel() < need:
+        if torch.cuda.is_current_stream_capturing():
+            # growing the cached workspace mid-capture would free the
+            # buffer an earlier-captured graph writes (the r1 replay
+            # fault class); warmup must size it first
+            raise RuntimeError(
+                "skinny_gemm workspace grown during graph capture — "
+                "run an eager warmup at the largest bucket first")
         ws = torch.empty(need, dtype=torch.float32, device=x.device)
         _SKINNY_WS[key] = ws
     out = torch.empty(M, N, dtype=x.dtype, device=x.device)
@@ -316,9 +323,8 @@ def _use_skinny(M: int, N: int, K: int) -> bool:
     max_m = int(os.environ.get("GLLM_SKINNY_MAX_M", "0")) or SKINNY_MAX_M
     if K % 64 != 0 or N < 1024 or M > max_m:
         return False
-    if M > 64 and torch.cuda.is_current_stream_capturing():
-        return False  # MB>=2 under graph capture faults on replay (see
-        # config.max_graph_bs note); capture uses the library instead
+    # MB>=2 under capture: r1 faulted with dynamic LDS; the kernel now
+    # uses static LDS and captures cleanly (scripts/graph_skinny_repro)
     if K >= 2 * N:
         return True
     return M > 64 and N <= 8192

@@ -70,7 +70,10 @@ __global__ __launch_bounds__(BLOCK) void skinny_gemm_kernel(
   const int l16 = lane & 15;
   const int lhi = lane >> 4;
 
-  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // static LDS: the r1 replay fault ("write to read-only page") on
+  // MB>=2 under hipGraph replay is suspected dynamic-LDS + graph
+  // interaction; sizes are template constants, so declare statically
+  __shared__ __attribute__((aligned(16))) char smem[RING * TILE_B];
 
   // ---- glds source mapping -------------------------------------------
   // Tile image: W rows [0,64) then x rows [0, MB*64), 128 B each, row
@@ -207,10 +210,8 @@ template <int MB, int RING>
 void launch_skinny(float *partial, const __hip_bfloat16 *x,
                    const __hip_bfloat16 *w, int M, int N, int K, int k_slice,
                    int splitk, hipStream_t stream) {
-  constexpr int TILE_B = (BN + MB * 64) * 2 * BK;
-  const int lds = RING * TILE_B;
   hipLaunchKernelGGL((skinny_gemm_kernel<MB, RING>),
-                     dim3((N + BN - 1) / BN, splitk), dim3(BLOCK), lds,
+                     dim3((N + BN - 1) / BN, splitk), dim3(BLOCK), 0,
                      stream, partial, x, w, M, N, K, k_slice);
 }
 
