@@ -45,15 +45,27 @@ class LinearBase(nn.Module):
         self._w_dq = None
 
     def _gemm(self, x, bias=None):
-        """GEMM entry for all subclasses. fp8-block / int4 weights are
-        dequantized once (lazily, cached) into the compute dtype — the
-        dequant-fused MFMA GEMMs consuming them natively are round 2
-        (layers/quantization/ docstrings)."""
+        """GEMM entry for all subclasses.
+
+        fp8-block weights on GPU run NATIVE: per-token-group activation
+        quant + the block-scale fp8 MFMA weight-streaming kernel at
+        decode sizes (weights stay e4m3-resident, halving the decode
+        weight stream); large-M prefill calls dequantize per call (no
+        resident cache — prefill is compute-bound and bursty). CPU and
+        int4 keep the dequant path."""
         w = self.weight
         if self.fp8_block is not None:
+            if (x.is_cuda and x.dim() == 2 and x.dtype == torch.bfloat16
+                    and x.shape[0] <= 256 and x.shape[1] % 128 == 0
+                    and self.fp8_block == (128, 128) and ops.has_kernels()):
+                return ops.fp8_linear(x, w, self.weight_scale_inv, bias)
+            from gllm_amd.layers.quantization.fp8 import dequant_block_fp8
+            if x.is_cuda:
+                # prefill burst: dequant per call, nothing cached
+                wd = dequant_block_fp8(w, self.weight_scale_inv,
+                                       self.fp8_block, x.dtype)
+                return ops.linear(x, wd, bias)
             if self._w_dq is None:
-                from gllm_amd.layers.quantization.fp8 import \
-                    dequant_block_fp8
                 self._w_dq = dequant_block_fp8(
                     w, self.weight_scale_inv, self.fp8_block,
                     x.dtype).to(x.device)

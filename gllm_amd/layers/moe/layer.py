@@ -128,12 +128,22 @@ class FusedMoE(nn.Module):
         ``sel.any()`` form synced once per expert: 256/layer on
         DeepSeek-V3). DP padding rows carry expert id -1 and land in
         segment 0 of the shifted bincount (skipped)."""
-        if (x.is_cuda and self.fp8_block is None and self.int4_cfg is None
+        if (x.is_cuda and self.int4_cfg is None
                 and x.dtype == torch.bfloat16 and ops.has_kernels()):
-            return ops.fused_moe(
-                x.contiguous(), self.w13_weight, self.w2_weight,
-                weights, ids, expert_start=self.expert_start,
-                num_global_experts=self.num_experts)
+            if self.fp8_block is None:
+                return ops.fused_moe(
+                    x.contiguous(), self.w13_weight, self.w2_weight,
+                    weights, ids, expert_start=self.expert_start,
+                    num_global_experts=self.num_experts)
+            if (self.fp8_block == (128, 128)
+                    and x.shape[1] % 128 == 0
+                    and self.intermediate_per_rank % 128 == 0):
+                # e4m3-resident grouped GEMM (no dequant cache at all)
+                return ops.fused_moe_fp8(
+                    x.contiguous(), self.w13_weight,
+                    self.w13_weight_scale_inv, self.w2_weight,
+                    self.w2_weight_scale_inv, weights, ids,
+                    expert_start=self.expert_start)
         T = x.shape[0]
         out = torch.zeros_like(x)
         flat_ids = ids.long().flatten()                    # [T*K]

@@ -182,6 +182,91 @@ def fused_moe(x: torch.Tensor, w13: torch.Tensor, w2: torch.Tensor,
     return out
 
 
+# --------------------------------------------------------------- fp8
+def per_token_group_quant_fp8(x: torch.Tensor, group: int = 128,
+                              ue8m0: bool = False):
+    """[T, K] bf16 -> (e4m3 [T, K], fp32 scales [T, K/group]).
+    GPU kernel; CPU callers use the torch math in quantization/fp8.py."""
+    assert x.is_cuda and group == 128
+    T, K = x.shape
+    q = torch.empty(T, K, dtype=torch.float8_e4m3fn, device=x.device)
+    scales = torch.empty(T, K // group, dtype=torch.float32,
+                         device=x.device)
+    _gpu_kernels().per_token_group_quant_fp8(
+        x.contiguous(), q.view(torch.uint8), scales, ue8m0)
+    return q, scales
+
+
+_FP8_WS: dict = {}
+
+
+def fp8_linear(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
+               bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out = x @ dequant(w_q).T via the fp8 weight-streaming kernel
+    (M <= 256): per-token-group activation quant + block-scale fp8
+    MFMA. Weights stream at 1 B/elem — ~2x the bf16 decode rate."""
+    M, K = x.shape
+    N = w_q.shape[0]
+    aq, as_ = per_token_group_quant_fp8(x.contiguous())
+    splitk = _skinny_splitk(M, N, K)
+    need = splitk * M * N
+    key = x.device.index or 0
+    ws = _FP8_WS.get(key)
+    if ws is None or ws.numel() < need:
+        if torch.cuda.is_current_stream_capturing():
+            raise RuntimeError("fp8 workspace grown during graph capture")
+        ws = torch.empty(need, dtype=torch.float32, device=x.device)
+        _FP8_WS[key] = ws
+    out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    _gpu_kernels().fp8_skinny_gemm(
+        out, aq.view(torch.uint8), as_, w_q.view(torch.uint8),
+        w_scale.contiguous(),
+        bias.float() if bias is not None else None, ws, splitk)
+    return out
+
+
+def fused_moe_fp8(x: torch.Tensor, w13: torch.Tensor, w13_scale,
+                  w2: torch.Tensor, w2_scale,
+                  topk_weights: torch.Tensor, topk_ids: torch.Tensor,
+                  expert_start: int = 0) -> torch.Tensor:
+    """fp8 grouped-GEMM MoE: same align/sum pipeline as fused_moe with
+    block-scale fp8 GEMMs; expert weights stay e4m3-resident."""
+    T, K = x.shape
+    E_local, two_i, _ = w13.shape
+    inter = two_i // 2
+    topk = topk_ids.shape[1]
+    n_pairs = T * topk
+    block_m = 16 if n_pairs < 8 * E_local or n_pairs <= 512 else 64
+    cap = n_pairs + E_local * (block_m - 1) + 1
+    max_blocks = (n_pairs + block_m - 1) // block_m + E_local
+    dev = x.device
+    k = _gpu_kernels()
+    ids32 = topk_ids.int().contiguous()
+    sorted_ids = torch.empty(cap, dtype=torch.int32, device=dev)
+    expert_blocks = torch.empty(max_blocks, dtype=torch.int32, device=dev)
+    n_post = torch.empty(1, dtype=torch.int32, device=dev)
+    k.moe_align(ids32, E_local, expert_start, block_m, sorted_ids,
+                expert_blocks, n_post)
+    rows_pad = cap - 1 + block_m
+    aq, as_ = per_token_group_quant_fp8(x.contiguous())
+    inter1 = torch.empty(rows_pad, two_i, dtype=x.dtype, device=dev)
+    k.moe_gemm_fp8(inter1, aq.view(torch.uint8), as_,
+                   w13.view(torch.uint8), w13_scale.contiguous(),
+                   sorted_ids, expert_blocks, n_post, None, n_pairs, topk,
+                   block_m, False)
+    act = silu_and_mul(inter1)
+    actq, act_s = per_token_group_quant_fp8(act)
+    pair_out = x.new_zeros(n_pairs, K)
+    k.moe_gemm_fp8(pair_out, actq.view(torch.uint8), act_s,
+                   w2.view(torch.uint8), w2_scale.contiguous(),
+                   sorted_ids, expert_blocks, n_post,
+                   topk_weights.float().contiguous(), n_pairs, topk,
+                   block_m, True)
+    out = torch.empty(T, K, dtype=x.dtype, device=dev)
+    k.moe_sum(out, pair_out, topk)
+    return out
+
+
 # --------------------------------------------------------------- sampling
 def topk_topp_filter(probs: torch.Tensor, top_ks: torch.Tensor,
                      top_ps: torch.Tensor,

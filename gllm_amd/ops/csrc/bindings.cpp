@@ -42,6 +42,18 @@ void moe_gemm(torch::Tensor C, torch::Tensor A, torch::Tensor W,
               c10::optional<torch::Tensor> topk_weights, long n_pairs,
               long topk, long block_m, bool scatter);
 void moe_sum(torch::Tensor out, torch::Tensor pair_out, long topk);
+void per_token_group_quant_fp8(torch::Tensor x, torch::Tensor q,
+                               torch::Tensor scales, bool ue8m0);
+void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor as,
+                     torch::Tensor w, torch::Tensor ws,
+                     c10::optional<torch::Tensor> bias,
+                     torch::Tensor workspace, long splitk);
+void moe_gemm_fp8(torch::Tensor C, torch::Tensor A, torch::Tensor As,
+                  torch::Tensor W, torch::Tensor Ws,
+                  torch::Tensor sorted_ids, torch::Tensor expert_blocks,
+                  torch::Tensor n_post_pad,
+                  c10::optional<torch::Tensor> topk_weights, long n_pairs,
+                  long topk, long block_m, bool scatter);
 void apply_repetition_penalty(torch::Tensor logits, torch::Tensor pool,
                               torch::Tensor slots,
                               torch::Tensor penalties);
@@ -78,6 +90,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_gemm", &moe_gemm,
         "grouped MFMA GEMM over sorted (token, expert) pairs");
   m.def("moe_sum", &moe_sum, "sum pair outputs over topk");
+  m.def("per_token_group_quant_fp8", &per_token_group_quant_fp8,
+        "per (token, 128-group) e4m3 quant + fp32 scales");
+  m.def("fp8_skinny_gemm", &fp8_skinny_gemm,
+        "block-scale fp8 weight-streaming GEMM (decode)");
+  m.def("moe_gemm_fp8", &moe_gemm_fp8,
+        "block-scale fp8 grouped MoE GEMM");
   m.def("apply_repetition_penalty", &apply_repetition_penalty,
         "scaling penalty vs persistent seen-token mask pool");
   m.def("topk_topp_filter", &topk_topp_filter,

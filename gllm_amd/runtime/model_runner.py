@@ -73,13 +73,14 @@ class ModelRunner:
                         if isinstance(m, FusedMoE)]
             if moe_mods and not (cfg.device.startswith("cuda")
                                  and ops.has_kernels()
-                                 and all(m.fp8_block is None
+                                 and all((m.fp8_block is None
+                                          or m.fp8_block == (128, 128))
                                          and m.int4_cfg is None
                                          for m in moe_mods)):
-                # bf16 MoE on GPU runs the device-resident grouped
-                # MFMA GEMM pipeline (ops.fused_moe) and is capture-
-                # safe; quantized banks still run the host segment
-                # loop (illegal sync under stream capture)
+                # bf16 and fp8(128-block) MoE on GPU run the device-
+                # resident grouped MFMA GEMM pipelines (ops.fused_moe /
+                # fused_moe_fp8) and are capture-safe; int4 banks still
+                # run the host segment loop (illegal sync under capture)
                 cfg.use_graph = False
         num_pages = num_pages_override or self._size_kv_cache()
         self._allocate_kv(num_pages)

@@ -384,3 +384,65 @@ def test_topk_topp_filter_mixed_rows(kernels):
     ops.topk_topp_filter(out, ks, ps, mps)
     assert ((out > 0) == (ref > 0)).all()
     assert torch.allclose(out, ref, atol=1e-5, rtol=1e-4)
+
+
+# ------------------------------------------------------------ fp8
+def test_per_token_group_quant(kernels):
+    torch.manual_seed(5)
+    from gllm_amd import ops
+    from gllm_amd.layers.quantization import fp8 as qfp8
+    x = torch.randn(37, 512, dtype=torch.bfloat16, device="cuda") * 3
+    q, s = ops.per_token_group_quant_fp8(x)
+    qr, sr = qfp8.per_token_group_quant_fp8(x.cpu())
+    assert torch.allclose(s.cpu(), sr, rtol=1e-4)
+    assert (q.cpu().float() - qr.float()).abs().max() <= 1.0
+
+
+@pytest.mark.parametrize("case", [
+    # (M, N, K)
+    (1, 1024, 512), (8, 896, 1024), (64, 2048, 896), (200, 512, 1280),
+])
+def test_fp8_linear(kernels, case):
+    M, N, K = case
+    torch.manual_seed(M + N)
+    from gllm_amd import ops
+    from gllm_amd.layers.quantization import fp8 as qfp8
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(N, K, dtype=torch.bfloat16) / math.sqrt(K)
+    wq, ws = qfp8.block_quant_fp8(w)
+    out = ops.fp8_linear(x, wq.cuda(), ws.cuda())
+    # oracle: dequantized weights x quantized activations in fp32
+    aq, as_ = qfp8.per_token_group_quant_fp8(x.cpu())
+    adq = aq.float().view(M, K // 128, 128) * as_.unsqueeze(-1)
+    wdq = qfp8.dequant_block_fp8(wq, ws, (128, 128), torch.float32)
+    ref = adq.view(M, K).float() @ wdq.T
+    assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=2e-3)
+
+
+def test_fused_moe_fp8(kernels):
+    T, E, topk, H, I = 45, 8, 2, 512, 256
+    torch.manual_seed(9)
+    from gllm_amd import ops
+    from gllm_amd.layers.quantization import fp8 as qfp8
+    x = torch.randn(T, H, dtype=torch.bfloat16, device="cuda") / 4
+    w13l, w2l, w13q, w2q, w13s, w2s = [], [], [], [], [], []
+    for e in range(E):
+        a = torch.randn(2 * I, H, dtype=torch.bfloat16) / math.sqrt(H)
+        b = torch.randn(H, I, dtype=torch.bfloat16) / math.sqrt(I)
+        qa, sa = qfp8.block_quant_fp8(a)
+        qb, sb = qfp8.block_quant_fp8(b)
+        w13q.append(qa); w2q.append(qb); w13s.append(sa); w2s.append(sb)
+        w13l.append(qfp8.dequant_block_fp8(qa, sa, (128, 128),
+                                           torch.bfloat16))
+        w2l.append(qfp8.dequant_block_fp8(qb, sb, (128, 128),
+                                          torch.bfloat16))
+    logits = torch.randn(T, E, device="cuda")
+    weights, ids = ops.topk_softmax(logits, topk, True)
+    out = ops.fused_moe_fp8(
+        x, torch.stack(w13q).cuda(), torch.stack(w13s).cuda(),
+        torch.stack(w2q).cuda(), torch.stack(w2s).cuda(), weights,
+        ids.int())
+    ref = _moe_ref(x, torch.stack(w13l).cuda(), torch.stack(w2l).cuda(),
+                   weights, ids)
+    # activation quant noise on top of weight quant: loose tolerance
+    assert_close_bf16(out, ref, atol=8e-2, rtol=8e-2, frac=5e-3)
