@@ -395,7 +395,13 @@ def test_per_token_group_quant(kernels):
     q, s = ops.per_token_group_quant_fp8(x)
     qr, sr = qfp8.per_token_group_quant_fp8(x.cpu())
     assert torch.allclose(s.cpu(), sr, rtol=1e-4)
-    assert (q.cpu().float() - qr.float()).abs().max() <= 1.0
+    # round trip within e4m3 quantization error (3-bit mantissa: half a
+    # quantum = amax/448 * 16 near the top of the range); the raw codes
+    # may differ by one step where mul-by-1/s vs div-by-s rounds apart
+    dq = (q.float().view(37, 4, 128) * s.unsqueeze(-1)).view(37, 512)
+    err = (dq.cpu() - x.float().cpu()).abs()
+    tol = (sr * 16.5).repeat_interleave(128, dim=1)
+    assert (err <= tol).all(), float((err - tol).max())
 
 
 @pytest.mark.parametrize("case", [
