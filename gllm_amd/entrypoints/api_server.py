@@ -212,9 +212,14 @@ def build_app():
             total_out += n_out
             message = ChatMessage(role="assistant", content=text)
             if req.tools:
+                from gllm_amd.tokenizers.deepseek_v32 import \
+                    load_dsv32_encoder
                 from gllm_amd.tokenizers.tool_parsers import \
                     parse_tool_calls
-                parsed_text, calls = parse_tool_calls(text, served_model)
+                parsed_text, calls = parse_tool_calls(
+                    text, served_model,
+                    tools=[t.model_dump() for t in req.tools],
+                    encoder=load_dsv32_encoder(engine.config.model))
                 if calls:
                     message = ChatMessage(role="assistant",
                                           content=parsed_text or None,
@@ -299,25 +304,64 @@ def build_app():
         first_for = set()
         n = req.n or 1
         n_out = 0
-        async for j, chunk in _merged_stream(raw, token_ids, sampling, n,
-                                             mm=mm):
-            delta = DeltaMessage(content=chunk.text)
-            if j not in first_for:
-                delta.role = "assistant"
-                first_for.add(j)
-            n_out += 1 if chunk.token_id >= 0 else 0
-            out = ChatCompletionStreamResponse(
+        # streaming tool-call parse: content before the first marker
+        # streams as it arrives; each completed call becomes one
+        # tool_calls delta (reference stream_parser semantics)
+        stream_parsers = {}
+        cum_text = {}
+        if req.tools:
+            from gllm_amd.tokenizers.deepseek_v32 import load_dsv32_encoder
+            from gllm_amd.tokenizers.tool_parsers import get_tool_parser
+            tool_schemas = [t.model_dump() for t in req.tools]
+            enc = load_dsv32_encoder(engine.config.model)
+            for j in range(n):
+                stream_parsers[j] = get_tool_parser(
+                    served_model, encoder=enc).stream(tool_schemas)
+                cum_text[j] = ""
+
+        def _mk_out(j, delta, finish, chunk=None):
+            return ChatCompletionStreamResponse(
                 model=req.model or served_model,
                 choices=[ChatCompletionStreamChoice(
-                    index=j, delta=delta,
-                    finish_reason=chunk.finish_reason,
+                    index=j, delta=delta, finish_reason=finish,
                     logprobs={"content": [_lp_entry(chunk)]}
-                    if chunk.logprob is not None else None)])
-            if resp_id is None:
-                resp_id = out.id
+                    if chunk is not None and chunk.logprob is not None
+                    else None)])
+
+        async for j, chunk in _merged_stream(raw, token_ids, sampling, n,
+                                             mm=mm):
+            n_out += 1 if chunk.token_id >= 0 else 0
+            deltas = []
+            if j in stream_parsers:
+                cum_text[j] += chunk.text or ""
+                for d in stream_parsers[j].feed(cum_text[j]):
+                    if "content" in d:
+                        deltas.append(DeltaMessage(content=d["content"]))
+                    else:
+                        deltas.append(DeltaMessage(
+                            tool_calls=[d["tool_call"]]))
+                finish = chunk.finish_reason
+                if finish and stream_parsers[j].emitted_tool_calls:
+                    finish = "tool_calls"
+                if not deltas and finish is None:
+                    continue
+                if not deltas:
+                    deltas = [DeltaMessage()]
             else:
-                out.id = resp_id
-            yield f"data: {out.model_dump_json(exclude_none=True)}\n\n"
+                deltas = [DeltaMessage(content=chunk.text)]
+                finish = chunk.finish_reason
+            if j not in first_for:
+                deltas[0].role = "assistant"
+                first_for.add(j)
+            for i, delta in enumerate(deltas):
+                out = _mk_out(j, delta,
+                              finish if i == len(deltas) - 1 else None,
+                              chunk if not stream_parsers else None)
+                if resp_id is None:
+                    resp_id = out.id
+                else:
+                    out.id = resp_id
+                yield f"data: {out.model_dump_json(exclude_none=True)}\n\n"
         if req.stream_options and req.stream_options.get("include_usage"):
             out = ChatCompletionStreamResponse(
                 id=resp_id or "", model=req.model or served_model,

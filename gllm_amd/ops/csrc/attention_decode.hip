@@ -51,11 +51,9 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   const int lane_per_tok = D / 8;            // 16 lanes for D=128
   const int toks_per_iter = BLOCK / lane_per_tok;
 
-  // LDS: double-buffered V tile + scores (phase-pipelined: chunk c+1's
-  // global loads are in flight while chunk c's softmax/PV run — the r1
-  // kernel serialized load/QK^T/softmax/PV phases and capped at ~3 TB/s)
-  __shared__ __hip_bfloat16 v_tile[2][CHUNK][D];
-  __shared__ float s_scores[2][G][CHUNK];
+  // LDS: V tile + scores + p + softmax state
+  __shared__ __hip_bfloat16 v_tile[CHUNK][D];
+  __shared__ float s_scores[G][CHUNK];
   __shared__ float s_m[G], s_l[G], s_alpha[G];
 
   const int tid = threadIdx.x;
@@ -92,52 +90,27 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
 
   const int *bt = block_table + (long)b * max_pages;
 
-  // ---- register staging: this thread's K/V rows of one chunk ----
-  constexpr int MAX_IT = (CHUNK + (BLOCK / (D / 8)) - 1) / (BLOCK / (D / 8));
-  shortx8 kreg[2][MAX_IT], vreg[2][MAX_IT];
-  auto issue_loads = [&](int c0, int set) {
-#pragma unroll
-    for (int it = 0; it < MAX_IT; ++it) {
-      const int tt = it * toks_per_iter + tok_slot;
-      const int tok = c0 + tt;
-      if (tok < t1) {
-        const long row = ((long)bt[tok / page_size] * page_size +
-                          tok % page_size);
-        kreg[set][it] = *reinterpret_cast<const shortx8 *>(
-            k_cache + (row * num_kv_heads + kvh) * D + d_off);
-        vreg[set][it] = *reinterpret_cast<const shortx8 *>(
-            v_cache + (row * num_kv_heads + kvh) * D + d_off);
-      } else {
-        kreg[set][it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
-        vreg[set][it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-    }
-  };
-
-  issue_loads(t0, 0);
-
-  int idx = 0;
-  for (int c0 = t0; c0 < t1; c0 += CHUNK, ++idx) {
+  for (int c0 = t0; c0 < t1; c0 += CHUNK) {
     const int c_len = min(CHUNK, t1 - c0);
-    const int buf = idx & 1;
-    const int set = idx & 1;
-    // issue-early: next chunk's global loads go in flight now; the
-    // compiler's counted waits let this chunk's registers be consumed
-    // while they fly
-    if (c0 + CHUNK < t1) issue_loads(c0 + CHUNK, set ^ 1);
-
-    // ---------- phase A: dots from registers + V staging ----------
-#pragma unroll
-    for (int it = 0; it < MAX_IT; ++it) {
+    // ---------- phase A: scores + V staging ----------
+    for (int it = 0; it < (c_len + toks_per_iter - 1) / toks_per_iter; ++it) {
       const int tt = it * toks_per_iter + tok_slot;  // token within chunk
       float dot[G];
 #pragma unroll
       for (int h = 0; h < G; ++h) dot[h] = 0.f;
       if (tt < c_len) {
+        const int tok = c0 + tt;
+        const long row = ((long)bt[tok / page_size] * page_size +
+                          tok % page_size);
+        const __hip_bfloat16 *kp =
+            k_cache + (row * num_kv_heads + kvh) * D + d_off;
+        const __hip_bfloat16 *vp =
+            v_cache + (row * num_kv_heads + kvh) * D + d_off;
+        shortx8 kv8 = *reinterpret_cast<const shortx8 *>(kp);
+        shortx8 vv8 = *reinterpret_cast<const shortx8 *>(vp);
         float kf[8];
-        unpack8<__hip_bfloat16>(kreg[set][it], kf);
-        *reinterpret_cast<shortx8 *>(&v_tile[buf][tt][d_off]) =
-            vreg[set][it];
+        unpack8<__hip_bfloat16>(kv8, kf);
+        *reinterpret_cast<shortx8 *>(&v_tile[tt][d_off]) = vv8;
 #pragma unroll
         for (int h = 0; h < G; ++h) {
 #pragma unroll
@@ -148,7 +121,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
 #pragma unroll
       for (int h = 0; h < G; ++h) {
         float v = group_reduce_sum<D / 8>(dot[h]);
-        if (dlane == 0 && tt < c_len) s_scores[buf][h][tt] = v * scale;
+        if (dlane == 0 && tt < c_len) s_scores[h][tt] = v * scale;
       }
     }
     __syncthreads();
@@ -164,8 +137,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
       const int j = tid % TPH;
       if (h < G) {
         float mx = -INFINITY;
-        for (int t = j; t < c_len; t += TPH)
-          mx = fmaxf(mx, s_scores[buf][h][t]);
+        for (int t = j; t < c_len; t += TPH) mx = fmaxf(mx, s_scores[h][t]);
 #pragma unroll
         for (int off = TPH / 2; off > 0; off >>= 1)
           mx = fmaxf(mx, __shfl_xor(mx, off, 64));
@@ -173,8 +145,8 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
         const float m_new = fmaxf(m_old, mx);
         float psum = 0.f;
         for (int t = j; t < c_len; t += TPH) {
-          const float p = __expf(s_scores[buf][h][t] - m_new);
-          s_scores[buf][h][t] = p;            // reuse scores LDS as p
+          const float p = __expf(s_scores[h][t] - m_new);
+          s_scores[h][t] = p;                 // reuse scores LDS as p
           psum += p;
         }
 #pragma unroll
@@ -202,9 +174,9 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
       floatx4 a = acc[i];
       a.x *= al; a.y *= al; a.z *= al; a.w *= al;
       for (int t = 0; t < c_len; ++t) {
-        const float p = s_scores[buf][h][t];
+        const float p = s_scores[h][t];
         const shortx4 v4 =
-            *reinterpret_cast<const shortx4 *>(&v_tile[buf][t][d0]);
+            *reinterpret_cast<const shortx4 *>(&v_tile[t][d0]);
         const __hip_bfloat16 *ve =
             reinterpret_cast<const __hip_bfloat16 *>(&v4);
         a.x += p * __bfloat162float(ve[0]);
@@ -214,8 +186,7 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
       }
       acc[i] = a;
     }
-    // no tail barrier: the next chunk's phase A writes the OTHER buffer,
-    // and its post-A barrier orders everything two chunks back
+    __syncthreads();
   }
 
   // ---------- epilogue: normalized split partial + lse ----------

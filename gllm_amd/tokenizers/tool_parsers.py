@@ -105,6 +105,9 @@ def _mk_call(name: str, arguments) -> ToolCall:
 # ----------------------------------------------------------------- base
 class ToolParser:
     name = "base"
+    # markers whose PREFIX at the end of the text must be held back
+    # from content streaming (a half-arrived "<tool_c" is not content)
+    start_markers: Tuple[str, ...] = ()
 
     def content_head(self, text: str) -> str:
         """Text before the first tool-call marker (streamed as plain
@@ -137,12 +140,26 @@ class StreamingToolParser:
     def emitted_tool_calls(self) -> bool:
         return self._calls_sent > 0
 
+    def _held(self, head: str) -> int:
+        """Length of the trailing piece that may be the start of a
+        marker (withheld until it resolves either way)."""
+        best = 0
+        for mk in self.parser.start_markers:
+            for k in range(min(len(mk) - 1, len(head)), 0, -1):
+                if head.endswith(mk[:k]):
+                    best = max(best, k)
+                    break
+        return best
+
     def feed(self, full_text: str) -> List[dict]:
         out: List[dict] = []
         head = self.parser.content_head(full_text)
-        if self._content_sent < len(head):
-            out.append({"content": head[self._content_sent:]})
-            self._content_sent = len(head)
+        emit_to = len(head)
+        if emit_to == len(full_text):  # no marker yet: hold a prefix
+            emit_to -= self._held(head)
+        if self._content_sent < emit_to:
+            out.append({"content": head[self._content_sent:emit_to]})
+            self._content_sent = emit_to
         _, calls = self.parser.parse(full_text, self.tools)
         while self._calls_sent < len(calls):
             c = calls[self._calls_sent]
@@ -163,6 +180,7 @@ class HermesToolParser(ToolParser):
 
     name = "hermes"
     _START = "<tool_call>"
+    start_markers = ("<tool_call>",)
     _RE = re.compile(r"<tool_call>\s*(\{.*?\})\s*</tool_call>", re.DOTALL)
 
     def content_head(self, text: str) -> str:
@@ -191,6 +209,7 @@ class Qwen3XmlToolParser(ToolParser):
 
     name = "qwen3_xml"
     _START = "<tool_call>"
+    start_markers = ("<tool_call>",)
     _FUNC = re.compile(r"<function=([^>\n]+)>(.*?)</function>", re.DOTALL)
     _PARAM = re.compile(
         r"<parameter=([^>\n]+)>(.*?)(?:</parameter>|(?=<parameter=)|\Z)",
@@ -219,6 +238,7 @@ class KimiToolParser(ToolParser):
 
     name = "kimi"
     _START = "<|tool_calls_section_begin|>"
+    start_markers = ("<|tool_calls_section_begin|>",)
     _CALL = re.compile(
         r"<\|tool_call_begin\|>\s*([^\s<]+?)\s*"
         r"<\|tool_call_argument_begin\|>\s*(.*?)\s*<\|tool_call_end\|>",
@@ -250,6 +270,7 @@ class DsmlToolParser(ToolParser):
 
     name = "dsml"
     _STARTS = ("<｜DSML｜function_calls", "<function_calls")
+    start_markers = _STARTS
     _INVOKE = re.compile(
         r"<(?:｜DSML｜)?invoke\s+name=\"([^\"]+)\">(.*?)"
         r"</(?:｜DSML｜)?invoke>", re.DOTALL)
@@ -307,6 +328,7 @@ class DsmlToolParser(ToolParser):
 class MistralToolParser(ToolParser):
     name = "mistral"
     _START = "[TOOL_CALLS]"
+    start_markers = ("[TOOL_CALLS]",)
     _RE = re.compile(r"\[TOOL_CALLS\]\s*(\[.*\])", re.DOTALL)
 
     def content_head(self, text: str) -> str:
@@ -332,7 +354,7 @@ def get_tool_parser(model_name: str = "", architecture: str = "",
     DSV3.2 bundled decoder (entrypoints inject it when the checkpoint
     ships one)."""
     key = f"{model_name} {architecture}".lower()
-    if "deepseekv32" in key.replace("_", "").replace("-", "") or \
+    if "deepseekv32" in key.replace("_", "").replace("-", "").replace(".", "") or \
             "dsml" in key:
         return DsmlToolParser(encoder)
     if "kimi" in key:
