@@ -17,6 +17,7 @@
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
+#include <cstdlib>
 #include "common.h"
 
 namespace {
@@ -247,11 +248,16 @@ void launch_decode(torch::Tensor &out, const torch::Tensor &q,
   const int Hkv = k_cache.size(2);
   constexpr int CHUNK = 128;
   // split heuristic: enough workgroups to fill 256 CUs x 8 XCDs
+  static int splits_env = [] {
+    const char *e = getenv("GLLM_DECODE_SPLITS");
+    return e ? atoi(e) : 0;
+  }();
   int splits = 1;
   const int base_wgs = B * Hkv;
   while (splits < 16 && base_wgs * splits < 640 &&
          splits * 2 <= (max_seq_len + CHUNK - 1) / CHUNK)
     splits *= 2;
+  if (splits_env > 0) splits = splits_env;
   auto opts = q.options().dtype(at::kFloat);
   auto partial = torch::empty({splits, B, Hq, D}, opts);
   auto lse = torch::empty({splits, B, Hq}, opts);

@@ -59,3 +59,35 @@ def decode_bench():
 
 if os.environ.get("ATTN_DECODE"):
     decode_bench()
+
+
+def decode_sweep():
+    """Decode attention TB/s across B/ctx/splits (GLLM_DECODE_SPLITS)."""
+    import itertools
+    Hq, Hkv, D, ps = 40, 8, 128, 16
+    for B, S in itertools.product([32, 64, 256], [1024, 4096]):
+        n_pages = B * (S // ps) + 1
+        torch.manual_seed(0)
+        k_cache = torch.randn(n_pages, ps, Hkv, D, dtype=torch.bfloat16,
+                              device="cuda")
+        v_cache = torch.randn(n_pages, ps, Hkv, D, dtype=torch.bfloat16,
+                              device="cuda")
+        bt = torch.arange(1, n_pages, dtype=torch.int32,
+                          device="cuda").reshape(B, S // ps)
+        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device="cuda")
+        seq_lens = torch.full((B,), S, dtype=torch.int32, device="cuda")
+        qsl = torch.arange(B + 1, dtype=torch.int32, device="cuda")
+        sc = D ** -0.5
+        for _ in range(5):
+            ops.paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl, sc)
+        torch.cuda.synchronize()
+        t0 = time.time()
+        it = 30
+        for _ in range(it):
+            ops.paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl, sc)
+        torch.cuda.synchronize()
+        dt = (time.time() - t0) / it
+        kv_bytes = 2.0 * B * S * Hkv * D * 2 * 2
+        print(f"decode B={B:4d} S={S:5d}: {dt*1e6:8.1f} us  "
+              f"KV {kv_bytes/dt/1e12:5.2f} TB/s")
+if __name__ == "__main__" and os.environ.get("DECODE_SWEEP"): decode_sweep()

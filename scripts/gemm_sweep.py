@@ -101,3 +101,27 @@ if os.environ.get("SWEEP_SKINNY"):
             dt = (time.time() - t0) / 20
             print(f"SK M={M:5d} {name:8s} {dt*1e6:9.1f} us  "
                   f"W-stream {2.0*N*K/dt/1e12:5.2f} TB/s")
+
+if os.environ.get("SWEEP_FP8"):
+    from gllm_amd import ops
+    from gllm_amd.layers.quantization import fp8 as qfp8
+    print("=== fp8_linear (quant+gemm end to end) ===")
+    for M in [32, 64, 128, 256]:
+        for name, N, K in SHAPES:
+            if K % 128 or N > 60000:
+                continue
+            x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+            wq = torch.randint(0, 255, (N, K), dtype=torch.uint8,
+                               device="cuda").view(torch.float8_e4m3fn)
+            ws = torch.rand(-(-N // 128), K // 128, device="cuda") * 0.01
+            for _ in range(3):
+                ops.fp8_linear(x, wq, ws)
+            torch.cuda.synchronize()
+            t0 = time.time()
+            for _ in range(20):
+                ops.fp8_linear(x, wq, ws)
+            torch.cuda.synchronize()
+            dt = (time.time() - t0) / 20
+            print(f"FP8 M={M:5d} {name:8s} {dt*1e6:9.1f} us  "
+                  f"W-stream {1.0*N*K/dt/1e12:5.2f} TB/s "
+                  f"(bf16-equiv {2.0*N*K/dt/1e12:5.2f})")
