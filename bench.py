@@ -192,31 +192,42 @@ def main():
         seqs.append(s)
     # ---- ramp: arrivals (burst or paced at --qps) until every seq has
     # produced its first token; TTFT measured from each seq's arrival ----
-    # Paced arrivals are wall-clock driven and would desynchronize the
-    # replicated schedulers across ranks: force burst for multi-GPU runs.
-    if world > 1:
-        args.qps = 0.0
+    # Paced arrivals under multi-GPU: rank 0's wall clock decides how
+    # many requests release each tick and BROADCASTS the count, so every
+    # replicated scheduler applies the identical release schedule
+    # (deterministic-by-construction; wall time never drives a
+    # non-rank-0 decision). BASELINE.json's "p50 TTFT at fixed QPS"
+    # comes from this paced ramp.
     ttfts = {}
     arrivals = {}
     if args.qps > 0:
         to_release = list(seqs)
         next_t = time.time()
-        while len(ttfts) < len(seqs):
+        while True:
+            k = 0
+            if rank == 0:
+                now = time.time()
+                while k < len(to_release) and now >= next_t:
+                    k += 1
+                    next_t += 1.0 / args.qps
+            if world > 1:
+                obj = [k]
+                dist.broadcast_object_list(obj, src=0)
+                k = int(obj[0])
             now = time.time()
-            while to_release and now >= next_t:
+            for _ in range(k):
                 s = to_release.pop(0)
                 arrivals[s.seq_id] = now
                 eng.add_requests([s])
-                next_t += 1.0 / args.qps
-                now = time.time()
             eng.step_tick()
             now = time.time()
             for s in seqs:
                 if s.seq_id in arrivals and s.seq_id not in ttfts \
                         and s.num_output_tokens > 0:
                     ttfts[s.seq_id] = (now - arrivals[s.seq_id]) * 1000.0
-            if not to_release and not eng.scheduler.has_work() and \
-                    not getattr(eng, "pending", None):
+            # deterministic exit: engine state is identical on every rank
+            if not to_release and \
+                    all(s.num_output_tokens > 0 for s in seqs):
                 break
     else:
         eng.add_requests(seqs)

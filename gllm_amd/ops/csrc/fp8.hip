@@ -100,9 +100,25 @@ __global__ __launch_bounds__(BLOCK) void fp8_skinny_kernel(
   const int l16 = lane & 15;
   const int lhi = lane >> 4;
 
-  __shared__ __attribute__((aligned(16))) char smem[RING * TILE_B];
+  __shared__ __attribute__((aligned(16)))
+      char smem[RING * ((BN + MB * 64) * ROW_B + MB * 256)];
+  // the whole k-slice's weight scales, staged ONCE (a per-tile global
+  // scale load would force a vmcnt drain that serializes the glds ring)
+  __shared__ float ws_lds[256];
+
+  const int wblk = n0 / 128;  // BN=64 tile sits inside one n scale block
+  for (int i = tid; i < nkt; i += BLOCK)
+    ws_lds[i] = ws[(long)wblk * kgroups + kb_begin + i];
+  __syncthreads();
 
   constexpr int GL_PER_WAVE = TILE_B / 1024 / 4;
+  // per-stage vm ops per wave: GL_PER_WAVE glds16 + MB glds4 (the
+  // activation scales gather through the SAME lds-dma pipeline so the
+  // counted waits stay exact — a plain global scale load would make the
+  // compiler emit a drain that serializes the ring)
+  constexpr int VM_PER_STAGE = GL_PER_WAVE + MB;
+  constexpr int ASC_OFF = TILE_B;  // asc area at the slot tail
+  constexpr int SLOT_B = TILE_B + MB * 64 * 4;
   const unsigned char *gsrc[GL_PER_WAVE];
   {
 #pragma unroll
@@ -122,7 +138,7 @@ __global__ __launch_bounds__(BLOCK) void fp8_skinny_kernel(
 
   auto stage = [&](int kt, int slot) {
     const long kb = (long)(kb_begin + kt) * BK;
-    char *base = smem + slot * TILE_B;
+    char *base = smem + slot * SLOT_B;
 #pragma unroll
     for (int j = 0; j < GL_PER_WAVE; ++j) {
       const int off = (wave * GL_PER_WAVE + j) * 1024;
@@ -130,6 +146,17 @@ __global__ __launch_bounds__(BLOCK) void fp8_skinny_kernel(
         glds16<2>(gsrc[j] + kb, base + off);
       else
         glds16<0>(gsrc[j] + kb, base + off);
+    }
+    // every wave issues the identical scale gather (same values land
+    // at the same LDS bytes) so per-wave vm counters stay uniform
+#pragma unroll
+    for (int mb = 0; mb < MB; ++mb) {
+      const int m = min(mb * 64 + lane, M - 1);
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const unsigned int *>(
+              as + (long)m * kgroups + kb_begin + kt),
+          reinterpret_cast<unsigned int *>(base + ASC_OFF + mb * 256), 4,
+          0, 0);
     }
   };
 
@@ -142,27 +169,29 @@ __global__ __launch_bounds__(BLOCK) void fp8_skinny_kernel(
   const int pre = min(RING - 1, nkt);
   for (int t = 0; t < pre; ++t) stage(t, t % RING);
 
-  const int wblk = n0 / 128;  // BN=64 tile sits inside one n scale block
-
   for (int kt = 0; kt < nkt; ++kt) {
     const int slot = kt % RING;
     if (kt + RING - 1 < nkt) stage(kt + RING - 1, (kt + RING - 1) % RING);
     const int ahead = min(nkt - 1 - kt, RING - 1);
     if (RING >= 4 && ahead == 3) {
-      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(3 * GL_PER_WAVE) : "memory");
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(3 * VM_PER_STAGE)
+                   : "memory");
     } else if (RING >= 3 && ahead == 2) {
-      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * GL_PER_WAVE) : "memory");
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * VM_PER_STAGE)
+                   : "memory");
     } else if (ahead == 1) {
-      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(1 * GL_PER_WAVE) : "memory");
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(1 * VM_PER_STAGE)
+                   : "memory");
     } else {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
     __builtin_amdgcn_s_barrier();
 
-    const char *wbase = smem + slot * TILE_B;
+    const char *wbase = smem + slot * SLOT_B;
     const char *xbase = wbase + BN * ROW_B;
-    const int kb = kb_begin + kt;
-    const float wsc = ws[(long)wblk * kgroups + kb];
+    const float *asc_lds =
+        reinterpret_cast<const float *>(wbase + ASC_OFF);
+    const float wsc = ws_lds[kt];
 
 #pragma unroll
     for (int mb = 0; mb < MB; ++mb) {
@@ -183,13 +212,10 @@ __global__ __launch_bounds__(BLOCK) void fp8_skinny_kernel(
               afrag, bfrag, sub[nt], 0, 0, 0);
         }
       }
-      // fold the scale block: rows lhi*4+r of this wave's m tile
       float asc[4];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int m = min(mb * 64 + wave * 16 + lhi * 4 + r, M - 1);
-        asc[r] = as[(long)m * kgroups + kb] * wsc;
-      }
+      for (int r = 0; r < 4; ++r)
+        asc[r] = asc_lds[mb * 64 + min(wave * 16 + lhi * 4 + r, 63)] * wsc;
 #pragma unroll
       for (int nt = 0; nt < BN / 16; ++nt)
 #pragma unroll
@@ -440,6 +466,7 @@ void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor as,
   const int M = aq.size(0), K = aq.size(1), N = w.size(0);
   TORCH_CHECK(aq.is_contiguous() && w.is_contiguous() && out.is_contiguous());
   TORCH_CHECK(K % BK == 0, "fp8 gemm: K must be a multiple of 128");
+  TORCH_CHECK(K / BK <= 256, "fp8 skinny: K <= 32768 (ws_lds)");
   TORCH_CHECK(M <= 256, "fp8 skinny: M <= 256");
   const int n_wg = (N + BN - 1) / BN;
   int splitk = (int)splitk_arg;
