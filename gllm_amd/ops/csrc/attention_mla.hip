@@ -5,25 +5,24 @@
 // ONE varlen kernel here. The absorbed latent cache holds a single
 // shared row per token [c_kv (512) | k_pe (64)] = 576 dims; attention is
 // MQA: every q head attends the same K row, and V is the first 512 dims
-// of that row (zero-copy view — this kernel never touches a separate V
-// cache, so each KV byte is read exactly once for ALL heads).
+// of that row (zero-copy view — each KV byte is read exactly once for
+// ALL heads, and the PV B-fragments read straight out of the K tile).
 //
-// MI355X-native design: because the KV row is shared, a (token, head)
-// pair is just a GEMM row — we flatten q to [T*H, 576] rows and tile
-// rows by 64 per workgroup, exactly like a dense attention q-tile. The
-// causal mask depends only on the row's token. One kernel then serves
-// decode (1 token x 128 heads = 2 row tiles), chunked prefill, prefix
-// hits and mixed batches; decode additionally split-KVs over blockIdx.z
-// with an LSE merge (fp32 partials), replacing the reference's separate
-// fa3/flashmla/triton decode backends and its bounded-workspace
-// chunked-context gather+merge loop (the paged read IS the gather).
+// MI355X-native design: a (token, head) pair is a GEMM row — q flattens
+// to [T*H, 576] rows tiled 64 per workgroup; the causal mask depends
+// only on the row's token, so one kernel serves decode (1 token x 128
+// heads = 2 row tiles), chunked prefill, prefix hits and mixed batches.
+// Decode split-KVs over blockIdx.z with an fp32 LSE merge.
 //
-// Per KV tile of 64 tokens: K [64, 576] staged to LDS (XOR-swizzled,
-// T2), V^T [512, 64] transposed at staging from the same global rows;
-// QK^T = 18 k-steps of mfma_f32_16x16x32_bf16 with 4 independent score
-// accumulators; online softmax per row; PV = 32 n-tiles x 2 k-steps.
-// Single-buffered tiles (156 KB LDS) with T14 issue-early register
-// staging: the next tile's global loads issue before this tile's MFMAs.
+// v2 execution structure (v1 ran 1 wave/SIMD and was latency-bound at
+// ~300 GB/s KV): 512 threads = 8 waves at 2 waves/SIMD. Wave w owns
+// row group w%4 (16 rows) and V half w/4 (256 dims) — the o
+// accumulator halves to 64 VGPRs so two waves co-reside per SIMD; the
+// QK^T S-tile is computed twice per row group (MFMA is not the bound).
+// K tiles stream through a 3-slot LDS ring via global_load_lds issued
+// by waves 0-3 with counted s_waitcnt vmcnt BEFORE the barrier (the
+// proven skinny-GEMM pipeline): consumers see landed tiles at barrier
+// release without ever waiting a vm counter of their own.
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
@@ -34,20 +33,21 @@ typedef __attribute__((ext_vector_type(4))) float mfma_f4;
 
 namespace {
 
-constexpr int BLOCK = 256;
+constexpr int BLOCK = 512;    // 8 waves
 constexpr int BQ = 64;        // flattened (token, head) rows per workgroup
-constexpr int BKV = 32;       // kv tokens per tile (staging regs: 9x16B/lane)
+constexpr int BKV = 32;       // kv tokens per ring slot
+constexpr int RING = 3;
 constexpr int DK = 576;       // latent dims (kv_lora 512 + rope 64)
-constexpr int DV = 512;       // value dims (first 512 of the K row)
+constexpr int DV = 512;
 constexpr int KT = DK / 32;   // QK^T k-steps (18)
-constexpr int NT = DV / 16;   // PV n-tiles (32)
-constexpr int NF = BKV / 16;  // score fragments per wave (4)
-constexpr int VT_STRIDE = BKV + 8;
-constexpr int CPR = DK / 8;   // 16-B chunks per K row (72)
-constexpr int N_IT = BKV * CPR / BLOCK;  // staging iterations (18)
+constexpr int NTW = 16;       // PV 16-wide n tiles per wave (256 dims)
+constexpr int NF = BKV / 16;  // score fragments per wave (2)
+constexpr int ROW_B = DK * 2; // K row bytes (1152)
+constexpr int TILE_B = BKV * ROW_B;          // 36864
+constexpr int GL_PER_WAVE = TILE_B / 1024 / 4;  // 9 (waves 0-3 stage)
+constexpr int PT_STRIDE = BKV + 8;
 
 DEV_INLINE int kswz(int row, int byte_off) {
-  // XOR swizzle within the 1152-B K row (mask 7 -> 8-chunk period)
   return byte_off ^ ((row & 7) << 4);
 }
 
@@ -79,6 +79,8 @@ __global__ __launch_bounds__(BLOCK, 1) void mla_attention_kernel(
   const int lane = tid & 63;
   const int l16 = lane & 15;
   const int lhi = lane >> 4;
+  const int rg = wave & 3;        // row group (16 rows)
+  const int vhalf = wave >> 2;    // V half (256 dims)
 
   // causal kv bound for this row tile, then the split's share of it
   const int last_tok = min(rows - 1, (tile + 1) * BQ - 1) / H;
@@ -92,7 +94,6 @@ __global__ __launch_bounds__(BLOCK, 1) void mla_attention_kernel(
 
   if (kv_lo >= kv_hi) {
     if (SPLIT) {
-      // empty split: -inf lse rows so the merge skips them
       for (int r = tid; r < BQ; r += BLOCK) {
         if (tile * BQ + r < rows)
           partial_lse[(long)split * th_total + row_base + r] = -INFINITY;
@@ -101,12 +102,11 @@ __global__ __launch_bounds__(BLOCK, 1) void mla_attention_kernel(
     return;
   }
 
-  __shared__ __hip_bfloat16 k_tile[BKV * DK];
-  __shared__ __hip_bfloat16 vt_tile[DV * VT_STRIDE];
-  __shared__ __hip_bfloat16 p_tile[4][16 * VT_STRIDE];
+  __shared__ __attribute__((aligned(16))) char k_ring[RING * TILE_B];
+  __shared__ __hip_bfloat16 p_tile[8][16 * PT_STRIDE];
 
-  // ---- Q fragments (A-operand): row = wave*16 + l16 ----
-  const int frow = tile * BQ + wave * 16 + l16;
+  // ---- Q fragments (A-operand): row = rg*16 + l16 ----
+  const int frow = tile * BQ + rg * 16 + l16;
   const bool row_valid = frow < rows;
   mfma_bf8 qfrag[KT];
   {
@@ -124,58 +124,62 @@ __global__ __launch_bounds__(BLOCK, 1) void mla_attention_kernel(
   float m_run[4], l_run[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
-  mfma_f4 o_acc[NT];
+  mfma_f4 o_acc[NTW];
 #pragma unroll
-  for (int nt = 0; nt < NT; ++nt) o_acc[nt] = mfma_f4{0, 0, 0, 0};
+  for (int nt = 0; nt < NTW; ++nt) o_acc[nt] = mfma_f4{0, 0, 0, 0};
 
   const int *bt = block_table + (long)b * max_pages;
 
-  // ---- staging: thread covers chunks idx, idx+256, ... of the tile ----
-  shortx8 kreg[N_IT];
-  auto stage_load = [&](int kv0) {
+  // ---- glds staging (waves 0-3): 9 chunks of 1 KiB per wave ----
+  auto stage = [&](int kt_idx, int slot) {
+    const int kv0 = kv_lo + kt_idx * BKV;
+    char *base = k_ring + slot * TILE_B;
 #pragma unroll
-    for (int it = 0; it < N_IT; ++it) {
-      const int idx = tid + it * BLOCK;
-      const int row = idx / CPR;
-      const int c = idx % CPR;
+    for (int j = 0; j < GL_PER_WAVE; ++j) {
+      const int p = (wave * GL_PER_WAVE + j) * 1024 + lane * 16;
+      const int row = p / ROW_B;
+      const int col = kswz(row, p % ROW_B);
       const int tok = kv0 + row;
-      if (tok < kv_hi) {
-        const long crow =
-            (long)bt[tok / page_size] * page_size + tok % page_size;
-        kreg[it] = *reinterpret_cast<const shortx8 *>(kc + crow * DK + c * 8);
+      // clamp to a valid cache row; masked later by the score mask
+      const int ctok = min(tok, seq_len - 1);
+      const long crow =
+          (long)bt[ctok / page_size] * page_size + ctok % page_size;
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const unsigned int *>(
+              reinterpret_cast<const char *>(kc) + crow * ROW_B + col),
+          reinterpret_cast<unsigned int *>(
+              base + (wave * GL_PER_WAVE + j) * 1024),
+          16, 0, 0);
+    }
+  };
+
+  const int nkt = (kv_hi - kv_lo + BKV - 1) / BKV;
+  const int pre = min(RING - 1, nkt);
+  if (wave < 4)
+    for (int t = 0; t < pre; ++t) stage(t, t % RING);
+
+  for (int kt_i = 0; kt_i < nkt; ++kt_i) {
+    const int kv0 = kv_lo + kt_i * BKV;
+    const int slot = kt_i % RING;
+    if (wave < 4) {
+      if (kt_i + RING - 1 < nkt)
+        stage(kt_i + RING - 1, (kt_i + RING - 1) % RING);
+      const int ahead = min(nkt - 1 - kt_i, RING - 1);
+      if (ahead >= 2) {
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * GL_PER_WAVE)
+                     : "memory");
+      } else if (ahead == 1) {
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(1 * GL_PER_WAVE)
+                     : "memory");
       } else {
-        kreg[it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
     }
-  };
-  auto stage_write = [&]() {
-#pragma unroll
-    for (int it = 0; it < N_IT; ++it) {
-      const int idx = tid + it * BLOCK;
-      const int row = idx / CPR;
-      const int c = idx % CPR;
-      *reinterpret_cast<shortx8 *>(
-          reinterpret_cast<char *>(&k_tile[row * DK]) + kswz(row, c * 16)) =
-          kreg[it];
-      if (c < DV / 8) {
-        const __hip_bfloat16 *ve =
-            reinterpret_cast<const __hip_bfloat16 *>(&kreg[it]);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          vt_tile[(c * 8 + j) * VT_STRIDE + row] = ve[j];
-      }
-    }
-  };
+    __syncthreads();  // stagers waited first -> tile kt_i is landed
 
-  stage_load(kv_lo);
-  stage_write();
-  __syncthreads();
+    const char *ktile = k_ring + slot * TILE_B;
 
-  for (int kv0 = kv_lo; kv0 < kv_hi; kv0 += BKV) {
-    const bool has_next = kv0 + BKV < kv_hi;
-    if (has_next) stage_load(kv0 + BKV);  // T14 issue-early
-
-    // ---------- QK^T ----------
+    // ---------- QK^T: S[16 rows x 32 tokens] ----------
     mfma_f4 s_frag[NF];
 #pragma unroll
     for (int f = 0; f < NF; ++f) s_frag[f] = mfma_f4{0, 0, 0, 0};
@@ -186,8 +190,7 @@ __global__ __launch_bounds__(BLOCK, 1) void mla_attention_kernel(
       for (int f = 0; f < NF; ++f) {
         const int krow = f * 16 + l16;
         mfma_bf8 bfrag = *reinterpret_cast<const mfma_bf8 *>(
-            reinterpret_cast<char *>(&k_tile[krow * DK]) +
-            kswz(krow, (kt * 32 + lhi * 8) * 2));
+            ktile + krow * ROW_B + kswz(krow, (kt * 32 + lhi * 8) * 2));
         s_frag[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             qfrag[kt], bfrag, s_frag[f], 0, 0, 0);
       }
@@ -199,8 +202,8 @@ __global__ __launch_bounds__(BLOCK, 1) void mla_attention_kernel(
     float alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int rr = tile * BQ + wave * 16 + lhi * 4 + r;
-      const int qpos = past + rr / H;  // mask depends on the token only
+      const int rr = tile * BQ + rg * 16 + lhi * 4 + r;
+      const int qpos = past + rr / H;
       float sv[NF];
       float mx = -INFINITY;
 #pragma unroll
@@ -232,67 +235,65 @@ __global__ __launch_bounds__(BLOCK, 1) void mla_attention_kernel(
       m_run[r] = m_new;
     }
 #pragma unroll
-    for (int nt = 0; nt < NT; ++nt)
+    for (int nt = 0; nt < NTW; ++nt)
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[nt][r] *= alpha[r];
 
-    // ---------- P -> LDS bounce -> A fragments ----------
+    // ---------- P -> LDS bounce -> A fragment ----------
     __hip_bfloat16 *pw = p_tile[wave];
 #pragma unroll
     for (int f = 0; f < NF; ++f)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        pw[(lhi * 4 + r) * VT_STRIDE + f * 16 + l16] =
+        pw[(lhi * 4 + r) * PT_STRIDE + f * 16 + l16] =
             __float2bfloat16(p_vals[f][r]);
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    mfma_bf8 pfrag[BKV / 32];
-#pragma unroll
-    for (int ks = 0; ks < BKV / 32; ++ks)
-      pfrag[ks] = *reinterpret_cast<const mfma_bf8 *>(
-          &pw[l16 * VT_STRIDE + ks * 32 + lhi * 8]);
+    mfma_bf8 pfrag = *reinterpret_cast<const mfma_bf8 *>(
+        &pw[l16 * PT_STRIDE + lhi * 8]);
 
-    // ---------- PV ----------
+    // ---------- PV: o += P[16x32] x V[32 x 256-half] ----------
+    // B-fragments gather from the K tile (V = first 512 dims of the
+    // row): lane l reads tokens lhi*8+j at dim vhalf*256 + nt*16 + l16
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int nt = 0; nt < NT; ++nt) {
+    for (int nt = 0; nt < NTW; ++nt) {
+      const int dim = vhalf * 256 + nt * 16 + l16;
+      mfma_bf8 vfrag;
+      __hip_bfloat16 *ve = reinterpret_cast<__hip_bfloat16 *>(&vfrag);
 #pragma unroll
-      for (int ks = 0; ks < BKV / 32; ++ks) {
-        mfma_bf8 vfrag = *reinterpret_cast<const mfma_bf8 *>(
-            &vt_tile[(nt * 16 + l16) * VT_STRIDE + ks * 32 + lhi * 8]);
-        o_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            pfrag[ks], vfrag, o_acc[nt], 0, 0, 0);
+      for (int j = 0; j < 8; ++j) {
+        const int vrow = lhi * 8 + j;
+        ve[j] = *reinterpret_cast<const __hip_bfloat16 *>(
+            ktile + vrow * ROW_B + kswz(vrow, dim * 2));
       }
+      o_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          pfrag, vfrag, o_acc[nt], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
-
-    __syncthreads();  // everyone done reading this tile
-    if (has_next) {
-      stage_write();
-      __syncthreads();
-    }
+    __syncthreads();  // everyone done with this ring slot
   }
 
   // ---------- epilogue ----------
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    const int rr = tile * BQ + wave * 16 + lhi * 4 + r;
+    const int rr = tile * BQ + rg * 16 + lhi * 4 + r;
     if (rr >= rows) continue;
     const float l = l_run[r];
+    const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
+    const long orow = row_base - tile * BQ + rr;
     if (SPLIT) {
-      const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
-      float *op =
-          partial_out + ((long)split * th_total + row_base - tile * BQ + rr) * DV;
+      float *op = partial_out +
+          ((long)split * th_total + orow) * DV + vhalf * 256;
 #pragma unroll
-      for (int nt = 0; nt < NT; ++nt)
+      for (int nt = 0; nt < NTW; ++nt)
         op[nt * 16 + l16] = o_acc[nt][r] * inv_l;
-      if (l16 == 0)
-        partial_lse[(long)split * th_total + row_base - tile * BQ + rr] =
+      if (l16 == 0 && vhalf == 0)
+        partial_lse[(long)split * th_total + orow] =
             (l > 0.f) ? m_run[r] + __logf(l) : -INFINITY;
     } else {
-      const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
-      __hip_bfloat16 *op = out + (row_base - tile * BQ + rr) * DV;
+      __hip_bfloat16 *op = out + orow * DV + vhalf * 256;
 #pragma unroll
-      for (int nt = 0; nt < NT; ++nt)
+      for (int nt = 0; nt < NTW; ++nt)
         op[nt * 16 + l16] = __float2bfloat16(o_acc[nt][r] * inv_l);
     }
   }
@@ -324,9 +325,9 @@ __global__ void mla_merge_kernel(__hip_bfloat16 *__restrict__ out,
 }
 
 // Scatter the per-token latent row [T, 1, DK] into the paged cache
-// (reference concat_and_cache_mla, cache_kernels.py:161 — here the
-// caller has already concatenated [c_kv | k_pe] so this is a single
-// vectorized row write; the v cache is a view of the first 512 dims).
+// (reference concat_and_cache_mla, cache_kernels.py:161 — the caller
+// has already concatenated [c_kv | k_pe]; the v cache is a view of the
+// first 512 dims so one write covers both).
 __global__ void cache_latent_kernel(const __hip_bfloat16 *__restrict__ k,
                                     __hip_bfloat16 *__restrict__ k_cache,
                                     const long *__restrict__ slot_mapping,
@@ -398,7 +399,7 @@ void mla_paged_attention(torch::Tensor out, torch::Tensor q,
                      (float)scale, splits, th_total);
   HIP_CHECK_KERNEL();
   hipLaunchKernelGGL((mla_merge_kernel), dim3((unsigned)th_total),
-                     dim3(BLOCK), 0, stream,
+                     dim3(256), 0, stream,
                      (__hip_bfloat16 *)out.data_ptr(),
                      partial.data_ptr<float>(), lse.data_ptr<float>(),
                      splits, th_total);
