@@ -146,6 +146,41 @@ class GatedDeltaNet(nn.Module):
         out_core = torch.empty(T, self.tp_v, self.head_v_dim,
                                dtype=hidden.dtype, device=hidden.device)
         qsl = fctx.host_qsl()
+        B = len(fctx.ssm_slots)
+
+        # ---- batched decode fast path (gfx950 kernels): every seq is a
+        # single-token step with carried state -> one conv-update launch
+        # + one fused recurrent launch for the whole batch (the per-seq
+        # loop below costs ~8 launches PER SEQUENCE)
+        from gllm_amd import ops as _ops
+        if (hidden.is_cuda and T == B and self.head_k_dim == 128
+                and self.conv1d_weight.shape[1] == 4
+                and all(bool(h) for h in fctx.ssm_has_init)
+                and _ops.has_kernels()):
+            slots_t = torch.as_tensor(
+                [int(s) for s in fctx.ssm_slots], dtype=torch.long,
+                device=hidden.device)
+            conv_out = _ops.gdn_conv_update(
+                mixed.contiguous(), self.conv1d_weight, conv_states,
+                slots_t)
+            qd, kd, vd = conv_out.split([k_tp, k_tp, v_tp], dim=-1)
+            G = self.tp_v // self.tp_k_heads
+            qn = gdn_ref.l2norm(
+                qd.view(T, self.tp_k_heads, self.head_k_dim).float()) \
+                * self.scale
+            kn = gdn_ref.l2norm(
+                kd.view(T, self.tp_k_heads, self.head_k_dim).float())
+            o = _ops.gdn_decode(
+                qn.repeat_interleave(G, dim=1),
+                kn.repeat_interleave(G, dim=1),
+                vd.view(T, self.tp_v, self.head_v_dim).float(),
+                g_all, beta_all, ssm_states, slots_t)
+            gated = _ops.rmsnorm_gated(
+                o.reshape(T * self.tp_v, self.head_v_dim),
+                z.reshape(T * self.tp_v, self.head_v_dim).contiguous(),
+                self.norm_weight, self.eps)
+            return self.out_proj(gated.reshape(T, -1))
+
         for i in range(len(fctx.ssm_slots)):
             s, e = qsl[i], qsl[i + 1]
             slot = int(fctx.ssm_slots[i])

@@ -267,6 +267,43 @@ def fused_moe_fp8(x: torch.Tensor, w13: torch.Tensor, w13_scale,
     return out
 
 
+# --------------------------------------------------------------- gdn
+def gdn_conv_update(x: torch.Tensor, weight: torch.Tensor,
+                    conv_state: torch.Tensor,
+                    slots: torch.Tensor) -> torch.Tensor:
+    """Batched causal-conv1d decode step: x [B, C] one token per seq,
+    conv_state pool rows indexed by ``slots`` roll in place."""
+    out = torch.empty_like(x)
+    _gpu_kernels().gdn_conv_update(out, x, weight, conv_state, slots)
+    return out
+
+
+def gdn_decode(qn: torch.Tensor, kn: torch.Tensor, v: torch.Tensor,
+               g: torch.Tensor, beta: torch.Tensor, state: torch.Tensor,
+               slots: torch.Tensor) -> torch.Tensor:
+    """Batched fused recurrent gated-delta-rule step (one token per
+    seq). qn/kn [B, Hv, 128] fp32 (normalized, q pre-scaled), v [B, Hv,
+    Dv] fp32; state pool [slots, Hv, Dv, 128] fp32 updated in place."""
+    B, Hv, Dv = v.shape
+    o = torch.empty(B, Hv, Dv, dtype=torch.bfloat16, device=v.device)
+    _gpu_kernels().gdn_decode(o, qn.contiguous(), kn.contiguous(),
+                              v.contiguous(), g.contiguous(),
+                              beta.contiguous(), state, slots)
+    return o
+
+
+def rmsnorm_gated(x: torch.Tensor, z: torch.Tensor, weight: torch.Tensor,
+                  eps: float) -> torch.Tensor:
+    """out = rmsnorm(x) * w * silu(z) (fused, GPU)."""
+    if not x.is_cuda:
+        from gllm_amd.ops import gdn_ref
+        return gdn_ref.rmsnorm_gated(x, z, weight, eps)
+    out = torch.empty_like(x)
+    _gpu_kernels().rmsnorm_gated(out, x.contiguous(), z.contiguous(),
+                                 weight, eps)
+    return out
+
+
 # --------------------------------------------------------------- sampling
 def topk_topp_filter(probs: torch.Tensor, top_ks: torch.Tensor,
                      top_ps: torch.Tensor,

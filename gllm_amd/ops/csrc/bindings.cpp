@@ -54,6 +54,14 @@ void moe_gemm_fp8(torch::Tensor C, torch::Tensor A, torch::Tensor As,
                   torch::Tensor n_post_pad,
                   c10::optional<torch::Tensor> topk_weights, long n_pairs,
                   long topk, long block_m, bool scatter);
+void gdn_conv_update(torch::Tensor out, torch::Tensor x,
+                     torch::Tensor weight, torch::Tensor conv_state,
+                     torch::Tensor slots);
+void gdn_decode(torch::Tensor o, torch::Tensor qn, torch::Tensor kn,
+                torch::Tensor v, torch::Tensor g, torch::Tensor beta,
+                torch::Tensor state, torch::Tensor slots);
+void rmsnorm_gated(torch::Tensor out, torch::Tensor x, torch::Tensor z,
+                   torch::Tensor w, double eps);
 void apply_repetition_penalty(torch::Tensor logits, torch::Tensor pool,
                               torch::Tensor slots,
                               torch::Tensor penalties);
@@ -96,6 +104,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "block-scale fp8 weight-streaming GEMM (decode)");
   m.def("moe_gemm_fp8", &moe_gemm_fp8,
         "block-scale fp8 grouped MoE GEMM");
+  m.def("gdn_conv_update", &gdn_conv_update,
+        "batched causal-conv1d decode step w/ state roll");
+  m.def("gdn_decode", &gdn_decode,
+        "batched fused recurrent gated-delta-rule decode step");
+  m.def("rmsnorm_gated", &rmsnorm_gated, "rmsnorm(x)*w*silu(z)");
   m.def("apply_repetition_penalty", &apply_repetition_penalty,
         "scaling penalty vs persistent seen-token mask pool");
   m.def("topk_topp_filter", &topk_topp_filter,

@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "attention_mla.hip"),
         os.path.join(CSRC, "moe.hip"),
         os.path.join(CSRC, "fp8.hip"),
+        os.path.join(CSRC, "gdn.hip"),
         os.path.join(CSRC, "skinny_gemm.hip"),
         os.path.join(CSRC, "sampling.hip"),
         os.path.join(CSRC, "custom_ar.hip"),

@@ -452,3 +452,66 @@ def test_fused_moe_fp8(kernels):
                    weights, ids)
     # activation quant noise on top of weight quant: loose tolerance
     assert_close_bf16(out, ref, atol=8e-2, rtol=8e-2, frac=5e-3)
+
+
+# ------------------------------------------------------------ gdn
+def test_gdn_decode_kernels(kernels):
+    """Batched conv-update + fused recurrent step vs the sequential
+    torch oracle (gdn_ref), including state roll parity."""
+    from gllm_amd import ops
+    from gllm_amd.ops import gdn_ref
+    torch.manual_seed(4)
+    B, Hk, Hv, Dk, Dv, K = 5, 2, 4, 128, 128, 4
+    C = 2 * Hk * Dk + Hv * Dv
+    slots = torch.tensor([3, 0, 7, 1, 5], dtype=torch.long, device="cuda")
+    n_slots = 8
+    x = torch.randn(B, C, dtype=torch.bfloat16, device="cuda") / 2
+    w = torch.randn(C, K, dtype=torch.bfloat16, device="cuda") / 2
+    conv_pool = torch.randn(n_slots, C, K - 1, dtype=torch.bfloat16,
+                            device="cuda") / 2
+    conv_ref = conv_pool.clone()
+    out = ops.gdn_conv_update(x, w, conv_pool, slots)
+    for i in range(B):
+        ref = gdn_ref.causal_conv1d_update(
+            x[i].cpu(), w.cpu(), conv_ref[slots[i]].cpu().clone())
+        st = conv_ref[slots[i]].cpu()
+        ctx = torch.cat([st.float(), x[i].cpu().float().unsqueeze(1)], 1)
+        conv_ref[slots[i]] = ctx[:, 1:].to(torch.bfloat16).cuda()
+        assert_close_bf16(out[i], ref.float(), atol=3e-2, rtol=3e-2)
+    assert torch.equal(conv_pool, conv_ref)
+
+    # recurrent step
+    state_pool = torch.randn(n_slots, Hv, Dv, Dk, device="cuda") / 8
+    state_ref = state_pool.clone()
+    q = torch.randn(B, Hk, Dk, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(B, Hk, Dk, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(B, Hv, Dv, dtype=torch.bfloat16, device="cuda")
+    g = -torch.rand(B, Hv, device="cuda") * 0.1
+    beta = torch.rand(B, Hv, device="cuda")
+    scale = Dk ** -0.5
+    G = Hv // Hk
+    qn = gdn_ref.l2norm(q.float()) * scale
+    kn = gdn_ref.l2norm(k.float())
+    o = ops.gdn_decode(qn.repeat_interleave(G, 1),
+                       kn.repeat_interleave(G, 1), v.float(), g, beta,
+                       state_pool, slots)
+    for i in range(B):
+        st = state_ref[slots[i]].cpu()
+        oref = gdn_ref.gated_delta_rule(
+            q[i:i + 1].cpu(), k[i:i + 1].cpu(), v[i:i + 1].cpu(),
+            g[i:i + 1].cpu(), beta[i:i + 1].cpu(), scale, st)
+        state_ref[slots[i]] = st.cuda()
+        assert_close_bf16(o[i], oref[0].float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(state_pool, state_ref, atol=1e-4, rtol=1e-4)
+
+
+def test_rmsnorm_gated(kernels):
+    from gllm_amd import ops
+    from gllm_amd.ops import gdn_ref
+    torch.manual_seed(6)
+    x = torch.randn(300, 128, dtype=torch.bfloat16, device="cuda")
+    z = torch.randn(300, 128, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(128, dtype=torch.bfloat16, device="cuda")
+    out = ops.rmsnorm_gated(x, z, w, 1e-6)
+    ref = gdn_ref.rmsnorm_gated(x.cpu(), z.cpu(), w.cpu(), 1e-6)
+    assert_close_bf16(out, ref.float())
