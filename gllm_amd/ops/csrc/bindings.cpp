@@ -69,6 +69,9 @@ void topk_topp_filter(torch::Tensor probs, torch::Tensor top_ks,
                       torch::Tensor top_ps,
                       c10::optional<torch::Tensor> min_ps);
 std::pair<int64_t, py::bytes> car_alloc(int64_t data_bytes);
+std::pair<int64_t, py::bytes> car_alloc_v2(int64_t max_bytes);
+void car_all_reduce_v2(torch::Tensor inout, std::vector<int64_t> ptrs,
+                       int64_t rank, int64_t world, int64_t max_bytes);
 int64_t car_open(py::bytes handle_bytes);
 void car_close(int64_t ptr);
 void car_free(int64_t ptr);
@@ -114,6 +117,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_topp_filter", &topk_topp_filter,
         "fused sorting-free top-k/top-p/min-p filter + renormalize");
   m.def("car_alloc", &car_alloc, "alloc hipIpc-shared AR buffer");
+  m.def("car_alloc_v2", &car_alloc_v2,
+        "alloc v2 AR buffer (device epoch + out slices)");
+  m.def("car_all_reduce_v2", &car_all_reduce_v2,
+        "two-shot/one-shot xGMI AR, device-epoch (graph-capturable)");
   m.def("car_open", &car_open, "map a peer's AR buffer");
   m.def("car_close", &car_close);
   m.def("car_free", &car_free);

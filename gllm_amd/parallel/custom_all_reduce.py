@@ -5,12 +5,15 @@ two-shot AR + cuda_wrapper.py ctypes IPC — here the hipIpc calls live
 in the extension, ops/csrc/custom_ar.hip, and the kernel reads peers
 straight over point-to-point xGMI links).
 
-Round-1 status: one-shot AR, eager-mode only (no graph capture),
-gated OFF by default — enable with GLLM_CUSTOM_AR=1. Validated by a
-2-process-on-1-GPU numerics test (tests/test_custom_ar_gpu.py); the
-multi-GPU xGMI latency win needs the 8-GPU tier (ROADMAP.md item 5).
-Eligibility mirrors the reference's should_custom_ar: bf16/fp32,
-payload <= max_bytes, world size 2..8, not during graph capture.
+Round-2 (v2): one-shot for small payloads + two-shot reduce-scatter/
+gather above ~256 KB, with the barrier epoch kept in a DEVICE cell so
+the collective is hipGraph-capturable without the reference's
+graph-buffer registration handshake (custom_all_reduce.py:266-391) —
+every rank issues the same collective sequence, so the cells stay in
+lock-step across replays and eager calls alike. ON by default for
+TP 2..8 (disable with GLLM_CUSTOM_AR=0); init failure (no IPC) falls
+back to RCCL cleanly. Eligibility mirrors should_custom_ar: bf16/fp32,
+payload <= max_bytes.
 """
 
 import os
@@ -32,7 +35,7 @@ class CustomAllReduce:
         self.rank = rank
         self.world = world
         self.max_bytes = max_bytes
-        self.my_ptr, handle = K.car_alloc(max_bytes)
+        self.my_ptr, handle = K.car_alloc_v2(max_bytes)
         # exchange 64-byte hipIpc handles over the group (byte tensors,
         # reference custom_all_reduce.py:57-78)
         h = torch.zeros(len(handle), dtype=torch.uint8)
@@ -57,16 +60,12 @@ class CustomAllReduce:
             return False
         if t.numel() * t.element_size() > self.max_bytes:
             return False
-        if torch.cuda.is_current_stream_capturing():
-            return False
         return True
 
     def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
         t = t.contiguous()
-        # the kernel issues barriers at epoch+1 and epoch+2
-        self.K.car_all_reduce(t, self.ptrs, self.rank, self.world,
-                              self.epoch + 1)
-        self.epoch += 2
+        self.K.car_all_reduce_v2(t, self.ptrs, self.rank, self.world,
+                                 self.max_bytes)
         return t
 
     def close(self):
@@ -80,7 +79,7 @@ def init_custom_all_reduce() -> Optional[CustomAllReduce]:
     init_distributed). No-op unless GLLM_CUSTOM_AR=1."""
     global _INSTANCE, _INIT_TRIED
     _INIT_TRIED = True
-    if os.environ.get("GLLM_CUSTOM_AR", "0") != "1":
+    if os.environ.get("GLLM_CUSTOM_AR", "1") == "0":
         return None
     if not torch.cuda.is_available():
         return None

@@ -29,7 +29,11 @@ def _ar_worker(rank, port, q):
     for trial, (n, dtype) in enumerate([(1024, torch.float32),
                                         (8192, torch.bfloat16),
                                         (333, torch.float32),
-                                        (65536, torch.bfloat16)]):
+                                        (65536, torch.bfloat16),
+                                        # two-shot sizes (>= 256 KB)
+                                        (1 << 18, torch.bfloat16),
+                                        (1 << 20, torch.bfloat16),
+                                        (1 << 18, torch.float32)]):
         g = torch.Generator(device="cpu").manual_seed(100 + trial)
         inputs = [torch.randn(n, generator=g).to(dtype) for _ in range(2)]
         ref = (inputs[0].float() + inputs[1].float()).to(dtype)
