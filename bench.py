@@ -102,8 +102,65 @@ DEEPSEEK_V3 = {  # BASELINE config: DeepSeek-V3 PP=8 + EP, absorbed MLA
     "eos_token_id": 1,
 }
 
+DEEPSEEK_V2_LITE = {  # real DeepSeek-V2-Lite dims: 1-GPU MLA+MoE bench
+    "architectures": ["DeepseekV2ForCausalLM"],
+    "model_type": "deepseek_v2",
+    "hidden_size": 2048,
+    "intermediate_size": 10944,
+    "moe_intermediate_size": 1408,
+    "num_hidden_layers": 27,
+    "first_k_dense_replace": 1,
+    "num_attention_heads": 16,
+    "num_key_value_heads": 16,
+    "n_routed_experts": 64,
+    "n_shared_experts": 2,
+    "num_experts_per_tok": 6,
+    "n_group": 1,
+    "topk_group": 1,
+    "routed_scaling_factor": 1.0,
+    "scoring_func": "softmax",
+    "norm_topk_prob": False,
+    "q_lora_rank": None,
+    "kv_lora_rank": 512,
+    "qk_nope_head_dim": 128,
+    "qk_rope_head_dim": 64,
+    "v_head_dim": 128,
+    "vocab_size": 102400,
+    "max_position_embeddings": 163840,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 10000.0,
+    "tie_word_embeddings": False,
+    "eos_token_id": 1,
+}
+
+QWEN3_NEXT_9B = {  # hybrid GDN + full attention, 1-GPU scale
+    "architectures": ["Qwen3NextForCausalLM"],
+    "model_type": "qwen3_next",
+    "hidden_size": 2048,
+    "intermediate_size": 5504,
+    "num_hidden_layers": 24,
+    "num_attention_heads": 16,
+    "num_key_value_heads": 2,
+    "head_dim": 128,
+    "linear_num_value_heads": 16,
+    "linear_num_key_heads": 8,
+    "linear_key_head_dim": 128,
+    "linear_value_head_dim": 128,
+    "linear_conv_kernel_dim": 4,
+    "full_attention_interval": 4,
+    "vocab_size": 151936,
+    "max_position_embeddings": 32768,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 1000000.0,
+    "partial_rotary_factor": 0.25,
+    "tie_word_embeddings": False,
+    "eos_token_id": 2,
+}
+
 MODELS = {
     "qwen2.5-32b": ("Qwen2.5-32B", QWEN25_32B),
+    "deepseek-v2-lite": ("DeepSeek-V2-Lite", DEEPSEEK_V2_LITE),
+    "qwen3-next-9b": ("Qwen3-Next-9B-hybrid", QWEN3_NEXT_9B),
     "debug": ("debug-0.2B", SMALL_DEBUG),
     "mixtral-8x7b": ("Mixtral-8x7B", MIXTRAL_8X7B),
     "mixtral-debug": ("mixtral-debug", MIXTRAL_DEBUG),
