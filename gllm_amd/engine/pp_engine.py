@@ -51,6 +51,14 @@ class PPEngine:
         self.device = config.device
         self.hidden = self.runner.hf_config.hidden_size
         self.dtype = config.torch_dtype()
+        # PP sends ride a dedicated HIP stream so the next micro-batch's
+        # compute never queues behind a send that is waiting for the
+        # receiver (reference dist_utils.py:8-22 uses isend; on RCCL a
+        # side stream gives the same overlap with plain send)
+        self._comm_stream = torch.cuda.Stream() \
+            if config.device.startswith("cuda") else None
+        from collections import deque
+        self._send_bufs = deque(maxlen=max(2, self.pp_size))
         # dedicated communicator for token broadcasts — one per DP
         # replica (a replica's pipeline spans all pp stages x tp ranks
         # of its dp index; tokens differ per replica, so the broadcast
@@ -90,26 +98,41 @@ class PPEngine:
             return
         if self.is_first:
             hidden, residual, _ = self.runner.step_first_stage(batch)
-            dist.send(hidden.contiguous(), dst=self.P.get_next_pp_rank())
-            dist.send(residual.contiguous(), dst=self.P.get_next_pp_rank())
+            self._send_pp(hidden, residual)
             self.inflight.append((batch, None))
         else:
             T = batch.num_tokens
-            shape = (T, self.hidden)
-            hidden = torch.empty(shape, dtype=self.dtype, device=self.device)
-            residual = torch.empty(shape, dtype=self.dtype,
-                                   device=self.device)
-            src = self.P.get_prev_pp_rank()
-            dist.recv(hidden, src=src)
-            dist.recv(residual, src=src)
+            hidden, residual = self._recv_pp(T)
             out = self.runner.step_mid_stage(batch, hidden, residual)
             if self.is_last:
                 self.inflight.append((batch, out))
             else:
                 h2, r2, _ = out
-                dist.send(h2.contiguous(), dst=self.P.get_next_pp_rank())
-                dist.send(r2.contiguous(), dst=self.P.get_next_pp_rank())
+                self._send_pp(h2, r2)
                 self.inflight.append((batch, None))
+
+    def _send_pp(self, hidden: torch.Tensor, residual: torch.Tensor):
+        """ONE fused [2T, H] message per hop (the reference sends hidden
+        and residual as two p2p ops), launched on the comm stream."""
+        msg = torch.cat([hidden, residual], dim=0).contiguous()
+        dst = self.P.get_next_pp_rank()
+        if self._comm_stream is not None:
+            evt = torch.cuda.Event()
+            evt.record()  # msg materialized on the compute stream
+            self._comm_stream.wait_event(evt)
+            with torch.cuda.stream(self._comm_stream):
+                dist.send(msg, dst=dst)
+            # keep the buffer alive until the send drains (bounded by
+            # <= pp_size micro-batches in flight)
+            self._send_bufs.append(msg)
+        else:
+            dist.send(msg, dst=dst)
+
+    def _recv_pp(self, T: int):
+        msg = torch.empty((2 * T, self.hidden), dtype=self.dtype,
+                          device=self.device)
+        dist.recv(msg, src=self.P.get_prev_pp_rank())
+        return msg[:T], msg[T:]
 
     @staticmethod
     def _stash_logprobs(batch, out) -> None:
