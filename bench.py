@@ -192,12 +192,12 @@ def main():
                          "each stage")
     ap.add_argument("--page-size", type=int, default=16)
     ap.add_argument("--schedule", type=str, default="token_throttling")
-    ap.add_argument("--qps", type=float, default=0.0,
+    ap.add_argument("--qps", type=float, default=32.0,
                     help=">0: pace request arrivals at this rate for the "
                          "TTFT measurement instead of a burst")
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--no-overlap", action="store_true")
-    ap.add_argument("--max-graph-bs", type=int, default=64)
+    ap.add_argument("--max-graph-bs", type=int, default=512)
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
