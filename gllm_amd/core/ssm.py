@@ -42,7 +42,11 @@ class SSMPool:
         self.ssm_state = torch.zeros(
             L, num_slots, spec.num_v_heads, spec.head_v_dim,
             spec.head_k_dim, dtype=torch.float32, device=device)
-        self.alloc = IDAllocator(num_slots)
+        # the LAST slot is the hipGraph dummy/scratch slot (padding rows
+        # of a captured decode bucket write their garbage there); the
+        # allocator never hands it out
+        self.dummy_slot = num_slots - 1
+        self.alloc = IDAllocator(num_slots - 1)
 
     def ensure(self, seq: Sequence) -> int:
         if seq.ssm_slot >= 0:

@@ -153,13 +153,18 @@ class GatedDeltaNet(nn.Module):
         # + one fused recurrent launch for the whole batch (the per-seq
         # loop below costs ~8 launches PER SEQUENCE)
         from gllm_amd import ops as _ops
-        if (hidden.is_cuda and T == B and self.head_k_dim == 128
+        graph_path = fctx.ssm_slots_dev is not None
+        if (hidden.is_cuda and self.head_k_dim == 128
                 and self.conv1d_weight.shape[1] == 4
-                and all(bool(h) for h in fctx.ssm_has_init)
-                and _ops.has_kernels()):
-            slots_t = torch.as_tensor(
-                [int(s) for s in fctx.ssm_slots], dtype=torch.long,
-                device=hidden.device)
+                and _ops.has_kernels()
+                and (graph_path or
+                     (T == B and all(bool(h) for h in fctx.ssm_has_init)))):
+            if graph_path:
+                slots_t = fctx.ssm_slots_dev[:T]
+            else:
+                slots_t = torch.as_tensor(
+                    [int(s) for s in fctx.ssm_slots], dtype=torch.long,
+                    device=hidden.device)
             conv_out = _ops.gdn_conv_update(
                 mixed.contiguous(), self.conv1d_weight, conv_states,
                 slots_t)

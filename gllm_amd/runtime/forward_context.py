@@ -36,6 +36,9 @@ class ForwardContext:
     ssm_pool: Optional[object] = None
     ssm_slots: Optional[List[int]] = None
     ssm_has_init: Optional[List[bool]] = None
+    # graph-replay path: persistent device slot buffer (all rows are
+    # initialized decode steps; padding rows point at the dummy slot)
+    ssm_slots_dev: Optional[object] = None
     # multimodal: rows of this batch whose embeddings come from the
     # vision tower (replaced after embed_tokens on the first stage)
     mm_rows: Optional[torch.Tensor] = None     # [N] long

@@ -53,9 +53,20 @@ class GraphRunner:
         self.pin_seq_lens = torch.zeros(B, dtype=torch.int32).pin_memory()
         self.pin_bt = torch.zeros((B, self.pages_cap),
                                   dtype=torch.int32).pin_memory()
+        # hybrid (GDN/SSM) models: persistent device slot buffer; all
+        # captured rows are initialized decode steps, padding rows point
+        # at the pool's dummy scratch slot
+        self.ssm_slots_dev = None
+        self.pin_ssm = None
+        if runner.ssm_pool is not None:
+            dummy = runner.ssm_pool.dummy_slot
+            self.ssm_slots_dev = torch.full((B,), dummy, dtype=torch.long,
+                                            device=dev)
+            self.pin_ssm = torch.full((B,), dummy,
+                                      dtype=torch.long).pin_memory()
 
     def _fctx_for(self, bs: int) -> ForwardContext:
-        return ForwardContext(
+        fctx = ForwardContext(
             num_tokens=bs,
             positions=self.positions[:bs],
             slot_mapping=self.slots[:bs],
@@ -65,6 +76,10 @@ class GraphRunner:
             max_query_len=1, max_seq_len=1,
             k_caches=self.runner.k_caches,
             v_caches=self.runner.v_caches)
+        if self.ssm_slots_dev is not None:
+            fctx.ssm_pool = self.runner.ssm_pool
+            fctx.ssm_slots_dev = self.ssm_slots_dev[:bs]
+        return fctx
 
     @torch.no_grad()
     def capture_all(self):
@@ -106,8 +121,16 @@ class GraphRunner:
         items = batch.items
         if len(items) > self.max_bs:
             return False
-        return all(it.num_tokens == 1 and it.start >= it.seq.prompt_len
-                   for it in items)
+        if not all(it.num_tokens == 1 and it.start >= it.seq.prompt_len
+                   for it in items):
+            return False
+        if self.ssm_slots_dev is not None:
+            # a pending recurrent-state restore (prefix hit on a fresh
+            # slot) needs the eager path's host-side copy
+            if not all(it.seq.ssm_slot >= 0 and it.seq.ssm_state_ready
+                       for it in items):
+                return False
+        return True
 
     def _bucket(self, n: int) -> int:
         for b in self.buckets:
@@ -144,6 +167,15 @@ class GraphRunner:
             bt[i, 0] = self.dummy_page
             slots[i] = dummy_base + i % ps
             seq_lens[i] = 1
+        if self.pin_ssm is not None:
+            sl = self.pin_ssm.numpy()
+            dummy = self.runner.ssm_pool.dummy_slot
+            for i, it in enumerate(items):
+                sl[i] = it.seq.ssm_slot
+            for i in range(B, bs):
+                sl[i] = dummy
+            self.ssm_slots_dev[:bs].copy_(self.pin_ssm[:bs],
+                                          non_blocking=True)
         # H2D into the captured buffers
         self.in_ids[:bs].copy_(self.pin_ids[:bs], non_blocking=True)
         self.positions[:bs].copy_(self.pin_pos[:bs], non_blocking=True)

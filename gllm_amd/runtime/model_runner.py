@@ -47,8 +47,15 @@ class ModelRunner:
         if getattr(self.model, "ssm_spec", None) is not None:
             # prefix caching stays ON: page-boundary recurrent-state
             # snapshots restore on hits (core/ssm.py + the hit_filter
-            # wired below); graphs are the round-2 item
-            cfg.use_graph = False
+            # wired below). Graphs capture the batched GDN decode path
+            # (device slot buffer + dummy scratch slot); batches with a
+            # pending state restore fall back to eager (graph_runner
+            # can_replay). Requires the gfx950 GDN kernels.
+            from gllm_amd import ops
+            spec = self.model.ssm_spec
+            if not (cfg.device.startswith("cuda") and ops.has_kernels()
+                    and spec.head_k_dim == 128 and spec.conv_kernel == 4):
+                cfg.use_graph = False
         self.uses_mrope = bool(getattr(self.model, "uses_mrope", False))
         if self.uses_mrope:
             # [3,B] graph position buffers are round 2; prefix caching
