@@ -48,30 +48,58 @@ class OverlapEngine(PPEngine):
         slot = self._slot
         self._slot = (self._slot + 1) % self.ring_slots
         out = self.runner.step_first_stage(batch)   # SamplerOutput, async
-        if out.logprobs is not None:
-            # a logprob-requesting batch syncs here (.tolist()) — the
-            # overlap win is traded for the feature on those requests
-            from gllm_amd.engine.pp_engine import PPEngine
-            PPEngine._stash_logprobs(batch, out)
         B = len(batch.items)
         ring = self.runner.token_ring
         ring[slot, :B].copy_(out.next_tokens)
         phs = [-(slot * self.maxd + i) - 1 for i in range(B)]
         records = self.scheduler.process_output_deferred(batch, phs)
+        lp_host = None
         if self.is_cuda:
             pinned = self._pinned[slot]
             pinned[:B].copy_(out.next_tokens, non_blocking=True)
+            if out.logprobs is not None:
+                # logprobs ride the same event-gated async D2H as the
+                # tokens (pinned staging per ring slot) — no sync in
+                # the launch path (r1 traded the overlap away here)
+                lp_host = self._lp_stage(slot, out)
             ev = torch.cuda.Event()
             ev.record()
         else:
             pinned = out.next_tokens
             ev = None
-        self.pending.append((batch, records, ev, pinned, B))
+            if out.logprobs is not None:
+                lp_host = (out.logprobs, out.topk_logprobs,
+                           out.topk_token_ids)
+        self.pending.append((batch, records, ev, pinned, B, lp_host))
+
+    def _lp_stage(self, slot, out):
+        """Async-copy the logprob tensors into per-slot pinned staging
+        (grown when a request raises top-k)."""
+        if not hasattr(self, "_lp_pinned"):
+            self._lp_pinned = {}
+        B, K = out.topk_logprobs.shape
+        cur = self._lp_pinned.get(slot)
+        if cur is None or cur[1].shape[1] < K:
+            cur = (torch.zeros(self.maxd).pin_memory(),
+                   torch.zeros(self.maxd, K).pin_memory(),
+                   torch.zeros(self.maxd, K, dtype=torch.long)
+                   .pin_memory())
+            self._lp_pinned[slot] = cur
+        lp, tv, ti = cur
+        lp[:B].copy_(out.logprobs.float(), non_blocking=True)
+        tv[:B, :K].copy_(out.topk_logprobs.float(), non_blocking=True)
+        ti[:B, :K].copy_(out.topk_token_ids, non_blocking=True)
+        return (lp[:B], tv[:B, :K], ti[:B, :K])
 
     def _collect_one(self) -> int:
-        batch, records, ev, pinned, B = self.pending.popleft()
+        batch, records, ev, pinned, B, lp_host = self.pending.popleft()
         if ev is not None:
             ev.synchronize()
+        if lp_host is not None:
+            from types import SimpleNamespace
+            self._stash_logprobs(batch, SimpleNamespace(
+                logprobs=lp_host[0], topk_logprobs=lp_host[1],
+                topk_token_ids=lp_host[2]))
         tokens = pinned[:B].tolist()
         finished = self.scheduler.finalize_output(batch, tokens, records)
         self._finished_since.extend(finished)
