@@ -145,8 +145,8 @@ class GatedDeltaNet(nn.Module):
 
         out_core = torch.empty(T, self.tp_v, self.head_v_dim,
                                dtype=hidden.dtype, device=hidden.device)
-        qsl = fctx.host_qsl()
-        B = len(fctx.ssm_slots)
+        # under graph capture/replay only the device slot buffer exists
+        B = len(fctx.ssm_slots) if fctx.ssm_slots is not None else T
 
         # ---- batched decode fast path (gfx950 kernels): every seq is a
         # single-token step with carried state -> one conv-update launch
@@ -186,6 +186,7 @@ class GatedDeltaNet(nn.Module):
                 self.norm_weight, self.eps)
             return self.out_proj(gated.reshape(T, -1))
 
+        qsl = fctx.host_qsl()
         for i in range(len(fctx.ssm_slots)):
             s, e = qsl[i], qsl[i + 1]
             slot = int(fctx.ssm_slots[i])
