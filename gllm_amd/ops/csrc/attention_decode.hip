@@ -165,6 +165,8 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
     __syncthreads();
 
     // ---------- phase B2: PV accumulation (vectorized V reads) ----------
+    // 4 independent partial accumulators break the serial FMA chain
+    // (the r1 single-chain loop was dependent-latency bound)
 #pragma unroll
     for (int i = 0; i < ACC; ++i) {
       const int quad = tid + i * BLOCK;       // (h, d/4) index
@@ -174,7 +176,26 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
       const float al = s_alpha[h];
       floatx4 a = acc[i];
       a.x *= al; a.y *= al; a.z *= al; a.w *= al;
-      for (int t = 0; t < c_len; ++t) {
+      floatx4 p1 = floatx4{0.f, 0.f, 0.f, 0.f};
+      floatx4 p2 = floatx4{0.f, 0.f, 0.f, 0.f};
+      floatx4 p3 = floatx4{0.f, 0.f, 0.f, 0.f};
+      int t = 0;
+      for (; t + 4 <= c_len; t += 4) {
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const float p = s_scores[h][t + u];
+          const shortx4 v4 =
+              *reinterpret_cast<const shortx4 *>(&v_tile[t + u][d0]);
+          const __hip_bfloat16 *ve =
+              reinterpret_cast<const __hip_bfloat16 *>(&v4);
+          floatx4 &dst = (u == 0) ? a : (u == 1) ? p1 : (u == 2) ? p2 : p3;
+          dst.x += p * __bfloat162float(ve[0]);
+          dst.y += p * __bfloat162float(ve[1]);
+          dst.z += p * __bfloat162float(ve[2]);
+          dst.w += p * __bfloat162float(ve[3]);
+        }
+      }
+      for (; t < c_len; ++t) {
         const float p = s_scores[h][t];
         const shortx4 v4 =
             *reinterpret_cast<const shortx4 *>(&v_tile[t][d0]);
@@ -185,6 +206,10 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
         a.z += p * __bfloat162float(ve[2]);
         a.w += p * __bfloat162float(ve[3]);
       }
+      a.x += p1.x + p2.x + p3.x;
+      a.y += p1.y + p2.y + p3.y;
+      a.z += p1.z + p2.z + p3.z;
+      a.w += p1.w + p2.w + p3.w;
       acc[i] = a;
     }
     __syncthreads();
