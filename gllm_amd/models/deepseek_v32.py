@@ -10,16 +10,21 @@ attention. For any causal horizon <= index_topk the top-k selects every
 key, so DSA is exactly dense there — the correctness oracle
 (reference deepseek_v32.py:455-470).
 
-MI355X round-1 scope: the fp32 torch selection path over a paged
-index-K cache parallel to the KV pool ([pages, page_size,
-index_head_dim], runtime/model_runner.py sizes it from
-``model.index_head_dim``), exact on CPU; selection returns per-query
-TOKEN POSITIONS (-1 padded) that ops.mla_paged_attention masks with.
-Round 2 (ROADMAP.md): the gfx950 tile-static scorer (graph-safe fixed
-position tiles like the reference's _INDEX_SCORE_TILE), the fp8 e4m3
-index cache + MFMA scoring kernel, and sparse gather-attention over
-physical slots. The indexer's rope is NON-interleaved (neox) on the
-first qk_rope dims — a deliberate reference quirk we match.
+MI355X scope: torch selection over a paged index-K cache parallel to
+the KV pool ([pages, page_size, index_head_dim],
+runtime/model_runner.py sizes it from ``model.index_head_dim``);
+decode batches run ONE padded gather + batched score + batched topk
+(r2 — the r1 per-seq loop cost ~5 launches/seq/layer). Selection
+returns per-query TOKEN POSITIONS (-1 padded) that
+ops.mla_paged_attention masks with.
+
+Deliberate deviation from the reference: the index-K cache stays BF16,
+not the reference's fp8-e4m3 656/132-byte packed layouts
+(memory_manager.py:291-362). Those exist to stretch H800-class HBM;
+on 288 GB MI355X the index cache is ~1-2 GB at 128k context and the
+bf16 path is both simpler and more accurate. The indexer's rope is
+NON-interleaved (neox) on the first qk_rope dims — a reference quirk
+we match.
 """
 
 from typing import Optional
@@ -119,6 +124,33 @@ class DSAMLAAttention(MLAAttention):
         out = torch.full((T, topk), -1, dtype=torch.int32, device=dev)
         qsl = fctx.host_qsl()
         host_lens = fctx.host_seq_lens()
+        B = len(host_lens)
+
+        # ---- batched decode selection: one padded gather + one batched
+        # score + one topk for the whole batch (the per-seq loop costs
+        # ~5 launches per sequence per layer)
+        if T == B and B > 1:
+            max_s = max(host_lens)
+            n_pages = -(-max_s // page_sz)
+            pages = fctx.block_table[:, :n_pages].long()     # [B, P]
+            keys = cache[pages.reshape(-1)].reshape(
+                B, n_pages * page_sz, -1)[:, :max_s]         # [B, S, D]
+            # scores[b, s] = sum_h w[b,h] relu(scale * q[b,h,:]·k[b,s,:])
+            s = torch.bmm(idx_q.float(),
+                          keys.float().transpose(1, 2))      # [B, Hi, S]
+            s = F.relu(s * self.indexer.softmax_scale)
+            logits = torch.einsum("bhs,bh->bs", s, weights)  # [B, S]
+            pos_k = torch.arange(max_s, device=dev).unsqueeze(0)
+            lens_t = fctx.seq_lens.unsqueeze(1)
+            logits = logits.masked_fill(pos_k >= lens_t, float("-inf"))
+            k_sel = min(topk, max_s)
+            top = logits.topk(k_sel, dim=-1)
+            sel = top.indices.to(torch.int32)
+            sel = torch.where(torch.isinf(top.values),
+                              sel.new_full((), -1), sel)
+            out[:, :k_sel] = sel
+            return out
+
         for b in range(len(host_lens)):
             qs, qe = qsl[b], qsl[b + 1]
             q_len = qe - qs
