@@ -271,7 +271,9 @@ void launch_decode(torch::Tensor &out, const torch::Tensor &q,
   const int B = q.size(0);
   const int Hq = q.size(1);
   const int Hkv = k_cache.size(2);
-  constexpr int CHUNK = 128;
+  // CHUNK=64 -> ~18 KB LDS -> 8 workgroups/CU co-resident (PMC:
+  // waves park 66% at phase barriers; co-residency absorbs it)
+  constexpr int CHUNK = 64;
   // split heuristic: enough workgroups to fill 256 CUs x 8 XCDs
   static int splits_env = [] {
     const char *e = getenv("GLLM_DECODE_SPLITS");
