@@ -237,6 +237,225 @@ __global__ __launch_bounds__(BLOCK) void paged_decode_kernel(
   }
 }
 
+// ================================================================ v3
+// MFMA decode kernel for D=128 (v2 above stays for D=64). PMC showed
+// the VALU path issue-bound: phase-A dots + B2 accumulates cost ~1.5k
+// scalar FMA per thread per 128-token chunk, saturating the SIMD issue
+// ports at 25% "active" with 66% barrier parking. Here both GEMM
+// phases run on the matrix pipe: per 64-token chunk each wave computes
+// one S-tile (rows = the GQA group's q heads, cols = 16 tokens,
+// 4 k-steps over D) and a PV tile (rows = heads, 32 of the 128 v-dims,
+// 2 k-steps over tokens), ~8 MFMAs/wave/chunk instead of ~800 VALU.
+typedef __attribute__((ext_vector_type(8))) __bf16 dec_bf8;
+typedef __attribute__((ext_vector_type(4))) float dec_f4;
+
+template <int G>
+__global__ __launch_bounds__(BLOCK) void paged_decode_mfma_kernel(
+    float *__restrict__ partial_out,  // [S, B, Hq, 128]
+    float *__restrict__ partial_lse,  // [S, B, Hq]
+    const __hip_bfloat16 *__restrict__ q,        // [B, Hq, 128]
+    const __hip_bfloat16 *__restrict__ k_cache,  // [P, ps, Hkv, 128]
+    const __hip_bfloat16 *__restrict__ v_cache,
+    const int *__restrict__ block_table, const int *__restrict__ seq_lens,
+    int max_pages, int page_size, int num_kv_heads, float scale,
+    int num_splits, long q_stride, int window) {
+  constexpr int D = 128;
+  constexpr int CK = 64;            // kv tokens per chunk
+  const int b = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int split = blockIdx.z;
+  const int Hq = num_kv_heads * G;
+  const int seq_len = seq_lens[b];
+  const int w_begin = (window > 0) ? max(0, seq_len - window) : 0;
+  const int visible = seq_len - w_begin;
+  const int split_len = (visible + num_splits - 1) / num_splits;
+  const int t0 = w_begin + split * split_len;
+  const int t1 = min(t0 + split_len, seq_len);
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int l16 = lane & 15;
+  const int lhi = lane >> 4;
+
+  if (t0 >= t1) {
+    if (tid < G)
+      partial_lse[(long)split * gridDim.x * Hq + (long)b * Hq + kvh * G +
+                  tid] = -INFINITY;
+    return;
+  }
+
+  // LDS: K tile (swizzled rows), V tile (linear), scores, p image
+  __shared__ __hip_bfloat16 k_tile[CK][D];
+  __shared__ __hip_bfloat16 v_tile[CK][D];
+  __shared__ float s_scores[G][CK];
+  __shared__ __hip_bfloat16 p_img[16][CK + 8];
+  __shared__ float s_m[G], s_l[G], s_alpha[G];
+
+  // q A-fragments: row l16 = head (pad to 16), dims lhi*8 + kt*32
+  dec_bf8 qfrag[4];
+#pragma unroll
+  for (int kt = 0; kt < 4; ++kt) {
+    if (l16 < G) {
+      const __hip_bfloat16 *qp =
+          q + (long)b * q_stride + (kvh * G + l16) * D + kt * 32 + lhi * 8;
+      qfrag[kt] = *reinterpret_cast<const dec_bf8 *>(qp);
+    } else {
+      qfrag[kt] = dec_bf8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  float m_run = -INFINITY, l_run = 0.f;  // per (head row) via s_m/s_l
+  if (tid < G) { s_m[tid] = -INFINITY; s_l[tid] = 0.f; }
+  // o accumulators: wave owns dims [wave*32, wave*32+32): 2 n-tiles
+  dec_f4 o_acc[2] = {dec_f4{0, 0, 0, 0}, dec_f4{0, 0, 0, 0}};
+  (void)m_run; (void)l_run;
+  __syncthreads();
+
+  const int *bt = block_table + (long)b * max_pages;
+  constexpr int SWZ_MASK = 15;  // 16 chunks of 16B per 256-B row
+
+  for (int c0 = t0; c0 < t1; c0 += CK) {
+    const int c_len = min(CK, t1 - c0);
+    // ---- stage K (xor-swizzled rows) + V (linear): 4 x 16B/thread ----
+    {
+      const int cpr = D / 8;              // 16-B chunks per row (16)
+      for (int idx = tid; idx < CK * cpr; idx += BLOCK) {
+        const int row = idx / cpr;
+        const int c = idx % cpr;
+        const int tok = c0 + row;
+        shortx8 kv{0, 0, 0, 0, 0, 0, 0, 0}, vv{0, 0, 0, 0, 0, 0, 0, 0};
+        if (tok < t1) {
+          const long crow = ((long)bt[tok / page_size] * page_size +
+                             tok % page_size);
+          kv = *reinterpret_cast<const shortx8 *>(
+              k_cache + (crow * num_kv_heads + kvh) * D + c * 8);
+          vv = *reinterpret_cast<const shortx8 *>(
+              v_cache + (crow * num_kv_heads + kvh) * D + c * 8);
+        }
+        *reinterpret_cast<shortx8 *>(
+            reinterpret_cast<char *>(&k_tile[row][0]) +
+            ((c * 16) ^ ((row & SWZ_MASK) << 4) & 255)) = kv;
+        *reinterpret_cast<shortx8 *>(&v_tile[row][c * 8]) = vv;
+      }
+    }
+    __syncthreads();
+
+    // ---- S-tile: wave's 16 tokens x 16 head-rows, 4 k-steps ----
+    dec_f4 s_frag = dec_f4{0, 0, 0, 0};
+#pragma unroll
+    for (int kt = 0; kt < 4; ++kt) {
+      const int krow = wave * 16 + l16;
+      dec_bf8 bfrag = *reinterpret_cast<const dec_bf8 *>(
+          reinterpret_cast<char *>(&k_tile[krow][0]) +
+          (((kt * 32 + lhi * 8) * 2) ^ ((krow & SWZ_MASK) << 4) & 255));
+      s_frag = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kt], bfrag,
+                                                       s_frag, 0, 0, 0);
+    }
+    // rows lhi*4+r = heads; col l16 = token (wave's 16-token slice)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int h = lhi * 4 + r;
+      const int tok = wave * 16 + l16;
+      if (h < G && tok < CK) {
+        float sv = s_frag[r] * scale;
+        if (c0 + tok >= t1) sv = -INFINITY;
+        s_scores[h][tok] = sv;
+      }
+    }
+    __syncthreads();
+
+    // ---- online softmax per head over the chunk (64 wide) ----
+    {
+      constexpr int TPH0 = BLOCK / G;
+      constexpr int TPH = TPH0 >= 64 ? 64 : (TPH0 >= 32 ? 32 :
+                          (TPH0 >= 16 ? 16 : (TPH0 >= 8 ? 8 : 4)));
+      const int h = tid / TPH;
+      const int j = tid % TPH;
+      if (h < G) {
+        float mx = -INFINITY;
+        for (int t = j; t < c_len; t += TPH)
+          mx = fmaxf(mx, s_scores[h][t]);
+#pragma unroll
+        for (int off = TPH / 2; off > 0; off >>= 1)
+          mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+        const float m_old = s_m[h];
+        const float m_new = fmaxf(m_old, mx);
+        float psum = 0.f;
+        for (int t = j; t < CK; t += TPH) {
+          const float p = (t < c_len && s_scores[h][t] != -INFINITY &&
+                           m_new != -INFINITY)
+                              ? __expf(s_scores[h][t] - m_new) : 0.f;
+          p_img[h][t] = __float2bfloat16(p);
+          psum += p;
+        }
+#pragma unroll
+        for (int off = TPH / 2; off > 0; off >>= 1)
+          psum += __shfl_xor(psum, off, 64);
+        if (j == 0) {
+          const float alpha =
+              (m_old == -INFINITY || m_new == -INFINITY)
+                  ? 0.f : __expf(m_old - m_new);
+          s_alpha[h] = alpha;
+          s_l[h] = s_l[h] * alpha + psum;
+          s_m[h] = m_new;
+        }
+      }
+      // zero the padded head rows of the p image once per chunk
+      for (int h = G + tid; h < 16; h += BLOCK)
+        for (int t = 0; t < CK; ++t) p_img[h][t] = __float2bfloat16(0.f);
+    }
+    __syncthreads();
+
+    // ---- PV: o[heads, wave's 32 dims] += P[heads, CK] x V[CK, dims] ----
+    const float al = s_alpha[min(lhi * 4, G - 1)];
+    (void)al;
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      // rescale by this chunk's alpha first (rows = heads)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int h = lhi * 4 + r;
+        o_acc[nt][r] *= (h < G) ? s_alpha[h] : 0.f;
+      }
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        dec_bf8 pfrag = *reinterpret_cast<const dec_bf8 *>(
+            &p_img[l16][ks * 32 + lhi * 8]);
+        const int dim = wave * 32 + nt * 16 + l16;
+        dec_bf8 vfrag;
+        __hip_bfloat16 *ve = reinterpret_cast<__hip_bfloat16 *>(&vfrag);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ve[j] = v_tile[ks * 32 + lhi * 8 + j][dim];
+        o_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pfrag, vfrag, o_acc[nt], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: fp32 partial + lse ----
+#pragma unroll
+  for (int nt = 0; nt < 2; ++nt) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int h = lhi * 4 + r;
+      if (h >= G) continue;
+      const float l = s_l[h];
+      const float inv = (l > 0.f) ? 1.f / l : 0.f;
+      const int dim = wave * 32 + nt * 16 + l16;
+      partial_out[(((long)split * gridDim.x + b) * Hq + kvh * G + h) * D +
+                  dim] = o_acc[nt][r] * inv;
+    }
+  }
+  if (tid < G) {
+    const float l = s_l[tid];
+    partial_lse[(long)split * gridDim.x * Hq + (long)b * Hq + kvh * G +
+                tid] = (l > 0.f) ? s_m[tid] + __logf(l) : -INFINITY;
+  }
+}
+
 // Merge split partials: out[b,h,:] = sum_s w_s * partial[s,b,h,:]
 template <int D>
 __global__ void decode_merge_kernel(
@@ -271,9 +490,7 @@ void launch_decode(torch::Tensor &out, const torch::Tensor &q,
   const int B = q.size(0);
   const int Hq = q.size(1);
   const int Hkv = k_cache.size(2);
-  // CHUNK=64 -> ~18 KB LDS -> 8 workgroups/CU co-resident (PMC:
-  // waves park 66% at phase barriers; co-residency absorbs it)
-  constexpr int CHUNK = 64;
+  constexpr int CHUNK = 128;
   // split heuristic: enough workgroups to fill 256 CUs x 8 XCDs
   static int splits_env = [] {
     const char *e = getenv("GLLM_DECODE_SPLITS");
@@ -289,15 +506,32 @@ void launch_decode(torch::Tensor &out, const torch::Tensor &q,
   auto partial = torch::empty({splits, B, Hq, D}, opts);
   auto lse = torch::empty({splits, B, Hq}, opts);
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL((paged_decode_kernel<D, G, CHUNK>),
-                     dim3(B, Hkv, splits), dim3(BLOCK), 0, stream,
-                     partial.data_ptr<float>(), lse.data_ptr<float>(),
-                     (const __hip_bfloat16 *)q.data_ptr(),
-                     (const __hip_bfloat16 *)k_cache.data_ptr(),
-                     (const __hip_bfloat16 *)v_cache.data_ptr(),
-                     block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                     (int)block_table.size(1), (int)k_cache.size(1), Hkv,
-                     scale, splits, q_stride, window);
+  static int no_mfma = [] {
+    const char *e = getenv("GLLM_DECODE_NO_MFMA");
+    return e ? atoi(e) : 0;
+  }();
+  if (D == 128 && G <= 16 && !no_mfma) {
+    hipLaunchKernelGGL((paged_decode_mfma_kernel<G>),
+                       dim3(B, Hkv, splits), dim3(BLOCK), 0, stream,
+                       partial.data_ptr<float>(), lse.data_ptr<float>(),
+                       (const __hip_bfloat16 *)q.data_ptr(),
+                       (const __hip_bfloat16 *)k_cache.data_ptr(),
+                       (const __hip_bfloat16 *)v_cache.data_ptr(),
+                       block_table.data_ptr<int>(),
+                       seq_lens.data_ptr<int>(), (int)block_table.size(1),
+                       (int)k_cache.size(1), Hkv, scale, splits, q_stride,
+                       window);
+  } else {
+    hipLaunchKernelGGL((paged_decode_kernel<D, G, CHUNK>),
+                       dim3(B, Hkv, splits), dim3(BLOCK), 0, stream,
+                       partial.data_ptr<float>(), lse.data_ptr<float>(),
+                       (const __hip_bfloat16 *)q.data_ptr(),
+                       (const __hip_bfloat16 *)k_cache.data_ptr(),
+                       (const __hip_bfloat16 *)v_cache.data_ptr(),
+                       block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
+                       (int)block_table.size(1), (int)k_cache.size(1), Hkv,
+                       scale, splits, q_stride, window);
+  }
   HIP_CHECK_KERNEL();
   hipLaunchKernelGGL((decode_merge_kernel<D>), dim3(B, Hq), dim3(D), 0,
                      stream, (__hip_bfloat16 *)out.data_ptr(),
