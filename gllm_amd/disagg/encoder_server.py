@@ -94,6 +94,7 @@ class EncoderServer:
         self._srv.close()
 
     def _serve_conn(self, conn: socket.socket) -> None:
+        pool = None  # per-connection GPU-direct mapping
         try:
             while not self._stop.is_set():
                 job = recv_msg(conn)
@@ -107,15 +108,38 @@ class EncoderServer:
                     send_msg(conn, {"ok": True})
                     self.stop()
                     return
-                send_msg(conn, self._run(job))
+                from gllm_amd.disagg.protocol import PoolRegistration
+                if isinstance(job, PoolRegistration):
+                    try:
+                        from gllm_amd.disagg.gpu_plane import RemotePool
+                        assert torch.cuda.is_available()
+                        pool = RemotePool(job)
+                        send_msg(conn, {"pool": True})
+                        logger.info("GPU-direct plane mapped "
+                                    "(%d slots x %d elems)",
+                                    job.n_slots, job.slot_elems)
+                    except Exception as e:
+                        logger.warning("pool mapping failed (%s); "
+                                       "TCP payload fallback", e)
+                        send_msg(conn, {"pool": False})
+                    continue
+                send_msg(conn, self._run(job, pool))
         except OSError:
             pass
         finally:
+            if pool is not None:
+                pool.close()
             conn.close()
 
-    def _run(self, job: EncoderJob) -> EncoderResult:
+    def _run(self, job: EncoderJob, pool=None) -> EncoderResult:
         emb = self.cache.get(job.content_hash)
         if emb is not None:
+            if pool is not None and job.slot is not None:
+                t, d = emb.shape
+                pool.write(job.slot, emb.cuda())
+                return EncoderResult(job.job_id, None, cached=True,
+                                     via_pool=True, n_tokens=t,
+                                     embed_dim=d)
             return EncoderResult(job.job_id, emb, cached=True)
         if job.pixel_values is None:
             return EncoderResult(job.job_id, None)  # probe miss
@@ -133,6 +157,13 @@ class EncoderServer:
             logger.exception("encoder job %d failed", job.job_id)
             return EncoderResult(job.job_id, None, error=str(e))
         self.cache.put(job.content_hash, emb)
+        if pool is not None and job.slot is not None:
+            # GPU-direct: WRITE into the client's slot; TCP carries only
+            # the readiness notification
+            t, d = emb.shape
+            pool.write(job.slot, emb.cuda())
+            return EncoderResult(job.job_id, None, via_pool=True,
+                                 n_tokens=t, embed_dim=d)
         return EncoderResult(job.job_id, emb)
 
 
@@ -149,6 +180,9 @@ class EncoderClient:
         import os
         self._addr = addr
         self._sock = None
+        self._pool = None
+        self._use_pool = os.environ.get("GLLM_DISAGG_GPU_DIRECT",
+                                        "1") != "0"
         self._connect()
         self._lock = threading.Lock()
         self._job_id = 0
@@ -164,22 +198,50 @@ class EncoderClient:
                 pass
         self._sock = socket.create_connection((host, int(port)),
                                               timeout=60)
+        self._register_pool()
+
+    def _register_pool(self):
+        """One-time hipIpc handshake per connection; downgrades to the
+        TCP payload path when the encoder can't map us (cross-node)."""
+        if not self._use_pool:
+            return
+        from gllm_amd.disagg.gpu_plane import try_make_pool
+        from gllm_amd.disagg.protocol import PoolRegistration
+        if self._pool is None:
+            self._pool = try_make_pool()
+        if self._pool is None:
+            self._use_pool = False
+            return
+        send_msg(self._sock, PoolRegistration(
+            bytes(self._pool.handle), self._pool.n_slots,
+            self._pool.slot_elems))
+        ack = recv_msg(self._sock)
+        if not (isinstance(ack, dict) and ack.get("pool")):
+            logger.info("encoder declined GPU-direct plane; TCP payload")
+            self._use_pool = False
 
     def _attempt(self, key, grids, pixel_values):
         self._job_id += 1
         jid = self._job_id
-        send_msg(self._sock, EncoderJob(jid, key, list(grids)))
+        slot = self._pool.acquire() if self._use_pool else None
+        send_msg(self._sock, EncoderJob(jid, key, list(grids), slot=slot))
         res: EncoderResult = recv_msg(self._sock)
         if res is None:
             raise ConnectionError("encoder connection closed")
-        if res.embeds is None and res.error is None:
+        if res.embeds is None and res.error is None and not res.via_pool:
             send_msg(self._sock,
-                     EncoderJob(jid, key, list(grids), pixel_values))
+                     EncoderJob(jid, key, list(grids), pixel_values,
+                                slot=slot))
             res = recv_msg(self._sock)
             if res is None:
                 raise ConnectionError("encoder connection closed")
         if res.error:
             raise RuntimeError(f"encoder job failed: {res.error}")
+        if res.via_pool:
+            # the notification gates the xGMI write; clone out of the
+            # slot so it can be reused
+            return self._pool.view(slot, res.n_tokens,
+                                   res.embed_dim).clone()
         return res.embeds
 
     def encode(self, pixel_values: torch.Tensor, grids) -> torch.Tensor:

@@ -5,12 +5,15 @@ notification wire format). Transport is the length-prefixed pickle
 framing shared with the multi-node control plane
 (engine/multinode.py send_msg/recv_msg).
 
-Round-1 data plane: embeddings return over the same TCP channel (CPU
-and single-node correct). Round-2 (ROADMAP.md): the MI355X data plane —
-encoder ranks write embeddings straight into per-LM-rank hipIpc slot
-pools over xGMI (the reference uses NIXL/UCX GPU WRITEs,
-nixl_transfer.py:122-298), with the TCP channel carrying only readiness
-notifications.
+Data planes:
+* TCP (always available, cross-node): embeddings ride the result
+  pickle.
+* GPU-direct (r2, intra-node): the LM side pre-registers a hipIpc slot
+  pool (disagg/gpu_plane.py); the encoder maps it once and WRITES each
+  job's embeddings straight into the designated slot over xGMI, and the
+  TCP result carries only (slot, n_tokens) — the reference's NIXL/UCX
+  GPU-WRITE role (nixl_transfer.py:122-298) on hipIpc. Falls back to
+  TCP when the mapping fails (cross-node / no GPU).
 """
 
 import dataclasses
@@ -36,12 +39,26 @@ class EncoderJob:
     grids: List[Tuple[int, int, int]]
     # None when the client believes the encoder has this hash cached
     pixel_values: Optional[torch.Tensor] = None
+    # GPU-direct plane: destination slot in the client's registered pool
+    slot: Optional[int] = None
 
 
 @dataclasses.dataclass
 class EncoderResult:
     job_id: int
     # None => cache miss on a pixel-less probe: resend with pixels
+    # (unless via_pool: the payload went over the GPU-direct plane)
     embeds: Optional[torch.Tensor] = None
     cached: bool = False
     error: Optional[str] = None
+    via_pool: bool = False
+    n_tokens: int = 0
+    embed_dim: int = 0
+
+
+@dataclasses.dataclass
+class PoolRegistration:
+    """One-time hipIpc handshake: LM client -> encoder server."""
+    handle: bytes
+    n_slots: int
+    slot_elems: int  # bf16 elements per slot

@@ -75,6 +75,7 @@ void car_all_reduce_v2(torch::Tensor inout, std::vector<int64_t> ptrs,
 int64_t car_open(py::bytes handle_bytes);
 void car_close(int64_t ptr);
 void car_free(int64_t ptr);
+torch::Tensor car_view_tensor(int64_t ptr, int64_t numel);
 void car_all_reduce(torch::Tensor inout, std::vector<int64_t> ptrs,
                     int64_t rank, int64_t world, int64_t epoch);
 
@@ -124,6 +125,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("car_open", &car_open, "map a peer's AR buffer");
   m.def("car_close", &car_close);
   m.def("car_free", &car_free);
+  m.def("car_view_tensor", &car_view_tensor,
+        "non-owning bf16 view over raw device memory");
   m.def("car_all_reduce", &car_all_reduce,
         "one-shot xGMI custom all-reduce (epoch advances by 2/call)");
 }

@@ -126,6 +126,15 @@ int64_t car_open(py::bytes handle_bytes) {
 }
 
 void car_close(int64_t ptr) { (void)hipIpcCloseMemHandle((void*)ptr); }
+
+// Non-owning bf16 tensor view over raw device memory (the disagg
+// GPU-direct slot pool reads/writes through this).
+torch::Tensor car_view_tensor(int64_t ptr, int64_t numel) {
+  auto opts = torch::TensorOptions()
+                  .dtype(at::kBFloat16)
+                  .device(at::kCUDA, at::cuda::current_device());
+  return torch::from_blob(reinterpret_cast<void*>(ptr), {numel}, opts);
+}
 void car_free(int64_t ptr) { (void)hipFree((void*)ptr); }
 
 void car_all_reduce(torch::Tensor inout, std::vector<int64_t> ptrs,
