@@ -87,7 +87,7 @@ def decode_sweep():
             ops.paged_attention(q, k_cache, v_cache, bt, seq_lens, qsl, sc)
         torch.cuda.synchronize()
         dt = (time.time() - t0) / it
-        kv_bytes = 2.0 * B * S * Hkv * D * 2 * 2
+        kv_bytes = 1.0 * B * S * Hkv * D * 2 * 2  # K+V, bf16
         print(f"decode B={B:4d} S={S:5d}: {dt*1e6:8.1f} us  "
               f"KV {kv_bytes/dt/1e12:5.2f} TB/s")
 if __name__ == "__main__" and os.environ.get("DECODE_SWEEP"): decode_sweep()

@@ -93,3 +93,62 @@ def test_repetition_penalty_kernel_matches_torch():
     torch.cuda.synchronize()
     assert torch.allclose(out, ref, atol=1e-5), \
         (out - ref).abs().max()
+
+
+def _pool_worker(rank, port, q):
+    os.environ.update(RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import torch
+    import torch.distributed as dist
+    torch.cuda.set_device(0)
+    dist.init_process_group("gloo", rank=rank, world_size=2)
+    from gllm_amd.disagg.gpu_plane import RemotePool, SlotPool
+    from gllm_amd.disagg.protocol import PoolRegistration
+    ok = True
+    if rank == 0:  # LM side: owns the pool
+        pool = SlotPool(n_slots=4, slot_elems=4096)
+        h = torch.zeros(len(pool.handle), dtype=torch.uint8)
+        h[:] = torch.tensor(list(pool.handle), dtype=torch.uint8)
+        dist.broadcast(h, src=0)
+        dist.barrier()  # writer finished
+        for slot, seed in [(1, 7), (3, 11)]:
+            g = torch.Generator().manual_seed(seed)
+            ref = torch.randn(32, 64, generator=g).to(torch.bfloat16)
+            got = pool.view(slot, 32, 64).cpu()
+            ok = ok and torch.equal(got, ref)
+        dist.barrier()
+        pool.close()
+    else:  # encoder side: maps and writes
+        h = torch.zeros(64, dtype=torch.uint8)
+        dist.broadcast(h, src=0)
+        reg = PoolRegistration(bytes(h.tolist()), 4, 4096)
+        rp = RemotePool(reg)
+        for slot, seed in [(1, 7), (3, 11)]:
+            g = torch.Generator().manual_seed(seed)
+            emb = torch.randn(32, 64, generator=g).to(torch.bfloat16).cuda()
+            rp.write(slot, emb)
+        dist.barrier()
+        dist.barrier()
+        rp.close()
+    q.put((rank, ok))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_disagg_gpu_plane_two_procs():
+    """hipIpc slot pool: process 1 (encoder role) writes embeddings
+    into process 0's (LM role) pool; contents land bit-exact."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_pool_worker, args=(r, 29883, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = {}
+    for _ in range(2):
+        rank, res = q.get(timeout=240)
+        got[rank] = res
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert all(got.values()), got
