@@ -191,6 +191,10 @@ def main():
                     help="expert parallelism over the dp*tp ranks of "
                          "each stage")
     ap.add_argument("--page-size", type=int, default=16)
+    ap.add_argument("--quant", type=str, default=None,
+                    choices=["fp8"],
+                    help="block-quantize the dummy weights (fp8 e4m3 "
+                         "128x128 blocks) and run the native fp8 path")
     ap.add_argument("--schedule", type=str, default="token_throttling")
     ap.add_argument("--qps", type=float, default=32.0,
                     help=">0: pace request arrivals at this rate for the "
@@ -210,6 +214,11 @@ def main():
     device = f"cuda:{int(os.environ.get('LOCAL_RANK', rank))}" if use_gpu \
         else "cpu"
     model_name, model_json = MODELS[args.model]
+    if args.quant == "fp8":
+        model_json = dict(model_json)
+        model_json["quantization_config"] = {
+            "quant_method": "fp8", "weight_block_size": [128, 128]}
+        model_name += "-fp8"
     model_dir = write_model_dir(model_json)
     assert n % args.tp == 0, (n, args.tp)
     pp = n // args.tp
@@ -333,7 +342,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": "bf16" if use_gpu else "fp32",
+            "dtype": ("fp8-w8a8" if args.quant == "fp8" else "bf16") if use_gpu else "fp32",
             "data": "synthetic",
             "ttft_p50_ms": round(ttft_p50, 1) if ttft_p50 else None,
             "ttft_qps": args.qps if args.qps > 0 else "burst",

@@ -170,7 +170,17 @@ def dummy_init(model: torch.nn.Module, seed: int = 0) -> None:
         gname = global_name(name)
         gen.manual_seed(seed ^ zlib.crc32(gname.encode()))
         with torch.no_grad():
-            if p.dim() >= 2:
+            if p.dtype == torch.float8_e4m3fn:
+                # fp8 dummy weights: draw bf16 then quantize (normal_
+                # has no float8 kernel); keeps the packed layout live
+                # on the --quant fp8 bench path
+                tmp = torch.empty(p.shape, dtype=torch.bfloat16,
+                                  device=p.device)
+                tmp.normal_(0.0, 0.02, generator=gen)
+                p.data.copy_(tmp.clamp(-0.4, 0.4).to(torch.float8_e4m3fn))
+            elif "scale_inv" in gname:
+                p.data.fill_(0.05)
+            elif p.dim() >= 2:
                 # draw in-place on device (32B params via a CPU RNG would
                 # take minutes); bf16 normal_ is supported on ROCm
                 p.data.normal_(0.0, 0.02, generator=gen)
