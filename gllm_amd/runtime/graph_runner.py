@@ -39,8 +39,10 @@ class GraphRunner:
 
         B = self.max_bs
         dev = self.device
+        self.uses_mrope = bool(getattr(runner, "uses_mrope", False))
         self.in_ids = torch.zeros(B, dtype=torch.long, device=dev)
-        self.positions = torch.zeros(B, dtype=torch.long, device=dev)
+        pshape = (3, B) if self.uses_mrope else (B,)
+        self.positions = torch.zeros(pshape, dtype=torch.long, device=dev)
         self.slots = torch.zeros(B, dtype=torch.long, device=dev)
         self.block_table = torch.zeros((B, self.pages_cap),
                                        dtype=torch.int32, device=dev)
@@ -48,7 +50,7 @@ class GraphRunner:
         self.qsl = torch.arange(B + 1, dtype=torch.int32, device=dev)
         # pinned staging (one flat int64 area reused per copy)
         self.pin_ids = torch.zeros(B, dtype=torch.long).pin_memory()
-        self.pin_pos = torch.zeros(B, dtype=torch.long).pin_memory()
+        self.pin_pos = torch.zeros(pshape, dtype=torch.long).pin_memory()
         self.pin_slots = torch.zeros(B, dtype=torch.long).pin_memory()
         self.pin_seq_lens = torch.zeros(B, dtype=torch.int32).pin_memory()
         self.pin_bt = torch.zeros((B, self.pages_cap),
@@ -66,9 +68,11 @@ class GraphRunner:
                                       dtype=torch.long).pin_memory()
 
     def _fctx_for(self, bs: int) -> ForwardContext:
+        pos = self.positions[:, :bs] if self.uses_mrope \
+            else self.positions[:bs]
         fctx = ForwardContext(
             num_tokens=bs,
-            positions=self.positions[:bs],
+            positions=pos,
             slot_mapping=self.slots[:bs],
             block_table=self.block_table[:bs],
             seq_lens=self.seq_lens[:bs],
@@ -98,15 +102,17 @@ class GraphRunner:
         for bs in reversed(self.buckets):
             fctx = self._fctx_for(bs)
             with torch.cuda.stream(stream):
+                pos = self.positions[:, :bs] if self.uses_mrope \
+                    else self.positions[:bs]
                 for _ in range(2):  # warmup outside capture
                     ids = runner.resolve_tokens(self.in_ids[:bs])
-                    runner.model(ids, self.positions[:bs], fctx)
+                    runner.model(ids, pos, fctx)
             torch.cuda.current_stream().wait_stream(stream)
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g, pool=pool, stream=stream):
                 ids = runner.resolve_tokens(self.in_ids[:bs])
-                hidden, _ = runner.model(ids, self.positions[:bs], fctx)
+                hidden, _ = runner.model(ids, pos, fctx)
             self.graphs[bs] = g
             self.hidden_out[bs] = hidden
         torch.cuda.synchronize()
@@ -153,7 +159,14 @@ class GraphRunner:
         for i, it in enumerate(items):
             seq = it.seq
             ids[i] = seq.token_ids[it.start]
-            pos[i] = it.start
+            if self.uses_mrope:
+                # decode rows: all three sections advance together from
+                # the prompt's mrope delta (batch_builder.py math)
+                pos[:, i] = (seq.mrope_delta + (it.start - seq.prompt_len)
+                             if seq.mrope_positions is not None
+                             else it.start)
+            else:
+                pos[i] = it.start
             page_tab = seq.page_table
             n_pages = len(page_tab)
             bt[i, :n_pages] = page_tab
@@ -163,7 +176,10 @@ class GraphRunner:
         dummy_base = self.dummy_page * ps
         for i in range(B, bs):
             ids[i] = 0
-            pos[i] = 0
+            if self.uses_mrope:
+                pos[:, i] = 0
+            else:
+                pos[i] = 0
             bt[i, 0] = self.dummy_page
             slots[i] = dummy_base + i % ps
             seq_lens[i] = 1
@@ -178,7 +194,12 @@ class GraphRunner:
                                           non_blocking=True)
         # H2D into the captured buffers
         self.in_ids[:bs].copy_(self.pin_ids[:bs], non_blocking=True)
-        self.positions[:bs].copy_(self.pin_pos[:bs], non_blocking=True)
+        if self.uses_mrope:
+            self.positions[:, :bs].copy_(self.pin_pos[:, :bs],
+                                         non_blocking=True)
+        else:
+            self.positions[:bs].copy_(self.pin_pos[:bs],
+                                      non_blocking=True)
         self.slots[:bs].copy_(self.pin_slots[:bs], non_blocking=True)
         self.seq_lens[:bs].copy_(self.pin_seq_lens[:bs], non_blocking=True)
         npg = max(1, max_pages_used)

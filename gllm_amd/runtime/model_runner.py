@@ -57,11 +57,10 @@ class ModelRunner:
                     and spec.head_k_dim == 128 and spec.conv_kernel == 4):
                 cfg.use_graph = False
         self.uses_mrope = bool(getattr(self.model, "uses_mrope", False))
-        if self.uses_mrope:
-            # [3,B] graph position buffers are round 2; prefix caching
-            # stays ON — image runs get content-hash cache keys
-            # (multimodal/prepare.py + core/kv_cache.py _key_ids)
-            cfg.use_graph = False
+        # mrope models capture with [3, B] position buffers (decode
+        # positions are a per-seq scalar across all three sections);
+        # prefix caching stays ON — image runs get content-hash cache
+        # keys (multimodal/prepare.py + core/kv_cache.py _key_ids)
         self.index_head_dim = getattr(self.model, "index_head_dim", None)
         if self.index_head_dim:
             # DSA selector round-1 runs the eager per-seq torch path;
