@@ -267,6 +267,7 @@ def main():
     ttfts = {}
     arrivals = {}
     if args.qps > 0:
+        import torch.distributed as dist
         to_release = list(seqs)
         next_t = time.time()
         while True:
