@@ -153,29 +153,40 @@ def fused_moe(x: torch.Tensor, w13: torch.Tensor, w2: torch.Tensor,
     inter = two_i // 2
     topk = topk_ids.shape[1]
     n_pairs = T * topk
-    # block_m heuristic: sparse decode routing -> small blocks
-    block_m = 16 if n_pairs < 8 * E_local or n_pairs <= 512 else 64
-    cap = n_pairs + E_local * (block_m - 1) + 1
-    max_blocks = (n_pairs + block_m - 1) // block_m + E_local
+    # tile heuristic: match block_m to the expected rows per expert so
+    # decode doesn't re-stream expert weights per m-block; large dense
+    # batches use the wide-BN prefill variant (gemm code 164 = BM 64 x
+    # BN 256 — quarters the A-tile re-reads that dominate prefill)
+    rpe = n_pairs / max(1, E_local)
+    if n_pairs >= 4096:
+        align_m, gemm_m = 64, 164
+    elif rpe < 12:
+        align_m = gemm_m = 16
+    elif rpe < 40:
+        align_m = gemm_m = 32
+    else:
+        align_m = gemm_m = 64
+    cap = n_pairs + E_local * (align_m - 1) + 1
+    max_blocks = (n_pairs + align_m - 1) // align_m + E_local
     dev = x.device
     k = _gpu_kernels()
     ids32 = topk_ids.int().contiguous()
     sorted_ids = torch.empty(cap, dtype=torch.int32, device=dev)
     expert_blocks = torch.empty(max_blocks, dtype=torch.int32, device=dev)
     n_post = torch.empty(1, dtype=torch.int32, device=dev)
-    k.moe_align(ids32, E_local, expert_start, block_m, sorted_ids,
+    k.moe_align(ids32, E_local, expert_start, align_m, sorted_ids,
                 expert_blocks, n_post)
-    rows_pad = cap - 1 + block_m  # >= n_post_pad upper bound
+    rows_pad = cap - 1 + align_m  # >= n_post_pad upper bound
     inter1 = torch.empty(rows_pad, two_i, dtype=x.dtype, device=dev)
     k.moe_gemm(inter1, x, w13, sorted_ids, expert_blocks, n_post, None,
-               n_pairs, topk, block_m, False)
+               n_pairs, topk, gemm_m, False)
     act = silu_and_mul(inter1)
     # zero-filled: pairs routed to non-local experts (EP shards) or to
     # the DP padding id -1 are dropped by the align kernel and must
     # contribute zeros
     pair_out = x.new_zeros(n_pairs, K)
     k.moe_gemm(pair_out, act, w2, sorted_ids, expert_blocks, n_post,
-               topk_weights.float().contiguous(), n_pairs, topk, block_m,
+               topk_weights.float().contiguous(), n_pairs, topk, gemm_m,
                True)
     out = torch.empty(T, K, dtype=x.dtype, device=dev)
     k.moe_sum(out, pair_out, topk)

@@ -76,11 +76,13 @@ __global__ __launch_bounds__(ALIGN_BLOCK) void moe_align_kernel(
 // GATHER (gemm1): row(m) = sorted_ids[g]/topk into x; C row = g (sorted).
 // SCATTER (gemm2): row(m) = g (sorted inter buffer); C row =
 //   sorted_ids[g] (pair index), scaled by topk_w[pair].
-constexpr int BN = 64;
 constexpr int BK = 64;
 constexpr int GEMM_BLOCK = 256;
 
-template <int BM, bool SCATTER>
+// BN is a template parameter: 64 for sparse decode routing, 256 for
+// dense prefill blocks (the wide tile quarters the A-tile re-reads
+// that dominate prefill: A is re-staged once per n-tile of the grid)
+template <int BM, int BN, bool SCATTER>
 __global__ __launch_bounds__(GEMM_BLOCK) void moe_gemm_kernel(
     __hip_bfloat16 *__restrict__ C,
     const __hip_bfloat16 *__restrict__ A,
@@ -100,11 +102,10 @@ __global__ __launch_bounds__(GEMM_BLOCK) void moe_gemm_kernel(
   const int l16 = lane & 15;
   const int lhi = lane >> 4;
 
-  // wave tiling over the BM x BN block: (BM/16) x 4 sixteen-wide tiles
-  // split over 4 waves -> each wave owns one m sub-tile row and
-  // BN/(4/WM*16) = WM consecutive n tiles
+  // wave tiling over the BM x BN block: (BM/16) x (BN/16) tiles over
+  // 4 waves; wave owns one m sub-tile row and BN*WM/64 n tiles
   constexpr int WM = BM / 16;            // m sub-tiles (1, 2 or 4)
-  constexpr int WN_TILES = WM;           // 16-wide n tiles per wave
+  constexpr int WN_TILES = BN * WM / 64; // 16-wide n tiles per wave
   const int wm = wave % WM;
   const int wn = wave / WM;              // n-tile group (4/WM groups)
 
@@ -290,15 +291,15 @@ void moe_gemm(torch::Tensor C, torch::Tensor A, torch::Tensor W,
   TORCH_CHECK(A.size(-1) == K);
   TORCH_CHECK(K % 8 == 0, "moe_gemm: K must be a multiple of 8");
   const int max_blocks = expert_blocks.numel();
-  const int n_tiles = (Nd + BN - 1) / BN;
   const float *tw = nullptr;
   if (topk_weights.has_value()) {
     TORCH_CHECK(topk_weights->scalar_type() == at::kFloat);
     tw = topk_weights->data_ptr<float>();
   }
   auto stream = at::cuda::getCurrentCUDAStream();
-#define LAUNCH(BM, SC)                                                      \
-  hipLaunchKernelGGL((moe_gemm_kernel<BM, SC>), dim3(max_blocks, n_tiles),  \
+#define LAUNCH(BM, BNV, SC)                                                 \
+  hipLaunchKernelGGL((moe_gemm_kernel<BM, BNV, SC>),                        \
+                     dim3(max_blocks, (Nd + BNV - 1) / BNV),                \
                      dim3(GEMM_BLOCK), 0, stream,                           \
                      (__hip_bfloat16 *)C.data_ptr(),                        \
                      (const __hip_bfloat16 *)A.data_ptr(),                  \
@@ -308,13 +309,15 @@ void moe_gemm(torch::Tensor C, torch::Tensor A, torch::Tensor W,
                      n_post_pad.data_ptr<int>(), tw, (int)n_pairs, K, Nd,   \
                      (int)topk)
   if (block_m == 16) {
-    if (scatter) LAUNCH(16, true); else LAUNCH(16, false);
+    if (scatter) LAUNCH(16, 64, true); else LAUNCH(16, 64, false);
   } else if (block_m == 32) {
-    if (scatter) LAUNCH(32, true); else LAUNCH(32, false);
+    if (scatter) LAUNCH(32, 64, true); else LAUNCH(32, 64, false);
   } else if (block_m == 64) {
-    if (scatter) LAUNCH(64, true); else LAUNCH(64, false);
+    if (scatter) LAUNCH(64, 64, true); else LAUNCH(64, 64, false);
+  } else if (block_m == 164) {  // BM=64, BN=256 (dense prefill)
+    if (scatter) LAUNCH(64, 256, true); else LAUNCH(64, 256, false);
   } else {
-    TORCH_CHECK(false, "moe_gemm: block_m must be 16/32/64");
+    TORCH_CHECK(false, "moe_gemm: block_m must be 16/32/64/164");
   }
 #undef LAUNCH
   HIP_CHECK_KERNEL();
