@@ -159,13 +159,17 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
     if (has_next) stage_load(kv0 + BKV);
 
     // ---------- QK^T: S[16 q x BKV] per wave ----------
+    // kt OUTER / f INNER: NF independent accumulator chains (PMC: the
+    // f-outer form spent 37% of wave cycles issue-stalled on the
+    // dependent same-fragment MFMA chain)
     mfma_f4 s_frag[NF];
+#pragma unroll
+    for (int f = 0; f < NF; ++f) s_frag[f] = mfma_f4{0, 0, 0, 0};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int f = 0; f < NF; ++f) {
-      s_frag[f] = mfma_f4{0, 0, 0, 0};
+    for (int kt = 0; kt < KT; ++kt) {
 #pragma unroll
-      for (int kt = 0; kt < KT; ++kt) {
+      for (int f = 0; f < NF; ++f) {
         const int krow = f * 16 + l16;
         mfma_bf8 bfrag = *reinterpret_cast<const mfma_bf8 *>(
             reinterpret_cast<char *>(&k_tile[cur][krow * D]) +
@@ -234,12 +238,12 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
       pfrag[ks] = *reinterpret_cast<const mfma_bf8 *>(
           &pw[l16 * VT_STRIDE + ks * 32 + lhi * 8]);
 
-    // ---------- PV ----------
+    // ---------- PV (ks outer / nt inner: NT independent chains) ------
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int nt = 0; nt < NT; ++nt) {
+    for (int ks = 0; ks < BKV / 32; ++ks) {
 #pragma unroll
-      for (int ks = 0; ks < BKV / 32; ++ks) {
+      for (int nt = 0; nt < NT; ++nt) {
         mfma_bf8 vfrag = *reinterpret_cast<const mfma_bf8 *>(
             &vt_tile[cur][(nt * 16 + l16) * VT_STRIDE + ks * 32 + lhi * 8]);
         o_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
