@@ -71,6 +71,23 @@ class LinearBase(nn.Module):
                     x.dtype).to(x.device)
             w = self._w_dq
         elif self.int4_cfg is not None:
+            if (x.is_cuda and x.dim() == 2 and x.dtype == torch.bfloat16
+                    and x.shape[0] <= 256 and x.shape[1] % 256 == 0
+                    and self.int4_cfg[1] == 128 and ops.has_kernels()):
+                # decode: fused-dequant nibble-streaming GEMM (weights
+                # stay packed; one-time canonical repack)
+                if getattr(self, "_i4_canon", None) is None:
+                    from gllm_amd.layers.quantization.int4 import \
+                        repack_canonical
+                    wq4, sb, grp = repack_canonical(self)
+                    self._i4_canon = (wq4.to(x.device),
+                                      sb.to(x.device), grp)
+                wq4, sb, grp = self._i4_canon
+                return ops.int4_linear(x, wq4, sb, grp, bias)
+            if x.is_cuda:
+                # prefill burst: dequant per call, nothing cached
+                from gllm_amd.layers.quantization.int4 import dequant_layer
+                return ops.linear(x, dequant_layer(self, x.dtype), bias)
             if self._w_dq is None:
                 from gllm_amd.layers.quantization.int4 import dequant_layer
                 self._w_dq = dequant_layer(self, x.dtype).to(x.device)

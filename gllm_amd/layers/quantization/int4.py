@@ -304,3 +304,32 @@ def dequant_layer(layer, dtype: torch.dtype) -> torch.Tensor:
     method, group = layer.int4_cfg
     fn = dequant_gptq if method == "gptq" else dequant_awq
     return fn(layer.qweight, layer.qzeros, layer.scales, group, dtype)
+
+
+# ------------------------------------------------- canonical repack (r2)
+def repack_canonical(layer, dtype=torch.bfloat16):
+    """One-time repack of a converted int4 linear's checkpoint tensors
+    into the gfx950 fused-dequant GEMM layout (ops/csrc/int4.hip):
+
+      wq4  uint8 [N, K/2]      (low nibble = even k, high = odd k)
+      sb   fp32  [N, K/g, 2]   ({scale, -zero*scale} per group)
+
+    Returns (wq4, sb, group). Runs in torch once (the reference's
+    gptq_marlin_repack role); caller caches the result and FREES the
+    original packed tensors.
+    """
+    method, group = layer.int4_cfg
+    if method == "gptq":
+        wq = _unpack_nibbles_k(layer.qweight)              # [K, N]
+        zeros = _unpack_nibbles_n(layer.qzeros) + 1        # [G, N]
+    else:
+        wq = _unpack_nibbles_n(layer.qweight, AWQ_ORDER)   # [K, N]
+        zeros = _unpack_nibbles_n(layer.qzeros, AWQ_ORDER)
+    scales = layer.scales.float()                          # [G, N]
+    wq = wq.t().contiguous().to(torch.uint8)               # [N, K]
+    N, K = wq.shape
+    wq4 = (wq[:, 0::2] | (wq[:, 1::2] << 4)).contiguous()  # [N, K/2]
+    s = scales.t().contiguous()                            # [N, G]
+    b = (-zeros.float() * scales).t().contiguous()         # [N, G]
+    sb = torch.stack([s, b], dim=-1).contiguous()          # [N, G, 2]
+    return wq4, sb, group

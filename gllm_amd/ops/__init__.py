@@ -315,6 +315,29 @@ def rmsnorm_gated(x: torch.Tensor, z: torch.Tensor, weight: torch.Tensor,
     return out
 
 
+def int4_linear(x: torch.Tensor, wq4: torch.Tensor, sb: torch.Tensor,
+                group: int,
+                bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out = x @ dequant(wq4).T via the fused-dequant int4 skinny GEMM
+    (M <= 256): packed nibbles stream at 0.5 B/elem."""
+    M, K = x.shape
+    N = wq4.shape[0]
+    splitk = _skinny_splitk(M, N, K)
+    need = splitk * M * N
+    key = x.device.index or 0
+    ws = _FP8_WS.get(key)  # shares the fp32 partial workspace pool
+    if ws is None or ws.numel() < need:
+        if torch.cuda.is_current_stream_capturing():
+            raise RuntimeError("int4 workspace grown during graph capture")
+        ws = torch.empty(need, dtype=torch.float32, device=x.device)
+        _FP8_WS[key] = ws
+    out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    _gpu_kernels().int4_skinny_gemm(
+        out, x.contiguous(), wq4, sb,
+        bias.float() if bias is not None else None, ws, group)
+    return out
+
+
 # --------------------------------------------------------------- sampling
 def topk_topp_filter(probs: torch.Tensor, top_ks: torch.Tensor,
                      top_ps: torch.Tensor,

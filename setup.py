@@ -27,6 +27,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "moe.hip"),
         os.path.join(CSRC, "fp8.hip"),
         os.path.join(CSRC, "gdn.hip"),
+        os.path.join(CSRC, "int4.hip"),
         os.path.join(CSRC, "skinny_gemm.hip"),
         os.path.join(CSRC, "sampling.hip"),
         os.path.join(CSRC, "custom_ar.hip"),

@@ -515,3 +515,24 @@ def test_rmsnorm_gated(kernels):
     out = ops.rmsnorm_gated(x, z, w, 1e-6)
     ref = gdn_ref.rmsnorm_gated(x.cpu(), z.cpu(), w.cpu(), 1e-6)
     assert_close_bf16(out, ref.float())
+
+
+# ------------------------------------------------------------ int4
+@pytest.mark.parametrize("case", [(8, 512, 1024), (64, 896, 2048),
+                                  (200, 1024, 512)])
+def test_int4_linear(kernels, case):
+    M, N, K = case
+    torch.manual_seed(M)
+    from types import SimpleNamespace
+    from gllm_amd import ops
+    from gllm_amd.layers.quantization import int4 as qi4
+    w = torch.randn(N, K) / math.sqrt(K)
+    qweight, qzeros, scales = qi4.pack_gptq(w, group_size=128)
+    layer = SimpleNamespace(int4_cfg=("gptq", 128), qweight=qweight,
+                            qzeros=qzeros, scales=scales)
+    wq4, sb, grp = qi4.repack_canonical(layer)
+    ref_w = qi4.dequant_gptq(qweight, qzeros, scales, 128, torch.float32)
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    out = ops.int4_linear(x, wq4.cuda(), sb.cuda(), grp)
+    ref = x.float().cpu() @ ref_w.T
+    assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=2e-3)
