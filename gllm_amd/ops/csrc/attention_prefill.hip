@@ -1,19 +1,20 @@
 // Varlen causal (chunked-)prefill attention over the paged KV cache,
 // MFMA-based, for gfx950.
 //
-// v2 structure: grid = (q_tiles, B, Hq), block = 256 threads (4 waves).
-// Each workgroup computes a 64-row Q tile for one q head; each wave owns
-// 16 q rows. K/V tiles of BKV=64 tokens double-buffered in LDS with the
-// T14 issue-early/write-late split (guide par.6 G15): tile t+1's global
-// loads issue before tile t's MFMAs, the LDS write lands after the
-// barrier, so HBM latency hides under compute. K image XOR-swizzled for
-// conflict-free ds_read_b128 (T2); V transposed at staging (padded
-// stride) so the PV B-fragment reads contiguous kv; P re-enters
-// A-fragment layout through a per-wave LDS bounce. s_setprio(1) wraps
-// the MFMA clusters (T5). Online softmax per row in registers.
+// v3 structure (v1/v2 ran 4 waves at 80 TF/s — PMC showed the time
+// split across issue-stalls and VALU/LDS staging overhead, not MFMA):
+// 512 threads = 8 waves at 2 waves/SIMD, BQ = 128 q rows per workgroup
+// (16 per wave), K/V tiles of BKV=64 streamed through a 2-slot
+// global_load_lds ring with counted s_waitcnt vmcnt (the proven
+// skinny/MLA pipeline — no register staging instructions, loads fly
+// under the previous tile's MFMAs). K staged XOR-swizzled via
+// source-address swizzle; V staged linear and PV B-fragments gather
+// straight from it (8 ds_read_b16 per fragment — replaces the v2
+// transpose staging). Fully-visible interior tiles skip the causal
+// mask lane work entirely.
 //
 // Capability parity: flash_attn_with_kvcache varlen semantics
-// (reference layers/attention.py:77-141) -- one kernel serves chunked
+// (reference layers/attention.py:77-141) — one kernel serves chunked
 // prefill, mixed prefill+decode batches, prefix-cache hits and sliding
 // windows.
 
@@ -26,22 +27,21 @@ typedef __attribute__((ext_vector_type(4))) float mfma_f4;
 
 namespace {
 
-constexpr int BLOCK = 256;
-constexpr int BQ = 64;       // q rows per workgroup
-constexpr int BKV = 64;      // kv tokens per tile
+constexpr int BLOCK = 512;   // 8 waves
+constexpr int BQ = 128;      // q rows per workgroup (16 per wave)
+constexpr int BKV = 64;      // kv tokens per ring slot
+constexpr int RING = 2;
 constexpr int NF = BKV / 16; // score fragments per wave (4)
-constexpr int VT_STRIDE = BKV + 8;  // padded kv stride of V^T / P images
+constexpr int PT_STRIDE = BKV + 8;
 
 template <int D>
 DEV_INLINE int kswz(int row, int byte_off) {
-  // XOR swizzle within a K row (T2). Mask keeps the offset inside the
-  // D*2-byte row (D=128: row&15 conflict-free; D=64: row&7 <=2-way).
   constexpr int MASK = (D * 2 / 16 - 1) & 15;
   return byte_off ^ ((row & MASK) << 4);
 }
 
 template <int D>
-__global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
+__global__ __launch_bounds__(BLOCK, 1) void paged_prefill_kernel(
     __hip_bfloat16 *__restrict__ out,            // [T, Hq, D]
     const __hip_bfloat16 *__restrict__ q,        // [T, Hq, D] (row stride)
     const __hip_bfloat16 *__restrict__ k_cache,  // [P, ps, Hkv, D]
@@ -64,16 +64,18 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
 
   constexpr int KT = D / 32;        // QK^T k-steps
   constexpr int NT = D / 16;        // PV n-tiles
+  constexpr int KROW_B = D * 2;     // K/V row bytes
+  constexpr int TILE_B = BKV * KROW_B;          // one K or V tile
+  constexpr int GL_PER_WAVE = 2 * TILE_B / 1024 / 8;  // K+V chunks/wave
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
   const int l16 = lane & 15;
   const int lhi = lane >> 4;
 
-  // ---- LDS (double-buffered K + V^T; per-wave P bounce) ----
-  __shared__ __hip_bfloat16 k_tile[2][BKV * D];
-  __shared__ __hip_bfloat16 vt_tile[2][D * VT_STRIDE];
-  __shared__ __hip_bfloat16 p_tile[4][16 * VT_STRIDE];
+  // ---- LDS ring: [K tile | V tile] per slot; per-wave P bounce ----
+  __shared__ __attribute__((aligned(16))) char kv_ring[RING * 2 * TILE_B];
+  __shared__ __hip_bfloat16 p_tile[8][16 * PT_STRIDE];
 
   // ---- Q fragments in registers (A-operand layout) ----
   const int qrow_local = tile * BQ + wave * 16 + l16;
@@ -104,64 +106,50 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
   if (window > 0)
     kv_lo = max(0, past + tile * BQ - window + 1) / BKV * BKV;
 
-  // ---- staging: per-thread slice of a K/V tile ----------------------
-  constexpr int LPT = D / 8;            // 16-B chunks per token row
-  constexpr int ROWS_PER_IT = BLOCK / LPT;
-  constexpr int N_IT = BKV / ROWS_PER_IT;
-  const int s_chunk = tid % LPT;
-  const int s_row0 = tid / LPT;
-
-  shortx8 kreg[N_IT], vreg[N_IT];
-  auto stage_load = [&](int kv0) {
+  // ---- glds staging: every wave stages GL_PER_WAVE 1-KiB chunks ----
+  auto stage = [&](int kv0, int slot) {
+    char *base = kv_ring + slot * 2 * TILE_B;
 #pragma unroll
-    for (int it = 0; it < N_IT; ++it) {
-      const int row = s_row0 + it * ROWS_PER_IT;
-      const int tok = kv0 + row;
-      if (tok < seq_len) {
-        const long crow =
-            ((long)bt[tok / page_size] * page_size + tok % page_size);
-        const __hip_bfloat16 *kp =
-            k_cache + (crow * num_kv_heads + kvh) * D + s_chunk * 8;
-        const __hip_bfloat16 *vp =
-            v_cache + (crow * num_kv_heads + kvh) * D + s_chunk * 8;
-        kreg[it] = *reinterpret_cast<const shortx8 *>(kp);
-        vreg[it] = *reinterpret_cast<const shortx8 *>(vp);
-      } else {
-        kreg[it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
-        vreg[it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-    }
-  };
-  auto stage_write = [&](int buf) {
-#pragma unroll
-    for (int it = 0; it < N_IT; ++it) {
-      const int row = s_row0 + it * ROWS_PER_IT;
-      *reinterpret_cast<shortx8 *>(
-          reinterpret_cast<char *>(&k_tile[buf][row * D]) +
-          kswz<D>(row, s_chunk * 16)) = kreg[it];
-      const __hip_bfloat16 *ve =
-          reinterpret_cast<const __hip_bfloat16 *>(&vreg[it]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        vt_tile[buf][(s_chunk * 8 + j) * VT_STRIDE + row] = ve[j];
+    for (int j = 0; j < GL_PER_WAVE; ++j) {
+      const int p = (wave * GL_PER_WAVE + j) * 1024 + lane * 16;
+      const bool is_k = p < TILE_B;
+      const int pp = is_k ? p : p - TILE_B;
+      const int row = pp / KROW_B;
+      const int tok = min(kv0 + row, seq_len - 1);
+      const long crow =
+          (long)bt[tok / page_size] * page_size + tok % page_size;
+      const __hip_bfloat16 *src = is_k ? k_cache : v_cache;
+      const int col = is_k ? kswz<D>(row, pp % KROW_B) : pp % KROW_B;
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const unsigned int *>(
+              reinterpret_cast<const char *>(
+                  src + (crow * num_kv_heads + kvh) * D) + col),
+          reinterpret_cast<unsigned int *>(base + p), 16, 0, 0);
     }
   };
 
-  // prologue: stage the first tile
-  stage_load(kv_lo);
-  stage_write(0);
-  __syncthreads();
+  const int nkt = (kv_max - kv_lo + BKV - 1) / BKV;
+  const int pre = min(RING - 1, nkt);
+  for (int t = 0; t < pre; ++t) stage(kv_lo + t * BKV, t % RING);
 
-  int cur = 0;
-  for (int kv0 = kv_lo; kv0 < kv_max; kv0 += BKV) {
-    const bool has_next = kv0 + BKV < kv_max;
-    // T14 issue-early: next tile's global loads start now
-    if (has_next) stage_load(kv0 + BKV);
+  for (int kt_i = 0; kt_i < nkt; ++kt_i) {
+    const int kv0 = kv_lo + kt_i * BKV;
+    const int slot = kt_i % RING;
+    if (kt_i + RING - 1 < nkt)
+      stage(kv_lo + (kt_i + RING - 1) * BKV, (kt_i + RING - 1) % RING);
+    const int ahead = min(nkt - 1 - kt_i, RING - 1);
+    if (ahead >= 1) {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"n"(GL_PER_WAVE) : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __syncthreads();
 
-    // ---------- QK^T: S[16 q x BKV] per wave ----------
-    // kt OUTER / f INNER: NF independent accumulator chains (PMC: the
-    // f-outer form spent 37% of wave cycles issue-stalled on the
-    // dependent same-fragment MFMA chain)
+    const char *ktile = kv_ring + slot * 2 * TILE_B;
+    const __hip_bfloat16 *vtile = reinterpret_cast<const __hip_bfloat16 *>(
+        ktile + TILE_B);
+
+    // ---------- QK^T: S[16 q x BKV] (kt outer: NF independent) -------
     mfma_f4 s_frag[NF];
 #pragma unroll
     for (int f = 0; f < NF; ++f) s_frag[f] = mfma_f4{0, 0, 0, 0};
@@ -172,15 +160,21 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
       for (int f = 0; f < NF; ++f) {
         const int krow = f * 16 + l16;
         mfma_bf8 bfrag = *reinterpret_cast<const mfma_bf8 *>(
-            reinterpret_cast<char *>(&k_tile[cur][krow * D]) +
-            kswz<D>(krow, (kt * 32 + lhi * 8) * 2));
+            ktile + krow * KROW_B + kswz<D>(krow, (kt * 32 + lhi * 8) * 2));
         s_frag[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             qfrag[kt], bfrag, s_frag[f], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // ---------- mask + online softmax (rows lhi*4+r per lane) ----------
+    // ---------- mask + online softmax (rows lhi*4+r per lane) --------
+    // interior tiles (every key visible to every row of this wave, no
+    // window cut) skip the per-element mask lanes entirely
+    const int qpos_min = past + tile * BQ + wave * 16;
+    const bool full_tile =
+        qrow_valid && (tile * BQ + wave * 16 + 15 < q_len) &&
+        (kv0 + BKV - 1 <= qpos_min) && (kv0 + BKV <= seq_len) &&
+        (window <= 0 || kv0 > qpos_min + 15 - window);
     float p_vals[NF][4];
     float alpha[4];
 #pragma unroll
@@ -189,14 +183,23 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
       const int qpos = past + qrow;
       float sv[NF];
       float mx = -INFINITY;
+      if (full_tile) {
 #pragma unroll
-      for (int f = 0; f < NF; ++f) {
-        const int kvp = kv0 + f * 16 + l16;
-        float s = s_frag[f][r] * scale;
-        if (qrow >= q_len || kvp > qpos || kvp >= seq_len) s = -INFINITY;
-        if (window > 0 && kvp <= qpos - window) s = -INFINITY;
-        sv[f] = s;
-        mx = fmaxf(mx, s);
+        for (int f = 0; f < NF; ++f) {
+          sv[f] = s_frag[f][r] * scale;
+          mx = fmaxf(mx, sv[f]);
+        }
+      } else {
+#pragma unroll
+        for (int f = 0; f < NF; ++f) {
+          const int kvp = kv0 + f * 16 + l16;
+          float s = s_frag[f][r] * scale;
+          if (qrow >= q_len || kvp > qpos || kvp >= seq_len)
+            s = -INFINITY;
+          if (window > 0 && kvp <= qpos - window) s = -INFINITY;
+          sv[f] = s;
+          mx = fmaxf(mx, s);
+        }
       }
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1)
@@ -229,36 +232,33 @@ __global__ __launch_bounds__(BLOCK) void paged_prefill_kernel(
     for (int f = 0; f < NF; ++f)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        pw[(lhi * 4 + r) * VT_STRIDE + f * 16 + l16] =
+        pw[(lhi * 4 + r) * PT_STRIDE + f * 16 + l16] =
             __float2bfloat16(p_vals[f][r]);
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     mfma_bf8 pfrag[BKV / 32];
 #pragma unroll
     for (int ks = 0; ks < BKV / 32; ++ks)
       pfrag[ks] = *reinterpret_cast<const mfma_bf8 *>(
-          &pw[l16 * VT_STRIDE + ks * 32 + lhi * 8]);
+          &pw[l16 * PT_STRIDE + ks * 32 + lhi * 8]);
 
-    // ---------- PV (ks outer / nt inner: NT independent chains) ------
+    // ---------- PV: B-fragments straight from the linear V tile ------
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < BKV / 32; ++ks) {
 #pragma unroll
       for (int nt = 0; nt < NT; ++nt) {
-        mfma_bf8 vfrag = *reinterpret_cast<const mfma_bf8 *>(
-            &vt_tile[cur][(nt * 16 + l16) * VT_STRIDE + ks * 32 + lhi * 8]);
+        const int dim = nt * 16 + l16;
+        mfma_bf8 vfrag;
+        __hip_bfloat16 *ve = reinterpret_cast<__hip_bfloat16 *>(&vfrag);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ve[j] = vtile[(ks * 32 + lhi * 8 + j) * D + dim];
         o_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             pfrag[ks], vfrag, o_acc[nt], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
-
-    // T14 write-late: land the next tile after everyone finished reading
-    __syncthreads();
-    if (has_next) {
-      stage_write(cur ^ 1);
-      cur ^= 1;
-      __syncthreads();
-    }
+    __syncthreads();  // everyone done with this ring slot
   }
 
   // ---------- epilogue ----------
