@@ -187,7 +187,15 @@ class GatedDeltaNet(nn.Module):
             return self.out_proj(gated.reshape(T, -1))
 
         qsl = fctx.host_qsl()
-        for i in range(len(fctx.ssm_slots)):
+        # ---- GPU mixed/prefill path: conv stays per segment (cheap),
+        # but ALL prefill segments run ONE padded batched WY pass
+        # (chunk=256; padding rows carry beta=0/g=0 so they are inert)
+        # instead of a per-seq python loop of ~20 launches per 64-token
+        # chunk — the r2 TTFT fix for hybrid models.
+        prefill_batched = hidden.is_cuda and B > 0
+        conv_outs = [None] * B
+        prefill_idx = []
+        for i in range(B):
             s, e = qsl[i], qsl[i + 1]
             slot = int(fctx.ssm_slots[i])
             has_init = bool(fctx.ssm_has_init[i])
@@ -202,9 +210,11 @@ class GatedDeltaNet(nn.Module):
                 conv_out = gdn_ref.causal_conv1d_prefill(
                     mixed[s:e], self.conv1d_weight, conv_states[slot],
                     has_init)
+                if prefill_batched:
+                    conv_outs[i] = conv_out
+                    prefill_idx.append(i)
+                    continue
             qd, kd, vd = conv_out.split([k_tp, k_tp, v_tp], dim=-1)
-            # decode: one sequential step; prefill: chunk-parallel WY
-            # form (O(T/64) sequential steps — the round-2 kernel shape)
             rule = gdn_ref.gated_delta_rule if e - s == 1 \
                 else gdn_ref.gated_delta_rule_chunked
             o = rule(
@@ -213,6 +223,36 @@ class GatedDeltaNet(nn.Module):
                 vd.view(e - s, self.tp_v, self.head_v_dim),
                 g_all[s:e], beta_all[s:e], self.scale, ssm_states[slot])
             out_core[s:e] = o
+
+        if prefill_idx:
+            Bp = len(prefill_idx)
+            Tmax = max(qsl[i + 1] - qsl[i] for i in prefill_idx)
+            dev = hidden.device
+            qb = torch.zeros(Bp, Tmax, self.tp_k_heads, self.head_k_dim,
+                             dtype=torch.float32, device=dev)
+            kb = torch.zeros_like(qb)
+            vb = torch.zeros(Bp, Tmax, self.tp_v, self.head_v_dim,
+                             dtype=torch.float32, device=dev)
+            gb = torch.zeros(Bp, Tmax, self.tp_v, dtype=torch.float32,
+                             device=dev)
+            bb = torch.zeros_like(gb)
+            slots_l = [int(fctx.ssm_slots[i]) for i in prefill_idx]
+            for j, i in enumerate(prefill_idx):
+                s, e = qsl[i], qsl[i + 1]
+                n = e - s
+                qd, kd, vd = conv_outs[i].split([k_tp, k_tp, v_tp], -1)
+                qb[j, :n] = qd.view(n, self.tp_k_heads, self.head_k_dim)
+                kb[j, :n] = kd.view(n, self.tp_k_heads, self.head_k_dim)
+                vb[j, :n] = vd.view(n, self.tp_v, self.head_v_dim)
+                gb[j, :n] = g_all[s:e]
+                bb[j, :n] = beta_all[s:e]
+            states_b = ssm_states[slots_l].contiguous()
+            ob = gdn_ref.gated_delta_rule_chunked_batched(
+                qb, kb, vb, gb, bb, self.scale, states_b)
+            ssm_states[slots_l] = states_b
+            for j, i in enumerate(prefill_idx):
+                s, e = qsl[i], qsl[i + 1]
+                out_core[s:e] = ob[j, :e - s].to(out_core.dtype)
         gated = gdn_ref.rmsnorm_gated(
             out_core.reshape(T * self.tp_v, self.head_v_dim),
             z.reshape(T * self.tp_v, self.head_v_dim),

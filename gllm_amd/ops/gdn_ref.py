@@ -156,6 +156,62 @@ def gated_delta_rule_chunked(q: torch.Tensor, k: torch.Tensor,
     return torch.cat(outs).to(v.dtype)
 
 
+def gated_delta_rule_chunked_batched(
+        q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+        g: torch.Tensor, beta: torch.Tensor, scale: float,
+        states: torch.Tensor, chunk: int = 256) -> torch.Tensor:
+    """Chunk-parallel WY delta rule over a PADDED batch of sequences.
+
+    q, k: [B, T, Hk, Dk]; v: [B, T, Hv, Dv]; g, beta: [B, T, Hv];
+    states: [B, Hv, Dv, Dk] fp32 (updated IN PLACE). Padding rows MUST
+    carry beta = 0 and g = 0: a zero-beta token has U row 0, so it
+    neither updates the state nor contributes to any output — the
+    padded math is exactly the ragged math. Returns o [B, T, Hv, Dv].
+
+    This removes the engine's per-sequence python loop (B x ~20
+    launches per chunk); with chunk=256 a 1k-token prompt runs 4
+    sequential chunk steps instead of 16.
+    """
+    B, T, Hk, Dk = q.shape
+    Hv, Dv = v.shape[2], v.shape[3]
+    G = Hv // Hk
+    qn = (l2norm(q.float()) * scale).repeat_interleave(G, dim=2)
+    kn = l2norm(k.float()).repeat_interleave(G, dim=2)  # [B, T, Hv, Dk]
+    vf = v.float()
+    gf = g.float()
+    bf = beta.float()
+    S = states.float()                                  # [B, Hv, Dv, Dk]
+    outs = []
+    for s in range(0, T, chunk):
+        e = min(s + chunk, T)
+        C = e - s
+        Q = qn[:, s:e].permute(0, 2, 1, 3)              # [B, Hv, C, Dk]
+        K = kn[:, s:e].permute(0, 2, 1, 3)
+        V = vf[:, s:e].permute(0, 2, 1, 3)              # [B, Hv, C, Dv]
+        b = gf[:, s:e].cumsum(1).transpose(1, 2)        # [B, Hv, C]
+        bt = bf[:, s:e].transpose(1, 2)                 # [B, Hv, C]
+        Bd = b.exp()
+        kk = torch.einsum("bhtd,bhid->bhti", K, K)
+        dec = (b.unsqueeze(-1) - b.unsqueeze(-2)).tril(-1).exp()
+        A = (bt.unsqueeze(-1) * dec * kk).tril(-1)
+        M = bt.unsqueeze(-1) * (
+            V - Bd.unsqueeze(-1) * torch.einsum("bhtd,bhvd->bhtv", K, S))
+        eye = torch.eye(C, device=A.device).expand_as(A)
+        U = torch.linalg.solve_triangular(eye + A, M, upper=False,
+                                          unitriangular=True)
+        qk = torch.einsum("bhtd,bhid->bhti", Q, K)
+        deci = (b.unsqueeze(-1) - b.unsqueeze(-2)).tril().exp()
+        att = (deci * qk).tril()
+        O = Bd.unsqueeze(-1) * torch.einsum("bhtd,bhvd->bhtv", Q, S) + \
+            torch.einsum("bhti,bhiv->bhtv", att, U)
+        outs.append(O.permute(0, 2, 1, 3))
+        wC = (b[..., C - 1:].expand_as(b) - b).exp()    # [B, Hv, C]
+        S = Bd[..., C - 1].view(B, Hv, 1, 1) * S + \
+            torch.einsum("bhtv,bhtd->bhvd", wC.unsqueeze(-1) * U, K)
+    states.copy_(S.to(states.dtype))
+    return torch.cat(outs, dim=1).to(v.dtype)
+
+
 def rmsnorm_gated(x: torch.Tensor, z: torch.Tensor, weight: torch.Tensor,
                   eps: float) -> torch.Tensor:
     """out = rmsnorm(x) * w * silu(z)  (norm applied before the gate;

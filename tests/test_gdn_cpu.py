@@ -321,3 +321,41 @@ def test_hybrid_prefix_multiturn_reuse(tmp_path):
         [p2], sp)[0].token_ids
     assert out2 == cold
     assert llm.runner.memory_manager.hit_tokens > hits_before
+
+
+def test_batched_chunked_matches_ragged():
+    """Padded batched WY (beta=0 padding) == per-seq chunked == the
+    sequential recurrence, including state carry."""
+    import torch
+    from gllm_amd.ops import gdn_ref
+    torch.manual_seed(3)
+    Hk, Hv, Dk, Dv = 2, 4, 128, 64
+    lens = [70, 1, 130]
+    B, Tmax = len(lens), max(lens)
+    scale = Dk ** -0.5
+    qb = torch.zeros(B, Tmax, Hk, Dk)
+    kb = torch.zeros(B, Tmax, Hk, Dk)
+    vb = torch.zeros(B, Tmax, Hv, Dv)
+    gb = torch.zeros(B, Tmax, Hv)
+    bb = torch.zeros(B, Tmax, Hv)
+    per = []
+    for i, n in enumerate(lens):
+        q = torch.randn(n, Hk, Dk)
+        k = torch.randn(n, Hk, Dk)
+        v = torch.randn(n, Hv, Dv) / 4
+        g = -torch.rand(n, Hv) * 0.1
+        b = torch.rand(n, Hv)
+        qb[i, :n], kb[i, :n], vb[i, :n] = q, k, v
+        gb[i, :n], bb[i, :n] = g, b
+        per.append((q, k, v, g, b))
+    states_b = torch.randn(B, Hv, Dv, Dk) / 8
+    states_ref = states_b.clone()
+    ob = gdn_ref.gated_delta_rule_chunked_batched(
+        qb, kb, vb, gb, bb, scale, states_b, chunk=64)
+    for i, n in enumerate(lens):
+        q, k, v, g, b = per[i]
+        st = states_ref[i].clone()
+        o_ref = gdn_ref.gated_delta_rule(q, k, v, g, b, scale, st)
+        assert torch.allclose(ob[i, :n].float(), o_ref.float(),
+                              atol=1e-3, rtol=1e-3), i
+        assert torch.allclose(states_b[i], st, atol=1e-3, rtol=1e-3), i
