@@ -247,24 +247,33 @@ def fused_moe_fp8(x: torch.Tensor, w13: torch.Tensor, w13_scale,
     inter = two_i // 2
     topk = topk_ids.shape[1]
     n_pairs = T * topk
-    block_m = 16 if n_pairs < 8 * E_local or n_pairs <= 512 else 64
-    cap = n_pairs + E_local * (block_m - 1) + 1
-    max_blocks = (n_pairs + block_m - 1) // block_m + E_local
+    # same tile heuristic as the bf16 path (164 = BM64 x BN256 prefill)
+    rpe = n_pairs / max(1, E_local)
+    if n_pairs >= 4096:
+        align_m, gemm_m = 64, 164
+    elif rpe < 12:
+        align_m = gemm_m = 16
+    elif rpe < 40:
+        align_m = gemm_m = 32
+    else:
+        align_m = gemm_m = 64
+    cap = n_pairs + E_local * (align_m - 1) + 1
+    max_blocks = (n_pairs + align_m - 1) // align_m + E_local
     dev = x.device
     k = _gpu_kernels()
     ids32 = topk_ids.int().contiguous()
     sorted_ids = torch.empty(cap, dtype=torch.int32, device=dev)
     expert_blocks = torch.empty(max_blocks, dtype=torch.int32, device=dev)
     n_post = torch.empty(1, dtype=torch.int32, device=dev)
-    k.moe_align(ids32, E_local, expert_start, block_m, sorted_ids,
+    k.moe_align(ids32, E_local, expert_start, align_m, sorted_ids,
                 expert_blocks, n_post)
-    rows_pad = cap - 1 + block_m
+    rows_pad = cap - 1 + align_m
     aq, as_ = per_token_group_quant_fp8(x.contiguous())
     inter1 = torch.empty(rows_pad, two_i, dtype=x.dtype, device=dev)
     k.moe_gemm_fp8(inter1, aq.view(torch.uint8), as_,
                    w13.view(torch.uint8), w13_scale.contiguous(),
                    sorted_ids, expert_blocks, n_post, None, n_pairs, topk,
-                   block_m, False)
+                   gemm_m, False)
     act = silu_and_mul(inter1)
     actq, act_s = per_token_group_quant_fp8(act)
     pair_out = x.new_zeros(n_pairs, K)
@@ -272,7 +281,7 @@ def fused_moe_fp8(x: torch.Tensor, w13: torch.Tensor, w13_scale,
                    w2.view(torch.uint8), w2_scale.contiguous(),
                    sorted_ids, expert_blocks, n_post,
                    topk_weights.float().contiguous(), n_pairs, topk,
-                   block_m, True)
+                   gemm_m, True)
     out = torch.empty(T, K, dtype=x.dtype, device=dev)
     k.moe_sum(out, pair_out, topk)
     return out

@@ -258,10 +258,9 @@ __global__ void fp8_reduce_kernel(__hip_bfloat16 *__restrict__ out,
 // ------------------------------------------------- fp8 grouped MoE GEMM
 // moe.hip's pipeline with fp8 operands + block scales. LDS-tiled like
 // the bf16 version (BKm = 128 elems = one scale block per k tile).
-constexpr int MOE_BN = 64;
 constexpr int MOE_BK = 128;
 
-template <int BM, bool SCATTER>
+template <int BM, int MOE_BN, bool SCATTER>
 __global__ __launch_bounds__(BLOCK) void moe_gemm_fp8_kernel(
     __hip_bfloat16 *__restrict__ C,
     const unsigned char *__restrict__ A,   // [Ta, K] e4m3
@@ -286,7 +285,7 @@ __global__ __launch_bounds__(BLOCK) void moe_gemm_fp8_kernel(
   const int lhi = lane >> 4;
 
   constexpr int WM = BM / 16;
-  constexpr int WN_TILES = WM;
+  constexpr int WN_TILES = MOE_BN * WM / 64;
   const int wm = wave % WM;
   const int wn = wave / WM;
 
@@ -516,13 +515,13 @@ void moe_gemm_fp8(torch::Tensor C, torch::Tensor A, torch::Tensor As,
   TORCH_CHECK(A.size(-1) == K);
   TORCH_CHECK(K % MOE_BK == 0, "fp8 moe: K must be a multiple of 128");
   const int max_blocks = expert_blocks.numel();
-  const int n_tiles = (Nd + MOE_BN - 1) / MOE_BN;
   const float *tw = nullptr;
   if (topk_weights.has_value()) tw = topk_weights->data_ptr<float>();
   auto stream = at::cuda::getCurrentCUDAStream();
-#define LAUNCH_M(BM, SC)                                                     \
-  hipLaunchKernelGGL((moe_gemm_fp8_kernel<BM, SC>),                          \
-                     dim3(max_blocks, n_tiles), dim3(BLOCK), 0, stream,      \
+#define LAUNCH_M(BM, BNV, SC)                                                \
+  hipLaunchKernelGGL((moe_gemm_fp8_kernel<BM, BNV, SC>),                     \
+                     dim3(max_blocks, (Nd + BNV - 1) / BNV),                 \
+                     dim3(BLOCK), 0, stream,                                 \
                      (__hip_bfloat16 *)C.data_ptr(),                         \
                      (const unsigned char *)A.data_ptr(),                    \
                      As.data_ptr<float>(),                                   \
@@ -532,13 +531,15 @@ void moe_gemm_fp8(torch::Tensor C, torch::Tensor A, torch::Tensor As,
                      n_post_pad.data_ptr<int>(), tw, (int)n_pairs, K, Nd,    \
                      (int)topk)
   if (block_m == 16) {
-    if (scatter) LAUNCH_M(16, true); else LAUNCH_M(16, false);
+    if (scatter) LAUNCH_M(16, 64, true); else LAUNCH_M(16, 64, false);
   } else if (block_m == 32) {
-    if (scatter) LAUNCH_M(32, true); else LAUNCH_M(32, false);
+    if (scatter) LAUNCH_M(32, 64, true); else LAUNCH_M(32, 64, false);
   } else if (block_m == 64) {
-    if (scatter) LAUNCH_M(64, true); else LAUNCH_M(64, false);
+    if (scatter) LAUNCH_M(64, 64, true); else LAUNCH_M(64, 64, false);
+  } else if (block_m == 164) {  // BM=64, BN=256 (dense prefill)
+    if (scatter) LAUNCH_M(64, 256, true); else LAUNCH_M(64, 256, false);
   } else {
-    TORCH_CHECK(false, "moe_gemm_fp8: block_m must be 16/32/64");
+    TORCH_CHECK(false, "moe_gemm_fp8: block_m must be 16/32/64/164");
   }
 #undef LAUNCH_M
   HIP_CHECK_KERNEL();
