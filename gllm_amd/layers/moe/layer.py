@@ -128,9 +128,8 @@ class FusedMoE(nn.Module):
         ``sel.any()`` form synced once per expert: 256/layer on
         DeepSeek-V3). DP padding rows carry expert id -1 and land in
         segment 0 of the shifted bincount (skipped)."""
-        if (x.is_cuda and self.int4_cfg is None
-                and x.dtype == torch.bfloat16 and ops.has_kernels()):
-            if self.fp8_block is None:
+        if (x.is_cuda and x.dtype == torch.bfloat16 and ops.has_kernels()):
+            if self.fp8_block is None and self.int4_cfg is None:
                 return ops.fused_moe(
                     x.contiguous(), self.w13_weight, self.w2_weight,
                     weights, ids, expert_start=self.expert_start,
@@ -143,6 +142,20 @@ class FusedMoE(nn.Module):
                     x.contiguous(), self.w13_weight,
                     self.w13_weight_scale_inv, self.w2_weight,
                     self.w2_weight_scale_inv, weights, ids,
+                    expert_start=self.expert_start)
+            if (self.int4_cfg is not None and self.int4_cfg[1] == 128
+                    and x.shape[1] % 128 == 0
+                    and self.intermediate_per_rank % 128 == 0):
+                # packed-nibble grouped GEMM (w4a16, no dequant banks)
+                if getattr(self, "_i4_banks", None) is None:
+                    from gllm_amd.layers.quantization.int4 import \
+                        repack_canonical_moe
+                    w13c, w13sb, w2c, w2sb, _ = repack_canonical_moe(self)
+                    self._i4_banks = (w13c.to(x.device), w13sb.to(x.device),
+                                      w2c.to(x.device), w2sb.to(x.device))
+                w13c, w13sb, w2c, w2sb = self._i4_banks
+                return ops.fused_moe_int4(
+                    x.contiguous(), w13c, w13sb, w2c, w2sb, weights, ids,
                     expert_start=self.expert_start)
         T = x.shape[0]
         out = torch.zeros_like(x)

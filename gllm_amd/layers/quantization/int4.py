@@ -333,3 +333,27 @@ def repack_canonical(layer, dtype=torch.bfloat16):
     b = (-zeros.float() * scales).t().contiguous()         # [N, G]
     sb = torch.stack([s, b], dim=-1).contiguous()          # [N, G, 2]
     return wq4, sb, group
+
+
+def repack_canonical_moe(moe):
+    """Per-expert canonical repack of a FusedMoE's int4 banks for the
+    grouped w4a16 GEMM (ops/csrc/int4.hip::moe_gemm_int4):
+    returns (w13_c [E, 2I, H/2] u8, w13_sb [E, 2I, H/g, 2] f32,
+             w2_c [E, H, I/2], w2_sb [E, H, I/g, 2], group)."""
+    from types import SimpleNamespace
+    method, group = moe.int4_cfg
+    banks = []
+    for name in ("w13", "w2"):
+        cs, sbs = [], []
+        for e in range(moe.num_local_experts):
+            shim = SimpleNamespace(
+                int4_cfg=moe.int4_cfg,
+                qweight=getattr(moe, f"{name}_qweight")[e],
+                qzeros=getattr(moe, f"{name}_qzeros")[e],
+                scales=getattr(moe, f"{name}_scales")[e])
+            wq4, sb, _ = repack_canonical(shim)
+            cs.append(wq4)
+            sbs.append(sb)
+        banks.append((torch.stack(cs).contiguous(),
+                      torch.stack(sbs).contiguous()))
+    return banks[0][0], banks[0][1], banks[1][0], banks[1][1], group

@@ -347,6 +347,43 @@ def int4_linear(x: torch.Tensor, wq4: torch.Tensor, sb: torch.Tensor,
     return out
 
 
+def fused_moe_int4(x: torch.Tensor, w13_c, w13_sb, w2_c, w2_sb,
+                   topk_weights: torch.Tensor, topk_ids: torch.Tensor,
+                   expert_start: int = 0) -> torch.Tensor:
+    """int4 (w4a16) grouped-GEMM MoE: packed-nibble expert banks with
+    in-fragment dequant — no bf16 materialization at all."""
+    T, K = x.shape
+    E_local, two_i, _ = w13_c.shape
+    topk = topk_ids.shape[1]
+    n_pairs = T * topk
+    rpe = n_pairs / max(1, E_local)
+    block_m = 16 if rpe < 12 else (32 if rpe < 40 else 64)
+    cap = n_pairs + E_local * (block_m - 1) + 1
+    max_blocks = (n_pairs + block_m - 1) // block_m + E_local
+    dev = x.device
+    k = _gpu_kernels()
+    ids32 = topk_ids.int().contiguous()
+    sorted_ids = torch.empty(cap, dtype=torch.int32, device=dev)
+    expert_blocks = torch.empty(max_blocks, dtype=torch.int32, device=dev)
+    n_post = torch.empty(1, dtype=torch.int32, device=dev)
+    k.moe_align(ids32, E_local, expert_start, block_m, sorted_ids,
+                expert_blocks, n_post)
+    rows_pad = cap - 1 + block_m
+    inter1 = torch.empty(rows_pad, two_i, dtype=x.dtype, device=dev)
+    k.moe_gemm_int4(inter1, x.contiguous(), w13_c, w13_sb, sorted_ids,
+                    expert_blocks, n_post, None, n_pairs, topk, block_m,
+                    False)
+    act = silu_and_mul(inter1)
+    pair_out = x.new_zeros(n_pairs, K)
+    k.moe_gemm_int4(pair_out, act, w2_c, w2_sb, sorted_ids,
+                    expert_blocks, n_post,
+                    topk_weights.float().contiguous(), n_pairs, topk,
+                    block_m, True)
+    out = torch.empty(T, K, dtype=x.dtype, device=dev)
+    k.moe_sum(out, pair_out, topk)
+    return out
+
+
 # --------------------------------------------------------------- sampling
 def topk_topp_filter(probs: torch.Tensor, top_ks: torch.Tensor,
                      top_ps: torch.Tensor,

@@ -57,6 +57,11 @@ void moe_gemm_fp8(torch::Tensor C, torch::Tensor A, torch::Tensor As,
 void int4_skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor wq,
                       torch::Tensor sb, c10::optional<torch::Tensor> bias,
                       torch::Tensor workspace, long group);
+void moe_gemm_int4(torch::Tensor C, torch::Tensor A, torch::Tensor W,
+                   torch::Tensor SB, torch::Tensor sorted_ids,
+                   torch::Tensor expert_blocks, torch::Tensor n_post_pad,
+                   c10::optional<torch::Tensor> topk_weights, long n_pairs,
+                   long topk, long block_m, bool scatter);
 void gdn_conv_update(torch::Tensor out, torch::Tensor x,
                      torch::Tensor weight, torch::Tensor conv_state,
                      torch::Tensor slots);
@@ -113,6 +118,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "block-scale fp8 grouped MoE GEMM");
   m.def("int4_skinny_gemm", &int4_skinny_gemm,
         "fused-dequant int4 weight-streaming GEMM (decode)");
+  m.def("moe_gemm_int4", &moe_gemm_int4,
+        "fused-dequant int4 grouped MoE GEMM (w4a16)");
   m.def("gdn_conv_update", &gdn_conv_update,
         "batched causal-conv1d decode step w/ state roll");
   m.def("gdn_decode", &gdn_decode,

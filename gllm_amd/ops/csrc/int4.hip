@@ -197,6 +197,186 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_kernel(
   }
 }
 
+// ------------------------------------------------- int4 grouped MoE
+// moe.hip's grouped-GEMM pipeline with packed-nibble weights (w4a16 —
+// activations stay bf16, matching the reference's moe_wna16 kernels).
+// BK = 128 elems = one quant group per k tile (group size 128).
+constexpr int MOE_BN = 64;
+constexpr int MOE_BK = 128;
+
+template <int BM, bool SCATTER>
+__global__ __launch_bounds__(BLOCK) void moe_gemm_int4_kernel(
+    __hip_bfloat16 *__restrict__ C,
+    const __hip_bfloat16 *__restrict__ A,   // [Ta, K] bf16
+    const unsigned char *__restrict__ W,    // [E, Nd, K/2] packed
+    const float *__restrict__ SB,           // [E, Nd, K/128, 2]
+    const int *__restrict__ sorted_ids, const int *__restrict__ expert_blocks,
+    const int *__restrict__ n_post_pad,
+    const float *__restrict__ topk_w, int n_pairs, int K, int Nd,
+    int topk) {
+  const int mb = blockIdx.x;
+  if (mb * BM >= n_post_pad[0]) return;
+  const int nb = blockIdx.y;
+  const int e = expert_blocks[mb];
+  const int kgroups = K / MOE_BK;
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int l16 = lane & 15;
+  const int lhi = lane >> 4;
+
+  constexpr int WM = BM / 16;
+  constexpr int WN_TILES = WM;
+  const int wm = wave % WM;
+  const int wn = wave / WM;
+
+  constexpr int APAD = 8;
+  __shared__ __hip_bfloat16 a_tile[2][BM * (MOE_BK + APAD)];
+  __shared__ unsigned char w_tile[2][MOE_BN * (MOE_BK / 2 + 8)];
+  __shared__ float sb_tile[2][MOE_BN * 2];
+
+  const int g0 = mb * BM;
+  constexpr int A_CH = BM * MOE_BK * 2 / 16;       // 16-B chunks (bf16)
+  constexpr int W_CH = MOE_BN * (MOE_BK / 2) / 16; // 16-B chunks (nibbles)
+  typedef __attribute__((ext_vector_type(4))) int int4v;
+  int4v areg[(A_CH + BLOCK - 1) / BLOCK];
+  int4v wreg[(W_CH + BLOCK - 1) / BLOCK];
+  float sreg[2];  // (s, b) staged by the first MOE_BN*2 threads
+
+  const long w_base = (long)e * Nd * (K / 2);
+  const long sb_base = (long)e * Nd * kgroups * 2;
+
+  auto load_tiles = [&](int kb) {
+#pragma unroll
+    for (int it = 0; it < (A_CH + BLOCK - 1) / BLOCK; ++it) {
+      const int idx = tid + it * BLOCK;
+      if (idx < A_CH) {
+        const int m = idx / (MOE_BK * 2 / 16);
+        const int c = idx % (MOE_BK * 2 / 16);
+        const int g = g0 + m;
+        const int pair = sorted_ids[g];
+        const bool valid = pair < n_pairs;
+        long arow = SCATTER ? g : (valid ? pair / topk : 0);
+        if (valid)
+          areg[it] = *reinterpret_cast<const int4v *>(
+              reinterpret_cast<const char *>(A + arow * (long)K +
+                                             kb * MOE_BK) + c * 16);
+        else
+          areg[it] = int4v{0, 0, 0, 0};
+      }
+    }
+#pragma unroll
+    for (int it = 0; it < (W_CH + BLOCK - 1) / BLOCK; ++it) {
+      const int idx = tid + it * BLOCK;
+      if (idx < W_CH) {
+        const int n = idx / (MOE_BK / 2 / 16);
+        const int c = idx % (MOE_BK / 2 / 16);
+        if (nb * MOE_BN + n < Nd)
+          wreg[it] = *reinterpret_cast<const int4v *>(
+              W + w_base + (long)(nb * MOE_BN + n) * (K / 2) +
+              kb * (MOE_BK / 2) + c * 16);
+        else
+          wreg[it] = int4v{0, 0, 0, 0};
+      }
+    }
+    if (tid < MOE_BN * 2) {
+      const int n = tid >> 1;
+      const int which = tid & 1;
+      const int nn = min(nb * MOE_BN + n, Nd - 1);
+      sreg[0] = SB[sb_base + ((long)nn * kgroups + kb) * 2 + which];
+    }
+  };
+  auto write_tiles = [&](int buf) {
+#pragma unroll
+    for (int it = 0; it < (A_CH + BLOCK - 1) / BLOCK; ++it) {
+      const int idx = tid + it * BLOCK;
+      if (idx < A_CH) {
+        const int m = idx / (MOE_BK * 2 / 16);
+        const int c = idx % (MOE_BK * 2 / 16);
+        *reinterpret_cast<int4v *>(
+            reinterpret_cast<char *>(&a_tile[buf][m * (MOE_BK + APAD)]) +
+            c * 16) = areg[it];
+      }
+    }
+#pragma unroll
+    for (int it = 0; it < (W_CH + BLOCK - 1) / BLOCK; ++it) {
+      const int idx = tid + it * BLOCK;
+      if (idx < W_CH) {
+        const int n = idx / (MOE_BK / 2 / 16);
+        const int c = idx % (MOE_BK / 2 / 16);
+        *reinterpret_cast<int4v *>(
+            &w_tile[buf][n * (MOE_BK / 2 + 8) + c * 16]) = wreg[it];
+      }
+    }
+    if (tid < MOE_BN * 2) sb_tile[buf][tid] = sreg[0];
+  };
+
+  i4_f4 acc[WN_TILES];
+#pragma unroll
+  for (int t = 0; t < WN_TILES; ++t) acc[t] = i4_f4{0, 0, 0, 0};
+
+  load_tiles(0);
+  write_tiles(0);
+  __syncthreads();
+
+  int cur = 0;
+  const int nkb = K / MOE_BK;
+  for (int kb = 0; kb < nkb; ++kb) {
+    if (kb + 1 < nkb) load_tiles(kb + 1);
+
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < MOE_BK / 32; ++ks) {
+      i4_bf8 afrag = *reinterpret_cast<const i4_bf8 *>(
+          &a_tile[cur][(wm * 16 + l16) * (MOE_BK + APAD) + ks * 32 +
+                       lhi * 8]);
+#pragma unroll
+      for (int t = 0; t < WN_TILES; ++t) {
+        const int brow = (wn * WN_TILES + t) * 16 + l16;
+        const unsigned int w4 = *reinterpret_cast<const unsigned int *>(
+            &w_tile[cur][brow * (MOE_BK / 2 + 8) + (ks * 32 + lhi * 8) / 2]);
+        const float s = sb_tile[cur][brow * 2];
+        const float bz = sb_tile[cur][brow * 2 + 1];
+        i4_bf8 bfrag;
+        __hip_bfloat16 *be = reinterpret_cast<__hip_bfloat16 *>(&bfrag);
+#pragma unroll
+        for (int ee = 0; ee < 8; ++ee) {
+          const float qv = (float)((w4 >> (4 * ee)) & 0xF);
+          be[ee] = __float2bfloat16(qv * s + bz);
+        }
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag, acc[t], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    __syncthreads();
+    if (kb + 1 < nkb) {
+      write_tiles(cur ^ 1);
+      cur ^= 1;
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = wm * 16 + lhi * 4 + r;
+    const int g = g0 + m;
+    const int pair = sorted_ids[g];
+    if (pair >= n_pairs) continue;
+    long crow = SCATTER ? (long)pair : (long)g;
+    float scalew = 1.f;
+    if (SCATTER && topk_w != nullptr) scalew = topk_w[pair];
+#pragma unroll
+    for (int t = 0; t < WN_TILES; ++t) {
+      const int n = nb * MOE_BN + (wn * WN_TILES + t) * 16 + l16;
+      if (n < Nd)
+        C[crow * (long)Nd + n] = __float2bfloat16(acc[t][r] * scalew);
+    }
+  }
+}
+
 __global__ void int4_reduce_kernel(__hip_bfloat16 *__restrict__ out,
                                    const float *__restrict__ partial,
                                    const float *__restrict__ bias, int M,
@@ -212,6 +392,46 @@ __global__ void int4_reduce_kernel(__hip_bfloat16 *__restrict__ out,
 }
 
 }  // namespace
+
+void moe_gemm_int4(torch::Tensor C, torch::Tensor A, torch::Tensor W,
+                   torch::Tensor SB, torch::Tensor sorted_ids,
+                   torch::Tensor expert_blocks, torch::Tensor n_post_pad,
+                   c10::optional<torch::Tensor> topk_weights, long n_pairs,
+                   long topk, long block_m, bool scatter) {
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16 && A.is_contiguous());
+  TORCH_CHECK(W.scalar_type() == at::kByte && W.is_contiguous());
+  TORCH_CHECK(SB.scalar_type() == at::kFloat && SB.is_contiguous());
+  const int Nd = W.size(1);
+  const int K = W.size(2) * 2;
+  TORCH_CHECK(A.size(-1) == K);
+  TORCH_CHECK(K % MOE_BK == 0, "int4 moe: K must be a multiple of 128");
+  const int max_blocks = expert_blocks.numel();
+  const int n_tiles = (Nd + MOE_BN - 1) / MOE_BN;
+  const float *tw = nullptr;
+  if (topk_weights.has_value()) tw = topk_weights->data_ptr<float>();
+  auto stream = at::cuda::getCurrentCUDAStream();
+#define LAUNCH_MI4(BM, SC)                                                  \
+  hipLaunchKernelGGL((moe_gemm_int4_kernel<BM, SC>),                        \
+                     dim3(max_blocks, n_tiles), dim3(BLOCK), 0, stream,     \
+                     (__hip_bfloat16 *)C.data_ptr(),                        \
+                     (const __hip_bfloat16 *)A.data_ptr(),                  \
+                     W.data_ptr<unsigned char>(), SB.data_ptr<float>(),     \
+                     sorted_ids.data_ptr<int>(),                            \
+                     expert_blocks.data_ptr<int>(),                         \
+                     n_post_pad.data_ptr<int>(), tw, (int)n_pairs, K, Nd,   \
+                     (int)topk)
+  if (block_m == 16) {
+    if (scatter) LAUNCH_MI4(16, true); else LAUNCH_MI4(16, false);
+  } else if (block_m == 32) {
+    if (scatter) LAUNCH_MI4(32, true); else LAUNCH_MI4(32, false);
+  } else if (block_m == 64) {
+    if (scatter) LAUNCH_MI4(64, true); else LAUNCH_MI4(64, false);
+  } else {
+    TORCH_CHECK(false, "moe_gemm_int4: block_m must be 16/32/64");
+  }
+#undef LAUNCH_MI4
+  HIP_CHECK_KERNEL();
+}
 
 void int4_skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor wq,
                       torch::Tensor sb, c10::optional<torch::Tensor> bias,

@@ -81,12 +81,14 @@ class ModelRunner:
                                  and ops.has_kernels()
                                  and all((m.fp8_block is None
                                           or m.fp8_block == (128, 128))
-                                         and m.int4_cfg is None
+                                         and (m.int4_cfg is None
+                                              or m.int4_cfg[1] == 128)
                                          for m in moe_mods)):
-                # bf16 and fp8(128-block) MoE on GPU run the device-
-                # resident grouped MFMA GEMM pipelines (ops.fused_moe /
-                # fused_moe_fp8) and are capture-safe; int4 banks still
-                # run the host segment loop (illegal sync under capture)
+                # bf16, fp8(128-block) and int4(group-128) MoE on GPU
+                # run the device-resident grouped MFMA GEMM pipelines
+                # (ops.fused_moe / fused_moe_fp8 / fused_moe_int4) and
+                # are capture-safe; other quant shapes keep the host
+                # segment loop (illegal sync under capture)
                 cfg.use_graph = False
         num_pages = num_pages_override or self._size_kv_cache()
         self._allocate_kv(num_pages)

@@ -536,3 +536,37 @@ def test_int4_linear(kernels, case):
     out = ops.int4_linear(x, wq4.cuda(), sb.cuda(), grp)
     ref = x.float().cpu() @ ref_w.T
     assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=2e-3)
+
+
+def test_fused_moe_int4(kernels):
+    """w4a16 grouped MoE vs the dequantized bf16 reference."""
+    from types import SimpleNamespace
+    from gllm_amd import ops
+    from gllm_amd.layers.quantization import int4 as qi4
+    torch.manual_seed(13)
+    T, E, topk, H, I = 37, 4, 2, 256, 256
+    w13q, w2q, w13d, w2d = [], [], [], []
+    for e in range(E):
+        a = torch.randn(2 * I, H) / math.sqrt(H)
+        b = torch.randn(H, I) / math.sqrt(I)
+        qa = qi4.pack_gptq(a, group_size=128)
+        qb = qi4.pack_gptq(b, group_size=128)
+        w13q.append(qa)
+        w2q.append(qb)
+        w13d.append(qi4.dequant_gptq(*qa, 128, torch.bfloat16))
+        w2d.append(qi4.dequant_gptq(*qb, 128, torch.bfloat16))
+    moe = SimpleNamespace(
+        int4_cfg=("gptq", 128), num_local_experts=E,
+        w13_qweight=[q[0] for q in w13q], w13_qzeros=[q[1] for q in w13q],
+        w13_scales=[q[2] for q in w13q],
+        w2_qweight=[q[0] for q in w2q], w2_qzeros=[q[1] for q in w2q],
+        w2_scales=[q[2] for q in w2q])
+    w13c, w13sb, w2c, w2sb, grp = qi4.repack_canonical_moe(moe)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device="cuda") / 4
+    logits = torch.randn(T, E, device="cuda")
+    weights, ids = ops.topk_softmax(logits, topk, True)
+    out = ops.fused_moe_int4(x, w13c.cuda(), w13sb.cuda(), w2c.cuda(),
+                             w2sb.cuda(), weights, ids.int())
+    ref = _moe_ref(x, torch.stack(w13d).cuda(), torch.stack(w2d).cuda(),
+                   weights, ids)
+    assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=3e-3)
