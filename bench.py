@@ -192,7 +192,7 @@ def main():
                          "each stage")
     ap.add_argument("--page-size", type=int, default=16)
     ap.add_argument("--quant", type=str, default=None,
-                    choices=["fp8"],
+                    choices=["fp8", "int4"],
                     help="block-quantize the dummy weights (fp8 e4m3 "
                          "128x128 blocks) and run the native fp8 path")
     ap.add_argument("--schedule", type=str, default="token_throttling")
@@ -219,6 +219,11 @@ def main():
         model_json["quantization_config"] = {
             "quant_method": "fp8", "weight_block_size": [128, 128]}
         model_name += "-fp8"
+    elif args.quant == "int4":
+        model_json = dict(model_json)
+        model_json["quantization_config"] = {
+            "quant_method": "gptq", "bits": 4, "group_size": 128}
+        model_name += "-int4"
     model_dir = write_model_dir(model_json)
     assert n % args.tp == 0, (n, args.tp)
     pp = n // args.tp
@@ -343,7 +348,8 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": ("fp8-w8a8" if args.quant == "fp8" else "bf16") if use_gpu else "fp32",
+            "dtype": ({"fp8": "fp8-w8a8", "int4": "int4-w4a16"}.get(
+                args.quant) or "bf16") if use_gpu else "fp32",
             "data": "synthetic",
             "ttft_p50_ms": round(ttft_p50, 1) if ttft_p50 else None,
             "ttft_qps": args.qps if args.qps > 0 else "burst",

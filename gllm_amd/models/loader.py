@@ -180,6 +180,11 @@ def dummy_init(model: torch.nn.Module, seed: int = 0) -> None:
                 p.data.copy_(tmp.clamp(-0.4, 0.4).to(torch.float8_e4m3fn))
             elif "scale_inv" in gname:
                 p.data.fill_(0.05)
+            elif p.dtype in (torch.int32, torch.uint8):
+                # packed int4 qweight/qzeros: random bits
+                p.data.random_(generator=gen)
+            elif "scales" in gname and p.dim() >= 2:
+                p.data.fill_(0.01)
             elif p.dim() >= 2:
                 # draw in-place on device (32B params via a CPU RNG would
                 # take minutes); bf16 normal_ is supported on ROCm
@@ -214,7 +219,8 @@ def load_model(engine_config, device: str = "cpu"):
     qcfg = getattr(cfg, "quantization_config", None)
     if qcfg is not None and not isinstance(qcfg, dict):
         qcfg = getattr(qcfg, "to_dict", lambda: vars(qcfg))()
-    if qcfg and engine_config.load_format != "dummy":
+    if qcfg:  # dummy loads convert too: quantized EXECUTION
+        # with random weights is exactly what bench --quant tests
         method = qcfg.get("quant_method")
         if method == "fp8":
             from gllm_amd.layers.quantization.fp8 import \
