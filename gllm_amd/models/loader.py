@@ -239,6 +239,10 @@ def load_model(engine_config, device: str = "cpu"):
                                        qcfg.get("q_group_size", 128))})
             logger.info("%s int4 checkpoint: converted %d linears",
                         method, n)
+    if qcfg:
+        # quant converters create replacement params on CPU; re-home
+        # them next to the rest of the model
+        model = model.to(device)
     if engine_config.load_format == "dummy":
         dummy_init(model, engine_config.seed)
     else:
