@@ -55,13 +55,21 @@ class LinearBase(nn.Module):
         int4 keep the dequant path."""
         w = self.weight
         if self.fp8_block is not None:
-            if (x.is_cuda and x.dim() == 2 and x.dtype == torch.bfloat16
-                    and x.shape[0] <= 256 and x.shape[1] % 128 == 0
-                    and self.fp8_block == (128, 128) and ops.has_kernels()):
-                return ops.fp8_linear(x, w, self.weight_scale_inv, bias)
+            native = (x.is_cuda and x.dim() == 2
+                      and x.dtype == torch.bfloat16
+                      and x.shape[1] % 128 == 0
+                      and self.fp8_block == (128, 128)
+                      and ops.has_kernels())
+            if native:
+                b32 = self._bias32()
+                if x.shape[0] <= 256:
+                    return ops.fp8_linear(x, w, self.weight_scale_inv, b32)
+                # prefill: identity-routed grouped fp8 GEMM (wide
+                # BM64xBN256 tile) — weights stay e4m3, no dequant
+                return ops.fp8_prefill_linear(x, w, self.weight_scale_inv,
+                                              self.bias)
             from gllm_amd.layers.quantization.fp8 import dequant_block_fp8
             if x.is_cuda:
-                # prefill burst: dequant per call, nothing cached
                 wd = dequant_block_fp8(w, self.weight_scale_inv,
                                        self.fp8_block, x.dtype)
                 return ops.linear(x, wd, bias)
@@ -71,21 +79,26 @@ class LinearBase(nn.Module):
                     x.dtype).to(x.device)
             w = self._w_dq
         elif self.int4_cfg is not None:
-            if (x.is_cuda and x.dim() == 2 and x.dtype == torch.bfloat16
-                    and x.shape[0] <= 256 and x.shape[1] % 256 == 0
-                    and self.int4_cfg[1] == 128 and ops.has_kernels()):
-                # decode: fused-dequant nibble-streaming GEMM (weights
-                # stay packed; one-time canonical repack)
+            native = (x.is_cuda and x.dim() == 2
+                      and x.dtype == torch.bfloat16
+                      and x.shape[1] % 256 == 0
+                      and self.int4_cfg[1] == 128 and ops.has_kernels())
+            if native:
+                # packed-nibble streaming GEMMs (one-time canonical
+                # repack; sbt = transposed (scale,bias) for the skinny)
                 if getattr(self, "_i4_canon", None) is None:
                     from gllm_amd.layers.quantization.int4 import \
                         repack_canonical
                     wq4, sb, grp = repack_canonical(self)
-                    self._i4_canon = (wq4.to(x.device),
-                                      sb.to(x.device), grp)
-                wq4, sb, grp = self._i4_canon
-                return ops.int4_linear(x, wq4, sb, grp, bias)
+                    sbt = sb.permute(1, 2, 0).contiguous()
+                    self._i4_canon = (wq4.to(x.device), sb.to(x.device),
+                                      sbt.to(x.device), grp)
+                wq4, sb, sbt, grp = self._i4_canon
+                if x.shape[0] <= 256:
+                    return ops.int4_linear(x, wq4, sbt, grp,
+                                           self._bias32())
+                return ops.int4_prefill_linear(x, wq4, sb, self.bias)
             if x.is_cuda:
-                # prefill burst: dequant per call, nothing cached
                 from gllm_amd.layers.quantization.int4 import dequant_layer
                 return ops.linear(x, dequant_layer(self, x.dtype), bias)
             if self._w_dq is None:
@@ -93,6 +106,14 @@ class LinearBase(nn.Module):
                 self._w_dq = dequant_layer(self, x.dtype).to(x.device)
             w = self._w_dq
         return ops.linear(x, w, bias)
+
+    def _bias32(self):
+        """fp32 bias cached once (the skinny epilogues add fp32)."""
+        if self.bias is None:
+            return None
+        if getattr(self, "_bias_f32", None) is None:
+            self._bias_f32 = self.bias.float()
+        return self._bias_f32
 
 
 class ReplicatedLinear(LinearBase):

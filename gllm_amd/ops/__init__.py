@@ -195,16 +195,24 @@ def fused_moe(x: torch.Tensor, w13: torch.Tensor, w2: torch.Tensor,
 
 # --------------------------------------------------------------- fp8
 def per_token_group_quant_fp8(x: torch.Tensor, group: int = 128,
-                              ue8m0: bool = False):
+                              ue8m0: bool = False,
+                              transposed: bool = False):
     """[T, K] bf16 -> (e4m3 [T, K], fp32 scales [T, K/group]).
-    GPU kernel; CPU callers use the torch math in quantization/fp8.py."""
+    GPU kernel; CPU callers use the torch math in quantization/fp8.py.
+    ``transposed=True`` additionally returns scales_t [K/group, T]
+    (column-major) — the skinny GEMM stages a k-group's scales as ONE
+    contiguous run instead of a strided 64-line gather."""
     assert x.is_cuda and group == 128
     T, K = x.shape
     q = torch.empty(T, K, dtype=torch.float8_e4m3fn, device=x.device)
     scales = torch.empty(T, K // group, dtype=torch.float32,
                          device=x.device)
+    st = torch.empty(K // group, T, dtype=torch.float32,
+                     device=x.device) if transposed else None
     _gpu_kernels().per_token_group_quant_fp8(
-        x.contiguous(), q.view(torch.uint8), scales, ue8m0)
+        x.contiguous(), q.view(torch.uint8), scales, st, ue8m0)
+    if transposed:
+        return q, scales, st
     return q, scales
 
 
@@ -218,7 +226,8 @@ def fp8_linear(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
     MFMA. Weights stream at 1 B/elem — ~2x the bf16 decode rate."""
     M, K = x.shape
     N = w_q.shape[0]
-    aq, as_ = per_token_group_quant_fp8(x.contiguous())
+    aq, _, ast = per_token_group_quant_fp8(x.contiguous(),
+                                           transposed=True)
     splitk = _skinny_splitk(M, N, K)
     need = splitk * M * N
     key = x.device.index or 0
@@ -230,9 +239,74 @@ def fp8_linear(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
         _FP8_WS[key] = ws
     out = torch.empty(M, N, dtype=x.dtype, device=x.device)
     _gpu_kernels().fp8_skinny_gemm(
-        out, aq.view(torch.uint8), as_, w_q.view(torch.uint8),
+        out, aq.view(torch.uint8), ast, w_q.view(torch.uint8),
         w_scale.contiguous(),
         bias.float() if bias is not None else None, ws, splitk)
+    return out
+
+
+_IDENT: dict = {}
+
+
+def _identity_tables(device, m_pad: int, block: int):
+    """Routing tables that make the grouped MoE GEMM a plain dense
+    GEMM: sorted_ids = arange (row i of the batch is 'pair' i, topk=1),
+    every block belongs to expert 0. Cached per (device, m_pad) — the
+    graph warmup pass populates the cache so capture never allocates."""
+    key = (device.index or 0, m_pad, block)
+    ent = _IDENT.get(key)
+    if ent is None:
+        if torch.cuda.is_current_stream_capturing():
+            raise RuntimeError("identity tables built during graph capture")
+        ids = torch.arange(m_pad, dtype=torch.int32, device=device)
+        blocks = torch.zeros(m_pad // block, dtype=torch.int32,
+                             device=device)
+        n_post = torch.full((1,), m_pad, dtype=torch.int32, device=device)
+        ent = (ids, blocks, n_post)
+        _IDENT[key] = ent
+    return ent
+
+
+def fp8_prefill_linear(x: torch.Tensor, w_q: torch.Tensor,
+                       w_scale: torch.Tensor,
+                       bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Dense fp8 GEMM for M > 256: the grouped MoE fp8 kernel (wide
+    BM64xBN256 prefill tile) with identity routing — weights stay e4m3,
+    no per-call dequant materialization."""
+    M, K = x.shape
+    N = w_q.shape[0]
+    aq, as_ = per_token_group_quant_fp8(x.contiguous())
+    m_pad = (M + 63) // 64 * 64
+    ids, blocks, n_post = _identity_tables(x.device, m_pad, 64)
+    out_pad = torch.empty(m_pad, N, dtype=x.dtype, device=x.device)
+    _gpu_kernels().moe_gemm_fp8(
+        out_pad, aq.view(torch.uint8), as_,
+        w_q.view(torch.uint8).view(1, N, K),
+        w_scale.contiguous().view(1, w_scale.shape[0], w_scale.shape[1]),
+        ids, blocks, n_post, None, M, 1, 164, False)
+    out = out_pad.narrow(0, 0, M)
+    if bias is not None:
+        out = out.add_(bias)
+    return out
+
+
+def int4_prefill_linear(x: torch.Tensor, wq4: torch.Tensor,
+                        sb: torch.Tensor,
+                        bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Dense int4 (w4a16) GEMM for M > 256 via the grouped MoE int4
+    kernel with identity routing — packed nibbles stream directly."""
+    M, K2 = x.shape[0], wq4.shape[1]
+    N = wq4.shape[0]
+    m_pad = (M + 63) // 64 * 64
+    ids, blocks, n_post = _identity_tables(x.device, m_pad, 64)
+    out_pad = torch.empty(m_pad, N, dtype=x.dtype, device=x.device)
+    _gpu_kernels().moe_gemm_int4(
+        out_pad, x.contiguous(), wq4.view(1, N, K2),
+        sb.view(1, N, -1, 2).contiguous() if sb.dim() == 3 else sb,
+        ids, blocks, n_post, None, M, 1, 64, False)
+    out = out_pad.narrow(0, 0, M)
+    if bias is not None:
+        out = out.add_(bias)
     return out
 
 
@@ -324,11 +398,13 @@ def rmsnorm_gated(x: torch.Tensor, z: torch.Tensor, weight: torch.Tensor,
     return out
 
 
-def int4_linear(x: torch.Tensor, wq4: torch.Tensor, sb: torch.Tensor,
+def int4_linear(x: torch.Tensor, wq4: torch.Tensor, sbt: torch.Tensor,
                 group: int,
                 bias: Optional[torch.Tensor] = None) -> torch.Tensor:
     """out = x @ dequant(wq4).T via the fused-dequant int4 skinny GEMM
-    (M <= 256): packed nibbles stream at 0.5 B/elem."""
+    (M <= 256): packed nibbles stream at 0.5 B/elem. ``sbt`` is the
+    TRANSPOSED (scale, bias) bank [K/group, 2, N] (lane-contiguous
+    staging; repack_canonical builds it)."""
     M, K = x.shape
     N = wq4.shape[0]
     splitk = _skinny_splitk(M, N, K)
@@ -342,7 +418,7 @@ def int4_linear(x: torch.Tensor, wq4: torch.Tensor, sb: torch.Tensor,
         _FP8_WS[key] = ws
     out = torch.empty(M, N, dtype=x.dtype, device=x.device)
     _gpu_kernels().int4_skinny_gemm(
-        out, x.contiguous(), wq4, sb,
+        out, x.contiguous(), wq4, sbt,
         bias.float() if bias is not None else None, ws, group)
     return out
 

@@ -43,7 +43,9 @@ void moe_gemm(torch::Tensor C, torch::Tensor A, torch::Tensor W,
               long topk, long block_m, bool scatter);
 void moe_sum(torch::Tensor out, torch::Tensor pair_out, long topk);
 void per_token_group_quant_fp8(torch::Tensor x, torch::Tensor q,
-                               torch::Tensor scales, bool ue8m0);
+                               torch::Tensor scales,
+                               c10::optional<torch::Tensor> scales_t,
+                               bool ue8m0);
 void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor as,
                      torch::Tensor w, torch::Tensor ws,
                      c10::optional<torch::Tensor> bias,

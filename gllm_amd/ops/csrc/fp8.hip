@@ -42,8 +42,8 @@ DEV_INLINE float fp8_to_f32(unsigned char v) {
 // (token, group): 2 elems per lane.
 __global__ void per_token_group_quant_kernel(
     const __hip_bfloat16 *__restrict__ x, unsigned char *__restrict__ q,
-    float *__restrict__ scales, long n_groups, int k_groups, int K,
-    bool ue8m0) {
+    float *__restrict__ scales, float *__restrict__ scales_t,
+    long n_groups, int k_groups, int K, bool ue8m0) {
   const long g0 = blockIdx.x * (long)(blockDim.x / 64) + (threadIdx.x >> 6);
   if (g0 >= n_groups) return;
   const int lane = threadIdx.x & 63;
@@ -64,7 +64,14 @@ __global__ void per_token_group_quant_kernel(
   unsigned char *dst = q + t * (long)K + kg * GROUP + lane * 2;
   dst[0] = q0.__x;
   dst[1] = q1.__x;
-  if (lane == 0) scales[t * (long)k_groups + kg] = scale;
+  if (lane == 0) {
+    scales[t * (long)k_groups + kg] = scale;
+    // column-major copy [k_groups, T] for the skinny GEMM: a stage's
+    // M scales become ONE contiguous glds line-run instead of a
+    // 64-line strided gather inside the counted-vmcnt ring
+    if (scales_t)
+      scales_t[(long)kg * (n_groups / k_groups) + t] = scale;
+  }
 }
 
 // ------------------------------------------------- skinny fp8 GEMM
@@ -81,10 +88,11 @@ template <int MB, int RING>
 __global__ __launch_bounds__(BLOCK) void fp8_skinny_kernel(
     float *__restrict__ partial,           // [SPLITK, M, N]
     const unsigned char *__restrict__ aq,  // [M, K] e4m3
-    const float *__restrict__ as,          // [M, K/128]
+    const float *__restrict__ ast,         // [K/128, M] (transposed)
     const unsigned char *__restrict__ w,   // [N, K] e4m3
     const float *__restrict__ ws,          // [N/128, K/128]
-    int M, int N, int K, int k_slice) {
+    __hip_bfloat16 *__restrict__ out,      // non-null iff splitk == 1
+    const float *__restrict__ bias, int M, int N, int K, int k_slice) {
   constexpr int TILE_B = (BN + MB * 64) * ROW_B;
   const int n0 = blockIdx.x * BN;
   const int z = blockIdx.y;
@@ -147,14 +155,17 @@ __global__ __launch_bounds__(BLOCK) void fp8_skinny_kernel(
       else
         glds16<0>(gsrc[j] + kb, base + off);
     }
-    // every wave issues the identical scale gather (same values land
-    // at the same LDS bytes) so per-wave vm counters stay uniform
+    // every wave issues the identical scale load (same values land
+    // at the same LDS bytes) so per-wave vm counters stay uniform.
+    // ast is [kgroups, M]: the stage's M scales are CONTIGUOUS — a
+    // handful of cache lines instead of the 64-line strided gather
+    // that used to double the ring's line-request rate.
 #pragma unroll
     for (int mb = 0; mb < MB; ++mb) {
       const int m = min(mb * 64 + lane, M - 1);
       __builtin_amdgcn_global_load_lds(
           reinterpret_cast<const unsigned int *>(
-              as + (long)m * kgroups + kb_begin + kt),
+              ast + (long)(kb_begin + kt) * M + m),
           reinterpret_cast<unsigned int *>(base + ASC_OFF + mb * 256), 4,
           0, 0);
     }
@@ -226,6 +237,27 @@ __global__ __launch_bounds__(BLOCK) void fp8_skinny_kernel(
     __builtin_amdgcn_s_barrier();
   }
 
+  if (out != nullptr) {
+    // splitk == 1: write bf16 directly (+bias) — skips the separate
+    // fp32-partial round-trip and the reduce launch entirely
+#pragma unroll
+    for (int mb = 0; mb < MB; ++mb) {
+#pragma unroll
+      for (int nt = 0; nt < BN / 16; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = mb * 64 + wave * 16 + lhi * 4 + r;
+          const int n = n0 + nt * 16 + l16;
+          if (m < M && n < N) {
+            float v = acc[mb][nt][r];
+            if (bias) v += bias[n];
+            out[(long)m * N + n] = __float2bfloat16(v);
+          }
+        }
+      }
+    }
+    return;
+  }
   float *base = partial + (long)z * M * N;
 #pragma unroll
   for (int mb = 0; mb < MB; ++mb) {
@@ -439,7 +471,9 @@ __global__ __launch_bounds__(BLOCK) void moe_gemm_fp8_kernel(
 }  // namespace
 
 void per_token_group_quant_fp8(torch::Tensor x, torch::Tensor q,
-                               torch::Tensor scales, bool ue8m0) {
+                               torch::Tensor scales,
+                               c10::optional<torch::Tensor> scales_t,
+                               bool ue8m0) {
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 && x.is_contiguous());
   TORCH_CHECK(q.is_contiguous() && scales.is_contiguous());
   const long T = x.size(0);
@@ -447,6 +481,12 @@ void per_token_group_quant_fp8(torch::Tensor x, torch::Tensor q,
   TORCH_CHECK(K % GROUP == 0, "K must be a multiple of 128");
   const int kg = K / GROUP;
   const long n_groups = T * kg;
+  float *st = nullptr;
+  if (scales_t.has_value()) {
+    TORCH_CHECK(scales_t->is_contiguous() &&
+                scales_t->numel() == n_groups);
+    st = scales_t->data_ptr<float>();
+  }
   const int wpb = 4;  // waves per block
   const long grid = (n_groups + wpb - 1) / wpb;
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -454,11 +494,11 @@ void per_token_group_quant_fp8(torch::Tensor x, torch::Tensor q,
                      dim3(wpb * 64), 0, stream,
                      (const __hip_bfloat16 *)x.data_ptr(),
                      (unsigned char *)q.data_ptr(),
-                     scales.data_ptr<float>(), n_groups, kg, K, ue8m0);
+                     scales.data_ptr<float>(), st, n_groups, kg, K, ue8m0);
   HIP_CHECK_KERNEL();
 }
 
-void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor as,
+void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor ast,
                      torch::Tensor w, torch::Tensor ws,
                      c10::optional<torch::Tensor> bias,
                      torch::Tensor workspace, long splitk_arg) {
@@ -467,6 +507,8 @@ void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor as,
   TORCH_CHECK(K % BK == 0, "fp8 gemm: K must be a multiple of 128");
   TORCH_CHECK(K / BK <= 256, "fp8 skinny: K <= 32768 (ws_lds)");
   TORCH_CHECK(M <= 256, "fp8 skinny: M <= 256");
+  TORCH_CHECK(ast.is_contiguous() && ast.numel() == (long)M * (K / BK),
+              "fp8 skinny: transposed scales [K/128, M]");
   const int n_wg = (N + BN - 1) / BN;
   int splitk = (int)splitk_arg;
   if (splitk <= 0) {
@@ -477,24 +519,28 @@ void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor as,
   int k_slice = (K + splitk - 1) / splitk;
   k_slice = ((k_slice + BK - 1) / BK) * BK;
   splitk = (K + k_slice - 1) / k_slice;
-  TORCH_CHECK(workspace.numel() >= (long)splitk * M * N);
+  TORCH_CHECK(splitk == 1 || workspace.numel() >= (long)splitk * M * N);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto *wsp = workspace.data_ptr<float>();
   auto *ap = (const unsigned char *)aq.data_ptr();
-  auto *asp = as.data_ptr<float>();
+  auto *asp = ast.data_ptr<float>();
   auto *wp = (const unsigned char *)w.data_ptr();
   auto *wsc = ws.data_ptr<float>();
+  const float *bias_ptr = nullptr;
+  if (bias.has_value()) bias_ptr = bias->data_ptr<float>();
+  // splitk == 1: the GEMM kernel writes bf16 (+bias) itself
+  __hip_bfloat16 *outp =
+      splitk == 1 ? (__hip_bfloat16 *)out.data_ptr() : nullptr;
 #define LAUNCH_SK(MB, RING)                                                  \
   hipLaunchKernelGGL((fp8_skinny_kernel<MB, RING>),                          \
                      dim3(n_wg, splitk), dim3(BLOCK), 0, stream, wsp, ap,    \
-                     asp, wp, wsc, M, N, K, k_slice)
+                     asp, wp, wsc, outp, bias_ptr, M, N, K, k_slice)
   if (M <= 64) LAUNCH_SK(1, 3);
   else if (M <= 128) LAUNCH_SK(2, 3);
   else LAUNCH_SK(4, 2);
 #undef LAUNCH_SK
   HIP_CHECK_KERNEL();
-  const float *bias_ptr = nullptr;
-  if (bias.has_value()) bias_ptr = bias->data_ptr<float>();
+  if (splitk == 1) return;
   const long total = (long)M * N;
   const long grid = std::min<long>((total + 1023) / 1024, 2048);
   hipLaunchKernelGGL(fp8_reduce_kernel, dim3(grid), dim3(256), 0, stream,

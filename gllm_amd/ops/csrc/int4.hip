@@ -46,7 +46,9 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_kernel(
     float *__restrict__ partial,            // [SPLITK, M, N]
     const __hip_bfloat16 *__restrict__ x,   // [M, K]
     const unsigned char *__restrict__ wq,   // [N, K/2]
-    const float *__restrict__ sb,           // [N, K/group, 2]
+    const float *__restrict__ sbt,          // [K/group, 2, N] transposed
+    __hip_bfloat16 *__restrict__ out,       // non-null iff splitk == 1
+    const float *__restrict__ bias,
     int M, int N, int K, int k_slice, int group) {
   constexpr int TILE_B = BN * WROW_B + MB * 64 * XROW_B;
   const int n0 = blockIdx.x * BN;
@@ -106,17 +108,18 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_kernel(
     }
     const int g0 = (kt_begin + kt) * 2;  // group 128, BK 256 -> gpt = 2
     const int n = min(n0 + lane, N - 1);
+    // sbt is [K/group, 2, N]: each (group, scale|bias) row is lane-
+    // contiguous — a few cache lines, not a 64-line strided gather
 #pragma unroll
     for (int g = 0; g < 2; ++g) {
-      // scales (4 B/lane), then biases (4 B/lane)
       __builtin_amdgcn_global_load_lds(
           reinterpret_cast<const unsigned int *>(
-              sb + ((long)n * kgroups + g0 + g) * 2),
+              sbt + ((long)(g0 + g) * 2 + 0) * N + n),
           reinterpret_cast<unsigned int *>(base + SB_OFF + g * 512), 4,
           0, 0);
       __builtin_amdgcn_global_load_lds(
           reinterpret_cast<const unsigned int *>(
-              sb + ((long)n * kgroups + g0 + g) * 2 + 1),
+              sbt + ((long)(g0 + g) * 2 + 1) * N + n),
           reinterpret_cast<unsigned int *>(base + SB_OFF + g * 512 + 256),
           4, 0, 0);
     }
@@ -182,6 +185,25 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_kernel(
     __builtin_amdgcn_s_barrier();
   }
 
+  if (out != nullptr) {  // splitk == 1: direct bf16 (+bias), no reduce
+#pragma unroll
+    for (int mb = 0; mb < MB; ++mb) {
+#pragma unroll
+      for (int nt = 0; nt < BN / 16; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = mb * 64 + wave * 16 + lhi * 4 + r;
+          const int n = n0 + nt * 16 + l16;
+          if (m < M && n < N) {
+            float v = acc[mb][nt][r];
+            if (bias) v += bias[n];
+            out[(long)m * N + n] = __float2bfloat16(v);
+          }
+        }
+      }
+    }
+    return;
+  }
   float *base = partial + (long)z * M * N;
 #pragma unroll
   for (int mb = 0; mb < MB; ++mb) {
@@ -434,12 +456,14 @@ void moe_gemm_int4(torch::Tensor C, torch::Tensor A, torch::Tensor W,
 }
 
 void int4_skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor wq,
-                      torch::Tensor sb, c10::optional<torch::Tensor> bias,
+                      torch::Tensor sbt, c10::optional<torch::Tensor> bias,
                       torch::Tensor workspace, long group) {
   const int M = x.size(0), K = x.size(1), N = wq.size(0);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 && x.is_contiguous());
   TORCH_CHECK(wq.scalar_type() == at::kByte && wq.is_contiguous());
-  TORCH_CHECK(sb.scalar_type() == at::kFloat && sb.is_contiguous());
+  TORCH_CHECK(sbt.scalar_type() == at::kFloat && sbt.is_contiguous());
+  TORCH_CHECK(sbt.numel() == (long)(K / group) * 2 * N,
+              "int4 skinny: transposed (scale,bias) [K/group, 2, N]");
   TORCH_CHECK(K % BK == 0, "int4 gemm: K must be a multiple of 256");
   TORCH_CHECK(group == 128, "int4 gemm: group size 128");
   TORCH_CHECK(M <= 256);
@@ -450,23 +474,26 @@ void int4_skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor wq,
   int k_slice = (K + splitk - 1) / splitk;
   k_slice = ((k_slice + BK - 1) / BK) * BK;
   splitk = (K + k_slice - 1) / k_slice;
-  TORCH_CHECK(workspace.numel() >= (long)splitk * M * N);
+  TORCH_CHECK(splitk == 1 || workspace.numel() >= (long)splitk * M * N);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto *ws = workspace.data_ptr<float>();
+  const float *bias_ptr = nullptr;
+  if (bias.has_value()) bias_ptr = bias->data_ptr<float>();
+  __hip_bfloat16 *outp =
+      splitk == 1 ? (__hip_bfloat16 *)out.data_ptr() : nullptr;
 #define LAUNCH_I4(MB, RING)                                                 \
   hipLaunchKernelGGL((int4_skinny_kernel<MB, RING>),                        \
                      dim3(n_wg, splitk), dim3(BLOCK), 0, stream, ws,        \
                      (const __hip_bfloat16 *)x.data_ptr(),                  \
-                     wq.data_ptr<unsigned char>(), sb.data_ptr<float>(),    \
-                     M, N, K, k_slice, (int)group)
+                     wq.data_ptr<unsigned char>(), sbt.data_ptr<float>(),   \
+                     outp, bias_ptr, M, N, K, k_slice, (int)group)
   // LDS: slot = 8K (W) + MB*32K (X); MB=4 fits only single-buffered
   if (M <= 64) LAUNCH_I4(1, 3);
   else if (M <= 128) LAUNCH_I4(2, 2);
   else LAUNCH_I4(4, 1);
 #undef LAUNCH_I4
   HIP_CHECK_KERNEL();
-  const float *bias_ptr = nullptr;
-  if (bias.has_value()) bias_ptr = bias->data_ptr<float>();
+  if (splitk == 1) return;
   const long total = (long)M * N;
   const long grid = std::min<long>((total + 1023) / 1024, 2048);
   hipLaunchKernelGGL(int4_reduce_kernel, dim3(grid), dim3(256), 0, stream,

@@ -531,10 +531,48 @@ def test_int4_linear(kernels, case):
     layer = SimpleNamespace(int4_cfg=("gptq", 128), qweight=qweight,
                             qzeros=qzeros, scales=scales)
     wq4, sb, grp = qi4.repack_canonical(layer)
+    sbt = sb.permute(1, 2, 0).contiguous()
     ref_w = qi4.dequant_gptq(qweight, qzeros, scales, 128, torch.float32)
     x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
-    out = ops.int4_linear(x, wq4.cuda(), sb.cuda(), grp)
+    out = ops.int4_linear(x, wq4.cuda(), sbt.cuda(), grp)
     ref = x.float().cpu() @ ref_w.T
+    assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=2e-3)
+
+
+def test_int4_prefill_linear(kernels):
+    """Identity-routed grouped int4 GEMM (dense M > 256 path)."""
+    M, N, K = 300, 896, 1024
+    torch.manual_seed(4)
+    from types import SimpleNamespace
+    from gllm_amd import ops
+    from gllm_amd.layers.quantization import int4 as qi4
+    w = torch.randn(N, K) / math.sqrt(K)
+    qweight, qzeros, scales = qi4.pack_gptq(w, group_size=128)
+    layer = SimpleNamespace(int4_cfg=("gptq", 128), qweight=qweight,
+                            qzeros=qzeros, scales=scales)
+    wq4, sb, _ = qi4.repack_canonical(layer)
+    ref_w = qi4.dequant_gptq(qweight, qzeros, scales, 128, torch.float32)
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    bias = torch.randn(N, dtype=torch.bfloat16, device="cuda")
+    out = ops.int4_prefill_linear(x, wq4.cuda(), sb.cuda(), bias)
+    ref = x.float().cpu() @ ref_w.T + bias.float().cpu()
+    assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=2e-3)
+
+
+def test_fp8_prefill_linear(kernels):
+    """Identity-routed grouped fp8 GEMM (dense M > 256 path)."""
+    M, N, K = 300, 1024, 512
+    torch.manual_seed(6)
+    from gllm_amd import ops
+    from gllm_amd.layers.quantization import fp8 as qfp8
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(N, K, dtype=torch.bfloat16) / math.sqrt(K)
+    wq, ws = qfp8.block_quant_fp8(w)
+    out = ops.fp8_prefill_linear(x, wq.cuda(), ws.cuda())
+    aq, as_ = qfp8.per_token_group_quant_fp8(x.cpu())
+    adq = aq.float().view(M, K // 128, 128) * as_.unsqueeze(-1)
+    wdq = qfp8.dequant_block_fp8(wq, ws, (128, 128), torch.float32)
+    ref = adq.view(M, K).float() @ wdq.T
     assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=2e-3)
 
 
