@@ -61,13 +61,10 @@ class LinearBase(nn.Module):
                       and self.fp8_block == (128, 128)
                       and ops.has_kernels())
             if native:
-                b32 = self._bias32()
                 if x.shape[0] <= 256:
-                    return ops.fp8_linear(x, w, self.weight_scale_inv, b32)
-                # prefill: identity-routed grouped fp8 GEMM (wide
-                # BM64xBN256 tile) — weights stay e4m3, no dequant
-                return ops.fp8_prefill_linear(x, w, self.weight_scale_inv,
-                                              self.bias)
+                    return ops.fp8_linear(x, w, self.weight_scale_inv,
+                                          self._bias32())
+                return self._prefill_quant(x, bias)
             from gllm_amd.layers.quantization.fp8 import dequant_block_fp8
             if x.is_cuda:
                 wd = dequant_block_fp8(w, self.weight_scale_inv,
@@ -97,7 +94,7 @@ class LinearBase(nn.Module):
                 if x.shape[0] <= 256:
                     return ops.int4_linear(x, wq4, sbt, grp,
                                            self._bias32())
-                return ops.int4_prefill_linear(x, wq4, sb, self.bias)
+                return self._prefill_quant(x, bias)
             if x.is_cuda:
                 from gllm_amd.layers.quantization.int4 import dequant_layer
                 return ops.linear(x, dequant_layer(self, x.dtype), bias)
@@ -114,6 +111,47 @@ class LinearBase(nn.Module):
         if getattr(self, "_bias_f32", None) is None:
             self._bias_f32 = self.bias.float()
         return self._bias_f32
+
+    def _prefill_quant(self, x, bias):
+        """Quantized prefill (M > 256), three modes via
+        GLLM_QUANT_PREFILL:
+
+        - "cache" (default): dequantize ONCE into a resident bf16 copy
+          and run the library GEMM. MI355X carries 288 GB HBM3E — for a
+          32B model the bf16 shadow costs 64 GB and buys bf16-identical
+          prefill speed, while decode keeps streaming the quantized
+          weights (where the bandwidth win lives). Built during the
+          profile run, so KV sizing accounts for it.
+        - "stream": identity-routed grouped quant GEMM — no shadow
+          copy; slower (dense M/64 weight re-reads) but memory-lean.
+        - "dequant": dequantize per call, nothing cached.
+        """
+        import os
+        from gllm_amd import ops
+        mode = os.environ.get("GLLM_QUANT_PREFILL", "cache")
+        if mode == "stream" and ops.has_kernels():
+            if self.fp8_block is not None:
+                return ops.fp8_prefill_linear(x, self.weight,
+                                              self.weight_scale_inv,
+                                              self.bias)
+            wq4, sb, _, _ = self._i4_canon
+            return ops.int4_prefill_linear(x, wq4, sb, self.bias)
+        if mode == "cache":
+            if self._w_dq is None:
+                if torch.cuda.is_current_stream_capturing():
+                    raise RuntimeError(
+                        "quant prefill cache built during graph capture")
+                self._w_dq = self._dequant_full(x.dtype).to(x.device)
+            return ops.linear(x, self._w_dq, bias)
+        return ops.linear(x, self._dequant_full(x.dtype), bias)
+
+    def _dequant_full(self, dtype):
+        if self.fp8_block is not None:
+            from gllm_amd.layers.quantization.fp8 import dequant_block_fp8
+            return dequant_block_fp8(self.weight, self.weight_scale_inv,
+                                     self.fp8_block, dtype)
+        from gllm_amd.layers.quantization.int4 import dequant_layer
+        return dequant_layer(self, dtype)
 
 
 class ReplicatedLinear(LinearBase):
