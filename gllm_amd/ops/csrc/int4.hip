@@ -219,6 +219,178 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_kernel(
   }
 }
 
+// --------------------------------------------- MB=1 decode variant
+// The generic tile stages the bf16 X rows through LDS: at MB=1 that is
+// a 40 KB slot (8 KB W + 32 KB X) -> RING*slot pins occupancy at ONE
+// block/CU and the X bytes cross LDS twice. Decode's X is tiny
+// (M<=64 rows, L2-resident after the first block touches it), so this
+// variant streams X straight to REGISTERS (double-buffered, static
+// names — no dynamic register indexing) and keeps only W + (scale,
+// bias) in the glds ring: 10 KB slots, ~4 blocks/CU.
+template <int RING>
+__global__ __launch_bounds__(BLOCK) void int4_skinny_mb1_kernel(
+    float *__restrict__ partial,            // [SPLITK, M, N]
+    const __hip_bfloat16 *__restrict__ x,   // [M, K]
+    const unsigned char *__restrict__ wq,   // [N, K/2]
+    const float *__restrict__ sbt,          // [K/group, 2, N]
+    __hip_bfloat16 *__restrict__ out,       // non-null iff splitk == 1
+    const float *__restrict__ bias,
+    int M, int N, int K, int k_slice) {
+  static_assert(RING == 2, "mb1 variant double-buffers X registers");
+  const int n0 = blockIdx.x * BN;
+  const int z = blockIdx.y;
+  const int kt_begin = z * (k_slice / BK);
+  const int kt_end = min(K / BK, kt_begin + k_slice / BK);
+  const int nkt = kt_end - kt_begin;
+  if (nkt <= 0) return;
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int l16 = lane & 15;
+  const int lhi = lane >> 4;
+
+  constexpr int W_B = BN * WROW_B;          // 8 KB packed W
+  constexpr int SB_OFF = W_B;
+  constexpr int SLOT_B = W_B + 4 * 512;
+  __shared__ __attribute__((aligned(16))) char smem[RING * SLOT_B];
+
+  constexpr int GL_W = W_B / 1024 / 4;      // glds16s per wave (W)
+  const unsigned char *gsrc[GL_W];
+  {
+#pragma unroll
+    for (int j = 0; j < GL_W; ++j) {
+      const int p = (wave * GL_W + j) * 1024 + lane * 16;
+      const int row = p / WROW_B;
+      const int col = swz(row, p % WROW_B);
+      const int n = min(n0 + row, N - 1);
+      gsrc[j] = wq + (long)n * (K / 2) + col;
+    }
+  }
+  const __hip_bfloat16 *xrow =
+      x + (long)min(wave * 16 + l16, M - 1) * K + (long)kt_begin * BK +
+      lhi * 8;
+
+  auto stage_w = [&](int kt, int slot) {
+    char *base = smem + slot * SLOT_B;
+    const long koff = (long)(kt_begin + kt) * WROW_B;
+#pragma unroll
+    for (int j = 0; j < GL_W; ++j)
+      glds16<2>(gsrc[j] + koff, base + (wave * GL_W + j) * 1024);
+    const int g0 = (kt_begin + kt) * 2;
+    const int n = min(n0 + lane, N - 1);
+#pragma unroll
+    for (int g = 0; g < 2; ++g) {
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const unsigned int *>(
+              sbt + ((long)(g0 + g) * 2 + 0) * N + n),
+          reinterpret_cast<unsigned int *>(base + SB_OFF + g * 512), 4,
+          0, 0);
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const unsigned int *>(
+              sbt + ((long)(g0 + g) * 2 + 1) * N + n),
+          reinterpret_cast<unsigned int *>(base + SB_OFF + g * 512 + 256),
+          4, 0, 0);
+    }
+  };
+  // X fragments for one BK tile: 8 x 16 B per lane (this wave's 16
+  // rows). Issued in the SAME per-wave vm chain as the W glds so the
+  // counted waits stay exact.
+  auto load_x = [&](int kt, i4_bf8 (&buf)[8]) {
+    const __hip_bfloat16 *src = xrow + (long)kt * BK;
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks)
+      buf[ks] = *reinterpret_cast<const i4_bf8 *>(src + ks * 32);
+  };
+  constexpr int VMS = GL_W + 4 + 8;  // W glds + sb glds + X reg loads
+
+  i4_f4 acc[BN / 16];
+#pragma unroll
+  for (int nt = 0; nt < BN / 16; ++nt) acc[nt] = i4_f4{0, 0, 0, 0};
+  i4_bf8 xa[8], xb[8];
+
+  auto consume = [&](const i4_bf8 (&xf)[8], const char *wbase) {
+    const float *sbb = reinterpret_cast<const float *>(wbase + SB_OFF);
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks) {
+      const int g = ks >> 2;
+#pragma unroll
+      for (int nt = 0; nt < BN / 16; ++nt) {
+        const int brow = nt * 16 + l16;
+        const unsigned int w4 = *reinterpret_cast<const unsigned int *>(
+            wbase + brow * WROW_B + swz(brow, (ks * 32 + lhi * 8) / 2));
+        const float s = sbb[g * 128 + brow];
+        const float bz = sbb[g * 128 + 64 + brow];
+        i4_bf8 bfrag;
+        __hip_bfloat16 *be = reinterpret_cast<__hip_bfloat16 *>(&bfrag);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float q = (float)((w4 >> (4 * e)) & 0xF);
+          be[e] = __float2bfloat16(q * s + bz);
+        }
+        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            xf[ks], bfrag, acc[nt], 0, 0, 0);
+      }
+    }
+  };
+
+  stage_w(0, 0);
+  load_x(0, xa);
+  if (nkt > 1) {
+    stage_w(1, 1);
+    load_x(1, xb);
+  }
+  for (int kt = 0; kt < nkt; ++kt) {
+    const int slot = kt & 1;
+    if (kt + 1 < nkt) {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"n"(VMS) : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    const char *wbase = smem + slot * SLOT_B;
+    if (kt & 1)
+      consume(xb, wbase);
+    else
+      consume(xa, wbase);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (kt + 2 < nkt) {
+      stage_w(kt + 2, slot);
+      if ((kt + 2) & 1)
+        load_x(kt + 2, xb);
+      else
+        load_x(kt + 2, xa);
+    }
+  }
+
+  const int m = wave * 16 + lhi * 4;
+  if (out != nullptr) {
+#pragma unroll
+    for (int nt = 0; nt < BN / 16; ++nt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int n = n0 + nt * 16 + l16;
+        if (m + r < M && n < N) {
+          float v = acc[nt][r];
+          if (bias) v += bias[n];
+          out[(long)(m + r) * N + n] = __float2bfloat16(v);
+        }
+      }
+    }
+    return;
+  }
+  float *base = partial + (long)z * M * N;
+#pragma unroll
+  for (int nt = 0; nt < BN / 16; ++nt) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int n = n0 + nt * 16 + l16;
+      if (m + r < M && n < N) base[(long)(m + r) * N + n] = acc[nt][r];
+    }
+  }
+}
+
 // ------------------------------------------------- int4 grouped MoE
 // moe.hip's grouped-GEMM pipeline with packed-nibble weights (w4a16 —
 // activations stay bf16, matching the reference's moe_wna16 kernels).
@@ -487,8 +659,14 @@ void int4_skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor wq,
                      (const __hip_bfloat16 *)x.data_ptr(),                  \
                      wq.data_ptr<unsigned char>(), sbt.data_ptr<float>(),   \
                      outp, bias_ptr, M, N, K, k_slice, (int)group)
-  // LDS: slot = 8K (W) + MB*32K (X); MB=4 fits only single-buffered
-  if (M <= 64) LAUNCH_I4(1, 3);
+  // LDS: slot = 8K (W) + MB*32K (X); MB=4 fits only single-buffered.
+  // M<=64 uses the register-X variant (10 KB slots, ~4 blocks/CU).
+  if (M <= 64)
+    hipLaunchKernelGGL((int4_skinny_mb1_kernel<2>),
+                       dim3(n_wg, splitk), dim3(BLOCK), 0, stream, ws,
+                       (const __hip_bfloat16 *)x.data_ptr(),
+                       wq.data_ptr<unsigned char>(), sbt.data_ptr<float>(),
+                       outp, bias_ptr, M, N, K, k_slice);
   else if (M <= 128) LAUNCH_I4(2, 2);
   else LAUNCH_I4(4, 1);
 #undef LAUNCH_I4
