@@ -41,6 +41,25 @@ DEV_INLINE void glds16(const unsigned char *gsrc, char *lds_ptr) {
       reinterpret_cast<unsigned int *>(lds_ptr), 16, 0, AUX);
 }
 
+// 8 nibbles of one packed dword -> bf16[8] as q*s + bz. Byte-wise
+// v_cvt_f32_ubyte on the masked even/odd nibble planes: ~2 AND + 8 cvt
+// + 8 fma + 8 cvt instead of 8x (bfe + cvt + fma + cvt) — the unpack
+// is the VALU hot loop of the whole kernel (PMC: 91% of issue).
+DEV_INLINE void unpack8(unsigned int w4, float s, float bz,
+                        __hip_bfloat16 *be) {
+  const unsigned int lo = w4 & 0x0F0F0F0Fu;         // elems 0,2,4,6
+  const unsigned int hi = (w4 >> 4) & 0x0F0F0F0Fu;  // elems 1,3,5,7
+  // uitofp(byte-extract) pattern-matches to v_cvt_f32_ubyte0..3
+  be[0] = __float2bfloat16((float)(lo & 0xffu) * s + bz);
+  be[2] = __float2bfloat16((float)((lo >> 8) & 0xffu) * s + bz);
+  be[4] = __float2bfloat16((float)((lo >> 16) & 0xffu) * s + bz);
+  be[6] = __float2bfloat16((float)(lo >> 24) * s + bz);
+  be[1] = __float2bfloat16((float)(hi & 0xffu) * s + bz);
+  be[3] = __float2bfloat16((float)((hi >> 8) & 0xffu) * s + bz);
+  be[5] = __float2bfloat16((float)((hi >> 16) & 0xffu) * s + bz);
+  be[7] = __float2bfloat16((float)(hi >> 24) * s + bz);
+}
+
 template <int MB, int RING>
 __global__ __launch_bounds__(BLOCK) void int4_skinny_kernel(
     float *__restrict__ partial,            // [SPLITK, M, N]
@@ -170,12 +189,8 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_kernel(
           const float s = sbb[g * 128 + brow];
           const float bz = sbb[g * 128 + 64 + brow];
           i4_bf8 bfrag;
-          __hip_bfloat16 *be = reinterpret_cast<__hip_bfloat16 *>(&bfrag);
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const float q = (float)((w4 >> (4 * e)) & 0xF);
-            be[e] = __float2bfloat16(q * s + bz);
-          }
+          unpack8(w4, s, bz,
+                  reinterpret_cast<__hip_bfloat16 *>(&bfrag));
           acc[mb][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag, bfrag, acc[mb][nt], 0, 0, 0);
         }
@@ -322,12 +337,8 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_mb1_kernel(
         const float s = sbb[g * 128 + brow];
         const float bz = sbb[g * 128 + 64 + brow];
         i4_bf8 bfrag;
-        __hip_bfloat16 *be = reinterpret_cast<__hip_bfloat16 *>(&bfrag);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const float q = (float)((w4 >> (4 * e)) & 0xF);
-          be[e] = __float2bfloat16(q * s + bz);
-        }
+        unpack8(w4, s, bz,
+                reinterpret_cast<__hip_bfloat16 *>(&bfrag));
         acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             xf[ks], bfrag, acc[nt], 0, 0, 0);
       }
@@ -533,12 +544,8 @@ __global__ __launch_bounds__(BLOCK) void moe_gemm_int4_kernel(
         const float s = sb_tile[cur][brow * 2];
         const float bz = sb_tile[cur][brow * 2 + 1];
         i4_bf8 bfrag;
-        __hip_bfloat16 *be = reinterpret_cast<__hip_bfloat16 *>(&bfrag);
-#pragma unroll
-        for (int ee = 0; ee < 8; ++ee) {
-          const float qv = (float)((w4 >> (4 * ee)) & 0xF);
-          be[ee] = __float2bfloat16(qv * s + bz);
-        }
+        unpack8(w4, s, bz,
+                reinterpret_cast<__hip_bfloat16 *>(&bfrag));
         acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag, bfrag, acc[t], 0, 0, 0);
       }
