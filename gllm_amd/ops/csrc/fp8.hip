@@ -273,173 +273,6 @@ __global__ __launch_bounds__(BLOCK) void fp8_skinny_kernel(
   }
 }
 
-// --------------------------------------------- MB=1 decode variant
-// PMC on the staged-tile kernel showed 69% SQ_WAIT at M<=64: the bf16
-// ring carried the A rows too, doubling the stage bytes and line
-// requests for data that is L2-resident after the first block. This
-// variant keeps only W (+ the stage's activation scales) in the glds
-// ring (8.25 KB slots) and streams each wave's 16 A rows straight to
-// registers (double-buffered, static names).
-template <int RING>
-__global__ __launch_bounds__(BLOCK) void fp8_skinny_mb1_kernel(
-    float *__restrict__ partial,           // [SPLITK, M, N]
-    const unsigned char *__restrict__ aq,  // [M, K] e4m3
-    const float *__restrict__ ast,         // [K/128, M]
-    const unsigned char *__restrict__ w,   // [N, K] e4m3
-    const float *__restrict__ ws,          // [N/128, K/128]
-    __hip_bfloat16 *__restrict__ out,      // non-null iff splitk == 1
-    const float *__restrict__ bias, int M, int N, int K, int k_slice) {
-  static_assert(RING == 2, "mb1 variant double-buffers A registers");
-  const int n0 = blockIdx.x * BN;
-  const int z = blockIdx.y;
-  const int kb_begin = z * (k_slice / BK);
-  const int kb_end = min(K / BK, kb_begin + k_slice / BK);
-  const int nkt = kb_end - kb_begin;
-  if (nkt <= 0) return;
-  const int kgroups = K / BK;
-
-  const int tid = threadIdx.x;
-  const int wave = tid >> 6;
-  const int lane = tid & 63;
-  const int l16 = lane & 15;
-  const int lhi = lane >> 4;
-
-  constexpr int W_B = BN * ROW_B;  // 8 KB
-  constexpr int ASC_OFF = W_B;
-  constexpr int SLOT_B = W_B + 256;
-  __shared__ __attribute__((aligned(16))) char smem[RING * SLOT_B];
-  __shared__ float ws_lds[256];
-
-  const int wblk = n0 / 128;
-  for (int i = tid; i < nkt; i += BLOCK)
-    ws_lds[i] = ws[(long)wblk * kgroups + kb_begin + i];
-  __syncthreads();
-
-  constexpr int GL_W = W_B / 1024 / 4;  // 2 glds16 per wave
-  const unsigned char *gsrc[GL_W];
-  {
-#pragma unroll
-    for (int j = 0; j < GL_W; ++j) {
-      const int p = (wave * GL_W + j) * 1024 + lane * 16;
-      const int row = p / ROW_B;
-      const int col = swz(row, p % ROW_B);
-      const int n = min(n0 + row, N - 1);
-      gsrc[j] = w + (long)n * K + col;
-    }
-  }
-  const unsigned char *arow =
-      aq + (long)min(wave * 16 + l16, M - 1) * K + (long)kb_begin * BK +
-      lhi * 8;
-
-  auto stage_w = [&](int kt, int slot) {
-    char *base = smem + slot * SLOT_B;
-    const long koff = (long)(kb_begin + kt) * ROW_B;
-#pragma unroll
-    for (int j = 0; j < GL_W; ++j)
-      glds16<2>(gsrc[j] + koff, base + (wave * GL_W + j) * 1024);
-    const int m = min(lane, M - 1);
-    __builtin_amdgcn_global_load_lds(
-        reinterpret_cast<const unsigned int *>(
-            ast + (long)(kb_begin + kt) * M + m),
-        reinterpret_cast<unsigned int *>(base + ASC_OFF), 4, 0, 0);
-  };
-  auto load_a = [&](int kt, long (&buf)[4]) {
-    const unsigned char *src = arow + (long)kt * BK;
-#pragma unroll
-    for (int ks = 0; ks < 4; ++ks)
-      buf[ks] = *reinterpret_cast<const long *>(src + ks * 32);
-  };
-  constexpr int VMS = GL_W + 1 + 4;  // W glds + asc glds + A reg loads
-
-  f8_f4 acc[BN / 16];
-#pragma unroll
-  for (int nt = 0; nt < BN / 16; ++nt) acc[nt] = f8_f4{0, 0, 0, 0};
-  long xa[4], xb[4];
-
-  auto consume = [&](const long (&af)[4], const char *wbase, int kt) {
-    const float *asc_lds = reinterpret_cast<const float *>(wbase + ASC_OFF);
-    const float wsc = ws_lds[kt];
-    f8_f4 sub[BN / 16];
-#pragma unroll
-    for (int nt = 0; nt < BN / 16; ++nt) sub[nt] = f8_f4{0, 0, 0, 0};
-#pragma unroll
-    for (int ks = 0; ks < 4; ++ks) {
-#pragma unroll
-      for (int nt = 0; nt < BN / 16; ++nt) {
-        const int brow = nt * 16 + l16;
-        const long bfrag = *reinterpret_cast<const long *>(
-            wbase + brow * ROW_B + swz(brow, ks * 32 + lhi * 8));
-        sub[nt] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-            af[ks], bfrag, sub[nt], 0, 0, 0);
-      }
-    }
-    float asc[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r)
-      asc[r] = asc_lds[min(wave * 16 + lhi * 4 + r, 63)] * wsc;
-#pragma unroll
-    for (int nt = 0; nt < BN / 16; ++nt)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) acc[nt][r] += sub[nt][r] * asc[r];
-  };
-
-  stage_w(0, 0);
-  load_a(0, xa);
-  if (nkt > 1) {
-    stage_w(1, 1);
-    load_a(1, xb);
-  }
-  for (int kt = 0; kt < nkt; ++kt) {
-    const int slot = kt & 1;
-    if (kt + 1 < nkt) {
-      asm volatile("s_waitcnt vmcnt(%0)" ::"n"(VMS) : "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
-    __builtin_amdgcn_s_barrier();
-    const char *wbase = smem + slot * SLOT_B;
-    if (kt & 1)
-      consume(xb, wbase, kt);
-    else
-      consume(xa, wbase, kt);
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-    if (kt + 2 < nkt) {
-      stage_w(kt + 2, slot);
-      if ((kt + 2) & 1)
-        load_a(kt + 2, xb);
-      else
-        load_a(kt + 2, xa);
-    }
-  }
-
-  const int m = wave * 16 + lhi * 4;
-  if (out != nullptr) {
-#pragma unroll
-    for (int nt = 0; nt < BN / 16; ++nt) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int n = n0 + nt * 16 + l16;
-        if (m + r < M && n < N) {
-          float v = acc[nt][r];
-          if (bias) v += bias[n];
-          out[(long)(m + r) * N + n] = __float2bfloat16(v);
-        }
-      }
-    }
-    return;
-  }
-  float *base = partial + (long)z * M * N;
-#pragma unroll
-  for (int nt = 0; nt < BN / 16; ++nt) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int n = n0 + nt * 16 + l16;
-      if (m + r < M && n < N) base[(long)(m + r) * N + n] = acc[nt][r];
-    }
-  }
-}
-
 // out[m, n] = bf16( sum_z partial[z, m, n] + bias[n] )
 __global__ void fp8_reduce_kernel(__hip_bfloat16 *__restrict__ out,
                                   const float *__restrict__ partial,
@@ -702,10 +535,16 @@ void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor ast,
   hipLaunchKernelGGL((fp8_skinny_kernel<MB, RING>),                          \
                      dim3(n_wg, splitk), dim3(BLOCK), 0, stream, wsp, ap,    \
                      asp, wp, wsc, outp, bias_ptr, M, N, K, k_slice)
-  if (M <= 64)
-    hipLaunchKernelGGL((fp8_skinny_mb1_kernel<2>),
-                       dim3(n_wg, splitk), dim3(BLOCK), 0, stream, wsp, ap,
-                       asp, wp, wsc, outp, bias_ptr, M, N, K, k_slice);
+  static int ring_env = [] {
+    const char *e = getenv("FP8_RING");
+    return e ? atoi(e) : 0;
+  }();
+  if (M <= 64) {
+    const int r = ring_env ? ring_env : 3;
+    if (r == 2) LAUNCH_SK(1, 2);
+    else if (r == 4) LAUNCH_SK(1, 4);
+    else LAUNCH_SK(1, 3);
+  }
   else if (M <= 128) LAUNCH_SK(2, 3);
   else LAUNCH_SK(4, 2);
 #undef LAUNCH_SK

@@ -72,5 +72,43 @@ def main():
         run(mode, name, iters)
 
 
+def splitk_sweep():
+    """Direct fp8_skinny_gemm calls with explicit splitk values."""
+    from gllm_amd.layers.quantization import fp8 as qfp8
+    from gllm_amd.ops import _gpu_kernels, per_token_group_quant_fp8
+    k = _gpu_kernels()
+    for name in SHAPES:
+        M, N, K = SHAPES[name]
+        x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(N, K, dtype=torch.bfloat16) / 32
+        wq, ws = qfp8.block_quant_fp8(w)
+        wq, ws = wq.cuda().view(torch.uint8), ws.cuda().contiguous()
+        aq, _, ast = per_token_group_quant_fp8(x, transposed=True)
+        aq = aq.view(torch.uint8)
+        out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+        for sk in (1, 2, 4, 8, 16):
+            if (K // sk) % 128:
+                continue
+            wsb = torch.empty(sk * M * N, dtype=torch.float32,
+                              device="cuda")
+            try:
+                k.fp8_skinny_gemm(out, aq, ast, wq, ws, None, wsb, sk)
+            except RuntimeError as e:
+                print(f"{name} sk={sk}: {e}")
+                continue
+            torch.cuda.synchronize()
+            import time
+            t0 = time.time()
+            for _ in range(30):
+                k.fp8_skinny_gemm(out, aq, ast, wq, ws, None, wsb, sk)
+            torch.cuda.synchronize()
+            dt = (time.time() - t0) / 30
+            print(f"{name:8s} splitk={sk:2d} {dt*1e6:7.1f} us "
+                  f"W-actual {N*K/dt/1e12:5.2f} TB/s")
+
+
 if __name__ == "__main__":
-    main()
+    if len(sys.argv) > 1 and sys.argv[1] == "sweep":
+        splitk_sweep()
+    else:
+        main()
