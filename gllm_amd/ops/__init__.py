@@ -228,7 +228,7 @@ def fp8_linear(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
     N = w_q.shape[0]
     aq, _, ast = per_token_group_quant_fp8(x.contiguous(),
                                            transposed=True)
-    splitk = _skinny_splitk(M, N, K)
+    splitk = _quant_splitk(N, K, 128)
     need = splitk * M * N
     key = x.device.index or 0
     ws = _FP8_WS.get(key)
@@ -407,7 +407,7 @@ def int4_linear(x: torch.Tensor, wq4: torch.Tensor, sbt: torch.Tensor,
     staging; repack_canonical builds it)."""
     M, K = x.shape
     N = wq4.shape[0]
-    splitk = _skinny_splitk(M, N, K)
+    splitk = _quant_splitk(N, K, 256)
     need = splitk * M * N
     key = x.device.index or 0
     ws = _FP8_WS.get(key)  # shares the fp32 partial workspace pool
@@ -558,6 +558,19 @@ def _skinny_splitk(M: int, N: int, K: int) -> int:
     while s < 16 and n_wg * s < 512 and (K // (s * 2)) >= 64:
         s *= 2
     k_slice = ((-(-K // s)) + 63) // 64 * 64
+    return -(-K // k_slice)
+
+
+def _quant_splitk(N: int, K: int, bk: int) -> int:
+    """splitk for the quant skinny GEMMs (fp8 bk=128, int4 bk=256):
+    deepest split that still leaves >= 8 ring stages per block, with
+    the grid capped near chip size (measured sweep, profiles/)."""
+    n_wg = -(-N // 64)
+    s = 1
+    while (s < 16 and n_wg * (s * 2) <= 1024
+           and (K // bk) // (s * 2) >= 8):
+        s *= 2
+    k_slice = ((-(-K // s)) + bk - 1) // bk * bk
     return -(-K // k_slice)
 
 

@@ -513,7 +513,8 @@ void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor ast,
   int splitk = (int)splitk_arg;
   if (splitk <= 0) {
     splitk = 1;
-    while (splitk < 16 && n_wg * splitk < 512 && (K / (splitk * 2)) >= BK)
+    while (splitk < 16 && n_wg * (splitk * 2) <= 1024 &&
+           (K / BK) / (splitk * 2) >= 8)
       splitk *= 2;
   }
   int k_slice = (K + splitk - 1) / splitk;
@@ -540,10 +541,12 @@ void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor ast,
     return e ? atoi(e) : 0;
   }();
   if (M <= 64) {
-    const int r = ring_env ? ring_env : 3;
-    if (r == 2) LAUNCH_SK(1, 2);
+    // RING=2 = 33 KB LDS = 4 blocks/CU: measured faster than the
+    // deeper 3-slot ring at 3 blocks/CU on every decode shape
+    const int r = ring_env ? ring_env : 2;
+    if (r == 3) LAUNCH_SK(1, 3);
     else if (r == 4) LAUNCH_SK(1, 4);
-    else LAUNCH_SK(1, 3);
+    else LAUNCH_SK(1, 2);
   }
   else if (M <= 128) LAUNCH_SK(2, 3);
   else LAUNCH_SK(4, 2);

@@ -648,7 +648,8 @@ void int4_skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor wq,
   TORCH_CHECK(M <= 256);
   const int n_wg = (N + BN - 1) / BN;
   int splitk = 1;
-  while (splitk < 16 && n_wg * splitk < 512 && (K / (splitk * 2)) >= BK)
+  while (splitk < 16 && n_wg * (splitk * 2) <= 1024 &&
+         (K / BK) / (splitk * 2) >= 8)
     splitk *= 2;
   int k_slice = (K + splitk - 1) / splitk;
   k_slice = ((k_slice + BK - 1) / BK) * BK;
