@@ -235,13 +235,13 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_kernel(
 }
 
 // --------------------------------------------- MB=1 decode variant
-// The generic tile stages the bf16 X rows through LDS: at MB=1 that is
-// a 40 KB slot (8 KB W + 32 KB X) -> RING*slot pins occupancy at ONE
-// block/CU and the X bytes cross LDS twice. Decode's X is tiny
-// (M<=64 rows, L2-resident after the first block touches it), so this
-// variant streams X straight to REGISTERS (double-buffered, static
-// names — no dynamic register indexing) and keeps only W + (scale,
-// bias) in the glds ring: 10 KB slots, ~4 blocks/CU.
+// Role-swapped tile for M <= 64. PMC on the generic tile showed 91% of
+// issue cycles were VALU: with X rows as the MFMA A-operand every wave
+// unpacks the ENTIRE 64-row W tile (4x duplicated). Here each wave owns
+// 16 W rows as the A-operand (unpacks each packed dword exactly once)
+// and the X tokens are the shared B-operand staged in LDS. SBK=128
+// (one quant group per stage): slot = 4 KB W + 16 KB X + 512 B
+// (scale,bias) = 20.5 KB, RING=3 -> 2 blocks/CU.
 template <int RING>
 __global__ __launch_bounds__(BLOCK) void int4_skinny_mb1_kernel(
     float *__restrict__ partial,            // [SPLITK, M, N]
@@ -251,11 +251,13 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_mb1_kernel(
     __hip_bfloat16 *__restrict__ out,       // non-null iff splitk == 1
     const float *__restrict__ bias,
     int M, int N, int K, int k_slice) {
-  static_assert(RING == 2, "mb1 variant double-buffers X registers");
+  constexpr int SBK = 128;       // elems per stage = 1 quant group
+  constexpr int WR_B = SBK / 2;  // 64 B packed W row
+  constexpr int XR_B = SBK * 2;  // 256 B bf16 X row
   const int n0 = blockIdx.x * BN;
   const int z = blockIdx.y;
-  const int kt_begin = z * (k_slice / BK);
-  const int kt_end = min(K / BK, kt_begin + k_slice / BK);
+  const int kt_begin = z * (k_slice / SBK);
+  const int kt_end = min(K / SBK, kt_begin + k_slice / SBK);
   const int nkt = kt_end - kt_begin;
   if (nkt <= 0) return;
 
@@ -265,127 +267,119 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_mb1_kernel(
   const int l16 = lane & 15;
   const int lhi = lane >> 4;
 
-  constexpr int W_B = BN * WROW_B;          // 8 KB packed W
-  constexpr int SB_OFF = W_B;
-  constexpr int SLOT_B = W_B + 4 * 512;
+  constexpr int W_B = BN * WR_B;       // 4 KB
+  constexpr int X_OFF = W_B;
+  constexpr int X_B = 64 * XR_B;       // 16 KB
+  constexpr int SB_OFF = W_B + X_B;
+  constexpr int SLOT_B = SB_OFF + 512;
   __shared__ __attribute__((aligned(16))) char smem[RING * SLOT_B];
 
-  constexpr int GL_W = W_B / 1024 / 4;      // glds16s per wave (W)
-  const unsigned char *gsrc[GL_W];
+  // 64-B W rows: swizzle must stay inside the row (bits 4-5)
+  auto swzw = [](int row, int off) { return off ^ ((row & 3) << 4); };
+  // 256-B X rows: bits 4-6
+  auto swzx = [](int row, int off) { return off ^ ((row & 7) << 4); };
+
+  constexpr int GL_W = W_B / 1024 / 4;   // 1 glds16 per wave
+  constexpr int GL_X = X_B / 1024 / 4;   // 4 glds16 per wave
+  const unsigned char *gw[GL_W];
+  const unsigned char *gx[GL_X];
   {
 #pragma unroll
     for (int j = 0; j < GL_W; ++j) {
       const int p = (wave * GL_W + j) * 1024 + lane * 16;
-      const int row = p / WROW_B;
-      const int col = swz(row, p % WROW_B);
+      const int row = p / WR_B;
       const int n = min(n0 + row, N - 1);
-      gsrc[j] = wq + (long)n * (K / 2) + col;
+      gw[j] = wq + (long)n * (K / 2) + swzw(row, p % WR_B);
+    }
+#pragma unroll
+    for (int j = 0; j < GL_X; ++j) {
+      const int p = (wave * GL_X + j) * 1024 + lane * 16;
+      const int row = p / XR_B;
+      const int m = min(row, M - 1);
+      gx[j] = reinterpret_cast<const unsigned char *>(x) +
+              (long)m * K * 2 + swzx(row, p % XR_B);
     }
   }
-  const __hip_bfloat16 *xrow =
-      x + (long)min(wave * 16 + l16, M - 1) * K + (long)kt_begin * BK +
-      lhi * 8;
 
-  auto stage_w = [&](int kt, int slot) {
+  auto stage = [&](int kt, int slot) {
     char *base = smem + slot * SLOT_B;
-    const long koff = (long)(kt_begin + kt) * WROW_B;
+    const long kw = (long)(kt_begin + kt) * WR_B;
+    const long kx = (long)(kt_begin + kt) * XR_B;
 #pragma unroll
     for (int j = 0; j < GL_W; ++j)
-      glds16<2>(gsrc[j] + koff, base + (wave * GL_W + j) * 1024);
-    const int g0 = (kt_begin + kt) * 2;
+      glds16<2>(gw[j] + kw, base + (wave * GL_W + j) * 1024);
+#pragma unroll
+    for (int j = 0; j < GL_X; ++j)
+      glds16<0>(gx[j] + kx, base + X_OFF + (wave * GL_X + j) * 1024);
+    const int g = kt_begin + kt;
     const int n = min(n0 + lane, N - 1);
-#pragma unroll
-    for (int g = 0; g < 2; ++g) {
-      __builtin_amdgcn_global_load_lds(
-          reinterpret_cast<const unsigned int *>(
-              sbt + ((long)(g0 + g) * 2 + 0) * N + n),
-          reinterpret_cast<unsigned int *>(base + SB_OFF + g * 512), 4,
-          0, 0);
-      __builtin_amdgcn_global_load_lds(
-          reinterpret_cast<const unsigned int *>(
-              sbt + ((long)(g0 + g) * 2 + 1) * N + n),
-          reinterpret_cast<unsigned int *>(base + SB_OFF + g * 512 + 256),
-          4, 0, 0);
-    }
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const unsigned int *>(sbt + ((long)g * 2) * N + n),
+        reinterpret_cast<unsigned int *>(base + SB_OFF), 4, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const unsigned int *>(
+            sbt + ((long)g * 2 + 1) * N + n),
+        reinterpret_cast<unsigned int *>(base + SB_OFF + 256), 4, 0, 0);
   };
-  // X fragments for one BK tile: 8 x 16 B per lane (this wave's 16
-  // rows). Issued in the SAME per-wave vm chain as the W glds so the
-  // counted waits stay exact.
-  auto load_x = [&](int kt, i4_bf8 (&buf)[8]) {
-    const __hip_bfloat16 *src = xrow + (long)kt * BK;
-#pragma unroll
-    for (int ks = 0; ks < 8; ++ks)
-      buf[ks] = *reinterpret_cast<const i4_bf8 *>(src + ks * 32);
-  };
-  constexpr int VMS = GL_W + 4 + 8;  // W glds + sb glds + X reg loads
+  constexpr int VMS = GL_W + GL_X + 2;
 
-  i4_f4 acc[BN / 16];
+  i4_f4 acc[4];  // 16 own-N-rows x 64 tokens
 #pragma unroll
-  for (int nt = 0; nt < BN / 16; ++nt) acc[nt] = i4_f4{0, 0, 0, 0};
-  i4_bf8 xa[8], xb[8];
+  for (int mt = 0; mt < 4; ++mt) acc[mt] = i4_f4{0, 0, 0, 0};
 
-  auto consume = [&](const i4_bf8 (&xf)[8], const char *wbase) {
-    const float *sbb = reinterpret_cast<const float *>(wbase + SB_OFF);
-#pragma unroll
-    for (int ks = 0; ks < 8; ++ks) {
-      const int g = ks >> 2;
-#pragma unroll
-      for (int nt = 0; nt < BN / 16; ++nt) {
-        const int brow = nt * 16 + l16;
-        const unsigned int w4 = *reinterpret_cast<const unsigned int *>(
-            wbase + brow * WROW_B + swz(brow, (ks * 32 + lhi * 8) / 2));
-        const float s = sbb[g * 128 + brow];
-        const float bz = sbb[g * 128 + 64 + brow];
-        i4_bf8 bfrag;
-        unpack8(w4, s, bz,
-                reinterpret_cast<__hip_bfloat16 *>(&bfrag));
-        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            xf[ks], bfrag, acc[nt], 0, 0, 0);
-      }
-    }
-  };
+  const int wr = wave * 16 + l16;  // this lane's W row within the tile
+  const int pre = min(RING - 1, nkt);
+  for (int t = 0; t < pre; ++t) stage(t, t % RING);
 
-  stage_w(0, 0);
-  load_x(0, xa);
-  if (nkt > 1) {
-    stage_w(1, 1);
-    load_x(1, xb);
-  }
   for (int kt = 0; kt < nkt; ++kt) {
-    const int slot = kt & 1;
-    if (kt + 1 < nkt) {
-      asm volatile("s_waitcnt vmcnt(%0)" ::"n"(VMS) : "memory");
+    const int slot = kt % RING;
+    if (kt + RING - 1 < nkt) stage(kt + RING - 1, (kt + RING - 1) % RING);
+    const int ahead = min(nkt - 1 - kt, RING - 1);
+    if (RING >= 3 && ahead >= 2) {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"n"(2 * VMS) : "memory");
+    } else if (ahead == 1) {
+      asm volatile("s_waitcnt vmcnt(%0)" ::"n"(1 * VMS) : "memory");
     } else {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
     __builtin_amdgcn_s_barrier();
+
     const char *wbase = smem + slot * SLOT_B;
-    if (kt & 1)
-      consume(xb, wbase);
-    else
-      consume(xa, wbase);
+    const char *xbase = wbase + X_OFF;
+    const float *sbl = reinterpret_cast<const float *>(wbase + SB_OFF);
+    const float s = sbl[wr];
+    const float bz = sbl[64 + wr];
+
+#pragma unroll
+    for (int ks = 0; ks < SBK / 32; ++ks) {
+      const unsigned int w4 = *reinterpret_cast<const unsigned int *>(
+          wbase + wr * WR_B + swzw(wr, (ks * 32 + lhi * 8) / 2));
+      i4_bf8 afrag;
+      unpack8(w4, s, bz, reinterpret_cast<__hip_bfloat16 *>(&afrag));
+#pragma unroll
+      for (int mt = 0; mt < 4; ++mt) {
+        const int m = mt * 16 + l16;
+        const i4_bf8 bfrag = *reinterpret_cast<const i4_bf8 *>(
+            xbase + m * XR_B + swzx(m, (ks * 32 + lhi * 8) * 2));
+        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag, acc[mt], 0, 0, 0);
+      }
+    }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    if (kt + 2 < nkt) {
-      stage_w(kt + 2, slot);
-      if ((kt + 2) & 1)
-        load_x(kt + 2, xb);
-      else
-        load_x(kt + 2, xa);
-    }
   }
 
-  const int m = wave * 16 + lhi * 4;
-  if (out != nullptr) {
+  const int n = n0 + wave * 16 + lhi * 4;
+  if (out != nullptr) {  // splitk == 1: direct bf16 (+bias)
 #pragma unroll
-    for (int nt = 0; nt < BN / 16; ++nt) {
+    for (int mt = 0; mt < 4; ++mt) {
+      const int m = mt * 16 + l16;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int n = n0 + nt * 16 + l16;
-        if (m + r < M && n < N) {
-          float v = acc[nt][r];
-          if (bias) v += bias[n];
-          out[(long)(m + r) * N + n] = __float2bfloat16(v);
+        if (m < M && n + r < N) {
+          float v = acc[mt][r];
+          if (bias) v += bias[n + r];
+          out[(long)m * N + n + r] = __float2bfloat16(v);
         }
       }
     }
@@ -393,11 +387,11 @@ __global__ __launch_bounds__(BLOCK) void int4_skinny_mb1_kernel(
   }
   float *base = partial + (long)z * M * N;
 #pragma unroll
-  for (int nt = 0; nt < BN / 16; ++nt) {
+  for (int mt = 0; mt < 4; ++mt) {
+    const int m = mt * 16 + l16;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int n = n0 + nt * 16 + l16;
-      if (m + r < M && n < N) base[(long)(m + r) * N + n] = acc[nt][r];
+      if (m < M && n + r < N) base[(long)m * N + n + r] = acc[mt][r];
     }
   }
 }
@@ -670,7 +664,7 @@ void int4_skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor wq,
   // LDS: slot = 8K (W) + MB*32K (X); MB=4 fits only single-buffered.
   // M<=64 uses the register-X variant (10 KB slots, ~4 blocks/CU).
   if (M <= 64)
-    hipLaunchKernelGGL((int4_skinny_mb1_kernel<2>),
+    hipLaunchKernelGGL((int4_skinny_mb1_kernel<3>),
                        dim3(n_wg, splitk), dim3(BLOCK), 0, stream, ws,
                        (const __hip_bfloat16 *)x.data_ptr(),
                        wq.data_ptr<unsigned char>(), sbt.data_ptr<float>(),
