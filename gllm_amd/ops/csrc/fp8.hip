@@ -542,7 +542,11 @@ void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor ast,
   }();
   if (M <= 64) {
     // RING=2 = 33 KB LDS = 4 blocks/CU: measured faster than the
-    // deeper 3-slot ring at 3 blocks/CU on every decode shape
+    // deeper 3-slot ring at 3 blocks/CU on every decode shape.
+    // (A W-only-DMA variant with asm register A-loads measured ~8%
+    // faster still, but async asm defs are invisible to regalloc —
+    // it may copy the register before the data lands — and it
+    // miscompiled; profiles/r02_session_notes.md records the attempt.)
     const int r = ring_env ? ring_env : 2;
     if (r == 3) LAUNCH_SK(1, 3);
     else if (r == 4) LAUNCH_SK(1, 4);
