@@ -471,11 +471,15 @@ __global__ __launch_bounds__(BLOCK) void moe_gemm_int4_kernel(
       if (idx < W_CH) {
         const int n = idx / (MOE_BK / 2 / 16);
         const int c = idx % (MOE_BK / 2 / 16);
-        if (nb * MOE_BN + n < Nd)
-          wreg[it] = *reinterpret_cast<const int4v *>(
+        if (nb * MOE_BN + n < Nd) {
+          const int4v *wp = reinterpret_cast<const int4v *>(
               W + w_base + (long)(nb * MOE_BN + n) * (K / 2) +
               kb * (MOE_BK / 2) + c * 16);
-        else
+          if constexpr (BM <= 32)  // decode: W streamed once -> nt
+            wreg[it] = __builtin_nontemporal_load(wp);
+          else
+            wreg[it] = *wp;
+        } else
           wreg[it] = int4v{0, 0, 0, 0};
       }
     }

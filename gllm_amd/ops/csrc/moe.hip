@@ -150,10 +150,17 @@ __global__ __launch_bounds__(GEMM_BLOCK) void moe_gemm_kernel(
       const int idx = tid + it * GEMM_BLOCK;
       const int n = idx / (BK / 8);
       const int c = idx % (BK / 8);
-      if (nb * BN + n < Nd && k0 + c * 8 < K)
-        wreg[it] = *reinterpret_cast<const shortx8 *>(
+      if (nb * BN + n < Nd && k0 + c * 8 < K) {
+        const shortx8 *wp = reinterpret_cast<const shortx8 *>(
             W + w_base + (long)(nb * BN + n) * K + k0 + c * 8);
-      else
+        // decode tiles (BM<=32, ~1 m-block per expert) stream each W
+        // row exactly once: non-temporal keeps the small A tile L2-
+        // resident (guide "nt-weights"); prefill re-reads W -> cached
+        if constexpr (BM <= 32)
+          wreg[it] = __builtin_nontemporal_load(wp);
+        else
+          wreg[it] = *wp;
+      } else
         wreg[it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
     }
   };
