@@ -362,10 +362,9 @@ __global__ __launch_bounds__(BLOCK) void moe_gemm_fp8_kernel(
           const int4v *wp = reinterpret_cast<const int4v *>(
               W + w_base + (long)(nb * MOE_BN + n) * K + kb * MOE_BK +
               c * 16);
-          if constexpr (BM <= 32)  // decode: W streamed once -> nt
-            wreg[it] = __builtin_nontemporal_load(wp);
-          else
-            wreg[it] = *wp;
+          // nt measured NEGATIVE here (DeepSeek fp8 MoE B256: many
+          // experts span 2 m-blocks, nt kills the W reuse) — cached
+          wreg[it] = *wp;
         } else
           wreg[it] = int4v{0, 0, 0, 0};
       }
