@@ -120,19 +120,14 @@ __global__ __launch_bounds__(GEMM_BLOCK) void moe_gemm_kernel(
   // A tile: BM rows x BK cols; W tile: BN rows x BK cols
   constexpr int A_CH = BM * BK / 8;         // 16-B chunks
   constexpr int W_CH = BN * BK / 8;
-  constexpr int A_IT = (A_CH + GEMM_BLOCK - 1) / GEMM_BLOCK;
-  constexpr int W_IT = W_CH / GEMM_BLOCK;
-  // TWO register sets: tile k+2 loads while tile k computes and tile
-  // k+1 spills to LDS — the single-set form waited out most of the
-  // load latency at the spill point every tile
-  shortx8 areg[2][A_IT];
-  shortx8 wreg[2][W_IT];
+  shortx8 areg[(A_CH + GEMM_BLOCK - 1) / GEMM_BLOCK];
+  shortx8 wreg[W_CH / GEMM_BLOCK];
 
   const long w_base = (long)e * Nd * K;
 
-  auto load_tiles = [&](int k0, shortx8 (&ar)[A_IT], shortx8 (&wr)[W_IT]) {
+  auto load_tiles = [&](int k0) {
 #pragma unroll
-    for (int it = 0; it < A_IT; ++it) {
+    for (int it = 0; it < (A_CH + GEMM_BLOCK - 1) / GEMM_BLOCK; ++it) {
       const int idx = tid + it * GEMM_BLOCK;
       if (idx < A_CH) {
         const int m = idx / (BK / 8);
@@ -144,14 +139,14 @@ __global__ __launch_bounds__(GEMM_BLOCK) void moe_gemm_kernel(
         if (SCATTER) arow = g;           // sorted-order inter buffer
         else arow = valid ? pair / topk : 0;
         if (valid && k0 + c * 8 < K)
-          ar[it] = *reinterpret_cast<const shortx8 *>(
+          areg[it] = *reinterpret_cast<const shortx8 *>(
               A + arow * (long)K + k0 + c * 8);
         else
-          ar[it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
+          areg[it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
       }
     }
 #pragma unroll
-    for (int it = 0; it < W_IT; ++it) {
+    for (int it = 0; it < W_CH / GEMM_BLOCK; ++it) {
       const int idx = tid + it * GEMM_BLOCK;
       const int n = idx / (BK / 8);
       const int c = idx % (BK / 8);
@@ -162,32 +157,31 @@ __global__ __launch_bounds__(GEMM_BLOCK) void moe_gemm_kernel(
         // row exactly once: non-temporal keeps the small A tile L2-
         // resident (guide "nt-weights"); prefill re-reads W -> cached
         if constexpr (BM <= 32)
-          wr[it] = __builtin_nontemporal_load(wp);
+          wreg[it] = __builtin_nontemporal_load(wp);
         else
-          wr[it] = *wp;
+          wreg[it] = *wp;
       } else
-        wr[it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
+        wreg[it] = shortx8{0, 0, 0, 0, 0, 0, 0, 0};
     }
   };
-  auto write_tiles = [&](int buf, shortx8 (&ar)[A_IT],
-                         shortx8 (&wr)[W_IT]) {
+  auto write_tiles = [&](int buf) {
 #pragma unroll
-    for (int it = 0; it < A_IT; ++it) {
+    for (int it = 0; it < (A_CH + GEMM_BLOCK - 1) / GEMM_BLOCK; ++it) {
       const int idx = tid + it * GEMM_BLOCK;
       if (idx < A_CH) {
         const int m = idx / (BK / 8);
         const int c = idx % (BK / 8);
         *reinterpret_cast<shortx8 *>(&a_tile[buf][m * (BK + APAD) + c * 8]) =
-            ar[it];
+            areg[it];
       }
     }
 #pragma unroll
-    for (int it = 0; it < W_IT; ++it) {
+    for (int it = 0; it < W_CH / GEMM_BLOCK; ++it) {
       const int idx = tid + it * GEMM_BLOCK;
       const int n = idx / (BK / 8);
       const int c = idx % (BK / 8);
       *reinterpret_cast<shortx8 *>(&w_tile[buf][n * (BK + APAD) + c * 8]) =
-          wr[it];
+          wreg[it];
     }
   };
 
@@ -195,21 +189,13 @@ __global__ __launch_bounds__(GEMM_BLOCK) void moe_gemm_kernel(
 #pragma unroll
   for (int t = 0; t < WN_TILES; ++t) acc[t] = mfma_f4{0, 0, 0, 0};
 
-  const int nkb = K / BK + (K % BK ? 1 : 0);
-  load_tiles(0, areg[0], wreg[0]);
-  write_tiles(0, areg[0], wreg[0]);
-  if (nkb > 1) load_tiles(BK, areg[1], wreg[1]);
+  load_tiles(0);
+  write_tiles(0);
   __syncthreads();
 
-  for (int kb = 0; kb < nkb; ++kb) {
-    const int cur = kb & 1;
-    // tile kb+2 into the register set whose tile is ALREADY in LDS
-    if (kb + 2 < nkb) {
-      if (kb & 1)
-        load_tiles((kb + 2) * BK, areg[1], wreg[1]);
-      else
-        load_tiles((kb + 2) * BK, areg[0], wreg[0]);
-    }
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    if (k0 + BK < K) load_tiles(k0 + BK);
 
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -228,12 +214,9 @@ __global__ __launch_bounds__(GEMM_BLOCK) void moe_gemm_kernel(
     __builtin_amdgcn_s_setprio(0);
 
     __syncthreads();
-    if (kb + 1 < nkb) {
-      // spill tile kb+1 (loaded one full compute ago — latency hidden)
-      if (kb & 1)
-        write_tiles(cur ^ 1, areg[0], wreg[0]);
-      else
-        write_tiles(cur ^ 1, areg[1], wreg[1]);
+    if (k0 + BK < K) {
+      write_tiles(cur ^ 1);
+      cur ^= 1;
       __syncthreads();
     }
   }
