@@ -395,6 +395,10 @@ def test_per_token_group_quant(kernels):
     q, s = ops.per_token_group_quant_fp8(x)
     qr, sr = qfp8.per_token_group_quant_fp8(x.cpu())
     assert torch.allclose(s.cpu(), sr, rtol=1e-4)
+    # transposed aux output (skinny GEMM staging layout)
+    q2, s2, st = ops.per_token_group_quant_fp8(x, transposed=True)
+    assert torch.equal(st, s2.t().contiguous())
+    assert torch.equal(q2.view(torch.uint8), q.view(torch.uint8))
     # round trip within e4m3 quantization error (3-bit mantissa: half a
     # quantum = amax/448 * 16 near the top of the range); the raw codes
     # may differ by one step where mul-by-1/s vs div-by-s rounds apart
