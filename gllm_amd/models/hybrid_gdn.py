@@ -23,7 +23,7 @@ import torch.nn as nn
 
 from gllm_amd.layers.attention import Attention
 from gllm_amd.layers.embedding import ParallelLMHead, VocabParallelEmbedding
-from gllm_amd.layers.layernorm import RMSNorm
+from gllm_amd.layers.layernorm import GemmaRMSNorm, RMSNorm
 from gllm_amd.layers.linear import (ColumnParallelLinear,
                                     MergedColumnParallelLinear,
                                     QKVParallelLinear, RowParallelLinear)
@@ -277,13 +277,17 @@ class HybridFullAttention(nn.Module):
         # q(+gate) | k | v fused, sharded by head (gate doubles q heads)
         self.qkv_proj = QKVParallelLinear(
             hidden, self.head_dim, self.total_heads * q_mult,
-            self.total_kv, bias=False, params_dtype=dtype)
+            self.total_kv,
+            bias=bool(getattr(cfg, "attention_bias", False)),
+            params_dtype=dtype)
         self.q_mult = q_mult
         self.o_proj = RowParallelLinear(self.total_heads * self.head_dim,
                                         hidden, params_dtype=dtype)
         eps = getattr(cfg, "rms_norm_eps", 1e-6)
-        self.q_norm = RMSNorm(self.head_dim, eps)
-        self.k_norm = RMSNorm(self.head_dim, eps)
+        # reference qwen3_5.py:564-565 uses the (1+w) Gemma convention
+        # for EVERY non-gated norm in this family (q/k, block, final)
+        self.q_norm = GemmaRMSNorm(self.head_dim, eps)
+        self.k_norm = GemmaRMSNorm(self.head_dim, eps)
         rot = int(self.head_dim *
                   getattr(cfg, "partial_rotary_factor", 1.0))
         self.rotary_emb = get_rope(
@@ -331,8 +335,8 @@ class HybridDecoderLayer(nn.Module):
                                                  dtype=dtype)
         self.mlp = mlp if mlp is not None else DenseMLP(
             cfg.hidden_size, cfg.intermediate_size, dtype=dtype)
-        self.input_layernorm = RMSNorm(cfg.hidden_size, eps)
-        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, eps)
+        self.input_layernorm = GemmaRMSNorm(cfg.hidden_size, eps)
+        self.post_attention_layernorm = GemmaRMSNorm(cfg.hidden_size, eps)
 
     def forward(self, positions, hidden, residual, fctx):
         if residual is None:
@@ -422,8 +426,8 @@ class Qwen3_5ForCausalLM(nn.Module):
             head_k_dim=cfg.linear_key_head_dim,
             head_v_dim=cfg.linear_value_head_dim)
         if self.is_last_stage:
-            self.norm = RMSNorm(cfg.hidden_size,
-                                getattr(cfg, "rms_norm_eps", 1e-6))
+            self.norm = GemmaRMSNorm(cfg.hidden_size,
+                                     getattr(cfg, "rms_norm_eps", 1e-6))
             self.lm_head = ParallelLMHead(cfg.vocab_size, cfg.hidden_size,
                                           params_dtype=dtype)
             if getattr(cfg, "tie_word_embeddings", False) and \
