@@ -293,7 +293,8 @@ class HybridFullAttention(nn.Module):
         self.rotary_emb = get_rope(
             self.head_dim, rot,
             getattr(cfg, "max_position_embeddings", 32768),
-            getattr(cfg, "rope_theta", 10000.0), is_neox=True)
+            getattr(cfg, "rope_theta", 10000.0), is_neox=True,
+            rope_scaling=getattr(cfg, "rope_scaling", None))
         self.num_heads = self.qkv_proj.num_heads // q_mult
         self.attn = Attention(kv_layer_idx, self.num_heads,
                               self.qkv_proj.num_kv_heads, self.head_dim,
@@ -450,10 +451,26 @@ class Qwen3_5ForCausalLM(nn.Module):
                 residual=None):
         if self.is_first_stage:
             hidden_states = self.embed_tokens(input_ids)
+            if fctx.mm_rows is not None:
+                # Qwen3.5-VL: vision embeddings replace image-pad rows
+                hidden_states = hidden_states.index_copy(
+                    0, fctx.mm_rows,
+                    fctx.mm_embeds.to(hidden_states.dtype))
             residual = None
-        for layer in self.layers:
+        ds = getattr(fctx, "mm_deepstack", None)
+        for li, layer in enumerate(self.layers):
             hidden_states, residual = layer(positions, hidden_states,
                                             residual, fctx)
+            if ds is not None:
+                # deepstack level li ADDED at the image rows after
+                # decoder layer li (reference qwen3_5.py:795-812 —
+                # same contract as Qwen3-VL, GDN layers included)
+                g = self.layer_start + li
+                H = hidden_states.shape[-1]
+                if g * H < ds.shape[1]:
+                    hidden_states = hidden_states.index_add(
+                        0, fctx.mm_rows,
+                        ds[:, g * H:(g + 1) * H].to(hidden_states.dtype))
         if self.is_last_stage:
             hidden_states, _ = self.norm(hidden_states, residual)
             return hidden_states, None

@@ -31,13 +31,16 @@ def get_arch_registry():
     from gllm_amd.models.hybrid_gdn import Qwen3_5ForCausalLM
     from gllm_amd.models.qwen2_vl import Qwen2VLForCausalLM
     from gllm_amd.models.qwen3_vl import (Qwen3VLForCausalLM,
-                                          Qwen3VLMoeForCausalLM)
+                                          Qwen3VLMoeForCausalLM,
+                                          Qwen3_5VLForCausalLM)
     return {
         "Qwen2VLForConditionalGeneration": Qwen2VLForCausalLM,
         "Qwen2_5_VLForConditionalGeneration": Qwen2VLForCausalLM,
         "Qwen3VLForConditionalGeneration": Qwen3VLForCausalLM,
         "Qwen3VLMoeForConditionalGeneration": Qwen3VLMoeForCausalLM,
         "Qwen3_5ForCausalLM": Qwen3_5ForCausalLM,
+        "Qwen3_5ForConditionalGeneration": Qwen3_5VLForCausalLM,
+        "Qwen3_5MoeForConditionalGeneration": Qwen3_5VLForCausalLM,
         "Qwen3NextForCausalLM": Qwen3_5ForCausalLM,
         "Qwen3_5MoeForCausalLM": Qwen3_5ForCausalLM,  # MoE via config
         "DeepseekV2ForCausalLM": DeepseekV2ForCausalLM,

@@ -12,6 +12,7 @@ from typing import Iterable, Tuple
 
 import torch
 
+from gllm_amd.models.hybrid_gdn import Qwen3_5ForCausalLM
 from gllm_amd.models.llama_family import Qwen3ForCausalLM
 from gllm_amd.models.moe_family import Qwen3MoeForCausalLM
 from gllm_amd.models.qwen3_vl_vision import Qwen3VisionTransformer
@@ -65,6 +66,16 @@ class Qwen3VLForCausalLM(_Qwen3VLMixin, Qwen3ForCausalLM):
 
 
 class Qwen3VLMoeForCausalLM(_Qwen3VLMixin, Qwen3MoeForCausalLM):
+    def __init__(self, cfg, engine_config):
+        super().__init__(cfg, engine_config)
+        self._init_vision(cfg, engine_config)
+
+
+class Qwen3_5VLForCausalLM(_Qwen3VLMixin, Qwen3_5ForCausalLM):
+    """Qwen3.5-VL: Qwen3-VL vision tower + the hybrid-GDN text LM
+    (reference qwen3_5.py:1093-1107 Qwen3_5ForConditionalGeneration —
+    the MoE flavour routes through the same class via config)."""
+
     def __init__(self, cfg, engine_config):
         super().__init__(cfg, engine_config)
         self._init_vision(cfg, engine_config)

@@ -111,3 +111,75 @@ def test_interleaved_mrope_differs_from_sectioned():
     qa2, _ = a(pos_eq, q.clone(), k.clone())
     qb2, _ = b(pos_eq, q.clone(), k.clone())
     assert torch.allclose(qa2, qb2, atol=1e-6)
+
+
+VL35_TINY = {
+    "architectures": ["Qwen3_5ForConditionalGeneration"],
+    "model_type": "qwen3_5",
+    "hidden_size": 64,
+    "intermediate_size": 128,
+    "num_hidden_layers": 4,
+    "full_attention_interval": 2,
+    "num_attention_heads": 4,
+    "num_key_value_heads": 2,
+    "head_dim": 16,
+    "attn_output_gate": True,
+    "partial_rotary_factor": 0.5,
+    "linear_num_value_heads": 4,
+    "linear_num_key_heads": 2,
+    "linear_key_head_dim": 8,
+    "linear_value_head_dim": 8,
+    "linear_conv_kernel_dim": 4,
+    "vocab_size": 160,
+    "image_token_id": 150,
+    "max_position_embeddings": 2048,
+    "rms_norm_eps": 1e-6,
+    "rope_theta": 10000.0,
+    "rope_scaling": {"type": "mrope", "mrope_section": [1, 1, 2],
+                     "mrope_interleaved": True},
+    "eos_token_id": 0,
+    "vision_config": VL3_TINY["vision_config"],
+}
+
+
+def _mk_llm35(tmp_path, name="v35", maxp=64):
+    d = tmp_path / name
+    d.mkdir(exist_ok=True)
+    with open(d / "config.json", "w") as f:
+        json.dump(VL35_TINY, f)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    cfg = EngineConfig(model=str(d), load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=maxp)
+    return LLM(config=cfg, num_pages_override=128)
+
+
+def test_qwen3_5_vl_generate_and_deepstack_matters(tmp_path):
+    """Qwen3.5-VL = Qwen3-VL tower + hybrid-GDN text LM (reference
+    qwen3_5.py Qwen3_5ForConditionalGeneration): deterministic
+    generation with image inputs, deepstack levels influence output."""
+    from gllm_amd.sequence import SamplingParams
+    llm = _mk_llm35(tmp_path)
+    toks, mm = _mm()
+    sp = [SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)]
+    o1 = llm.generate([toks], sp, mm_inputs=[mm])[0].token_ids
+    o2 = llm.generate([toks], sp, mm_inputs=[mm])[0].token_ids
+    assert len(o1) == 6 and o1 == o2
+    emb = llm.runner.model.encode_images(mm["pixel_values"], mm["grids"])
+    emb2 = emb.clone()
+    emb2[:, 64:] = 0
+    o3 = llm.generate([toks], sp,
+                      mm_inputs=[{"embeds": emb2.detach(),
+                                  "grids": mm["grids"]}])[0].token_ids
+    assert o3 != o1, "deepstack levels must influence generation"
+
+
+def test_qwen3_5_vl_chunked_prefill_matches_full(tmp_path):
+    from gllm_amd.sequence import SamplingParams
+    toks, mm = _mm()
+    sp = [SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)]
+    ref = _mk_llm35(tmp_path, name="f35").generate(
+        [toks], sp, mm_inputs=[mm])[0].token_ids
+    out = _mk_llm35(tmp_path, name="c35", maxp=3).generate(
+        [toks], sp, mm_inputs=[mm])[0].token_ids
+    assert out == ref
