@@ -54,9 +54,11 @@ class FusedMoE(nn.Module):
             requires_grad=False)
         self.w13_weight.weight_loader = self._load_w13
         self.w2_weight.weight_loader = self._load_w2
-        # set by quantization/ converters (convert_moe_to_{fp8,int4})
+        # set by quantization/ converters (convert_moe_to_{fp8,int4}
+        # / convert_moe_to_int4_packed)
         self.fp8_block = None
         self.int4_cfg = None
+        self.int4_packed = None
         self._dq_cache = None
 
     # ---- loading: per-expert pulls with EP ownership / TP sharding ----
@@ -129,6 +131,27 @@ class FusedMoE(nn.Module):
         DeepSeek-V3). DP padding rows carry expert id -1 and land in
         segment 0 of the shifted bincount (skipped)."""
         if (x.is_cuda and x.dtype == torch.bfloat16 and ops.has_kernels()):
+            if self.int4_packed is not None:
+                # compressed-tensors int4 experts (Kimi-K2.5): run the
+                # bf16 grouped MFMA GEMM over a one-time dequantized
+                # shadow bank (288 GB HBM covers it; the group-32
+                # native packed kernel is a ROADMAP item)
+                if getattr(self, "_dq_stack", None) is None:
+                    from gllm_amd.layers.quantization.int4 import                         dequant_ct_int4
+                    g = self.int4_packed[1]
+                    self._dq_stack = tuple(
+                        torch.stack([
+                            dequant_ct_int4(wp[e], ws[e], g, x.dtype)
+                            for e in range(self.num_local_experts)
+                        ]).contiguous()
+                        for wp, ws in
+                        ((self.w13_weight_packed, self.w13_weight_scale),
+                         (self.w2_weight_packed, self.w2_weight_scale)))
+                w13s, w2s = self._dq_stack
+                return ops.fused_moe(
+                    x.contiguous(), w13s, w2s, weights, ids,
+                    expert_start=self.expert_start,
+                    num_global_experts=self.num_experts)
             if self.fp8_block is None and self.int4_cfg is None:
                 return ops.fused_moe(
                     x.contiguous(), self.w13_weight, self.w2_weight,
@@ -177,7 +200,8 @@ class FusedMoE(nn.Module):
             start += c
             rows = flat_rows[sel]
             xe = x.index_select(0, rows)
-            if self.fp8_block is not None or self.int4_cfg is not None:
+            if (self.fp8_block is not None or self.int4_cfg is not None
+                    or self.int4_packed is not None):
                 w13, w2 = self._dequant_expert(lid, x.dtype, x.device)
             else:
                 w13, w2 = self.w13_weight[lid], self.w2_weight[lid]
@@ -203,7 +227,17 @@ class FusedMoE(nn.Module):
                     self._dequant_expert(other, dtype, device)
         ent = self._dq_cache[lid]
         if ent is None:
-            if self.fp8_block is not None:
+            if self.int4_packed is not None:
+                from gllm_amd.layers.quantization.int4 import \
+                    dequant_ct_int4
+                g = self.int4_packed[1]
+                w13 = dequant_ct_int4(self.w13_weight_packed[lid],
+                                      self.w13_weight_scale[lid], g,
+                                      dtype).to(device)
+                w2 = dequant_ct_int4(self.w2_weight_packed[lid],
+                                     self.w2_weight_scale[lid], g,
+                                     dtype).to(device)
+            elif self.fp8_block is not None:
                 from gllm_amd.layers.quantization.fp8 import \
                     dequant_block_fp8
                 w13 = dequant_block_fp8(

@@ -357,3 +357,96 @@ def repack_canonical_moe(moe):
         banks.append((torch.stack(cs).contiguous(),
                       torch.stack(sbs).contiguous()))
     return banks[0][0], banks[0][1], banks[1][0], banks[1][1], group
+
+
+# ------------------------------------------------------------------
+# compressed-tensors "pack-quantized" int4 (Kimi-K2.5 routed experts;
+# reference model_loader.py:538-591 _normalize_kimi_quant_config +
+# fused_moe_triton/layer.py:229-545 Int4MarlinMoEMethod). Layout per
+# expert: weight_packed int32 [N, K/8] (8 x int4 along K, nibble j at
+# bits 4j), SYMMETRIC with offset 8 (uint4b8: value = q - 8), and
+# weight_scale [N, K/group] group scales. No zero points.
+
+def pack_ct_int4(w: torch.Tensor, group_size: int = 32):
+    """Quantize [N, K] -> (weight_packed int32 [N, K/8],
+    weight_scale [N, K/group]) in the compressed-tensors symmetric
+    int4 layout (test/tooling helper; checkpoints arrive packed)."""
+    N, K = w.shape
+    assert K % group_size == 0 and K % 8 == 0
+    wg = w.float().view(N, K // group_size, group_size)
+    scale = wg.abs().amax(-1).clamp(min=1e-8) / 7.0
+    q = torch.clamp(torch.round(wg / scale.unsqueeze(-1)), -8, 7)
+    qu = (q + 8).to(torch.int32).view(N, K // 8, 8)
+    packed = torch.zeros(N, K // 8, dtype=torch.int32)
+    for j in range(8):
+        packed |= qu[:, :, j] << (4 * j)
+    return packed, scale.to(w.dtype)
+
+
+def dequant_ct_int4(packed: torch.Tensor, scale: torch.Tensor,
+                    group_size: int, dtype=torch.bfloat16) -> torch.Tensor:
+    """(int32 [N, K/8], scales [N, K/group]) -> [N, K] dtype."""
+    N, Kp = packed.shape
+    K = Kp * 8
+    shifts = torch.arange(8, device=packed.device) * 4
+    q = (packed.unsqueeze(-1) >> shifts) & 0xF      # [N, K/8, 8]
+    q = q.view(N, K).float() - 8.0
+    s = scale.float().repeat_interleave(group_size, dim=1)
+    return (q * s).to(dtype)
+
+
+def convert_moe_to_int4_packed(model, mq: dict) -> int:
+    """Swap every FusedMoE's bf16 expert banks for compressed-tensors
+    packed-int4 parameters (+ group scales) with TP/EP-aware loaders.
+    Execution: dequantized bf16 banks feed the grouped MFMA MoE GEMM
+    (memory for the shadow bank comes out of the 288 GB pool — the
+    group-32 native kernel variant is a ROADMAP item)."""
+    import torch.nn as nn
+    from gllm_amd.layers.moe.layer import FusedMoE
+    group = int(mq.get("group_size", 32))
+    n = 0
+    for moe in model.modules():
+        if not isinstance(moe, FusedMoE):
+            continue
+        E = moe.num_local_experts
+        I = moe.intermediate_per_rank
+        H = moe.hidden_size
+        assert H % 8 == 0 and I % 8 == 0 and H % group == 0 \
+            and I % group == 0
+        dt = moe.w13_weight.dtype
+        dev = moe.w13_weight.device
+        del moe.w13_weight, moe.w2_weight
+
+        def mk(shape, dtype):
+            return nn.Parameter(torch.empty(*shape, dtype=dtype,
+                                            device=dev),
+                                requires_grad=False)
+
+        moe.w13_weight_packed = mk((E, 2 * I, H // 8), torch.int32)
+        moe.w13_weight_scale = mk((E, 2 * I, H // group), dt)
+        moe.w2_weight_packed = mk((E, H, I // 8), torch.int32)
+        moe.w2_weight_scale = mk((E, H, I // group), dt)
+        # w13: rows are the (gate|up) output dim — the existing row
+        # loaders shard dim 0, packing is along K (dim 1): reuse as-is
+        moe.w13_weight_packed.weight_loader = moe._load_w13
+        moe.w13_weight_scale.weight_loader = moe._load_w13
+
+        def load_w2_cols(param, loaded, expert_id, div, _moe=moe):
+            lid = _moe._local_expert(expert_id)
+            if lid is None:
+                return
+            cols = _moe.intermediate_per_rank // div
+            if _moe.use_ep:
+                shard = loaded
+            else:
+                from gllm_amd.parallel import get_tp_rank
+                shard = loaded.narrow(1, get_tp_rank() * cols, cols)
+            param.data[lid].copy_(shard)
+
+        moe.w2_weight_packed.weight_loader = \
+            lambda p, w, e, _f=load_w2_cols: _f(p, w, e, 8)
+        moe.w2_weight_scale.weight_loader = \
+            lambda p, w, e, _f=load_w2_cols: _f(p, w, e, group)
+        moe.int4_packed = (4, group)
+        n += 1
+    return n

@@ -186,7 +186,8 @@ def dummy_init(model: torch.nn.Module, seed: int = 0) -> None:
             elif p.dtype in (torch.int32, torch.uint8):
                 # packed int4 qweight/qzeros: random bits
                 p.data.random_(generator=gen)
-            elif "scales" in gname and p.dim() >= 2:
+            elif ("scales" in gname or "weight_scale" in gname) \
+                    and p.dim() >= 2:
                 p.data.fill_(0.01)
             elif p.dim() >= 2:
                 # draw in-place on device (32B params via a CPU RNG would
@@ -215,13 +216,35 @@ def create_model(cfg, engine_config, device: str = "cpu") -> torch.nn.Module:
     return model.eval()
 
 
+def _normalize_compressed_tensors(cfg, qcfg):
+    """Kimi-K2.5 ships compressed-tensors ``pack-quantized`` int4 on the
+    ROUTED EXPERTS only (dense/shared layers stay bf16). Translate it to
+    a moe_quantization_config hint and clear the dense-layer config
+    (reference model_loader.py:538-591)."""
+    num_bits, group = 4, 32
+    for grp in (qcfg.get("config_groups") or {}).values():
+        wc = (grp or {}).get("weights") or {}
+        num_bits = int(wc.get("num_bits", num_bits))
+        group = int(wc.get("group_size", group))
+        break
+    if num_bits != 4:
+        raise ValueError(f"compressed-tensors: only int4, got {num_bits}")
+    cfg.moe_quantization_config = {"quant_method": "int4_moe",
+                                   "num_bits": 4, "group_size": group,
+                                   "symmetric": True}
+    cfg.quantization_config = None
+    return None
+
+
 def load_model(engine_config, device: str = "cpu"):
     t0 = time.time()
     cfg = load_hf_config(engine_config.model)
-    model = create_model(cfg, engine_config, device)
     qcfg = getattr(cfg, "quantization_config", None)
     if qcfg is not None and not isinstance(qcfg, dict):
         qcfg = getattr(qcfg, "to_dict", lambda: vars(qcfg))()
+    if qcfg and qcfg.get("quant_method") == "compressed-tensors":
+        qcfg = _normalize_compressed_tensors(cfg, qcfg)
+    model = create_model(cfg, engine_config, device)
     if qcfg:  # dummy loads convert too: quantized EXECUTION
         # with random weights is exactly what bench --quant tests
         method = qcfg.get("quant_method")
@@ -242,7 +265,14 @@ def load_model(engine_config, device: str = "cpu"):
                                        qcfg.get("q_group_size", 128))})
             logger.info("%s int4 checkpoint: converted %d linears",
                         method, n)
-    if qcfg:
+    mq = getattr(cfg, "moe_quantization_config", None)
+    if mq and mq.get("quant_method") == "int4_moe":
+        from gllm_amd.layers.quantization.int4 import \
+            convert_moe_to_int4_packed
+        n = convert_moe_to_int4_packed(model, mq)
+        logger.info("compressed-tensors int4 MoE: converted %d layers "
+                    "(group %d); dense layers bf16", n, mq["group_size"])
+    if qcfg or mq:
         # quant converters create replacement params on CPU; re-home
         # them next to the rest of the model
         model = model.to(device)

@@ -169,6 +169,10 @@ class MoEFamilyForCausalLM(LlamaFamilyForCausalLM):
                 if wname == ckpt:
                     if suffix == "weight_scale_inv":  # fp8 expert scales
                         fused = fused + "_scale_inv"
+                    elif suffix == "weight_packed":  # ct int4 experts
+                        fused = fused + "_packed"
+                    elif suffix == "weight_scale":
+                        fused = fused + "_scale"
                     elif suffix in ("qweight", "qzeros", "scales"):
                         fused = fused.replace("weight", suffix)  # int4
                     p = params[f"{prefix}.{fused}"]
