@@ -44,6 +44,10 @@ class EngineConfig:
     # encoder disaggregation (disagg/): remote vision-encoder address
     # "host:port", or a discovery server to resolve one from
     mm_encoder_addr: Optional[str] = None
+    # LM node of an encoder-disaggregated deployment: skip loading the
+    # vision tower (reference lm_server.py --skip-visual); embeddings
+    # arrive from the remote encoder
+    skip_visual: bool = False
     discovery_addr: Optional[str] = None
 
     # --- MLA (DeepSeek family) ---

@@ -489,6 +489,10 @@ def make_arg_parser():
     p.add_argument("--discovery-addr", type=str, default=None,
                    help="discovery server 'host:port' to resolve an "
                         "encoder from")
+    p.add_argument("--skip-visual", action="store_true", default=None,
+                   help="LM node of an encoder-disagg deployment: do "
+                        "not load the vision tower (default: on when "
+                        "an encoder addr/discovery is configured)")
     p.add_argument("--mla-mode", choices=["absorbed", "decompressed"],
                    default="absorbed",
                    help="MLA execution form (reference --mla-backend): "
@@ -515,6 +519,9 @@ def config_from_args(args) -> EngineConfig:
         if args.worker_ranks else None,
         relay_port=args.relay_port,
         mm_encoder_addr=args.mm_encoder_addr,
+        skip_visual=(args.skip_visual if args.skip_visual is not None
+                     else bool(args.mm_encoder_addr
+                               or args.discovery_addr)),
         discovery_addr=args.discovery_addr, seed=args.seed,
         mla_mode=args.mla_mode,
         device="cuda" if _has_gpu() else "cpu")

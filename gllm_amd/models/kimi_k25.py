@@ -28,7 +28,8 @@ class KimiK25ForCausalLM(DeepseekV2ForCausalLM):
             vcfg = dict(vars(vcfg))
         self.image_token_id = getattr(cfg, "media_placeholder_token_id",
                                       None)
-        if vcfg is not None and self.is_first_stage:
+        if (vcfg is not None and self.is_first_stage
+                and not getattr(engine_config, "skip_visual", False)):
             dtype = engine_config.torch_dtype()
             vcfg.setdefault("text_hidden_size", cfg.hidden_size)
             vcfg.setdefault("mm_hidden_size", vcfg["vt_hidden_size"])

@@ -31,7 +31,8 @@ class Qwen2VLForCausalLM(Qwen2ForCausalLM):
         self.image_token_id = getattr(cfg, "image_token_id", None)
         self.spatial_merge_size = getattr(vcfg, "spatial_merge_size", 2) \
             if vcfg is not None else 2
-        if self.is_first_stage and vcfg is not None:
+        if (self.is_first_stage and vcfg is not None
+                and not getattr(engine_config, "skip_visual", False)):
             # Qwen2.5-VL configs carry window attention fields; the
             # plain Qwen2-VL tower is full-attention LayerNorm/QuickGELU
             if getattr(vcfg, "window_size", None) is not None:

@@ -28,7 +28,8 @@ class _Qwen3VLMixin:
         self.image_token_id = getattr(cfg, "image_token_id", None)
         self.spatial_merge_size = getattr(vcfg, "spatial_merge_size", 2) \
             if vcfg is not None else 2
-        if self.is_first_stage and vcfg is not None:
+        if (self.is_first_stage and vcfg is not None
+                and not getattr(engine_config, "skip_visual", False)):
             self.visual = Qwen3VisionTransformer(
                 vcfg, dtype=engine_config.torch_dtype())
         else:
