@@ -22,17 +22,25 @@ class RMSNorm(nn.Module):
 
 
 class GemmaRMSNorm(nn.Module):
-    """(1 + w) convention."""
+    """(1 + w) convention (Qwen3.5/Gemma checkpoints store w - 1).
+
+    The effective scale (1 + w) is cached after load so the hot path
+    runs the same fused add+rmsnorm kernel as RMSNorm."""
 
     def __init__(self, hidden_size: int, eps: float = 1e-6):
         super().__init__()
         self.weight = nn.Parameter(torch.zeros(hidden_size))
         self.eps = eps
+        self._w1 = None
+
+    def _scale(self):
+        if self._w1 is None or self._w1.device != self.weight.device:
+            self._w1 = (self.weight + 1.0).contiguous()
+        return self._w1
 
     def forward(self, x: torch.Tensor,
                 residual: Optional[torch.Tensor] = None):
-        w = self.weight + 1.0
         if residual is not None:
-            residual.add_(x)
-            return ops.rmsnorm(residual, w, self.eps), residual
-        return ops.rmsnorm(x, w, self.eps)
+            return ops.fused_add_rmsnorm(x, residual, self._scale(),
+                                         self.eps)
+        return ops.rmsnorm(x, self._scale(), self.eps)
