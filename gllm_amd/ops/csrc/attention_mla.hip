@@ -144,12 +144,14 @@ __global__ __launch_bounds__(BLOCK, 1) void mla_attention_kernel(
       const int ctok = min(tok, seq_len - 1);
       const long crow =
           (long)bt[ctok / page_size] * page_size + ctok % page_size;
+      // aux=2 (nt): each latent row is streamed once per split —
+      // non-temporal lands faster and leaves L2 to shared data
       __builtin_amdgcn_global_load_lds(
           reinterpret_cast<const unsigned int *>(
               reinterpret_cast<const char *>(kc) + crow * ROW_B + col),
           reinterpret_cast<unsigned int *>(
               base + (wave * GL_PER_WAVE + j) * 1024),
-          16, 0, 0);
+          16, 0, 2);
     }
   };
 
