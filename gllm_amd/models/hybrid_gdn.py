@@ -247,8 +247,8 @@ class GatedDeltaNet(nn.Module):
                 gb[j, :n] = g_all[s:e]
                 bb[j, :n] = beta_all[s:e]
             states_b = ssm_states[slots_l].contiguous()
-            ob = gdn_ref.gated_delta_rule_chunked_batched(
-                qb, kb, vb, gb, bb, self.scale, states_b)
+            ob = _ops.gdn_chunk_prefill(
+                qb, kb, vb, gb, bb, states_b, self.scale)
             ssm_states[slots_l] = states_b
             for j, i in enumerate(prefill_idx):
                 s, e = qsl[i], qsl[i + 1]

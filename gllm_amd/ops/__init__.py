@@ -386,6 +386,39 @@ def gdn_decode(qn: torch.Tensor, kn: torch.Tensor, v: torch.Tensor,
     return o
 
 
+def gdn_chunk_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                      g: torch.Tensor, beta: torch.Tensor,
+                      states: torch.Tensor, scale: float) -> torch.Tensor:
+    """Fused WY chunk-parallel gated delta rule over a padded batch
+    (one launch per layer; replaces ~25 torch launches per 256-token
+    chunk). q/k [B, T, Hk, 128], v [B, T, Hv, 128], g/beta [B, T, Hv]
+    fp32, states [B, Hv, 128, 128] fp32 updated IN PLACE. Falls back
+    to the torch oracle off-GPU or for non-128 head dims."""
+    B, T, Hk, Dk = q.shape
+    Hv, Dv = v.shape[2], v.shape[3]
+    if not (q.is_cuda and has_kernels() and Dk == 128 and Dv == 128
+            and Hv % Hk == 0):
+        from gllm_amd.ops import gdn_ref
+        return gdn_ref.gated_delta_rule_chunked_batched(
+            q, k, v, g, beta, scale, states)
+    Tp = (T + 63) // 64 * 64
+    if Tp != T:  # zero padding rows carry g = 0, beta = 0 -> inert
+        pad = (0, 0, 0, 0, 0, Tp - T)
+        q = torch.nn.functional.pad(q, pad)
+        k = torch.nn.functional.pad(k, pad)
+        v = torch.nn.functional.pad(v, pad)
+        g = torch.nn.functional.pad(g, (0, 0, 0, Tp - T))
+        beta = torch.nn.functional.pad(beta, (0, 0, 0, Tp - T))
+    o = torch.empty(B, Tp, Hv, Dv, dtype=torch.bfloat16, device=q.device)
+    _gpu_kernels().gdn_chunk_prefill(
+        o, q.to(torch.bfloat16).contiguous(),
+        k.to(torch.bfloat16).contiguous(),
+        v.to(torch.bfloat16).contiguous(),
+        g.float().contiguous(), beta.float().contiguous(),
+        states, float(scale))
+    return o[:, :T]
+
+
 def rmsnorm_gated(x: torch.Tensor, z: torch.Tensor, weight: torch.Tensor,
                   eps: float) -> torch.Tensor:
     """out = rmsnorm(x) * w * silu(z) (fused, GPU)."""

@@ -67,6 +67,9 @@ void moe_gemm_int4(torch::Tensor C, torch::Tensor A, torch::Tensor W,
 void gdn_conv_update(torch::Tensor out, torch::Tensor x,
                      torch::Tensor weight, torch::Tensor conv_state,
                      torch::Tensor slots);
+void gdn_chunk_prefill(torch::Tensor o, torch::Tensor q, torch::Tensor k,
+                       torch::Tensor v, torch::Tensor g, torch::Tensor beta,
+                       torch::Tensor states, double scale);
 void gdn_decode(torch::Tensor o, torch::Tensor qn, torch::Tensor kn,
                 torch::Tensor v, torch::Tensor g, torch::Tensor beta,
                 torch::Tensor state, torch::Tensor slots);
@@ -124,6 +127,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused-dequant int4 grouped MoE GEMM (w4a16)");
   m.def("gdn_conv_update", &gdn_conv_update,
         "batched causal-conv1d decode step w/ state roll");
+  m.def("gdn_chunk_prefill", &gdn_chunk_prefill,
+        "fused WY chunk-parallel gated delta rule prefill");
   m.def("gdn_decode", &gdn_decode,
         "batched fused recurrent gated-delta-rule decode step");
   m.def("rmsnorm_gated", &rmsnorm_gated, "rmsnorm(x)*w*silu(z)");

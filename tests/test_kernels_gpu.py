@@ -612,3 +612,40 @@ def test_fused_moe_int4(kernels):
     ref = _moe_ref(x, torch.stack(w13d).cuda(), torch.stack(w2d).cuda(),
                    weights, ids)
     assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=3e-3)
+
+
+def test_gdn_chunk_prefill(kernels):
+    """Fused WY chunk prefill vs the fp32 torch oracle: padded batch,
+    multi-chunk sequences, ragged lengths (inert padding), GQA repeat,
+    and in-place state update."""
+    import copy
+    from gllm_amd import ops
+    from gllm_amd.ops import gdn_ref
+    torch.manual_seed(11)
+    B, T, Hk, Hv, D = 3, 192, 2, 4, 128
+    lens = [192, 130, 64]
+    q = torch.randn(B, T, Hk, D, device="cuda") * 0.5
+    k = torch.randn(B, T, Hk, D, device="cuda") * 0.5
+    v = torch.randn(B, T, Hv, D, device="cuda") * 0.5
+    g = -torch.rand(B, T, Hv, device="cuda") * 0.1
+    beta = torch.rand(B, T, Hv, device="cuda") * 0.9 + 0.05
+    for b, n in enumerate(lens):  # padding rows must be inert
+        g[b, n:] = 0
+        beta[b, n:] = 0
+        q[b, n:] = 0
+        k[b, n:] = 0
+        v[b, n:] = 0
+    st = torch.randn(B, Hv, D, D, device="cuda") * 0.05
+    st_hip = st.clone()
+    st_ref = st.clone()
+    # oracle consumes the same bf16-rounded inputs the kernel reads
+    qh = q.to(torch.bfloat16).float()
+    kh = k.to(torch.bfloat16).float()
+    vh = v.to(torch.bfloat16).float()
+    out = ops.gdn_chunk_prefill(q, k, v, g, beta, st_hip, 0.5)
+    ref = gdn_ref.gated_delta_rule_chunked_batched(
+        qh, kh, vh, g, beta, 0.5, st_ref, chunk=64)
+    for b, n in enumerate(lens):
+        assert_close_bf16(out[b, :n].float(), ref[b, :n].float(),
+                          atol=2e-2, rtol=2e-2, frac=1e-3)
+    assert torch.allclose(st_hip, st_ref, atol=2e-2, rtol=2e-2)
