@@ -409,23 +409,28 @@ def test_per_token_group_quant(kernels):
 
 
 @pytest.mark.parametrize("case", [
-    # (M, N, K)
-    (1, 1024, 512), (8, 896, 1024), (64, 2048, 896), (200, 512, 1280),
+    # (M, N, K, bias?) — bias exercises the fused splitk==1 epilogue
+    # and the reduce-kernel bias path
+    (1, 1024, 512, False), (8, 896, 1024, True), (64, 2048, 896, False),
+    (64, 55296 // 9, 1152, True), (200, 512, 1280, True),
 ])
 def test_fp8_linear(kernels, case):
-    M, N, K = case
+    M, N, K, with_bias = case
     torch.manual_seed(M + N)
     from gllm_amd import ops
     from gllm_amd.layers.quantization import fp8 as qfp8
     x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
     w = torch.randn(N, K, dtype=torch.bfloat16) / math.sqrt(K)
+    bias = torch.randn(N, dtype=torch.float32, device="cuda")         if with_bias else None
     wq, ws = qfp8.block_quant_fp8(w)
-    out = ops.fp8_linear(x, wq.cuda(), ws.cuda())
+    out = ops.fp8_linear(x, wq.cuda(), ws.cuda(), bias)
     # oracle: dequantized weights x quantized activations in fp32
     aq, as_ = qfp8.per_token_group_quant_fp8(x.cpu())
     adq = aq.float().view(M, K // 128, 128) * as_.unsqueeze(-1)
     wdq = qfp8.dequant_block_fp8(wq, ws, (128, 128), torch.float32)
     ref = adq.view(M, K).float() @ wdq.T
+    if bias is not None:
+        ref = ref + bias.cpu().float()
     assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=2e-3)
 
 
@@ -522,10 +527,12 @@ def test_rmsnorm_gated(kernels):
 
 
 # ------------------------------------------------------------ int4
-@pytest.mark.parametrize("case", [(8, 512, 1024), (64, 896, 2048),
-                                  (200, 1024, 512)])
+@pytest.mark.parametrize("case", [(8, 512, 1024, False),
+                                  (64, 896, 2048, True),
+                                  (64, 4096, 1280, True),
+                                  (200, 1024, 512, True)])
 def test_int4_linear(kernels, case):
-    M, N, K = case
+    M, N, K, with_bias = case
     torch.manual_seed(M)
     from types import SimpleNamespace
     from gllm_amd import ops
@@ -538,8 +545,11 @@ def test_int4_linear(kernels, case):
     sbt = sb.permute(1, 2, 0).contiguous()
     ref_w = qi4.dequant_gptq(qweight, qzeros, scales, 128, torch.float32)
     x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
-    out = ops.int4_linear(x, wq4.cuda(), sbt.cuda(), grp)
+    bias = torch.randn(N, dtype=torch.float32, device="cuda")         if with_bias else None
+    out = ops.int4_linear(x, wq4.cuda(), sbt.cuda(), grp, bias)
     ref = x.float().cpu() @ ref_w.T
+    if bias is not None:
+        ref = ref + bias.cpu().float()
     assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=2e-3)
 
 
