@@ -412,7 +412,8 @@ def test_per_token_group_quant(kernels):
     # (M, N, K, bias?) — bias exercises the fused splitk==1 epilogue
     # and the reduce-kernel bias path
     (1, 1024, 512, False), (8, 896, 1024, True), (64, 2048, 896, False),
-    (64, 55296 // 9, 1152, True), (200, 512, 1280, True),
+    (64, 55296 // 9, 1152, True), (100, 768, 1024, True),
+    (200, 512, 1280, True),
 ])
 def test_fp8_linear(kernels, case):
     M, N, K, with_bias = case
@@ -530,6 +531,7 @@ def test_rmsnorm_gated(kernels):
 @pytest.mark.parametrize("case", [(8, 512, 1024, False),
                                   (64, 896, 2048, True),
                                   (64, 4096, 1280, True),
+                                  (100, 768, 1536, True),
                                   (200, 1024, 512, True)])
 def test_int4_linear(kernels, case):
     M, N, K, with_bias = case
