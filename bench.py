@@ -46,6 +46,22 @@ SMALL_DEBUG = {  # --model debug: quick bring-up config
     "num_key_value_heads": 2, "vocab_size": 32000,
 }
 
+LLAMA3_8B = {  # BASELINE config: Llama-3 8B PP=1 bf16
+    "architectures": ["LlamaForCausalLM"],
+    "model_type": "llama",
+    "hidden_size": 4096,
+    "intermediate_size": 14336,
+    "num_hidden_layers": 32,
+    "num_attention_heads": 32,
+    "num_key_value_heads": 8,
+    "vocab_size": 128256,
+    "max_position_embeddings": 8192,
+    "rms_norm_eps": 1e-5,
+    "rope_theta": 500000.0,
+    "tie_word_embeddings": False,
+    "eos_token_id": 128001,
+}
+
 MIXTRAL_8X7B = {  # BASELINE config: Mixtral 8x7B PP + EP over xGMI
     "architectures": ["MixtralForCausalLM"],
     "model_type": "mixtral",
@@ -162,6 +178,7 @@ MODELS = {
     "deepseek-v2-lite": ("DeepSeek-V2-Lite", DEEPSEEK_V2_LITE),
     "qwen3-next-9b": ("Qwen3-Next-9B-hybrid", QWEN3_NEXT_9B),
     "debug": ("debug-0.2B", SMALL_DEBUG),
+    "llama3-8b": ("Llama-3-8B", LLAMA3_8B),
     "mixtral-8x7b": ("Mixtral-8x7B", MIXTRAL_8X7B),
     "mixtral-debug": ("mixtral-debug", MIXTRAL_DEBUG),
     "deepseek-v3": ("DeepSeek-V3", DEEPSEEK_V3),
