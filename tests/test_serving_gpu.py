@@ -99,3 +99,48 @@ def test_gpu_vl_mrope_graph_decode(tmp_path):
     got_graph = run(True)
     got_eager = run(False)
     assert got_graph == got_eager
+
+
+@pytest.mark.timeout(600)
+def test_gpu_hybrid_chunked_prefill_matches_full(tmp_path):
+    """Hybrid GDN model on GPU: chunked prefill (small maxp -> the WY
+    chunk kernel runs across SEVERAL engine iterations with state
+    continuation) must match one-shot prefill."""
+    d = tmp_path / "hy"
+    d.mkdir()
+    cfg_json = {
+        "architectures": ["Qwen3_5ForCausalLM"], "model_type": "qwen3_5",
+        "hidden_size": 512, "intermediate_size": 1024,
+        "num_hidden_layers": 4, "full_attention_interval": 2,
+        "num_attention_heads": 8, "num_key_value_heads": 2,
+        "head_dim": 128, "attn_output_gate": True,
+        "partial_rotary_factor": 0.25,
+        "linear_num_value_heads": 4, "linear_num_key_heads": 2,
+        "linear_key_head_dim": 128, "linear_value_head_dim": 128,
+        "linear_conv_kernel_dim": 4,
+        "vocab_size": 32000, "max_position_embeddings": 4096,
+        "rms_norm_eps": 1e-6, "rope_theta": 10000.0, "eos_token_id": 0,
+    }
+    with open(d / "config.json", "w") as f:
+        json.dump(cfg_json, f)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    from gllm_amd.sequence import SamplingParams
+
+    def run(maxp):
+        cfg = EngineConfig(model=str(d), load_format="dummy",
+                           device="cuda", dtype="bfloat16", page_size=16,
+                           max_graph_bs=32, gpu_memory_util=0.2,
+                           maxp=maxp, enable_prefix_caching=False)
+        llm = LLM(config=cfg, num_pages_override=512)
+        outs = llm.generate(
+            [list(range(1, 150)), list(range(5, 80))],
+            SamplingParams(temperature=0.0, max_tokens=8,
+                           ignore_eos=True))
+        del llm
+        torch.cuda.empty_cache()
+        return [list(o.token_ids) for o in outs]
+
+    full = run(4096)
+    chunked = run(64)
+    assert full == chunked
