@@ -555,7 +555,10 @@ void fp8_skinny_gemm(torch::Tensor out, torch::Tensor aq, torch::Tensor ast,
     else if (r == 4) LAUNCH_SK(1, 4);
     else LAUNCH_SK(1, 2);
   }
-  else if (M <= 128) LAUNCH_SK(2, 3);
+  else if (M <= 128) {
+    if (ring_env == 2) LAUNCH_SK(2, 2);
+    else LAUNCH_SK(2, 3);
+  }
   else LAUNCH_SK(4, 2);
 #undef LAUNCH_SK
   HIP_CHECK_KERNEL();
