@@ -61,7 +61,11 @@ class LinearBase(nn.Module):
                       and self.fp8_block == (128, 128)
                       and ops.has_kernels())
             if native:
-                if x.shape[0] <= 256:
+                # measured crossover (profiles/): the weight-streaming
+                # skinny wins only at M <= 64; above that the bf16
+                # shadow + library GEMM is faster (M=256 gate_up:
+                # 329 us skinny vs ~103 us library)
+                if x.shape[0] <= 64:
                     return ops.fp8_linear(x, w, self.weight_scale_inv,
                                           self._bias32())
                 return self._prefill_quant(x, bias)
@@ -91,7 +95,7 @@ class LinearBase(nn.Module):
                     self._i4_canon = (wq4.to(x.device), sb.to(x.device),
                                       sbt.to(x.device), grp)
                 wq4, sb, sbt, grp = self._i4_canon
-                if x.shape[0] <= 256:
+                if x.shape[0] <= 64:
                     return ops.int4_linear(x, wq4, sbt, grp,
                                            self._bias32())
                 return self._prefill_quant(x, bias)
