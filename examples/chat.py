@@ -1,4 +1,9 @@
 """Interactive offline chat (reference: examples/chat.py)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import argparse
 
 from gllm_amd.engine.llm import LLM

@@ -5,6 +5,11 @@
 
 Images are sent as base64 data: URLs (this deployment has no egress,
 so http image URLs are rejected server-side)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 
 import argparse
 import base64

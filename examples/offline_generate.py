@@ -1,4 +1,9 @@
 """Offline batch generation (reference: examples/batch_inference.py)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import argparse
 
 from gllm_amd.engine.llm import LLM

@@ -1,5 +1,10 @@
 """OpenAI-API client against a running api_server
 (reference: examples/client.py / chat_client.py)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import argparse
 import json
 

@@ -183,3 +183,27 @@ def test_qwen3_5_vl_chunked_prefill_matches_full(tmp_path):
     out = _mk_llm35(tmp_path, name="c35", maxp=3).generate(
         [toks], sp, mm_inputs=[mm])[0].token_ids
     assert out == ref
+
+
+def test_skip_visual_lm_node_accepts_embeds(tmp_path):
+    """Encoder-disagg LM node: with skip_visual the tower is not built,
+    and requests carrying ready embeddings still generate (reference
+    lm_server.py --skip-visual)."""
+    d = tmp_path / "sv"
+    d.mkdir()
+    with open(d / "config.json", "w") as f:
+        json.dump(VL3_TINY, f)
+    from gllm_amd.config import EngineConfig
+    from gllm_amd.engine.llm import LLM
+    from gllm_amd.sequence import SamplingParams
+    cfg = EngineConfig(model=str(d), load_format="dummy", device="cpu",
+                       dtype="float32", page_size=4, maxp=64,
+                       skip_visual=True)
+    llm = LLM(config=cfg, num_pages_override=128)
+    assert llm.runner.model.visual is None
+    toks, _ = _mm()
+    emb = torch.randn(4, 64 * 3)  # 4 merged tokens, 1 + 2 deepstack
+    sp = [SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)]
+    out = llm.generate([toks], sp,
+                       mm_inputs=[{"embeds": emb, "grids": [(1, 4, 4)]}])
+    assert len(out[0].token_ids) == 5
