@@ -626,15 +626,16 @@ def test_fused_moe_int4(kernels):
     assert_close_bf16(out, ref, atol=5e-2, rtol=5e-2, frac=3e-3)
 
 
-def test_gdn_chunk_prefill(kernels):
+@pytest.mark.parametrize("heads", [(2, 4), (4, 4)])
+def test_gdn_chunk_prefill(kernels, heads):
     """Fused WY chunk prefill vs the fp32 torch oracle: padded batch,
-    multi-chunk sequences, ragged lengths (inert padding), GQA repeat,
-    and in-place state update."""
-    import copy
+    multi-chunk sequences, ragged lengths (inert padding), GQA repeat
+    (and the G=1 Hk==Hv case), and in-place state update."""
     from gllm_amd import ops
     from gllm_amd.ops import gdn_ref
     torch.manual_seed(11)
-    B, T, Hk, Hv, D = 3, 192, 2, 4, 128
+    Hk, Hv = heads
+    B, T, D = 3, 192, 128
     lens = [192, 130, 64]
     q = torch.randn(B, T, Hk, D, device="cuda") * 0.5
     k = torch.randn(B, T, Hk, D, device="cuda") * 0.5
