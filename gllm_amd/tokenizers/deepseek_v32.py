@@ -9,33 +9,34 @@ surface to it. When absent, callers fall back to
 ``apply_chat_template``.
 """
 
-import importlib.util
 import json
-import os
-from typing import Any, Dict, List, Optional
+import pathlib
+from functools import lru_cache
+from importlib import util as _imp
+from typing import Any, List, Optional
 
-_ENCODER_CACHE: Dict[str, Optional[Any]] = {}
 
-
+@lru_cache(maxsize=16)
 def load_dsv32_encoder(model_path: str) -> Optional[Any]:
-    """Import ``<model_path>/encoding/encoding_dsv32.py`` (cached).
+    """Import ``<model_path>/encoding/encoding_dsv32.py``.
     Returns the module (must expose ``encode_messages``) or None."""
-    if model_path in _ENCODER_CACHE:
-        return _ENCODER_CACHE[model_path]
-    enc_path = os.path.join(model_path, "encoding", "encoding_dsv32.py")
-    module: Optional[Any] = None
-    if os.path.isfile(enc_path):
-        try:
-            spec = importlib.util.spec_from_file_location(
-                "gllm_amd_dsv32_encoding", enc_path)
-            module = importlib.util.module_from_spec(spec)
-            spec.loader.exec_module(module)
-            if not hasattr(module, "encode_messages"):
-                module = None
-        except Exception:
-            module = None
-    _ENCODER_CACHE[model_path] = module
-    return module
+    src = pathlib.Path(model_path) / "encoding" / "encoding_dsv32.py"
+    if not src.is_file():
+        return None
+    try:
+        spec = _imp.spec_from_file_location("gllm_amd_dsv32_encoding",
+                                            str(src))
+        mod = _imp.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+    except Exception:
+        return None
+    return mod if callable(getattr(mod, "encode_messages", None)) else None
+
+
+def _as_plain_dict(m) -> dict:
+    if hasattr(m, "model_dump"):
+        return m.model_dump(mode="json", exclude_none=True)
+    return json.loads(json.dumps(m, default=list))
 
 
 def apply_dsv32_chat_template(encoder: Any, messages: List[dict],
@@ -52,19 +53,13 @@ def apply_dsv32_chat_template(encoder: Any, messages: List[dict],
     """
     thinking = bool(kwargs.get("thinking", False)
                     or kwargs.get("enable_thinking", False))
-    norm = []
-    for m in messages:
-        if hasattr(m, "model_dump"):
-            norm.append(m.model_dump(mode="json", exclude_none=True))
-        else:
-            norm.append(json.loads(json.dumps(m, default=list)))
-    messages = norm
+    convo = [_as_plain_dict(m) for m in messages]
     if tools:
-        messages.insert(0, {"role": "system", "tools": tools})
-    drop_thinking = bool(messages) and messages[-1].get("role") == "user"
+        convo = [{"role": "system", "tools": tools}] + convo
+    fresh_user_turn = bool(convo) and convo[-1].get("role") == "user"
     prompt = encoder.encode_messages(
-        messages, thinking_mode="thinking" if thinking else "chat",
-        drop_thinking=drop_thinking)
+        convo, thinking_mode="thinking" if thinking else "chat",
+        drop_thinking=fresh_user_turn)
     if not tokenize:
         return prompt
     return tokenizer.encode(prompt, add_special_tokens=False)
