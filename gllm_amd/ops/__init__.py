@@ -75,6 +75,15 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     return torch_ref.silu_and_mul(x)
 
 
+def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        d = x.shape[-1] // 2
+        out = torch.empty(x.shape[:-1] + (d,), dtype=x.dtype, device=x.device)
+        _gpu_kernels().gelu_and_mul(out, x)
+        return out
+    return torch_ref.gelu_and_mul(x)
+
+
 # --------------------------------------------------------------- rope
 def rotary_embedding(positions: torch.Tensor, q: torch.Tensor,
                      k: torch.Tensor, head_dim: int,

@@ -6,6 +6,7 @@ void rmsnorm(torch::Tensor out, torch::Tensor in, torch::Tensor weight,
 void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
                        torch::Tensor weight, double eps);
 void silu_and_mul(torch::Tensor out, torch::Tensor x);
+void gelu_and_mul(torch::Tensor out, torch::Tensor x);
 void rotary_embedding(torch::Tensor positions, torch::Tensor q,
                       torch::Tensor k, long head_dim,
                       torch::Tensor cos_sin_cache, bool is_neox);
@@ -97,6 +98,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
         "residual += x; x = rmsnorm(residual)");
   m.def("silu_and_mul", &silu_and_mul, "silu(x[:d]) * x[d:]");
+  m.def("gelu_and_mul", &gelu_and_mul, "gelu(x[:d]) * x[d:]");
   m.def("rotary_embedding", &rotary_embedding, "in-place RoPE on q,k");
   m.def("reshape_and_cache", &reshape_and_cache,
         "scatter K/V into paged cache");

@@ -32,6 +32,32 @@ __global__ void silu_and_mul_kernel(T *__restrict__ out,
   }
 }
 
+// gelu(x[:, :d]) * x[:, d:] — erf-based GELU (reference gelu_and_mul,
+// the MoE activation="gelu" path in fused_moe.py:943).
+template <typename T>
+__global__ void gelu_and_mul_kernel(T *__restrict__ out,
+                                    const T *__restrict__ x, long rows,
+                                    int d) {
+  const int nvec = d / 8;
+  const long total = rows * nvec;
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const long row = idx / nvec;
+    const int i = idx % nvec;
+    shortx8 a = reinterpret_cast<const shortx8 *>(x + row * 2 * d)[i];
+    shortx8 b = reinterpret_cast<const shortx8 *>(x + row * 2 * d + d)[i];
+    float va[8], vb[8];
+    unpack8<T>(a, va);
+    unpack8<T>(b, vb);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float g = 0.5f * va[j] * (1.f + erff(va[j] * 0.70710678f));
+      va[j] = g * vb[j];
+    }
+    reinterpret_cast<shortx8 *>(out + row * d)[i] = pack8<T>(va);
+  }
+}
+
 // In-place neox / interleaved RoPE on q [T, Hq*D] and k [T, Hk*D].
 // cos_sin_cache: [max_pos, rot_dim] fp32, [cos | sin] halves (host-built
 // — no device trig, guide App. B). One block per token.
@@ -130,6 +156,30 @@ void silu_and_mul(torch::Tensor out, torch::Tensor x) {
                        x.data_ptr<float>(), rows, d);
   } else {
     TORCH_CHECK(false, "silu_and_mul: unsupported dtype");
+  }
+  HIP_CHECK_KERNEL();
+}
+
+void gelu_and_mul(torch::Tensor out, torch::Tensor x) {
+  const int d = x.size(-1) / 2;
+  const long rows = x.numel() / (2 * d);
+  TORCH_CHECK(d % 8 == 0);
+  TORCH_CHECK(x.is_contiguous() && out.is_contiguous());
+  const long total = rows * (d / 8);
+  const int block = 256;
+  const long grid = std::min<long>((total + block - 1) / block, 2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((gelu_and_mul_kernel<__hip_bfloat16>), dim3(grid),
+                       dim3(block), 0, stream,
+                       (__hip_bfloat16 *)out.data_ptr(),
+                       (const __hip_bfloat16 *)x.data_ptr(), rows, d);
+  } else if (x.scalar_type() == at::kHalf) {
+    hipLaunchKernelGGL((gelu_and_mul_kernel<__half>), dim3(grid), dim3(block),
+                       0, stream, (__half *)out.data_ptr(),
+                       (const __half *)x.data_ptr(), rows, d);
+  } else {
+    TORCH_CHECK(false, "gelu_and_mul: unsupported dtype");
   }
   HIP_CHECK_KERNEL();
 }

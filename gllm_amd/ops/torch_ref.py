@@ -34,6 +34,14 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
             * x[..., d:].float()).to(x.dtype)
 
 
+
+def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    """gelu(x[:, :d]) * x[:, d:] (erf-based, reference gelu_and_mul)."""
+    d = x.shape[-1] // 2
+    a = x[..., :d].float()
+    return (torch.nn.functional.gelu(a) *
+            x[..., d:].float()).to(x.dtype)
+
 def rotary_embedding(positions: torch.Tensor, q: torch.Tensor,
                      k: torch.Tensor, head_dim: int,
                      cos_sin_cache: torch.Tensor, is_neox: bool = True):

@@ -662,3 +662,11 @@ def test_gdn_chunk_prefill(kernels, heads):
         assert_close_bf16(out[b, :n].float(), ref[b, :n].float(),
                           atol=2e-2, rtol=2e-2, frac=1e-3)
     assert torch.allclose(st_hip, st_ref, atol=2e-2, rtol=2e-2)
+
+
+def test_gelu_and_mul(kernels):
+    torch.manual_seed(6)
+    x = torch.randn(97, 2 * 1024, dtype=torch.bfloat16, device="cuda")
+    out = kernels.gelu_and_mul(x)
+    ref = R.gelu_and_mul(x.float().cpu())
+    assert_close_bf16(out, ref)
